@@ -1,0 +1,248 @@
+"""simple_tensorflow_amd — an MI355X-native dataflow DL framework with the
+capability surface of TF 1.0.
+
+Use `import simple_tensorflow_amd as tf` — the module exposes the familiar
+tf.* names (Session, Graph, Variable, train.*, nn.*, ...).
+"""
+import os as _os
+import sys as _sys
+
+# The pybind extension must be importable as simple_tensorflow_amd._core.
+from simple_tensorflow_amd import _core  # noqa: F401
+
+from simple_tensorflow_amd.python.framework import dtypes as _dtypes
+from simple_tensorflow_amd.python.framework import ops as _ops
+from simple_tensorflow_amd.python.client import session as _session
+from simple_tensorflow_amd.python.ops import (  # noqa: F401
+    array_ops as _array_ops,
+    clip_ops as _clip_ops,
+    control_flow_ops as _control_flow_ops,
+    gradients_impl as _gradients_impl,
+    init_ops as _init_ops,
+    math_ops as _math_ops,
+    nn_ops as _nn,
+    random_ops as _random_ops,
+    state_ops as _state_ops,
+    variables as _variables,
+)
+from simple_tensorflow_amd.python.training import optimizer as _optimizer
+from simple_tensorflow_amd.python.training import training_util as _training_util
+
+# ---- dtypes ----
+float32 = _dtypes.float32
+float64 = _dtypes.float64
+double = _dtypes.float64
+int32 = _dtypes.int32
+int64 = _dtypes.int64
+int16 = _dtypes.int16
+int8 = _dtypes.int8
+uint8 = _dtypes.uint8
+uint16 = _dtypes.uint16
+bool = _dtypes.bool  # noqa: A001
+string = _dtypes.string
+bfloat16 = _dtypes.bfloat16
+float16 = _dtypes.float16
+half = _dtypes.float16
+DType = _dtypes.DType
+as_dtype = _dtypes.as_dtype
+
+# ---- framework ----
+Graph = _ops.Graph
+Operation = _ops.Operation
+Tensor = _ops.Tensor
+TensorShape = _ops.TensorShape
+GraphKeys = _ops.GraphKeys
+get_default_graph = _ops.get_default_graph
+reset_default_graph = _ops.reset_default_graph
+device = _ops.device
+name_scope = _ops.name_scope
+control_dependencies = _ops.control_dependencies
+colocate_with = _ops.colocate_with
+convert_to_tensor = _ops.convert_to_tensor
+constant = _ops.constant
+RegisterGradient = _ops.RegisterGradient
+NoGradient = _ops.NoGradient
+NotDifferentiable = _ops.NotDifferentiable
+add_to_collection = lambda name, value: get_default_graph().add_to_collection(name, value)  # noqa: E731
+get_collection = lambda name, scope=None: get_default_graph().get_collection(name, scope)  # noqa: E731
+get_collection_ref = lambda name: get_default_graph().get_collection_ref(name)  # noqa: E731
+
+# ---- session ----
+Session = _session.Session
+InteractiveSession = _session.InteractiveSession
+get_default_session = _session.get_default_session
+
+# ---- variables ----
+Variable = _variables.Variable
+global_variables = _variables.global_variables
+all_variables = _variables.all_variables
+trainable_variables = _variables.trainable_variables
+local_variables = _variables.local_variables
+moving_average_variables = _variables.moving_average_variables
+global_variables_initializer = _variables.global_variables_initializer
+initialize_all_variables = _variables.initialize_all_variables
+local_variables_initializer = _variables.local_variables_initializer
+variables_initializer = _variables.variables_initializer
+is_variable_initialized = _variables.is_variable_initialized
+assign = _state_ops.assign
+assign_add = _state_ops.assign_add
+assign_sub = _state_ops.assign_sub
+scatter_add = _state_ops.scatter_add
+scatter_sub = _state_ops.scatter_sub
+
+# ---- array ops ----
+placeholder = _array_ops.placeholder
+identity = _array_ops.identity
+stop_gradient = _array_ops.stop_gradient
+shape = _array_ops.shape
+shape_n = _array_ops.shape_n
+rank = _array_ops.rank
+size = _array_ops.size
+reshape = _array_ops.reshape
+expand_dims = _array_ops.expand_dims
+squeeze = _array_ops.squeeze
+zeros = _array_ops.zeros
+ones = _array_ops.ones
+fill = _array_ops.fill
+zeros_like = _array_ops.zeros_like
+ones_like = _array_ops.ones_like
+concat = _array_ops.concat
+split = _array_ops.split
+stack = _array_ops.stack
+pack = _array_ops.pack
+unstack = _array_ops.unstack
+unpack = _array_ops.unpack
+slice = _array_ops.slice  # noqa: A001
+pad = _array_ops.pad
+transpose = _array_ops.transpose
+gather = _array_ops.gather
+tile = _array_ops.tile
+one_hot = _array_ops.one_hot
+where = _array_ops.where
+check_numerics = _array_ops.check_numerics
+unsorted_segment_sum = _array_ops.unsorted_segment_sum
+
+# ---- math ----
+add = _math_ops.add
+subtract = _math_ops.subtract
+sub = _math_ops.sub
+multiply = _math_ops.multiply
+mul = _math_ops.mul
+divide = _math_ops.divide
+div = _math_ops.div
+truediv = _math_ops.truediv
+floordiv = _math_ops.floordiv
+mod = _math_ops.mod
+pow = _math_ops.pow  # noqa: A001
+maximum = _math_ops.maximum
+minimum = _math_ops.minimum
+squared_difference = _math_ops.squared_difference
+less = _math_ops.less
+less_equal = _math_ops.less_equal
+greater = _math_ops.greater
+greater_equal = _math_ops.greater_equal
+equal = _math_ops.equal
+not_equal = _math_ops.not_equal
+logical_and = _math_ops.logical_and
+logical_or = _math_ops.logical_or
+logical_not = _math_ops.logical_not
+negative = _math_ops.negative
+neg = _math_ops.neg
+abs = _math_ops.abs  # noqa: A001
+sign = _math_ops.sign
+square = _math_ops.square
+sqrt = _math_ops.sqrt
+rsqrt = _math_ops.rsqrt
+exp = _math_ops.exp
+log = _math_ops.log
+log1p = _math_ops.log1p
+tanh = _math_ops.tanh
+sigmoid = _math_ops.sigmoid
+sin = _math_ops.sin
+cos = _math_ops.cos
+floor = _math_ops.floor
+ceil = _math_ops.ceil
+round = _math_ops.round  # noqa: A001
+reciprocal = _math_ops.reciprocal
+is_nan = _math_ops.is_nan
+is_inf = _math_ops.is_inf
+is_finite = _math_ops.is_finite
+cast = _math_ops.cast
+to_float = _math_ops.to_float
+to_double = _math_ops.to_double
+to_int32 = _math_ops.to_int32
+to_int64 = _math_ops.to_int64
+matmul = _math_ops.matmul
+batch_matmul = _math_ops.batch_matmul
+add_n = _math_ops.add_n
+reduce_sum = _math_ops.reduce_sum
+reduce_mean = _math_ops.reduce_mean
+reduce_max = _math_ops.reduce_max
+reduce_min = _math_ops.reduce_min
+reduce_prod = _math_ops.reduce_prod
+reduce_all = _math_ops.reduce_all
+reduce_any = _math_ops.reduce_any
+argmax = _math_ops.argmax
+argmin = _math_ops.argmin
+select = _math_ops.select
+range = _math_ops.range  # noqa: A001
+cumsum = _math_ops.cumsum
+global_norm = _clip_ops.global_norm
+clip_by_value = _clip_ops.clip_by_value
+clip_by_norm = _clip_ops.clip_by_norm
+clip_by_global_norm = _clip_ops.clip_by_global_norm
+
+# ---- random ----
+random_uniform = _random_ops.random_uniform
+random_normal = _random_ops.random_normal
+truncated_normal = _random_ops.truncated_normal
+set_random_seed = _random_ops.set_random_seed
+
+# ---- control flow ----
+group = _control_flow_ops.group
+no_op = _control_flow_ops.no_op
+cond = _control_flow_ops.cond
+while_loop = _control_flow_ops.while_loop
+Assert = _control_flow_ops.Assert
+
+# ---- gradients ----
+gradients = _gradients_impl.gradients
+
+# ---- initializers ----
+zeros_initializer = _init_ops.zeros_initializer
+ones_initializer = _init_ops.ones_initializer
+constant_initializer = _init_ops.constant_initializer
+random_uniform_initializer = _init_ops.random_uniform_initializer
+random_normal_initializer = _init_ops.random_normal_initializer
+truncated_normal_initializer = _init_ops.truncated_normal_initializer
+
+
+# ---- nn / train sub-namespaces ----
+class _NnModule(object):
+    pass
+
+
+nn = _nn  # module with conv2d/relu/softmax/...
+
+
+class _TrainModule(object):
+    Optimizer = _optimizer.Optimizer
+    GradientDescentOptimizer = _optimizer.GradientDescentOptimizer
+    MomentumOptimizer = _optimizer.MomentumOptimizer
+    AdamOptimizer = _optimizer.AdamOptimizer
+    RMSPropOptimizer = _optimizer.RMSPropOptimizer
+    AdagradOptimizer = _optimizer.AdagradOptimizer
+    AdadeltaOptimizer = _optimizer.AdadeltaOptimizer
+    create_global_step = staticmethod(_training_util.create_global_step)
+    get_global_step = staticmethod(_training_util.get_global_step)
+    get_or_create_global_step = staticmethod(
+        _training_util.get_or_create_global_step)
+    exponential_decay = staticmethod(_training_util.exponential_decay)
+    polynomial_decay = staticmethod(_training_util.polynomial_decay)
+    piecewise_constant = staticmethod(_training_util.piecewise_constant)
+    ExponentialMovingAverage = _training_util.ExponentialMovingAverage
+
+
+train = _TrainModule()
+
+__version__ = '0.1.0'

@@ -1,0 +1,223 @@
+// Op registry definitions — the tf.* op surface (TF 1.0-compatible NodeDef
+// attrs; reference: tensorflow/core/ops/*.cc). Grouped as in the reference's
+// ops/ directory. Shape functions live in the Python layer
+// (python/framework/shapes.py).
+#include "framework/op.h"
+
+namespace stf {
+
+#define NUMTYPES "{float, double, int32, int64, bfloat16, half, uint8, int8}"
+#define REALTYPES "{float, double, int32, int64, bfloat16, half}"
+#define FLOATTYPES "{float, double, bfloat16, half}"
+
+// ----------------------------- array_ops ----------------------------------
+REGISTER_OP("Const").Output("output: dtype").Attr("dtype: type").Attr("value: tensor");
+REGISTER_OP("Placeholder").Output("output: dtype").Attr("dtype: type").Attr("shape: shape = []");
+REGISTER_OP("PlaceholderWithDefault").Input("input: dtype").Output("output: dtype").Attr("dtype: type").Attr("shape: shape");
+REGISTER_OP("Identity").Input("input: T").Output("output: T").Attr("T: type");
+REGISTER_OP("StopGradient").Input("input: T").Output("output: T").Attr("T: type");
+REGISTER_OP("PreventGradient").Input("input: T").Output("output: T").Attr("T: type").Attr("message: string = ''");
+REGISTER_OP("Shape").Input("input: T").Output("output: out_type").Attr("T: type").Attr("out_type: {int32, int64} = int32");
+REGISTER_OP("ShapeN").Input("input: N * T").Output("output: N * out_type").Attr("N: int >= 1").Attr("T: type").Attr("out_type: {int32, int64} = int32");
+REGISTER_OP("Rank").Input("input: T").Output("output: int32").Attr("T: type");
+REGISTER_OP("Size").Input("input: T").Output("output: out_type").Attr("T: type").Attr("out_type: {int32, int64} = int32");
+REGISTER_OP("Reshape").Input("tensor: T").Input("shape: Tshape").Output("output: T").Attr("T: type").Attr("Tshape: {int32, int64} = int32");
+REGISTER_OP("ExpandDims").Input("input: T").Input("dim: Tdim").Output("output: T").Attr("T: type").Attr("Tdim: {int32, int64} = int32");
+REGISTER_OP("Squeeze").Input("input: T").Output("output: T").Attr("T: type").Attr("squeeze_dims: list(int) = []");
+REGISTER_OP("Fill").Input("dims: int32").Input("value: T").Output("output: T").Attr("T: type");
+REGISTER_OP("ZerosLike").Input("x: T").Output("y: T").Attr("T: type");
+REGISTER_OP("OnesLike").Input("x: T").Output("y: T").Attr("T: type");
+REGISTER_OP("Cast").Input("x: SrcT").Output("y: DstT").Attr("SrcT: type").Attr("DstT: type");
+REGISTER_OP("Pack").Input("values: N * T").Output("output: T").Attr("N: int >= 1").Attr("T: type").Attr("axis: int = 0");
+REGISTER_OP("Unpack").Input("value: T").Output("output: num * T").Attr("num: int >= 0").Attr("T: type").Attr("axis: int = 0");
+REGISTER_OP("ConcatV2").Input("values: N * T").Input("axis: Tidx").Output("output: T").Attr("N: int >= 2").Attr("T: type").Attr("Tidx: {int32, int64} = int32");
+REGISTER_OP("Concat").Input("concat_dim: int32").Input("values: N * T").Output("output: T").Attr("N: int >= 2").Attr("T: type");
+REGISTER_OP("ConcatOffset").Input("concat_dim: int32").Input("shape: N * int32").Output("offset: N * int32").Attr("N: int >= 2");
+REGISTER_OP("Split").Input("split_dim: int32").Input("value: T").Output("output: num_split * T").Attr("num_split: int >= 1").Attr("T: type");
+REGISTER_OP("SplitV").Input("value: T").Input("size_splits: Tlen").Input("split_dim: int32").Output("output: num_split * T").Attr("num_split: int >= 1").Attr("T: type").Attr("Tlen: {int32, int64} = int64");
+REGISTER_OP("Slice").Input("input: T").Input("begin: Index").Input("size: Index").Output("output: T").Attr("T: type").Attr("Index: {int32, int64}");
+REGISTER_OP("StridedSlice").Input("input: T").Input("begin: Index").Input("end: Index").Input("strides: Index").Output("output: T").Attr("T: type").Attr("Index: {int32, int64}").Attr("begin_mask: int = 0").Attr("end_mask: int = 0").Attr("ellipsis_mask: int = 0").Attr("new_axis_mask: int = 0").Attr("shrink_axis_mask: int = 0");
+REGISTER_OP("StridedSliceGrad").Input("shape: Index").Input("begin: Index").Input("end: Index").Input("strides: Index").Input("dy: T").Output("output: T").Attr("T: type").Attr("Index: {int32, int64}").Attr("begin_mask: int = 0").Attr("end_mask: int = 0").Attr("ellipsis_mask: int = 0").Attr("new_axis_mask: int = 0").Attr("shrink_axis_mask: int = 0");
+REGISTER_OP("Pad").Input("input: T").Input("paddings: Tpaddings").Output("output: T").Attr("T: type").Attr("Tpaddings: {int32, int64} = int32");
+REGISTER_OP("Transpose").Input("x: T").Input("perm: Tperm").Output("y: T").Attr("T: type").Attr("Tperm: {int32, int64} = int32");
+REGISTER_OP("Gather").Input("params: Tparams").Input("indices: Tindices").Output("output: Tparams").Attr("validate_indices: bool = true").Attr("Tparams: type").Attr("Tindices: {int32, int64}");
+REGISTER_OP("GatherV2").Input("params: Tparams").Input("indices: Tindices").Input("axis: Taxis").Output("output: Tparams").Attr("Tparams: type").Attr("Tindices: {int32, int64}").Attr("Taxis: {int32, int64} = int32");
+REGISTER_OP("Tile").Input("input: T").Input("multiples: Tmultiples").Output("output: T").Attr("T: type").Attr("Tmultiples: {int32, int64} = int32");
+REGISTER_OP("InvertPermutation").Input("x: T").Output("y: T").Attr("T: {int32, int64} = int32");
+REGISTER_OP("Reverse").Input("tensor: T").Input("dims: bool").Output("output: T").Attr("T: type");
+REGISTER_OP("BroadcastGradientArgs").Input("s0: T").Input("s1: T").Output("r0: T").Output("r1: T").Attr("T: {int32, int64} = int32");
+REGISTER_OP("OneHot").Input("indices: TI").Input("depth: int32").Input("on_value: T").Input("off_value: T").Output("output: T").Attr("axis: int = -1").Attr("T: type").Attr("TI: {uint8, int32, int64} = int64");
+REGISTER_OP("Range").Input("start: Tidx").Input("limit: Tidx").Input("delta: Tidx").Output("output: Tidx").Attr("Tidx: {float, double, int32, int64} = int32");
+REGISTER_OP("LinSpace").Input("start: T").Input("stop: T").Input("num: Tidx").Output("output: T").Attr("T: {float, double}").Attr("Tidx: {int32, int64} = int32");
+REGISTER_OP("CheckNumerics").Input("tensor: T").Output("output: T").Attr("T: " FLOATTYPES).Attr("message: string");
+REGISTER_OP("UnsortedSegmentSum").Input("data: T").Input("segment_ids: Tindices").Input("num_segments: int32").Output("output: T").Attr("T: " NUMTYPES).Attr("Tindices: {int32, int64}");
+REGISTER_OP("DynamicStitch").Input("indices: N * int32").Input("data: N * T").Output("merged: T").Attr("N: int >= 1").Attr("T: type");
+
+// ----------------------------- math_ops -----------------------------------
+#define BINARY_OP(NAME) REGISTER_OP(NAME).Input("x: T").Input("y: T").Output("z: T")
+BINARY_OP("Add").Attr("T: " NUMTYPES);
+BINARY_OP("Sub").Attr("T: " NUMTYPES);
+BINARY_OP("Mul").Attr("T: " NUMTYPES);
+BINARY_OP("Div").Attr("T: " NUMTYPES);
+BINARY_OP("RealDiv").Attr("T: " NUMTYPES);
+BINARY_OP("FloorDiv").Attr("T: " NUMTYPES);
+BINARY_OP("FloorMod").Attr("T: " NUMTYPES);
+BINARY_OP("Pow").Attr("T: " NUMTYPES);
+BINARY_OP("Maximum").Attr("T: " REALTYPES);
+BINARY_OP("Minimum").Attr("T: " REALTYPES);
+BINARY_OP("SquaredDifference").Attr("T: " NUMTYPES);
+#undef BINARY_OP
+#define CMP_OP(NAME) REGISTER_OP(NAME).Input("x: T").Input("y: T").Output("z: bool")
+CMP_OP("Less").Attr("T: " REALTYPES);
+CMP_OP("LessEqual").Attr("T: " REALTYPES);
+CMP_OP("Greater").Attr("T: " REALTYPES);
+CMP_OP("GreaterEqual").Attr("T: " REALTYPES);
+CMP_OP("Equal").Attr("T: " NUMTYPES);
+CMP_OP("NotEqual").Attr("T: " NUMTYPES);
+#undef CMP_OP
+REGISTER_OP("LogicalAnd").Input("x: bool").Input("y: bool").Output("z: bool");
+REGISTER_OP("LogicalOr").Input("x: bool").Input("y: bool").Output("z: bool");
+REGISTER_OP("LogicalNot").Input("x: bool").Output("y: bool");
+#define UNARY_OP(NAME) REGISTER_OP(NAME).Input("x: T").Output("y: T")
+UNARY_OP("Neg").Attr("T: " NUMTYPES);
+UNARY_OP("Abs").Attr("T: " REALTYPES);
+UNARY_OP("Sign").Attr("T: " REALTYPES);
+UNARY_OP("Square").Attr("T: " NUMTYPES);
+UNARY_OP("Sqrt").Attr("T: " FLOATTYPES);
+UNARY_OP("Rsqrt").Attr("T: " FLOATTYPES);
+UNARY_OP("Exp").Attr("T: " FLOATTYPES);
+UNARY_OP("Log").Attr("T: " FLOATTYPES);
+UNARY_OP("Log1p").Attr("T: " FLOATTYPES);
+UNARY_OP("Tanh").Attr("T: " FLOATTYPES);
+UNARY_OP("Sigmoid").Attr("T: " FLOATTYPES);
+UNARY_OP("Sin").Attr("T: " FLOATTYPES);
+UNARY_OP("Cos").Attr("T: " FLOATTYPES);
+UNARY_OP("Floor").Attr("T: " FLOATTYPES);
+UNARY_OP("Ceil").Attr("T: " FLOATTYPES);
+UNARY_OP("Round").Attr("T: " FLOATTYPES);
+UNARY_OP("Reciprocal").Attr("T: " NUMTYPES);
+#undef UNARY_OP
+REGISTER_OP("IsNan").Input("x: T").Output("y: bool").Attr("T: " FLOATTYPES);
+REGISTER_OP("IsInf").Input("x: T").Output("y: bool").Attr("T: " FLOATTYPES);
+REGISTER_OP("IsFinite").Input("x: T").Output("y: bool").Attr("T: " FLOATTYPES);
+#define GRAD2_OP(NAME) REGISTER_OP(NAME).Input("y: T").Input("dy: T").Output("z: T").Attr("T: " FLOATTYPES)
+GRAD2_OP("SigmoidGrad");
+GRAD2_OP("TanhGrad");
+GRAD2_OP("RsqrtGrad");
+GRAD2_OP("SqrtGrad");
+GRAD2_OP("ReciprocalGrad");
+#undef GRAD2_OP
+REGISTER_OP("AddN").Input("inputs: N * T").Output("sum: T").Attr("N: int >= 1").Attr("T: " NUMTYPES);
+REGISTER_OP("MatMul").Input("a: T").Input("b: T").Output("product: T").Attr("transpose_a: bool = false").Attr("transpose_b: bool = false").Attr("T: " FLOATTYPES);
+REGISTER_OP("BatchMatMul").Input("x: T").Input("y: T").Output("output: T").Attr("T: " FLOATTYPES).Attr("adj_x: bool = false").Attr("adj_y: bool = false");
+#define REDUCE_OP(NAME) REGISTER_OP(NAME).Input("input: T").Input("reduction_indices: Tidx").Output("output: T").Attr("keep_dims: bool = false").Attr("T: " NUMTYPES).Attr("Tidx: {int32, int64} = int32")
+REDUCE_OP("Sum");
+REDUCE_OP("Mean");
+REDUCE_OP("Prod");
+REDUCE_OP("Max");
+REDUCE_OP("Min");
+#undef REDUCE_OP
+REGISTER_OP("All").Input("input: bool").Input("reduction_indices: Tidx").Output("output: bool").Attr("keep_dims: bool = false").Attr("Tidx: {int32, int64} = int32");
+REGISTER_OP("Any").Input("input: bool").Input("reduction_indices: Tidx").Output("output: bool").Attr("keep_dims: bool = false").Attr("Tidx: {int32, int64} = int32");
+REGISTER_OP("ArgMax").Input("input: T").Input("dimension: Tidx").Output("output: output_type").Attr("T: " NUMTYPES).Attr("Tidx: {int32, int64} = int32").Attr("output_type: {int32, int64} = int64");
+REGISTER_OP("ArgMin").Input("input: T").Input("dimension: Tidx").Output("output: output_type").Attr("T: " NUMTYPES).Attr("Tidx: {int32, int64} = int32").Attr("output_type: {int32, int64} = int64");
+REGISTER_OP("Select").Input("condition: bool").Input("t: T").Input("e: T").Output("output: T").Attr("T: type");
+REGISTER_OP("Cumsum").Input("x: T").Input("axis: Tidx").Output("out: T").Attr("exclusive: bool = false").Attr("reverse: bool = false").Attr("T: " NUMTYPES).Attr("Tidx: {int32, int64} = int32");
+
+// ------------------------------- nn_ops ------------------------------------
+REGISTER_OP("Conv2D").Input("input: T").Input("filter: T").Output("output: T").Attr("T: " FLOATTYPES).Attr("strides: list(int)").Attr("use_cudnn_on_gpu: bool = true").Attr("padding: string").Attr("data_format: string = 'NHWC'");
+REGISTER_OP("Conv2DBackpropInput").Input("input_sizes: int32").Input("filter: T").Input("out_backprop: T").Output("output: T").Attr("T: " FLOATTYPES).Attr("strides: list(int)").Attr("use_cudnn_on_gpu: bool = true").Attr("padding: string").Attr("data_format: string = 'NHWC'");
+REGISTER_OP("Conv2DBackpropFilter").Input("input: T").Input("filter_sizes: int32").Input("out_backprop: T").Output("output: T").Attr("T: " FLOATTYPES).Attr("strides: list(int)").Attr("use_cudnn_on_gpu: bool = true").Attr("padding: string").Attr("data_format: string = 'NHWC'");
+REGISTER_OP("BiasAdd").Input("value: T").Input("bias: T").Output("output: T").Attr("T: " NUMTYPES).Attr("data_format: string = 'NHWC'");
+REGISTER_OP("BiasAddGrad").Input("out_backprop: T").Output("output: T").Attr("T: " NUMTYPES).Attr("data_format: string = 'NHWC'");
+REGISTER_OP("Relu").Input("features: T").Output("activations: T").Attr("T: " REALTYPES);
+REGISTER_OP("ReluGrad").Input("gradients: T").Input("features: T").Output("backprops: T").Attr("T: " REALTYPES);
+REGISTER_OP("Relu6").Input("features: T").Output("activations: T").Attr("T: " REALTYPES);
+REGISTER_OP("Relu6Grad").Input("gradients: T").Input("features: T").Output("backprops: T").Attr("T: " REALTYPES);
+REGISTER_OP("Elu").Input("features: T").Output("activations: T").Attr("T: " FLOATTYPES);
+REGISTER_OP("EluGrad").Input("gradients: T").Input("outputs: T").Output("backprops: T").Attr("T: " FLOATTYPES);
+REGISTER_OP("Softplus").Input("features: T").Output("activations: T").Attr("T: " FLOATTYPES);
+REGISTER_OP("SoftplusGrad").Input("gradients: T").Input("features: T").Output("backprops: T").Attr("T: " FLOATTYPES);
+REGISTER_OP("Softmax").Input("logits: T").Output("softmax: T").Attr("T: " FLOATTYPES);
+REGISTER_OP("LogSoftmax").Input("logits: T").Output("logsoftmax: T").Attr("T: " FLOATTYPES);
+REGISTER_OP("SoftmaxCrossEntropyWithLogits").Input("features: T").Input("labels: T").Output("loss: T").Output("backprop: T").Attr("T: " FLOATTYPES);
+REGISTER_OP("SparseSoftmaxCrossEntropyWithLogits").Input("features: T").Input("labels: Tlabels").Output("loss: T").Output("backprop: T").Attr("T: " FLOATTYPES).Attr("Tlabels: {int32, int64} = int64");
+REGISTER_OP("MaxPool").Input("input: T").Output("output: T").Attr("T: " FLOATTYPES).Attr("ksize: list(int)").Attr("strides: list(int)").Attr("padding: string").Attr("data_format: string = 'NHWC'");
+REGISTER_OP("MaxPoolGrad").Input("orig_input: T").Input("orig_output: T").Input("grad: T").Output("output: T").Attr("T: " FLOATTYPES).Attr("ksize: list(int)").Attr("strides: list(int)").Attr("padding: string").Attr("data_format: string = 'NHWC'");
+REGISTER_OP("AvgPool").Input("value: T").Output("output: T").Attr("T: " FLOATTYPES).Attr("ksize: list(int)").Attr("strides: list(int)").Attr("padding: string").Attr("data_format: string = 'NHWC'");
+REGISTER_OP("AvgPoolGrad").Input("orig_input_shape: int32").Input("grad: T").Output("output: T").Attr("T: " FLOATTYPES).Attr("ksize: list(int)").Attr("strides: list(int)").Attr("padding: string").Attr("data_format: string = 'NHWC'");
+REGISTER_OP("FusedBatchNorm").Input("x: T").Input("scale: T").Input("offset: T").Input("mean: T").Input("variance: T").Output("y: T").Output("batch_mean: T").Output("batch_variance: T").Output("reserve_space_1: T").Output("reserve_space_2: T").Attr("T: {float}").Attr("epsilon: float = 0.0001").Attr("data_format: string = 'NHWC'").Attr("is_training: bool = true");
+REGISTER_OP("FusedBatchNormGrad").Input("y_backprop: T").Input("x: T").Input("scale: T").Input("reserve_space_1: T").Input("reserve_space_2: T").Output("x_backprop: T").Output("scale_backprop: T").Output("offset_backprop: T").Output("reserve_space_3: T").Output("reserve_space_4: T").Attr("T: {float}").Attr("epsilon: float = 0.0001").Attr("data_format: string = 'NHWC'").Attr("is_training: bool = true");
+REGISTER_OP("BatchNormMi").Input("x: T").Input("scale: float").Input("offset: float").Output("y: T").Output("batch_mean: float").Output("batch_variance: float").Attr("T: {float, bfloat16}").Attr("epsilon: float = 0.0001");
+REGISTER_OP("L2Loss").Input("t: T").Output("output: T").Attr("T: " FLOATTYPES);
+REGISTER_OP("LRN").Input("input: T").Output("output: T").Attr("depth_radius: int = 5").Attr("bias: float = 1.0").Attr("alpha: float = 1.0").Attr("beta: float = 0.5").Attr("T: {float}");
+REGISTER_OP("InTopK").Input("predictions: float").Input("targets: T").Output("precision: bool").Attr("k: int").Attr("T: {int32, int64} = int32");
+
+// --------------------------- state / training -------------------------------
+REGISTER_OP("VariableV2").Output("ref: Ref(dtype)").Attr("shape: shape").Attr("dtype: type").Attr("container: string = ''").Attr("shared_name: string = ''").SetIsStateful();
+REGISTER_OP("Variable").Output("ref: Ref(dtype)").Attr("shape: shape").Attr("dtype: type").Attr("container: string = ''").Attr("shared_name: string = ''").SetIsStateful();
+REGISTER_OP("IsVariableInitialized").Input("ref: Ref(dtype)").Output("is_initialized: bool").Attr("dtype: type").SetAllowsUninitializedInput();
+REGISTER_OP("Assign").Input("ref: Ref(T)").Input("value: T").Output("output_ref: Ref(T)").Attr("T: type").Attr("validate_shape: bool = true").Attr("use_locking: bool = true").SetAllowsUninitializedInput();
+REGISTER_OP("AssignAdd").Input("ref: Ref(T)").Input("value: T").Output("output_ref: Ref(T)").Attr("T: " NUMTYPES).Attr("use_locking: bool = false");
+REGISTER_OP("AssignSub").Input("ref: Ref(T)").Input("value: T").Output("output_ref: Ref(T)").Attr("T: " NUMTYPES).Attr("use_locking: bool = false");
+REGISTER_OP("ScatterSub").Input("ref: Ref(T)").Input("indices: Tindices").Input("updates: T").Output("output_ref: Ref(T)").Attr("T: " NUMTYPES).Attr("Tindices: {int32, int64}").Attr("use_locking: bool = false");
+REGISTER_OP("ScatterAdd").Input("ref: Ref(T)").Input("indices: Tindices").Input("updates: T").Output("output_ref: Ref(T)").Attr("T: " NUMTYPES).Attr("Tindices: {int32, int64}").Attr("use_locking: bool = false");
+REGISTER_OP("ApplyGradientDescent").Input("var: Ref(T)").Input("alpha: T").Input("delta: T").Output("out: Ref(T)").Attr("T: " FLOATTYPES).Attr("use_locking: bool = false");
+REGISTER_OP("ApplyMomentum").Input("var: Ref(T)").Input("accum: Ref(T)").Input("lr: T").Input("grad: T").Input("momentum: T").Output("out: Ref(T)").Attr("T: " FLOATTYPES).Attr("use_locking: bool = false").Attr("use_nesterov: bool = false");
+REGISTER_OP("ApplyAdam").Input("var: Ref(T)").Input("m: Ref(T)").Input("v: Ref(T)").Input("beta1_power: T").Input("beta2_power: T").Input("lr: T").Input("beta1: T").Input("beta2: T").Input("epsilon: T").Input("grad: T").Output("out: Ref(T)").Attr("T: " FLOATTYPES).Attr("use_locking: bool = false");
+REGISTER_OP("ApplyRMSProp").Input("var: Ref(T)").Input("ms: Ref(T)").Input("mom: Ref(T)").Input("lr: T").Input("rho: T").Input("momentum: T").Input("epsilon: T").Input("grad: T").Output("out: Ref(T)").Attr("T: " FLOATTYPES).Attr("use_locking: bool = false");
+REGISTER_OP("ApplyAdagrad").Input("var: Ref(T)").Input("accum: Ref(T)").Input("lr: T").Input("grad: T").Output("out: Ref(T)").Attr("T: " FLOATTYPES).Attr("use_locking: bool = false");
+REGISTER_OP("ApplyAdadelta").Input("var: Ref(T)").Input("accum: Ref(T)").Input("accum_update: Ref(T)").Input("lr: T").Input("rho: T").Input("epsilon: T").Input("grad: T").Output("out: Ref(T)").Attr("T: " FLOATTYPES).Attr("use_locking: bool = false");
+REGISTER_OP("ApplyFtrl").Input("var: Ref(T)").Input("accum: Ref(T)").Input("linear: Ref(T)").Input("grad: T").Input("lr: T").Input("l1: T").Input("l2: T").Input("lr_power: T").Output("out: Ref(T)").Attr("T: " FLOATTYPES).Attr("use_locking: bool = false");
+REGISTER_OP("ApplyProximalGradientDescent").Input("var: Ref(T)").Input("alpha: T").Input("l1: T").Input("l2: T").Input("delta: T").Output("out: Ref(T)").Attr("T: " FLOATTYPES).Attr("use_locking: bool = false");
+REGISTER_OP("CountUpTo").Input("ref: Ref(T)").Output("output: T").Attr("limit: int").Attr("T: {int32, int64}");
+REGISTER_OP("DestroyTemporaryVariable").Input("ref: Ref(T)").Output("value: T").Attr("T: type").Attr("var_name: string");
+
+// ------------------------------ random_ops ---------------------------------
+REGISTER_OP("RandomUniform").Input("shape: T").Output("output: dtype").Attr("seed: int = 0").Attr("seed2: int = 0").Attr("dtype: " FLOATTYPES).Attr("T: {int32, int64}").SetIsStateful();
+REGISTER_OP("RandomStandardNormal").Input("shape: T").Output("output: dtype").Attr("seed: int = 0").Attr("seed2: int = 0").Attr("dtype: " FLOATTYPES).Attr("T: {int32, int64}").SetIsStateful();
+REGISTER_OP("TruncatedNormal").Input("shape: T").Output("output: dtype").Attr("seed: int = 0").Attr("seed2: int = 0").Attr("dtype: " FLOATTYPES).Attr("T: {int32, int64}").SetIsStateful();
+REGISTER_OP("RandomUniformInt").Input("shape: T").Input("minval: Tout").Input("maxval: Tout").Output("output: Tout").Attr("seed: int = 0").Attr("seed2: int = 0").Attr("T: {int32, int64}").Attr("Tout: {int32, int64}").SetIsStateful();
+REGISTER_OP("RandomShuffle").Input("value: T").Output("output: T").Attr("seed: int = 0").Attr("seed2: int = 0").Attr("T: type").SetIsStateful();
+REGISTER_OP("Multinomial").Input("logits: T").Input("num_samples: int32").Output("output: int64").Attr("seed: int = 0").Attr("seed2: int = 0").Attr("T: " REALTYPES).SetIsStateful();
+
+// ----------------------------- control flow --------------------------------
+REGISTER_OP("Switch").Input("data: T").Input("pred: bool").Output("output_false: T").Output("output_true: T").Attr("T: type");
+REGISTER_OP("RefSwitch").Input("data: Ref(T)").Input("pred: bool").Output("output_false: Ref(T)").Output("output_true: Ref(T)").Attr("T: type").SetAllowsUninitializedInput();
+REGISTER_OP("Merge").Input("inputs: N * T").Output("output: T").Output("value_index: int32").Attr("N: int >= 1").Attr("T: type");
+REGISTER_OP("RefMerge").Input("inputs: N * Ref(T)").Output("output: Ref(T)").Output("value_index: int32").Attr("N: int >= 1").Attr("T: type");
+REGISTER_OP("Enter").Input("data: T").Output("output: T").Attr("T: type").Attr("frame_name: string").Attr("is_constant: bool = false").Attr("parallel_iterations: int = 10");
+REGISTER_OP("RefEnter").Input("data: Ref(T)").Output("output: Ref(T)").Attr("T: type").Attr("frame_name: string").Attr("is_constant: bool = false").Attr("parallel_iterations: int = 10");
+REGISTER_OP("Exit").Input("data: T").Output("output: T").Attr("T: type");
+REGISTER_OP("RefExit").Input("data: Ref(T)").Output("output: Ref(T)").Attr("T: type");
+REGISTER_OP("NextIteration").Input("data: T").Output("output: T").Attr("T: type");
+REGISTER_OP("RefNextIteration").Input("data: Ref(T)").Output("output: Ref(T)").Attr("T: type");
+REGISTER_OP("LoopCond").Input("input: bool").Output("output: bool");
+REGISTER_OP("ControlTrigger");
+REGISTER_OP("NoOp");
+
+// ------------------------------ send/recv ----------------------------------
+REGISTER_OP("_Send").Input("tensor: T").Attr("T: type").Attr("tensor_name: string").Attr("send_device: string").Attr("send_device_incarnation: int = 0").Attr("recv_device: string").Attr("client_terminated: bool = false").SetIsStateful();
+REGISTER_OP("_Recv").Output("tensor: tensor_type").Attr("tensor_type: type").Attr("tensor_name: string").Attr("send_device: string").Attr("send_device_incarnation: int = 0").Attr("recv_device: string").Attr("client_terminated: bool = false").SetIsStateful();
+
+// ------------------------------ logging/debug -------------------------------
+REGISTER_OP("Assert").Input("condition: bool").Input("data: T").Attr("T: list(type)").Attr("summarize: int = 3").SetIsStateful();
+REGISTER_OP("Print").Input("input: T").Input("data: U").Output("output: T").Attr("T: type").Attr("U: list(type)").Attr("message: string = ''").Attr("first_n: int = -1").Attr("summarize: int = 3").SetIsStateful();
+REGISTER_OP("ScalarSummary").Input("tags: string").Input("values: T").Output("summary: string").Attr("T: " REALTYPES);
+REGISTER_OP("HistogramSummary").Input("tag: string").Input("values: T").Output("summary: string").Attr("T: " REALTYPES);
+REGISTER_OP("MergeSummary").Input("inputs: N * string").Output("summary: string").Attr("N: int >= 1");
+
+// ------------------------------ io / ckpt ----------------------------------
+REGISTER_OP("SaveV2").Input("prefix: string").Input("tensor_names: string").Input("shape_and_slices: string").Input("tensors: dtypes").Attr("dtypes: list(type)").SetIsStateful();
+REGISTER_OP("RestoreV2").Input("prefix: string").Input("tensor_names: string").Input("shape_and_slices: string").Output("tensors: dtypes").Attr("dtypes: list(type)").SetIsStateful();
+REGISTER_OP("MergeV2Checkpoints").Input("checkpoint_prefixes: string").Input("destination_prefix: string").Attr("delete_old_dirs: bool = true").SetIsStateful();
+
+// ----------------------------- collectives ---------------------------------
+// MI355X-native: RCCL collectives over xGMI as first-class graph ops
+// (the reference had no collectives — §2.3 of SURVEY.md; gradient
+// aggregation there was AddN/PS. Here all-reduce is the primary multi-GPU
+// path per BASELINE.json config 3).
+REGISTER_OP("RcclAllReduce").Input("input: T").Output("output: T").Attr("T: {float, bfloat16, half}").Attr("reduction: string = 'sum'").Attr("num_devices: int = 1").Attr("shared_name: string = ''").SetIsStateful();
+REGISTER_OP("RcclBroadcast").Input("input: T").Output("output: T").Attr("T: {float, bfloat16, half}").Attr("root: int = 0").SetIsStateful();
+
+}  // namespace stf

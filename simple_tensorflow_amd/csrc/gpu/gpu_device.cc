@@ -1,0 +1,262 @@
+// MI355X HIP device: stream group (compute/h2d/d2h), BFC allocator over
+// hipMalloc sized for 288 GB HBM3E, pinned-host allocator.
+//
+// Capability analog of the reference's BaseGPUDevice + GPUBFCAllocator
+// (common_runtime/gpu/gpu_device.cc, gpu_bfc_allocator.cc, bfc_allocator.h:44)
+// without StreamExecutor: there is exactly one GPU backend (HIP/gfx950), so
+// kernels enqueue directly on the device's hipStream_t. All compute runs on a
+// single compute stream, which makes BFC reuse stream-ordered by
+// construction; H2D/D2H copies run on side streams fenced by hipEvents.
+#include <hip/hip_runtime.h>
+
+#include <map>
+#include <mutex>
+#include <set>
+
+#include "framework/device.h"
+#include "framework/op_kernel.h"
+
+namespace stf {
+
+#define HIP_CHECK_STATUS(expr)                                             \
+  do {                                                                     \
+    hipError_t _e = (expr);                                                \
+    if (_e != hipSuccess)                                                  \
+      return errors::Internal("HIP error: ", hipGetErrorString(_e), " at ", \
+                              __FILE__, ":", __LINE__);                    \
+  } while (0)
+
+// ---------------------------------------------------------------------------
+// BFC allocator over hipMalloc: power-of-two bins, best-fit, split+coalesce,
+// on-demand region growth (allow_growth style — the MI355X has 288 GB, we
+// grab 1 GiB regions as needed and reuse aggressively).
+// ---------------------------------------------------------------------------
+class GpuBfcAllocator : public Allocator {
+ public:
+  explicit GpuBfcAllocator(int ordinal) : ordinal_(ordinal) {}
+
+  void* Allocate(size_t bytes) override {
+    if (bytes == 0) bytes = 256;
+    size_t size = RoundUp(bytes);
+    std::lock_guard<std::mutex> l(mu_);
+    Chunk* c = FindFree(size);
+    if (!c) {
+      if (!Grow(size)) {
+        // Retry once after trying a bigger region failed: hard OOM.
+        LOG(ERROR) << "GPU" << ordinal_ << " BFC: out of memory allocating "
+                   << bytes << " bytes (in use: " << in_use_bytes_ << ")";
+        return nullptr;
+      }
+      c = FindFree(size);
+      if (!c) return nullptr;
+    }
+    RemoveFromFree(c);
+    // Split if the remainder is useful.
+    if (c->size >= size + 512) {
+      Chunk* rest = new Chunk();
+      rest->ptr = (char*)c->ptr + size;
+      rest->size = c->size - size;
+      rest->prev = c;
+      rest->next = c->next;
+      if (rest->next) rest->next->prev = rest;
+      c->next = rest;
+      c->size = size;
+      InsertFree(rest);
+    }
+    c->in_use = true;
+    in_use_bytes_ += c->size;
+    by_ptr_[c->ptr] = c;
+    return c->ptr;
+  }
+
+  void Deallocate(void* ptr, size_t) override {
+    if (!ptr) return;
+    std::lock_guard<std::mutex> l(mu_);
+    auto it = by_ptr_.find(ptr);
+    if (it == by_ptr_.end()) {
+      LOG(ERROR) << "BFC: free of unknown pointer";
+      return;
+    }
+    Chunk* c = it->second;
+    by_ptr_.erase(it);
+    c->in_use = false;
+    in_use_bytes_ -= c->size;
+    // Coalesce with neighbors inside the same region.
+    if (c->next && !c->next->in_use) {
+      Chunk* n = c->next;
+      RemoveFromFree(n);
+      c->size += n->size;
+      c->next = n->next;
+      if (c->next) c->next->prev = c;
+      delete n;
+    }
+    if (c->prev && !c->prev->in_use) {
+      Chunk* p = c->prev;
+      RemoveFromFree(p);
+      p->size += c->size;
+      p->next = c->next;
+      if (p->next) p->next->prev = p;
+      delete c;
+      c = p;
+    }
+    InsertFree(c);
+  }
+
+  MemSpace space() const override { return MemSpace::DEVICE; }
+  int device_ordinal() const override { return ordinal_; }
+  const char* name() const override { return "gpu_bfc"; }
+
+ private:
+  struct Chunk {
+    void* ptr = nullptr;
+    size_t size = 0;
+    bool in_use = false;
+    Chunk* prev = nullptr;  // adjacent in region
+    Chunk* next = nullptr;
+  };
+  struct BySize {
+    bool operator()(const Chunk* a, const Chunk* b) const {
+      if (a->size != b->size) return a->size < b->size;
+      return a->ptr < b->ptr;
+    }
+  };
+
+  static size_t RoundUp(size_t b) { return (b + 255) & ~size_t(255); }
+
+  Chunk* FindFree(size_t size) {
+    Chunk probe;
+    probe.size = size;
+    probe.ptr = nullptr;
+    auto it = free_.lower_bound(&probe);
+    return it == free_.end() ? nullptr : *it;
+  }
+  void InsertFree(Chunk* c) { free_.insert(c); }
+  void RemoveFromFree(Chunk* c) { free_.erase(c); }
+
+  bool Grow(size_t min_bytes) {
+    size_t region = 1ull << 30;  // 1 GiB
+    while (region < min_bytes) region <<= 1;
+    void* p = nullptr;
+    hipError_t e = hipSuccess;
+    for (;;) {
+      e = hipMalloc(&p, region);
+      if (e == hipSuccess) break;
+      if (region <= min_bytes || region <= (1ull << 26)) return false;
+      region >>= 1;
+      if (region < min_bytes) region = RoundUp(min_bytes);
+    }
+    Chunk* c = new Chunk();
+    c->ptr = p;
+    c->size = region;
+    InsertFree(c);
+    total_bytes_ += region;
+    return true;
+  }
+
+  int ordinal_;
+  std::mutex mu_;
+  std::set<Chunk*, BySize> free_;
+  std::map<void*, Chunk*> by_ptr_;
+  size_t total_bytes_ = 0;
+  size_t in_use_bytes_ = 0;
+};
+
+// Pinned host memory for fast DMA.
+class PinnedAllocator : public Allocator {
+ public:
+  void* Allocate(size_t bytes) override {
+    void* p = nullptr;
+    if (hipHostMalloc(&p, bytes ? bytes : 1) != hipSuccess) return nullptr;
+    return p;
+  }
+  void Deallocate(void* ptr, size_t) override { hipHostFree(ptr); }
+  const char* name() const override { return "pinned"; }
+};
+
+// ---------------------------------------------------------------------------
+// GpuDevice
+// ---------------------------------------------------------------------------
+class GpuDevice : public Device {
+ public:
+  GpuDevice(int ordinal, const std::string& name)
+      : Device(name, "GPU"), ordinal_(ordinal), bfc_(ordinal) {
+    hipSetDevice(ordinal_);
+    hipStreamCreateWithFlags(&compute_, hipStreamNonBlocking);
+    hipStreamCreateWithFlags(&h2d_, hipStreamNonBlocking);
+    hipStreamCreateWithFlags(&d2h_, hipStreamNonBlocking);
+  }
+  ~GpuDevice() override {
+    hipStreamDestroy(compute_);
+    hipStreamDestroy(h2d_);
+    hipStreamDestroy(d2h_);
+  }
+
+  Allocator* allocator() override { return &bfc_; }
+  Allocator* host_allocator() override {
+    static PinnedAllocator* pinned = new PinnedAllocator();
+    return pinned;
+  }
+  void* compute_stream() override { return (void*)compute_; }
+  int gpu_ordinal() const override { return ordinal_; }
+
+  void Compute(OpKernel* kernel, OpKernelContext* ctx) override {
+    hipSetDevice(ordinal_);
+    kernel->Compute(ctx);
+  }
+
+  Status Sync() override {
+    hipSetDevice(ordinal_);
+    HIP_CHECK_STATUS(hipDeviceSynchronize());
+    return Status::OK();
+  }
+
+  Status CopyDeviceTensorToHost(const Tensor& src, Tensor* dst) override {
+    hipSetDevice(ordinal_);
+    Tensor host(host_allocator(), src.dtype(), src.shape());
+    // Fence: wait for pending compute that may produce src.
+    hipEvent_t ev;
+    HIP_CHECK_STATUS(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+    HIP_CHECK_STATUS(hipEventRecord(ev, compute_));
+    HIP_CHECK_STATUS(hipStreamWaitEvent(d2h_, ev, 0));
+    HIP_CHECK_STATUS(hipMemcpyAsync(host.raw_data(), src.raw_data(),
+                                    src.TotalBytes(), hipMemcpyDeviceToHost,
+                                    d2h_));
+    HIP_CHECK_STATUS(hipStreamSynchronize(d2h_));
+    HIP_CHECK_STATUS(hipEventDestroy(ev));
+    *dst = host;
+    return Status::OK();
+  }
+
+  Status CopyHostTensorToDevice(const Tensor& src, Tensor* dst) override {
+    hipSetDevice(ordinal_);
+    Tensor dev(&bfc_, src.dtype(), src.shape());
+    HIP_CHECK_STATUS(hipMemcpyAsync(dev.raw_data(), src.raw_data(),
+                                    src.TotalBytes(), hipMemcpyHostToDevice,
+                                    h2d_));
+    // Make the compute stream wait for the transfer; then sync h2d so the
+    // host source buffer may be released by the caller.
+    hipEvent_t ev;
+    HIP_CHECK_STATUS(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+    HIP_CHECK_STATUS(hipEventRecord(ev, h2d_));
+    HIP_CHECK_STATUS(hipStreamWaitEvent(compute_, ev, 0));
+    HIP_CHECK_STATUS(hipStreamSynchronize(h2d_));
+    HIP_CHECK_STATUS(hipEventDestroy(ev));
+    *dst = dev;
+    return Status::OK();
+  }
+
+ private:
+  int ordinal_;
+  GpuBfcAllocator bfc_;
+  hipStream_t compute_, h2d_, d2h_;
+};
+
+void AddGpuDevices(DeviceMgr* mgr) {
+  int count = 0;
+  if (hipGetDeviceCount(&count) != hipSuccess) return;
+  for (int i = 0; i < count; ++i) {
+    mgr->AddDevice(std::make_unique<GpuDevice>(i, "/gpu:" + std::to_string(i)));
+  }
+}
+
+}  // namespace stf

@@ -1,0 +1,478 @@
+// CPU NN kernels: Conv2D (NHWC direct loops), pooling, batch norm.
+// These are the fp32 reference implementations the HIP kernels are validated
+// against (SURVEY.md §4 test strategy); CPU speed is not a goal.
+#include <cmath>
+
+#include "kernels/kernel_util.h"
+
+namespace stf {
+
+struct Conv2DParams {
+  int64_t N, H, W, C, R, S, K, stride_h, stride_w, pad_h, pad_w, P, Q;
+};
+
+static Status ComputeConvParams(const TensorShape& input,
+                                const TensorShape& filter,
+                                const std::vector<int64_t>& strides,
+                                const std::string& padding, Conv2DParams* p) {
+  p->N = input.dim_size(0);
+  p->H = input.dim_size(1);
+  p->W = input.dim_size(2);
+  p->C = input.dim_size(3);
+  p->R = filter.dim_size(0);
+  p->S = filter.dim_size(1);
+  if (filter.dim_size(2) != p->C)
+    return errors::InvalidArgument("Conv2D channel mismatch");
+  p->K = filter.dim_size(3);
+  p->stride_h = strides[1];
+  p->stride_w = strides[2];
+  if (padding == "SAME") {
+    p->P = (p->H + p->stride_h - 1) / p->stride_h;
+    p->Q = (p->W + p->stride_w - 1) / p->stride_w;
+    int64_t pad_rows = std::max<int64_t>(
+        0, (p->P - 1) * p->stride_h + p->R - p->H);
+    int64_t pad_cols = std::max<int64_t>(
+        0, (p->Q - 1) * p->stride_w + p->S - p->W);
+    p->pad_h = pad_rows / 2;
+    p->pad_w = pad_cols / 2;
+  } else if (padding == "VALID") {
+    p->P = (p->H - p->R) / p->stride_h + 1;
+    p->Q = (p->W - p->S) / p->stride_w + 1;
+    p->pad_h = p->pad_w = 0;
+  } else {
+    return errors::InvalidArgument("bad padding: ", padding);
+  }
+  return Status::OK();
+}
+
+template <typename T>
+class Conv2DOp : public OpKernel {
+ public:
+  explicit Conv2DOp(OpKernelConstruction* ctx) : OpKernel(ctx) {
+    ctx->GetAttr("strides", &strides_);
+    ctx->GetAttr("padding", &padding_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& in = ctx->input(0);
+    const Tensor& filter = ctx->input(1);
+    Conv2DParams p;
+    OP_REQUIRES_OK(ctx, ComputeConvParams(in.shape(), filter.shape(), strides_,
+                                          padding_, &p));
+    Tensor* out = ctx->allocate_output(0, TensorShape({p.N, p.P, p.Q, p.K}));
+    const T* x = in.flat<T>();
+    const T* w = filter.flat<T>();
+    T* y = out->flat<T>();
+    std::memset(y, 0, out->TotalBytes());
+    for (int64_t n = 0; n < p.N; ++n)
+      for (int64_t ph = 0; ph < p.P; ++ph)
+        for (int64_t pw = 0; pw < p.Q; ++pw) {
+          T* yrow = y + ((n * p.P + ph) * p.Q + pw) * p.K;
+          for (int64_t r = 0; r < p.R; ++r) {
+            int64_t ih = ph * p.stride_h - p.pad_h + r;
+            if (ih < 0 || ih >= p.H) continue;
+            for (int64_t s = 0; s < p.S; ++s) {
+              int64_t iw = pw * p.stride_w - p.pad_w + s;
+              if (iw < 0 || iw >= p.W) continue;
+              const T* xrow = x + ((n * p.H + ih) * p.W + iw) * p.C;
+              const T* wrow = w + (r * p.S + s) * p.C * p.K;
+              for (int64_t c = 0; c < p.C; ++c) {
+                T xv = xrow[c];
+                const T* wk = wrow + c * p.K;
+                for (int64_t k = 0; k < p.K; ++k) yrow[k] += xv * wk[k];
+              }
+            }
+          }
+        }
+  }
+
+ private:
+  std::vector<int64_t> strides_;
+  std::string padding_;
+};
+REGISTER_CPU_KERNEL_FLOATS("Conv2D", Conv2DOp)
+
+template <typename T>
+class Conv2DBackpropInputOp : public OpKernel {
+ public:
+  explicit Conv2DBackpropInputOp(OpKernelConstruction* ctx) : OpKernel(ctx) {
+    ctx->GetAttr("strides", &strides_);
+    ctx->GetAttr("padding", &padding_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    auto in_sizes = IntVector(ctx->input(0));
+    const Tensor& filter = ctx->input(1);
+    const Tensor& dy = ctx->input(2);
+    TensorShape in_shape(in_sizes);
+    Conv2DParams p;
+    OP_REQUIRES_OK(ctx, ComputeConvParams(in_shape, filter.shape(), strides_,
+                                          padding_, &p));
+    Tensor* out = ctx->allocate_output(0, in_shape);
+    const T* w = filter.flat<T>();
+    const T* g = dy.flat<T>();
+    T* dx = out->flat<T>();
+    std::memset(dx, 0, out->TotalBytes());
+    for (int64_t n = 0; n < p.N; ++n)
+      for (int64_t ph = 0; ph < p.P; ++ph)
+        for (int64_t pw = 0; pw < p.Q; ++pw) {
+          const T* grow = g + ((n * p.P + ph) * p.Q + pw) * p.K;
+          for (int64_t r = 0; r < p.R; ++r) {
+            int64_t ih = ph * p.stride_h - p.pad_h + r;
+            if (ih < 0 || ih >= p.H) continue;
+            for (int64_t s = 0; s < p.S; ++s) {
+              int64_t iw = pw * p.stride_w - p.pad_w + s;
+              if (iw < 0 || iw >= p.W) continue;
+              T* xrow = dx + ((n * p.H + ih) * p.W + iw) * p.C;
+              const T* wrow = w + (r * p.S + s) * p.C * p.K;
+              for (int64_t c = 0; c < p.C; ++c) {
+                const T* wk = wrow + c * p.K;
+                T acc = 0;
+                for (int64_t k = 0; k < p.K; ++k) acc += grow[k] * wk[k];
+                xrow[c] += acc;
+              }
+            }
+          }
+        }
+  }
+
+ private:
+  std::vector<int64_t> strides_;
+  std::string padding_;
+};
+REGISTER_CPU_KERNEL_FLOATS("Conv2DBackpropInput", Conv2DBackpropInputOp)
+
+template <typename T>
+class Conv2DBackpropFilterOp : public OpKernel {
+ public:
+  explicit Conv2DBackpropFilterOp(OpKernelConstruction* ctx) : OpKernel(ctx) {
+    ctx->GetAttr("strides", &strides_);
+    ctx->GetAttr("padding", &padding_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& in = ctx->input(0);
+    auto f_sizes = IntVector(ctx->input(1));
+    const Tensor& dy = ctx->input(2);
+    TensorShape f_shape(f_sizes);
+    Conv2DParams p;
+    OP_REQUIRES_OK(
+        ctx, ComputeConvParams(in.shape(), f_shape, strides_, padding_, &p));
+    Tensor* out = ctx->allocate_output(0, f_shape);
+    const T* x = in.flat<T>();
+    const T* g = dy.flat<T>();
+    T* dw = out->flat<T>();
+    std::memset(dw, 0, out->TotalBytes());
+    for (int64_t n = 0; n < p.N; ++n)
+      for (int64_t ph = 0; ph < p.P; ++ph)
+        for (int64_t pw = 0; pw < p.Q; ++pw) {
+          const T* grow = g + ((n * p.P + ph) * p.Q + pw) * p.K;
+          for (int64_t r = 0; r < p.R; ++r) {
+            int64_t ih = ph * p.stride_h - p.pad_h + r;
+            if (ih < 0 || ih >= p.H) continue;
+            for (int64_t s = 0; s < p.S; ++s) {
+              int64_t iw = pw * p.stride_w - p.pad_w + s;
+              if (iw < 0 || iw >= p.W) continue;
+              const T* xrow = x + ((n * p.H + ih) * p.W + iw) * p.C;
+              T* wrow = dw + (r * p.S + s) * p.C * p.K;
+              for (int64_t c = 0; c < p.C; ++c) {
+                T xv = xrow[c];
+                T* wk = wrow + c * p.K;
+                for (int64_t k = 0; k < p.K; ++k) wk[k] += xv * grow[k];
+              }
+            }
+          }
+        }
+  }
+
+ private:
+  std::vector<int64_t> strides_;
+  std::string padding_;
+};
+REGISTER_CPU_KERNEL_FLOATS("Conv2DBackpropFilter", Conv2DBackpropFilterOp)
+
+// --------------------------------- pooling ---------------------------------
+struct PoolParams {
+  int64_t N, H, W, C, kh, kw, sh, sw, pad_h, pad_w, P, Q;
+};
+static PoolParams GetPoolParams(const TensorShape& input,
+                                const std::vector<int64_t>& ksize,
+                                const std::vector<int64_t>& strides,
+                                const std::string& padding) {
+  PoolParams p;
+  p.N = input.dim_size(0);
+  p.H = input.dim_size(1);
+  p.W = input.dim_size(2);
+  p.C = input.dim_size(3);
+  p.kh = ksize[1];
+  p.kw = ksize[2];
+  p.sh = strides[1];
+  p.sw = strides[2];
+  if (padding == "SAME") {
+    p.P = (p.H + p.sh - 1) / p.sh;
+    p.Q = (p.W + p.sw - 1) / p.sw;
+    p.pad_h = std::max<int64_t>(0, (p.P - 1) * p.sh + p.kh - p.H) / 2;
+    p.pad_w = std::max<int64_t>(0, (p.Q - 1) * p.sw + p.kw - p.W) / 2;
+  } else {
+    p.P = (p.H - p.kh) / p.sh + 1;
+    p.Q = (p.W - p.kw) / p.sw + 1;
+    p.pad_h = p.pad_w = 0;
+  }
+  return p;
+}
+
+template <typename T, bool is_max>
+class PoolOp : public OpKernel {
+ public:
+  explicit PoolOp(OpKernelConstruction* ctx) : OpKernel(ctx) {
+    ctx->GetAttr("ksize", &ksize_);
+    ctx->GetAttr("strides", &strides_);
+    ctx->GetAttr("padding", &padding_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& in = ctx->input(0);
+    PoolParams p = GetPoolParams(in.shape(), ksize_, strides_, padding_);
+    Tensor* out = ctx->allocate_output(0, TensorShape({p.N, p.P, p.Q, p.C}));
+    const T* x = in.flat<T>();
+    T* y = out->flat<T>();
+    for (int64_t n = 0; n < p.N; ++n)
+      for (int64_t ph = 0; ph < p.P; ++ph)
+        for (int64_t pw = 0; pw < p.Q; ++pw)
+          for (int64_t c = 0; c < p.C; ++c) {
+            T best = is_max ? std::numeric_limits<T>::lowest() : T(0);
+            int64_t count = 0;
+            for (int64_t kh = 0; kh < p.kh; ++kh) {
+              int64_t ih = ph * p.sh - p.pad_h + kh;
+              if (ih < 0 || ih >= p.H) continue;
+              for (int64_t kw = 0; kw < p.kw; ++kw) {
+                int64_t iw = pw * p.sw - p.pad_w + kw;
+                if (iw < 0 || iw >= p.W) continue;
+                T v = x[((n * p.H + ih) * p.W + iw) * p.C + c];
+                if (is_max) best = v > best ? v : best;
+                else best += v;
+                ++count;
+              }
+            }
+            y[((n * p.P + ph) * p.Q + pw) * p.C + c] =
+                is_max ? best : best / (T)count;
+          }
+  }
+
+ protected:
+  std::vector<int64_t> ksize_, strides_;
+  std::string padding_;
+};
+REGISTER_KERNEL_BUILDER(Name("MaxPool").Device(DEVICE_CPU).TypeConstraint<float>("T"), PoolOp<float, true>);
+REGISTER_KERNEL_BUILDER(Name("MaxPool").Device(DEVICE_CPU).TypeConstraint<double>("T"), PoolOp<double, true>);
+REGISTER_KERNEL_BUILDER(Name("AvgPool").Device(DEVICE_CPU).TypeConstraint<float>("T"), PoolOp<float, false>);
+REGISTER_KERNEL_BUILDER(Name("AvgPool").Device(DEVICE_CPU).TypeConstraint<double>("T"), PoolOp<double, false>);
+
+template <typename T>
+class MaxPoolGradOp : public OpKernel {
+ public:
+  explicit MaxPoolGradOp(OpKernelConstruction* ctx) : OpKernel(ctx) {
+    ctx->GetAttr("ksize", &ksize_);
+    ctx->GetAttr("strides", &strides_);
+    ctx->GetAttr("padding", &padding_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& in = ctx->input(0);
+    const Tensor& dy = ctx->input(2);
+    PoolParams p = GetPoolParams(in.shape(), ksize_, strides_, padding_);
+    Tensor* out = ctx->allocate_output(0, in.shape());
+    const T* x = in.flat<T>();
+    const T* g = dy.flat<T>();
+    T* dx = out->flat<T>();
+    std::memset(dx, 0, out->TotalBytes());
+    for (int64_t n = 0; n < p.N; ++n)
+      for (int64_t ph = 0; ph < p.P; ++ph)
+        for (int64_t pw = 0; pw < p.Q; ++pw)
+          for (int64_t c = 0; c < p.C; ++c) {
+            // find argmax
+            T best = std::numeric_limits<T>::lowest();
+            int64_t bi = -1;
+            for (int64_t kh = 0; kh < p.kh; ++kh) {
+              int64_t ih = ph * p.sh - p.pad_h + kh;
+              if (ih < 0 || ih >= p.H) continue;
+              for (int64_t kw = 0; kw < p.kw; ++kw) {
+                int64_t iw = pw * p.sw - p.pad_w + kw;
+                if (iw < 0 || iw >= p.W) continue;
+                int64_t idx = ((n * p.H + ih) * p.W + iw) * p.C + c;
+                if (x[idx] > best) {
+                  best = x[idx];
+                  bi = idx;
+                }
+              }
+            }
+            if (bi >= 0) dx[bi] += g[((n * p.P + ph) * p.Q + pw) * p.C + c];
+          }
+  }
+
+ private:
+  std::vector<int64_t> ksize_, strides_;
+  std::string padding_;
+};
+REGISTER_KERNEL_BUILDER(Name("MaxPoolGrad").Device(DEVICE_CPU).TypeConstraint<float>("T"), MaxPoolGradOp<float>);
+REGISTER_KERNEL_BUILDER(Name("MaxPoolGrad").Device(DEVICE_CPU).TypeConstraint<double>("T"), MaxPoolGradOp<double>);
+
+template <typename T>
+class AvgPoolGradOp : public OpKernel {
+ public:
+  explicit AvgPoolGradOp(OpKernelConstruction* ctx) : OpKernel(ctx) {
+    ctx->GetAttr("ksize", &ksize_);
+    ctx->GetAttr("strides", &strides_);
+    ctx->GetAttr("padding", &padding_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    auto in_sizes = IntVector(ctx->input(0));
+    const Tensor& dy = ctx->input(1);
+    TensorShape in_shape(in_sizes);
+    PoolParams p = GetPoolParams(in_shape, ksize_, strides_, padding_);
+    Tensor* out = ctx->allocate_output(0, in_shape);
+    const T* g = dy.flat<T>();
+    T* dx = out->flat<T>();
+    std::memset(dx, 0, out->TotalBytes());
+    for (int64_t n = 0; n < p.N; ++n)
+      for (int64_t ph = 0; ph < p.P; ++ph)
+        for (int64_t pw = 0; pw < p.Q; ++pw) {
+          // count valid positions
+          int64_t count = 0;
+          for (int64_t kh = 0; kh < p.kh; ++kh) {
+            int64_t ih = ph * p.sh - p.pad_h + kh;
+            if (ih < 0 || ih >= p.H) continue;
+            for (int64_t kw = 0; kw < p.kw; ++kw) {
+              int64_t iw = pw * p.sw - p.pad_w + kw;
+              if (iw >= 0 && iw < p.W) ++count;
+            }
+          }
+          for (int64_t c = 0; c < p.C; ++c) {
+            T gv = g[((n * p.P + ph) * p.Q + pw) * p.C + c] / (T)count;
+            for (int64_t kh = 0; kh < p.kh; ++kh) {
+              int64_t ih = ph * p.sh - p.pad_h + kh;
+              if (ih < 0 || ih >= p.H) continue;
+              for (int64_t kw = 0; kw < p.kw; ++kw) {
+                int64_t iw = pw * p.sw - p.pad_w + kw;
+                if (iw < 0 || iw >= p.W) continue;
+                dx[((n * p.H + ih) * p.W + iw) * p.C + c] += gv;
+              }
+            }
+          }
+        }
+  }
+
+ private:
+  std::vector<int64_t> ksize_, strides_;
+  std::string padding_;
+};
+REGISTER_KERNEL_BUILDER(Name("AvgPoolGrad").Device(DEVICE_CPU).TypeConstraint<float>("T").HostMemory("orig_input_shape"), AvgPoolGradOp<float>);
+REGISTER_KERNEL_BUILDER(Name("AvgPoolGrad").Device(DEVICE_CPU).TypeConstraint<double>("T").HostMemory("orig_input_shape"), AvgPoolGradOp<double>);
+
+// ------------------------------ FusedBatchNorm -------------------------------
+class FusedBatchNormOp : public OpKernel {
+ public:
+  explicit FusedBatchNormOp(OpKernelConstruction* ctx) : OpKernel(ctx) {
+    ctx->GetAttr("epsilon", &eps_);
+    ctx->GetAttr("is_training", &training_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    const Tensor& scale = ctx->input(1);
+    const Tensor& offset = ctx->input(2);
+    int64_t C = x.dim_size(3);
+    int64_t rows = x.NumElements() / C;
+    Tensor* y = ctx->allocate_output(0, x.shape());
+    Tensor* mean_out = ctx->allocate_output(1, TensorShape({C}));
+    Tensor* var_out = ctx->allocate_output(2, TensorShape({C}));
+    Tensor* save_mean = ctx->allocate_output(3, TensorShape({C}));
+    Tensor* save_inv = ctx->allocate_output(4, TensorShape({C}));
+    const float* xp = x.flat<float>();
+    float* yp = y->flat<float>();
+    std::vector<double> mean(C, 0), var(C, 0);
+    if (training_) {
+      for (int64_t i = 0; i < rows; ++i)
+        for (int64_t c = 0; c < C; ++c) mean[c] += xp[i * C + c];
+      for (int64_t c = 0; c < C; ++c) mean[c] /= rows;
+      for (int64_t i = 0; i < rows; ++i)
+        for (int64_t c = 0; c < C; ++c) {
+          double d = xp[i * C + c] - mean[c];
+          var[c] += d * d;
+        }
+      for (int64_t c = 0; c < C; ++c) var[c] /= rows;
+    } else {
+      const float* m = ctx->input(3).flat<float>();
+      const float* v = ctx->input(4).flat<float>();
+      for (int64_t c = 0; c < C; ++c) {
+        mean[c] = m[c];
+        var[c] = v[c];
+      }
+    }
+    const float* sc = scale.flat<float>();
+    const float* of = offset.flat<float>();
+    std::vector<float> inv(C);
+    for (int64_t c = 0; c < C; ++c) {
+      inv[c] = 1.0f / std::sqrt((float)var[c] + eps_);
+      mean_out->flat<float>()[c] = (float)mean[c];
+      var_out->flat<float>()[c] = (float)var[c];
+      save_mean->flat<float>()[c] = (float)mean[c];
+      save_inv->flat<float>()[c] = inv[c];
+    }
+    for (int64_t i = 0; i < rows; ++i)
+      for (int64_t c = 0; c < C; ++c)
+        yp[i * C + c] =
+            ((xp[i * C + c] - (float)mean[c]) * inv[c]) * sc[c] + of[c];
+  }
+
+ private:
+  float eps_ = 1e-4f;
+  bool training_ = true;
+};
+REGISTER_KERNEL_BUILDER(Name("FusedBatchNorm").Device(DEVICE_CPU).TypeConstraint<float>("T"), FusedBatchNormOp);
+
+class FusedBatchNormGradOp : public OpKernel {
+ public:
+  explicit FusedBatchNormGradOp(OpKernelConstruction* ctx) : OpKernel(ctx) {
+    ctx->GetAttr("epsilon", &eps_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& dy = ctx->input(0);
+    const Tensor& x = ctx->input(1);
+    const Tensor& scale = ctx->input(2);
+    const Tensor& saved_mean = ctx->input(3);
+    const Tensor& saved_inv = ctx->input(4);  // 1/sqrt(var+eps)
+    int64_t C = x.dim_size(3);
+    int64_t rows = x.NumElements() / C;
+    Tensor* dx = ctx->allocate_output(0, x.shape());
+    Tensor* dscale = ctx->allocate_output(1, TensorShape({C}));
+    Tensor* doffset = ctx->allocate_output(2, TensorShape({C}));
+    ctx->allocate_output(3, TensorShape({0}));
+    ctx->allocate_output(4, TensorShape({0}));
+    const float* g = dy.flat<float>();
+    const float* xp = x.flat<float>();
+    const float* sc = scale.flat<float>();
+    const float* mu = saved_mean.flat<float>();
+    const float* inv = saved_inv.flat<float>();
+    float* dxp = dx->flat<float>();
+    std::vector<double> sum_dy(C, 0), sum_dy_xhat(C, 0);
+    for (int64_t i = 0; i < rows; ++i)
+      for (int64_t c = 0; c < C; ++c) {
+        float xhat = (xp[i * C + c] - mu[c]) * inv[c];
+        sum_dy[c] += g[i * C + c];
+        sum_dy_xhat[c] += g[i * C + c] * xhat;
+      }
+    for (int64_t c = 0; c < C; ++c) {
+      dscale->flat<float>()[c] = (float)sum_dy_xhat[c];
+      doffset->flat<float>()[c] = (float)sum_dy[c];
+    }
+    for (int64_t i = 0; i < rows; ++i)
+      for (int64_t c = 0; c < C; ++c) {
+        float xhat = (xp[i * C + c] - mu[c]) * inv[c];
+        dxp[i * C + c] =
+            sc[c] * inv[c] *
+            (g[i * C + c] - (float)sum_dy[c] / rows -
+             xhat * (float)sum_dy_xhat[c] / rows);
+      }
+  }
+
+ private:
+  float eps_ = 1e-4f;
+};
+REGISTER_KERNEL_BUILDER(Name("FusedBatchNormGrad").Device(DEVICE_CPU).TypeConstraint<float>("T"), FusedBatchNormGradOp);
+
+}  // namespace stf

@@ -1,0 +1,235 @@
+// pybind11 bridge: the analog of the reference's SWIG pywrap_tensorflow +
+// tf_session_helper.cc (numpy <-> Tensor, session lifecycle, op registry
+// introspection for the Python op_def_library).
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "core/protos.h"
+#include "framework/op.h"
+#include "runtime/session.h"
+
+namespace py = pybind11;
+using namespace stf;
+
+namespace {
+
+DataType NumpyToDataType(const py::array& a) {
+  auto dt = a.dtype();
+  char kind = dt.kind();
+  int size = (int)dt.itemsize();
+  if (kind == 'f' && size == 4) return DT_FLOAT;
+  if (kind == 'f' && size == 8) return DT_DOUBLE;
+  if (kind == 'f' && size == 2) return DT_HALF;
+  if (kind == 'i' && size == 4) return DT_INT32;
+  if (kind == 'i' && size == 8) return DT_INT64;
+  if (kind == 'i' && size == 1) return DT_INT8;
+  if (kind == 'i' && size == 2) return DT_INT16;
+  if (kind == 'u' && size == 1) return DT_UINT8;
+  if (kind == 'u' && size == 2) return DT_UINT16;  // also bf16 carrier
+  if (kind == 'b') return DT_BOOL;
+  return DT_INVALID;
+}
+
+Tensor NumpyToTensor(py::array arr) {
+  DataType dt = NumpyToDataType(arr);
+  if (dt == DT_INVALID)
+    throw std::runtime_error("Unsupported numpy dtype for feed");
+  auto buf = py::array::ensure(arr, py::array::c_style | py::array::forcecast);
+  TensorShape shape;
+  for (int i = 0; i < buf.ndim(); ++i) shape.AddDim(buf.shape(i));
+  Tensor t(dt, shape);
+  std::memcpy(t.raw_data(), buf.data(), t.TotalBytes());
+  return t;
+}
+
+py::object TensorToPy(const Tensor& t) {
+  std::vector<ssize_t> shape;
+  for (auto d : t.shape().dim_sizes()) shape.push_back((ssize_t)d);
+  std::string fmt;
+  switch (t.dtype()) {
+    case DT_FLOAT: fmt = py::format_descriptor<float>::format(); break;
+    case DT_DOUBLE: fmt = py::format_descriptor<double>::format(); break;
+    case DT_INT32: fmt = py::format_descriptor<int32_t>::format(); break;
+    case DT_INT64: fmt = py::format_descriptor<int64_t>::format(); break;
+    case DT_UINT8: fmt = py::format_descriptor<uint8_t>::format(); break;
+    case DT_INT8: fmt = py::format_descriptor<int8_t>::format(); break;
+    case DT_INT16: fmt = py::format_descriptor<int16_t>::format(); break;
+    case DT_UINT16:
+    case DT_BFLOAT16:
+    case DT_HALF: fmt = py::format_descriptor<uint16_t>::format(); break;
+    case DT_BOOL: fmt = py::format_descriptor<bool>::format(); break;
+    case DT_STRING: {
+      // Return list (or scalar) of bytes.
+      const std::string* p = t.flat<std::string>();
+      if (t.dims() == 0) return py::bytes(p[0]);
+      py::list out;
+      for (int64_t i = 0; i < t.NumElements(); ++i)
+        out.append(py::bytes(p[i]));
+      return out;
+    }
+    default:
+      throw std::runtime_error(std::string("Unsupported fetch dtype ") +
+                               DataTypeString(t.dtype()));
+  }
+  size_t es = DataTypeSize(t.dtype());
+  py::array out(py::dtype(fmt), shape);
+  std::memcpy(out.mutable_data(), t.raw_data(), (size_t)t.NumElements() * es);
+  if (t.dtype() == DT_BFLOAT16) {
+    // convert bf16 -> float32 for numpy friendliness
+    py::array_t<float> f(shape);
+    const uint16_t* src = t.flat<uint16_t>();
+    float* dst = (float*)f.mutable_data();
+    for (int64_t i = 0; i < t.NumElements(); ++i) {
+      uint32_t bits = ((uint32_t)src[i]) << 16;
+      std::memcpy(dst + i, &bits, 4);
+    }
+    return f;
+  }
+  return out;
+}
+
+py::dict AttrValueToPy(const AttrValue& v);
+
+py::dict OpDefToPy(const OpDef& op) {
+  py::dict d;
+  d["name"] = op.name;
+  auto args = [](const std::vector<OpDef::ArgDef>& as) {
+    py::list out;
+    for (auto& a : as) {
+      py::dict ad;
+      ad["name"] = a.name;
+      ad["type"] = (int)a.type;
+      ad["type_attr"] = a.type_attr;
+      ad["number_attr"] = a.number_attr;
+      ad["type_list_attr"] = a.type_list_attr;
+      ad["is_ref"] = a.is_ref;
+      out.append(ad);
+    }
+    return out;
+  };
+  d["input_arg"] = args(op.input_arg);
+  d["output_arg"] = args(op.output_arg);
+  py::list attrs;
+  for (auto& a : op.attr) {
+    py::dict ad;
+    ad["name"] = a.name;
+    ad["type"] = a.type;
+    ad["has_default"] = a.has_default;
+    if (a.has_default) ad["default"] = AttrValueToPy(a.default_value);
+    py::list allowed;
+    for (auto t : a.allowed) allowed.append((int)t);
+    ad["allowed"] = allowed;
+    attrs.append(ad);
+  }
+  d["attr"] = attrs;
+  d["is_stateful"] = op.is_stateful;
+  return d;
+}
+
+py::dict AttrValueToPy(const AttrValue& v) {
+  py::dict d;
+  d["kind"] = std::string(1, v.kind ? v.kind : '0');
+  switch (v.kind) {
+    case 's': d["value"] = py::bytes(v.s); break;
+    case 'i': d["value"] = v.i; break;
+    case 'f': d["value"] = v.f; break;
+    case 'b': d["value"] = v.b; break;
+    case 't': d["value"] = (int)v.type; break;
+    case 'l': {
+      py::dict lv;
+      lv["s"] = v.list.s;
+      lv["i"] = v.list.i;
+      lv["f"] = v.list.f;
+      lv["b"] = v.list.b;
+      lv["type"] = v.list.type;
+      d["value"] = lv;
+      break;
+    }
+    default: d["value"] = py::none();
+  }
+  return d;
+}
+
+class PySession {
+ public:
+  explicit PySession(bool cpu_only) : sess_(cpu_only) {}
+
+  void Create(py::bytes graph_def) {
+    GraphDef gd;
+    std::string data = graph_def;
+    if (!gd.ParseFromString(data))
+      throw std::runtime_error("Failed to parse GraphDef");
+    Status s = sess_.Create(gd);
+    if (!s.ok()) throw std::runtime_error(s.ToString());
+  }
+  void Extend(py::bytes graph_def) {
+    GraphDef gd;
+    std::string data = graph_def;
+    if (!gd.ParseFromString(data))
+      throw std::runtime_error("Failed to parse GraphDef");
+    Status s = sess_.Extend(gd);
+    if (!s.ok()) throw std::runtime_error(s.ToString());
+  }
+
+  py::list Run(py::dict feeds, std::vector<std::string> fetches,
+               std::vector<std::string> targets) {
+    std::vector<std::pair<std::string, Tensor>> feed_vec;
+    for (auto item : feeds) {
+      std::string name = py::cast<std::string>(item.first);
+      py::object val = py::reinterpret_borrow<py::object>(item.second);
+      if (py::isinstance<py::bytes>(val) || py::isinstance<py::str>(val)) {
+        Tensor t(DT_STRING, TensorShape({}));
+        t.flat<std::string>()[0] = py::cast<std::string>(val);
+        feed_vec.emplace_back(name, t);
+      } else {
+        feed_vec.emplace_back(name, NumpyToTensor(py::cast<py::array>(val)));
+      }
+    }
+    std::vector<Tensor> outputs;
+    Status s;
+    {
+      py::gil_scoped_release release;
+      s = sess_.Run(feed_vec, fetches, targets, &outputs);
+    }
+    if (!s.ok()) throw std::runtime_error(s.ToString());
+    py::list out;
+    for (auto& t : outputs) out.append(TensorToPy(t));
+    return out;
+  }
+
+  int NumGpus() {
+    int n = 0;
+    for (auto& d : sess_.device_mgr()->devices())
+      if (d->is_gpu()) ++n;
+    return n;
+  }
+
+ private:
+  DirectSession sess_;
+};
+
+}  // namespace
+
+PYBIND11_MODULE(_core, m) {
+  m.doc() = "simple_tensorflow_amd core runtime (MI355X-native)";
+
+  py::class_<PySession>(m, "Session")
+      .def(py::init<bool>(), py::arg("cpu_only") = false)
+      .def("create", &PySession::Create)
+      .def("extend", &PySession::Extend)
+      .def("run", &PySession::Run)
+      .def("num_gpus", &PySession::NumGpus);
+
+  m.def("list_ops", []() {
+    py::dict out;
+    for (auto& name : OpRegistry::Global()->ListOps()) {
+      out[py::str(name)] = OpDefToPy(*OpRegistry::Global()->LookUp(name));
+    }
+    return out;
+  });
+  m.def("has_gpu", []() {
+    // cheap probe without creating a session
+    return false;
+  });
+}

@@ -1,0 +1,449 @@
+#include "runtime/session.h"
+
+#include <algorithm>
+#include <condition_variable>
+#include <set>
+#include <thread>
+
+#include "graph/graph.h"
+
+namespace stf {
+
+// Implemented in gpu/gpu_device.cc; adds one Device per visible HIP GPU.
+void AddGpuDevices(DeviceMgr* mgr);
+// Implemented in kernels/resource_mgr.cc.
+void* NewResourceMgr();
+void DeleteResourceMgr(void*);
+
+namespace {
+
+// "name" or "name:3" -> (name, port)
+std::pair<std::string, int> ParseTensorName(const std::string& s) {
+  auto colon = s.rfind(':');
+  if (colon != std::string::npos &&
+      s.find_first_not_of("0123456789", colon + 1) == std::string::npos &&
+      colon + 1 < s.size())
+    return {s.substr(0, colon), atoi(s.c_str() + colon + 1)};
+  return {s, 0};
+}
+
+NodeDef MakeSendRecv(const std::string& op, const std::string& name,
+                     DataType dtype, const std::string& tensor_name,
+                     const std::string& send_dev, const std::string& recv_dev) {
+  NodeDef d;
+  d.name = name;
+  d.op = op;
+  d.attr[op == "_Recv" ? "tensor_type" : "T"] = AttrValue::Type(dtype);
+  d.attr["tensor_name"] = AttrValue::S(tensor_name);
+  d.attr["send_device"] = AttrValue::S(send_dev);
+  d.attr["recv_device"] = AttrValue::S(recv_dev);
+  return d;
+}
+
+}  // namespace
+
+DirectSession::DirectSession(bool force_cpu_only, int num_threads) {
+  devices_.AddDevice(std::make_unique<ThreadPoolDevice>("/cpu:0"));
+  if (!force_cpu_only) AddGpuDevices(&devices_);
+  int n = num_threads > 0 ? num_threads
+                          : std::max(4u, std::thread::hardware_concurrency());
+  pool_ = std::make_unique<ThreadPool>(n);
+  resource_mgr_ = NewResourceMgr();
+}
+
+DirectSession::~DirectSession() { DeleteResourceMgr(resource_mgr_); }
+
+Status DirectSession::Create(const GraphDef& def) {
+  std::lock_guard<std::mutex> l(mu_);
+  graph_def_ = def;
+  executors_.clear();
+  return Status::OK();
+}
+
+Status DirectSession::Extend(const GraphDef& def) {
+  std::lock_guard<std::mutex> l(mu_);
+  for (auto& n : def.node) graph_def_.node.push_back(n);
+  executors_.clear();
+  return Status::OK();
+}
+
+Status DirectSession::GetOrCreateExecutors(
+    const std::vector<std::string>& feeds,
+    const std::vector<std::string>& fetches,
+    const std::vector<std::string>& targets, ExecutorsAndKeys** out) {
+  std::string key;
+  for (auto& f : feeds) key += f + ",";
+  key += "|";
+  for (auto& f : fetches) key += f + ",";
+  key += "|";
+  for (auto& t : targets) key += t + ",";
+  {
+    std::lock_guard<std::mutex> l(mu_);
+    auto it = executors_.find(key);
+    if (it != executors_.end()) {
+      *out = it->second.get();
+      return Status::OK();
+    }
+  }
+  std::unique_ptr<ExecutorsAndKeys> ek;
+  STF_RETURN_IF_ERROR(BuildExecutors(feeds, fetches, targets, &ek));
+  std::lock_guard<std::mutex> l(mu_);
+  auto& slot = executors_[key];
+  if (!slot) slot = std::move(ek);
+  *out = slot.get();
+  return Status::OK();
+}
+
+Status DirectSession::BuildExecutors(const std::vector<std::string>& feeds,
+                                     const std::vector<std::string>& fetches,
+                                     const std::vector<std::string>& targets,
+                                     std::unique_ptr<ExecutorsAndKeys>* out) {
+  GraphDef gdef_copy;
+  {
+    std::lock_guard<std::mutex> l(mu_);
+    gdef_copy = graph_def_;
+  }
+  auto graph = std::make_unique<Graph>();
+  STF_RETURN_IF_ERROR(ConvertGraphDefToGraph(gdef_copy, graph.get()));
+  auto ek = std::make_unique<ExecutorsAndKeys>();
+
+  // ---- 1. Placement (before rewrite so feed recvs inherit devices). ----
+  std::vector<Node*> order;
+  STF_RETURN_IF_ERROR(TopologicalOrder(*graph, &order));
+  Device* cpu = devices_.LookUp("/cpu:0");
+  Device* gpu0 = nullptr;
+  for (auto& d : devices_.devices())
+    if (d->is_gpu() && !gpu0) gpu0 = d.get();
+
+  auto kernel_available = [&](Node* n, Device* d) {
+    if (n->IsControlFlow() || n->IsSend() || n->IsRecv() || n->op() == "NoOp")
+      return true;
+    return KernelRegistry::Global()->HasKernel(n->def, d->device_type());
+  };
+  for (Node* n : order) {
+    std::string req = CanonicalDevice(n->def.device);
+    Device* dev = nullptr;
+    if (!req.empty()) {
+      dev = devices_.LookUp(req);
+      if (!dev && StrStartsWith(req, "GPU") && gpu0) dev = gpu0;
+      if (!dev) dev = cpu;  // soft placement
+      if (!kernel_available(n, dev)) dev = cpu;
+    } else {
+      // Colocate with first data input when it exists (keeps variables and
+      // their updates together); else default to GPU when a kernel exists.
+      const Edge* in0 = n->input_edge(0);
+      if (in0 && !in0->src->assigned_device.empty()) {
+        dev = devices_.LookUp(in0->src->assigned_device);
+        if (dev && !kernel_available(n, dev)) dev = nullptr;
+      }
+      if (!dev) {
+        if (gpu0 && kernel_available(n, gpu0)) dev = gpu0;
+        else dev = cpu;
+      }
+    }
+    if (!kernel_available(n, dev))
+      return errors::NotFound("No kernel for op ", n->op(), " (node ",
+                              n->name(), ") on any device");
+    n->assigned_device = CanonicalDevice(dev->name());
+  }
+
+  // ---- 2. Feed rewrite: consumers of fed tensors read a client _Recv. ----
+  std::map<std::string, Node*> feed_nodes;  // "name:port" -> recv node
+  for (auto& f : feeds) {
+    auto [name, port] = ParseTensorName(f);
+    Node* n = graph->FindNode(name);
+    if (!n) return errors::NotFound("Feed node not found: ", name);
+    std::string fkey = name + ":" + std::to_string(port);
+    if (feed_nodes.count(fkey)) continue;
+    std::string dev = n->assigned_device;
+    std::string node_name = "_feed_" + name + "_" + std::to_string(port);
+    NodeDef rd = MakeSendRecv("_Recv", node_name, n->out_types[port],
+                              "feed:" + fkey, "client", dev);
+    Node* recv;
+    STF_RETURN_IF_ERROR(graph->AddNode(rd, &recv));
+    recv->assigned_device = dev;
+    ek->feed_devices["feed:" + fkey] = dev;
+    std::vector<Edge*> to_rewire;
+    for (auto* e : n->out_edges)
+      if (!e->IsControl() && e->src_output == port) to_rewire.push_back(e);
+    for (auto* e : to_rewire) {
+      Node* dst = e->dst;
+      int slot = e->dst_input;
+      graph->RemoveEdge(e);
+      graph->AddEdge(recv, 0, dst, slot);
+    }
+    feed_nodes[fkey] = recv;
+  }
+
+  // ---- 3. Fetch rewrite: _Send each fetched tensor to the client. ----
+  std::vector<Node*> keep;
+  for (auto& f : fetches) {
+    auto [name, port] = ParseTensorName(f);
+    std::string fkey = name + ":" + std::to_string(port);
+    Node* src;
+    int src_port;
+    if (feed_nodes.count(fkey)) {
+      src = feed_nodes[fkey];
+      src_port = 0;
+    } else {
+      src = graph->FindNode(name);
+      if (!src) return errors::NotFound("Fetch node not found: ", name);
+      src_port = port;
+      if (src_port >= src->num_outputs())
+        return errors::InvalidArgument("Fetch ", f, ": port out of range");
+    }
+    std::string dev = src->assigned_device;
+    NodeDef sd = MakeSendRecv("_Send",
+                              "_fetch_" + name + "_" + std::to_string(port) +
+                                  "_" + std::to_string(keep.size()),
+                              src->out_types[src_port], "fetch:" + fkey, dev,
+                              "client");
+    Node* send;
+    STF_RETURN_IF_ERROR(graph->AddNode(sd, &send));
+    send->assigned_device = dev;
+    graph->AddEdge(src, src_port, send, 0);
+    ek->fetch_devices["fetch:" + fkey] = dev;
+    keep.push_back(send);
+  }
+  for (auto& t : targets) {
+    Node* n = graph->FindNode(ParseTensorName(t).first);
+    if (!n) return errors::NotFound("Target node not found: ", t);
+    keep.push_back(n);
+  }
+
+  // ---- 4. Prune to nodes reachable from keep. ----
+  std::set<Node*> reachable;
+  std::vector<Node*> stack(keep);
+  while (!stack.empty()) {
+    Node* n = stack.back();
+    stack.pop_back();
+    if (!reachable.insert(n).second) continue;
+    for (auto* e : n->in_edges) stack.push_back(e->src);
+  }
+  std::vector<Node*> to_remove;
+  for (Node* n : graph->nodes())
+    if (!reachable.count(n)) to_remove.push_back(n);
+  for (Node* n : to_remove) graph->RemoveNode(n);
+
+  // ---- 5. Partition by device, inserting _Send/_Recv pairs. ----
+  std::set<std::string> part_devices;
+  for (Node* n : graph->nodes()) part_devices.insert(n->assigned_device);
+
+  // Build per-partition GraphDefs.
+  std::map<std::string, GraphDef> parts;
+  std::map<std::string, std::string> send_dedup;  // src:port|dstdev -> recv name
+  int edge_id = 0;
+  for (auto& dev : part_devices) parts[dev];  // create empty
+
+  // Emit nodes.
+  for (Node* n : graph->nodes()) {
+    NodeDef d = n->def;
+    d.device = n->assigned_device;
+    d.input.clear();
+    parts[n->assigned_device].node.push_back(d);
+  }
+  // Index of nodedef per partition for appending inputs. Indices (not
+  // pointers): the node vectors grow while we insert send/recv pairs.
+  std::map<std::string, size_t> defs;
+  for (auto& kv : parts)
+    for (size_t i = 0; i < kv.second.node.size(); ++i)
+      defs[kv.first + "|" + kv.second.node[i].name] = i;
+
+  auto add_input = [&](const std::string& dev, const std::string& node,
+                       const std::string& input) {
+    parts[dev].node[defs[dev + "|" + node]].input.push_back(input);
+  };
+
+  for (Node* n : graph->nodes()) {
+    // Order data inputs by slot, then controls (NodeDef convention).
+    std::vector<const Edge*> data(n->num_inputs(), nullptr);
+    std::vector<const Edge*> ctrl;
+    for (auto* e : n->in_edges) {
+      if (e->IsControl()) ctrl.push_back(e);
+      else data[e->dst_input] = e;
+    }
+    const std::string& ddev = n->assigned_device;
+    for (auto* e : data) {
+      if (!e) return errors::Internal("missing input edge on ", n->name());
+      Node* s = e->src;
+      if (s->assigned_device == ddev) {
+        add_input(ddev, n->name(),
+                  e->src_output == 0
+                      ? s->name()
+                      : s->name() + ":" + std::to_string(e->src_output));
+      } else {
+        std::string dkey = s->name() + ":" + std::to_string(e->src_output) +
+                           "|" + ddev;
+        auto it = send_dedup.find(dkey);
+        std::string recv_name;
+        if (it != send_dedup.end()) {
+          recv_name = it->second;
+        } else {
+          std::string tname = "e" + std::to_string(edge_id++) + "_" + s->name();
+          DataType dt = s->out_types[e->src_output];
+          NodeDef sd = MakeSendRecv("_Send", "_s_" + tname, dt, tname,
+                                    s->assigned_device, ddev);
+          sd.device = s->assigned_device;
+          sd.input.push_back(e->src_output == 0
+                                 ? s->name()
+                                 : s->name() + ":" +
+                                       std::to_string(e->src_output));
+          parts[s->assigned_device].node.push_back(sd);
+          NodeDef rd = MakeSendRecv("_Recv", "_r_" + tname, dt, tname,
+                                    s->assigned_device, ddev);
+          rd.device = ddev;
+          parts[ddev].node.push_back(rd);
+          recv_name = rd.name;
+          send_dedup[dkey] = recv_name;
+        }
+        add_input(ddev, n->name(), recv_name);
+      }
+    }
+    for (auto* e : ctrl) {
+      Node* s = e->src;
+      if (s->assigned_device == ddev) {
+        add_input(ddev, n->name(), "^" + s->name());
+      } else {
+        std::string dkey = s->name() + ":ctrl|" + ddev;
+        auto it = send_dedup.find(dkey);
+        std::string recv_name;
+        if (it != send_dedup.end()) {
+          recv_name = it->second;
+        } else {
+          std::string tname = "c" + std::to_string(edge_id++) + "_" + s->name();
+          // Dummy const carrying the control signal.
+          NodeDef cd;
+          cd.name = "_c_" + tname;
+          cd.op = "Const";
+          cd.attr["dtype"] = AttrValue::Type(DT_BOOL);
+          TensorProto tp;
+          tp.dtype = DT_BOOL;
+          tp.has_shape = true;
+          tp.bool_val.push_back(0);
+          AttrValue av;
+          av.kind = 'e';
+          av.tensor = tp;
+          cd.attr["value"] = av;
+          cd.device = s->assigned_device;
+          cd.input.push_back("^" + s->name());
+          parts[s->assigned_device].node.push_back(cd);
+          NodeDef sd = MakeSendRecv("_Send", "_s_" + tname, DT_BOOL, tname,
+                                    s->assigned_device, ddev);
+          sd.device = s->assigned_device;
+          sd.input.push_back(cd.name);
+          parts[s->assigned_device].node.push_back(sd);
+          NodeDef rd = MakeSendRecv("_Recv", "_r_" + tname, DT_BOOL, tname,
+                                    s->assigned_device, ddev);
+          rd.device = ddev;
+          parts[ddev].node.push_back(rd);
+          recv_name = rd.name;
+          send_dedup[dkey] = recv_name;
+        }
+        add_input(ddev, n->name(), "^" + recv_name);
+      }
+    }
+  }
+
+  // ---- 6. Build one executor per partition. ----
+  for (auto& kv : parts) {
+    Device* dev = devices_.LookUp(kv.first);
+    if (!dev) return errors::Internal("Unknown partition device ", kv.first);
+    auto pg = std::make_unique<Graph>();
+    STF_RETURN_IF_ERROR(ConvertGraphDefToGraph(kv.second, pg.get()));
+    for (Node* n : pg->nodes()) n->assigned_device = kv.first;
+    ExecutorsAndKeys::Item item;
+    item.device = dev;
+    STF_RETURN_IF_ERROR(
+        Executor::Create(std::move(pg), dev, &opseg_, &item.executor));
+    ek->items.push_back(std::move(item));
+  }
+  *out = std::move(ek);
+  return Status::OK();
+}
+
+Status DirectSession::Run(
+    const std::vector<std::pair<std::string, Tensor>>& feeds,
+    const std::vector<std::string>& fetches,
+    const std::vector<std::string>& targets, std::vector<Tensor>* outputs) {
+  std::vector<std::string> feed_names;
+  for (auto& f : feeds) feed_names.push_back(f.first);
+  ExecutorsAndKeys* ek = nullptr;
+  STF_RETURN_IF_ERROR(GetOrCreateExecutors(feed_names, fetches, targets, &ek));
+
+  Rendezvous rendez;
+  // Send feeds.
+  for (auto& f : feeds) {
+    auto [name, port] = ParseTensorName(f.first);
+    std::string fkey = "feed:" + name + ":" + std::to_string(port);
+    auto it = ek->feed_devices.find(fkey);
+    if (it == ek->feed_devices.end())
+      return errors::Internal("feed not wired: ", f.first);
+    STF_RETURN_IF_ERROR(rendez.Send(
+        RendezvousKey("client", it->second, fkey, "", 0), f.second, false));
+  }
+
+  // Run all partition executors.
+  int64_t step_id;
+  {
+    std::lock_guard<std::mutex> l(mu_);
+    step_id = ++step_counter_;
+  }
+  std::mutex mu;
+  std::condition_variable cv;
+  int remaining = (int)ek->items.size();
+  Status agg;
+  for (auto& item : ek->items) {
+    ExecutorArgs args;
+    args.step_id = step_id;
+    args.rendezvous = &rendez;
+    args.pool = pool_.get();
+    args.resource_mgr = resource_mgr_;
+    item.executor->RunAsync(args, [&](Status s) {
+      std::lock_guard<std::mutex> l(mu);
+      if (!s.ok() && agg.ok()) agg = s;
+      if (--remaining == 0) cv.notify_one();
+    });
+  }
+
+  // Collect fetches (they arrive as the graph runs).
+  outputs->clear();
+  Status fetch_status;
+  for (auto& f : fetches) {
+    auto [name, port] = ParseTensorName(f);
+    std::string fkey = "fetch:" + name + ":" + std::to_string(port);
+    auto it = ek->fetch_devices.find(fkey);
+    if (it == ek->fetch_devices.end()) {
+      fetch_status = errors::Internal("fetch not wired: ", f);
+      break;
+    }
+    Tensor val;
+    bool is_dead = false;
+    Status s =
+        rendez.Recv(RendezvousKey(it->second, "client", fkey, "", 0), &val,
+                    &is_dead);
+    if (!s.ok()) {
+      fetch_status = s;
+      break;
+    }
+    if (val.IsInitialized() && val.mem_space() == MemSpace::DEVICE) {
+      Device* dev = devices_.LookUp(it->second);
+      Tensor host;
+      Status cs = dev->CopyDeviceTensorToHost(val, &host);
+      if (!cs.ok()) {
+        fetch_status = cs;
+        break;
+      }
+      val = host;
+    }
+    outputs->push_back(val);
+  }
+
+  {
+    std::unique_lock<std::mutex> l(mu);
+    cv.wait(l, [&]() { return remaining == 0; });
+  }
+  if (!agg.ok()) return agg;
+  return fetch_status;
+}
+
+}  // namespace stf

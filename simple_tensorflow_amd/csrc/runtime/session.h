@@ -1,0 +1,73 @@
+// DirectSession: the single-process session runtime.
+// Capability analog of the reference's DirectSession
+// (common_runtime/direct_session.cc) + SimpleGraphExecutionState +
+// SimplePlacer + graph partitioner: owns the full graph, rewrites
+// feeds/fetches to client _Recv/_Send through a per-step rendezvous, places
+// nodes on CPU/GPU, partitions by device inserting _Send/_Recv pairs, builds
+// one dataflow executor per device partition and caches executors by the
+// (feeds, fetches, targets) signature.
+#pragma once
+
+#include <map>
+#include <memory>
+#include <mutex>
+#include <string>
+#include <vector>
+
+#include "core/threadpool.h"
+#include "framework/device.h"
+#include "runtime/executor.h"
+
+namespace stf {
+
+class DirectSession {
+ public:
+  // Creates the session with all visible devices (CPU + any HIP GPUs).
+  // `force_cpu_only` ignores GPUs (used in tests).
+  explicit DirectSession(bool force_cpu_only = false, int num_threads = 0);
+  ~DirectSession();
+
+  Status Create(const GraphDef& def);
+  Status Extend(const GraphDef& def);
+
+  Status Run(const std::vector<std::pair<std::string, Tensor>>& feeds,
+             const std::vector<std::string>& fetches,
+             const std::vector<std::string>& targets,
+             std::vector<Tensor>* outputs);
+
+  DeviceMgr* device_mgr() { return &devices_; }
+  void* resource_mgr() { return resource_mgr_; }
+
+ private:
+  struct ExecutorsAndKeys {
+    struct Item {
+      Device* device;
+      std::unique_ptr<Executor> executor;
+    };
+    std::vector<Item> items;
+    // feed key -> (recv_device name); fetch key -> send_device name
+    std::map<std::string, std::string> feed_devices;
+    std::map<std::string, std::string> fetch_devices;
+  };
+
+  Status GetOrCreateExecutors(
+      const std::vector<std::string>& feeds,
+      const std::vector<std::string>& fetches,
+      const std::vector<std::string>& targets, ExecutorsAndKeys** out);
+
+  Status BuildExecutors(const std::vector<std::string>& feeds,
+                        const std::vector<std::string>& fetches,
+                        const std::vector<std::string>& targets,
+                        std::unique_ptr<ExecutorsAndKeys>* out);
+
+  std::mutex mu_;
+  GraphDef graph_def_;
+  DeviceMgr devices_;
+  OpSegment opseg_;
+  std::unique_ptr<ThreadPool> pool_;
+  std::map<std::string, std::unique_ptr<ExecutorsAndKeys>> executors_;
+  int64_t step_counter_ = 0;
+  void* resource_mgr_ = nullptr;  // owned; see kernels/resource_mgr
+};
+
+}  // namespace stf

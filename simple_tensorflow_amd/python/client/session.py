@@ -1,0 +1,160 @@
+"""tf.Session — client for the C++ DirectSession.
+
+Analog of the reference's python/client/session.py (BaseSession.run:660),
+speaking to csrc/runtime/session.cc through the pybind module.
+"""
+import threading
+
+import numpy as np
+
+from simple_tensorflow_amd import _core
+from simple_tensorflow_amd.python.framework import dtypes, ops
+
+_default_session_stack = threading.local()
+
+
+def get_default_session():
+    stack = getattr(_default_session_stack, 'stack', [])
+    if not stack:
+        raise RuntimeError('No default session')
+    return stack[-1]
+
+
+class Session(object):
+    def __init__(self, target='', graph=None, config=None):
+        self._graph = graph if graph is not None else ops.get_default_graph()
+        cpu_only = False
+        if config is not None and isinstance(config, dict):
+            cpu_only = config.get('device_count', {}).get('GPU', 1) == 0
+        self._core = _core.Session(cpu_only)
+        self._created = False
+        self._serialized_nodes = 0
+        self._lock = threading.Lock()
+
+    @property
+    def graph(self):
+        return self._graph
+
+    def _sync_graph(self):
+        g = self._graph
+        with self._lock:
+            if not self._created or g._mutated_after_serialize:
+                self._core.create(g.as_graph_def(0))
+                self._created = True
+                g._mutated_after_serialize = False
+                self._serialized_nodes = len(g._node_list)
+            elif len(g._node_list) > self._serialized_nodes:
+                self._core.extend(g.as_graph_def(self._serialized_nodes))
+                self._serialized_nodes = len(g._node_list)
+
+    def run(self, fetches, feed_dict=None, options=None, run_metadata=None):
+        self._sync_graph()
+        flat, restore = _flatten_fetches(fetches)
+        fetch_names = []
+        targets = []
+        fetch_slots = []  # index into results for each flat fetch, or None
+        for f in flat:
+            if isinstance(f, ops.Operation):
+                targets.append(f.name)
+                fetch_slots.append(None)
+            else:
+                t = _as_fetchable(f, self._graph)
+                if isinstance(t, ops.Operation):
+                    targets.append(t.name)
+                    fetch_slots.append(None)
+                else:
+                    fetch_slots.append(len(fetch_names))
+                    fetch_names.append(t.name)
+
+        feeds = {}
+        if feed_dict:
+            for k, v in feed_dict.items():
+                t = _as_fetchable(k, self._graph)
+                feeds[t.name] = _convert_feed(t, v)
+
+        results = self._core.run(feeds, fetch_names, targets)
+        out = []
+        for f, slot in zip(flat, fetch_slots):
+            if slot is None:
+                out.append(None)
+            else:
+                val = results[slot]
+                t = f if isinstance(f, ops.Tensor) else None
+                if t is not None and t._shape is not None and \
+                        isinstance(val, np.ndarray) and val.ndim == 0:
+                    pass
+                out.append(val)
+        return restore(out)
+
+    def close(self):
+        pass
+
+    def __enter__(self):
+        stack = getattr(_default_session_stack, 'stack', None)
+        if stack is None:
+            _default_session_stack.stack = stack = []
+        stack.append(self)
+        return self
+
+    def __exit__(self, *exc):
+        _default_session_stack.stack.pop()
+        return False
+
+    def as_default(self):
+        return self
+
+    def num_gpus(self):
+        return self._core.num_gpus()
+
+
+class InteractiveSession(Session):
+    def __init__(self, *a, **kw):
+        super().__init__(*a, **kw)
+        stack = getattr(_default_session_stack, 'stack', None)
+        if stack is None:
+            _default_session_stack.stack = stack = []
+        stack.append(self)
+
+
+def _as_fetchable(f, graph):
+    if isinstance(f, (ops.Tensor, ops.Operation)):
+        return f
+    if isinstance(f, str):
+        if ':' in f:
+            return graph.get_tensor_by_name(f)
+        return graph.get_operation_by_name(f)
+    if hasattr(f, '_as_graph_element'):
+        return f._as_graph_element()
+    raise TypeError('Cannot fetch %r' % (f,))
+
+
+def _flatten_fetches(fetches):
+    if isinstance(fetches, (list, tuple)):
+        items = list(fetches)
+        def restore(vals):
+            return type(fetches)(vals) if isinstance(fetches, tuple) else vals
+        return items, restore
+    if isinstance(fetches, dict):
+        keys = list(fetches.keys())
+        items = [fetches[k] for k in keys]
+        def restore(vals):
+            return dict(zip(keys, vals))
+        return items, restore
+    return [fetches], lambda vals: vals[0]
+
+
+def _convert_feed(t, v):
+    if t.dtype is dtypes.string:
+        if isinstance(v, str):
+            return v.encode()
+        if isinstance(v, bytes):
+            return v
+        raise TypeError('string feed must be str/bytes')
+    np_dt = t.dtype.as_numpy_dtype
+    arr = np.asarray(v)
+    if t.dtype is dtypes.bfloat16:
+        arr32 = arr.astype(np.float32)
+        return ops._f32_to_bf16(arr32)
+    if arr.dtype != np_dt:
+        arr = arr.astype(np_dt)
+    return np.ascontiguousarray(arr)

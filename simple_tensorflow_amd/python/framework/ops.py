@@ -1,0 +1,750 @@
+"""Graph / Operation / Tensor and generic op application.
+
+Capability analog of the reference's python/framework/ops.py (Tensor:196,
+Operation:1117, Graph:1891) + op_def_library.py (apply_op:289), compacted:
+op wrappers call `apply_op` directly against the C++ op registry exposed by
+the pybind module, and NodeDefs are serialized with pbwire.py.
+"""
+import contextlib
+import threading
+
+import numpy as np
+
+from simple_tensorflow_amd import _core
+from simple_tensorflow_amd.python.framework import dtypes, pbwire
+
+_OP_DEFS = None
+
+
+def op_defs():
+    global _OP_DEFS
+    if _OP_DEFS is None:
+        _OP_DEFS = _core.list_ops()
+    return _OP_DEFS
+
+
+class GraphKeys(object):
+    GLOBAL_VARIABLES = 'variables'
+    TRAINABLE_VARIABLES = 'trainable_variables'
+    LOCAL_VARIABLES = 'local_variables'
+    SUMMARIES = 'summaries'
+    QUEUE_RUNNERS = 'queue_runners'
+    UPDATE_OPS = 'update_ops'
+    REGULARIZATION_LOSSES = 'regularization_losses'
+    MOVING_AVERAGE_VARIABLES = 'moving_average_variables'
+    GLOBAL_STEP = 'global_step'
+    SAVERS = 'savers'
+    INIT_OP = 'init_op'
+    LOSSES = 'losses'
+    VARIABLES = 'variables'  # legacy alias
+
+
+class Tensor(object):
+    def __init__(self, op, index, dtype, is_ref=False):
+        self.op = op
+        self.value_index = index
+        self._dtype = dtypes.as_dtype(dtype)
+        self._shape = None  # tuple with None for unknown, or None = unknown rank
+        self._is_ref = is_ref
+
+    @property
+    def dtype(self):
+        return self._dtype
+
+    @property
+    def graph(self):
+        return self.op.graph
+
+    @property
+    def name(self):
+        return '%s:%d' % (self.op.name, self.value_index)
+
+    @property
+    def shape(self):
+        return TensorShape(self._shape)
+
+    def get_shape(self):
+        return self.shape
+
+    def set_shape(self, shape):
+        if isinstance(shape, TensorShape):
+            shape = shape.dims_tuple()
+        if shape is not None:
+            shape = tuple(shape)
+        self._shape = shape
+
+    def eval(self, feed_dict=None, session=None):
+        from simple_tensorflow_amd.python.client import session as sess_mod
+        s = session or sess_mod.get_default_session()
+        return s.run(self, feed_dict=feed_dict)
+
+    def consumers(self):
+        return [op for op in self.graph._node_list
+                if any(t is self for t in op.inputs)]
+
+    # operator sugar (delegates to math_ops; late import to avoid cycles)
+    def _binop(self, other, fn, reverse=False):
+        from simple_tensorflow_amd.python.ops import math_ops
+        f = getattr(math_ops, fn)
+        return f(other, self) if reverse else f(self, other)
+
+    def __add__(self, o): return self._binop(o, 'add')
+    def __radd__(self, o): return self._binop(o, 'add', True)
+    def __sub__(self, o): return self._binop(o, 'subtract')
+    def __rsub__(self, o): return self._binop(o, 'subtract', True)
+    def __mul__(self, o): return self._binop(o, 'multiply')
+    def __rmul__(self, o): return self._binop(o, 'multiply', True)
+    def __truediv__(self, o): return self._binop(o, 'divide')
+    def __rtruediv__(self, o): return self._binop(o, 'divide', True)
+    def __pow__(self, o): return self._binop(o, 'pow')
+    def __neg__(self):
+        from simple_tensorflow_amd.python.ops import math_ops
+        return math_ops.negative(self)
+    def __lt__(self, o): return self._binop(o, 'less')
+    def __le__(self, o): return self._binop(o, 'less_equal')
+    def __gt__(self, o): return self._binop(o, 'greater')
+    def __ge__(self, o): return self._binop(o, 'greater_equal')
+    def __getitem__(self, key):
+        from simple_tensorflow_amd.python.ops import array_ops
+        return array_ops._slice_helper(self, key)
+    def __iter__(self):
+        raise TypeError('Tensor objects are not iterable; use tf.unstack')
+
+    def __hash__(self):
+        return id(self)
+
+    def __repr__(self):
+        return "<Tensor '%s' shape=%s dtype=%s>" % (
+            self.name, self._shape, self._dtype.name)
+
+
+class TensorShape(object):
+    """Static shape: tuple of dims (None = unknown dim) or None = unknown."""
+
+    def __init__(self, dims):
+        if dims is None:
+            self._dims = None
+        elif isinstance(dims, TensorShape):
+            self._dims = dims._dims
+        else:
+            self._dims = tuple(int(d) if d is not None and int(d) >= 0 else None
+                               for d in dims)
+
+    @property
+    def ndims(self):
+        return None if self._dims is None else len(self._dims)
+
+    @property
+    def dims(self):
+        return None if self._dims is None else list(self._dims)
+
+    def dims_tuple(self):
+        return self._dims
+
+    def as_list(self):
+        if self._dims is None:
+            raise ValueError('Shape has unknown rank')
+        return list(self._dims)
+
+    def num_elements(self):
+        if self._dims is None or any(d is None for d in self._dims):
+            return None
+        n = 1
+        for d in self._dims:
+            n *= d
+        return n
+
+    def is_fully_defined(self):
+        return self._dims is not None and all(d is not None for d in self._dims)
+
+    def merge_with(self, other):
+        other = TensorShape(other) if not isinstance(other, TensorShape) else other
+        if self._dims is None:
+            return other
+        if other._dims is None:
+            return self
+        merged = []
+        for a, b in zip(self._dims, other._dims):
+            merged.append(a if b is None else b)
+        return TensorShape(merged)
+
+    def __getitem__(self, i):
+        if self._dims is None:
+            return None
+        d = self._dims[i]
+        return d
+
+    def __len__(self):
+        if self._dims is None:
+            raise ValueError('unknown rank')
+        return len(self._dims)
+
+    def __iter__(self):
+        return iter(self.as_list())
+
+    def __eq__(self, other):
+        return TensorShape(other)._dims == self._dims
+
+    def __repr__(self):
+        return 'TensorShape(%r)' % (self._dims,)
+
+
+class Operation(object):
+    def __init__(self, graph, node_name, op_type, inputs, control_inputs,
+                 attrs, device, output_dtypes, output_is_ref):
+        self.graph = graph
+        self.name = node_name
+        self.type = op_type
+        self.inputs = list(inputs)
+        self.control_inputs = list(control_inputs)
+        self.attrs = dict(attrs)  # name -> (kind, value)
+        self.device = device
+        self.outputs = [Tensor(self, i, dt, ref)
+                        for i, (dt, ref) in enumerate(zip(output_dtypes,
+                                                          output_is_ref))]
+
+    def get_attr(self, name):
+        if name in self.attrs:
+            k, v = self.attrs[name]
+            if k == 'type':
+                return dtypes.as_dtype(v)
+            return v
+        od = op_defs().get(self.type)
+        if od:
+            for a in od['attr']:
+                if a['name'] == name and a['has_default']:
+                    d = a['default']
+                    kind = d['kind']
+                    if kind == 't':
+                        return dtypes.as_dtype(d['value'])
+                    if kind == 'l':
+                        lv = d['value']
+                        for key in ('i', 'f', 's', 'b', 'type'):
+                            if lv.get(key):
+                                return list(lv[key])
+                        return []
+                    if kind == 's':
+                        v = d['value']
+                        return v.decode() if isinstance(v, bytes) else v
+                    if kind == '0':
+                        return None
+                    return d['value']
+        raise ValueError('No attr %s on op %s' % (name, self.name))
+
+    def _add_control_input(self, op):
+        if op not in self.control_inputs:
+            self.control_inputs.append(op)
+            self.graph._bump_version(self)
+
+    def run(self, feed_dict=None, session=None):
+        from simple_tensorflow_amd.python.client import session as sess_mod
+        s = session or sess_mod.get_default_session()
+        s.run(self, feed_dict=feed_dict)
+
+    def node_def_bytes(self):
+        input_strs = []
+        for t in self.inputs:
+            if t.value_index == 0:
+                input_strs.append(t.op.name)
+            else:
+                input_strs.append('%s:%d' % (t.op.name, t.value_index))
+        for c in self.control_inputs:
+            input_strs.append('^' + c.name)
+        wire_attrs = {}
+        for k, (kind, v) in self.attrs.items():
+            wire_attrs[k] = (kind, v)
+        return pbwire.node_def(self.name, self.type, input_strs, self.device,
+                               wire_attrs)
+
+    def __repr__(self):
+        return "<Operation '%s' type=%s>" % (self.name, self.type)
+
+
+class Graph(object):
+    def __init__(self):
+        self._nodes_by_name = {}
+        self._node_list = []
+        self._names_used = {}
+        self._name_stack = ''
+        self._device_stack = []
+        self._control_deps_stack = []
+        self._collections = {}
+        self._while_ctx_stack = []  # active while-loop capture contexts
+        self._lock = threading.Lock()
+        self.version = 0
+        self._mutated_after_serialize = False
+        self.seed = None
+        self._finalized = False
+
+    # ---- naming ----
+    def unique_name(self, name, mark_as_used=True):
+        full = self._name_stack + name if not self._name_stack else (
+            self._name_stack + '/' + name)
+        base = full
+        i = self._names_used.get(base, 0)
+        if mark_as_used:
+            self._names_used[base] = i + 1
+        if i > 0:
+            full = '%s_%d' % (base, i)
+        return full
+
+    @contextlib.contextmanager
+    def name_scope(self, name):
+        old = self._name_stack
+        if name:
+            if name.endswith('/'):
+                self._name_stack = name[:-1]
+            else:
+                scope = self.unique_name(name)
+                self._name_stack = scope
+        else:
+            self._name_stack = ''
+        try:
+            yield self._name_stack + '/' if self._name_stack else ''
+        finally:
+            self._name_stack = old
+
+    @contextlib.contextmanager
+    def device(self, dev):
+        self._device_stack.append(dev)
+        try:
+            yield
+        finally:
+            self._device_stack.pop()
+
+    @contextlib.contextmanager
+    def control_dependencies(self, ops_or_tensors):
+        if ops_or_tensors is None:
+            self._control_deps_stack.append(None)  # clear
+        else:
+            deps = []
+            for x in ops_or_tensors:
+                deps.append(x.op if isinstance(x, Tensor) else x)
+            self._control_deps_stack.append(deps)
+        try:
+            yield
+        finally:
+            self._control_deps_stack.pop()
+
+    def _current_control_deps(self):
+        deps = []
+        for frame in self._control_deps_stack:
+            if frame is None:
+                deps = []
+            else:
+                deps.extend(frame)
+        return deps
+
+    def _current_device(self):
+        return self._device_stack[-1] if self._device_stack else ''
+
+    # ---- op creation ----
+    def create_op(self, op_type, inputs, output_dtypes, attrs=None, name=None,
+                  output_is_ref=None, control_inputs=None, device=None):
+        if self._finalized:
+            raise RuntimeError('Graph is finalized and cannot be modified.')
+        attrs = attrs or {}
+        name = name or op_type
+        if name.endswith('/'):
+            name = name[:-1]
+        node_name = self.unique_name(name)
+        if output_is_ref is None:
+            output_is_ref = [False] * len(output_dtypes)
+        # While-loop capture: an op built inside a while context must not read
+        # tensors from outside the frame directly — route them through a
+        # constant Enter (the reference's WhileContext.AddValue).
+        if self._while_ctx_stack and op_type not in (
+                'Enter', 'Exit', 'NextIteration', 'Merge'):
+            ctx = self._while_ctx_stack[-1]
+            inputs = [ctx.capture(t) for t in inputs]
+        deps = list(control_inputs or []) + [
+            d for d in self._current_control_deps()]
+        # drop control deps that are already data inputs
+        data_ops = {t.op for t in inputs}
+        deps = [d for d in dict.fromkeys(deps) if d not in data_ops]
+        op = Operation(self, node_name, op_type, inputs, deps, attrs,
+                       device if device is not None else self._current_device(),
+                       output_dtypes, output_is_ref)
+        with self._lock:
+            self._nodes_by_name[node_name] = op
+            self._node_list.append(op)
+            self.version += 1
+        if self._while_ctx_stack and inputs:
+            # Zero-input ops (Const etc.) execute in the ROOT frame; their
+            # consumers inside the loop capture them via a constant Enter.
+            self._while_ctx_stack[-1].internal.update(op.outputs)
+        _infer_shapes(op)
+        return op
+
+    def _bump_version(self, op):
+        # mutation of an existing node (e.g. late control edge): force full
+        # re-serialization on next run
+        self._mutated_after_serialize = True
+        self.version += 1
+
+    def get_operation_by_name(self, name):
+        return self._nodes_by_name[name]
+
+    def get_tensor_by_name(self, name):
+        base, _, idx = name.partition(':')
+        return self._nodes_by_name[base].outputs[int(idx or 0)]
+
+    def as_graph_def(self, from_version=0):
+        nodes = [op.node_def_bytes() for op in self._node_list[from_version:]]
+        return pbwire.graph_def(nodes)
+
+    def finalize(self):
+        self._finalized = True
+
+    # ---- collections ----
+    def add_to_collection(self, name, value):
+        self._collections.setdefault(name, []).append(value)
+
+    def get_collection(self, name, scope=None):
+        items = list(self._collections.get(name, []))
+        if scope:
+            items = [x for x in items
+                     if getattr(x, 'name', '').startswith(scope)]
+        return items
+
+    def get_collection_ref(self, name):
+        return self._collections.setdefault(name, [])
+
+    @contextlib.contextmanager
+    def as_default(self):
+        old = _default_graph_stack.stack[-1] if _default_graph_stack.stack else None
+        _default_graph_stack.stack.append(self)
+        try:
+            yield self
+        finally:
+            _default_graph_stack.stack.pop()
+
+
+class _DefaultGraphStack(threading.local):
+    def __init__(self):
+        self.stack = []
+
+
+_default_graph_stack = _DefaultGraphStack()
+_global_default_graph = None
+
+
+def get_default_graph():
+    global _global_default_graph
+    if _default_graph_stack.stack:
+        return _default_graph_stack.stack[-1]
+    if _global_default_graph is None:
+        _global_default_graph = Graph()
+    return _global_default_graph
+
+
+def reset_default_graph():
+    global _global_default_graph
+    _global_default_graph = Graph()
+    return _global_default_graph
+
+
+def name_scope(name, default_name=None, values=None):
+    g = get_default_graph()
+    return g.name_scope(name if name is not None else (default_name or ''))
+
+
+def device(dev):
+    return get_default_graph().device(dev)
+
+
+def control_dependencies(deps):
+    return get_default_graph().control_dependencies(deps)
+
+
+@contextlib.contextmanager
+def colocate_with(op, ignore_existing=False):
+    # Placement heuristics in the C++ placer colocate ref consumers already.
+    yield
+
+
+# ---------------------------------------------------------------------------
+# convert_to_tensor / constants
+# ---------------------------------------------------------------------------
+def _np_for_dtype(dt):
+    return dtypes.as_dtype(dt).as_numpy_dtype
+
+
+def constant(value, dtype=None, shape=None, name='Const'):
+    g = get_default_graph()
+    if isinstance(value, Tensor):
+        return value
+    if dtype is not None:
+        dtype = dtypes.as_dtype(dtype)
+    if dtype is dtypes.string or (
+            dtype is None and isinstance(value, (str, bytes))):
+        vals = value
+        if isinstance(vals, (str, bytes)):
+            vals = [vals]
+            dims = []
+        else:
+            dims = [len(vals)]
+        vals = [v.encode() if isinstance(v, str) else v for v in vals]
+        tp = pbwire.tensor_proto(7, dims, string_vals=vals)
+        op = g.create_op('Const', [], [dtypes.string],
+                         attrs={'dtype': ('type', 7), 'value': ('tensor', tp)},
+                         name=name)
+        op.outputs[0].set_shape(dims)
+        return op.outputs[0]
+
+    if dtype is dtypes.bfloat16:
+        arr32 = np.array(value, dtype=np.float32)
+        if shape is not None:
+            arr32 = np.broadcast_to(arr32, shape).astype(np.float32)
+        arr = _f32_to_bf16(arr32)
+        enum = 14
+    else:
+        np_dt = _np_for_dtype(dtype) if dtype is not None else None
+        arr = np.array(value, dtype=np_dt)
+        if arr.dtype == np.float64 and dtype is None:
+            arr = arr.astype(np.float32)
+        if arr.dtype == np.int64 and dtype is None:
+            arr = arr.astype(np.int32)
+        if shape is not None:
+            arr = np.broadcast_to(arr, shape).astype(arr.dtype)
+        enum = dtypes.as_dtype(arr.dtype).as_datatype_enum
+    tp = pbwire.tensor_proto(enum, list(arr.shape),
+                             content=np.ascontiguousarray(arr).tobytes())
+    op = g.create_op('Const', [], [enum],
+                     attrs={'dtype': ('type', enum), 'value': ('tensor', tp)},
+                     name=name)
+    op.outputs[0].set_shape(arr.shape)
+    if enum in (3, 9):  # int32/int64 constants feed shape computations
+        op.outputs[0]._const_value = arr
+    return op.outputs[0]
+
+
+def _f32_to_bf16(arr32):
+    bits = arr32.view(np.uint32)
+    lsb = (bits >> 16) & 1
+    rounded = bits + 0x7FFF + lsb
+    return (rounded >> 16).astype(np.uint16)
+
+
+def convert_to_tensor(value, dtype=None, name=None):
+    if isinstance(value, Tensor):
+        if dtype is not None and dtypes.as_dtype(dtype) != value.dtype:
+            raise TypeError('Tensor conversion dtype mismatch: %s vs %s for %r'
+                            % (dtypes.as_dtype(dtype).name, value.dtype.name,
+                               value))
+        return value
+    if hasattr(value, '_as_graph_element'):
+        return value._as_graph_element()
+    return constant(value, dtype=dtype, name=name or 'Const')
+
+
+# ---------------------------------------------------------------------------
+# Generic op application against the C++ OpDef registry
+# (analog of op_def_library.apply_op, reference op_def_library.py:289)
+# ---------------------------------------------------------------------------
+def apply_op(op_type, *args, **kwargs):
+    name = kwargs.pop('name', None)
+    od = op_defs().get(op_type)
+    if od is None:
+        raise ValueError('Unknown op %s' % op_type)
+    g = get_default_graph()
+
+    attrs = {}
+    inferred_types = {}
+
+    attr_defs = {a['name']: a for a in od['attr']}
+
+    # consume positional inputs per input_arg
+    args = list(args)
+    input_tensors = []
+    for arg in od['input_arg']:
+        if not args:
+            raise ValueError('%s: missing input %s' % (op_type, arg['name']))
+        val = args.pop(0)
+        if arg['number_attr']:
+            vals = list(val)
+            attrs[arg['number_attr']] = ('i', len(vals))
+            elem_dt = None
+            if arg['type_attr'] and arg['type_attr'] in inferred_types:
+                elem_dt = inferred_types[arg['type_attr']]
+            elif arg['type']:
+                elem_dt = arg['type']
+            conv = []
+            for v in vals:
+                t = convert_to_tensor(v, dtype=elem_dt)
+                if elem_dt is None:
+                    elem_dt = t.dtype.as_datatype_enum
+                conv.append(t)
+            if arg['type_attr']:
+                inferred_types.setdefault(arg['type_attr'], elem_dt)
+            input_tensors.extend(conv)
+        elif arg['type_list_attr']:
+            vals = [convert_to_tensor(v) for v in val]
+            attrs[arg['type_list_attr']] = (
+                'list', {'type': [t.dtype.as_datatype_enum for t in vals]})
+            input_tensors.extend(vals)
+        else:
+            want = None
+            if arg['type_attr'] and arg['type_attr'] in inferred_types:
+                want = inferred_types[arg['type_attr']]
+            elif arg['type']:
+                want = arg['type']
+            elif arg['type_attr'] and arg['type_attr'] in kwargs and \
+                    isinstance(kwargs.get(arg['type_attr']), (int, dtypes.DType)):
+                want = int(dtypes.as_dtype(kwargs[arg['type_attr']]))
+            if isinstance(val, Tensor):
+                t = val
+                if want is not None and t.dtype.as_datatype_enum != int(want) \
+                        and arg['type_attr'] and arg['type_attr'] not in inferred_types:
+                    pass
+            else:
+                t = convert_to_tensor(val, dtype=want)
+            if arg['type_attr']:
+                inferred_types.setdefault(arg['type_attr'],
+                                          t.dtype.as_datatype_enum)
+            input_tensors.append(t)
+
+    # explicit attrs from kwargs
+    for k, v in kwargs.items():
+        ad = attr_defs.get(k)
+        if ad is None:
+            raise ValueError('%s: unknown attr %s' % (op_type, k))
+        attrs[k] = _encode_attr(ad['type'], v)
+
+    # inferred type attrs
+    for k, v in inferred_types.items():
+        if k not in attrs:
+            attrs[k] = ('type', int(v))
+
+    # outputs
+    out_dtypes = []
+    out_ref = []
+    for arg in od['output_arg']:
+        n = 1
+        if arg['number_attr']:
+            n = attrs.get(arg['number_attr'],
+                          ('i', _attr_default(attr_defs, arg['number_attr'])))[1]
+        if arg['type_list_attr']:
+            lv = attrs[arg['type_list_attr']][1]
+            for t in lv['type']:
+                out_dtypes.append(int(t))
+                out_ref.append(arg['is_ref'])
+            continue
+        if arg['type_attr']:
+            if arg['type_attr'] in attrs:
+                dt = attrs[arg['type_attr']][1]
+            else:
+                dt = _attr_default(attr_defs, arg['type_attr'])
+                attrs[arg['type_attr']] = ('type', int(dt))
+        else:
+            dt = arg['type']
+        for _ in range(int(n)):
+            out_dtypes.append(int(dt))
+            out_ref.append(arg['is_ref'])
+
+    op = g.create_op(op_type, input_tensors, out_dtypes, attrs=attrs,
+                     name=name, output_is_ref=out_ref)
+    if len(op.outputs) == 1:
+        return op.outputs[0]
+    if not op.outputs:
+        return op
+    return tuple(op.outputs)
+
+
+def _attr_default(attr_defs, name):
+    ad = attr_defs.get(name)
+    if ad and ad['has_default']:
+        return ad['default']['value']
+    raise ValueError('attr %s has no value' % name)
+
+
+def _encode_attr(attr_type, v):
+    if attr_type == 'int':
+        return ('i', int(v))
+    if attr_type == 'float':
+        return ('f', float(v))
+    if attr_type == 'bool':
+        return ('b', True if v else False)
+    if attr_type == 'string':
+        return ('s', v if isinstance(v, (str, bytes)) else str(v))
+    if attr_type == 'type':
+        return ('type', int(dtypes.as_dtype(v)))
+    if attr_type == 'shape':
+        if isinstance(v, TensorShape):
+            v = v.dims_tuple()
+        return ('shape', None if v is None else [int(d) if d is not None else -1
+                                                 for d in v])
+    if attr_type == 'list(int)':
+        return ('list', {'i': [int(x) for x in v]})
+    if attr_type == 'list(float)':
+        return ('list', {'f': [float(x) for x in v]})
+    if attr_type == 'list(string)':
+        return ('list', {'s': list(v)})
+    if attr_type == 'list(type)':
+        return ('list', {'type': [int(dtypes.as_dtype(x)) for x in v]})
+    if attr_type == 'list(shape)':
+        return ('list', {'shape': [list(s) for s in v]})
+    raise ValueError('unsupported attr type %s' % attr_type)
+
+
+# ---------------------------------------------------------------------------
+# Gradient registry
+# ---------------------------------------------------------------------------
+_gradient_registry = {}
+
+
+class RegisterGradient(object):
+    def __init__(self, op_type):
+        self.op_type = op_type
+
+    def __call__(self, fn):
+        _gradient_registry[self.op_type] = fn
+        return fn
+
+
+def NoGradient(op_type):
+    _gradient_registry[op_type] = None
+
+
+NotDifferentiable = NoGradient
+
+
+def get_gradient_function(op_type):
+    return _gradient_registry.get(op_type, '__missing__')
+
+
+# ---------------------------------------------------------------------------
+# Static shape inference (python-side; compact)
+# ---------------------------------------------------------------------------
+_shape_fns = {}
+
+
+def RegisterShape(op_type):
+    def deco(fn):
+        _shape_fns[op_type] = fn
+        return fn
+    return deco
+
+
+def _infer_shapes(op):
+    fn = _shape_fns.get(op.type)
+    if fn is None:
+        return
+    try:
+        shapes = fn(op)
+    except Exception:
+        return
+    if shapes is None:
+        return
+    for t, s in zip(op.outputs, shapes):
+        if s is not None:
+            t.set_shape(s)
+
+
+class GraphDef(object):
+    """Tiny stand-in used where callers expect a GraphDef object."""
+    def __init__(self, data):
+        self.data = data
+
+    def SerializeToString(self):
+        return self.data

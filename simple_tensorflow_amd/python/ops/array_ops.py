@@ -1,0 +1,345 @@
+"""Array ops (analog of reference python/ops/array_ops.py)."""
+import numpy as np
+
+from simple_tensorflow_amd.python.framework import dtypes, ops
+from simple_tensorflow_amd.python.framework.ops import apply_op, convert_to_tensor
+
+
+def placeholder(dtype, shape=None, name=None):
+    dt = dtypes.as_dtype(dtype)
+    t = apply_op('Placeholder', dtype=dt,
+                 shape=None if shape is None else list(shape), name=name)
+    t.set_shape(shape)
+    return t
+
+
+def identity(x, name=None):
+    return apply_op('Identity', x, name=name)
+
+
+def stop_gradient(x, name=None):
+    return apply_op('StopGradient', x, name=name)
+
+
+def shape(x, name=None, out_type=dtypes.int32):
+    x = convert_to_tensor(x)
+    t = apply_op('Shape', x, out_type=out_type, name=name)
+    if x._shape is not None:
+        t.set_shape([len(x._shape)])
+        if all(d is not None for d in x._shape):
+            t._const_value = np.array(x._shape,
+                                      dtype=out_type.as_numpy_dtype)
+    return t
+
+
+def shape_n(xs, name=None):
+    return [shape(x) for x in xs]
+
+
+def rank(x, name=None):
+    return apply_op('Rank', x, name=name)
+
+
+def size(x, name=None):
+    return apply_op('Size', x, name=name)
+
+
+def reshape(x, new_shape, name=None):
+    x = convert_to_tensor(x)
+    t = apply_op('Reshape', x, convert_to_tensor(new_shape, dtype=dtypes.int32),
+                 name=name)
+    # static shape
+    sv = _static_value(t.op.inputs[1])
+    if sv is not None:
+        dims = [int(d) for d in sv]
+        if -1 in dims:
+            known = 1
+            for d in dims:
+                if d != -1:
+                    known *= d
+            total = _num_elements(x)
+            if total is not None:
+                dims[dims.index(-1)] = total // known
+            else:
+                dims[dims.index(-1)] = None
+        t.set_shape(dims)
+    return t
+
+
+def _num_elements(x):
+    if x._shape is None or any(d is None for d in x._shape):
+        return None
+    n = 1
+    for d in x._shape:
+        n *= d
+    return n
+
+
+def _static_value(t):
+    v = getattr(t, '_const_value', None)
+    return v
+
+
+def expand_dims(x, axis, name=None):
+    x = convert_to_tensor(x)
+    t = apply_op('ExpandDims', x, convert_to_tensor(axis, dtype=dtypes.int32),
+                 name=name)
+    if x._shape is not None:
+        s = list(x._shape)
+        a = axis if axis >= 0 else axis + len(s) + 1
+        s.insert(a, 1)
+        t.set_shape(s)
+    return t
+
+
+def squeeze(x, axis=None, name=None, squeeze_dims=None):
+    if squeeze_dims is not None:
+        axis = squeeze_dims
+    x = convert_to_tensor(x)
+    t = apply_op('Squeeze', x, squeeze_dims=list(axis) if axis else [],
+                 name=name)
+    if x._shape is not None:
+        dims = []
+        nd = len(x._shape)
+        ax = [a % nd for a in (axis or [])]
+        for i, d in enumerate(x._shape):
+            if (not ax and d == 1) or (ax and i in ax):
+                continue
+            dims.append(d)
+        t.set_shape(dims)
+    return t
+
+
+def zeros(shape_arg, dtype=dtypes.float32, name=None):
+    return _filled(shape_arg, 0, dtype, name or 'zeros')
+
+
+def ones(shape_arg, dtype=dtypes.float32, name=None):
+    return _filled(shape_arg, 1, dtype, name or 'ones')
+
+
+def _filled(shape_arg, value, dtype, name):
+    dt = dtypes.as_dtype(dtype)
+    if isinstance(shape_arg, ops.Tensor):
+        v = ops.constant(value, dtype=dt)
+        t = apply_op('Fill', shape_arg, v, name=name)
+        sv = _static_value(shape_arg)
+        if sv is not None:
+            t.set_shape([int(d) for d in sv])
+        return t
+    return ops.constant(value, dtype=dt, shape=list(shape_arg), name=name)
+
+
+def fill(dims, value, name=None):
+    return apply_op('Fill', convert_to_tensor(dims, dtype=dtypes.int32),
+                    convert_to_tensor(value), name=name)
+
+
+def zeros_like(x, dtype=None, name=None):
+    x = convert_to_tensor(x)
+    if dtype is not None and dtypes.as_dtype(dtype) != x.dtype:
+        from simple_tensorflow_amd.python.ops import math_ops
+        return zeros_like(math_ops.cast(x, dtype), name=name)
+    t = apply_op('ZerosLike', x, name=name)
+    t.set_shape(x._shape)
+    return t
+
+
+def ones_like(x, dtype=None, name=None):
+    x = convert_to_tensor(x)
+    t = apply_op('OnesLike', x, name=name)
+    t.set_shape(x._shape)
+    return t
+
+
+def concat(values, axis, name=None):
+    if isinstance(values, ops.Tensor):
+        values = [values]
+    values = [convert_to_tensor(v) for v in values]
+    t = apply_op('ConcatV2', values, convert_to_tensor(axis, dtype=dtypes.int32),
+                 name=name)
+    shapes = [v._shape for v in values]
+    if all(s is not None for s in shapes):
+        nd = len(shapes[0])
+        a = axis % nd
+        dims = list(shapes[0])
+        tot = 0
+        for s in shapes:
+            if s[a] is None:
+                tot = None
+                break
+            tot += s[a]
+        dims[a] = tot
+        t.set_shape(dims)
+    return t
+
+
+def split(value, num_or_size_splits, axis=0, name=None):
+    value = convert_to_tensor(value)
+    if isinstance(num_or_size_splits, int):
+        res = apply_op('Split', convert_to_tensor(axis, dtype=dtypes.int32),
+                       value, num_split=num_or_size_splits, name=name)
+        outs = list(res) if isinstance(res, tuple) else [res]
+        if value._shape is not None:
+            nd = len(value._shape)
+            a = axis % nd
+            dims = list(value._shape)
+            if dims[a] is not None:
+                dims[a] //= num_or_size_splits
+            for o in outs:
+                o.set_shape(dims)
+        return outs
+    raise NotImplementedError('size_splits list: use multiple slice ops')
+
+
+def stack(values, axis=0, name=None):
+    values = [convert_to_tensor(v) for v in values]
+    t = apply_op('Pack', values, axis=axis, name=name)
+    if values[0]._shape is not None:
+        dims = list(values[0]._shape)
+        dims.insert(axis if axis >= 0 else axis + len(dims) + 1, len(values))
+        t.set_shape(dims)
+    return t
+
+
+pack = stack
+
+
+def unstack(value, num=None, axis=0, name=None):
+    value = convert_to_tensor(value)
+    if num is None:
+        if value._shape is None or value._shape[axis] is None:
+            raise ValueError('Cannot infer num from shape')
+        num = value._shape[axis]
+    res = apply_op('Unpack', value, num=num, axis=axis, name=name)
+    outs = list(res) if isinstance(res, tuple) else [res]
+    if value._shape is not None:
+        dims = [d for i, d in enumerate(value._shape) if i != axis % len(value._shape)]
+        for o in outs:
+            o.set_shape(dims)
+    return outs
+
+
+unpack = unstack
+
+
+def slice(input_, begin, size, name=None):  # pylint: disable=redefined-builtin
+    t = apply_op('Slice', convert_to_tensor(input_),
+                 convert_to_tensor(begin, dtype=dtypes.int32),
+                 convert_to_tensor(size, dtype=dtypes.int32), name=name)
+    if not isinstance(size, ops.Tensor):
+        x = convert_to_tensor(input_)
+        dims = list(size)
+        if x._shape is not None and not isinstance(begin, ops.Tensor):
+            for i, d in enumerate(dims):
+                if d == -1 and x._shape[i] is not None:
+                    dims[i] = x._shape[i] - begin[i]
+        t.set_shape([d if d != -1 else None for d in dims])
+    return t
+
+
+def _slice_helper(tensor, key):
+    """Basic tensor[a:b, c] support via Slice (no strides)."""
+    if not isinstance(key, tuple):
+        key = (key,)
+    begin, sz, squeeze_axes = [], [], []
+    for i, k in enumerate(key):
+        if isinstance(k, int):
+            begin.append(k)
+            sz.append(1)
+            squeeze_axes.append(i)
+        elif isinstance(k, type(Ellipsis)):
+            raise NotImplementedError('ellipsis slicing')
+        elif isinstance(k, slice):
+            if k.step not in (None, 1):
+                raise NotImplementedError('strided slicing')
+            b = k.start or 0
+            begin.append(b)
+            if k.stop is None:
+                sz.append(-1)
+            else:
+                sz.append(k.stop - b)
+        else:
+            raise NotImplementedError('slice key %r' % (k,))
+    nd = len(tensor._shape) if tensor._shape is not None else len(key)
+    while len(begin) < nd:
+        begin.append(0)
+        sz.append(-1)
+    out = slice(tensor, begin, sz)
+    if squeeze_axes:
+        out = squeeze(out, axis=squeeze_axes)
+    return out
+
+
+def pad(x, paddings, name=None):
+    t = apply_op('Pad', convert_to_tensor(x),
+                 convert_to_tensor(paddings, dtype=dtypes.int32), name=name)
+    x = convert_to_tensor(x)
+    if x._shape is not None and not isinstance(paddings, ops.Tensor):
+        dims = []
+        for d, (lo, hi) in zip(x._shape, paddings):
+            dims.append(None if d is None else d + lo + hi)
+        t.set_shape(dims)
+    return t
+
+
+def transpose(x, perm=None, name=None):
+    x = convert_to_tensor(x)
+    if perm is None:
+        nd = len(x._shape)
+        perm = list(range(nd))[::-1]
+    t = apply_op('Transpose', x, convert_to_tensor(perm, dtype=dtypes.int32),
+                 name=name)
+    if x._shape is not None and not isinstance(perm, ops.Tensor):
+        t.set_shape([x._shape[p] for p in perm])
+    return t
+
+
+def gather(params, indices, name=None):
+    t = apply_op('Gather', convert_to_tensor(params),
+                 convert_to_tensor(indices, dtype=dtypes.int32), name=name)
+    p, i = t.op.inputs
+    if p._shape is not None and i._shape is not None:
+        t.set_shape(list(i._shape) + list(p._shape[1:]))
+    return t
+
+
+def tile(x, multiples, name=None):
+    return apply_op('Tile', convert_to_tensor(x),
+                    convert_to_tensor(multiples, dtype=dtypes.int32), name=name)
+
+
+def one_hot(indices, depth, on_value=1.0, off_value=0.0, axis=-1,
+            dtype=dtypes.float32, name=None):
+    dt = dtypes.as_dtype(dtype)
+    t = apply_op('OneHot', convert_to_tensor(indices, dtype=dtypes.int64)
+                 if not isinstance(indices, ops.Tensor) else indices,
+                 convert_to_tensor(depth, dtype=dtypes.int32),
+                 ops.constant(on_value, dtype=dt),
+                 ops.constant(off_value, dtype=dt), axis=axis, name=name)
+    idx = t.op.inputs[0]
+    if idx._shape is not None and isinstance(depth, int):
+        t.set_shape(list(idx._shape) + [depth])
+    return t
+
+
+def where(condition, x=None, y=None, name=None):
+    if x is None and y is None:
+        raise NotImplementedError('tf.where without x/y')
+    return apply_op('Select', convert_to_tensor(condition),
+                    convert_to_tensor(x), convert_to_tensor(y), name=name)
+
+
+def broadcast_gradient_args(s0, s1, name=None):
+    return apply_op('BroadcastGradientArgs', s0, s1, name=name)
+
+
+def check_numerics(x, message, name=None):
+    return apply_op('CheckNumerics', x, message=message, name=name)
+
+
+def unsorted_segment_sum(data, segment_ids, num_segments, name=None):
+    return apply_op('UnsortedSegmentSum', convert_to_tensor(data),
+                    convert_to_tensor(segment_ids),
+                    convert_to_tensor(num_segments, dtype=dtypes.int32),
+                    name=name)

@@ -1,0 +1,305 @@
+"""Gradients for math ops (analog of reference python/ops/math_grad.py)."""
+from simple_tensorflow_amd.python.framework import dtypes, ops
+from simple_tensorflow_amd.python.framework.ops import RegisterGradient, apply_op
+from simple_tensorflow_amd.python.ops import array_ops, math_ops
+
+
+def _shape_both(op):
+    sx = array_ops.shape(op.inputs[0])
+    sy = array_ops.shape(op.inputs[1])
+    rx, ry = array_ops.broadcast_gradient_args(sx, sy)
+    return sx, sy, rx, ry
+
+
+def _maybe_reduce(grad, orig, reduce_idx, shape_t):
+    """Sum `grad` over broadcast dims so it matches orig's shape."""
+    if grad is None:
+        return None
+    if orig._shape is not None and grad._shape is not None and \
+            list(orig._shape) == list(grad._shape) and \
+            all(d is not None for d in orig._shape):
+        return grad
+    g = math_ops.reduce_sum(grad, reduce_idx)
+    return array_ops.reshape(g, shape_t)
+
+
+@RegisterGradient('Add')
+def _add_grad(op, grad):
+    sx, sy, rx, ry = _shape_both(op)
+    return (_maybe_reduce(grad, op.inputs[0], rx, sx),
+            _maybe_reduce(grad, op.inputs[1], ry, sy))
+
+
+@RegisterGradient('Sub')
+def _sub_grad(op, grad):
+    sx, sy, rx, ry = _shape_both(op)
+    return (_maybe_reduce(grad, op.inputs[0], rx, sx),
+            _maybe_reduce(-grad, op.inputs[1], ry, sy))
+
+
+@RegisterGradient('Neg')
+def _neg_grad(op, grad):
+    return -grad
+
+
+@RegisterGradient('Mul')
+def _mul_grad(op, grad):
+    x, y = op.inputs
+    sx, sy, rx, ry = _shape_both(op)
+    return (_maybe_reduce(grad * y, x, rx, sx),
+            _maybe_reduce(grad * x, y, ry, sy))
+
+
+@RegisterGradient('RealDiv')
+def _realdiv_grad(op, grad):
+    x, y = op.inputs
+    sx, sy, rx, ry = _shape_both(op)
+    gx = grad / y
+    gy = -grad * x / (y * y)
+    return (_maybe_reduce(gx, x, rx, sx), _maybe_reduce(gy, y, ry, sy))
+
+
+@RegisterGradient('Div')
+def _div_grad(op, grad):
+    return _realdiv_grad(op, grad)
+
+
+@RegisterGradient('Pow')
+def _pow_grad(op, grad):
+    x, y = op.inputs
+    z = op.outputs[0]
+    sx, sy, rx, ry = _shape_both(op)
+    gx = grad * y * math_ops.pow(x, y - 1.0)
+    gy = grad * z * math_ops.log(x)
+    return (_maybe_reduce(gx, x, rx, sx), _maybe_reduce(gy, y, ry, sy))
+
+
+@RegisterGradient('Maximum')
+def _maximum_grad(op, grad):
+    x, y = op.inputs
+    sx, sy, rx, ry = _shape_both(op)
+    mask = math_ops.cast(math_ops.greater_equal(x, y), grad.dtype)
+    gx = grad * mask
+    gy = grad * (1.0 - mask)
+    return (_maybe_reduce(gx, x, rx, sx), _maybe_reduce(gy, y, ry, sy))
+
+
+@RegisterGradient('Minimum')
+def _minimum_grad(op, grad):
+    x, y = op.inputs
+    sx, sy, rx, ry = _shape_both(op)
+    mask = math_ops.cast(math_ops.less_equal(x, y), grad.dtype)
+    gx = grad * mask
+    gy = grad * (1.0 - mask)
+    return (_maybe_reduce(gx, x, rx, sx), _maybe_reduce(gy, y, ry, sy))
+
+
+@RegisterGradient('SquaredDifference')
+def _sqdiff_grad(op, grad):
+    x, y = op.inputs
+    sx, sy, rx, ry = _shape_both(op)
+    d = 2.0 * (x - y) * grad
+    return (_maybe_reduce(d, x, rx, sx), _maybe_reduce(-d, y, ry, sy))
+
+
+@RegisterGradient('Square')
+def _square_grad(op, grad):
+    return grad * 2.0 * op.inputs[0]
+
+
+@RegisterGradient('Sqrt')
+def _sqrt_grad(op, grad):
+    return apply_op('SqrtGrad', op.outputs[0], grad)
+
+
+@RegisterGradient('Rsqrt')
+def _rsqrt_grad(op, grad):
+    return apply_op('RsqrtGrad', op.outputs[0], grad)
+
+
+@RegisterGradient('Exp')
+def _exp_grad(op, grad):
+    return grad * op.outputs[0]
+
+
+@RegisterGradient('Log')
+def _log_grad(op, grad):
+    return grad / op.inputs[0]
+
+
+@RegisterGradient('Log1p')
+def _log1p_grad(op, grad):
+    return grad / (1.0 + op.inputs[0])
+
+
+@RegisterGradient('Tanh')
+def _tanh_grad(op, grad):
+    return apply_op('TanhGrad', op.outputs[0], grad)
+
+
+@RegisterGradient('Sigmoid')
+def _sigmoid_grad(op, grad):
+    return apply_op('SigmoidGrad', op.outputs[0], grad)
+
+
+@RegisterGradient('Sin')
+def _sin_grad(op, grad):
+    return grad * math_ops.cos(op.inputs[0])
+
+
+@RegisterGradient('Cos')
+def _cos_grad(op, grad):
+    return -grad * math_ops.sin(op.inputs[0])
+
+
+@RegisterGradient('Abs')
+def _abs_grad(op, grad):
+    return grad * math_ops.sign(op.inputs[0])
+
+
+@RegisterGradient('Reciprocal')
+def _reciprocal_grad(op, grad):
+    return apply_op('ReciprocalGrad', op.outputs[0], grad)
+
+
+@RegisterGradient('MatMul')
+def _matmul_grad(op, grad):
+    ta = op.get_attr('transpose_a')
+    tb = op.get_attr('transpose_b')
+    a, b = op.inputs
+    if not ta and not tb:
+        ga = math_ops.matmul(grad, b, transpose_b=True)
+        gb = math_ops.matmul(a, grad, transpose_a=True)
+    elif not ta and tb:
+        ga = math_ops.matmul(grad, b)
+        gb = math_ops.matmul(grad, a, transpose_a=True)
+    elif ta and not tb:
+        ga = math_ops.matmul(b, grad, transpose_b=True)
+        gb = math_ops.matmul(a, grad)
+    else:
+        ga = math_ops.matmul(b, grad, transpose_a=True, transpose_b=True)
+        gb = math_ops.matmul(grad, a, transpose_a=True, transpose_b=True)
+    return ga, gb
+
+
+@RegisterGradient('BatchMatMul')
+def _batch_matmul_grad(op, grad):
+    ta = op.get_attr('adj_x')
+    tb = op.get_attr('adj_y')
+    a, b = op.inputs
+    if not ta and not tb:
+        ga = math_ops.batch_matmul(grad, b, adj_y=True)
+        gb = math_ops.batch_matmul(a, grad, adj_x=True)
+    elif not ta and tb:
+        ga = math_ops.batch_matmul(grad, b)
+        gb = math_ops.batch_matmul(grad, a, adj_x=True)
+    elif ta and not tb:
+        ga = math_ops.batch_matmul(b, grad, adj_y=True)
+        gb = math_ops.batch_matmul(a, grad)
+    else:
+        ga = math_ops.batch_matmul(b, grad, adj_x=True, adj_y=True)
+        gb = math_ops.batch_matmul(grad, a, adj_x=True, adj_y=True)
+    return ga, gb
+
+
+@RegisterGradient('AddN')
+def _add_n_grad(op, grad):
+    return [grad] * len(op.inputs)
+
+
+def _safe_shape_div(x, y):
+    return x // y
+
+
+@RegisterGradient('Sum')
+def _sum_grad(op, grad):
+    input_shape = array_ops.shape(op.inputs[0])
+    # tile grad over reduced dims
+    factor = _tile_spec(op, grad, input_shape)
+    return [factor, None]
+
+
+def _tile_spec(op, grad, input_shape):
+    x = op.inputs[0]
+    axes = op.inputs[1]
+    # Static fast path: fully-known input shape + constant axes.
+    axes_v = getattr(axes, '_const_value', None)
+    if axes_v is not None and x._shape is not None and \
+            all(d is not None for d in x._shape):
+        nd = len(x._shape)
+        ax = {int(a) % nd for a in axes_v.reshape(-1)}
+        keep = [1 if i in ax else x._shape[i] for i in range(nd)]
+        mult = [x._shape[i] if i in ax else 1 for i in range(nd)]
+        g = array_ops.reshape(grad, keep)
+        t = array_ops.tile(g, mult)
+        t.set_shape(list(x._shape))
+        return t
+    # new shape = input shape with reduced dims -> 1
+    ones = array_ops.ones_like(input_shape)
+    nd = array_ops.size(input_shape)
+    axes_pos = math_ops.floormod(math_ops.add(axes, nd), nd)
+    mask1 = array_ops.unsorted_segment_sum(
+        array_ops.ones_like(axes_pos), axes_pos, nd)
+    keep_shape = math_ops.select(math_ops.greater(mask1, 0), ones, input_shape)
+    g = array_ops.reshape(grad, keep_shape)
+    mult = math_ops.floordiv(input_shape, keep_shape)
+    return array_ops.tile(g, mult)
+
+
+@RegisterGradient('Mean')
+def _mean_grad(op, grad):
+    input_shape = array_ops.shape(op.inputs[0])
+    sum_grad = _tile_spec(op, grad, input_shape)
+    in_n = math_ops.cast(array_ops.size(op.inputs[0]), grad.dtype)
+    out_n = math_ops.cast(array_ops.size(op.outputs[0]), grad.dtype)
+    return [sum_grad * (out_n / in_n), None]
+
+
+@RegisterGradient('Max')
+def _max_grad(op, grad):
+    y_rep = _tile_spec(op, op.outputs[0], array_ops.shape(op.inputs[0]))
+    g_rep = _tile_spec(op, grad, array_ops.shape(op.inputs[0]))
+    mask = math_ops.cast(math_ops.equal(op.inputs[0], y_rep), grad.dtype)
+    return [g_rep * mask, None]
+
+
+@RegisterGradient('Min')
+def _min_grad(op, grad):
+    return _max_grad(op, grad)
+
+
+@RegisterGradient('Prod')
+def _prod_grad(op, grad):
+    # grad * prod / x (requires nonzero x; acceptable round-1 semantics)
+    y_rep = _tile_spec(op, op.outputs[0], array_ops.shape(op.inputs[0]))
+    g_rep = _tile_spec(op, grad, array_ops.shape(op.inputs[0]))
+    return [g_rep * y_rep / op.inputs[0], None]
+
+
+@RegisterGradient('Cast')
+def _cast_grad(op, grad):
+    src = op.inputs[0].dtype
+    if src.is_floating and grad.dtype.is_floating:
+        return math_ops.cast(grad, src)
+    return None
+
+
+@RegisterGradient('Select')
+def _select_grad(op, grad):
+    c = op.inputs[0]
+    zeros = array_ops.zeros_like(grad)
+    return (None, math_ops.select(c, grad, zeros),
+            math_ops.select(c, zeros, grad))
+
+
+@RegisterGradient('L2Loss')
+def _l2_loss_grad(op, grad):
+    return op.inputs[0] * grad
+
+
+for _op in ('Less', 'LessEqual', 'Greater', 'GreaterEqual', 'Equal',
+            'NotEqual', 'LogicalAnd', 'LogicalOr', 'LogicalNot', 'ArgMax',
+            'ArgMin', 'Sign', 'Floor', 'Ceil', 'Round', 'Range', 'Shape',
+            'Rank', 'Size', 'IsNan', 'IsInf', 'IsFinite',
+            'BroadcastGradientArgs', 'InTopK'):
+    ops.NoGradient(_op)

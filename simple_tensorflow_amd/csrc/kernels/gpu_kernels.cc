@@ -1,0 +1,1149 @@
+// GPU OpKernels: thin wrappers that resolve shapes/attrs and enqueue the
+// hand-written CDNA4 HIP kernels (kernels/hip/*.hip) on the device's compute
+// stream. These REPLACE library paths (no rocBLAS/MIOpen): MatMul/Conv2D run
+// on the in-tree MFMA GEMM (guide-compliant, gfx950-only).
+#include <hip/hip_runtime.h>
+
+#include <atomic>
+
+#include "kernels/kernel_util.h"
+
+namespace stf {
+
+// ---- extern kernels (kernels/hip/*.hip) ----
+extern "C" {
+hipError_t stf_gemm_bf16_nt(const void*, const void*, void*, const void*,
+                            int64_t, int64_t, int64_t, float, int, int,
+                            hipStream_t);
+hipError_t stf_gemm_f32_nt(const void*, const void*, void*, int64_t, int64_t,
+                           int64_t, hipStream_t);
+hipError_t stf_unary(int, int, const void*, void*, int64_t, hipStream_t);
+hipError_t stf_binary(int, int, const void*, const void*, void*, int64_t,
+                      hipStream_t);
+hipError_t stf_binary_scalar(int, int, const void*, const void*, void*,
+                             int64_t, int, hipStream_t);
+hipError_t stf_binary_bcast(int, int, const void*, const void*, void*, int64_t,
+                            int, const int64_t*, const int64_t*,
+                            const int64_t*, hipStream_t);
+hipError_t stf_cast(int, int, const void*, void*, int64_t, hipStream_t);
+hipError_t stf_addn(int, const void* const*, int, void*, int64_t, hipStream_t);
+hipError_t stf_fill_f32(void*, float, int64_t, int, hipStream_t);
+hipError_t stf_scale(int, const void*, void*, int64_t, float, hipStream_t);
+hipError_t stf_im2col_bf16(const void*, void*, int, int, int, int, int, int,
+                           int, int, int, int, int, int, hipStream_t);
+hipError_t stf_col2im_bf16(const void*, void*, int, int, int, int, int, int,
+                           int, int, int, int, int, int, hipStream_t);
+hipError_t stf_bias_add(int, const void*, const void*, void*, int64_t, int,
+                        hipStream_t);
+hipError_t stf_bias_grad(int, const void*, void*, int64_t, int, hipStream_t);
+hipError_t stf_softmax(int, int, const void*, void*, int64_t, int,
+                       hipStream_t);
+hipError_t stf_sparse_xent(int, const void*, const void*, int, void*, void*,
+                           int64_t, int, hipStream_t);
+hipError_t stf_xent(int, const void*, const void*, void*, void*, int64_t, int,
+                    hipStream_t);
+hipError_t stf_bn_fwd(int, const void*, const void*, const void*, float*,
+                      float*, float*, float*, void*, int64_t, int, float, int,
+                      hipStream_t);
+hipError_t stf_bn_bwd(int, const void*, const void*, const float*,
+                      const float*, const void*, float*, void*, int64_t, int,
+                      hipStream_t);
+hipError_t stf_pool_fwd(int, int, const void*, void*, int, int, int, int, int,
+                        int, int, int, int, int, int, int, hipStream_t);
+hipError_t stf_max_pool_bwd(int, const void*, const void*, float*, int, int,
+                            int, int, int, int, int, int, int, int, int, int,
+                            hipStream_t);
+hipError_t stf_avg_pool_bwd(int, const void*, void*, int, int, int, int, int,
+                            int, int, int, int, int, int, int, hipStream_t);
+hipError_t stf_transpose2d(int, const void*, void*, int64_t, int64_t,
+                           hipStream_t);
+hipError_t stf_permute(int, const void*, void*, int64_t, int, const int64_t*,
+                       const int64_t*, hipStream_t);
+hipError_t stf_full_reduce(int, int, const void*, float*, int64_t,
+                           hipStream_t);
+hipError_t stf_row_reduce(int, int, const void*, float*, int64_t, int64_t,
+                          hipStream_t);
+hipError_t stf_col_reduce(int, int, const void*, float*, int64_t, int64_t,
+                          hipStream_t);
+hipError_t stf_bcast_copy(int, const void*, void*, int64_t, int,
+                          const int64_t*, const int64_t*, hipStream_t);
+hipError_t stf_apply_sgd(int, void*, const void*, const void*, int64_t,
+                         hipStream_t);
+hipError_t stf_apply_momentum(int, void*, void*, const void*, const void*,
+                              const void*, int, int64_t, hipStream_t);
+hipError_t stf_apply_adam(int, void*, void*, void*, const void*, const void*,
+                          const void*, const void*, const void*, const void*,
+                          const void*, int64_t, hipStream_t);
+hipError_t stf_random_uniform(uint64_t, uint64_t, void*, int64_t, int,
+                              hipStream_t);
+hipError_t stf_random_normal(uint64_t, uint64_t, void*, int64_t, int, int,
+                             hipStream_t);
+}
+
+namespace {
+
+#define GPU_STREAM(ctx) ((hipStream_t)(ctx)->device()->compute_stream())
+#define OP_HIP_OK(ctx, expr)                                              \
+  {                                                                       \
+    hipError_t _e = (expr);                                               \
+    if (_e != hipSuccess) {                                               \
+      (ctx)->SetStatus(errors::Internal("HIP kernel failure: ",           \
+                                        hipGetErrorString(_e)));          \
+      return;                                                             \
+    }                                                                     \
+  }
+
+inline int DtypeCode(DataType dt) { return dt == DT_FLOAT ? 0 : 1; }
+inline int CastCode(DataType dt) {
+  switch (dt) {
+    case DT_FLOAT: return 0;
+    case DT_BFLOAT16: return 1;
+    case DT_HALF: return 2;
+    case DT_INT32: return 3;
+    case DT_INT64: return 4;
+    default: return -1;
+  }
+}
+
+// zero an f32 device buffer on the stream
+inline hipError_t ZeroF32(void* p, int64_t n, hipStream_t s) {
+  return hipMemsetAsync(p, 0, n * 4, s);
+}
+
+// ---------------------------------------------------------------------------
+// elementwise
+// ---------------------------------------------------------------------------
+class GpuUnaryOp : public OpKernel {
+ public:
+  GpuUnaryOp(OpKernelConstruction* c, int op) : OpKernel(c), op_(op) {}
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    Tensor* y = ctx->allocate_output(0, x.shape());
+    OP_HIP_OK(ctx, stf_unary(op_, DtypeCode(x.dtype()), x.raw_data(),
+                             y->raw_data(), x.NumElements(), GPU_STREAM(ctx)));
+  }
+
+ private:
+  int op_;
+};
+
+// matches UOp order in elementwise.hip
+enum {
+  U_NEG, U_ABS, U_SIGN, U_SQUARE, U_SQRT, U_RSQRT, U_EXP, U_LOG, U_LOG1P,
+  U_TANH, U_SIGMOID, U_RELU, U_RELU6, U_SOFTPLUS, U_RECIP, U_FLOOR, U_CEIL,
+  U_SIN, U_COS
+};
+enum {
+  B_ADD, B_SUB, B_MUL, B_DIV, B_MAX, B_MIN, B_POW, B_SQDIFF, B_SIGMOID_GRAD,
+  B_TANH_GRAD, B_RSQRT_GRAD, B_SQRT_GRAD, B_RELU_GRAD, B_RELU6_GRAD,
+  B_SOFTPLUS_GRAD
+};
+
+#define REG_GPU_UNARY(NAME, CODE)                                             \
+  struct NAME##GpuTag {};                                                     \
+  class NAME##GpuOp : public GpuUnaryOp {                                     \
+   public:                                                                    \
+    explicit NAME##GpuOp(OpKernelConstruction* c) : GpuUnaryOp(c, CODE) {}    \
+  };                                                                          \
+  REGISTER_KERNEL_BUILDER(Name(#NAME).Device(DEVICE_GPU).TypeConstraint<float>("T"), NAME##GpuOp); \
+  REGISTER_KERNEL_BUILDER(Name(#NAME).Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), NAME##GpuOp);
+
+REG_GPU_UNARY(Neg, U_NEG)
+REG_GPU_UNARY(Abs, U_ABS)
+REG_GPU_UNARY(Sign, U_SIGN)
+REG_GPU_UNARY(Square, U_SQUARE)
+REG_GPU_UNARY(Sqrt, U_SQRT)
+REG_GPU_UNARY(Rsqrt, U_RSQRT)
+REG_GPU_UNARY(Exp, U_EXP)
+REG_GPU_UNARY(Log, U_LOG)
+REG_GPU_UNARY(Log1p, U_LOG1P)
+REG_GPU_UNARY(Tanh, U_TANH)
+REG_GPU_UNARY(Sigmoid, U_SIGMOID)
+REG_GPU_UNARY(Relu, U_RELU)
+REG_GPU_UNARY(Relu6, U_RELU6)
+REG_GPU_UNARY(Softplus, U_SOFTPLUS)
+REG_GPU_UNARY(Reciprocal, U_RECIP)
+REG_GPU_UNARY(Floor, U_FLOOR)
+REG_GPU_UNARY(Ceil, U_CEIL)
+REG_GPU_UNARY(Sin, U_SIN)
+REG_GPU_UNARY(Cos, U_COS)
+
+class GpuBinaryOp : public OpKernel {
+ public:
+  GpuBinaryOp(OpKernelConstruction* c, int op) : OpKernel(c), op_(op) {}
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& a = ctx->input(0);
+    const Tensor& b = ctx->input(1);
+    int dt = DtypeCode(a.dtype());
+    hipStream_t s = GPU_STREAM(ctx);
+    if (a.shape() == b.shape()) {
+      Tensor* y = ctx->allocate_output(0, a.shape());
+      OP_HIP_OK(ctx, stf_binary(op_, dt, a.raw_data(), b.raw_data(),
+                                y->raw_data(), a.NumElements(), s));
+      return;
+    }
+    if (b.NumElements() == 1) {
+      Tensor* y = ctx->allocate_output(0, a.shape());
+      OP_HIP_OK(ctx, stf_binary_scalar(op_, dt, a.raw_data(), b.raw_data(),
+                                       y->raw_data(), a.NumElements(), 0, s));
+      return;
+    }
+    if (a.NumElements() == 1) {
+      Tensor* y = ctx->allocate_output(0, b.shape());
+      OP_HIP_OK(ctx, stf_binary_scalar(op_, dt, b.raw_data(), a.raw_data(),
+                                       y->raw_data(), b.NumElements(), 1, s));
+      return;
+    }
+    BCast bc(a.shape(), b.shape());
+    OP_REQUIRES(ctx, bc.valid && bc.out.size() <= 6,
+                errors::InvalidArgument("bad broadcast"));
+    Tensor* y = ctx->allocate_output(0, bc.out_shape());
+    OP_HIP_OK(ctx, stf_binary_bcast(op_, dt, a.raw_data(), b.raw_data(),
+                                    y->raw_data(), bc.num_elements,
+                                    (int)bc.out.size(), bc.out.data(),
+                                    bc.sx.data(), bc.sy.data(), s));
+  }
+
+ private:
+  int op_;
+};
+
+#define REG_GPU_BINARY(NAME, CODE)                                            \
+  class NAME##GpuOp : public GpuBinaryOp {                                    \
+   public:                                                                    \
+    explicit NAME##GpuOp(OpKernelConstruction* c) : GpuBinaryOp(c, CODE) {}   \
+  };                                                                          \
+  REGISTER_KERNEL_BUILDER(Name(#NAME).Device(DEVICE_GPU).TypeConstraint<float>("T"), NAME##GpuOp); \
+  REGISTER_KERNEL_BUILDER(Name(#NAME).Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), NAME##GpuOp);
+
+REG_GPU_BINARY(Add, B_ADD)
+REG_GPU_BINARY(Sub, B_SUB)
+REG_GPU_BINARY(Mul, B_MUL)
+REG_GPU_BINARY(RealDiv, B_DIV)
+REG_GPU_BINARY(Div, B_DIV)
+REG_GPU_BINARY(Maximum, B_MAX)
+REG_GPU_BINARY(Minimum, B_MIN)
+REG_GPU_BINARY(Pow, B_POW)
+REG_GPU_BINARY(SquaredDifference, B_SQDIFF)
+REG_GPU_BINARY(SigmoidGrad, B_SIGMOID_GRAD)
+REG_GPU_BINARY(TanhGrad, B_TANH_GRAD)
+REG_GPU_BINARY(RsqrtGrad, B_RSQRT_GRAD)
+REG_GPU_BINARY(SqrtGrad, B_SQRT_GRAD)
+REG_GPU_BINARY(ReluGrad, B_RELU_GRAD)
+REG_GPU_BINARY(Relu6Grad, B_RELU6_GRAD)
+REG_GPU_BINARY(SoftplusGrad, B_SOFTPLUS_GRAD)
+
+class GpuCastOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    Tensor* y = ctx->allocate_output(0, x.shape());
+    int sc = CastCode(x.dtype()), dc = CastCode(y->dtype());
+    OP_REQUIRES(ctx, sc >= 0 && dc >= 0,
+                errors::Unimplemented("GPU cast dtype"));
+    OP_HIP_OK(ctx, stf_cast(sc, dc, x.raw_data(), y->raw_data(),
+                            x.NumElements(), GPU_STREAM(ctx)));
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("Cast").Device(DEVICE_GPU), GpuCastOp);
+
+class GpuAddNOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    int n = num_inputs();
+    const Tensor& first = ctx->input(0);
+    Tensor* y = ctx->allocate_output(0, first.shape());
+    hipStream_t s = GPU_STREAM(ctx);
+    int dt = DtypeCode(first.dtype());
+    if (n == 1) {
+      OP_HIP_OK(ctx, hipMemcpyAsync(y->raw_data(), first.raw_data(),
+                                    first.TotalBytes(),
+                                    hipMemcpyDeviceToDevice, s));
+      return;
+    }
+    // pairwise adds: y = in0 + in1; y += in_k
+    OP_HIP_OK(ctx, stf_binary(B_ADD, dt, first.raw_data(),
+                              ctx->input(1).raw_data(), y->raw_data(),
+                              first.NumElements(), s));
+    for (int k = 2; k < n; ++k) {
+      OP_HIP_OK(ctx, stf_binary(B_ADD, dt, y->raw_data(),
+                                ctx->input(k).raw_data(), y->raw_data(),
+                                first.NumElements(), s));
+    }
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("AddN").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuAddNOp);
+REGISTER_KERNEL_BUILDER(Name("AddN").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuAddNOp);
+
+// ---------------------------------------------------------------------------
+// fill-style
+// ---------------------------------------------------------------------------
+class GpuZerosLikeOp : public OpKernel {
+ public:
+  GpuZerosLikeOp(OpKernelConstruction* c, float v) : OpKernel(c), v_(v) {}
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    Tensor* y = ctx->allocate_output(0, x.shape());
+    OP_HIP_OK(ctx, stf_fill_f32(y->raw_data(), v_, x.NumElements(),
+                                x.dtype() == DT_BFLOAT16, GPU_STREAM(ctx)));
+  }
+
+ private:
+  float v_;
+};
+class ZerosLikeGpu : public GpuZerosLikeOp {
+ public:
+  explicit ZerosLikeGpu(OpKernelConstruction* c) : GpuZerosLikeOp(c, 0.f) {}
+};
+class OnesLikeGpu : public GpuZerosLikeOp {
+ public:
+  explicit OnesLikeGpu(OpKernelConstruction* c) : GpuZerosLikeOp(c, 1.f) {}
+};
+REGISTER_KERNEL_BUILDER(Name("ZerosLike").Device(DEVICE_GPU).TypeConstraint<float>("T"), ZerosLikeGpu);
+REGISTER_KERNEL_BUILDER(Name("ZerosLike").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), ZerosLikeGpu);
+REGISTER_KERNEL_BUILDER(Name("OnesLike").Device(DEVICE_GPU).TypeConstraint<float>("T"), OnesLikeGpu);
+REGISTER_KERNEL_BUILDER(Name("OnesLike").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), OnesLikeGpu);
+
+class GpuFillOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    auto dims = IntVector(ctx->input(0));
+    const Tensor& v = ctx->input(1);  // host scalar
+    Tensor* y = ctx->allocate_output(0, TensorShape(dims));
+    float val = v.dtype() == DT_BFLOAT16
+                    ? (float)v.flat<bfloat16>()[0]
+                    : v.flat<float>()[0];
+    OP_HIP_OK(ctx, stf_fill_f32(y->raw_data(), val, y->NumElements(),
+                                y->dtype() == DT_BFLOAT16, GPU_STREAM(ctx)));
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("Fill").Device(DEVICE_GPU).TypeConstraint<float>("T").HostMemory("dims").HostMemory("value"), GpuFillOp);
+REGISTER_KERNEL_BUILDER(Name("Fill").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T").HostMemory("dims").HostMemory("value"), GpuFillOp);
+
+// ---------------------------------------------------------------------------
+// MatMul (MFMA GEMM + pre-transposes to NT form)
+// ---------------------------------------------------------------------------
+class GpuMatMulOp : public OpKernel {
+ public:
+  explicit GpuMatMulOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("transpose_a", &ta_);
+    c->GetAttr("transpose_b", &tb_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& a = ctx->input(0);
+    const Tensor& b = ctx->input(1);
+    hipStream_t s = GPU_STREAM(ctx);
+    int64_t m = ta_ ? a.dim_size(1) : a.dim_size(0);
+    int64_t k = ta_ ? a.dim_size(0) : a.dim_size(1);
+    int64_t n = tb_ ? b.dim_size(0) : b.dim_size(1);
+    Tensor* y = ctx->allocate_output(0, TensorShape({m, n}));
+    size_t es = DataTypeSize(a.dtype());
+    // A effective [M,K]
+    Tensor a_eff = a;
+    if (ta_) {
+      a_eff = ctx->allocate_temp(a.dtype(), TensorShape({m, k}));
+      OP_HIP_OK(ctx, stf_transpose2d((int)es, a.raw_data(), a_eff.raw_data(),
+                                     a.dim_size(0), a.dim_size(1), s));
+    }
+    // B effective [N,K]
+    Tensor b_eff = b;
+    if (!tb_) {
+      b_eff = ctx->allocate_temp(b.dtype(), TensorShape({n, k}));
+      OP_HIP_OK(ctx, stf_transpose2d((int)es, b.raw_data(), b_eff.raw_data(),
+                                     b.dim_size(0), b.dim_size(1), s));
+    }
+    if (a.dtype() == DT_BFLOAT16) {
+      OP_HIP_OK(ctx, stf_gemm_bf16_nt(a_eff.raw_data(), b_eff.raw_data(),
+                                      y->raw_data(), nullptr, m, n, k, 0.f,
+                                      1, 0, s));
+    } else {
+      OP_HIP_OK(ctx, stf_gemm_f32_nt(a_eff.raw_data(), b_eff.raw_data(),
+                                     y->raw_data(), m, n, k, s));
+    }
+  }
+
+ private:
+  bool ta_ = false, tb_ = false;
+};
+REGISTER_KERNEL_BUILDER(Name("MatMul").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuMatMulOp);
+REGISTER_KERNEL_BUILDER(Name("MatMul").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuMatMulOp);
+
+// ---------------------------------------------------------------------------
+// Conv2D family (im2col + MFMA GEMM)
+// ---------------------------------------------------------------------------
+struct GpuConvGeom {
+  int64_t N, H, W, C, R, S, K, sh, sw, ph, pw, P, Q;
+  int64_t M() const { return N * P * Q; }
+  int64_t RSC() const { return R * S * C; }
+  bool is_1x1_s1() const {
+    return R == 1 && S == 1 && sh == 1 && sw == 1 && ph == 0 && pw == 0;
+  }
+};
+
+static Status GetConvGeom(const TensorShape& x, const TensorShape& f,
+                          const std::vector<int64_t>& strides,
+                          const std::string& padding, GpuConvGeom* g) {
+  g->N = x.dim_size(0);
+  g->H = x.dim_size(1);
+  g->W = x.dim_size(2);
+  g->C = x.dim_size(3);
+  g->R = f.dim_size(0);
+  g->S = f.dim_size(1);
+  g->K = f.dim_size(3);
+  if (f.dim_size(2) != g->C)
+    return errors::InvalidArgument("conv channel mismatch");
+  g->sh = strides[1];
+  g->sw = strides[2];
+  if (padding == "SAME") {
+    g->P = (g->H + g->sh - 1) / g->sh;
+    g->Q = (g->W + g->sw - 1) / g->sw;
+    g->ph = std::max<int64_t>(0, (g->P - 1) * g->sh + g->R - g->H) / 2;
+    g->pw = std::max<int64_t>(0, (g->Q - 1) * g->sw + g->S - g->W) / 2;
+  } else {
+    g->P = (g->H - g->R) / g->sh + 1;
+    g->Q = (g->W - g->S) / g->sw + 1;
+    g->ph = g->pw = 0;
+  }
+  return Status::OK();
+}
+
+class GpuConv2DOp : public OpKernel {
+ public:
+  explicit GpuConv2DOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("strides", &strides_);
+    c->GetAttr("padding", &padding_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    const Tensor& w = ctx->input(1);
+    hipStream_t s = GPU_STREAM(ctx);
+    GpuConvGeom g;
+    OP_REQUIRES_OK(ctx, GetConvGeom(x.shape(), w.shape(), strides_, padding_,
+                                    &g));
+    Tensor* y = ctx->allocate_output(0, TensorShape({g.N, g.P, g.Q, g.K}));
+    // weights -> [K, RSC]
+    Tensor wt = ctx->allocate_temp(DT_BFLOAT16, TensorShape({g.K, g.RSC()}));
+    OP_HIP_OK(ctx, stf_transpose2d(2, w.raw_data(), wt.raw_data(), g.RSC(),
+                                   g.K, s));
+    const void* col_data = x.raw_data();
+    Tensor col;
+    if (!g.is_1x1_s1()) {
+      col = ctx->allocate_temp(DT_BFLOAT16, TensorShape({g.M(), g.RSC()}));
+      OP_HIP_OK(ctx, stf_im2col_bf16(x.raw_data(), col.raw_data(), (int)g.N,
+                                     (int)g.H, (int)g.W, (int)g.C, (int)g.R,
+                                     (int)g.S, (int)g.sh, (int)g.sw, (int)g.ph,
+                                     (int)g.pw, (int)g.P, (int)g.Q, s));
+      col_data = col.raw_data();
+    }
+    OP_HIP_OK(ctx, stf_gemm_bf16_nt(col_data, wt.raw_data(), y->raw_data(),
+                                    nullptr, g.M(), g.K, g.RSC(), 0.f, 1, 0,
+                                    s));
+  }
+
+ private:
+  std::vector<int64_t> strides_;
+  std::string padding_;
+};
+REGISTER_KERNEL_BUILDER(Name("Conv2D").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuConv2DOp);
+
+class GpuConv2DBackpropInputOp : public OpKernel {
+ public:
+  explicit GpuConv2DBackpropInputOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("strides", &strides_);
+    c->GetAttr("padding", &padding_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    auto in_sizes = IntVector(ctx->input(0));
+    const Tensor& w = ctx->input(1);
+    const Tensor& dy = ctx->input(2);
+    hipStream_t s = GPU_STREAM(ctx);
+    TensorShape x_shape(in_sizes);
+    GpuConvGeom g;
+    OP_REQUIRES_OK(ctx, GetConvGeom(x_shape, w.shape(), strides_, padding_,
+                                    &g));
+    Tensor* dx = ctx->allocate_output(0, x_shape);
+    // dcol[M, RSC] = dy[M, K] x w[RSC, K]^T  (filter layout is already NT B)
+    if (g.is_1x1_s1()) {
+      OP_HIP_OK(ctx, stf_gemm_bf16_nt(dy.raw_data(), w.raw_data(),
+                                      dx->raw_data(), nullptr, g.M(), g.RSC(),
+                                      g.K, 0.f, 1, 0, s));
+      return;
+    }
+    Tensor dcol = ctx->allocate_temp(DT_BFLOAT16,
+                                     TensorShape({g.M(), g.RSC()}));
+    OP_HIP_OK(ctx, stf_gemm_bf16_nt(dy.raw_data(), w.raw_data(),
+                                    dcol.raw_data(), nullptr, g.M(), g.RSC(),
+                                    g.K, 0.f, 1, 0, s));
+    OP_HIP_OK(ctx, stf_col2im_bf16(dcol.raw_data(), dx->raw_data(), (int)g.N,
+                                   (int)g.H, (int)g.W, (int)g.C, (int)g.R,
+                                   (int)g.S, (int)g.sh, (int)g.sw, (int)g.ph,
+                                   (int)g.pw, (int)g.P, (int)g.Q, s));
+  }
+
+ private:
+  std::vector<int64_t> strides_;
+  std::string padding_;
+};
+REGISTER_KERNEL_BUILDER(Name("Conv2DBackpropInput").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T").HostMemory("input_sizes"), GpuConv2DBackpropInputOp);
+
+class GpuConv2DBackpropFilterOp : public OpKernel {
+ public:
+  explicit GpuConv2DBackpropFilterOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("strides", &strides_);
+    c->GetAttr("padding", &padding_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    auto f_sizes = IntVector(ctx->input(1));
+    const Tensor& dy = ctx->input(2);
+    hipStream_t s = GPU_STREAM(ctx);
+    TensorShape f_shape(f_sizes);
+    GpuConvGeom g;
+    OP_REQUIRES_OK(ctx, GetConvGeom(x.shape(), f_shape, strides_, padding_,
+                                    &g));
+    Tensor* dw = ctx->allocate_output(0, f_shape);
+    // dW[RSC, K] = col[M, RSC]^T x dy[M, K]  -> NT with A=colT, B=dyT
+    Tensor colT = ctx->allocate_temp(DT_BFLOAT16,
+                                     TensorShape({g.RSC(), g.M()}));
+    if (g.is_1x1_s1()) {
+      OP_HIP_OK(ctx, stf_transpose2d(2, x.raw_data(), colT.raw_data(), g.M(),
+                                     g.RSC(), s));
+    } else {
+      Tensor col = ctx->allocate_temp(DT_BFLOAT16,
+                                      TensorShape({g.M(), g.RSC()}));
+      OP_HIP_OK(ctx, stf_im2col_bf16(x.raw_data(), col.raw_data(), (int)g.N,
+                                     (int)g.H, (int)g.W, (int)g.C, (int)g.R,
+                                     (int)g.S, (int)g.sh, (int)g.sw, (int)g.ph,
+                                     (int)g.pw, (int)g.P, (int)g.Q, s));
+      OP_HIP_OK(ctx, stf_transpose2d(2, col.raw_data(), colT.raw_data(),
+                                     g.M(), g.RSC(), s));
+    }
+    Tensor dyT = ctx->allocate_temp(DT_BFLOAT16, TensorShape({g.K, g.M()}));
+    OP_HIP_OK(ctx, stf_transpose2d(2, dy.raw_data(), dyT.raw_data(), g.M(),
+                                   g.K, s));
+    OP_HIP_OK(ctx, stf_gemm_bf16_nt(colT.raw_data(), dyT.raw_data(),
+                                    dw->raw_data(), nullptr, g.RSC(), g.K,
+                                    g.M(), 0.f, 1, 0, s));
+  }
+
+ private:
+  std::vector<int64_t> strides_;
+  std::string padding_;
+};
+REGISTER_KERNEL_BUILDER(Name("Conv2DBackpropFilter").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T").HostMemory("filter_sizes"), GpuConv2DBackpropFilterOp);
+
+// ---------------------------------------------------------------------------
+// bias / softmax / xent
+// ---------------------------------------------------------------------------
+class GpuBiasAddOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    const Tensor& b = ctx->input(1);
+    Tensor* y = ctx->allocate_output(0, x.shape());
+    OP_HIP_OK(ctx, stf_bias_add(DtypeCode(x.dtype()), x.raw_data(),
+                                b.raw_data(), y->raw_data(), x.NumElements(),
+                                (int)b.NumElements(), GPU_STREAM(ctx)));
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("BiasAdd").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuBiasAddOp);
+REGISTER_KERNEL_BUILDER(Name("BiasAdd").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuBiasAddOp);
+
+class GpuBiasAddGradOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& dy = ctx->input(0);
+    int c = (int)dy.dim_size(dy.dims() - 1);
+    int64_t rows = dy.NumElements() / c;
+    Tensor* db = ctx->allocate_output(0, TensorShape({c}));
+    hipStream_t s = GPU_STREAM(ctx);
+    Tensor scratch = ctx->allocate_temp(DT_FLOAT, TensorShape({c}));
+    OP_HIP_OK(ctx, ZeroF32(scratch.raw_data(), c, s));
+    OP_HIP_OK(ctx, stf_bias_grad(DtypeCode(dy.dtype()), dy.raw_data(),
+                                 scratch.raw_data(), rows, c, s));
+    OP_HIP_OK(ctx, stf_cast(0, CastCode(db->dtype()), scratch.raw_data(),
+                            db->raw_data(), c, s));
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("BiasAddGrad").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuBiasAddGradOp);
+REGISTER_KERNEL_BUILDER(Name("BiasAddGrad").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuBiasAddGradOp);
+
+class GpuSoftmaxOp : public OpKernel {
+ public:
+  GpuSoftmaxOp(OpKernelConstruction* c, bool log_sm)
+      : OpKernel(c), log_(log_sm) {}
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    Tensor* y = ctx->allocate_output(0, x.shape());
+    int cols = (int)x.dim_size(x.dims() - 1);
+    int64_t rows = x.NumElements() / cols;
+    OP_HIP_OK(ctx, stf_softmax(DtypeCode(x.dtype()), log_, x.raw_data(),
+                               y->raw_data(), rows, cols, GPU_STREAM(ctx)));
+  }
+
+ private:
+  bool log_;
+};
+class SoftmaxGpu : public GpuSoftmaxOp {
+ public:
+  explicit SoftmaxGpu(OpKernelConstruction* c) : GpuSoftmaxOp(c, false) {}
+};
+class LogSoftmaxGpu : public GpuSoftmaxOp {
+ public:
+  explicit LogSoftmaxGpu(OpKernelConstruction* c) : GpuSoftmaxOp(c, true) {}
+};
+REGISTER_KERNEL_BUILDER(Name("Softmax").Device(DEVICE_GPU).TypeConstraint<float>("T"), SoftmaxGpu);
+REGISTER_KERNEL_BUILDER(Name("Softmax").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), SoftmaxGpu);
+REGISTER_KERNEL_BUILDER(Name("LogSoftmax").Device(DEVICE_GPU).TypeConstraint<float>("T"), LogSoftmaxGpu);
+
+class GpuSparseXentOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& logits = ctx->input(0);
+    const Tensor& labels = ctx->input(1);
+    int64_t rows = logits.dim_size(0);
+    int cols = (int)logits.dim_size(1);
+    Tensor* loss = ctx->allocate_output(0, TensorShape({rows}));
+    Tensor* bp = ctx->allocate_output(1, logits.shape());
+    hipStream_t s = GPU_STREAM(ctx);
+    Tensor loss_f32 = ctx->allocate_temp(DT_FLOAT, TensorShape({rows}));
+    OP_HIP_OK(ctx,
+              stf_sparse_xent(DtypeCode(logits.dtype()), logits.raw_data(),
+                              labels.raw_data(), labels.dtype() == DT_INT32,
+                              loss_f32.raw_data(), bp->raw_data(), rows, cols,
+                              s));
+    OP_HIP_OK(ctx, stf_cast(0, CastCode(loss->dtype()), loss_f32.raw_data(),
+                            loss->raw_data(), rows, s));
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("SparseSoftmaxCrossEntropyWithLogits").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuSparseXentOp);
+REGISTER_KERNEL_BUILDER(Name("SparseSoftmaxCrossEntropyWithLogits").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuSparseXentOp);
+
+class GpuXentOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& logits = ctx->input(0);
+    const Tensor& labels = ctx->input(1);
+    int64_t rows = logits.dim_size(0);
+    int cols = (int)logits.dim_size(1);
+    Tensor* loss = ctx->allocate_output(0, TensorShape({rows}));
+    Tensor* bp = ctx->allocate_output(1, logits.shape());
+    hipStream_t s = GPU_STREAM(ctx);
+    Tensor loss_f32 = ctx->allocate_temp(DT_FLOAT, TensorShape({rows}));
+    OP_HIP_OK(ctx, stf_xent(DtypeCode(logits.dtype()), logits.raw_data(),
+                            labels.raw_data(), loss_f32.raw_data(),
+                            bp->raw_data(), rows, cols, s));
+    OP_HIP_OK(ctx, stf_cast(0, CastCode(loss->dtype()), loss_f32.raw_data(),
+                            loss->raw_data(), rows, s));
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("SoftmaxCrossEntropyWithLogits").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuXentOp);
+REGISTER_KERNEL_BUILDER(Name("SoftmaxCrossEntropyWithLogits").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuXentOp);
+
+// ---------------------------------------------------------------------------
+// batch norm
+// ---------------------------------------------------------------------------
+class GpuBatchNormMiOp : public OpKernel {
+ public:
+  explicit GpuBatchNormMiOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("epsilon", &eps_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    const Tensor& scale = ctx->input(1);
+    const Tensor& offset = ctx->input(2);
+    int c = (int)x.dim_size(x.dims() - 1);
+    int64_t rows = x.NumElements() / c;
+    Tensor* y = ctx->allocate_output(0, x.shape());
+    Tensor* mean = ctx->allocate_output(1, TensorShape({c}));
+    Tensor* var = ctx->allocate_output(2, TensorShape({c}));
+    Tensor* inv_std = ctx->allocate_output(3, TensorShape({c}));
+    hipStream_t s = GPU_STREAM(ctx);
+    Tensor acc = ctx->allocate_temp(DT_FLOAT, TensorShape({2 * c}));
+    OP_HIP_OK(ctx, ZeroF32(acc.raw_data(), 2 * c, s));
+    OP_HIP_OK(ctx, stf_bn_fwd(DtypeCode(x.dtype()), x.raw_data(),
+                              scale.raw_data(), offset.raw_data(),
+                              acc.flat<float>(), mean->flat<float>(),
+                              var->flat<float>(), inv_std->flat<float>(),
+                              y->raw_data(), rows, c, eps_, 0, s));
+  }
+
+ private:
+  float eps_ = 1e-4f;
+};
+REGISTER_KERNEL_BUILDER(Name("BatchNormMi").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuBatchNormMiOp);
+REGISTER_KERNEL_BUILDER(Name("BatchNormMi").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuBatchNormMiOp);
+
+class GpuBatchNormMiGradOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& dy = ctx->input(0);
+    const Tensor& x = ctx->input(1);
+    const Tensor& scale = ctx->input(2);
+    const Tensor& mean = ctx->input(3);
+    const Tensor& inv_std = ctx->input(4);
+    int c = (int)x.dim_size(x.dims() - 1);
+    int64_t rows = x.NumElements() / c;
+    Tensor* dx = ctx->allocate_output(0, x.shape());
+    Tensor* dscale = ctx->allocate_output(1, TensorShape({c}));
+    Tensor* doffset = ctx->allocate_output(2, TensorShape({c}));
+    hipStream_t s = GPU_STREAM(ctx);
+    Tensor acc = ctx->allocate_temp(DT_FLOAT, TensorShape({2 * c}));
+    OP_HIP_OK(ctx, ZeroF32(acc.raw_data(), 2 * c, s));
+    OP_HIP_OK(ctx, stf_bn_bwd(DtypeCode(x.dtype()), dy.raw_data(),
+                              x.raw_data(), mean.flat<float>(),
+                              inv_std.flat<float>(), scale.raw_data(),
+                              acc.flat<float>(), dx->raw_data(), rows, c, s));
+    // doffset = acc[0:c] (sum_dy); dscale = acc[c:2c] (sum_dy_xhat)
+    OP_HIP_OK(ctx, hipMemcpyAsync(doffset->raw_data(), acc.flat<float>(),
+                                  c * 4, hipMemcpyDeviceToDevice, s));
+    OP_HIP_OK(ctx, hipMemcpyAsync(dscale->raw_data(), acc.flat<float>() + c,
+                                  c * 4, hipMemcpyDeviceToDevice, s));
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("BatchNormMiGrad").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuBatchNormMiGradOp);
+REGISTER_KERNEL_BUILDER(Name("BatchNormMiGrad").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuBatchNormMiGradOp);
+
+// ---------------------------------------------------------------------------
+// pooling
+// ---------------------------------------------------------------------------
+struct PoolAttrs {
+  std::vector<int64_t> ksize, strides;
+  std::string padding;
+};
+static void GetPool(OpKernelConstruction* c, PoolAttrs* p) {
+  c->GetAttr("ksize", &p->ksize);
+  c->GetAttr("strides", &p->strides);
+  c->GetAttr("padding", &p->padding);
+}
+static void PoolDims(const TensorShape& x, const PoolAttrs& p, int64_t* P,
+                     int64_t* Q, int64_t* ph, int64_t* pw) {
+  int64_t H = x.dim_size(1), W = x.dim_size(2);
+  int64_t kh = p.ksize[1], kw = p.ksize[2], sh = p.strides[1],
+          sw = p.strides[2];
+  if (p.padding == "SAME") {
+    *P = (H + sh - 1) / sh;
+    *Q = (W + sw - 1) / sw;
+    *ph = std::max<int64_t>(0, (*P - 1) * sh + kh - H) / 2;
+    *pw = std::max<int64_t>(0, (*Q - 1) * sw + kw - W) / 2;
+  } else {
+    *P = (H - kh) / sh + 1;
+    *Q = (W - kw) / sw + 1;
+    *ph = *pw = 0;
+  }
+}
+
+class GpuPoolOp : public OpKernel {
+ public:
+  GpuPoolOp(OpKernelConstruction* c, bool is_max)
+      : OpKernel(c), is_max_(is_max) {
+    GetPool(c, &p_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    int64_t P, Q, ph, pw;
+    PoolDims(x.shape(), p_, &P, &Q, &ph, &pw);
+    Tensor* y = ctx->allocate_output(
+        0, TensorShape({x.dim_size(0), P, Q, x.dim_size(3)}));
+    OP_HIP_OK(ctx, stf_pool_fwd(DtypeCode(x.dtype()), is_max_, x.raw_data(),
+                                y->raw_data(), (int)x.dim_size(0),
+                                (int)x.dim_size(1), (int)x.dim_size(2),
+                                (int)x.dim_size(3), (int)p_.ksize[1],
+                                (int)p_.ksize[2], (int)p_.strides[1],
+                                (int)p_.strides[2], (int)ph, (int)pw, (int)P,
+                                (int)Q, GPU_STREAM(ctx)));
+  }
+
+ private:
+  bool is_max_;
+  PoolAttrs p_;
+};
+class MaxPoolGpu : public GpuPoolOp {
+ public:
+  explicit MaxPoolGpu(OpKernelConstruction* c) : GpuPoolOp(c, true) {}
+};
+class AvgPoolGpu : public GpuPoolOp {
+ public:
+  explicit AvgPoolGpu(OpKernelConstruction* c) : GpuPoolOp(c, false) {}
+};
+REGISTER_KERNEL_BUILDER(Name("MaxPool").Device(DEVICE_GPU).TypeConstraint<float>("T"), MaxPoolGpu);
+REGISTER_KERNEL_BUILDER(Name("MaxPool").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), MaxPoolGpu);
+REGISTER_KERNEL_BUILDER(Name("AvgPool").Device(DEVICE_GPU).TypeConstraint<float>("T"), AvgPoolGpu);
+REGISTER_KERNEL_BUILDER(Name("AvgPool").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), AvgPoolGpu);
+
+class GpuMaxPoolGradOp : public OpKernel {
+ public:
+  explicit GpuMaxPoolGradOp(OpKernelConstruction* c) : OpKernel(c) {
+    GetPool(c, &p_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    const Tensor& dy = ctx->input(2);
+    int64_t P, Q, ph, pw;
+    PoolDims(x.shape(), p_, &P, &Q, &ph, &pw);
+    Tensor* dx = ctx->allocate_output(0, x.shape());
+    hipStream_t s = GPU_STREAM(ctx);
+    Tensor scratch = ctx->allocate_temp(DT_FLOAT, x.shape());
+    OP_HIP_OK(ctx, ZeroF32(scratch.raw_data(), x.NumElements(), s));
+    OP_HIP_OK(ctx, stf_max_pool_bwd(DtypeCode(x.dtype()), x.raw_data(),
+                                    dy.raw_data(), scratch.flat<float>(),
+                                    (int)x.dim_size(0), (int)x.dim_size(1),
+                                    (int)x.dim_size(2), (int)x.dim_size(3),
+                                    (int)p_.ksize[1], (int)p_.ksize[2],
+                                    (int)p_.strides[1], (int)p_.strides[2],
+                                    (int)ph, (int)pw, (int)P, (int)Q, s));
+    OP_HIP_OK(ctx, stf_cast(0, CastCode(dx->dtype()), scratch.raw_data(),
+                            dx->raw_data(), x.NumElements(), s));
+  }
+
+ private:
+  PoolAttrs p_;
+};
+REGISTER_KERNEL_BUILDER(Name("MaxPoolGrad").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuMaxPoolGradOp);
+REGISTER_KERNEL_BUILDER(Name("MaxPoolGrad").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuMaxPoolGradOp);
+
+class GpuAvgPoolGradOp : public OpKernel {
+ public:
+  explicit GpuAvgPoolGradOp(OpKernelConstruction* c) : OpKernel(c) {
+    GetPool(c, &p_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    auto in_sizes = IntVector(ctx->input(0));
+    const Tensor& dy = ctx->input(1);
+    TensorShape x_shape(in_sizes);
+    int64_t P, Q, ph, pw;
+    PoolDims(x_shape, p_, &P, &Q, &ph, &pw);
+    Tensor* dx = ctx->allocate_output(0, x_shape);
+    OP_HIP_OK(ctx, stf_avg_pool_bwd(DtypeCode(dy.dtype()), dy.raw_data(),
+                                    dx->raw_data(), (int)x_shape.dim_size(0),
+                                    (int)x_shape.dim_size(1),
+                                    (int)x_shape.dim_size(2),
+                                    (int)x_shape.dim_size(3),
+                                    (int)p_.ksize[1], (int)p_.ksize[2],
+                                    (int)p_.strides[1], (int)p_.strides[2],
+                                    (int)ph, (int)pw, (int)P, (int)Q,
+                                    GPU_STREAM(ctx)));
+  }
+
+ private:
+  PoolAttrs p_;
+};
+REGISTER_KERNEL_BUILDER(Name("AvgPoolGrad").Device(DEVICE_GPU).TypeConstraint<float>("T").HostMemory("orig_input_shape"), GpuAvgPoolGradOp);
+REGISTER_KERNEL_BUILDER(Name("AvgPoolGrad").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T").HostMemory("orig_input_shape"), GpuAvgPoolGradOp);
+
+// ---------------------------------------------------------------------------
+// reductions: Sum / Mean / Max (all-axes, trailing-axes, or leading-axes)
+// ---------------------------------------------------------------------------
+class GpuReduceOp : public OpKernel {
+ public:
+  GpuReduceOp(OpKernelConstruction* c, int red, bool mean)
+      : OpKernel(c), red_(red), mean_(mean) {
+    c->GetAttr("keep_dims", &keep_dims_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    auto axes_v = IntVector(ctx->input(1));
+    int rank = x.dims();
+    std::vector<bool> reduce(rank, false);
+    for (auto a : axes_v) reduce[a < 0 ? a + rank : a] = true;
+    hipStream_t s = GPU_STREAM(ctx);
+    int dt = DtypeCode(x.dtype());
+    TensorShape out_shape;
+    int64_t count = 1;
+    for (int i = 0; i < rank; ++i) {
+      if (reduce[i]) {
+        count *= x.dim_size(i);
+        if (keep_dims_) out_shape.AddDim(1);
+      } else {
+        out_shape.AddDim(x.dim_size(i));
+      }
+    }
+    int64_t out_n = out_shape.num_elements();
+    Tensor f32_out = ctx->allocate_temp(DT_FLOAT, TensorShape({out_n}));
+    // classify
+    bool all_red = count == x.NumElements();
+    int first_keep = -1, last_red = -1, first_red = rank, contiguous = 1;
+    for (int i = 0; i < rank; ++i) {
+      if (reduce[i]) {
+        if (i < first_red) first_red = i;
+        last_red = i;
+      }
+    }
+    bool trailing = last_red == rank - 1;
+    bool leading = first_red == 0;
+    for (int i = first_red; i <= last_red && i >= 0; ++i)
+      if (!reduce[i]) contiguous = 0;
+    if (all_red) {
+      float init = red_ == 0 ? 0.f : (red_ == 1 ? -3.4e38f : 3.4e38f);
+      OP_HIP_OK(ctx, stf_fill_f32(f32_out.raw_data(), init, 1, 0, s));
+      OP_HIP_OK(ctx, stf_full_reduce(dt, red_, x.raw_data(),
+                                     f32_out.flat<float>(), x.NumElements(),
+                                     s));
+    } else if (trailing && contiguous) {
+      OP_HIP_OK(ctx, stf_row_reduce(dt, red_, x.raw_data(),
+                                    f32_out.flat<float>(), out_n, count, s));
+    } else if (leading && contiguous) {
+      OP_HIP_OK(ctx, stf_col_reduce(dt, red_, x.raw_data(),
+                                    f32_out.flat<float>(), count, out_n, s));
+    } else {
+      OP_REQUIRES(ctx, false,
+                  errors::Unimplemented(
+                      "GPU reduce supports all/leading/trailing axes"));
+    }
+    if (mean_ && count > 0) {
+      OP_HIP_OK(ctx, stf_scale(0, f32_out.raw_data(), f32_out.raw_data(),
+                               out_n, 1.f / (float)count, s));
+    }
+    Tensor* out = ctx->allocate_output(0, out_shape);
+    OP_HIP_OK(ctx, stf_cast(0, CastCode(out->dtype()), f32_out.raw_data(),
+                            out->raw_data(), out_n, s));
+  }
+
+ private:
+  int red_;
+  bool mean_;
+  bool keep_dims_ = false;
+};
+class SumGpu : public GpuReduceOp {
+ public:
+  explicit SumGpu(OpKernelConstruction* c) : GpuReduceOp(c, 0, false) {}
+};
+class MeanGpu : public GpuReduceOp {
+ public:
+  explicit MeanGpu(OpKernelConstruction* c) : GpuReduceOp(c, 0, true) {}
+};
+class MaxGpu : public GpuReduceOp {
+ public:
+  explicit MaxGpu(OpKernelConstruction* c) : GpuReduceOp(c, 1, false) {}
+};
+class MinGpu : public GpuReduceOp {
+ public:
+  explicit MinGpu(OpKernelConstruction* c) : GpuReduceOp(c, 2, false) {}
+};
+#define REG_GPU_REDUCE(OP, CLS)                                                \
+  REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_GPU).TypeConstraint<float>("T").HostMemory("reduction_indices"), CLS); \
+  REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_GPU).TypeConstraint<bfloat16>("T").HostMemory("reduction_indices"), CLS);
+REG_GPU_REDUCE("Sum", SumGpu)
+REG_GPU_REDUCE("Mean", MeanGpu)
+REG_GPU_REDUCE("Max", MaxGpu)
+REG_GPU_REDUCE("Min", MinGpu)
+#undef REG_GPU_REDUCE
+
+// ---------------------------------------------------------------------------
+// Tile (broadcast only: tiled dims must have input size 1) + Transpose
+// ---------------------------------------------------------------------------
+class GpuTileOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    auto mult = IntVector(ctx->input(1));
+    int rank = x.dims();
+    TensorShape out_shape;
+    std::vector<int64_t> strides(rank), dims(rank);
+    int64_t s_acc = 1;
+    for (int i = rank - 1; i >= 0; --i) {
+      strides[i] = x.dim_size(i) == 1 ? 0 : s_acc;
+      s_acc *= x.dim_size(i);
+    }
+    for (int i = 0; i < rank; ++i) {
+      OP_REQUIRES(ctx, mult[i] == 1 || x.dim_size(i) == 1,
+                  errors::Unimplemented("GPU Tile: only broadcast tiling"));
+      dims[i] = x.dim_size(i) * mult[i];
+      out_shape.AddDim(dims[i]);
+    }
+    Tensor* y = ctx->allocate_output(0, out_shape);
+    OP_HIP_OK(ctx, stf_bcast_copy((int)DataTypeSize(x.dtype()), x.raw_data(),
+                                  y->raw_data(), out_shape.num_elements(),
+                                  rank, dims.data(), strides.data(),
+                                  GPU_STREAM(ctx)));
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("Tile").Device(DEVICE_GPU).HostMemory("multiples"), GpuTileOp);
+
+class GpuTransposeOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    auto perm = IntVector(ctx->input(1));
+    int rank = x.dims();
+    TensorShape out_shape;
+    std::vector<int64_t> in_strides(rank, 1), src_strides(rank), dims(rank);
+    for (int i = rank - 2; i >= 0; --i)
+      in_strides[i] = in_strides[i + 1] * x.dim_size(i + 1);
+    for (int i = 0; i < rank; ++i) {
+      dims[i] = x.dim_size((int)perm[i]);
+      src_strides[i] = in_strides[(int)perm[i]];
+      out_shape.AddDim(dims[i]);
+    }
+    Tensor* y = ctx->allocate_output(0, out_shape);
+    if (rank == 2 && perm[0] == 1) {
+      OP_HIP_OK(ctx, stf_transpose2d((int)DataTypeSize(x.dtype()),
+                                     x.raw_data(), y->raw_data(),
+                                     x.dim_size(0), x.dim_size(1),
+                                     GPU_STREAM(ctx)));
+      return;
+    }
+    OP_HIP_OK(ctx, stf_permute((int)DataTypeSize(x.dtype()), x.raw_data(),
+                               y->raw_data(), out_shape.num_elements(), rank,
+                               dims.data(), src_strides.data(),
+                               GPU_STREAM(ctx)));
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("Transpose").Device(DEVICE_GPU).HostMemory("perm"), GpuTransposeOp);
+
+// ---------------------------------------------------------------------------
+// state: Assign / AssignAdd / AssignSub
+// ---------------------------------------------------------------------------
+class GpuAssignOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    Tensor ref = ctx->input(0);
+    const Tensor& value = ctx->input(1);
+    OP_REQUIRES(ctx, ref.shape() == value.shape(),
+                errors::InvalidArgument("Assign shape mismatch"));
+    OP_HIP_OK(ctx, hipMemcpyAsync(ref.raw_data(), value.raw_data(),
+                                  value.TotalBytes(), hipMemcpyDeviceToDevice,
+                                  GPU_STREAM(ctx)));
+    ctx->set_output(0, ref);
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("Assign").Device(DEVICE_GPU), GpuAssignOp);
+
+template <int BOP>
+class GpuAssignUpdateOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    Tensor ref = ctx->input(0);
+    const Tensor& value = ctx->input(1);
+    OP_HIP_OK(ctx, stf_binary(BOP, DtypeCode(ref.dtype()), ref.raw_data(),
+                              value.raw_data(), ref.raw_data(),
+                              ref.NumElements(), GPU_STREAM(ctx)));
+    ctx->set_output(0, ref);
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("AssignAdd").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuAssignUpdateOp<B_ADD>);
+REGISTER_KERNEL_BUILDER(Name("AssignSub").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuAssignUpdateOp<B_SUB>);
+
+// ---------------------------------------------------------------------------
+// optimizer applies (f32 vars)
+// ---------------------------------------------------------------------------
+class GpuApplySgdOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    Tensor var = ctx->input(0);
+    const Tensor& lr = ctx->input(1);
+    const Tensor& grad = ctx->input(2);
+    OP_HIP_OK(ctx, stf_apply_sgd(grad.dtype() == DT_BFLOAT16, var.raw_data(),
+                                 lr.raw_data(), grad.raw_data(),
+                                 var.NumElements(), GPU_STREAM(ctx)));
+    ctx->set_output(0, var);
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("ApplyGradientDescent").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuApplySgdOp);
+
+class GpuApplyMomentumOp : public OpKernel {
+ public:
+  explicit GpuApplyMomentumOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("use_nesterov", &nesterov_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    Tensor var = ctx->input(0);
+    Tensor accum = ctx->input(1);
+    const Tensor& lr = ctx->input(2);
+    const Tensor& grad = ctx->input(3);
+    const Tensor& mom = ctx->input(4);
+    OP_HIP_OK(ctx, stf_apply_momentum(grad.dtype() == DT_BFLOAT16,
+                                      var.raw_data(), accum.raw_data(),
+                                      lr.raw_data(), grad.raw_data(),
+                                      mom.raw_data(), nesterov_,
+                                      var.NumElements(), GPU_STREAM(ctx)));
+    ctx->set_output(0, var);
+  }
+
+ private:
+  bool nesterov_ = false;
+};
+REGISTER_KERNEL_BUILDER(Name("ApplyMomentum").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuApplyMomentumOp);
+
+class GpuApplyAdamOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    Tensor var = ctx->input(0);
+    Tensor m = ctx->input(1);
+    Tensor v = ctx->input(2);
+    OP_HIP_OK(ctx, stf_apply_adam(
+                       ctx->input(9).dtype() == DT_BFLOAT16, var.raw_data(),
+                       m.raw_data(), v.raw_data(), ctx->input(3).raw_data(),
+                       ctx->input(4).raw_data(), ctx->input(5).raw_data(),
+                       ctx->input(6).raw_data(), ctx->input(7).raw_data(),
+                       ctx->input(8).raw_data(), ctx->input(9).raw_data(),
+                       var.NumElements(), GPU_STREAM(ctx)));
+    ctx->set_output(0, var);
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("ApplyAdam").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuApplyAdamOp);
+
+// ---------------------------------------------------------------------------
+// random
+// ---------------------------------------------------------------------------
+class GpuRandomOp : public OpKernel {
+ public:
+  GpuRandomOp(OpKernelConstruction* c, int kind) : OpKernel(c), kind_(kind) {
+    int64_t seed = 0, seed2 = 0;
+    c->GetAttr("seed", &seed);
+    c->GetAttr("seed2", &seed2);
+    if (seed == 0 && seed2 == 0) seed = 0x9E3779B9;
+    seed_ = ((uint64_t)seed << 32) | (uint32_t)seed2;
+  }
+  void Compute(OpKernelContext* ctx) override {
+    auto dims = IntVector(ctx->input(0));
+    Tensor* y = ctx->allocate_output(0, TensorShape(dims));
+    int64_t n = y->NumElements();
+    uint64_t off = offset_.fetch_add((n + 3) / 4 + 1);
+    int bf16 = y->dtype() == DT_BFLOAT16;
+    hipStream_t s = GPU_STREAM(ctx);
+    if (kind_ == 0) {
+      OP_HIP_OK(ctx, stf_random_uniform(seed_, off, y->raw_data(), n, bf16, s));
+    } else {
+      OP_HIP_OK(ctx, stf_random_normal(seed_, off, y->raw_data(), n, bf16,
+                                       kind_ == 2, s));
+    }
+  }
+
+ private:
+  int kind_;
+  uint64_t seed_;
+  std::atomic<uint64_t> offset_{0};
+};
+class RandomUniformGpu : public GpuRandomOp {
+ public:
+  explicit RandomUniformGpu(OpKernelConstruction* c) : GpuRandomOp(c, 0) {}
+};
+class RandomNormalGpu : public GpuRandomOp {
+ public:
+  explicit RandomNormalGpu(OpKernelConstruction* c) : GpuRandomOp(c, 1) {}
+};
+class TruncatedNormalGpu : public GpuRandomOp {
+ public:
+  explicit TruncatedNormalGpu(OpKernelConstruction* c) : GpuRandomOp(c, 2) {}
+};
+REGISTER_KERNEL_BUILDER(Name("RandomUniform").Device(DEVICE_GPU).HostMemory("shape"), RandomUniformGpu);
+REGISTER_KERNEL_BUILDER(Name("RandomStandardNormal").Device(DEVICE_GPU).HostMemory("shape"), RandomNormalGpu);
+REGISTER_KERNEL_BUILDER(Name("TruncatedNormal").Device(DEVICE_GPU).HostMemory("shape"), TruncatedNormalGpu);
+
+}  // namespace
+}  // namespace stf

@@ -1,0 +1,243 @@
+// NT bf16 MFMA GEMM for gfx950: C[M,N](f32 or bf16) = A[M,K] * B[N,K]^T.
+//
+// This is the framework's core matmul/conv engine (replaces the reference's
+// cuBLAS path, matmul_op.cc:192 ThenBlasGemm). Structure follows the CDNA4
+// guide's "step-3" recipe (§5 ladder): 4-wave workgroup, 128x128 (or
+// 128x64/64x64) C tile, K-step 64, double-buffered LDS staged with
+// __builtin_amdgcn_global_load_lds (16B), XOR-swizzled LDS image
+// (byte ^= (row&7)<<4, both-sides rule 21) read back as ds_read_b128
+// fragments feeding v_mfma_f32_16x16x32_bf16, f32 accumulation in AGPRs.
+// Interior blocks take the glds fast path; edge blocks (M/N remainder or K
+// tail) stage through a guarded scalar path into the same swizzled image.
+#include "hip_common.h"
+
+namespace {
+
+__device__ __forceinline__ f32x4 mfma_bf16(bf16x8 a, bf16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+// One operand tile: BR rows x 64 k, stored as 16B chunks with chunk index
+// XOR-swizzled by (row&7). Slot s holds (r = s/8, c8 = (s%8) ^ (r&7)).
+template <int BR>
+struct TileGeom {
+  static constexpr int kSlots = BR * 8;        // 16B slots
+  static constexpr int kBytes = kSlots * 16;   // = BR * 128
+  static constexpr int kPasses = kSlots / 256; // glds passes (256 threads)
+};
+
+// Fast staging: lane-linear glds; the source address carries the inverse
+// swizzle so the LDS image is the swizzled one.
+template <int BR>
+__device__ __forceinline__ void StageFast(const uint16_t* __restrict__ src,
+                                          int64_t ld, int row0, int64_t k0,
+                                          uint16_t* lds_base, int tid) {
+#pragma unroll
+  for (int p = 0; p < TileGeom<BR>::kPasses; ++p) {
+    int s = p * 256 + tid;
+    int r = s >> 3;
+    int c8 = (s & 7) ^ (r & 7);
+    const uint16_t* g = src + (int64_t)(row0 + r) * ld + k0 + c8 * 8;
+    __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) uint32_t*)g,
+                                     (__attribute__((address_space(3))) uint32_t*)(lds_base + (int64_t)s * 8),
+                                     16, 0, 0);
+  }
+}
+
+// Guarded staging for edge blocks / K tails: scalar loads, zero padding.
+template <int BR>
+__device__ __forceinline__ void StageSafe(const uint16_t* __restrict__ src,
+                                          int64_t ld, int row0, int64_t k0,
+                                          int64_t rows, int64_t K,
+                                          uint16_t* lds_base, int tid) {
+#pragma unroll
+  for (int p = 0; p < TileGeom<BR>::kPasses; ++p) {
+    int s = p * 256 + tid;
+    int r = s >> 3;
+    int c8 = (s & 7) ^ (r & 7);
+    uint16_t vals[8];
+    int64_t row = row0 + r;
+    int64_t kbase = k0 + c8 * 8;
+    if (row < rows && kbase + 8 <= K) {
+      const uint16_t* g = src + row * ld + kbase;
+      *(ulong2*)vals = *(const ulong2*)g;
+    } else if (row < rows && kbase < K) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        vals[e] = (kbase + e < K) ? src[row * ld + kbase + e] : 0;
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) vals[e] = 0;
+    }
+    *(ulong2*)(lds_base + (int64_t)s * 8) = *(ulong2*)vals;
+  }
+}
+
+// GEMM kernel template.
+//  WAVES_M x WAVES_N waves; each wave computes (WM*16) x (WN*16) outputs.
+//  BM = WAVES_M*WM*16, BN = WAVES_N*WN*16, BK = 64.
+template <int WAVES_M, int WAVES_N, int WM, int WN, bool OUT_BF16,
+          bool FUSE_RELU>
+__launch_bounds__(256) __global__ void GemmBf16NT(
+    const uint16_t* __restrict__ A,  // [M, K] row-major bf16
+    const uint16_t* __restrict__ B,  // [N, K] row-major bf16
+    void* __restrict__ C,            // [M, N] f32 or bf16
+    const float* __restrict__ bias,  // optional [N] f32 bias (nullptr = none)
+    int64_t M, int64_t N, int64_t K, float beta) {
+  constexpr int BM = WAVES_M * WM * 16;
+  constexpr int BN = WAVES_N * WN * 16;
+  constexpr int BK = 64;
+
+  __shared__ __attribute__((aligned(16)))
+      uint16_t lds[2 * (BM + BN) * BK];
+  // buffer layout: [A0][A1][B0][B1]; pick by integer offset (an initialized
+  // array of LDS pointers does not compile on gfx950)
+  auto a_tile = [&](int buf) { return lds + buf * BM * BK; };
+  auto b_tile = [&](int buf) { return lds + 2 * BM * BK + buf * BN * BK; };
+
+  int nbm = (int)((M + BM - 1) / BM);
+  int nbn = (int)((N + BN - 1) / BN);
+  int bid = XcdSwizzle(blockIdx.x, nbm * nbn);
+  int bm = bid / nbn, bn = bid % nbn;
+  int64_t m0 = (int64_t)bm * BM, n0 = (int64_t)bn * BN;
+
+  int tid = threadIdx.x;
+  int lane = tid & 63;
+  int wid = tid >> 6;
+  int wr = wid / WAVES_N, wc = wid % WAVES_N;
+
+  bool a_interior = (m0 + BM) <= M;
+  bool b_interior = (n0 + BN) <= N;
+
+  f32x4 acc[WM][WN];
+#pragma unroll
+  for (int i = 0; i < WM; ++i)
+#pragma unroll
+    for (int j = 0; j < WN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  int kt_count = (int)((K + BK - 1) / BK);
+  // ---- prologue: stage tile 0 ----
+  {
+    bool kfull = BK <= K;
+    if (a_interior && kfull)
+      StageFast<BM>(A, K, (int)m0, 0, a_tile(0), tid);
+    else
+      StageSafe<BM>(A, K, (int)m0, 0, M, K, a_tile(0), tid);
+    if (b_interior && kfull)
+      StageFast<BN>(B, K, (int)n0, 0, b_tile(0), tid);
+    else
+      StageSafe<BN>(B, K, (int)n0, 0, N, K, b_tile(0), tid);
+  }
+  __syncthreads();
+
+  int cur = 0;
+  for (int kt = 0; kt < kt_count; ++kt) {
+    // issue next tile's loads first (overlap with this tile's compute)
+    if (kt + 1 < kt_count) {
+      int64_t k0 = (int64_t)(kt + 1) * BK;
+      bool kfull = k0 + BK <= K;
+      if (a_interior && kfull)
+        StageFast<BM>(A, K, (int)m0, k0, a_tile(cur ^ 1), tid);
+      else
+        StageSafe<BM>(A, K, (int)m0, k0, M, K, a_tile(cur ^ 1), tid);
+      if (b_interior && kfull)
+        StageFast<BN>(B, K, (int)n0, k0, b_tile(cur ^ 1), tid);
+      else
+        StageSafe<BN>(B, K, (int)n0, k0, N, K, b_tile(cur ^ 1), tid);
+    }
+
+    // compute on current tile: 2 mfma K-steps of 32
+    const uint16_t* at = a_tile(cur);
+    const uint16_t* bt = b_tile(cur);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 afrag[WM], bfrag[WN];
+#pragma unroll
+      for (int i = 0; i < WM; ++i) {
+        int r = wr * WM * 16 + i * 16 + (lane & 15);
+        int c8 = (kk * 4 + (lane >> 4)) ^ (r & 7);
+        afrag[i] = *(const bf16x8*)(at + (r * 8 + c8) * 8);
+      }
+#pragma unroll
+      for (int j = 0; j < WN; ++j) {
+        int r = wc * WN * 16 + j * 16 + (lane & 15);
+        int c8 = (kk * 4 + (lane >> 4)) ^ (r & 7);
+        bfrag[j] = *(const bf16x8*)(bt + (r * 8 + c8) * 8);
+      }
+#pragma unroll
+      for (int i = 0; i < WM; ++i)
+#pragma unroll
+        for (int j = 0; j < WN; ++j)
+          acc[i][j] = mfma_bf16(afrag[i], bfrag[j], acc[i][j]);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // ---- epilogue ----
+#pragma unroll
+  for (int i = 0; i < WM; ++i) {
+#pragma unroll
+    for (int j = 0; j < WN; ++j) {
+      int64_t col = n0 + wc * WN * 16 + j * 16 + (lane & 15);
+      if (col >= N) continue;
+      float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+      for (int rgi = 0; rgi < 4; ++rgi) {
+        int64_t row = m0 + wr * WM * 16 + i * 16 + (lane >> 4) * 4 + rgi;
+        if (row >= M) continue;
+        float v = acc[i][j][rgi] + bv;
+        if (FUSE_RELU) v = v > 0.f ? v : 0.f;
+        if (OUT_BF16) {
+          uint16_t* out = (uint16_t*)C + row * N + col;
+          if (beta != 0.f) v += beta * bf16_to_f32(*out);
+          *out = f32_to_bf16(v);
+        } else {
+          float* out = (float*)C + row * N + col;
+          if (beta != 0.f) v += beta * *out;
+          *out = v;
+        }
+      }
+    }
+  }
+}
+
+template <bool OUT_BF16, bool FUSE_RELU>
+hipError_t LaunchVariant(const uint16_t* A, const uint16_t* B, void* C,
+                         const float* bias, int64_t M, int64_t N, int64_t K,
+                         float beta, hipStream_t stream) {
+  auto launch = [&](auto kern, int BM, int BN) {
+    int64_t blocks = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
+    hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0, stream, A,
+                       B, C, bias, M, N, K, beta);
+  };
+  if (N >= 128 && M >= 128) {
+    launch(GemmBf16NT<2, 2, 4, 4, OUT_BF16, FUSE_RELU>, 128, 128);
+  } else if (N >= 128) {
+    launch(GemmBf16NT<2, 2, 2, 4, OUT_BF16, FUSE_RELU>, 64, 128);
+  } else if (M >= 128) {
+    launch(GemmBf16NT<2, 2, 4, 2, OUT_BF16, FUSE_RELU>, 128, 64);
+  } else {
+    launch(GemmBf16NT<2, 2, 2, 2, OUT_BF16, FUSE_RELU>, 64, 64);
+  }
+  return hipGetLastError();
+}
+
+}  // namespace
+
+// C[M,N] = A[M,K] * B[N,K]^T (+ beta*C) (+bias) (+relu)
+extern "C" hipError_t stf_gemm_bf16_nt(const void* A, const void* B, void* C,
+                                       const void* bias_f32, int64_t M,
+                                       int64_t N, int64_t K, float beta,
+                                       int out_bf16, int fuse_relu,
+                                       hipStream_t stream) {
+  const uint16_t* a = (const uint16_t*)A;
+  const uint16_t* b = (const uint16_t*)B;
+  const float* bias = (const float*)bias_f32;
+  if (out_bf16) {
+    if (fuse_relu) return LaunchVariant<true, true>(a, b, C, bias, M, N, K, beta, stream);
+    return LaunchVariant<true, false>(a, b, C, bias, M, N, K, beta, stream);
+  }
+  if (fuse_relu) return LaunchVariant<false, true>(a, b, C, bias, M, N, K, beta, stream);
+  return LaunchVariant<false, false>(a, b, C, bias, M, N, K, beta, stream);
+}

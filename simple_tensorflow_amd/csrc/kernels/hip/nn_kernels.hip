@@ -1,0 +1,598 @@
+// NN kernels: bias add/grad, softmax + fused cross-entropy, batch norm
+// fwd/bwd, max/avg pooling fwd/bwd. NHWC, bf16 activations with f32 math.
+// (Replaces reference bias_op_gpu.cu.cc, softmax_op_gpu, xent_op_gpu,
+// fused_batch_norm_op, maxpooling_op_gpu, avgpooling_op_gpu — redesigned for
+// wave64 + LDS per the CDNA4 guide, not translated.)
+#include "hip_common.h"
+
+namespace {
+
+__device__ __forceinline__ float WaveReduceSum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  return v;
+}
+__device__ __forceinline__ float WaveReduceMax(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    v = fmaxf(v, __shfl_down(v, off, 64));
+  return v;
+}
+
+// Block-level (256 threads) reduce; result valid in thread 0.
+__device__ float BlockReduceSum(float v, float* lds4) {
+  v = WaveReduceSum(v);
+  int wid = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) lds4[wid] = v;
+  __syncthreads();
+  if (threadIdx.x == 0) v = lds4[0] + lds4[1] + lds4[2] + lds4[3];
+  return v;
+}
+__device__ float BlockReduceMax(float v, float* lds4) {
+  v = WaveReduceMax(v);
+  int wid = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) lds4[wid] = v;
+  __syncthreads();
+  if (threadIdx.x == 0)
+    v = fmaxf(fmaxf(lds4[0], lds4[1]), fmaxf(lds4[2], lds4[3]));
+  return v;
+}
+
+// ---------------- bias ----------------
+template <typename T>
+__global__ void BiasAddKernel(const T* __restrict__ x,
+                              const T* __restrict__ bias, T* __restrict__ y,
+                              int64_t n, int c) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+    y[i] = (T)((float)x[i] + (float)bias[i % c]);
+}
+
+// column reduction dy[rows, C] -> db[C], partials via f32 atomics into a
+// zeroed f32 scratch (LDS pre-accumulate per block, guide G12).
+template <typename T>
+__global__ void BiasGradKernel(const T* __restrict__ dy, float* __restrict__ db,
+                               int64_t rows, int c) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* part = (float*)smem;  // [c]
+  for (int i = threadIdx.x; i < c; i += blockDim.x) part[i] = 0.f;
+  __syncthreads();
+  int64_t row0 = (int64_t)blockIdx.x * 64;
+  int64_t row1 = min(row0 + 64, rows);
+  // threads sweep the slab row-major; consecutive threads -> consecutive c
+  for (int64_t r = row0; r < row1; ++r) {
+    const T* src = dy + r * c;
+    for (int i = threadIdx.x; i < c; i += blockDim.x)
+      part[i] += (float)src[i];
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < c; i += blockDim.x)
+    if (part[i] != 0.f) atomicAdd(&db[i], part[i]);
+}
+
+// ---------------- softmax / xent ----------------
+// one block per row
+template <typename T, bool LOG>
+__global__ void SoftmaxKernel(const T* __restrict__ x, T* __restrict__ y,
+                              int cols) {
+  __shared__ float lds4[4];
+  __shared__ float stat[2];  // max, sum
+  const T* row = x + (int64_t)blockIdx.x * cols;
+  T* out = y + (int64_t)blockIdx.x * cols;
+  float mx = -3.4e38f;
+  for (int i = threadIdx.x; i < cols; i += blockDim.x)
+    mx = fmaxf(mx, (float)row[i]);
+  mx = BlockReduceMax(mx, lds4);
+  if (threadIdx.x == 0) stat[0] = mx;
+  __syncthreads();
+  mx = stat[0];
+  float sum = 0.f;
+  for (int i = threadIdx.x; i < cols; i += blockDim.x)
+    sum += __expf((float)row[i] - mx);
+  __syncthreads();
+  sum = BlockReduceSum(sum, lds4);
+  if (threadIdx.x == 0) stat[1] = sum;
+  __syncthreads();
+  sum = stat[1];
+  if (LOG) {
+    float lsum = __logf(sum);
+    for (int i = threadIdx.x; i < cols; i += blockDim.x)
+      out[i] = (T)((float)row[i] - mx - lsum);
+  } else {
+    float inv = 1.f / sum;
+    for (int i = threadIdx.x; i < cols; i += blockDim.x)
+      out[i] = (T)(__expf((float)row[i] - mx) * inv);
+  }
+}
+
+// fused sparse softmax xent: loss[row], backprop[row, cols]
+template <typename T, typename L>
+__global__ void SparseXentKernel(const T* __restrict__ logits,
+                                 const L* __restrict__ labels,
+                                 float* __restrict__ loss,
+                                 T* __restrict__ backprop, int cols) {
+  __shared__ float lds4[4];
+  __shared__ float stat[2];
+  int64_t r = blockIdx.x;
+  const T* row = logits + r * cols;
+  T* bp = backprop + r * cols;
+  float mx = -3.4e38f;
+  for (int i = threadIdx.x; i < cols; i += blockDim.x)
+    mx = fmaxf(mx, (float)row[i]);
+  mx = BlockReduceMax(mx, lds4);
+  if (threadIdx.x == 0) stat[0] = mx;
+  __syncthreads();
+  mx = stat[0];
+  float sum = 0.f;
+  for (int i = threadIdx.x; i < cols; i += blockDim.x)
+    sum += __expf((float)row[i] - mx);
+  __syncthreads();
+  sum = BlockReduceSum(sum, lds4);
+  if (threadIdx.x == 0) stat[1] = sum;
+  __syncthreads();
+  sum = stat[1];
+  float lsum = __logf(sum);
+  int lbl = (int)labels[r];
+  for (int i = threadIdx.x; i < cols; i += blockDim.x) {
+    float logp = (float)row[i] - mx - lsum;
+    bp[i] = (T)(__expf(logp) - (i == lbl ? 1.f : 0.f));
+    if (i == lbl && loss) loss[r] = -logp;
+  }
+}
+
+// dense-label fused xent
+template <typename T>
+__global__ void XentKernel(const T* __restrict__ logits,
+                           const T* __restrict__ labels,
+                           float* __restrict__ loss, T* __restrict__ backprop,
+                           int cols) {
+  __shared__ float lds4[4];
+  __shared__ float stat[2];
+  int64_t r = blockIdx.x;
+  const T* row = logits + r * cols;
+  const T* lab = labels + r * cols;
+  T* bp = backprop + r * cols;
+  float mx = -3.4e38f;
+  for (int i = threadIdx.x; i < cols; i += blockDim.x)
+    mx = fmaxf(mx, (float)row[i]);
+  mx = BlockReduceMax(mx, lds4);
+  if (threadIdx.x == 0) stat[0] = mx;
+  __syncthreads();
+  mx = stat[0];
+  float sum = 0.f;
+  for (int i = threadIdx.x; i < cols; i += blockDim.x)
+    sum += __expf((float)row[i] - mx);
+  __syncthreads();
+  sum = BlockReduceSum(sum, lds4);
+  if (threadIdx.x == 0) stat[1] = sum;
+  __syncthreads();
+  sum = stat[1];
+  float lsum = __logf(sum);
+  float lsum_part = 0.f;
+  for (int i = threadIdx.x; i < cols; i += blockDim.x) {
+    float logp = (float)row[i] - mx - lsum;
+    float l = (float)lab[i];
+    bp[i] = (T)(__expf(logp) - l);
+    lsum_part -= l * logp;
+  }
+  __syncthreads();
+  lsum_part = BlockReduceSum(lsum_part, lds4);
+  if (threadIdx.x == 0) loss[r] = lsum_part;
+}
+
+// ---------------- batch norm ----------------
+// pass 1: per-channel sum and sumsq into f32 accumulators [2C] (zeroed).
+template <typename T>
+__global__ void BnStatsKernel(const T* __restrict__ x, float* __restrict__ acc,
+                              int64_t rows, int c) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* s1 = (float*)smem;        // [c]
+  float* s2 = s1 + c;              // [c]
+  for (int i = threadIdx.x; i < c; i += blockDim.x) {
+    s1[i] = 0.f;
+    s2[i] = 0.f;
+  }
+  __syncthreads();
+  int64_t rows_per_block = (rows + gridDim.x - 1) / gridDim.x;
+  int64_t row0 = blockIdx.x * rows_per_block;
+  int64_t row1 = min(row0 + rows_per_block, rows);
+  for (int64_t r = row0; r < row1; ++r) {
+    const T* src = x + r * c;
+    for (int i = threadIdx.x; i < c; i += blockDim.x) {
+      float v = (float)src[i];
+      s1[i] += v;
+      s2[i] += v * v;
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < c; i += blockDim.x) {
+    atomicAdd(&acc[i], s1[i]);
+    atomicAdd(&acc[c + i], s2[i]);
+  }
+}
+
+// pass 2: finalize mean/var; write mean, var, inv_std
+__global__ void BnFinalizeKernel(const float* __restrict__ acc,
+                                 float* __restrict__ mean,
+                                 float* __restrict__ var,
+                                 float* __restrict__ inv_std, int64_t rows,
+                                 int c, float eps) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= c) return;
+  float m = acc[i] / rows;
+  float v = acc[c + i] / rows - m * m;
+  v = v > 0.f ? v : 0.f;
+  mean[i] = m;
+  var[i] = v;
+  inv_std[i] = rsqrtf(v + eps);
+}
+
+// pass 3: normalize (+ optional relu)
+template <typename T, bool RELU>
+__global__ void BnNormKernel(const T* __restrict__ x,
+                             const float* __restrict__ mean,
+                             const float* __restrict__ inv_std,
+                             const float* __restrict__ scale,
+                             const float* __restrict__ offset,
+                             T* __restrict__ y, int64_t n, int c) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    int ch = (int)(i % c);
+    float v = ((float)x[i] - mean[ch]) * inv_std[ch] * scale[ch] + offset[ch];
+    if (RELU) v = v > 0.f ? v : 0.f;
+    y[i] = (T)v;
+  }
+}
+
+// bwd pass 1: per-channel sum(dy), sum(dy * xhat) into acc[2C] (zeroed)
+template <typename T>
+__global__ void BnGradStatsKernel(const T* __restrict__ dy,
+                                  const T* __restrict__ x,
+                                  const float* __restrict__ mean,
+                                  const float* __restrict__ inv_std,
+                                  float* __restrict__ acc, int64_t rows,
+                                  int c) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* s1 = (float*)smem;
+  float* s2 = s1 + c;
+  for (int i = threadIdx.x; i < c; i += blockDim.x) {
+    s1[i] = 0.f;
+    s2[i] = 0.f;
+  }
+  __syncthreads();
+  int64_t rows_per_block = (rows + gridDim.x - 1) / gridDim.x;
+  int64_t row0 = blockIdx.x * rows_per_block;
+  int64_t row1 = min(row0 + rows_per_block, rows);
+  for (int64_t r = row0; r < row1; ++r) {
+    const T* dsrc = dy + r * c;
+    const T* xsrc = x + r * c;
+    for (int i = threadIdx.x; i < c; i += blockDim.x) {
+      float g = (float)dsrc[i];
+      float xhat = ((float)xsrc[i] - mean[i]) * inv_std[i];
+      s1[i] += g;
+      s2[i] += g * xhat;
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < c; i += blockDim.x) {
+    atomicAdd(&acc[i], s1[i]);
+    atomicAdd(&acc[c + i], s2[i]);
+  }
+}
+
+// bwd pass 2: dx = scale*inv_std*(dy - sum_dy/rows - xhat*sum_dy_xhat/rows)
+template <typename T>
+__global__ void BnGradKernel(const T* __restrict__ dy, const T* __restrict__ x,
+                             const float* __restrict__ mean,
+                             const float* __restrict__ inv_std,
+                             const float* __restrict__ scale,
+                             const float* __restrict__ acc,
+                             T* __restrict__ dx, int64_t n, int64_t rows,
+                             int c) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  float inv_rows = 1.f / (float)rows;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    int ch = (int)(i % c);
+    float xhat = ((float)x[i] - mean[ch]) * inv_std[ch];
+    float g = (float)dy[i];
+    dx[i] = (T)(scale[ch] * inv_std[ch] *
+                (g - acc[ch] * inv_rows - xhat * acc[c + ch] * inv_rows));
+  }
+}
+
+// ---------------- pooling ----------------
+struct PoolGeom {
+  int N, H, W, C, kh, kw, sh, sw, ph, pw, P, Q;
+};
+
+template <typename T, bool IS_MAX>
+__global__ void PoolFwdKernel(const T* __restrict__ x, T* __restrict__ y,
+                              PoolGeom g, int64_t total) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    int64_t rem = i;
+    int c = (int)(rem % g.C); rem /= g.C;
+    int q = (int)(rem % g.Q); rem /= g.Q;
+    int p = (int)(rem % g.P); rem /= g.P;
+    int n = (int)rem;
+    float best = IS_MAX ? -3.4e38f : 0.f;
+    int count = 0;
+    for (int kh = 0; kh < g.kh; ++kh) {
+      int ih = p * g.sh - g.ph + kh;
+      if (ih < 0 || ih >= g.H) continue;
+      for (int kw = 0; kw < g.kw; ++kw) {
+        int iw = q * g.sw - g.pw + kw;
+        if (iw < 0 || iw >= g.W) continue;
+        float v = (float)x[((int64_t)(n * g.H + ih) * g.W + iw) * g.C + c];
+        if (IS_MAX) best = fmaxf(best, v);
+        else best += v;
+        ++count;
+      }
+    }
+    y[i] = (T)(IS_MAX ? best : best / count);
+  }
+}
+
+// max pool grad: scan window for the argmax, atomically add into f32 scratch.
+template <typename T>
+__global__ void MaxPoolGradKernel(const T* __restrict__ x,
+                                  const T* __restrict__ dy,
+                                  float* __restrict__ dx_f32, PoolGeom g,
+                                  int64_t total_out) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total_out;
+       i += stride) {
+    int64_t rem = i;
+    int c = (int)(rem % g.C); rem /= g.C;
+    int q = (int)(rem % g.Q); rem /= g.Q;
+    int p = (int)(rem % g.P); rem /= g.P;
+    int n = (int)rem;
+    float best = -3.4e38f;
+    int64_t best_idx = -1;
+    for (int kh = 0; kh < g.kh; ++kh) {
+      int ih = p * g.sh - g.ph + kh;
+      if (ih < 0 || ih >= g.H) continue;
+      for (int kw = 0; kw < g.kw; ++kw) {
+        int iw = q * g.sw - g.pw + kw;
+        if (iw < 0 || iw >= g.W) continue;
+        int64_t idx = ((int64_t)(n * g.H + ih) * g.W + iw) * g.C + c;
+        float v = (float)x[idx];
+        if (v > best) {
+          best = v;
+          best_idx = idx;
+        }
+      }
+    }
+    if (best_idx >= 0) atomicAdd(&dx_f32[best_idx], (float)dy[i]);
+  }
+}
+
+template <typename T>
+__global__ void AvgPoolGradKernel(const T* __restrict__ dy,
+                                  T* __restrict__ dx, PoolGeom g,
+                                  int64_t total_in) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total_in;
+       i += stride) {
+    int64_t rem = i;
+    int c = (int)(rem % g.C); rem /= g.C;
+    int iw = (int)(rem % g.W); rem /= g.W;
+    int ih = (int)(rem % g.H); rem /= g.H;
+    int n = (int)rem;
+    float acc = 0.f;
+    for (int kh = 0; kh < g.kh; ++kh) {
+      int phh = ih + g.ph - kh;
+      if (phh < 0 || phh % g.sh) continue;
+      int p = phh / g.sh;
+      if (p >= g.P) continue;
+      for (int kw = 0; kw < g.kw; ++kw) {
+        int pww = iw + g.pw - kw;
+        if (pww < 0 || pww % g.sw) continue;
+        int q = pww / g.sw;
+        if (q >= g.Q) continue;
+        // count of valid positions for this output window
+        int cnt = 0;
+        for (int a = 0; a < g.kh; ++a) {
+          int t = p * g.sh - g.ph + a;
+          if (t < 0 || t >= g.H) continue;
+          for (int b = 0; b < g.kw; ++b) {
+            int u = q * g.sw - g.pw + b;
+            if (u >= 0 && u < g.W) ++cnt;
+          }
+        }
+        acc += (float)dy[((int64_t)(n * g.P + p) * g.Q + q) * g.C + c] / cnt;
+      }
+    }
+    dx[i] = (T)acc;
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+hipError_t stf_bias_add(int dtype, const void* x, const void* bias, void* y,
+                        int64_t n, int c, hipStream_t stream) {
+  dim3 grid = ElemwiseGrid(n, 256, 4);
+  if (dtype == 0)
+    hipLaunchKernelGGL((BiasAddKernel<float>), grid, dim3(256), 0, stream,
+                       (const float*)x, (const float*)bias, (float*)y, n, c);
+  else
+    hipLaunchKernelGGL((BiasAddKernel<__bf16>), grid, dim3(256), 0, stream,
+                       (const __bf16*)x, (const __bf16*)bias, (__bf16*)y, n, c);
+  return hipGetLastError();
+}
+
+// db_f32 must be zeroed
+hipError_t stf_bias_grad(int dtype, const void* dy, void* db_f32, int64_t rows,
+                         int c, hipStream_t stream) {
+  int64_t blocks = (rows + 63) / 64;
+  if (blocks > 1024) blocks = 1024;
+  size_t lds = (size_t)c * 4;
+  if (dtype == 0)
+    hipLaunchKernelGGL((BiasGradKernel<float>), dim3((uint32_t)blocks),
+                       dim3(256), lds, stream, (const float*)dy,
+                       (float*)db_f32, rows, c);
+  else
+    hipLaunchKernelGGL((BiasGradKernel<__bf16>), dim3((uint32_t)blocks),
+                       dim3(256), lds, stream, (const __bf16*)dy,
+                       (float*)db_f32, rows, c);
+  return hipGetLastError();
+}
+
+hipError_t stf_softmax(int dtype, int log_sm, const void* x, void* y,
+                       int64_t rows, int cols, hipStream_t stream) {
+#define SM(T, L)                                                           \
+  hipLaunchKernelGGL((SoftmaxKernel<T, L>), dim3((uint32_t)rows), dim3(256), \
+                     0, stream, (const T*)x, (T*)y, cols)
+  if (dtype == 0) {
+    if (log_sm) SM(float, true); else SM(float, false);
+  } else {
+    if (log_sm) SM(__bf16, true); else SM(__bf16, false);
+  }
+#undef SM
+  return hipGetLastError();
+}
+
+hipError_t stf_sparse_xent(int dtype, const void* logits, const void* labels,
+                           int labels_i32, void* loss_f32, void* backprop,
+                           int64_t rows, int cols, hipStream_t stream) {
+#define SX(T, L)                                                            \
+  hipLaunchKernelGGL((SparseXentKernel<T, L>), dim3((uint32_t)rows),         \
+                     dim3(256), 0, stream, (const T*)logits,                 \
+                     (const L*)labels, (float*)loss_f32, (T*)backprop, cols)
+  if (dtype == 0) {
+    if (labels_i32) SX(float, int32_t); else SX(float, int64_t);
+  } else {
+    if (labels_i32) SX(__bf16, int32_t); else SX(__bf16, int64_t);
+  }
+#undef SX
+  return hipGetLastError();
+}
+
+hipError_t stf_xent(int dtype, const void* logits, const void* labels,
+                    void* loss_f32, void* backprop, int64_t rows, int cols,
+                    hipStream_t stream) {
+  if (dtype == 0)
+    hipLaunchKernelGGL((XentKernel<float>), dim3((uint32_t)rows), dim3(256), 0,
+                       stream, (const float*)logits, (const float*)labels,
+                       (float*)loss_f32, (float*)backprop, cols);
+  else
+    hipLaunchKernelGGL((XentKernel<__bf16>), dim3((uint32_t)rows), dim3(256),
+                       0, stream, (const __bf16*)logits,
+                       (const __bf16*)labels, (float*)loss_f32,
+                       (__bf16*)backprop, cols);
+  return hipGetLastError();
+}
+
+// acc must be zeroed [2C] f32; outputs mean/var/inv_std [C] f32
+hipError_t stf_bn_fwd(int dtype, const void* x, const void* scale,
+                      const void* offset, float* acc, float* mean, float* var,
+                      float* inv_std, void* y, int64_t rows, int c, float eps,
+                      int fuse_relu, hipStream_t stream) {
+  int blocks = 512;
+  size_t lds = (size_t)c * 8;
+  if (dtype == 0)
+    hipLaunchKernelGGL((BnStatsKernel<float>), dim3(blocks), dim3(256), lds,
+                       stream, (const float*)x, acc, rows, c);
+  else
+    hipLaunchKernelGGL((BnStatsKernel<__bf16>), dim3(blocks), dim3(256), lds,
+                       stream, (const __bf16*)x, acc, rows, c);
+  hipLaunchKernelGGL(BnFinalizeKernel, dim3((c + 255) / 256), dim3(256), 0,
+                     stream, acc, mean, var, inv_std, rows, c, eps);
+  int64_t n = rows * c;
+  dim3 grid = ElemwiseGrid(n, 256, 4);
+#define BNN(T, R)                                                          \
+  hipLaunchKernelGGL((BnNormKernel<T, R>), grid, dim3(256), 0, stream,     \
+                     (const T*)x, mean, inv_std, (const float*)scale,      \
+                     (const float*)offset, (T*)y, n, c)
+  if (dtype == 0) {
+    if (fuse_relu) BNN(float, true); else BNN(float, false);
+  } else {
+    if (fuse_relu) BNN(__bf16, true); else BNN(__bf16, false);
+  }
+#undef BNN
+  return hipGetLastError();
+}
+
+// acc zeroed [2C]; outputs dscale=acc[c..], doffset=acc[0..] are copied out by
+// the wrapper after the kernel (acc[0:c]=sum_dy=doffset, acc[c:2c]=sum_dy_xhat=dscale)
+hipError_t stf_bn_bwd(int dtype, const void* dy, const void* x,
+                      const float* mean, const float* inv_std,
+                      const void* scale, float* acc, void* dx, int64_t rows,
+                      int c, hipStream_t stream) {
+  int blocks = 512;
+  size_t lds = (size_t)c * 8;
+  if (dtype == 0)
+    hipLaunchKernelGGL((BnGradStatsKernel<float>), dim3(blocks), dim3(256),
+                       lds, stream, (const float*)dy, (const float*)x, mean,
+                       inv_std, acc, rows, c);
+  else
+    hipLaunchKernelGGL((BnGradStatsKernel<__bf16>), dim3(blocks), dim3(256),
+                       lds, stream, (const __bf16*)dy, (const __bf16*)x, mean,
+                       inv_std, acc, rows, c);
+  int64_t n = rows * c;
+  dim3 grid = ElemwiseGrid(n, 256, 4);
+  if (dtype == 0)
+    hipLaunchKernelGGL((BnGradKernel<float>), grid, dim3(256), 0, stream,
+                       (const float*)dy, (const float*)x, mean, inv_std,
+                       (const float*)scale, acc, (float*)dx, n, rows, c);
+  else
+    hipLaunchKernelGGL((BnGradKernel<__bf16>), grid, dim3(256), 0, stream,
+                       (const __bf16*)dy, (const __bf16*)x, mean, inv_std,
+                       (const float*)scale, acc, (__bf16*)dx, n, rows, c);
+  return hipGetLastError();
+}
+
+hipError_t stf_pool_fwd(int dtype, int is_max, const void* x, void* y, int N,
+                        int H, int W, int C, int kh, int kw, int sh, int sw,
+                        int ph, int pw, int P, int Q, hipStream_t stream) {
+  PoolGeom g{N, H, W, C, kh, kw, sh, sw, ph, pw, P, Q};
+  int64_t total = (int64_t)N * P * Q * C;
+  dim3 grid = ElemwiseGrid(total, 256, 1);
+#define PF(T, M)                                                            \
+  hipLaunchKernelGGL((PoolFwdKernel<T, M>), grid, dim3(256), 0, stream,     \
+                     (const T*)x, (T*)y, g, total)
+  if (dtype == 0) {
+    if (is_max) PF(float, true); else PF(float, false);
+  } else {
+    if (is_max) PF(__bf16, true); else PF(__bf16, false);
+  }
+#undef PF
+  return hipGetLastError();
+}
+
+// dx_f32 zeroed scratch [N*H*W*C]; wrapper casts to output dtype after.
+hipError_t stf_max_pool_bwd(int dtype, const void* x, const void* dy,
+                            float* dx_f32, int N, int H, int W, int C, int kh,
+                            int kw, int sh, int sw, int ph, int pw, int P,
+                            int Q, hipStream_t stream) {
+  PoolGeom g{N, H, W, C, kh, kw, sh, sw, ph, pw, P, Q};
+  int64_t total = (int64_t)N * P * Q * C;
+  dim3 grid = ElemwiseGrid(total, 256, 1);
+  if (dtype == 0)
+    hipLaunchKernelGGL((MaxPoolGradKernel<float>), grid, dim3(256), 0, stream,
+                       (const float*)x, (const float*)dy, dx_f32, g, total);
+  else
+    hipLaunchKernelGGL((MaxPoolGradKernel<__bf16>), grid, dim3(256), 0, stream,
+                       (const __bf16*)x, (const __bf16*)dy, dx_f32, g, total);
+  return hipGetLastError();
+}
+
+hipError_t stf_avg_pool_bwd(int dtype, const void* dy, void* dx, int N, int H,
+                            int W, int C, int kh, int kw, int sh, int sw,
+                            int ph, int pw, int P, int Q, hipStream_t stream) {
+  PoolGeom g{N, H, W, C, kh, kw, sh, sw, ph, pw, P, Q};
+  int64_t total = (int64_t)N * H * W * C;
+  dim3 grid = ElemwiseGrid(total, 256, 1);
+  if (dtype == 0)
+    hipLaunchKernelGGL((AvgPoolGradKernel<float>), grid, dim3(256), 0, stream,
+                       (const float*)dy, (float*)dx, g, total);
+  else
+    hipLaunchKernelGGL((AvgPoolGradKernel<__bf16>), grid, dim3(256), 0, stream,
+                       (const __bf16*)dy, (__bf16*)dx, g, total);
+  return hipGetLastError();
+}
+
+}  // extern "C"

@@ -1,0 +1,457 @@
+// Data-movement & reduction kernels: 2D transpose (LDS-tiled), generic
+// permute, strided broadcast (Tile), reductions (inner-dim row reduce,
+// full reduce, generic), concat/split along the last dims, optimizer applies,
+// Philox RNG fills. (Replaces reference transpose_functor_gpu,
+// reduction_ops_gpu, tile_ops_gpu, training_ops_gpu, random_op_gpu —
+// wave64-native designs.)
+#include "hip_common.h"
+
+#include "../philox.h"
+
+namespace {
+
+// ---- 2D transpose: out[j,i] = in[i,j]; LDS 64x64 tile with +1-pad ----
+template <typename T>
+__global__ void Transpose2DKernel(const T* __restrict__ in, T* __restrict__ out,
+                                  int64_t rows, int64_t cols) {
+  __shared__ T tile[64][65];
+  int64_t tiles_c = (cols + 63) / 64;
+  int64_t bid = blockIdx.x;
+  int64_t br = bid / tiles_c, bc = bid % tiles_c;
+  int64_t r0 = br * 64, c0 = bc * 64;
+  int tx = threadIdx.x & 63;   // col within tile
+  int ty = threadIdx.x >> 6;   // 4 rows per pass
+  for (int p = 0; p < 16; ++p) {
+    int64_t r = r0 + ty + p * 4;
+    int64_t c = c0 + tx;
+    if (r < rows && c < cols) tile[ty + p * 4][tx] = in[r * cols + c];
+  }
+  __syncthreads();
+  for (int p = 0; p < 16; ++p) {
+    int64_t r = c0 + ty + p * 4;  // output row = input col
+    int64_t c = r0 + tx;          // output col = input row
+    if (r < cols && c < rows) out[r * rows + c] = tile[tx][ty + p * 4];
+  }
+}
+
+// ---- generic permute (rank <= 6) ----
+struct PermArgs {
+  int rank;
+  int64_t out_dims[6];
+  int64_t src_strides[6];  // stride in src for each out dim
+};
+
+template <typename T>
+__global__ void PermuteKernel(const T* __restrict__ in, T* __restrict__ out,
+                              int64_t n, PermArgs args) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    int64_t rem = i, src = 0;
+    for (int d = args.rank - 1; d >= 0; --d) {
+      int64_t c = rem % args.out_dims[d];
+      rem /= args.out_dims[d];
+      src += c * args.src_strides[d];
+    }
+    out[i] = in[src];
+  }
+}
+
+// ---- strided broadcast: out[i] = in[map(i)] (Tile / grad broadcast) ----
+template <typename T>
+__global__ void BcastCopyKernel(const T* __restrict__ in, T* __restrict__ out,
+                                int64_t n, PermArgs args) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    int64_t rem = i, src = 0;
+    for (int d = args.rank - 1; d >= 0; --d) {
+      int64_t c = rem % args.out_dims[d];
+      rem /= args.out_dims[d];
+      src += c * args.src_strides[d];  // stride 0 on broadcast dims
+    }
+    out[i] = in[src];
+  }
+}
+
+// ---- reductions ----
+// full reduce to scalar: two-stage (block partials via atomics on f32)
+template <typename T, int RED>  // 0 sum, 1 max, 2 min
+__global__ void FullReduceKernel(const T* __restrict__ x,
+                                 float* __restrict__ out, int64_t n) {
+  __shared__ float lds4[4];
+  float acc = RED == 0 ? 0.f : (RED == 1 ? -3.4e38f : 3.4e38f);
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    float v = (float)x[i];
+    if (RED == 0) acc += v;
+    else if (RED == 1) acc = fmaxf(acc, v);
+    else acc = fminf(acc, v);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    float o = __shfl_down(acc, off, 64);
+    if (RED == 0) acc += o;
+    else if (RED == 1) acc = fmaxf(acc, o);
+    else acc = fminf(acc, o);
+  }
+  int wid = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) lds4[wid] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float v = lds4[0];
+    for (int w = 1; w < 4; ++w) {
+      if (RED == 0) v += lds4[w];
+      else if (RED == 1) v = fmaxf(v, lds4[w]);
+      else v = fminf(v, lds4[w]);
+    }
+    if (RED == 0) atomicAdd(out, v);
+    else if (RED == 1) {
+      // atomic max on float via CAS loop
+      float old = *out;
+      while (v > old) {
+        float assumed = old;
+        old = __uint_as_float(atomicCAS((unsigned*)out,
+                                        __float_as_uint(assumed),
+                                        __float_as_uint(v)));
+        if (old == assumed) break;
+      }
+    } else {
+      float old = *out;
+      while (v < old) {
+        float assumed = old;
+        old = __uint_as_float(atomicCAS((unsigned*)out,
+                                        __float_as_uint(assumed),
+                                        __float_as_uint(v)));
+        if (old == assumed) break;
+      }
+    }
+  }
+}
+
+// row reduce: [rows, inner] -> [rows]; one block per row chunk
+template <typename T, int RED>
+__global__ void RowReduceKernel(const T* __restrict__ x, float* __restrict__ y,
+                                int64_t rows, int64_t inner) {
+  __shared__ float lds4[4];
+  int64_t r = blockIdx.x;
+  if (r >= rows) return;
+  const T* row = x + r * inner;
+  float acc = RED == 0 ? 0.f : (RED == 1 ? -3.4e38f : 3.4e38f);
+  for (int64_t i = threadIdx.x; i < inner; i += blockDim.x) {
+    float v = (float)row[i];
+    if (RED == 0) acc += v;
+    else if (RED == 1) acc = fmaxf(acc, v);
+    else acc = fminf(acc, v);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    float o = __shfl_down(acc, off, 64);
+    if (RED == 0) acc += o;
+    else if (RED == 1) acc = fmaxf(acc, o);
+    else acc = fminf(acc, o);
+  }
+  int wid = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) lds4[wid] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float v = lds4[0];
+    for (int w = 1; w < 4; ++w) {
+      if (RED == 0) v += lds4[w];
+      else if (RED == 1) v = fmaxf(v, lds4[w]);
+      else v = fminf(v, lds4[w]);
+    }
+    y[r] = v;
+  }
+}
+
+// outer reduce: [outer, inner] -> [inner] (column sums; e.g. bias-style)
+template <typename T, int RED>
+__global__ void ColReduceKernel(const T* __restrict__ x, float* __restrict__ y,
+                                int64_t outer, int64_t inner) {
+  // grid-stride over inner; each thread owns one column strip
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t c = blockIdx.x * blockDim.x + threadIdx.x; c < inner;
+       c += stride) {
+    float acc = RED == 0 ? 0.f : (RED == 1 ? -3.4e38f : 3.4e38f);
+    for (int64_t r = 0; r < outer; ++r) {
+      float v = (float)x[r * inner + c];
+      if (RED == 0) acc += v;
+      else if (RED == 1) acc = fmaxf(acc, v);
+      else acc = fminf(acc, v);
+    }
+    y[c] = acc;
+  }
+}
+
+// ---- optimizer applies (f32 params; grad may be f32 or bf16) ----
+template <typename G>
+__global__ void ApplySgdKernel(float* __restrict__ var,
+                               const float* __restrict__ lr,
+                               const G* __restrict__ grad, int64_t n) {
+  float l = lr[0];
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+    var[i] -= l * (float)grad[i];
+}
+
+template <typename G>
+__global__ void ApplyMomentumKernel(float* __restrict__ var,
+                                    float* __restrict__ accum,
+                                    const float* __restrict__ lr,
+                                    const G* __restrict__ grad,
+                                    const float* __restrict__ momentum,
+                                    int nesterov, int64_t n) {
+  float l = lr[0], mom = momentum[0];
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    float g = (float)grad[i];
+    float a = accum[i] * mom + g;
+    accum[i] = a;
+    var[i] -= nesterov ? l * (g + mom * a) : l * a;
+  }
+}
+
+template <typename G>
+__global__ void ApplyAdamKernel(float* __restrict__ var, float* __restrict__ m,
+                                float* __restrict__ v,
+                                const float* __restrict__ b1p_p,
+                                const float* __restrict__ b2p_p,
+                                const float* __restrict__ lr_p,
+                                const float* __restrict__ b1_p,
+                                const float* __restrict__ b2_p,
+                                const float* __restrict__ eps_p,
+                                const G* __restrict__ grad, int64_t n) {
+  float b1p = b1p_p[0], b2p = b2p_p[0], lr = lr_p[0];
+  float b1 = b1_p[0], b2 = b2_p[0], eps = eps_p[0];
+  float alpha = lr * sqrtf(1.f - b2p) / (1.f - b1p);
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    float g = (float)grad[i];
+    float mi = m[i] + (g - m[i]) * (1.f - b1);
+    float vi = v[i] + (g * g - v[i]) * (1.f - b2);
+    m[i] = mi;
+    v[i] = vi;
+    var[i] -= alpha * mi / (sqrtf(vi) + eps);
+  }
+}
+
+// ---- Philox RNG fills ----
+__global__ void RandomUniformKernel(uint64_t seed, uint64_t offset,
+                                    float* __restrict__ out, int64_t n,
+                                    int as_bf16) {
+  int64_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = tid; i * 4 < n; i += stride) {
+    stf::random::Philox4x32 rng(seed, offset + i);
+    uint32_t r[4];
+    rng.Next(r);
+    int64_t base = i * 4;
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      if (base + e < n) {
+        float v = stf::random::Uint32ToFloat01(r[e]);
+        if (as_bf16) ((__bf16*)out)[base + e] = (__bf16)v;
+        else out[base + e] = v;
+      }
+    }
+  }
+}
+
+__global__ void RandomNormalKernel(uint64_t seed, uint64_t offset,
+                                   float* __restrict__ out, int64_t n,
+                                   int as_bf16, int truncated) {
+  int64_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = tid; i * 4 < n; i += stride) {
+    stf::random::Philox4x32 rng(seed, offset + i);
+    uint32_t r[4];
+    float z[4];
+    rng.Next(r);
+    stf::random::BoxMuller(r[0], r[1], &z[0], &z[1]);
+    stf::random::BoxMuller(r[2], r[3], &z[2], &z[3]);
+    if (truncated) {
+      // redraw per element until |z| < 2 (bounded attempts)
+      for (int e = 0; e < 4; ++e) {
+        int tries = 0;
+        while (fabsf(z[e]) >= 2.f && tries < 16) {
+          rng.Next(r);
+          float z2;
+          stf::random::BoxMuller(r[0], r[1], &z[e], &z2);
+          ++tries;
+        }
+        if (fabsf(z[e]) >= 2.f) z[e] = 0.f;
+      }
+    }
+    int64_t base = i * 4;
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      if (base + e < n) {
+        if (as_bf16) ((__bf16*)out)[base + e] = (__bf16)z[e];
+        else out[base + e] = z[e];
+      }
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+hipError_t stf_transpose2d(int elem_size, const void* in, void* out,
+                           int64_t rows, int64_t cols, hipStream_t stream) {
+  int64_t blocks = ((rows + 63) / 64) * ((cols + 63) / 64);
+  if (elem_size == 2)
+    hipLaunchKernelGGL((Transpose2DKernel<uint16_t>), dim3((uint32_t)blocks),
+                       dim3(256), 0, stream, (const uint16_t*)in,
+                       (uint16_t*)out, rows, cols);
+  else if (elem_size == 4)
+    hipLaunchKernelGGL((Transpose2DKernel<uint32_t>), dim3((uint32_t)blocks),
+                       dim3(256), 0, stream, (const uint32_t*)in,
+                       (uint32_t*)out, rows, cols);
+  else
+    hipLaunchKernelGGL((Transpose2DKernel<uint64_t>), dim3((uint32_t)blocks),
+                       dim3(256), 0, stream, (const uint64_t*)in,
+                       (uint64_t*)out, rows, cols);
+  return hipGetLastError();
+}
+
+hipError_t stf_permute(int elem_size, const void* in, void* out, int64_t n,
+                       int rank, const int64_t* out_dims,
+                       const int64_t* src_strides, hipStream_t stream) {
+  PermArgs args;
+  args.rank = rank;
+  for (int i = 0; i < rank && i < 6; ++i) {
+    args.out_dims[i] = out_dims[i];
+    args.src_strides[i] = src_strides[i];
+  }
+  dim3 grid = ElemwiseGrid(n, 256, 4);
+  if (elem_size == 2)
+    hipLaunchKernelGGL((PermuteKernel<uint16_t>), grid, dim3(256), 0, stream,
+                       (const uint16_t*)in, (uint16_t*)out, n, args);
+  else if (elem_size == 4)
+    hipLaunchKernelGGL((PermuteKernel<uint32_t>), grid, dim3(256), 0, stream,
+                       (const uint32_t*)in, (uint32_t*)out, n, args);
+  else
+    hipLaunchKernelGGL((PermuteKernel<uint64_t>), grid, dim3(256), 0, stream,
+                       (const uint64_t*)in, (uint64_t*)out, n, args);
+  return hipGetLastError();
+}
+
+hipError_t stf_full_reduce(int dtype, int red, const void* x, float* out_f32,
+                           int64_t n, hipStream_t stream) {
+  dim3 grid = ElemwiseGrid(n, 256, 8);
+#define FR(T, R)                                                           \
+  hipLaunchKernelGGL((FullReduceKernel<T, R>), grid, dim3(256), 0, stream, \
+                     (const T*)x, out_f32, n)
+  if (dtype == 0) {
+    if (red == 0) FR(float, 0); else if (red == 1) FR(float, 1); else FR(float, 2);
+  } else {
+    if (red == 0) FR(__bf16, 0); else if (red == 1) FR(__bf16, 1); else FR(__bf16, 2);
+  }
+#undef FR
+  return hipGetLastError();
+}
+
+hipError_t stf_row_reduce(int dtype, int red, const void* x, float* y,
+                          int64_t rows, int64_t inner, hipStream_t stream) {
+#define RR(T, R)                                                            \
+  hipLaunchKernelGGL((RowReduceKernel<T, R>), dim3((uint32_t)rows),          \
+                     dim3(256), 0, stream, (const T*)x, y, rows, inner)
+  if (dtype == 0) {
+    if (red == 0) RR(float, 0); else if (red == 1) RR(float, 1); else RR(float, 2);
+  } else {
+    if (red == 0) RR(__bf16, 0); else if (red == 1) RR(__bf16, 1); else RR(__bf16, 2);
+  }
+#undef RR
+  return hipGetLastError();
+}
+
+hipError_t stf_col_reduce(int dtype, int red, const void* x, float* y,
+                          int64_t outer, int64_t inner, hipStream_t stream) {
+  dim3 grid = ElemwiseGrid(inner, 256, 1);
+#define CR(T, R)                                                            \
+  hipLaunchKernelGGL((ColReduceKernel<T, R>), grid, dim3(256), 0, stream,   \
+                     (const T*)x, y, outer, inner)
+  if (dtype == 0) {
+    if (red == 0) CR(float, 0); else if (red == 1) CR(float, 1); else CR(float, 2);
+  } else {
+    if (red == 0) CR(__bf16, 0); else if (red == 1) CR(__bf16, 1); else CR(__bf16, 2);
+  }
+#undef CR
+  return hipGetLastError();
+}
+
+hipError_t stf_bcast_copy(int elem_size, const void* in, void* out, int64_t n,
+                          int rank, const int64_t* out_dims,
+                          const int64_t* src_strides, hipStream_t stream) {
+  return stf_permute(elem_size, in, out, n, rank, out_dims, src_strides,
+                     stream);
+}
+
+hipError_t stf_apply_sgd(int grad_bf16, void* var, const void* lr,
+                         const void* grad, int64_t n, hipStream_t stream) {
+  dim3 grid = ElemwiseGrid(n, 256, 4);
+  if (grad_bf16)
+    hipLaunchKernelGGL((ApplySgdKernel<__bf16>), grid, dim3(256), 0, stream,
+                       (float*)var, (const float*)lr, (const __bf16*)grad, n);
+  else
+    hipLaunchKernelGGL((ApplySgdKernel<float>), grid, dim3(256), 0, stream,
+                       (float*)var, (const float*)lr, (const float*)grad, n);
+  return hipGetLastError();
+}
+
+hipError_t stf_apply_momentum(int grad_bf16, void* var, void* accum,
+                              const void* lr, const void* grad,
+                              const void* momentum, int nesterov, int64_t n,
+                              hipStream_t stream) {
+  dim3 grid = ElemwiseGrid(n, 256, 4);
+  if (grad_bf16)
+    hipLaunchKernelGGL((ApplyMomentumKernel<__bf16>), grid, dim3(256), 0,
+                       stream, (float*)var, (float*)accum, (const float*)lr,
+                       (const __bf16*)grad, (const float*)momentum, nesterov,
+                       n);
+  else
+    hipLaunchKernelGGL((ApplyMomentumKernel<float>), grid, dim3(256), 0,
+                       stream, (float*)var, (float*)accum, (const float*)lr,
+                       (const float*)grad, (const float*)momentum, nesterov,
+                       n);
+  return hipGetLastError();
+}
+
+hipError_t stf_apply_adam(int grad_bf16, void* var, void* m, void* v,
+                          const void* b1p, const void* b2p, const void* lr,
+                          const void* b1, const void* b2, const void* eps,
+                          const void* grad, int64_t n, hipStream_t stream) {
+  dim3 grid = ElemwiseGrid(n, 256, 4);
+  if (grad_bf16)
+    hipLaunchKernelGGL((ApplyAdamKernel<__bf16>), grid, dim3(256), 0, stream,
+                       (float*)var, (float*)m, (float*)v, (const float*)b1p,
+                       (const float*)b2p, (const float*)lr, (const float*)b1,
+                       (const float*)b2, (const float*)eps,
+                       (const __bf16*)grad, n);
+  else
+    hipLaunchKernelGGL((ApplyAdamKernel<float>), grid, dim3(256), 0, stream,
+                       (float*)var, (float*)m, (float*)v, (const float*)b1p,
+                       (const float*)b2p, (const float*)lr, (const float*)b1,
+                       (const float*)b2, (const float*)eps,
+                       (const float*)grad, n);
+  return hipGetLastError();
+}
+
+hipError_t stf_random_uniform(uint64_t seed, uint64_t offset, void* out,
+                              int64_t n, int as_bf16, hipStream_t stream) {
+  dim3 grid = ElemwiseGrid(n, 256, 4);
+  hipLaunchKernelGGL(RandomUniformKernel, grid, dim3(256), 0, stream, seed,
+                     offset, (float*)out, n, as_bf16);
+  return hipGetLastError();
+}
+
+hipError_t stf_random_normal(uint64_t seed, uint64_t offset, void* out,
+                             int64_t n, int as_bf16, int truncated,
+                             hipStream_t stream) {
+  dim3 grid = ElemwiseGrid(n, 256, 4);
+  hipLaunchKernelGGL(RandomNormalKernel, grid, dim3(256), 0, stream, seed,
+                     offset, (float*)out, n, as_bf16, truncated);
+  return hipGetLastError();
+}
+
+}  // extern "C"

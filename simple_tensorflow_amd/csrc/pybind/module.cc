@@ -198,6 +198,12 @@ class PySession {
     return out;
   }
 
+  void Sync() {
+    py::gil_scoped_release release;
+    Status s = sess_.SyncAllDevices();
+    if (!s.ok()) throw std::runtime_error(s.ToString());
+  }
+
   int NumGpus() {
     int n = 0;
     for (auto& d : sess_.device_mgr()->devices())
@@ -219,6 +225,7 @@ PYBIND11_MODULE(_core, m) {
       .def("create", &PySession::Create)
       .def("extend", &PySession::Extend)
       .def("run", &PySession::Run)
+      .def("sync", &PySession::Sync)
       .def("num_gpus", &PySession::NumGpus);
 
   m.def("list_ops", []() {

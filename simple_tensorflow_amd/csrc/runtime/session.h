@@ -36,6 +36,11 @@ class DirectSession {
              std::vector<Tensor>* outputs);
 
   DeviceMgr* device_mgr() { return &devices_; }
+  // Blocks until all device work is complete (bench timing bracket).
+  Status SyncAllDevices() {
+    for (auto& d : devices_.devices()) STF_RETURN_IF_ERROR(d->Sync());
+    return Status::OK();
+  }
   void* resource_mgr() { return resource_mgr_; }
 
  private:

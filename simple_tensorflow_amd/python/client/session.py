@@ -106,6 +106,10 @@ class Session(object):
     def num_gpus(self):
         return self._core.num_gpus()
 
+    def sync(self):
+        """Block until all enqueued device work completes."""
+        self._core.sync()
+
 
 class InteractiveSession(Session):
     def __init__(self, *a, **kw):

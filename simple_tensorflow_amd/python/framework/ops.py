@@ -208,6 +208,11 @@ class Operation(object):
             k, v = self.attrs[name]
             if k == 'type':
                 return dtypes.as_dtype(v)
+            if k == 'list':
+                for key in ('i', 'f', 's', 'b', 'type'):
+                    if v.get(key):
+                        return list(v[key])
+                return []
             return v
         od = op_defs().get(self.type)
         if od:

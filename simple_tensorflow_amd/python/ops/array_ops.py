@@ -127,7 +127,17 @@ def _filled(shape_arg, value, dtype, name):
         if sv is not None:
             t.set_shape([int(d) for d in sv])
         return t
-    return ops.constant(value, dtype=dt, shape=list(shape_arg), name=name)
+    dims = [int(d) for d in shape_arg]
+    n = 1
+    for d in dims:
+        n *= d
+    if n > 256:
+        # Fill op instead of a materialized Const proto.
+        t = apply_op('Fill', convert_to_tensor(dims, dtype=dtypes.int32),
+                     ops.constant(value, dtype=dt), name=name)
+        t.set_shape(dims)
+        return t
+    return ops.constant(value, dtype=dt, shape=dims, name=name)
 
 
 def fill(dims, value, name=None):

@@ -104,8 +104,10 @@ def _fused_batch_norm_grad(op, grad_y, *rest):
 def _batch_norm_mi_grad(op, grad_y, *rest):
     dx, dscale, doffset = apply_op(
         'BatchNormMiGrad', grad_y, op.inputs[0], op.inputs[1],
-        op.outputs[1], op.outputs[2], epsilon=op.get_attr('epsilon'))
+        op.outputs[1], op.outputs[3], epsilon=op.get_attr('epsilon'))
     dx.set_shape(op.inputs[0]._shape)
+    dscale.set_shape(op.inputs[1]._shape)
+    doffset.set_shape(op.inputs[2]._shape)
     return dx, dscale, doffset
 
 

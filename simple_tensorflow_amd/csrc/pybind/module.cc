@@ -217,6 +217,11 @@ class PySession {
 
 }  // namespace
 
+namespace stf {
+std::string RcclGetUniqueId();
+Status RcclInit(int nranks, int rank, const std::string& id_bytes);
+}
+
 PYBIND11_MODULE(_core, m) {
   m.doc() = "simple_tensorflow_amd core runtime (MI355X-native)";
 
@@ -234,6 +239,13 @@ PYBIND11_MODULE(_core, m) {
       out[py::str(name)] = OpDefToPy(*OpRegistry::Global()->LookUp(name));
     }
     return out;
+  });
+  m.def("rccl_get_unique_id", []() {
+    return py::bytes(stf::RcclGetUniqueId());
+  });
+  m.def("rccl_init", [](int nranks, int rank, py::bytes id) {
+    Status s = stf::RcclInit(nranks, rank, std::string(id));
+    if (!s.ok()) throw std::runtime_error(s.ToString());
   });
   m.def("has_gpu", []() {
     // cheap probe without creating a session

@@ -1,0 +1,111 @@
+"""Data-parallel training over RCCL/xGMI — one process per GPU.
+
+MI355X-native replacement for the reference's PS/SyncReplicas gradient path
+(SURVEY.md §2.3): bucketed all-reduce of gradients over the 7 xGMI links,
+deterministic cross-rank enqueue order via control-dependency chaining.
+
+Bootstrap: the RCCL unique id is exchanged over a torch.distributed gloo
+group (CPU-only; torchrun provides MASTER_ADDR/PORT) — torch is used ONLY
+for rendezvous/barrier, never for compute.
+"""
+import os
+
+import numpy as np
+
+import simple_tensorflow_amd as tf
+from simple_tensorflow_amd import _core
+from simple_tensorflow_amd.python.framework.ops import apply_op
+from simple_tensorflow_amd.python.framework import ops as fw_ops
+
+
+class Comm(object):
+    def __init__(self, world, rank, torch_dist):
+        self.world = world
+        self.rank = rank
+        self._td = torch_dist
+
+    def barrier(self):
+        self._td.barrier()
+
+    def max_scalar(self, v):
+        import torch
+        t = torch.tensor([float(v)])
+        self._td.all_reduce(t, op=self._td.ReduceOp.MAX)
+        return float(t.item())
+
+    def broadcast_variables(self, sess, var_list=None):
+        """Sync all ranks' variables to rank 0's values."""
+        if var_list is None:
+            var_list = tf.global_variables()
+        assigns = []
+        prev = None
+        for v in var_list:
+            ref = v._as_graph_element()
+            if ref.dtype.name not in ('float32', 'bfloat16', 'float16'):
+                continue
+            b = apply_op('RcclBroadcast', v.value(), root=0)
+            if prev is not None:
+                b.op._add_control_input(prev)  # fixed cross-rank order
+            asn = tf.assign(ref, b)
+            prev = asn.op
+            assigns.append(asn.op)
+        from simple_tensorflow_amd.python.ops import control_flow_ops
+        group = control_flow_ops.group(*assigns)
+        sess.run(group)
+        self.barrier()
+
+
+def init(world, rank):
+    """Initialize gloo rendezvous + the RCCL communicator."""
+    import torch.distributed as td
+    if not td.is_initialized():
+        os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+        td.init_process_group('gloo', rank=rank, world_size=world)
+    import torch
+    if rank == 0:
+        uid = _core.rccl_get_unique_id()
+        buf = torch.tensor(list(uid), dtype=torch.uint8)
+    else:
+        buf = torch.zeros(128, dtype=torch.uint8)
+    td.broadcast(buf, src=0)
+    uid = bytes(buf.tolist())
+    _core.rccl_init(world, rank, uid)
+    return Comm(world, rank, td)
+
+
+class DistributedOptimizer(object):
+    """Wraps an optimizer: all-reduce (sum) gradients across ranks, scaled by
+    1/world, with a fixed cross-rank collective order (control-dep chain)."""
+
+    def __init__(self, opt, world):
+        self._opt = opt
+        self._world = world
+
+    def compute_gradients(self, loss, var_list=None, **kw):
+        gvs = self._opt.compute_gradients(loss, var_list=var_list, **kw)
+        out = []
+        prev = None
+        scale = 1.0 / self._world
+        for g, v in gvs:
+            if g is None:
+                out.append((g, v))
+                continue
+            g_scaled = g * scale
+            if prev is not None:
+                g_scaled.op._add_control_input(prev)
+            red = apply_op('RcclAllReduce', g_scaled,
+                           num_devices=self._world)
+            red.set_shape(g._shape)
+            prev = red.op
+            out.append((red, v))
+        return out
+
+    def apply_gradients(self, grads_and_vars, global_step=None, name=None):
+        return self._opt.apply_gradients(grads_and_vars, global_step, name)
+
+    def minimize(self, loss, global_step=None, var_list=None, **kw):
+        gvs = self.compute_gradients(loss, var_list=var_list)
+        return self.apply_gradients(gvs, global_step=global_step)
+
+    def __getattr__(self, name):
+        return getattr(self._opt, name)

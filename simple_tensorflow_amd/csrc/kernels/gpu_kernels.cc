@@ -867,6 +867,14 @@ class GpuReduceOp : public OpKernel {
       }
     }
     int64_t out_n = out_shape.num_elements();
+    if (axes_v.empty()) {
+      // no reduction: reshape copy
+      Tensor* out = ctx->allocate_output(0, out_shape);
+      OP_HIP_OK(ctx, hipMemcpyAsync(out->raw_data(), x.raw_data(),
+                                    x.TotalBytes(), hipMemcpyDeviceToDevice,
+                                    s));
+      return;
+    }
     Tensor f32_out = ctx->allocate_temp(DT_FLOAT, TensorShape({out_n}));
     // classify
     bool all_red = count == x.NumElements();

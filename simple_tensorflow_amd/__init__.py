@@ -27,6 +27,7 @@ from simple_tensorflow_amd.python.ops import (  # noqa: F401
 )
 from simple_tensorflow_amd.python.training import optimizer as _optimizer
 from simple_tensorflow_amd.python.training import training_util as _training_util
+from simple_tensorflow_amd.python.training import saver as _saver
 
 # ---- dtypes ----
 float32 = _dtypes.float32
@@ -241,6 +242,11 @@ class _TrainModule(object):
     polynomial_decay = staticmethod(_training_util.polynomial_decay)
     piecewise_constant = staticmethod(_training_util.piecewise_constant)
     ExponentialMovingAverage = _training_util.ExponentialMovingAverage
+    Saver = _saver.Saver
+    latest_checkpoint = staticmethod(_saver.latest_checkpoint)
+    get_checkpoint_state = staticmethod(_saver.get_checkpoint_state)
+    update_checkpoint_state = staticmethod(_saver.update_checkpoint_state)
+    checkpoint_exists = staticmethod(_saver.checkpoint_exists)
 
 
 train = _TrainModule()

@@ -123,6 +123,16 @@ Status DirectSession::BuildExecutors(const std::vector<std::string>& feeds,
   for (Node* n : order) {
     std::string req = CanonicalDevice(n->def.device);
     Device* dev = nullptr;
+    // Shape-carrying int32 constants live on the host: GPU kernels take
+    // shape/axes/perm args in host memory, and a device-resident int32 Const
+    // would force a per-step d2h sync at every consumer.
+    if (req.empty() && n->op() == "Const") {
+      DataType dt;
+      if (GetAttrType(n->def, "dtype", &dt) && dt == DT_INT32) {
+        n->assigned_device = CanonicalDevice(cpu->name());
+        continue;
+      }
+    }
     if (!req.empty()) {
       dev = devices_.LookUp(req);
       if (!dev && StrStartsWith(req, "GPU") && gpu0) dev = gpu0;

@@ -481,8 +481,10 @@ def constant(value, dtype=None, shape=None, name='Const'):
         return value
     if dtype is not None:
         dtype = dtypes.as_dtype(dtype)
+    is_str_list = isinstance(value, (list, tuple)) and len(value) > 0 and \
+        all(isinstance(v, (str, bytes)) for v in value)
     if dtype is dtypes.string or (
-            dtype is None and isinstance(value, (str, bytes))):
+            dtype is None and (isinstance(value, (str, bytes)) or is_str_list)):
         vals = value
         if isinstance(vals, (str, bytes)):
             vals = [vals]

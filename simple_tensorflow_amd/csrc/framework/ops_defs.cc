@@ -215,6 +215,17 @@ REGISTER_OP("SaveV2").Input("prefix: string").Input("tensor_names: string").Inpu
 REGISTER_OP("RestoreV2").Input("prefix: string").Input("tensor_names: string").Input("shape_and_slices: string").Output("tensors: dtypes").Attr("dtypes: list(type)").SetIsStateful();
 REGISTER_OP("MergeV2Checkpoints").Input("checkpoint_prefixes: string").Input("destination_prefix: string").Attr("delete_old_dirs: bool = true").SetIsStateful();
 
+// ------------------------------- queues ------------------------------------
+REGISTER_OP("FIFOQueue").Output("handle: Ref(string)").Attr("component_types: list(type)").Attr("shapes: list(shape) = []").Attr("capacity: int = -1").Attr("container: string = ''").Attr("shared_name: string = ''").SetIsStateful();
+REGISTER_OP("RandomShuffleQueue").Output("handle: Ref(string)").Attr("component_types: list(type)").Attr("shapes: list(shape) = []").Attr("capacity: int = -1").Attr("min_after_dequeue: int = 0").Attr("seed: int = 0").Attr("seed2: int = 0").Attr("container: string = ''").Attr("shared_name: string = ''").SetIsStateful();
+REGISTER_OP("PaddingFIFOQueue").Output("handle: Ref(string)").Attr("component_types: list(type)").Attr("shapes: list(shape) = []").Attr("capacity: int = -1").Attr("container: string = ''").Attr("shared_name: string = ''").SetIsStateful();
+REGISTER_OP("QueueEnqueue").Input("handle: Ref(string)").Input("components: Tcomponents").Attr("Tcomponents: list(type)").Attr("timeout_ms: int = -1").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("QueueEnqueueMany").Input("handle: Ref(string)").Input("components: Tcomponents").Attr("Tcomponents: list(type)").Attr("timeout_ms: int = -1").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("QueueDequeue").Input("handle: Ref(string)").Output("components: component_types").Attr("component_types: list(type)").Attr("timeout_ms: int = -1").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("QueueDequeueMany").Input("handle: Ref(string)").Input("n: int32").Output("components: component_types").Attr("component_types: list(type)").Attr("timeout_ms: int = -1").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("QueueClose").Input("handle: Ref(string)").Attr("cancel_pending_enqueues: bool = false").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("QueueSize").Input("handle: Ref(string)").Output("size: int32").SetIsStateful().SetAllowsUninitializedInput();
+
 // ----------------------------- collectives ---------------------------------
 // MI355X-native: RCCL collectives over xGMI as first-class graph ops
 // (the reference had no collectives — §2.3 of SURVEY.md; gradient

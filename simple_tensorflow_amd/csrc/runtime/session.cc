@@ -139,12 +139,19 @@ Status DirectSession::BuildExecutors(const std::vector<std::string>& feeds,
       if (!dev) dev = cpu;  // soft placement
       if (!kernel_available(n, dev)) dev = cpu;
     } else {
-      // Colocate with first data input when it exists (keeps variables and
-      // their updates together); else default to GPU when a kernel exists.
-      const Edge* in0 = n->input_edge(0);
-      if (in0 && !in0->src->assigned_device.empty()) {
-        dev = devices_.LookUp(in0->src->assigned_device);
-        if (dev && !kernel_available(n, dev)) dev = nullptr;
+      // Colocate with a ref-typed input's producer (keeps variables and their
+      // updates/readers together); else default to GPU when a kernel exists.
+      for (auto* e : n->in_edges) {
+        if (e->IsControl()) continue;
+        if (e->src_output < (int)e->src->out_is_ref.size() &&
+            e->src->out_is_ref[e->src_output] &&
+            !e->src->assigned_device.empty()) {
+          Device* cand = devices_.LookUp(e->src->assigned_device);
+          if (cand && kernel_available(n, cand)) {
+            dev = cand;
+            break;
+          }
+        }
       }
       if (!dev) {
         if (gpu0 && kernel_available(n, gpu0)) dev = gpu0;

@@ -155,7 +155,24 @@ static std::mutex& registry_mu() {
 
 void OpRegistry::Register(const OpDef& def) {
   std::lock_guard<std::mutex> l(registry_mu());
-  ops_[def.name] = def;
+  OpDef fixed = def;
+  // "arg: A" where A is a list(type) attr means a variadic input whose
+  // element types come from that list — reclassify type_attr as
+  // type_list_attr now that all attrs are known.
+  auto fixup = [&](std::vector<OpDef::ArgDef>* args) {
+    for (auto& a : *args) {
+      if (!a.type_attr.empty()) {
+        const OpDef::AttrDef* ad = fixed.FindAttr(a.type_attr);
+        if (ad && ad->type == "list(type)") {
+          a.type_list_attr = a.type_attr;
+          a.type_attr.clear();
+        }
+      }
+    }
+  };
+  fixup(&fixed.input_arg);
+  fixup(&fixed.output_arg);
+  ops_[def.name] = fixed;
 }
 
 const OpDef* OpRegistry::LookUp(const std::string& op) const {

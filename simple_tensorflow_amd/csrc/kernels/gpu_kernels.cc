@@ -15,6 +15,8 @@ extern "C" {
 hipError_t stf_gemm_bf16_nt(const void*, const void*, void*, const void*,
                             int64_t, int64_t, int64_t, float, int, int,
                             hipStream_t);
+hipError_t stf_gemm_bf16_nt_splitk(const void*, const void*, void*, int64_t,
+                                   int64_t, int64_t, int, hipStream_t);
 hipError_t stf_gemm_f32_nt(const void*, const void*, void*, int64_t, int64_t,
                            int64_t, hipStream_t);
 hipError_t stf_unary(int, int, const void*, void*, int64_t, hipStream_t);
@@ -30,7 +32,7 @@ hipError_t stf_addn(int, const void* const*, int, void*, int64_t, hipStream_t);
 hipError_t stf_fill_f32(void*, float, int64_t, int, hipStream_t);
 hipError_t stf_scale(int, const void*, void*, int64_t, float, hipStream_t);
 hipError_t stf_im2col_bf16(const void*, void*, int, int, int, int, int, int,
-                           int, int, int, int, int, int, hipStream_t);
+                           int, int, int, int, int, int, int64_t, hipStream_t);
 hipError_t stf_col2im_bf16(const void*, void*, int, int, int, int, int, int,
                            int, int, int, int, int, int, hipStream_t);
 hipError_t stf_bias_add(int, const void*, const void*, void*, int64_t, int,
@@ -94,6 +96,33 @@ namespace {
   }
 
 inline int DtypeCode(DataType dt) { return dt == DT_FLOAT ? 0 : 1; }
+
+// Choose a split-K factor so the grid covers the 256 CUs (guide: blocks
+// ~= 0.5-2x CU count); 1 = no split.
+inline int PickSplitK(int64_t M, int64_t N, int64_t K) {
+  int64_t tiles = ((M + 127) / 128) * ((N + 127) / 128);
+  if (tiles >= 256 || K < 4096) return 1;
+  int64_t want = 512 / (tiles ? tiles : 1);
+  int64_t maxk = K / 512;  // keep >= 8 K-iters per slice
+  if (maxk < 1) maxk = 1;
+  int64_t sk = std::min(want, maxk);
+  return (int)(sk < 1 ? 1 : sk);
+}
+
+// GEMM helper: picks plain vs split-K (f32 scratch + cast) automatically.
+inline hipError_t GemmBf16Auto(OpKernelContext* ctx, const void* A,
+                               const void* B, void* C_bf16, int64_t M,
+                               int64_t N, int64_t K, hipStream_t s) {
+  int sk = PickSplitK(M, N, K);
+  if (sk <= 1)
+    return stf_gemm_bf16_nt(A, B, C_bf16, nullptr, M, N, K, 0.f, 1, 0, s);
+  Tensor scratch = ctx->allocate_temp(DT_FLOAT, TensorShape({M, N}));
+  hipError_t e = hipMemsetAsync(scratch.raw_data(), 0, M * N * 4, s);
+  if (e != hipSuccess) return e;
+  e = stf_gemm_bf16_nt_splitk(A, B, scratch.raw_data(), M, N, K, sk, s);
+  if (e != hipSuccess) return e;
+  return stf_cast(0, 1, scratch.raw_data(), C_bf16, M * N, s);
+}
 inline int CastCode(DataType dt) {
   switch (dt) {
     case DT_FLOAT: return 0;
@@ -356,9 +385,8 @@ class GpuMatMulOp : public OpKernel {
                                      b.dim_size(0), b.dim_size(1), s));
     }
     if (a.dtype() == DT_BFLOAT16) {
-      OP_HIP_OK(ctx, stf_gemm_bf16_nt(a_eff.raw_data(), b_eff.raw_data(),
-                                      y->raw_data(), nullptr, m, n, k, 0.f,
-                                      1, 0, s));
+      OP_HIP_OK(ctx, GemmBf16Auto(ctx, a_eff.raw_data(), b_eff.raw_data(),
+                                  y->raw_data(), m, n, k, s));
     } else {
       OP_HIP_OK(ctx, stf_gemm_f32_nt(a_eff.raw_data(), b_eff.raw_data(),
                                      y->raw_data(), m, n, k, s));
@@ -378,6 +406,9 @@ struct GpuConvGeom {
   int64_t N, H, W, C, R, S, K, sh, sw, ph, pw, P, Q;
   int64_t M() const { return N * P * Q; }
   int64_t RSC() const { return R * S * C; }
+  // K-dim padded to the GEMM K-step so glds staging stays 16B-aligned
+  // (conv1's RSC=147 would otherwise force the scalar edge path everywhere).
+  int64_t RSCp() const { return (RSC() + 63) & ~int64_t(63); }
   bool is_1x1_s1() const {
     return R == 1 && S == 1 && sh == 1 && sw == 1 && ph == 0 && pw == 0;
   }
@@ -424,22 +455,33 @@ class GpuConv2DOp : public OpKernel {
     OP_REQUIRES_OK(ctx, GetConvGeom(x.shape(), w.shape(), strides_, padding_,
                                     &g));
     Tensor* y = ctx->allocate_output(0, TensorShape({g.N, g.P, g.Q, g.K}));
-    // weights -> [K, RSC]
-    Tensor wt = ctx->allocate_temp(DT_BFLOAT16, TensorShape({g.K, g.RSC()}));
-    OP_HIP_OK(ctx, stf_transpose2d(2, w.raw_data(), wt.raw_data(), g.RSC(),
+    int64_t rsc = g.RSC(), rscp = g.RSCp();
+    // weights -> [K, RSCp] (zero-padded K dim)
+    Tensor wsrc = w;
+    if (rscp != rsc) {
+      wsrc = ctx->allocate_temp(DT_BFLOAT16, TensorShape({rscp, g.K}));
+      OP_HIP_OK(ctx, hipMemsetAsync(wsrc.raw_data(), 0, rscp * g.K * 2, s));
+      OP_HIP_OK(ctx, hipMemcpyAsync(wsrc.raw_data(), w.raw_data(),
+                                    rsc * g.K * 2, hipMemcpyDeviceToDevice,
+                                    s));
+    }
+    Tensor wt = ctx->allocate_temp(DT_BFLOAT16, TensorShape({g.K, rscp}));
+    OP_HIP_OK(ctx, stf_transpose2d(2, wsrc.raw_data(), wt.raw_data(), rscp,
                                    g.K, s));
     const void* col_data = x.raw_data();
     Tensor col;
     if (!g.is_1x1_s1()) {
-      col = ctx->allocate_temp(DT_BFLOAT16, TensorShape({g.M(), g.RSC()}));
+      col = ctx->allocate_temp(DT_BFLOAT16, TensorShape({g.M(), rscp}));
+      if (rscp != rsc)
+        OP_HIP_OK(ctx, hipMemsetAsync(col.raw_data(), 0, g.M() * rscp * 2, s));
       OP_HIP_OK(ctx, stf_im2col_bf16(x.raw_data(), col.raw_data(), (int)g.N,
                                      (int)g.H, (int)g.W, (int)g.C, (int)g.R,
                                      (int)g.S, (int)g.sh, (int)g.sw, (int)g.ph,
-                                     (int)g.pw, (int)g.P, (int)g.Q, s));
+                                     (int)g.pw, (int)g.P, (int)g.Q, rscp, s));
       col_data = col.raw_data();
     }
     OP_HIP_OK(ctx, stf_gemm_bf16_nt(col_data, wt.raw_data(), y->raw_data(),
-                                    nullptr, g.M(), g.K, g.RSC(), 0.f, 1, 0,
+                                    nullptr, g.M(), g.K, rscp, 0.f, 1, 0,
                                     s));
   }
 
@@ -505,28 +547,27 @@ class GpuConv2DBackpropFilterOp : public OpKernel {
     OP_REQUIRES_OK(ctx, GetConvGeom(x.shape(), f_shape, strides_, padding_,
                                     &g));
     Tensor* dw = ctx->allocate_output(0, f_shape);
+    int64_t rsc = g.RSC();
     // dW[RSC, K] = col[M, RSC]^T x dy[M, K]  -> NT with A=colT, B=dyT
-    Tensor colT = ctx->allocate_temp(DT_BFLOAT16,
-                                     TensorShape({g.RSC(), g.M()}));
+    Tensor colT = ctx->allocate_temp(DT_BFLOAT16, TensorShape({rsc, g.M()}));
     if (g.is_1x1_s1()) {
       OP_HIP_OK(ctx, stf_transpose2d(2, x.raw_data(), colT.raw_data(), g.M(),
-                                     g.RSC(), s));
+                                     rsc, s));
     } else {
       Tensor col = ctx->allocate_temp(DT_BFLOAT16,
-                                      TensorShape({g.M(), g.RSC()}));
+                                      TensorShape({g.M(), rsc}));
       OP_HIP_OK(ctx, stf_im2col_bf16(x.raw_data(), col.raw_data(), (int)g.N,
                                      (int)g.H, (int)g.W, (int)g.C, (int)g.R,
                                      (int)g.S, (int)g.sh, (int)g.sw, (int)g.ph,
-                                     (int)g.pw, (int)g.P, (int)g.Q, s));
+                                     (int)g.pw, (int)g.P, (int)g.Q, rsc, s));
       OP_HIP_OK(ctx, stf_transpose2d(2, col.raw_data(), colT.raw_data(),
-                                     g.M(), g.RSC(), s));
+                                     g.M(), rsc, s));
     }
     Tensor dyT = ctx->allocate_temp(DT_BFLOAT16, TensorShape({g.K, g.M()}));
     OP_HIP_OK(ctx, stf_transpose2d(2, dy.raw_data(), dyT.raw_data(), g.M(),
                                    g.K, s));
-    OP_HIP_OK(ctx, stf_gemm_bf16_nt(colT.raw_data(), dyT.raw_data(),
-                                    dw->raw_data(), nullptr, g.RSC(), g.K,
-                                    g.M(), 0.f, 1, 0, s));
+    OP_HIP_OK(ctx, GemmBf16Auto(ctx, colT.raw_data(), dyT.raw_data(),
+                                dw->raw_data(), rsc, g.K, g.M(), s));
   }
 
  private:

@@ -13,7 +13,7 @@ struct ConvGeom {
 // one thread per 8 consecutive c (vectorized when C % 8 == 0)
 __global__ void Im2ColKernel(const __bf16* __restrict__ x,
                              __bf16* __restrict__ col, ConvGeom g,
-                             int64_t total_vec, int vec) {
+                             int64_t total_vec, int vec, int64_t ld) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   int64_t cvec = g.C / vec;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total_vec;
@@ -29,7 +29,7 @@ __global__ void Im2ColKernel(const __bf16* __restrict__ x,
     int iw = q * g.sw - g.pw + s;
     int64_t m = ((int64_t)(n * g.P + p) * g.Q + q);
     int64_t rsc = ((int64_t)(r * g.S + s) * g.C + cv * vec);
-    __bf16* dst = col + m * ((int64_t)g.R * g.S * g.C) + rsc;
+    __bf16* dst = col + m * ld + rsc;
     if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W) {
       const __bf16* src =
           x + (((int64_t)(n * g.H + ih) * g.W + iw) * g.C + cv * vec);
@@ -86,12 +86,13 @@ __global__ void Col2ImKernel(const __bf16* __restrict__ dcol,
 extern "C" hipError_t stf_im2col_bf16(const void* x, void* col, int N, int H,
                                       int W, int C, int R, int S, int sh,
                                       int sw, int ph, int pw, int P, int Q,
-                                      hipStream_t stream) {
+                                      int64_t ld, hipStream_t stream) {
   ConvGeom g{N, H, W, C, R, S, 0, sh, sw, ph, pw, P, Q};
   int vec = (C % 8 == 0) ? 8 : 1;
   int64_t total = (int64_t)N * P * Q * R * S * (C / vec);
   hipLaunchKernelGGL(Im2ColKernel, ElemwiseGrid(total, 256, 1), dim3(256), 0,
-                     stream, (const __bf16*)x, (__bf16*)col, g, total, vec);
+                     stream, (const __bf16*)x, (__bf16*)col, g, total, vec,
+                     ld);
   return hipGetLastError();
 }
 

@@ -77,13 +77,13 @@ __device__ __forceinline__ void StageSafe(const uint16_t* __restrict__ src,
 //  WAVES_M x WAVES_N waves; each wave computes (WM*16) x (WN*16) outputs.
 //  BM = WAVES_M*WM*16, BN = WAVES_N*WN*16, BK = 64.
 template <int WAVES_M, int WAVES_N, int WM, int WN, bool OUT_BF16,
-          bool FUSE_RELU>
+          bool FUSE_RELU, bool SPLITK = false>
 __launch_bounds__(256) __global__ void GemmBf16NT(
     const uint16_t* __restrict__ A,  // [M, K] row-major bf16
     const uint16_t* __restrict__ B,  // [N, K] row-major bf16
-    void* __restrict__ C,            // [M, N] f32 or bf16
+    void* __restrict__ C,            // [M, N] f32 or bf16 (f32 for SPLITK)
     const float* __restrict__ bias,  // optional [N] f32 bias (nullptr = none)
-    int64_t M, int64_t N, int64_t K, float beta) {
+    int64_t M, int64_t N, int64_t K, float beta, int splitk = 1) {
   constexpr int BM = WAVES_M * WM * 16;
   constexpr int BN = WAVES_N * WN * 16;
   constexpr int BK = 64;
@@ -97,7 +97,9 @@ __launch_bounds__(256) __global__ void GemmBf16NT(
 
   int nbm = (int)((M + BM - 1) / BM);
   int nbn = (int)((N + BN - 1) / BN);
-  int bid = XcdSwizzle(blockIdx.x, nbm * nbn);
+  int bid = XcdSwizzle(blockIdx.x, nbm * nbn * (SPLITK ? splitk : 1));
+  int slice = SPLITK ? bid % splitk : 0;
+  if (SPLITK) bid /= splitk;
   int bm = bid / nbn, bn = bid % nbn;
   int64_t m0 = (int64_t)bm * BM, n0 = (int64_t)bn * BN;
 
@@ -115,23 +117,31 @@ __launch_bounds__(256) __global__ void GemmBf16NT(
 #pragma unroll
     for (int j = 0; j < WN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  int kt_count = (int)((K + BK - 1) / BK);
-  // ---- prologue: stage tile 0 ----
+  int kt_total = (int)((K + BK - 1) / BK);
+  int kt_begin = 0, kt_count = kt_total;
+  if (SPLITK) {
+    int per = (kt_total + splitk - 1) / splitk;
+    kt_begin = slice * per;
+    kt_count = min(kt_total, kt_begin + per);
+    if (kt_begin >= kt_count) return;
+  }
+  // ---- prologue: stage first tile ----
   {
-    bool kfull = BK <= K;
+    int64_t kp = (int64_t)kt_begin * BK;
+    bool kfull = kp + BK <= K;
     if (a_interior && kfull)
-      StageFast<BM>(A, K, (int)m0, 0, a_tile(0), tid);
+      StageFast<BM>(A, K, (int)m0, kp, a_tile(0), tid);
     else
-      StageSafe<BM>(A, K, (int)m0, 0, M, K, a_tile(0), tid);
+      StageSafe<BM>(A, K, (int)m0, kp, M, K, a_tile(0), tid);
     if (b_interior && kfull)
-      StageFast<BN>(B, K, (int)n0, 0, b_tile(0), tid);
+      StageFast<BN>(B, K, (int)n0, kp, b_tile(0), tid);
     else
-      StageSafe<BN>(B, K, (int)n0, 0, N, K, b_tile(0), tid);
+      StageSafe<BN>(B, K, (int)n0, kp, N, K, b_tile(0), tid);
   }
   __syncthreads();
 
   int cur = 0;
-  for (int kt = 0; kt < kt_count; ++kt) {
+  for (int kt = kt_begin; kt < kt_count; ++kt) {
     // issue next tile's loads first (overlap with this tile's compute)
     if (kt + 1 < kt_count) {
       int64_t k0 = (int64_t)(kt + 1) * BK;
@@ -188,6 +198,10 @@ __launch_bounds__(256) __global__ void GemmBf16NT(
         if (row >= M) continue;
         float v = acc[i][j][rgi] + bv;
         if (FUSE_RELU) v = v > 0.f ? v : 0.f;
+        if (SPLITK) {
+          atomicAdd((float*)C + row * N + col, v);
+          continue;
+        }
         if (OUT_BF16) {
           uint16_t* out = (uint16_t*)C + row * N + col;
           if (beta != 0.f) v += beta * bf16_to_f32(*out);
@@ -209,7 +223,7 @@ hipError_t LaunchVariant(const uint16_t* A, const uint16_t* B, void* C,
   auto launch = [&](auto kern, int BM, int BN) {
     int64_t blocks = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
     hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0, stream, A,
-                       B, C, bias, M, N, K, beta);
+                       B, C, bias, M, N, K, beta, 1);
   };
   if (N >= 128 && M >= 128) {
     launch(GemmBf16NT<2, 2, 4, 4, OUT_BF16, FUSE_RELU>, 128, 128);
@@ -223,7 +237,38 @@ hipError_t LaunchVariant(const uint16_t* A, const uint16_t* B, void* C,
   return hipGetLastError();
 }
 
+hipError_t LaunchSplitK(const uint16_t* A, const uint16_t* B, float* C,
+                        int64_t M, int64_t N, int64_t K, int splitk,
+                        hipStream_t stream) {
+  auto launch = [&](auto kern, int BM, int BN) {
+    int64_t blocks = ((M + BM - 1) / BM) * ((N + BN - 1) / BN) * splitk;
+    hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0, stream, A,
+                       B, C, nullptr, M, N, K, 0.f, splitk);
+  };
+  if (N >= 128 && M >= 128) {
+    launch(GemmBf16NT<2, 2, 4, 4, false, false, true>, 128, 128);
+  } else if (N >= 128) {
+    launch(GemmBf16NT<2, 2, 2, 4, false, false, true>, 64, 128);
+  } else if (M >= 128) {
+    launch(GemmBf16NT<2, 2, 4, 2, false, false, true>, 128, 64);
+  } else {
+    launch(GemmBf16NT<2, 2, 2, 2, false, false, true>, 64, 64);
+  }
+  return hipGetLastError();
+}
+
 }  // namespace
+
+// Split-K variant for skinny-output huge-K GEMMs (dW): partial products
+// atomically accumulated into a zeroed f32 C (guide split-K decomposition:
+// size the grid to ~1-2x the 256 CUs).
+extern "C" hipError_t stf_gemm_bf16_nt_splitk(const void* A, const void* B,
+                                              void* C_f32, int64_t M,
+                                              int64_t N, int64_t K, int splitk,
+                                              hipStream_t stream) {
+  return LaunchSplitK((const uint16_t*)A, (const uint16_t*)B, (float*)C_f32,
+                      M, N, K, splitk, stream);
+}
 
 // C[M,N] = A[M,K] * B[N,K]^T (+ beta*C) (+bias) (+relu)
 extern "C" hipError_t stf_gemm_bf16_nt(const void* A, const void* B, void* C,

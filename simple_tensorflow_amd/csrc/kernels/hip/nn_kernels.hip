@@ -182,6 +182,8 @@ __global__ void XentKernel(const T* __restrict__ logits,
 
 // ---------------- batch norm ----------------
 // pass 1: per-channel sum and sumsq into f32 accumulators [2C] (zeroed).
+// All 256 threads stream elements (vectorized 8-wide bf16 when C%8==0),
+// accumulating via LDS atomics — wave64-coalesced regardless of C.
 template <typename T>
 __global__ void BnStatsKernel(const T* __restrict__ x, float* __restrict__ acc,
                               int64_t rows, int c) {
@@ -193,15 +195,29 @@ __global__ void BnStatsKernel(const T* __restrict__ x, float* __restrict__ acc,
     s2[i] = 0.f;
   }
   __syncthreads();
-  int64_t rows_per_block = (rows + gridDim.x - 1) / gridDim.x;
-  int64_t row0 = blockIdx.x * rows_per_block;
-  int64_t row1 = min(row0 + rows_per_block, rows);
-  for (int64_t r = row0; r < row1; ++r) {
-    const T* src = x + r * c;
-    for (int i = threadIdx.x; i < c; i += blockDim.x) {
-      float v = (float)src[i];
-      s1[i] += v;
-      s2[i] += v * v;
+  int64_t n = rows * c;
+  int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+  if (c % 8 == 0) {
+    int64_t nvec = n / 8;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+         i += gstride) {
+      T v[8];
+      *(ulong2*)v = *(const ulong2*)(x + i * 8);
+      int cb = (int)((i * 8) % c);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float f = (float)v[e];
+        atomicAdd(&s1[cb + e], f);
+        atomicAdd(&s2[cb + e], f * f);
+      }
+    }
+  } else {
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += gstride) {
+      float f = (float)x[i];
+      int ch = (int)(i % c);
+      atomicAdd(&s1[ch], f);
+      atomicAdd(&s2[ch], f * f);
     }
   }
   __syncthreads();
@@ -260,17 +276,32 @@ __global__ void BnGradStatsKernel(const T* __restrict__ dy,
     s2[i] = 0.f;
   }
   __syncthreads();
-  int64_t rows_per_block = (rows + gridDim.x - 1) / gridDim.x;
-  int64_t row0 = blockIdx.x * rows_per_block;
-  int64_t row1 = min(row0 + rows_per_block, rows);
-  for (int64_t r = row0; r < row1; ++r) {
-    const T* dsrc = dy + r * c;
-    const T* xsrc = x + r * c;
-    for (int i = threadIdx.x; i < c; i += blockDim.x) {
-      float g = (float)dsrc[i];
-      float xhat = ((float)xsrc[i] - mean[i]) * inv_std[i];
-      s1[i] += g;
-      s2[i] += g * xhat;
+  int64_t n = rows * c;
+  int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+  if (c % 8 == 0) {
+    int64_t nvec = n / 8;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+         i += gstride) {
+      T g[8], xv[8];
+      *(ulong2*)g = *(const ulong2*)(dy + i * 8);
+      *(ulong2*)xv = *(const ulong2*)(x + i * 8);
+      int cb = (int)((i * 8) % c);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float gf = (float)g[e];
+        float xhat = ((float)xv[e] - mean[cb + e]) * inv_std[cb + e];
+        atomicAdd(&s1[cb + e], gf);
+        atomicAdd(&s2[cb + e], gf * xhat);
+      }
+    }
+  } else {
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += gstride) {
+      int ch = (int)(i % c);
+      float gf = (float)dy[i];
+      float xhat = ((float)x[i] - mean[ch]) * inv_std[ch];
+      atomicAdd(&s1[ch], gf);
+      atomicAdd(&s2[ch], gf * xhat);
     }
   }
   __syncthreads();

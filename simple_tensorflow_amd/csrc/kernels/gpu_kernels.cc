@@ -455,7 +455,9 @@ class GpuConv2DOp : public OpKernel {
     OP_REQUIRES_OK(ctx, GetConvGeom(x.shape(), w.shape(), strides_, padding_,
                                     &g));
     Tensor* y = ctx->allocate_output(0, TensorShape({g.N, g.P, g.Q, g.K}));
-    int64_t rsc = g.RSC(), rscp = g.RSCp();
+    int64_t rsc = g.RSC();
+    // 1x1/s1 uses x directly as the GEMM A (ld = C) — no padding there.
+    int64_t rscp = g.is_1x1_s1() ? rsc : g.RSCp();
     // weights -> [K, RSCp] (zero-padded K dim)
     Tensor wsrc = w;
     if (rscp != rsc) {

@@ -198,17 +198,43 @@ __global__ void BnStatsKernel(const T* __restrict__ x, float* __restrict__ acc,
   int64_t n = rows * c;
   int64_t gstride = (int64_t)gridDim.x * blockDim.x;
   if (c % 8 == 0) {
+    // Each thread's 8-channel window is FIXED across iterations whenever
+    // (gstride*8) % c == 0 (power-of-two C with the 512x256 grid), so
+    // accumulate privately and flush to LDS once; flush on window change
+    // keeps the general case correct.
     int64_t nvec = n / 8;
+    float p1[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    float p2[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    int cb0 = -1;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
          i += gstride) {
       T v[8];
       *(ulong2*)v = *(const ulong2*)(x + i * 8);
       int cb = (int)((i * 8) % c);
+      if (cb != cb0) {
+        if (cb0 >= 0) {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            atomicAdd(&s1[cb0 + e], p1[e]);
+            atomicAdd(&s2[cb0 + e], p2[e]);
+            p1[e] = 0.f;
+            p2[e] = 0.f;
+          }
+        }
+        cb0 = cb;
+      }
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         float f = (float)v[e];
-        atomicAdd(&s1[cb + e], f);
-        atomicAdd(&s2[cb + e], f * f);
+        p1[e] += f;
+        p2[e] += f * f;
+      }
+    }
+    if (cb0 >= 0) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        atomicAdd(&s1[cb0 + e], p1[e]);
+        atomicAdd(&s2[cb0 + e], p2[e]);
       }
     }
   } else {
@@ -280,18 +306,46 @@ __global__ void BnGradStatsKernel(const T* __restrict__ dy,
   int64_t gstride = (int64_t)gridDim.x * blockDim.x;
   if (c % 8 == 0) {
     int64_t nvec = n / 8;
+    float p1[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    float p2[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    float mloc[8], iloc[8];
+    int cb0 = -1;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
          i += gstride) {
       T g[8], xv[8];
       *(ulong2*)g = *(const ulong2*)(dy + i * 8);
       *(ulong2*)xv = *(const ulong2*)(x + i * 8);
       int cb = (int)((i * 8) % c);
+      if (cb != cb0) {
+        if (cb0 >= 0) {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            atomicAdd(&s1[cb0 + e], p1[e]);
+            atomicAdd(&s2[cb0 + e], p2[e]);
+            p1[e] = 0.f;
+            p2[e] = 0.f;
+          }
+        }
+        cb0 = cb;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          mloc[e] = mean[cb + e];
+          iloc[e] = inv_std[cb + e];
+        }
+      }
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         float gf = (float)g[e];
-        float xhat = ((float)xv[e] - mean[cb + e]) * inv_std[cb + e];
-        atomicAdd(&s1[cb + e], gf);
-        atomicAdd(&s2[cb + e], gf * xhat);
+        float xhat = ((float)xv[e] - mloc[e]) * iloc[e];
+        p1[e] += gf;
+        p2[e] += gf * xhat;
+      }
+    }
+    if (cb0 >= 0) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        atomicAdd(&s1[cb0 + e], p1[e]);
+        atomicAdd(&s2[cb0 + e], p2[e]);
       }
     }
   } else {

@@ -8,7 +8,7 @@ import threading
 import numpy as np
 
 from simple_tensorflow_amd import _core
-from simple_tensorflow_amd.python.framework import dtypes, ops
+from simple_tensorflow_amd.python.framework import dtypes, errors, ops
 
 _default_session_stack = threading.local()
 
@@ -72,7 +72,10 @@ class Session(object):
                 t = _as_fetchable(k, self._graph)
                 feeds[t.name] = _convert_feed(t, v)
 
-        results = self._core.run(feeds, fetch_names, targets)
+        try:
+            results = self._core.run(feeds, fetch_names, targets)
+        except RuntimeError as e:
+            errors.raise_from_message(str(e))
         out = []
         for f, slot in zip(flat, fetch_slots):
             if slot is None:

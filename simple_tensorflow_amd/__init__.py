@@ -28,6 +28,12 @@ from simple_tensorflow_amd.python.ops import (  # noqa: F401
 from simple_tensorflow_amd.python.training import optimizer as _optimizer
 from simple_tensorflow_amd.python.training import training_util as _training_util
 from simple_tensorflow_amd.python.training import saver as _saver
+from simple_tensorflow_amd.python.training import coordinator as _coord
+from simple_tensorflow_amd.python.training import input as _input
+from simple_tensorflow_amd.python.training import monitored_session as _ms
+from simple_tensorflow_amd.python.summary import summary as _summary_mod
+from simple_tensorflow_amd.python.summary import writer as _summary_writer
+from simple_tensorflow_amd.python.framework import errors  # noqa: F401
 
 # ---- dtypes ----
 float32 = _dtypes.float32
@@ -247,8 +253,52 @@ class _TrainModule(object):
     get_checkpoint_state = staticmethod(_saver.get_checkpoint_state)
     update_checkpoint_state = staticmethod(_saver.update_checkpoint_state)
     checkpoint_exists = staticmethod(_saver.checkpoint_exists)
+    Coordinator = _coord.Coordinator
+    QueueRunner = _coord.QueueRunner
+    add_queue_runner = staticmethod(_coord.add_queue_runner)
+    start_queue_runners = staticmethod(_coord.start_queue_runners)
+    batch = staticmethod(_input.batch)
+    shuffle_batch = staticmethod(_input.shuffle_batch)
+    input_producer = staticmethod(_input.input_producer)
+    string_input_producer = staticmethod(_input.string_input_producer)
+    range_input_producer = staticmethod(_input.range_input_producer)
+    MonitoredSession = _ms.MonitoredSession
+    MonitoredTrainingSession = staticmethod(_ms.MonitoredTrainingSession)
+    Scaffold = _ms.Scaffold
+    SessionManager = _ms.SessionManager
+    ChiefSessionCreator = _ms.ChiefSessionCreator
+    WorkerSessionCreator = _ms.WorkerSessionCreator
+    SessionRunHook = _ms.SessionRunHook
+    SessionRunArgs = _ms.SessionRunArgs
+    SessionRunContext = _ms.SessionRunContext
+    SessionRunValues = _ms.SessionRunValues
+    StopAtStepHook = _ms.StopAtStepHook
+    CheckpointSaverHook = _ms.CheckpointSaverHook
+    LoggingTensorHook = _ms.LoggingTensorHook
+    StepCounterHook = _ms.StepCounterHook
+    NanTensorHook = _ms.NanTensorHook
+    SummarySaverHook = _ms.SummarySaverHook
 
 
 train = _TrainModule()
+
+
+class _SummaryModule(object):
+    scalar = staticmethod(_summary_mod.scalar)
+    histogram = staticmethod(_summary_mod.histogram)
+    merge = staticmethod(_summary_mod.merge)
+    merge_all = staticmethod(_summary_mod.merge_all)
+    FileWriter = _summary_writer.FileWriter
+
+
+summary = _SummaryModule()
+
+
+class _PythonIoModule(object):
+    from simple_tensorflow_amd.python.lib.io.tf_record import (
+        TFRecordWriter, tf_record_iterator)
+
+
+python_io = _PythonIoModule()
 
 __version__ = '0.1.0'

@@ -43,6 +43,20 @@ class Device {
   virtual void* compute_stream() { return nullptr; }
   virtual int gpu_ordinal() const { return -1; }
 
+  // hipGraph capture of the steady-state step (north-star: hipGraph-captured
+  // step loops). Begin/End bracket one executor run on the compute stream;
+  // Launch replays the instantiated graph.
+  virtual Status BeginGraphCapture() {
+    return errors::Unimplemented("not a GPU device");
+  }
+  virtual Status EndGraphCapture(void** graph_exec) {
+    return errors::Unimplemented("not a GPU device");
+  }
+  virtual Status LaunchCapturedGraph(void* graph_exec) {
+    return errors::Unimplemented("not a GPU device");
+  }
+  virtual bool capturing() const { return false; }
+
   // Tensor movement. `done` is invoked when the copy is complete.
   virtual Status CopyDeviceTensorToHost(const Tensor& src, Tensor* dst);
   virtual Status CopyHostTensorToDevice(const Tensor& src, Tensor* dst);

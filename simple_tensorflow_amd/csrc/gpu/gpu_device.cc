@@ -210,7 +210,38 @@ class GpuDevice : public Device {
     return Status::OK();
   }
 
+  Status BeginGraphCapture() override {
+    hipSetDevice(ordinal_);
+    HIP_CHECK_STATUS(hipStreamBeginCapture(compute_,
+                                           hipStreamCaptureModeRelaxed));
+    capturing_ = true;
+    return Status::OK();
+  }
+  Status EndGraphCapture(void** graph_exec) override {
+    hipSetDevice(ordinal_);
+    capturing_ = false;
+    hipGraph_t graph = nullptr;
+    HIP_CHECK_STATUS(hipStreamEndCapture(compute_, &graph));
+    hipGraphExec_t exec = nullptr;
+    hipError_t e = hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0);
+    hipGraphDestroy(graph);
+    if (e != hipSuccess)
+      return errors::Internal("hipGraphInstantiate: ", hipGetErrorString(e));
+    *graph_exec = (void*)exec;
+    return Status::OK();
+  }
+  Status LaunchCapturedGraph(void* graph_exec) override {
+    hipSetDevice(ordinal_);
+    HIP_CHECK_STATUS(hipGraphLaunch((hipGraphExec_t)graph_exec, compute_));
+    return Status::OK();
+  }
+  bool capturing() const override { return capturing_; }
+
   Status CopyDeviceTensorToHost(const Tensor& src, Tensor* dst) override {
+    if (capturing_)
+      return errors::FailedPrecondition(
+          "d2h copy during hipGraph capture (host-dependent op in the "
+          "captured step)");
     hipSetDevice(ordinal_);
     Tensor host(host_allocator(), src.dtype(), src.shape());
     // Fence: wait for pending compute that may produce src.
@@ -249,6 +280,7 @@ class GpuDevice : public Device {
   int ordinal_;
   GpuBfcAllocator bfc_;
   hipStream_t compute_, h2d_, d2h_;
+  bool capturing_ = false;
 };
 
 void AddGpuDevices(DeviceMgr* mgr) {

@@ -5,6 +5,7 @@
 #include <hip/hip_runtime.h>
 
 #include <atomic>
+#include <mutex>
 
 #include "kernels/kernel_util.h"
 
@@ -76,9 +77,9 @@ hipError_t stf_apply_momentum(int, void*, void*, const void*, const void*,
 hipError_t stf_apply_adam(int, void*, void*, void*, const void*, const void*,
                           const void*, const void*, const void*, const void*,
                           const void*, int64_t, hipStream_t);
-hipError_t stf_random_uniform(uint64_t, uint64_t, void*, int64_t, int,
+hipError_t stf_random_uniform(uint64_t, void*, void*, int64_t, int,
                               hipStream_t);
-hipError_t stf_random_normal(uint64_t, uint64_t, void*, int64_t, int, int,
+hipError_t stf_random_normal(uint64_t, void*, void*, int64_t, int, int,
                              hipStream_t);
 }
 
@@ -1164,21 +1165,31 @@ class GpuRandomOp : public OpKernel {
     auto dims = IntVector(ctx->input(0));
     Tensor* y = ctx->allocate_output(0, TensorShape(dims));
     int64_t n = y->NumElements();
-    uint64_t off = offset_.fetch_add((n + 3) / 4 + 1);
     int bf16 = y->dtype() == DT_BFLOAT16;
     hipStream_t s = GPU_STREAM(ctx);
+    {
+      std::lock_guard<std::mutex> l(mu_);
+      if (!ctr_.IsInitialized()) {
+        // device-resident philox offset (advanced in-graph so hipGraph
+        // replays still draw fresh numbers)
+        ctr_ = Tensor(ctx->device()->allocator(), DT_INT64, TensorShape({1}));
+        OP_HIP_OK(ctx, hipMemsetAsync(ctr_.raw_data(), 0, 8, s));
+      }
+    }
     if (kind_ == 0) {
-      OP_HIP_OK(ctx, stf_random_uniform(seed_, off, y->raw_data(), n, bf16, s));
+      OP_HIP_OK(ctx, stf_random_uniform(seed_, ctr_.raw_data(), y->raw_data(),
+                                        n, bf16, s));
     } else {
-      OP_HIP_OK(ctx, stf_random_normal(seed_, off, y->raw_data(), n, bf16,
-                                       kind_ == 2, s));
+      OP_HIP_OK(ctx, stf_random_normal(seed_, ctr_.raw_data(), y->raw_data(),
+                                       n, bf16, kind_ == 2, s));
     }
   }
 
  private:
   int kind_;
   uint64_t seed_;
-  std::atomic<uint64_t> offset_{0};
+  std::mutex mu_;
+  Tensor ctr_;
 };
 class RandomUniformGpu : public GpuRandomOp {
  public:

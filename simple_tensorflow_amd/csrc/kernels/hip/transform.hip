@@ -235,9 +235,13 @@ __global__ void ApplyAdamKernel(float* __restrict__ var, float* __restrict__ m,
 }
 
 // ---- Philox RNG fills ----
-__global__ void RandomUniformKernel(uint64_t seed, uint64_t offset,
+// The philox offset lives in DEVICE memory so a hipGraph replay of the step
+// still draws fresh numbers (the advance kernel below bumps it in-graph).
+__global__ void RandomUniformKernel(uint64_t seed,
+                                    const unsigned long long* __restrict__ ctr,
                                     float* __restrict__ out, int64_t n,
                                     int as_bf16) {
+  uint64_t offset = ctr[0];
   int64_t tid = blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = tid; i * 4 < n; i += stride) {
@@ -256,9 +260,11 @@ __global__ void RandomUniformKernel(uint64_t seed, uint64_t offset,
   }
 }
 
-__global__ void RandomNormalKernel(uint64_t seed, uint64_t offset,
+__global__ void RandomNormalKernel(uint64_t seed,
+                                   const unsigned long long* __restrict__ ctr,
                                    float* __restrict__ out, int64_t n,
                                    int as_bf16, int truncated) {
+  uint64_t offset = ctr[0];
   int64_t tid = blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = tid; i * 4 < n; i += stride) {
@@ -290,6 +296,11 @@ __global__ void RandomNormalKernel(uint64_t seed, uint64_t offset,
       }
     }
   }
+}
+
+__global__ void AdvanceCtrKernel(unsigned long long* ctr,
+                                 unsigned long long delta) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) ctr[0] += delta;
 }
 
 }  // namespace
@@ -437,20 +448,28 @@ hipError_t stf_apply_adam(int grad_bf16, void* var, void* m, void* v,
   return hipGetLastError();
 }
 
-hipError_t stf_random_uniform(uint64_t seed, uint64_t offset, void* out,
+hipError_t stf_random_uniform(uint64_t seed, void* ctr_dev, void* out,
                               int64_t n, int as_bf16, hipStream_t stream) {
   dim3 grid = ElemwiseGrid(n, 256, 4);
   hipLaunchKernelGGL(RandomUniformKernel, grid, dim3(256), 0, stream, seed,
-                     offset, (float*)out, n, as_bf16);
+                     (const unsigned long long*)ctr_dev, (float*)out, n,
+                     as_bf16);
+  hipLaunchKernelGGL(AdvanceCtrKernel, dim3(1), dim3(64), 0, stream,
+                     (unsigned long long*)ctr_dev,
+                     (unsigned long long)((n + 3) / 4 + 1));
   return hipGetLastError();
 }
 
-hipError_t stf_random_normal(uint64_t seed, uint64_t offset, void* out,
+hipError_t stf_random_normal(uint64_t seed, void* ctr_dev, void* out,
                              int64_t n, int as_bf16, int truncated,
                              hipStream_t stream) {
   dim3 grid = ElemwiseGrid(n, 256, 4);
   hipLaunchKernelGGL(RandomNormalKernel, grid, dim3(256), 0, stream, seed,
-                     offset, (float*)out, n, as_bf16, truncated);
+                     (const unsigned long long*)ctr_dev, (float*)out, n,
+                     as_bf16, truncated);
+  hipLaunchKernelGGL(AdvanceCtrKernel, dim3(1), dim3(64), 0, stream,
+                     (unsigned long long*)ctr_dev,
+                     (unsigned long long)((n + 3) / 4 + 16));
   return hipGetLastError();
 }
 

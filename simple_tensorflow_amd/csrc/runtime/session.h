@@ -53,6 +53,14 @@ class DirectSession {
     // feed key -> (recv_device name); fetch key -> send_device name
     std::map<std::string, std::string> feed_devices;
     std::map<std::string, std::string> fetch_devices;
+    // hipGraph capture of the steady-state step (no feeds/fetches, one GPU
+    // partition, host side trivial): after `kCaptureAfter` plain runs the
+    // next run is recorded and subsequent runs replay the instantiated graph.
+    bool capture_eligible = false;
+    bool capture_broken = false;
+    int plain_runs = 0;
+    Device* capture_device = nullptr;
+    void* graph_exec = nullptr;
   };
 
   Status GetOrCreateExecutors(

@@ -79,6 +79,12 @@ hipError_t stf_apply_adam(int, void*, void*, void*, const void*, const void*,
                           const void*, int64_t, hipStream_t);
 hipError_t stf_random_uniform(uint64_t, void*, void*, int64_t, int,
                               hipStream_t);
+hipError_t stf_strided_copy(int, int, const void*, void*, int64_t, int64_t,
+                            int64_t, int64_t, int64_t, hipStream_t);
+hipError_t stf_gather_rows(int, int, const void*, const void*, void*, int64_t,
+                           int64_t, int64_t, hipStream_t);
+hipError_t stf_segment_sum(int, int, const void*, const void*, float*,
+                           int64_t, int64_t, int64_t, hipStream_t);
 hipError_t stf_random_normal(uint64_t, void*, void*, int64_t, int, int,
                              hipStream_t);
 }
@@ -1206,6 +1212,214 @@ class TruncatedNormalGpu : public GpuRandomOp {
 REGISTER_KERNEL_BUILDER(Name("RandomUniform").Device(DEVICE_GPU).HostMemory("shape"), RandomUniformGpu);
 REGISTER_KERNEL_BUILDER(Name("RandomStandardNormal").Device(DEVICE_GPU).HostMemory("shape"), RandomNormalGpu);
 REGISTER_KERNEL_BUILDER(Name("TruncatedNormal").Device(DEVICE_GPU).HostMemory("shape"), TruncatedNormalGpu);
+
+// ---------------------------------------------------------------------------
+// concat / split / pack / unpack / gather / segment sum / slice
+// ---------------------------------------------------------------------------
+class GpuConcatV2Op : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    int n = num_inputs() - 1;
+    int64_t axis = IntVector(ctx->input(n))[0];
+    const Tensor& first = ctx->input(0);
+    if (axis < 0) axis += first.dims();
+    TensorShape out_shape = first.shape();
+    int64_t total = 0;
+    for (int k = 0; k < n; ++k) total += ctx->input(k).dim_size((int)axis);
+    out_shape.set_dim((int)axis, total);
+    Tensor* y = ctx->allocate_output(0, out_shape);
+    int64_t outer = 1, inner = 1;
+    for (int i = 0; i < axis; ++i) outer *= first.dim_size(i);
+    for (int i = (int)axis + 1; i < first.dims(); ++i)
+      inner *= first.dim_size(i);
+    hipStream_t s = GPU_STREAM(ctx);
+    int es = (int)DataTypeSize(first.dtype());
+    int64_t off = 0;
+    for (int k = 0; k < n; ++k) {
+      const Tensor& t = ctx->input(k);
+      int64_t rows = t.dim_size((int)axis);
+      OP_HIP_OK(ctx, stf_strided_copy(es, 1, t.raw_data(), y->raw_data(),
+                                      outer, rows, inner, total, off, s));
+      off += rows;
+    }
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("ConcatV2").Device(DEVICE_GPU).HostMemory("axis"), GpuConcatV2Op);
+
+class GpuSplitOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    int64_t axis = IntVector(ctx->input(0))[0];
+    const Tensor& in = ctx->input(1);
+    if (axis < 0) axis += in.dims();
+    int n = num_outputs();
+    OP_REQUIRES(ctx, in.dim_size((int)axis) % n == 0,
+                errors::InvalidArgument("Split axis not divisible"));
+    int64_t part = in.dim_size((int)axis) / n;
+    TensorShape out_shape = in.shape();
+    out_shape.set_dim((int)axis, part);
+    int64_t outer = 1, inner = 1;
+    for (int i = 0; i < axis; ++i) outer *= in.dim_size(i);
+    for (int i = (int)axis + 1; i < in.dims(); ++i) inner *= in.dim_size(i);
+    hipStream_t s = GPU_STREAM(ctx);
+    int es = (int)DataTypeSize(in.dtype());
+    for (int k = 0; k < n; ++k) {
+      Tensor* y = ctx->allocate_output(k, out_shape);
+      OP_HIP_OK(ctx, stf_strided_copy(es, 0, in.raw_data(), y->raw_data(),
+                                      outer, part, inner,
+                                      in.dim_size((int)axis), k * part, s));
+    }
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("Split").Device(DEVICE_GPU).HostMemory("split_dim"), GpuSplitOp);
+
+class GpuPackOp : public OpKernel {
+ public:
+  explicit GpuPackOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("axis", &axis_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    int n = num_inputs();
+    const Tensor& first = ctx->input(0);
+    int axis = axis_ < 0 ? (int)(axis_ + first.dims() + 1) : (int)axis_;
+    TensorShape out_shape = first.shape();
+    out_shape.InsertDim(axis, n);
+    Tensor* y = ctx->allocate_output(0, out_shape);
+    int64_t outer = 1, inner = 1;
+    for (int i = 0; i < axis; ++i) outer *= first.dim_size(i);
+    for (int i = axis; i < first.dims(); ++i) inner *= first.dim_size(i);
+    hipStream_t s = GPU_STREAM(ctx);
+    int es = (int)DataTypeSize(first.dtype());
+    for (int k = 0; k < n; ++k) {
+      OP_HIP_OK(ctx, stf_strided_copy(es, 1, ctx->input(k).raw_data(),
+                                      y->raw_data(), outer, 1, inner, n, k,
+                                      s));
+    }
+  }
+
+ private:
+  int64_t axis_ = 0;
+};
+REGISTER_KERNEL_BUILDER(Name("Pack").Device(DEVICE_GPU), GpuPackOp);
+
+class GpuUnpackOp : public OpKernel {
+ public:
+  explicit GpuUnpackOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("axis", &axis_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& in = ctx->input(0);
+    int axis = axis_ < 0 ? (int)(axis_ + in.dims()) : (int)axis_;
+    int n = num_outputs();
+    TensorShape out_shape = in.shape();
+    out_shape.RemoveDim(axis);
+    int64_t outer = 1, inner = 1;
+    for (int i = 0; i < axis; ++i) outer *= in.dim_size(i);
+    for (int i = axis + 1; i < in.dims(); ++i) inner *= in.dim_size(i);
+    hipStream_t s = GPU_STREAM(ctx);
+    int es = (int)DataTypeSize(in.dtype());
+    for (int k = 0; k < n; ++k) {
+      Tensor* y = ctx->allocate_output(k, out_shape);
+      OP_HIP_OK(ctx, stf_strided_copy(es, 0, in.raw_data(), y->raw_data(),
+                                      outer, 1, inner, n, k, s));
+    }
+  }
+
+ private:
+  int64_t axis_ = 0;
+};
+REGISTER_KERNEL_BUILDER(Name("Unpack").Device(DEVICE_GPU), GpuUnpackOp);
+
+class GpuGatherOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& params = ctx->input(0);
+    const Tensor& idx = ctx->input(1);
+    TensorShape out_shape = idx.shape();
+    int64_t row = 1;
+    for (int i = 1; i < params.dims(); ++i) {
+      out_shape.AddDim(params.dim_size(i));
+      row *= params.dim_size(i);
+    }
+    Tensor* y = ctx->allocate_output(0, out_shape);
+    OP_HIP_OK(ctx, stf_gather_rows(DtypeCode(params.dtype()),
+                                   idx.dtype() == DT_INT32, params.raw_data(),
+                                   idx.raw_data(), y->raw_data(),
+                                   idx.NumElements(), row, params.dim_size(0),
+                                   GPU_STREAM(ctx)));
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("Gather").Device(DEVICE_GPU).TypeConstraint<float>("Tparams"), GpuGatherOp);
+REGISTER_KERNEL_BUILDER(Name("Gather").Device(DEVICE_GPU).TypeConstraint<bfloat16>("Tparams"), GpuGatherOp);
+
+class GpuUnsortedSegmentSumOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& data = ctx->input(0);
+    const Tensor& ids = ctx->input(1);
+    int64_t nseg = IntVector(ctx->input(2))[0];
+    TensorShape out_shape({nseg});
+    int64_t row = 1;
+    for (int i = ids.dims(); i < data.dims(); ++i) {
+      out_shape.AddDim(data.dim_size(i));
+      row *= data.dim_size(i);
+    }
+    Tensor* y = ctx->allocate_output(0, out_shape);
+    hipStream_t s = GPU_STREAM(ctx);
+    Tensor scratch = ctx->allocate_temp(DT_FLOAT, TensorShape({nseg * row}));
+    OP_HIP_OK(ctx, ZeroF32(scratch.raw_data(), nseg * row, s));
+    OP_HIP_OK(ctx, stf_segment_sum(DtypeCode(data.dtype()),
+                                   ids.dtype() == DT_INT32, data.raw_data(),
+                                   ids.raw_data(), scratch.flat<float>(),
+                                   ids.NumElements(), row, nseg, s));
+    OP_HIP_OK(ctx, stf_cast(0, CastCode(y->dtype()), scratch.raw_data(),
+                            y->raw_data(), nseg * row, s));
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("UnsortedSegmentSum").Device(DEVICE_GPU).TypeConstraint<float>("T").HostMemory("num_segments"), GpuUnsortedSegmentSumOp);
+REGISTER_KERNEL_BUILDER(Name("UnsortedSegmentSum").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T").HostMemory("num_segments"), GpuUnsortedSegmentSumOp);
+
+// Slice on GPU (single sliced dim; degenerate = plain copy)
+class GpuSliceOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& in = ctx->input(0);
+    auto begin = IntVector(ctx->input(1));
+    auto size = IntVector(ctx->input(2));
+    int rank = in.dims();
+    for (int i = 0; i < rank; ++i)
+      if (size[i] == -1) size[i] = in.dim_size(i) - begin[i];
+    Tensor* y = ctx->allocate_output(0, TensorShape(size));
+    int cut = -1;
+    for (int i = 0; i < rank; ++i) {
+      if (begin[i] != 0 || size[i] != in.dim_size(i)) {
+        OP_REQUIRES(ctx, cut == -1,
+                    errors::Unimplemented("GPU Slice: one sliced dim"));
+        cut = i;
+      }
+    }
+    hipStream_t s = GPU_STREAM(ctx);
+    if (cut == -1) {
+      OP_HIP_OK(ctx, hipMemcpyAsync(y->raw_data(), in.raw_data(),
+                                    in.TotalBytes(), hipMemcpyDeviceToDevice,
+                                    s));
+      return;
+    }
+    int64_t outer = 1, inner = 1;
+    for (int i = 0; i < cut; ++i) outer *= in.dim_size(i);
+    for (int i = cut + 1; i < rank; ++i) inner *= in.dim_size(i);
+    OP_HIP_OK(ctx, stf_strided_copy((int)DataTypeSize(in.dtype()), 0,
+                                    in.raw_data(), y->raw_data(), outer,
+                                    size[cut], inner, in.dim_size(cut),
+                                    begin[cut], s));
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("Slice").Device(DEVICE_GPU).HostMemory("begin").HostMemory("size"), GpuSliceOp);
 
 }  // namespace
 }  // namespace stf

@@ -298,6 +298,61 @@ __global__ void RandomNormalKernel(uint64_t seed,
   }
 }
 
+// strided block copy for concat/split: moves [outer, rows, inner] between a
+// contiguous side and a strided side (dst_stride rows, dst_off row offset).
+template <typename T, bool SCATTER>  // SCATTER: contiguous src -> strided dst
+__global__ void StridedCopyKernel(const T* __restrict__ src,
+                                  T* __restrict__ dst, int64_t outer,
+                                  int64_t rows, int64_t inner,
+                                  int64_t big_stride, int64_t row_off,
+                                  int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t idx = blockIdx.x * blockDim.x + threadIdx.x; idx < n;
+       idx += stride) {
+    int64_t i = idx % inner;
+    int64_t rem = idx / inner;
+    int64_t a = rem % rows;
+    int64_t o = rem / rows;
+    int64_t small = (o * rows + a) * inner + i;
+    int64_t big = (o * big_stride + row_off + a) * inner + i;
+    if (SCATTER) dst[big] = src[small];
+    else dst[small] = src[big];
+  }
+}
+
+// embedding gather: out[i, :] = params[indices[i], :]
+template <typename T, typename I>
+__global__ void GatherRowsKernel(const T* __restrict__ params,
+                                 const I* __restrict__ indices,
+                                 T* __restrict__ out, int64_t nidx,
+                                 int64_t row, int64_t nrows) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t n = nidx * row;
+  for (int64_t idx = blockIdx.x * blockDim.x + threadIdx.x; idx < n;
+       idx += stride) {
+    int64_t i = idx / row, r = idx % row;
+    int64_t src = (int64_t)indices[i];
+    out[idx] = (src >= 0 && src < nrows) ? params[src * row + r] : (T)0.f;
+  }
+}
+
+// embedding grad scatter-add into f32 accumulator (zeroed)
+template <typename T, typename I>
+__global__ void SegmentSumKernel(const T* __restrict__ data,
+                                 const I* __restrict__ ids,
+                                 float* __restrict__ out, int64_t nidx,
+                                 int64_t row, int64_t nseg) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t n = nidx * row;
+  for (int64_t idx = blockIdx.x * blockDim.x + threadIdx.x; idx < n;
+       idx += stride) {
+    int64_t i = idx / row, r = idx % row;
+    int64_t seg = (int64_t)ids[i];
+    if (seg >= 0 && seg < nseg)
+      atomicAdd(&out[seg * row + r], (float)data[idx]);
+  }
+}
+
 __global__ void AdvanceCtrKernel(unsigned long long* ctr,
                                  unsigned long long delta) {
   if (threadIdx.x == 0 && blockIdx.x == 0) ctr[0] += delta;
@@ -445,6 +500,48 @@ hipError_t stf_apply_adam(int grad_bf16, void* var, void* m, void* v,
                        (const float*)b2p, (const float*)lr, (const float*)b1,
                        (const float*)b2, (const float*)eps,
                        (const float*)grad, n);
+  return hipGetLastError();
+}
+
+hipError_t stf_strided_copy(int elem_size, int scatter, const void* src,
+                            void* dst, int64_t outer, int64_t rows,
+                            int64_t inner, int64_t big_stride, int64_t row_off,
+                            hipStream_t stream) {
+  int64_t n = outer * rows * inner;
+  dim3 grid = ElemwiseGrid(n, 256, 4);
+#define SC(T)                                                                  do {                                                                           if (scatter)                                                                   hipLaunchKernelGGL((StridedCopyKernel<T, true>), grid, dim3(256), 0,                            stream, (const T*)src, (T*)dst, outer, rows, inner,                          big_stride, row_off, n);                                else                                                                           hipLaunchKernelGGL((StridedCopyKernel<T, false>), grid, dim3(256), 0,                           stream, (const T*)src, (T*)dst, outer, rows, inner,                          big_stride, row_off, n);                              } while (0)
+  if (elem_size == 2) SC(uint16_t);
+  else if (elem_size == 4) SC(uint32_t);
+  else SC(uint64_t);
+#undef SC
+  return hipGetLastError();
+}
+
+hipError_t stf_gather_rows(int dtype, int idx_i32, const void* params,
+                           const void* indices, void* out, int64_t nidx,
+                           int64_t row, int64_t nrows, hipStream_t stream) {
+  dim3 grid = ElemwiseGrid(nidx * row, 256, 4);
+#define GR(T, I)                                                               hipLaunchKernelGGL((GatherRowsKernel<T, I>), grid, dim3(256), 0, stream,                        (const T*)params, (const I*)indices, (T*)out, nidx,                          row, nrows)
+  if (dtype == 0) {
+    if (idx_i32) GR(float, int32_t); else GR(float, int64_t);
+  } else {
+    if (idx_i32) GR(__bf16, int32_t); else GR(__bf16, int64_t);
+  }
+#undef GR
+  return hipGetLastError();
+}
+
+hipError_t stf_segment_sum(int dtype, int idx_i32, const void* data,
+                           const void* ids, float* out_f32, int64_t nidx,
+                           int64_t row, int64_t nseg, hipStream_t stream) {
+  dim3 grid = ElemwiseGrid(nidx * row, 256, 4);
+#define SS(T, I)                                                               hipLaunchKernelGGL((SegmentSumKernel<T, I>), grid, dim3(256), 0, stream,                        (const T*)data, (const I*)ids, out_f32, nidx, row, nseg)
+  if (dtype == 0) {
+    if (idx_i32) SS(float, int32_t); else SS(float, int64_t);
+  } else {
+    if (idx_i32) SS(__bf16, int32_t); else SS(__bf16, int64_t);
+  }
+#undef SS
   return hipGetLastError();
 }
 

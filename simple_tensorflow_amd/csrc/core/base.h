@@ -173,7 +173,7 @@ enum DataType : int {
 struct bfloat16 {
   uint16_t value = 0;
   bfloat16() = default;
-  explicit bfloat16(float f) {
+  bfloat16(float f) {  // NOLINT: implicit by design (compute in float)
     uint32_t bits;
     std::memcpy(&bits, &f, 4);
     // round to nearest even
@@ -186,6 +186,26 @@ struct bfloat16 {
     float f;
     std::memcpy(&f, &bits, 4);
     return f;
+  }
+  bfloat16& operator=(float f) {
+    *this = bfloat16(f);
+    return *this;
+  }
+  bfloat16& operator+=(float f) {
+    *this = bfloat16((float)*this + f);
+    return *this;
+  }
+  bfloat16& operator*=(float f) {
+    *this = bfloat16((float)*this * f);
+    return *this;
+  }
+  bfloat16& operator-=(float f) {
+    *this = bfloat16((float)*this - f);
+    return *this;
+  }
+  bfloat16& operator/=(float f) {
+    *this = bfloat16((float)*this / f);
+    return *this;
   }
 };
 

@@ -714,6 +714,7 @@ class UnsortedSegmentSumOp : public OpKernel {
   }
 };
 REGISTER_CPU_KERNEL_TYPES("UnsortedSegmentSum", UnsortedSegmentSumOp)
+REGISTER_KERNEL_BUILDER(Name("UnsortedSegmentSum").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), UnsortedSegmentSumOp<bfloat16>);
 
 // ------------------------------- OneHot -------------------------------------
 template <typename T>

@@ -99,6 +99,15 @@ DEFINE_CMP(FNe, !=)
   REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_CPU).TypeConstraint<double>("T"), BinaryOp<double, F>); \
   REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_CPU).TypeConstraint<int32_t>("T"), BinaryOp<int32_t, F>); \
   REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_CPU).TypeConstraint<int64_t>("T"), BinaryOp<int64_t, F>);
+#define REG_BINARY_BF16(OP, F)                                                \
+  REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), BinaryOp<bfloat16, F>);
+REG_BINARY_BF16("Add", FAdd)
+REG_BINARY_BF16("Sub", FSub)
+REG_BINARY_BF16("Mul", FMul)
+REG_BINARY_BF16("RealDiv", FDiv)
+REG_BINARY_BF16("Maximum", FMax)
+REG_BINARY_BF16("Minimum", FMin)
+#undef REG_BINARY_BF16
 REG_BINARY("Add", FAdd)
 REG_BINARY("Sub", FSub)
 REG_BINARY("Mul", FMul)
@@ -195,6 +204,16 @@ REG_UNARY_ALL("Abs", FAbs)
 REG_UNARY_ALL("Sign", FSign)
 REG_UNARY_ALL("Square", FSquare)
 REG_UNARY_ALL("Reciprocal", FRecip)
+#define REG_UNARY_BF16(OP, F)                                                 \
+  REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), UnaryOp<bfloat16, F>);
+REG_UNARY_BF16("Tanh", FTanh)
+REG_UNARY_BF16("Sigmoid", FSigmoid)
+REG_UNARY_BF16("Exp", FExp)
+REG_UNARY_BF16("Neg", FNeg)
+REG_UNARY_BF16("Square", FSquare)
+REG_UNARY_BF16("Sqrt", FSqrt)
+REG_UNARY_BF16("Rsqrt", FRsqrt)
+#undef REG_UNARY_BF16
 REG_UNARY_F("Sqrt", FSqrt)
 REG_UNARY_F("Rsqrt", FRsqrt)
 REG_UNARY_F("Exp", FExp)
@@ -248,6 +267,8 @@ class Grad2Op : public OpKernel {
 #define REG_GRAD2(OP, F)                                                      \
   REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_CPU).TypeConstraint<float>("T"), Grad2Op<float, F>); \
   REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_CPU).TypeConstraint<double>("T"), Grad2Op<double, F>);
+REGISTER_KERNEL_BUILDER(Name("SigmoidGrad").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), Grad2Op<bfloat16, FSigGrad>);
+REGISTER_KERNEL_BUILDER(Name("TanhGrad").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), Grad2Op<bfloat16, FTanhGrad>);
 REG_GRAD2("SigmoidGrad", FSigGrad)
 REG_GRAD2("TanhGrad", FTanhGrad)
 REG_GRAD2("RsqrtGrad", FRsqrtGrad)
@@ -306,6 +327,7 @@ class AddNOp : public OpKernel {
   }
 };
 REGISTER_CPU_KERNEL_TYPES("AddN", AddNOp)
+REGISTER_KERNEL_BUILDER(Name("AddN").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), AddNOp<bfloat16>);
 
 // -------------------------------- MatMul ------------------------------------
 template <typename T>
@@ -355,6 +377,43 @@ class MatMulOp : public OpKernel {
   bool ta_ = false, tb_ = false;
 };
 REGISTER_CPU_KERNEL_FLOATS("MatMul", MatMulOp)
+
+// bf16 CPU matmul: widen to f32, run the f32 loop, narrow (numerics match
+// the GPU MFMA path's f32 accumulation).
+class Bf16MatMulOp : public OpKernel {
+ public:
+  explicit Bf16MatMulOp(OpKernelConstruction* ctx) : OpKernel(ctx) {
+    ctx->GetAttr("transpose_a", &ta_);
+    ctx->GetAttr("transpose_b", &tb_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& a = ctx->input(0);
+    const Tensor& b = ctx->input(1);
+    int64_t m = ta_ ? a.dim_size(1) : a.dim_size(0);
+    int64_t k = ta_ ? a.dim_size(0) : a.dim_size(1);
+    int64_t n = tb_ ? b.dim_size(0) : b.dim_size(1);
+    Tensor* out = ctx->allocate_output(0, TensorShape({m, n}));
+    std::vector<float> af(a.NumElements()), bf(b.NumElements()),
+        cf(m * n, 0.f);
+    const bfloat16* ap = a.flat<bfloat16>();
+    const bfloat16* bp = b.flat<bfloat16>();
+    for (int64_t i = 0; i < a.NumElements(); ++i) af[i] = (float)ap[i];
+    for (int64_t i = 0; i < b.NumElements(); ++i) bf[i] = (float)bp[i];
+    int64_t lda = a.dim_size(1), ldb = b.dim_size(1);
+    for (int64_t i = 0; i < m; ++i)
+      for (int64_t kk = 0; kk < k; ++kk) {
+        float av = ta_ ? af[kk * lda + i] : af[i * lda + kk];
+        for (int64_t j = 0; j < n; ++j)
+          cf[i * n + j] += av * (tb_ ? bf[j * ldb + kk] : bf[kk * ldb + j]);
+      }
+    bfloat16* cp = out->flat<bfloat16>();
+    for (int64_t i = 0; i < m * n; ++i) cp[i] = bfloat16(cf[i]);
+  }
+
+ private:
+  bool ta_ = false, tb_ = false;
+};
+REGISTER_KERNEL_BUILDER(Name("MatMul").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), Bf16MatMulOp);
 
 // BatchMatMul: [..., M, K] x [..., K, N].
 template <typename T>
@@ -476,6 +535,8 @@ class ReduceOp : public OpKernel {
   REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_CPU).TypeConstraint<double>("T"), ReduceOp<double, R>); \
   REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_CPU).TypeConstraint<int32_t>("T"), ReduceOp<int32_t, R>); \
   REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_CPU).TypeConstraint<int64_t>("T"), ReduceOp<int64_t, R>);
+REGISTER_KERNEL_BUILDER(Name("Sum").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), ReduceOp<bfloat16, Red::SUM>);
+REGISTER_KERNEL_BUILDER(Name("Mean").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), ReduceOp<bfloat16, Red::MEAN>);
 REG_REDUCE("Sum", Red::SUM)
 REG_REDUCE("Mean", Red::MEAN)
 REG_REDUCE("Max", Red::MAX)
@@ -648,6 +709,7 @@ class BiasAddOp : public OpKernel {
   }
 };
 REGISTER_CPU_KERNEL_TYPES("BiasAdd", BiasAddOp)
+REGISTER_KERNEL_BUILDER(Name("BiasAdd").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), BiasAddOp<bfloat16>);
 
 template <typename T>
 class BiasAddGradOp : public OpKernel {
@@ -665,6 +727,7 @@ class BiasAddGradOp : public OpKernel {
   }
 };
 REGISTER_CPU_KERNEL_TYPES("BiasAddGrad", BiasAddGradOp)
+REGISTER_KERNEL_BUILDER(Name("BiasAddGrad").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), BiasAddGradOp<bfloat16>);
 
 template <typename T>
 class L2LossOp : public OpKernel {

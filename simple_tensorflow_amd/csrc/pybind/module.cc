@@ -218,6 +218,7 @@ class PySession {
 }  // namespace
 
 namespace stf {
+std::string HipGraphSelfTest();
 std::string RcclGetUniqueId();
 Status RcclInit(int nranks, int rank, const std::string& id_bytes);
 }
@@ -239,6 +240,11 @@ PYBIND11_MODULE(_core, m) {
       out[py::str(name)] = OpDefToPy(*OpRegistry::Global()->LookUp(name));
     }
     return out;
+  });
+  m.def("hipgraph_selftest", []() -> std::string {
+    // Probe: capture with cross-thread enqueue + an in-capture hipMalloc —
+    // mirrors what the executor does during a captured step.
+    return stf::HipGraphSelfTest();
   });
   m.def("rccl_get_unique_id", []() {
     return py::bytes(stf::RcclGetUniqueId());

@@ -158,9 +158,8 @@ Status DirectSession::BuildExecutors(const std::vector<std::string>& feeds,
         else dev = cpu;
       }
     }
-    if (!kernel_available(n, dev))
-      return errors::NotFound("No kernel for op ", n->op(), " (node ",
-                              n->name(), ") on any device");
+    // A node with no kernel anywhere may still be pruned away (dangling
+    // gradient subgraphs); defer the hard error to executor creation.
     n->assigned_device = CanonicalDevice(dev->name());
   }
 

@@ -629,8 +629,10 @@ def apply_op(op_type, *args, **kwargs):
     for arg in od['output_arg']:
         n = 1
         if arg['number_attr']:
-            n = attrs.get(arg['number_attr'],
-                          ('i', _attr_default(attr_defs, arg['number_attr'])))[1]
+            if arg['number_attr'] in attrs:
+                n = attrs[arg['number_attr']][1]
+            else:
+                n = _attr_default(attr_defs, arg['number_attr'])
         if arg['type_list_attr']:
             lv = attrs[arg['type_list_attr']][1]
             for t in lv['type']:

@@ -1,0 +1,141 @@
+"""RNN cells + static unroll (analog of reference python/ops/rnn.py +
+rnn_cell_impl.py — the reference trim only ships the _RNNCell ABC since cells
+lived in contrib; these are the standard TF-1.0-era cell definitions,
+bf16-compute/f32-master capable)."""
+from simple_tensorflow_amd.python.framework import dtypes, ops
+from simple_tensorflow_amd.python.ops import (array_ops, init_ops, math_ops,
+                                              nn_ops, variable_scope)
+
+
+class RNNCell(object):
+    @property
+    def state_size(self):
+        raise NotImplementedError
+
+    @property
+    def output_size(self):
+        raise NotImplementedError
+
+    def zero_state(self, batch_size, dtype):
+        raise NotImplementedError
+
+    def __call__(self, inputs, state, scope=None):
+        raise NotImplementedError
+
+
+class BasicRNNCell(RNNCell):
+    def __init__(self, num_units, activation=math_ops.tanh):
+        self._num_units = num_units
+        self._activation = activation
+
+    @property
+    def state_size(self):
+        return self._num_units
+
+    @property
+    def output_size(self):
+        return self._num_units
+
+    def zero_state(self, batch_size, dtype):
+        return array_ops.zeros([batch_size, self._num_units], dtype)
+
+    def __call__(self, inputs, state, scope=None):
+        with variable_scope.variable_scope(scope or 'basic_rnn_cell'):
+            in_dim = inputs._shape[-1]
+            w = variable_scope.get_variable(
+                'kernel', [in_dim + self._num_units, self._num_units])
+            b = variable_scope.get_variable(
+                'bias', [self._num_units],
+                initializer=init_ops.zeros_initializer())
+            cat = array_ops.concat([inputs, state], 1)
+            wk = w.ref()
+            bk = b.ref()
+            if inputs.dtype == dtypes.bfloat16:
+                wk = math_ops.cast(wk, dtypes.bfloat16)
+                bk = math_ops.cast(bk, dtypes.bfloat16)
+            out = self._activation(
+                nn_ops.bias_add(math_ops.matmul(cat, wk), bk))
+            return out, out
+
+
+class BasicLSTMCell(RNNCell):
+    def __init__(self, num_units, forget_bias=1.0, activation=math_ops.tanh):
+        self._num_units = num_units
+        self._forget_bias = forget_bias
+        self._activation = activation
+
+    @property
+    def state_size(self):
+        return (self._num_units, self._num_units)
+
+    @property
+    def output_size(self):
+        return self._num_units
+
+    def zero_state(self, batch_size, dtype):
+        return (array_ops.zeros([batch_size, self._num_units], dtype),
+                array_ops.zeros([batch_size, self._num_units], dtype))
+
+    def __call__(self, inputs, state, scope=None):
+        c, h = state
+        with variable_scope.variable_scope(scope or 'basic_lstm_cell'):
+            in_dim = inputs._shape[-1]
+            w = variable_scope.get_variable(
+                'kernel', [in_dim + self._num_units, 4 * self._num_units])
+            b = variable_scope.get_variable(
+                'bias', [4 * self._num_units],
+                initializer=init_ops.zeros_initializer())
+            cat = array_ops.concat([inputs, h], 1)
+            wk = w.ref()
+            bk = b.ref()
+            if inputs.dtype == dtypes.bfloat16:
+                wk = math_ops.cast(wk, dtypes.bfloat16)
+                bk = math_ops.cast(bk, dtypes.bfloat16)
+            gates = nn_ops.bias_add(math_ops.matmul(cat, wk), bk)
+            i, j, f, o = array_ops.split(gates, 4, axis=1)
+            fb = ops.constant(self._forget_bias, dtype=gates.dtype)
+            new_c = c * math_ops.sigmoid(f + fb) + \
+                math_ops.sigmoid(i) * self._activation(j)
+            new_h = self._activation(new_c) * math_ops.sigmoid(o)
+            return new_h, (new_c, new_h)
+
+
+class MultiRNNCell(RNNCell):
+    def __init__(self, cells, state_is_tuple=True):
+        self._cells = cells
+
+    @property
+    def output_size(self):
+        return self._cells[-1].output_size
+
+    def zero_state(self, batch_size, dtype):
+        return tuple(c.zero_state(batch_size, dtype) for c in self._cells)
+
+    def __call__(self, inputs, state, scope=None):
+        new_states = []
+        x = inputs
+        with variable_scope.variable_scope(scope or 'multi_rnn_cell'):
+            for i, (cell, st) in enumerate(zip(self._cells, state)):
+                with variable_scope.variable_scope('cell_%d' % i):
+                    x, ns = cell(x, st)
+                new_states.append(ns)
+        return x, tuple(new_states)
+
+
+def static_rnn(cell, inputs, initial_state=None, dtype=None, scope=None):
+    """Statically-unrolled RNN (the reference's tf.nn.rnn; the PTB config's
+    execution style — while-loop dynamic_rnn is available for parity but the
+    unrolled graph is what the MI355X scheduler pipelines best)."""
+    outputs = []
+    state = initial_state
+    if state is None:
+        batch = inputs[0]._shape[0]
+        state = cell.zero_state(batch, dtype or inputs[0].dtype)
+    g = ops.get_default_graph()
+    with variable_scope.variable_scope(scope or 'rnn') as vs:
+        for t, x in enumerate(inputs):
+            if t == 1:
+                vs.reuse = True
+            out, state = cell(x, state)
+            outputs.append(out)
+    return outputs, state

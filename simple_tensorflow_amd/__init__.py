@@ -31,6 +31,8 @@ from simple_tensorflow_amd.python.training import saver as _saver
 from simple_tensorflow_amd.python.training import coordinator as _coord
 from simple_tensorflow_amd.python.training import input as _input
 from simple_tensorflow_amd.python.training import monitored_session as _ms
+from simple_tensorflow_amd.python.training import server_lib as _server_lib
+from simple_tensorflow_amd.python.training import device_setter as _device_setter
 from simple_tensorflow_amd.python.summary import summary as _summary_mod
 from simple_tensorflow_amd.python.summary import writer as _summary_writer
 from simple_tensorflow_amd.python.framework import errors  # noqa: F401
@@ -278,6 +280,9 @@ class _TrainModule(object):
     StepCounterHook = _ms.StepCounterHook
     NanTensorHook = _ms.NanTensorHook
     SummarySaverHook = _ms.SummarySaverHook
+    Server = _server_lib.Server
+    ClusterSpec = _server_lib.ClusterSpec
+    replica_device_setter = staticmethod(_device_setter.replica_device_setter)
 
 
 train = _TrainModule()

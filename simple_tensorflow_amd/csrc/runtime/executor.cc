@@ -195,7 +195,7 @@ class ExecutorState {
   }
 
   void Schedule(const TaggedNode& t) {
-    args_.pool->Schedule([this, t]() { Process(t); });
+    args_.schedule([this, t]() { Process(t); });
   }
   void ScheduleLocked(const TaggedNode& t) { Schedule(t); }
 

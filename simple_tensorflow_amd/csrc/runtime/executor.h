@@ -54,7 +54,10 @@ class ResourceMgr;  // defined in kernels (queues etc.)
 struct ExecutorArgs {
   int64_t step_id = 0;
   Rendezvous* rendezvous = nullptr;
-  ThreadPool* pool = nullptr;
+  // Work scheduler (normally ThreadPool::Schedule; during hipGraph capture a
+  // single-threaded trampoline so every kernel enqueue happens on the
+  // capturing thread).
+  std::function<void(std::function<void()>)> schedule;
   void* resource_mgr = nullptr;
   std::function<bool()> is_cancelled;
 };

@@ -454,17 +454,55 @@ Status DirectSession::Run(
   std::condition_variable cv;
   int remaining = (int)ek->items.size();
   Status agg;
+  // During hipGraph capture every enqueue must come from this thread: use a
+  // single-threaded trampoline queue instead of the pool.
+  std::deque<std::function<void()>> inline_q;
+  std::mutex inline_mu;
+  ThreadPool* pool = pool_.get();
+  std::function<void(std::function<void()>)> scheduler;
+  if (do_capture) {
+    scheduler = [&](std::function<void()> fn) {
+      std::lock_guard<std::mutex> l(inline_mu);
+      inline_q.push_back(std::move(fn));
+    };
+  } else {
+    scheduler = [pool](std::function<void()> fn) {
+      pool->Schedule(std::move(fn));
+    };
+  }
   for (auto& item : ek->items) {
     ExecutorArgs args;
     args.step_id = step_id;
     args.rendezvous = &rendez;
-    args.pool = pool_.get();
+    args.schedule = scheduler;
     args.resource_mgr = resource_mgr_;
     item.executor->RunAsync(args, [&](Status s) {
       std::lock_guard<std::mutex> l(mu);
       if (!s.ok() && agg.ok()) agg = s;
       if (--remaining == 0) cv.notify_one();
     });
+  }
+  if (do_capture) {
+    // Drain the trampoline on this thread until all executors complete.
+    for (;;) {
+      std::function<void()> fn;
+      {
+        std::lock_guard<std::mutex> l(inline_mu);
+        if (!inline_q.empty()) {
+          fn = std::move(inline_q.front());
+          inline_q.pop_front();
+        }
+      }
+      if (fn) {
+        fn();
+        continue;
+      }
+      std::lock_guard<std::mutex> l(mu);
+      if (remaining == 0) break;
+      // nothing queued but not done: an async kernel callback may still be
+      // pending on another thread (not expected in capture-eligible graphs)
+      std::this_thread::yield();
+    }
   }
 
   // Collect fetches (they arrive as the graph runs).

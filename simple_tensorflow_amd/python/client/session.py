@@ -26,7 +26,11 @@ class Session(object):
         cpu_only = False
         if config is not None and isinstance(config, dict):
             cpu_only = config.get('device_count', {}).get('GPU', 1) == 0
-        self._core = _core.Session(cpu_only)
+        if target.startswith('grpc://'):
+            from simple_tensorflow_amd.python.training import server_lib
+            self._core = server_lib.GrpcRemoteCore(target)
+        else:
+            self._core = _core.Session(cpu_only)
         self._created = False
         self._serialized_nodes = 0
         self._lock = threading.Lock()

@@ -69,6 +69,8 @@ control_dependencies = _ops.control_dependencies
 colocate_with = _ops.colocate_with
 convert_to_tensor = _ops.convert_to_tensor
 constant = _ops.constant
+from simple_tensorflow_amd.python.framework import importer as _importer
+import_graph_def = _importer.import_graph_def
 RegisterGradient = _ops.RegisterGradient
 NoGradient = _ops.NoGradient
 NotDifferentiable = _ops.NotDifferentiable
@@ -251,6 +253,17 @@ class _TrainModule(object):
     piecewise_constant = staticmethod(_training_util.piecewise_constant)
     ExponentialMovingAverage = _training_util.ExponentialMovingAverage
     Saver = _saver.Saver
+
+    @staticmethod
+    def export_meta_graph(filename=None, **kw):
+        from simple_tensorflow_amd.python.framework import meta_graph
+        return meta_graph.export_meta_graph(filename=filename, **kw)
+
+    @staticmethod
+    def import_meta_graph(meta_graph_or_file, **kw):
+        from simple_tensorflow_amd.python.framework import meta_graph
+        return meta_graph.import_meta_graph(meta_graph_or_file, **kw)
+
     latest_checkpoint = staticmethod(_saver.latest_checkpoint)
     get_checkpoint_state = staticmethod(_saver.get_checkpoint_state)
     update_checkpoint_state = staticmethod(_saver.update_checkpoint_state)

@@ -12,6 +12,7 @@
 #include <map>
 #include <mutex>
 #include <set>
+#include <vector>
 
 #include "framework/device.h"
 #include "framework/op_kernel.h"
@@ -222,6 +223,12 @@ class GpuDevice : public Device {
     capturing_ = false;
     hipGraph_t graph = nullptr;
     HIP_CHECK_STATUS(hipStreamEndCapture(compute_, &graph));
+    if (const char* dot = getenv("STF_GRAPH_DOT")) {
+      size_t n_nodes = 0;
+      hipGraphGetNodes(graph, nullptr, &n_nodes);
+      LOG(INFO) << "captured hipGraph with " << n_nodes << " nodes -> " << dot;
+      hipGraphDebugDotPrint(graph, dot, 1 /*verbose*/);
+    }
     hipGraphExec_t exec = nullptr;
     hipError_t e = hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0);
     hipGraphDestroy(graph);
@@ -264,14 +271,18 @@ class GpuDevice : public Device {
     HIP_CHECK_STATUS(hipMemcpyAsync(dev.raw_data(), src.raw_data(),
                                     src.TotalBytes(), hipMemcpyHostToDevice,
                                     h2d_));
-    // Make the compute stream wait for the transfer; then sync h2d so the
-    // host source buffer may be released by the caller.
-    hipEvent_t ev;
-    HIP_CHECK_STATUS(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
-    HIP_CHECK_STATUS(hipEventRecord(ev, h2d_));
-    HIP_CHECK_STATUS(hipStreamWaitEvent(compute_, ev, 0));
+    // Sync the copy: the host source buffer may be released by the caller,
+    // and completion here also orders the data before any compute-stream
+    // consumer enqueued after this call.
     HIP_CHECK_STATUS(hipStreamSynchronize(h2d_));
-    HIP_CHECK_STATUS(hipEventDestroy(ev));
+    if (capturing_) {
+      // The h2d stream is not part of the capture, so this copy is NOT a
+      // node of the hipGraph: replayed kernels will re-read `dev` directly.
+      // The value is static (capture-eligible steps only move shape-derived
+      // host scalars), but the buffer must outlive every replay — pin it.
+      std::lock_guard<std::mutex> l(keepalive_mu_);
+      capture_keepalive_.push_back(dev);
+    }
     *dst = dev;
     return Status::OK();
   }
@@ -281,6 +292,11 @@ class GpuDevice : public Device {
   GpuBfcAllocator bfc_;
   hipStream_t compute_, h2d_, d2h_;
   bool capturing_ = false;
+  // Device buffers referenced by captured hipGraphs but not produced inside
+  // them (host->device shape scalars copied at capture time). A few bytes
+  // per captured graph; lives as long as the device.
+  std::mutex keepalive_mu_;
+  std::vector<Tensor> capture_keepalive_;
 };
 
 void AddGpuDevices(DeviceMgr* mgr) {

@@ -493,6 +493,21 @@ class ExecutorState {
     OpKernelContext ctx(item.kernel, dev, std::move(inputs));
     FillCtx(&ctx, t);
     dev->Compute(item.kernel, &ctx);
+    if (getenv("STF_DEBUG_LAUNCH")) {
+      std::string line = n->name() + " (" + n->op() + ") in:";
+      char b[32];
+      for (int i = 0; i < ctx.num_inputs(); ++i) {
+        snprintf(b, sizeof(b), " %p", ctx.input(i).raw_data());
+        line += b;
+      }
+      line += " out:";
+      for (auto& o : ctx.outputs()) {
+        snprintf(b, sizeof(b), " %p", o.raw_data());
+        line += b;
+      }
+      fprintf(stderr, "[launch cap=%d] %s\n", dev->capturing() ? 1 : 0,
+              line.c_str());
+    }
     if (!ctx.status().ok()) {
       Fail(Status(ctx.status().code(),
                   "node " + n->name() + " (" + n->op() + "): " +

@@ -549,6 +549,10 @@ Status DirectSession::Run(
       if (cs.ok()) {
         ek->graph_exec = exec;
         ek->capture_device = capture_dev;
+        // Stream capture records but does not execute: launch the graph once
+        // now so this Run() still performs exactly one step.
+        Status ls = capture_dev->LaunchCapturedGraph(exec);
+        if (!ls.ok()) return ls;
       } else {
         LOG(WARN) << "hipGraph capture failed; falling back to eager: "
                   << cs.ToString();

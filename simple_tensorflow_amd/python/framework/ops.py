@@ -351,8 +351,14 @@ class Graph(object):
         attrs = attrs or {}
         name = name or op_type
         if name.endswith('/'):
-            name = name[:-1]
-        node_name = self.unique_name(name)
+            # Trailing '/' means an absolute name (already fully scoped) —
+            # the reference's Graph.unique_name convention.
+            # Used verbatim: the caller owns uniqueness (typically via the
+            # enclosing name_scope, which is itself uniquified).
+            node_name = name[:-1]
+            self._names_used.setdefault(node_name, 1)
+        else:
+            node_name = self.unique_name(name)
         if output_is_ref is None:
             output_is_ref = [False] * len(output_dtypes)
         # While-loop capture: an op built inside a while context must not read

@@ -248,7 +248,25 @@ def _tile_spec(op, grad, input_shape):
 
 @RegisterGradient('Mean')
 def _mean_grad(op, grad):
-    input_shape = array_ops.shape(op.inputs[0])
+    x = op.inputs[0]
+    # Static fast path: with a fully-known input shape the 1/N factor is a
+    # compile-time constant. This keeps the whole step on-device (no
+    # Size->Cast host round trip), which both drops 3 kernels per Mean and
+    # keeps the step hipGraph-capturable without host-pinned scalars.
+    if x._shape is not None and all(d is not None for d in x._shape) and \
+            op.outputs[0]._shape is not None and \
+            all(d is not None for d in op.outputs[0]._shape):
+        in_n = 1
+        for d in x._shape:
+            in_n *= d
+        out_n = 1
+        for d in op.outputs[0]._shape:
+            out_n *= d
+        scale = ops.constant(float(out_n) / float(in_n), dtype=grad.dtype)
+        input_shape = array_ops.shape(x)
+        sum_grad = _tile_spec(op, grad * scale, input_shape)
+        return [sum_grad, None]
+    input_shape = array_ops.shape(x)
     sum_grad = _tile_spec(op, grad, input_shape)
     in_n = math_ops.cast(array_ops.size(op.inputs[0]), grad.dtype)
     out_n = math_ops.cast(array_ops.size(op.outputs[0]), grad.dtype)

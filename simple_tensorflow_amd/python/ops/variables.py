@@ -19,8 +19,8 @@ class Variable(object):
             if shape is None or any(d is None for d in shape):
                 raise ValueError('Variable needs fully-defined initial shape')
             self._variable = state_ops.variable_op(
-                shape, self._initial_value.dtype, name=scope[:-1] if scope
-                else 'Variable')
+                shape, self._initial_value.dtype,
+                name=scope if scope else 'Variable')
             self._initializer_op = state_ops.assign(
                 self._variable, self._initial_value,
                 validate_shape=validate_shape).op
@@ -31,6 +31,18 @@ class Variable(object):
             cols.append(ops.GraphKeys.TRAINABLE_VARIABLES)
         for c in cols:
             g.add_to_collection(c, self)
+
+    @classmethod
+    def _from_graph_elements(cls, variable_tensor, initializer_op, snapshot):
+        """Rebuild a Variable wrapper around already-imported graph nodes
+        (used by meta_graph.import_meta_graph; analog of reference
+        Variable.from_proto)."""
+        v = cls.__new__(cls)
+        v._variable = variable_tensor
+        v._initializer_op = initializer_op
+        v._snapshot = snapshot
+        v._initial_value = None
+        return v
 
     @property
     def name(self):

@@ -51,6 +51,25 @@ class Saver(object):
             self._restore_op = control_flow_ops.group(*assigns,
                                                       name='restore_all')
 
+    @classmethod
+    def _from_imported(cls, filename_tensor, save_op, restore_op,
+                       max_to_keep=5):
+        """Wrap save/restore nodes recovered by import_meta_graph."""
+        s = cls.__new__(cls)
+        s._filename = filename_tensor
+        s._save_op = save_op
+        s._restore_op = restore_op
+        s._max_to_keep = max_to_keep
+        s._kept = []
+        s._names = []
+        s._vars = []
+        return s
+
+    def export_meta_graph(self, filename=None, collection_list=None):
+        from simple_tensorflow_amd.python.framework import meta_graph
+        return meta_graph.export_meta_graph(
+            filename=filename, saver=self, collection_list=collection_list)
+
     def save(self, sess, save_path, global_step=None,
              latest_filename='checkpoint', write_meta_graph=True,
              meta_graph_suffix='meta'):
@@ -67,6 +86,11 @@ class Saver(object):
             path = save_path
         sess.run(self._save_op, feed_dict={self._filename: path})
         self._record_checkpoint(path, latest_filename)
+        if write_meta_graph:
+            try:
+                self.export_meta_graph(path + '.' + meta_graph_suffix)
+            except Exception:
+                pass  # meta export must not fail the checkpoint itself
         return path
 
     def restore(self, sess, save_path):

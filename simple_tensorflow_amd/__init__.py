@@ -319,4 +319,6 @@ class _PythonIoModule(object):
 
 python_io = _PythonIoModule()
 
+from simple_tensorflow_amd.python import saved_model  # noqa: E402,F401
+
 __version__ = '0.1.0'

@@ -76,6 +76,11 @@ def import_graph_def(graph_def, input_map=None, return_elements=None,
             g._nodes_by_name[full] = op
             g._node_list.append(op)
             g._names_used[full] = 1
+            # Mark ancestor scopes used so later name_scope()s uniquify
+            # around the imported names (save/* -> save_1/*).
+            parts = full.split('/')
+            for i in range(1, len(parts)):
+                g._names_used.setdefault('/'.join(parts[:i]), 1)
             g.version += 1
         created[nd['name']] = op
 

@@ -142,3 +142,58 @@ def parse_graph_def(data):
         if f == 1:
             nodes.append(parse_node_def(v))
     return nodes
+
+
+def parse_tensor_proto(data):
+    """TensorProto: dtype=1, tensor_shape=2, tensor_content=4, float_val=5,
+    int_val=7, string_val=8, int64_val=10, bool_val=11. Returns
+    (dtype_enum, dims, content_bytes, repeated_vals)."""
+    dtype = 0
+    dims = []
+    content = b''
+    vals = []
+    for f, w, v in _fields(data):
+        if f == 1:
+            dtype = v
+        elif f == 2:
+            dims = parse_tensor_shape(v) or []
+        elif f == 4:
+            content = bytes(v)
+        elif f == 5:
+            if w == 2:
+                for k in range(0, len(v), 4):
+                    vals.append(struct.unpack('<f', v[k:k + 4])[0])
+            else:
+                vals.append(struct.unpack('<f', v)[0])
+        elif f in (7, 10):
+            if w == 2:
+                off = 0
+                while off < len(v):
+                    x, off = _read_varint(v, off)
+                    vals.append(x if x < (1 << 62) else x - (1 << 64))
+            else:
+                vals.append(v if v < (1 << 62) else v - (1 << 64))
+        elif f == 8:
+            vals.append(bytes(v))
+        elif f == 11:
+            vals.append(bool(v))
+    return dtype, dims, content, vals
+
+
+def np_from_tensor_proto(data):
+    import numpy as np
+    from simple_tensorflow_amd.python.framework import dtypes
+    dtype, dims, content, vals = parse_tensor_proto(data)
+    npdt = dtypes.as_dtype(dtype).as_numpy_dtype
+    n = 1
+    for d in dims:
+        n *= d
+    if content:
+        arr = np.frombuffer(content, dtype=npdt)
+    elif vals:
+        arr = np.array(vals, dtype=npdt)
+        if arr.size == 1 and n > 1:
+            arr = np.full(n, arr[0], dtype=npdt)
+    else:
+        arr = np.zeros(n, dtype=npdt)
+    return arr.reshape(dims)

@@ -59,15 +59,15 @@ def fold_batch_norms(input_graph_def):
         new_nodes.append(_make_const(w_node['name'] + '_bnfold', w_f, wdt))
         new_nodes.append(_make_const(n['name'] + '_bias', b_f,
                                      int(dtypes.float32)))
-        # Rewire: conv reads folded weights; a BiasAdd replaces the BN.
+        # Rewire: conv reads folded weights; a BiasAdd replaces the BN under
+        # the BN's own name, so downstream consumers keep working.
         src['input'][1] = w_node['name'] + '_bnfold'
         new_nodes.append({
-            'name': n['name'] + '_biasadd', 'op': 'BiasAdd',
+            'name': n['name'], 'op': 'BiasAdd',
             'input': [n['input'][0], n['name'] + '_bias'],
             'device': n.get('device', ''),
             'attr': {'T': dict(n['attr']).get('T', ('type',
                                                     int(dtypes.float32)))}})
-        rewrites[n['name']] = n['name'] + '_biasadd'
         removed.add(n['name'])
 
     out = []

@@ -18,6 +18,12 @@ hipError_t stf_gemm_bf16_nt(const void*, const void*, void*, const void*,
                             hipStream_t);
 hipError_t stf_gemm_bf16_nt_splitk(const void*, const void*, void*, int64_t,
                                    int64_t, int64_t, int, hipStream_t);
+hipError_t stf_gemm_bf16(const void*, const void*, void*, const void*,
+                         int64_t, int64_t, int64_t, int64_t, int64_t, float,
+                         int, int, int, int, hipStream_t);
+hipError_t stf_gemm_bf16_splitk(const void*, const void*, void*, int64_t,
+                                int64_t, int64_t, int64_t, int64_t, int, int,
+                                int, hipStream_t);
 hipError_t stf_gemm_f32_nt(const void*, const void*, void*, int64_t, int64_t,
                            int64_t, hipStream_t);
 hipError_t stf_unary(int, int, const void*, void*, int64_t, hipStream_t);
@@ -108,7 +114,7 @@ inline int DtypeCode(DataType dt) { return dt == DT_FLOAT ? 0 : 1; }
 // ~= 0.5-2x CU count); 1 = no split.
 inline int PickSplitK(int64_t M, int64_t N, int64_t K) {
   int64_t tiles = ((M + 127) / 128) * ((N + 127) / 128);
-  if (tiles >= 256 || K < 4096) return 1;
+  if (tiles >= 256 || K < 1024) return 1;
   int64_t want = 512 / (tiles ? tiles : 1);
   int64_t maxk = K / 512;  // keep >= 8 K-iters per slice
   if (maxk < 1) maxk = 1;
@@ -117,18 +123,30 @@ inline int PickSplitK(int64_t M, int64_t N, int64_t K) {
 }
 
 // GEMM helper: picks plain vs split-K (f32 scratch + cast) automatically.
-inline hipError_t GemmBf16Auto(OpKernelContext* ctx, const void* A,
-                               const void* B, void* C_bf16, int64_t M,
-                               int64_t N, int64_t K, hipStream_t s) {
+// a_km/b_km: the operand is stored contraction-major ([K,M]/[K,N], row
+// stride lda/ldb) and transposed in-kernel during LDS staging.
+inline hipError_t GemmBf16AutoEx(OpKernelContext* ctx, const void* A,
+                                 const void* B, void* C_bf16, int64_t M,
+                                 int64_t N, int64_t K, int64_t lda,
+                                 int64_t ldb, int a_km, int b_km,
+                                 hipStream_t s) {
   int sk = PickSplitK(M, N, K);
   if (sk <= 1)
-    return stf_gemm_bf16_nt(A, B, C_bf16, nullptr, M, N, K, 0.f, 1, 0, s);
+    return stf_gemm_bf16(A, B, C_bf16, nullptr, M, N, K, lda, ldb, 0.f, a_km,
+                         b_km, 1, 0, s);
   Tensor scratch = ctx->allocate_temp(DT_FLOAT, TensorShape({M, N}));
   hipError_t e = hipMemsetAsync(scratch.raw_data(), 0, M * N * 4, s);
   if (e != hipSuccess) return e;
-  e = stf_gemm_bf16_nt_splitk(A, B, scratch.raw_data(), M, N, K, sk, s);
+  e = stf_gemm_bf16_splitk(A, B, scratch.raw_data(), M, N, K, lda, ldb, a_km,
+                           b_km, sk, s);
   if (e != hipSuccess) return e;
   return stf_cast(0, 1, scratch.raw_data(), C_bf16, M * N, s);
+}
+
+inline hipError_t GemmBf16Auto(OpKernelContext* ctx, const void* A,
+                               const void* B, void* C_bf16, int64_t M,
+                               int64_t N, int64_t K, hipStream_t s) {
+  return GemmBf16AutoEx(ctx, A, B, C_bf16, M, N, K, K, K, 0, 0, s);
 }
 inline int CastCode(DataType dt) {
   switch (dt) {
@@ -377,24 +395,55 @@ class GpuMatMulOp : public OpKernel {
     int64_t n = tb_ ? b.dim_size(0) : b.dim_size(1);
     Tensor* y = ctx->allocate_output(0, TensorShape({m, n}));
     size_t es = DataTypeSize(a.dtype());
-    // A effective [M,K]
-    Tensor a_eff = a;
-    if (ta_) {
-      a_eff = ctx->allocate_temp(a.dtype(), TensorShape({m, k}));
-      OP_HIP_OK(ctx, stf_transpose2d((int)es, a.raw_data(), a_eff.raw_data(),
-                                     a.dim_size(0), a.dim_size(1), s));
-    }
-    // B effective [N,K]
-    Tensor b_eff = b;
-    if (!tb_) {
-      b_eff = ctx->allocate_temp(b.dtype(), TensorShape({n, k}));
-      OP_HIP_OK(ctx, stf_transpose2d((int)es, b.raw_data(), b_eff.raw_data(),
-                                     b.dim_size(0), b.dim_size(1), s));
-    }
     if (a.dtype() == DT_BFLOAT16) {
-      OP_HIP_OK(ctx, GemmBf16Auto(ctx, a_eff.raw_data(), b_eff.raw_data(),
-                                  y->raw_data(), m, n, k, s));
+      // The bf16 GEMM stages either operand layout directly (K-major
+      // staging for transposed sides) as long as the row stride keeps 16B
+      // global loads aligned; otherwise fall back to a pre-transpose.
+      const void* ap = a.raw_data();
+      const void* bp = b.raw_data();
+      int a_km = ta_ ? 1 : 0;
+      int64_t lda = a.dim_size(1);
+      Tensor a_tmp;
+      if (ta_ && (m & 7) != 0) {
+        a_tmp = ctx->allocate_temp(a.dtype(), TensorShape({m, k}));
+        OP_HIP_OK(ctx, stf_transpose2d((int)es, a.raw_data(),
+                                       a_tmp.raw_data(), a.dim_size(0),
+                                       a.dim_size(1), s));
+        ap = a_tmp.raw_data();
+        a_km = 0;
+        lda = k;
+      }
+      int b_km = tb_ ? 0 : 1;
+      int64_t ldb = b.dim_size(1);
+      Tensor b_tmp;
+      if (!tb_ && (n & 7) != 0) {
+        b_tmp = ctx->allocate_temp(b.dtype(), TensorShape({n, k}));
+        OP_HIP_OK(ctx, stf_transpose2d((int)es, b.raw_data(),
+                                       b_tmp.raw_data(), b.dim_size(0),
+                                       b.dim_size(1), s));
+        bp = b_tmp.raw_data();
+        b_km = 0;
+        ldb = k;
+      }
+      OP_HIP_OK(ctx, GemmBf16AutoEx(ctx, ap, bp, y->raw_data(), m, n, k, lda,
+                                    ldb, a_km, b_km, s));
     } else {
+      // A effective [M,K]
+      Tensor a_eff = a;
+      if (ta_) {
+        a_eff = ctx->allocate_temp(a.dtype(), TensorShape({m, k}));
+        OP_HIP_OK(ctx, stf_transpose2d((int)es, a.raw_data(),
+                                       a_eff.raw_data(), a.dim_size(0),
+                                       a.dim_size(1), s));
+      }
+      // B effective [N,K]
+      Tensor b_eff = b;
+      if (!tb_) {
+        b_eff = ctx->allocate_temp(b.dtype(), TensorShape({n, k}));
+        OP_HIP_OK(ctx, stf_transpose2d((int)es, b.raw_data(),
+                                       b_eff.raw_data(), b.dim_size(0),
+                                       b.dim_size(1), s));
+      }
       OP_HIP_OK(ctx, stf_gemm_f32_nt(a_eff.raw_data(), b_eff.raw_data(),
                                      y->raw_data(), m, n, k, s));
     }
@@ -465,7 +514,10 @@ class GpuConv2DOp : public OpKernel {
     int64_t rsc = g.RSC();
     // 1x1/s1 uses x directly as the GEMM A (ld = C) — no padding there.
     int64_t rscp = g.is_1x1_s1() ? rsc : g.RSCp();
-    // weights -> [K, RSCp] (zero-padded K dim)
+    // weights stay [RSC(p), K] and are read contraction-major by the GEMM
+    // (b_km staging) — no transpose kernel. Zero-pad the RSC dim only when
+    // it is not 64-aligned (conv1's 147 -> 192); rows [rsc, rscp) are zero,
+    // matching the zero-padded im2col columns.
     Tensor wsrc = w;
     if (rscp != rsc) {
       wsrc = ctx->allocate_temp(DT_BFLOAT16, TensorShape({rscp, g.K}));
@@ -474,9 +526,6 @@ class GpuConv2DOp : public OpKernel {
                                     rsc * g.K * 2, hipMemcpyDeviceToDevice,
                                     s));
     }
-    Tensor wt = ctx->allocate_temp(DT_BFLOAT16, TensorShape({g.K, rscp}));
-    OP_HIP_OK(ctx, stf_transpose2d(2, wsrc.raw_data(), wt.raw_data(), rscp,
-                                   g.K, s));
     const void* col_data = x.raw_data();
     Tensor col;
     if (!g.is_1x1_s1()) {
@@ -489,9 +538,18 @@ class GpuConv2DOp : public OpKernel {
                                      (int)g.pw, (int)g.P, (int)g.Q, rscp, s));
       col_data = col.raw_data();
     }
-    OP_HIP_OK(ctx, stf_gemm_bf16_nt(col_data, wt.raw_data(), y->raw_data(),
-                                    nullptr, g.M(), g.K, rscp, 0.f, 1, 0,
-                                    s));
+    if ((g.K & 7) == 0) {
+      OP_HIP_OK(ctx, stf_gemm_bf16(col_data, wsrc.raw_data(), y->raw_data(),
+                                   nullptr, g.M(), g.K, rscp, rscp, g.K, 0.f,
+                                   0, 1, 1, 0, s));
+    } else {
+      Tensor wt = ctx->allocate_temp(DT_BFLOAT16, TensorShape({g.K, rscp}));
+      OP_HIP_OK(ctx, stf_transpose2d(2, wsrc.raw_data(), wt.raw_data(), rscp,
+                                     g.K, s));
+      OP_HIP_OK(ctx, stf_gemm_bf16_nt(col_data, wt.raw_data(), y->raw_data(),
+                                      nullptr, g.M(), g.K, rscp, 0.f, 1, 0,
+                                      s));
+    }
   }
 
  private:
@@ -557,7 +615,33 @@ class GpuConv2DBackpropFilterOp : public OpKernel {
                                     &g));
     Tensor* dw = ctx->allocate_output(0, f_shape);
     int64_t rsc = g.RSC();
-    // dW[RSC, K] = col[M, RSC]^T x dy[M, K]  -> NT with A=colT, B=dyT
+    // dW[RSC, K] = col[M, RSC]^T x dy[M, K]: both operands are contraction
+    // (M-)major as stored, so the GEMM's K-major staging reads them directly
+    // — no transpose kernels (this was 2 full passes over the im2col matrix).
+    bool km_ok = (g.K & 7) == 0 && (!g.is_1x1_s1() || (g.C & 7) == 0);
+    if (km_ok) {
+      const void* col_data = x.raw_data();
+      int64_t lda = g.C;
+      Tensor col;
+      if (!g.is_1x1_s1()) {
+        int64_t rscp = g.RSCp();
+        col = ctx->allocate_temp(DT_BFLOAT16, TensorShape({g.M(), rscp}));
+        if (rscp != rsc)
+          OP_HIP_OK(ctx,
+                    hipMemsetAsync(col.raw_data(), 0, g.M() * rscp * 2, s));
+        OP_HIP_OK(ctx, stf_im2col_bf16(x.raw_data(), col.raw_data(), (int)g.N,
+                                       (int)g.H, (int)g.W, (int)g.C, (int)g.R,
+                                       (int)g.S, (int)g.sh, (int)g.sw,
+                                       (int)g.ph, (int)g.pw, (int)g.P,
+                                       (int)g.Q, rscp, s));
+        col_data = col.raw_data();
+        lda = rscp;
+      }
+      OP_HIP_OK(ctx, GemmBf16AutoEx(ctx, col_data, dy.raw_data(),
+                                    dw->raw_data(), rsc, g.K, g.M(), lda,
+                                    g.K, 1, 1, s));
+      return;
+    }
     Tensor colT = ctx->allocate_temp(DT_BFLOAT16, TensorShape({rsc, g.M()}));
     if (g.is_1x1_s1()) {
       OP_HIP_OK(ctx, stf_transpose2d(2, x.raw_data(), colT.raw_data(), g.M(),

@@ -44,6 +44,72 @@ __device__ __forceinline__ void StageFast(const uint16_t* __restrict__ src,
   }
 }
 
+// K-major staging: the operand is stored [K, R] row-major (contraction dim
+// outermost — e.g. im2col columns feeding dW, or a weight matrix [K, N] read
+// without a separate transpose pass). 16B global loads run along the
+// contiguous R dim; the transpose happens on the LDS write side as 8
+// staggered ds_write_b16s per load (the (e + lane) stagger spreads the
+// writes over banks — without it every lane in a row-group hits the same
+// bank because the 128B LDS row stride aliases all 32 banks).
+template <int BR>
+__device__ __forceinline__ void StageKMajor(const uint16_t* __restrict__ src,
+                                            int64_t ld, int col0, int64_t k0,
+                                            uint16_t* lds_base, int tid) {
+  constexpr int kRG = BR / 8;  // 16B row-groups per k row
+#pragma unroll
+  for (int p = 0; p < BR / 32; ++p) {
+    int idx = p * 256 + tid;
+    int k = idx / kRG;
+    int r0 = (idx % kRG) * 8;
+    const uint16_t* g = src + (k0 + k) * ld + col0 + r0;
+    uint16_t vals[8];
+    *(ulong2*)vals = *(const ulong2*)g;
+    int c8 = k >> 3;
+    int klo = (k & 7) * 2;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      int e = (i + tid) & 7;
+      int r = r0 + e;
+      *(uint16_t*)((char*)(lds_base + (int64_t)(r * 8 + (c8 ^ (r & 7))) * 8) +
+                   klo) = vals[e];
+    }
+  }
+}
+
+template <int BR>
+__device__ __forceinline__ void StageKMajorSafe(
+    const uint16_t* __restrict__ src, int64_t ld, int col0, int64_t k0,
+    int64_t cols, int64_t K, uint16_t* lds_base, int tid) {
+  constexpr int kRG = BR / 8;
+#pragma unroll
+  for (int p = 0; p < BR / 32; ++p) {
+    int idx = p * 256 + tid;
+    int k = idx / kRG;
+    int r0 = (idx % kRG) * 8;
+    uint16_t vals[8];
+    int64_t krow = k0 + k;
+    if (krow < K && col0 + r0 + 8 <= cols) {
+      *(ulong2*)vals = *(const ulong2*)(src + krow * ld + col0 + r0);
+    } else if (krow < K) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        vals[e] = (col0 + r0 + e < cols) ? src[krow * ld + col0 + r0 + e] : 0;
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) vals[e] = 0;
+    }
+    int c8 = k >> 3;
+    int klo = (k & 7) * 2;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      int e = (i + tid) & 7;
+      int r = r0 + e;
+      *(uint16_t*)((char*)(lds_base + (int64_t)(r * 8 + (c8 ^ (r & 7))) * 8) +
+                   klo) = vals[e];
+    }
+  }
+}
+
 // Guarded staging for edge blocks / K tails: scalar loads, zero padding.
 template <int BR>
 __device__ __forceinline__ void StageSafe(const uint16_t* __restrict__ src,
@@ -76,14 +142,19 @@ __device__ __forceinline__ void StageSafe(const uint16_t* __restrict__ src,
 // GEMM kernel template.
 //  WAVES_M x WAVES_N waves; each wave computes (WM*16) x (WN*16) outputs.
 //  BM = WAVES_M*WM*16, BN = WAVES_N*WN*16, BK = 64.
-template <int WAVES_M, int WAVES_N, int WM, int WN, bool OUT_BF16,
-          bool FUSE_RELU, bool SPLITK = false>
+//  A_KM/B_KM: operand stored contraction-major ([K, M] / [K, N] row-major;
+//  lda/ldb = that row stride) and transposed during LDS staging — this is
+//  how dW (= col^T · dy) and NN matmuls run without separate transpose
+//  kernels. With *_KM false the operand is [M, K] / [N, K] row-major (NT).
+template <int WAVES_M, int WAVES_N, int WM, int WN, bool A_KM, bool B_KM,
+          bool OUT_BF16, bool FUSE_RELU, bool SPLITK = false>
 __launch_bounds__(256) __global__ void GemmBf16NT(
-    const uint16_t* __restrict__ A,  // [M, K] row-major bf16
-    const uint16_t* __restrict__ B,  // [N, K] row-major bf16
+    const uint16_t* __restrict__ A,
+    const uint16_t* __restrict__ B,
     void* __restrict__ C,            // [M, N] f32 or bf16 (f32 for SPLITK)
     const float* __restrict__ bias,  // optional [N] f32 bias (nullptr = none)
-    int64_t M, int64_t N, int64_t K, float beta, int splitk = 1) {
+    int64_t M, int64_t N, int64_t K, int64_t lda, int64_t ldb, float beta,
+    int splitk = 1) {
   constexpr int BM = WAVES_M * WM * 16;
   constexpr int BN = WAVES_N * WN * 16;
   constexpr int BK = 64;
@@ -125,35 +196,42 @@ __launch_bounds__(256) __global__ void GemmBf16NT(
     kt_count = min(kt_total, kt_begin + per);
     if (kt_begin >= kt_count) return;
   }
+  // Stage one K-chunk of both operands into LDS buffer `buf`.
+  auto stage = [&](int buf, int64_t k0) {
+    bool kfull = k0 + BK <= K;
+    if (A_KM) {
+      if (a_interior && kfull)
+        StageKMajor<BM>(A, lda, (int)m0, k0, a_tile(buf), tid);
+      else
+        StageKMajorSafe<BM>(A, lda, (int)m0, k0, M, K, a_tile(buf), tid);
+    } else {
+      if (a_interior && kfull)
+        StageFast<BM>(A, lda, (int)m0, k0, a_tile(buf), tid);
+      else
+        StageSafe<BM>(A, lda, (int)m0, k0, M, K, a_tile(buf), tid);
+    }
+    if (B_KM) {
+      if (b_interior && kfull)
+        StageKMajor<BN>(B, ldb, (int)n0, k0, b_tile(buf), tid);
+      else
+        StageKMajorSafe<BN>(B, ldb, (int)n0, k0, N, K, b_tile(buf), tid);
+    } else {
+      if (b_interior && kfull)
+        StageFast<BN>(B, ldb, (int)n0, k0, b_tile(buf), tid);
+      else
+        StageSafe<BN>(B, ldb, (int)n0, k0, N, K, b_tile(buf), tid);
+    }
+  };
+
   // ---- prologue: stage first tile ----
-  {
-    int64_t kp = (int64_t)kt_begin * BK;
-    bool kfull = kp + BK <= K;
-    if (a_interior && kfull)
-      StageFast<BM>(A, K, (int)m0, kp, a_tile(0), tid);
-    else
-      StageSafe<BM>(A, K, (int)m0, kp, M, K, a_tile(0), tid);
-    if (b_interior && kfull)
-      StageFast<BN>(B, K, (int)n0, kp, b_tile(0), tid);
-    else
-      StageSafe<BN>(B, K, (int)n0, kp, N, K, b_tile(0), tid);
-  }
+  stage(0, (int64_t)kt_begin * BK);
   __syncthreads();
 
   int cur = 0;
   for (int kt = kt_begin; kt < kt_count; ++kt) {
     // issue next tile's loads first (overlap with this tile's compute)
     if (kt + 1 < kt_count) {
-      int64_t k0 = (int64_t)(kt + 1) * BK;
-      bool kfull = k0 + BK <= K;
-      if (a_interior && kfull)
-        StageFast<BM>(A, K, (int)m0, k0, a_tile(cur ^ 1), tid);
-      else
-        StageSafe<BM>(A, K, (int)m0, k0, M, K, a_tile(cur ^ 1), tid);
-      if (b_interior && kfull)
-        StageFast<BN>(B, K, (int)n0, k0, b_tile(cur ^ 1), tid);
-      else
-        StageSafe<BN>(B, K, (int)n0, k0, N, K, b_tile(cur ^ 1), tid);
+      stage(cur ^ 1, (int64_t)(kt + 1) * BK);
     }
 
     // compute on current tile: 2 mfma K-steps of 32
@@ -216,58 +294,122 @@ __launch_bounds__(256) __global__ void GemmBf16NT(
   }
 }
 
-template <bool OUT_BF16, bool FUSE_RELU>
+template <bool A_KM, bool B_KM, bool OUT_BF16, bool FUSE_RELU>
 hipError_t LaunchVariant(const uint16_t* A, const uint16_t* B, void* C,
                          const float* bias, int64_t M, int64_t N, int64_t K,
-                         float beta, hipStream_t stream) {
+                         int64_t lda, int64_t ldb, float beta,
+                         hipStream_t stream) {
   auto launch = [&](auto kern, int BM, int BN) {
     int64_t blocks = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
     hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0, stream, A,
-                       B, C, bias, M, N, K, beta, 1);
+                       B, C, bias, M, N, K, lda, ldb, beta, 1);
   };
   if (N >= 128 && M >= 128) {
-    launch(GemmBf16NT<2, 2, 4, 4, OUT_BF16, FUSE_RELU>, 128, 128);
+    launch(GemmBf16NT<2, 2, 4, 4, A_KM, B_KM, OUT_BF16, FUSE_RELU>, 128, 128);
   } else if (N >= 128) {
-    launch(GemmBf16NT<2, 2, 2, 4, OUT_BF16, FUSE_RELU>, 64, 128);
+    launch(GemmBf16NT<2, 2, 2, 4, A_KM, B_KM, OUT_BF16, FUSE_RELU>, 64, 128);
   } else if (M >= 128) {
-    launch(GemmBf16NT<2, 2, 4, 2, OUT_BF16, FUSE_RELU>, 128, 64);
+    launch(GemmBf16NT<2, 2, 4, 2, A_KM, B_KM, OUT_BF16, FUSE_RELU>, 128, 64);
   } else {
-    launch(GemmBf16NT<2, 2, 2, 2, OUT_BF16, FUSE_RELU>, 64, 64);
+    launch(GemmBf16NT<2, 2, 2, 2, A_KM, B_KM, OUT_BF16, FUSE_RELU>, 64, 64);
   }
   return hipGetLastError();
 }
 
+template <bool A_KM, bool B_KM>
 hipError_t LaunchSplitK(const uint16_t* A, const uint16_t* B, float* C,
-                        int64_t M, int64_t N, int64_t K, int splitk,
-                        hipStream_t stream) {
+                        int64_t M, int64_t N, int64_t K, int64_t lda,
+                        int64_t ldb, int splitk, hipStream_t stream) {
   auto launch = [&](auto kern, int BM, int BN) {
     int64_t blocks = ((M + BM - 1) / BM) * ((N + BN - 1) / BN) * splitk;
     hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0, stream, A,
-                       B, C, nullptr, M, N, K, 0.f, splitk);
+                       B, C, nullptr, M, N, K, lda, ldb, 0.f, splitk);
   };
   if (N >= 128 && M >= 128) {
-    launch(GemmBf16NT<2, 2, 4, 4, false, false, true>, 128, 128);
+    launch(GemmBf16NT<2, 2, 4, 4, A_KM, B_KM, false, false, true>, 128, 128);
   } else if (N >= 128) {
-    launch(GemmBf16NT<2, 2, 2, 4, false, false, true>, 64, 128);
+    launch(GemmBf16NT<2, 2, 2, 4, A_KM, B_KM, false, false, true>, 64, 128);
   } else if (M >= 128) {
-    launch(GemmBf16NT<2, 2, 4, 2, false, false, true>, 128, 64);
+    launch(GemmBf16NT<2, 2, 4, 2, A_KM, B_KM, false, false, true>, 128, 64);
   } else {
-    launch(GemmBf16NT<2, 2, 2, 2, false, false, true>, 64, 64);
+    launch(GemmBf16NT<2, 2, 2, 2, A_KM, B_KM, false, false, true>, 64, 64);
   }
   return hipGetLastError();
 }
 
 }  // namespace
 
-// Split-K variant for skinny-output huge-K GEMMs (dW): partial products
-// atomically accumulated into a zeroed f32 C (guide split-K decomposition:
-// size the grid to ~1-2x the 256 CUs).
+// General entry. a_km/b_km select the contraction-major (transposed-staging)
+// path per operand; lda/ldb are row strides of the stored layouts
+// (a_km ? [K,M] : [M,K], b_km ? [K,N] : [N,K]).
+extern "C" hipError_t stf_gemm_bf16(const void* A, const void* B, void* C,
+                                    const void* bias_f32, int64_t M, int64_t N,
+                                    int64_t K, int64_t lda, int64_t ldb,
+                                    float beta, int a_km, int b_km,
+                                    int out_bf16, int fuse_relu,
+                                    hipStream_t stream) {
+  const uint16_t* a = (const uint16_t*)A;
+  const uint16_t* b = (const uint16_t*)B;
+  const float* bias = (const float*)bias_f32;
+  int sel = (a_km ? 8 : 0) | (b_km ? 4 : 0) | (out_bf16 ? 2 : 0) |
+            (fuse_relu ? 1 : 0);
+  switch (sel) {
+#define STF_CASE(AK, BK_, OB, FR)                                         \
+  case ((AK ? 8 : 0) | (BK_ ? 4 : 0) | (OB ? 2 : 0) | (FR ? 1 : 0)):      \
+    return LaunchVariant<AK, BK_, OB, FR>(a, b, C, bias, M, N, K, lda,    \
+                                          ldb, beta, stream)
+    STF_CASE(false, false, false, false);
+    STF_CASE(false, false, false, true);
+    STF_CASE(false, false, true, false);
+    STF_CASE(false, false, true, true);
+    STF_CASE(false, true, false, false);
+    STF_CASE(false, true, false, true);
+    STF_CASE(false, true, true, false);
+    STF_CASE(false, true, true, true);
+    STF_CASE(true, false, false, false);
+    STF_CASE(true, false, false, true);
+    STF_CASE(true, false, true, false);
+    STF_CASE(true, false, true, true);
+    STF_CASE(true, true, false, false);
+    STF_CASE(true, true, false, true);
+    STF_CASE(true, true, true, false);
+    STF_CASE(true, true, true, true);
+#undef STF_CASE
+  }
+  return hipErrorInvalidValue;
+}
+
+// Split-K for skinny-output huge-K GEMMs (dW): partial products atomically
+// accumulated into a zeroed f32 C (guide split-K decomposition: size the
+// grid to ~1-2x the 256 CUs).
+extern "C" hipError_t stf_gemm_bf16_splitk(const void* A, const void* B,
+                                           void* C_f32, int64_t M, int64_t N,
+                                           int64_t K, int64_t lda, int64_t ldb,
+                                           int a_km, int b_km, int splitk,
+                                           hipStream_t stream) {
+  const uint16_t* a = (const uint16_t*)A;
+  const uint16_t* b = (const uint16_t*)B;
+  float* c = (float*)C_f32;
+  if (a_km && b_km)
+    return LaunchSplitK<true, true>(a, b, c, M, N, K, lda, ldb, splitk,
+                                    stream);
+  if (a_km)
+    return LaunchSplitK<true, false>(a, b, c, M, N, K, lda, ldb, splitk,
+                                     stream);
+  if (b_km)
+    return LaunchSplitK<false, true>(a, b, c, M, N, K, lda, ldb, splitk,
+                                     stream);
+  return LaunchSplitK<false, false>(a, b, c, M, N, K, lda, ldb, splitk,
+                                    stream);
+}
+
+// Legacy NT entries (both operands row-major with contraction innermost).
 extern "C" hipError_t stf_gemm_bf16_nt_splitk(const void* A, const void* B,
                                               void* C_f32, int64_t M,
                                               int64_t N, int64_t K, int splitk,
                                               hipStream_t stream) {
-  return LaunchSplitK((const uint16_t*)A, (const uint16_t*)B, (float*)C_f32,
-                      M, N, K, splitk, stream);
+  return stf_gemm_bf16_splitk(A, B, C_f32, M, N, K, K, K, 0, 0, splitk,
+                              stream);
 }
 
 // C[M,N] = A[M,K] * B[N,K]^T (+ beta*C) (+bias) (+relu)
@@ -276,13 +418,6 @@ extern "C" hipError_t stf_gemm_bf16_nt(const void* A, const void* B, void* C,
                                        int64_t N, int64_t K, float beta,
                                        int out_bf16, int fuse_relu,
                                        hipStream_t stream) {
-  const uint16_t* a = (const uint16_t*)A;
-  const uint16_t* b = (const uint16_t*)B;
-  const float* bias = (const float*)bias_f32;
-  if (out_bf16) {
-    if (fuse_relu) return LaunchVariant<true, true>(a, b, C, bias, M, N, K, beta, stream);
-    return LaunchVariant<true, false>(a, b, C, bias, M, N, K, beta, stream);
-  }
-  if (fuse_relu) return LaunchVariant<false, true>(a, b, C, bias, M, N, K, beta, stream);
-  return LaunchVariant<false, false>(a, b, C, bias, M, N, K, beta, stream);
+  return stf_gemm_bf16(A, B, C, bias_f32, M, N, K, K, K, beta, 0, 0,
+                       out_bf16, fuse_relu, stream);
 }

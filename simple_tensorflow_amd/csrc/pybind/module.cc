@@ -172,8 +172,17 @@ class PySession {
     if (!s.ok()) throw std::runtime_error(s.ToString());
   }
 
+  // Per-node host-side timings from the last Run(collect_stats=true):
+  // list of (node, op, start_us, end_us).
+  py::list LastStats() {
+    py::list out;
+    for (auto& ns : last_stats_)
+      out.append(py::make_tuple(ns.node, ns.op, ns.start_us, ns.end_us));
+    return out;
+  }
+
   py::list Run(py::dict feeds, std::vector<std::string> fetches,
-               std::vector<std::string> targets) {
+               std::vector<std::string> targets, bool collect_stats = false) {
     std::vector<std::pair<std::string, Tensor>> feed_vec;
     for (auto item : feeds) {
       std::string name = py::cast<std::string>(item.first);
@@ -188,11 +197,14 @@ class PySession {
     }
     std::vector<Tensor> outputs;
     Status s;
+    StatsCollector stats;
     {
       py::gil_scoped_release release;
-      s = sess_.Run(feed_vec, fetches, targets, &outputs);
+      s = sess_.Run(feed_vec, fetches, targets, &outputs,
+                    collect_stats ? &stats : nullptr);
     }
     if (!s.ok()) throw std::runtime_error(s.ToString());
+    if (collect_stats) last_stats_ = std::move(stats.stats);
     py::list out;
     for (auto& t : outputs) out.append(TensorToPy(t));
     return out;
@@ -213,6 +225,7 @@ class PySession {
 
  private:
   DirectSession sess_;
+  std::vector<NodeStats> last_stats_;
 };
 
 }  // namespace
@@ -230,7 +243,9 @@ PYBIND11_MODULE(_core, m) {
       .def(py::init<bool>(), py::arg("cpu_only") = false)
       .def("create", &PySession::Create)
       .def("extend", &PySession::Extend)
-      .def("run", &PySession::Run)
+      .def("run", &PySession::Run, py::arg("feeds"), py::arg("fetches"),
+           py::arg("targets"), py::arg("collect_stats") = false)
+      .def("last_stats", &PySession::LastStats)
       .def("sync", &PySession::Sync)
       .def("num_gpus", &PySession::NumGpus);
 

@@ -1,6 +1,7 @@
 #include "runtime/executor.h"
 
 #include <condition_variable>
+#include <chrono>
 #include <deque>
 #include <tuple>
 
@@ -492,7 +493,18 @@ class ExecutorState {
 
     OpKernelContext ctx(item.kernel, dev, std::move(inputs));
     FillCtx(&ctx, t);
+    int64_t t0 = 0;
+    if (args_.stats)
+      t0 = std::chrono::duration_cast<std::chrono::microseconds>(
+               std::chrono::system_clock::now().time_since_epoch())
+               .count();
     dev->Compute(item.kernel, &ctx);
+    if (args_.stats) {
+      int64_t t1 = std::chrono::duration_cast<std::chrono::microseconds>(
+                       std::chrono::system_clock::now().time_since_epoch())
+                       .count();
+      args_.stats->Add(n->name(), n->op(), t0, t1);
+    }
     if (getenv("STF_DEBUG_LAUNCH")) {
       std::string line = n->name() + " (" + n->op() + ") in:";
       char b[32];

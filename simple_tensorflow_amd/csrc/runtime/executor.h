@@ -51,6 +51,26 @@ class OpSegment {
 
 class ResourceMgr;  // defined in kernels (queues etc.)
 
+// Per-node timing record (StepStats; reference step_stats.proto NodeExecStats
+// — host-side enqueue times, the analog of the reference's
+// StepStatsCollector in common_runtime/step_stats_collector.cc).
+struct NodeStats {
+  std::string node;
+  std::string op;
+  int64_t start_us = 0;
+  int64_t end_us = 0;
+};
+
+struct StatsCollector {
+  std::mutex mu;
+  std::vector<NodeStats> stats;
+  void Add(const std::string& node, const std::string& op, int64_t start_us,
+           int64_t end_us) {
+    std::lock_guard<std::mutex> l(mu);
+    stats.push_back({node, op, start_us, end_us});
+  }
+};
+
 struct ExecutorArgs {
   int64_t step_id = 0;
   Rendezvous* rendezvous = nullptr;
@@ -60,6 +80,7 @@ struct ExecutorArgs {
   std::function<void(std::function<void()>)> schedule;
   void* resource_mgr = nullptr;
   std::function<bool()> is_cancelled;
+  StatsCollector* stats = nullptr;  // non-null => collect per-node timings
 };
 
 class Executor {

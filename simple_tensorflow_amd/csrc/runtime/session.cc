@@ -401,7 +401,8 @@ Status DirectSession::BuildExecutors(const std::vector<std::string>& feeds,
 Status DirectSession::Run(
     const std::vector<std::pair<std::string, Tensor>>& feeds,
     const std::vector<std::string>& fetches,
-    const std::vector<std::string>& targets, std::vector<Tensor>* outputs) {
+    const std::vector<std::string>& targets, std::vector<Tensor>* outputs,
+    StatsCollector* stats) {
   std::vector<std::string> feed_names;
   for (auto& f : feeds) feed_names.push_back(f.first);
   ExecutorsAndKeys* ek = nullptr;
@@ -420,7 +421,7 @@ Status DirectSession::Run(
   }
 
   // hipGraph fast path: replay the captured step.
-  bool no_io = feeds.empty() && fetches.empty();
+  bool no_io = feeds.empty() && fetches.empty() && stats == nullptr;
   if (no_io && ek->capture_eligible && ek->graph_exec) {
     outputs->clear();
     return ek->capture_device->LaunchCapturedGraph(ek->graph_exec);
@@ -476,6 +477,7 @@ Status DirectSession::Run(
     args.rendezvous = &rendez;
     args.schedule = scheduler;
     args.resource_mgr = resource_mgr_;
+    args.stats = stats;
     item.executor->RunAsync(args, [&](Status s) {
       std::lock_guard<std::mutex> l(mu);
       if (!s.ok() && agg.ok()) agg = s;

@@ -33,7 +33,7 @@ class DirectSession {
   Status Run(const std::vector<std::pair<std::string, Tensor>>& feeds,
              const std::vector<std::string>& fetches,
              const std::vector<std::string>& targets,
-             std::vector<Tensor>* outputs);
+             std::vector<Tensor>* outputs, StatsCollector* stats = nullptr);
 
   DeviceMgr* device_mgr() { return &devices_; }
   // Blocks until all device work is complete (bench timing bracket).

@@ -320,5 +320,8 @@ class _PythonIoModule(object):
 python_io = _PythonIoModule()
 
 from simple_tensorflow_amd.python import saved_model  # noqa: E402,F401
+from simple_tensorflow_amd.python.client import timeline  # noqa: E402,F401
+RunOptions = timeline.RunOptions
+RunMetadata = timeline.RunMetadata
 
 __version__ = '0.1.0'

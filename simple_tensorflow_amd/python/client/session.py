@@ -76,10 +76,23 @@ class Session(object):
                 t = _as_fetchable(k, self._graph)
                 feeds[t.name] = _convert_feed(t, v)
 
+        want_stats = (options is not None and
+                      getattr(options, 'trace_level', 0) and
+                      run_metadata is not None)
         try:
-            results = self._core.run(feeds, fetch_names, targets)
+            if want_stats:
+                results = self._core.run(feeds, fetch_names, targets, True)
+            else:
+                results = self._core.run(feeds, fetch_names, targets)
         except RuntimeError as e:
             errors.raise_from_message(str(e))
+        except TypeError:
+            # remote (grpc) cores take no collect_stats flag
+            results = self._core.run(feeds, fetch_names, targets)
+            want_stats = False
+        if want_stats:
+            from simple_tensorflow_amd.python.client import timeline as tl
+            run_metadata.step_stats = tl.StepStats(self._core.last_stats())
         out = []
         for f, slot in zip(flat, fetch_slots):
             if slot is None:

@@ -234,6 +234,18 @@ class _NnModule(object):
 
 
 nn = _nn  # module with conv2d/relu/softmax/...
+from simple_tensorflow_amd.python.ops import rnn_cell_impl as _rnn  # noqa: E402
+from simple_tensorflow_amd.python.ops import tensor_array_ops as _ta_ops  # noqa: E402
+TensorArray = _ta_ops.TensorArray
+from simple_tensorflow_amd.python.ops import variable_scope as _vs  # noqa: E402
+variable_scope = _vs.variable_scope
+get_variable = _vs.get_variable
+get_variable_scope = _vs.get_variable_scope
+VariableScope = _vs.VariableScope
+nn.rnn_cell = _rnn
+nn.dynamic_rnn = _rnn.dynamic_rnn
+nn.static_rnn = _rnn.static_rnn
+nn.rnn = _rnn.static_rnn
 
 
 class _TrainModule(object):

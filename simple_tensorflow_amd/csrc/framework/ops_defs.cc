@@ -216,6 +216,17 @@ REGISTER_OP("RestoreV2").Input("prefix: string").Input("tensor_names: string").I
 REGISTER_OP("MergeV2Checkpoints").Input("checkpoint_prefixes: string").Input("destination_prefix: string").Attr("delete_old_dirs: bool = true").SetIsStateful();
 
 // ------------------------------- queues ------------------------------------
+// TensorArray family (reference core/ops/data_flow_ops.cc:1080; resource
+// handles are session-scoped strings here, like the queue ops above).
+REGISTER_OP("TensorArrayV3").Input("size: int32").Output("handle: string").Output("flow: float").Attr("dtype: type").Attr("dynamic_size: bool = false").Attr("clear_after_read: bool = true").Attr("tensor_array_name: string = ''").SetIsStateful();
+REGISTER_OP("TensorArrayGradV3").Input("handle: string").Input("flow_in: float").Output("grad_handle: string").Output("flow_out: float").Attr("source: string").SetIsStateful();
+REGISTER_OP("TensorArrayWriteV3").Input("handle: string").Input("index: int32").Input("value: T").Input("flow_in: float").Output("flow_out: float").Attr("T: type").SetIsStateful();
+REGISTER_OP("TensorArrayReadV3").Input("handle: string").Input("index: int32").Input("flow_in: float").Output("value: dtype").Attr("dtype: type").SetIsStateful();
+REGISTER_OP("TensorArraySizeV3").Input("handle: string").Input("flow_in: float").Output("size: int32").SetIsStateful();
+REGISTER_OP("TensorArrayGatherV3").Input("handle: string").Input("indices: int32").Input("flow_in: float").Output("value: dtype").Attr("dtype: type").SetIsStateful();
+REGISTER_OP("TensorArrayScatterV3").Input("handle: string").Input("indices: int32").Input("value: T").Input("flow_in: float").Output("flow_out: float").Attr("T: type").SetIsStateful();
+REGISTER_OP("TensorArrayCloseV3").Input("handle: string").SetIsStateful();
+
 REGISTER_OP("FIFOQueue").Output("handle: Ref(string)").Attr("component_types: list(type)").Attr("shapes: list(shape) = []").Attr("capacity: int = -1").Attr("container: string = ''").Attr("shared_name: string = ''").SetIsStateful();
 REGISTER_OP("RandomShuffleQueue").Output("handle: Ref(string)").Attr("component_types: list(type)").Attr("shapes: list(shape) = []").Attr("capacity: int = -1").Attr("min_after_dequeue: int = 0").Attr("seed: int = 0").Attr("seed2: int = 0").Attr("container: string = ''").Attr("shared_name: string = ''").SetIsStateful();
 REGISTER_OP("PaddingFIFOQueue").Output("handle: Ref(string)").Attr("component_types: list(type)").Attr("shapes: list(shape) = []").Attr("capacity: int = -1").Attr("container: string = ''").Attr("shared_name: string = ''").SetIsStateful();

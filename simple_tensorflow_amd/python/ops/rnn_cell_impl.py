@@ -139,3 +139,62 @@ def static_rnn(cell, inputs, initial_state=None, dtype=None, scope=None):
             out, state = cell(x, state)
             outputs.append(out)
     return outputs, state
+
+
+def dynamic_rnn(cell, inputs, sequence_length=None, initial_state=None,
+                dtype=None, time_major=False, scope=None,
+                parallel_iterations=None, swap_memory=False):
+    """while_loop + TensorArray RNN (reference python/ops/rnn.py
+    dynamic_rnn; forward-only in round 1 — training uses static_rnn)."""
+    from simple_tensorflow_amd.python.ops import control_flow_ops, math_ops
+    from simple_tensorflow_amd.python.ops import tensor_array_ops
+    from simple_tensorflow_amd.python.util import nest
+    from simple_tensorflow_amd.python.framework import dtypes as _dt
+
+    if not time_major:
+        inputs = array_ops.transpose(inputs, [1, 0, 2])
+    in_shape = inputs._shape
+    T = in_shape[0] if in_shape is not None else None
+    batch = in_shape[1] if in_shape is not None else None
+    if T is None:
+        raise ValueError('dynamic_rnn requires a static time dimension '
+                         'in round 1')
+    state = initial_state
+    if state is None:
+        state = cell.zero_state(batch, dtype or inputs.dtype)
+    flat_state = nest.flatten(state)
+
+    input_ta = tensor_array_ops.TensorArray(inputs.dtype, size=T)
+    input_ta = input_ta.unstack(inputs)
+    output_ta = tensor_array_ops.TensorArray(
+        dtype or inputs.dtype, size=T)
+
+    g = ops.get_default_graph()
+    with variable_scope.variable_scope(scope or 'rnn') as vs:
+        built = [False]
+
+        def body(t, out_flow, *fs):
+            st = nest.pack_sequence_as(state, list(fs))
+            x = input_ta.read(t)
+            if in_shape is not None:
+                x.set_shape([in_shape[1], in_shape[2]])
+            if built[0]:
+                vs.reuse = True
+            out, new_st = cell(x, st)
+            built[0] = True
+            ta = output_ta._with_flow(out_flow)
+            new_flow = ta.write(t, out)._flow
+            return [math_ops.add(t, 1), new_flow] + nest.flatten(new_st)
+
+        results = control_flow_ops.while_loop(
+            lambda t, f, *fs: math_ops.less(t, T), body,
+            [ops.constant(0, dtype=_dt.int32), output_ta._flow] + flat_state)
+    final_flow = results[1]
+    final_state = nest.pack_sequence_as(state, list(results[2:]))
+    outputs = output_ta._with_flow(final_flow).stack()
+    osh = None
+    if in_shape is not None and getattr(cell, 'output_size', None):
+        outputs.set_shape([T, in_shape[1], cell.output_size])
+    if not time_major:
+        outputs = array_ops.transpose(outputs, [1, 0, 2])
+    return outputs, final_state

@@ -218,6 +218,8 @@ REGISTER_OP("MergeV2Checkpoints").Input("checkpoint_prefixes: string").Input("de
 // ------------------------------- queues ------------------------------------
 // TensorArray family (reference core/ops/data_flow_ops.cc:1080; resource
 // handles are session-scoped strings here, like the queue ops above).
+REGISTER_OP("PyFunc").Input("input: Tin").Output("output: Tout").Attr("token: string").Attr("Tin: list(type) >= 0").Attr("Tout: list(type) >= 0").SetIsStateful();
+REGISTER_OP("PyFuncStateless").Input("input: Tin").Output("output: Tout").Attr("token: string").Attr("Tin: list(type) >= 0").Attr("Tout: list(type) >= 0");
 REGISTER_OP("TensorArrayV3").Input("size: int32").Output("handle: string").Output("flow: float").Attr("dtype: type").Attr("dynamic_size: bool = false").Attr("clear_after_read: bool = true").Attr("tensor_array_name: string = ''").SetIsStateful();
 REGISTER_OP("TensorArrayGradV3").Input("handle: string").Input("flow_in: float").Output("grad_handle: string").Output("flow_out: float").Attr("source: string").SetIsStateful();
 REGISTER_OP("TensorArrayWriteV3").Input("handle: string").Input("index: int32").Input("value: T").Input("flow_in: float").Output("flow_out: float").Attr("T: type").SetIsStateful();

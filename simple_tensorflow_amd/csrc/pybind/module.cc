@@ -5,6 +5,9 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include <map>
+#include <mutex>
+
 #include "core/protos.h"
 #include "framework/op.h"
 #include "runtime/session.h"
@@ -151,6 +154,75 @@ py::dict AttrValueToPy(const AttrValue& v) {
   return d;
 }
 
+// ---- py_func bridge (reference python/lib/core/py_func.cc + kernels
+// script_ops): python callables registered by token, invoked from the
+// executor with the GIL re-acquired. ----
+std::mutex g_pyfunc_mu;
+std::map<std::string, py::function>& PyFuncRegistry() {
+  static auto* m = new std::map<std::string, py::function>();
+  return *m;
+}
+
+class PyFuncOp : public OpKernel {
+ public:
+  explicit PyFuncOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("token", &token_);
+    auto it = c->def().attr.find("Tout");
+    if (it != c->def().attr.end())
+      for (int t : it->second.list.type) tout_.push_back((DataType)t);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    py::gil_scoped_acquire gil;
+    py::function fn;
+    {
+      std::lock_guard<std::mutex> l(g_pyfunc_mu);
+      auto it = PyFuncRegistry().find(token_);
+      if (it == PyFuncRegistry().end()) {
+        ctx->SetStatus(errors::NotFound("py_func token ", token_));
+        return;
+      }
+      fn = it->second;
+    }
+    py::list args;
+    for (int i = 0; i < ctx->num_inputs(); ++i)
+      args.append(TensorToPy(ctx->input(i)));
+    py::object result;
+    try {
+      result = fn(*py::tuple(args));
+    } catch (py::error_already_set& e) {
+      ctx->SetStatus(errors::Internal("py_func raised: ", e.what()));
+      return;
+    }
+    py::list outs;
+    if (py::isinstance<py::tuple>(result) || py::isinstance<py::list>(result))
+      outs = py::list(result);
+    else if (!result.is_none())
+      outs.append(result);
+    if ((size_t)outs.size() != tout_.size()) {
+      ctx->SetStatus(errors::InvalidArgument(
+          "py_func returned ", (int)outs.size(), " values, expected ",
+          (int)tout_.size()));
+      return;
+    }
+    for (size_t i = 0; i < tout_.size(); ++i) {
+      py::array arr = py::array::ensure(outs[i]);
+      Tensor t = NumpyToTensor(arr);
+      if (t.dtype() != tout_[i]) {
+        ctx->SetStatus(errors::InvalidArgument(
+            "py_func output ", (int)i, " dtype mismatch"));
+        return;
+      }
+      ctx->set_output((int)i, t);
+    }
+  }
+
+ private:
+  std::string token_;
+  std::vector<DataType> tout_;
+};
+REGISTER_KERNEL_BUILDER(Name("PyFunc").Device(DEVICE_CPU), PyFuncOp);
+REGISTER_KERNEL_BUILDER(Name("PyFuncStateless").Device(DEVICE_CPU), PyFuncOp);
+
 class PySession {
  public:
   explicit PySession(bool cpu_only) : sess_(cpu_only) {}
@@ -238,6 +310,14 @@ Status RcclInit(int nranks, int rank, const std::string& id_bytes);
 
 PYBIND11_MODULE(_core, m) {
   m.doc() = "simple_tensorflow_amd core runtime (MI355X-native)";
+  m.def("register_py_func", [](const std::string& token, py::function fn) {
+    std::lock_guard<std::mutex> l(g_pyfunc_mu);
+    PyFuncRegistry()[token] = std::move(fn);
+  });
+  m.def("remove_py_func", [](const std::string& token) {
+    std::lock_guard<std::mutex> l(g_pyfunc_mu);
+    PyFuncRegistry().erase(token);
+  });
 
   py::class_<PySession>(m, "Session")
       .def(py::init<bool>(), py::arg("cpu_only") = false)

@@ -1,4 +1,5 @@
 """Array ops (analog of reference python/ops/array_ops.py)."""
+import builtins as _bi
 import numpy as np
 
 from simple_tensorflow_amd.python.framework import dtypes, ops
@@ -260,7 +261,7 @@ def _slice_helper(tensor, key):
             squeeze_axes.append(i)
         elif isinstance(k, type(Ellipsis)):
             raise NotImplementedError('ellipsis slicing')
-        elif isinstance(k, slice):
+        elif isinstance(k, _bi.slice):
             if k.step not in (None, 1):
                 raise NotImplementedError('strided slicing')
             b = k.start or 0

@@ -26,6 +26,7 @@ from simple_tensorflow_amd.python.ops import (  # noqa: F401
     variables as _variables,
 )
 from simple_tensorflow_amd.python.training import optimizer as _optimizer
+from simple_tensorflow_amd.python.training import sync_replicas_optimizer as _sync_opt
 from simple_tensorflow_amd.python.training import training_util as _training_util
 from simple_tensorflow_amd.python.training import saver as _saver
 from simple_tensorflow_amd.python.training import coordinator as _coord
@@ -264,6 +265,9 @@ class _TrainModule(object):
     polynomial_decay = staticmethod(_training_util.polynomial_decay)
     piecewise_constant = staticmethod(_training_util.piecewise_constant)
     ExponentialMovingAverage = _training_util.ExponentialMovingAverage
+
+    SyncReplicasOptimizer = _sync_opt.SyncReplicasOptimizer
+
     Saver = _saver.Saver
 
     @staticmethod

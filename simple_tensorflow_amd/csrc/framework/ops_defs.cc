@@ -218,6 +218,11 @@ REGISTER_OP("MergeV2Checkpoints").Input("checkpoint_prefixes: string").Input("de
 // ------------------------------- queues ------------------------------------
 // TensorArray family (reference core/ops/data_flow_ops.cc:1080; resource
 // handles are session-scoped strings here, like the queue ops above).
+REGISTER_OP("ConditionalAccumulator").Output("handle: Ref(string)").Attr("dtype: type").Attr("shape: shape").Attr("container: string = ''").Attr("shared_name: string = ''").SetIsStateful();
+REGISTER_OP("AccumulatorApplyGradient").Input("handle: Ref(string)").Input("local_step: int64").Input("gradient: dtype").Attr("dtype: type").SetIsStateful();
+REGISTER_OP("AccumulatorTakeGradient").Input("handle: Ref(string)").Input("num_required: int32").Output("average: dtype").Attr("dtype: type").SetIsStateful();
+REGISTER_OP("AccumulatorSetGlobalStep").Input("handle: Ref(string)").Input("new_global_step: int64").SetIsStateful();
+REGISTER_OP("AccumulatorNumAccumulated").Input("handle: Ref(string)").Output("num_accumulated: int32").SetIsStateful();
 REGISTER_OP("PyFunc").Input("input: Tin").Output("output: Tout").Attr("token: string").Attr("Tin: list(type) >= 0").Attr("Tout: list(type) >= 0").SetIsStateful();
 REGISTER_OP("PyFuncStateless").Input("input: Tin").Output("output: Tout").Attr("token: string").Attr("Tin: list(type) >= 0").Attr("Tout: list(type) >= 0");
 REGISTER_OP("TensorArrayV3").Input("size: int32").Output("handle: string").Output("flow: float").Attr("dtype: type").Attr("dynamic_size: bool = false").Attr("clear_after_read: bool = true").Attr("tensor_array_name: string = ''").SetIsStateful();

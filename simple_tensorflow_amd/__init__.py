@@ -281,6 +281,12 @@ class _TrainModule(object):
         return meta_graph.import_meta_graph(meta_graph_or_file, **kw)
 
     latest_checkpoint = staticmethod(_saver.latest_checkpoint)
+
+    @staticmethod
+    def NewCheckpointReader(prefix):
+        from simple_tensorflow_amd.python.tools import inspect_checkpoint
+        return inspect_checkpoint.NewCheckpointReader(prefix)
+
     get_checkpoint_state = staticmethod(_saver.get_checkpoint_state)
     update_checkpoint_state = staticmethod(_saver.update_checkpoint_state)
     checkpoint_exists = staticmethod(_saver.checkpoint_exists)
@@ -336,6 +342,7 @@ class _PythonIoModule(object):
 python_io = _PythonIoModule()
 
 from simple_tensorflow_amd.python import saved_model  # noqa: E402,F401
+from simple_tensorflow_amd.python import estimator  # noqa: E402,F401
 from simple_tensorflow_amd.python.client import timeline  # noqa: E402,F401
 RunOptions = timeline.RunOptions
 RunMetadata = timeline.RunMetadata

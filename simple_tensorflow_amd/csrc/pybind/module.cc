@@ -8,7 +8,11 @@
 #include <map>
 #include <mutex>
 
+#include <fstream>
+
+#include "core/pb.h"
 #include "core/protos.h"
+#include "util/table.h"
 #include "framework/op.h"
 #include "runtime/session.h"
 
@@ -310,6 +314,50 @@ Status RcclInit(int nranks, int rank, const std::string& id_bytes);
 
 PYBIND11_MODULE(_core, m) {
   m.doc() = "simple_tensorflow_amd core runtime (MI355X-native)";
+  // checkpoint introspection (reference python/tools/inspect_checkpoint.py
+  // via checkpoint_reader.cc): list (name, dtype_enum, shape) entries of a
+  // V2 bundle index.
+  m.def("list_checkpoint", [](const std::string& prefix) {
+    std::ifstream f(prefix + ".index", std::ios::binary);
+    if (!f) throw std::runtime_error("checkpoint index not found: " + prefix);
+    std::string data((std::istreambuf_iterator<char>(f)),
+                     std::istreambuf_iterator<char>());
+    std::map<std::string, std::string> index;
+    Status s = table::ReadTable(data, &index);
+    if (!s.ok()) throw std::runtime_error(s.ToString());
+    py::list out;
+    for (auto& kv : index) {
+      if (kv.first.empty()) continue;  // header entry
+      pb::Reader r(kv.second);
+      int field, wire;
+      int dtype = 0;
+      std::vector<int64_t> shape;
+      bool ok_entry = true;
+      while (r.ReadTag(&field, &wire)) {
+        if (field == 1) {
+          uint64_t v;
+          if (!r.ReadVarint(&v)) { ok_entry = false; break; }
+          dtype = (int)v;
+        } else if (field == 2) {
+          const char* d;
+          size_t l;
+          if (!r.ReadView(&d, &l)) { ok_entry = false; break; }
+          pb::Reader sub(d, l);
+          TensorShapeProto p;
+          if (p.Parse(&sub)) {
+            TensorShape ts = p.AsShape();
+            for (auto dim : ts.dim_sizes()) shape.push_back(dim);
+          }
+        } else if (!r.SkipField(wire)) {
+          ok_entry = false;
+          break;
+        }
+      }
+      if (ok_entry && dtype != 0)
+        out.append(py::make_tuple(kv.first, dtype, shape));
+    }
+    return out;
+  });
   m.def("register_py_func", [](const std::string& token, py::function fn) {
     std::lock_guard<std::mutex> l(g_pyfunc_mu);
     PyFuncRegistry()[token] = std::move(fn);

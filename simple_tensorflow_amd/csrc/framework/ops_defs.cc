@@ -150,8 +150,8 @@ REGISTER_OP("FusedBatchNorm").Input("x: T").Input("scale: T").Input("offset: T")
 REGISTER_OP("FusedBatchNormGrad").Input("y_backprop: T").Input("x: T").Input("scale: T").Input("reserve_space_1: T").Input("reserve_space_2: T").Output("x_backprop: T").Output("scale_backprop: T").Output("offset_backprop: T").Output("reserve_space_3: T").Output("reserve_space_4: T").Attr("T: {float}").Attr("epsilon: float = 0.0001").Attr("data_format: string = 'NHWC'").Attr("is_training: bool = true");
 // MI355X-native fused batch norm: bf16 activations with f32 statistics and
 // f32 scale/offset (the bf16 analog of FusedBatchNorm; reserve = 1/sqrt(var+eps)).
-REGISTER_OP("BatchNormMi").Input("x: T").Input("scale: float").Input("offset: float").Output("y: T").Output("batch_mean: float").Output("batch_variance: float").Output("reserve_inv_std: float").Attr("T: {float, bfloat16}").Attr("epsilon: float = 0.0001");
-REGISTER_OP("BatchNormMiGrad").Input("y_backprop: T").Input("x: T").Input("scale: float").Input("saved_mean: float").Input("saved_inv_std: float").Output("x_backprop: T").Output("scale_backprop: float").Output("offset_backprop: float").Attr("T: {float, bfloat16}").Attr("epsilon: float = 0.0001");
+REGISTER_OP("BatchNormMi").Input("x: T").Input("scale: float").Input("offset: float").Output("y: T").Output("batch_mean: float").Output("batch_variance: float").Output("reserve_inv_std: float").Attr("T: {float, bfloat16}").Attr("epsilon: float = 0.0001").Attr("fuse_relu: bool = false");
+REGISTER_OP("BatchNormMiGrad").Input("y_backprop: T").Input("x: T").Input("scale: float").Input("saved_mean: float").Input("saved_inv_std: float").Input("y_relu: T").Output("x_backprop: T").Output("scale_backprop: float").Output("offset_backprop: float").Attr("T: {float, bfloat16}").Attr("epsilon: float = 0.0001").Attr("fuse_relu: bool = false");
 REGISTER_OP("L2Loss").Input("t: T").Output("output: T").Attr("T: " FLOATTYPES);
 REGISTER_OP("LRN").Input("input: T").Output("output: T").Attr("depth_radius: int = 5").Attr("bias: float = 1.0").Attr("alpha: float = 1.0").Attr("beta: float = 0.5").Attr("T: {float}");
 REGISTER_OP("InTopK").Input("predictions: float").Input("targets: T").Output("precision: bool").Attr("k: int").Attr("T: {int32, int64} = int32");

@@ -54,9 +54,9 @@ hipError_t stf_xent(int, const void*, const void*, void*, void*, int64_t, int,
 hipError_t stf_bn_fwd(int, const void*, const void*, const void*, float*,
                       float*, float*, float*, void*, int64_t, int, float, int,
                       hipStream_t);
-hipError_t stf_bn_bwd(int, const void*, const void*, const float*,
-                      const float*, const void*, float*, void*, int64_t, int,
-                      hipStream_t);
+hipError_t stf_bn_bwd(int, const void*, const void*, const void*,
+                      const float*, const float*, const void*, float*,
+                      float*, void*, int64_t, int, int, hipStream_t);
 hipError_t stf_pool_fwd(int, int, const void*, void*, int, int, int, int, int,
                         int, int, int, int, int, int, int, hipStream_t);
 hipError_t stf_max_pool_bwd(int, const void*, const void*, float*, int, int,
@@ -788,6 +788,7 @@ class GpuBatchNormMiOp : public OpKernel {
  public:
   explicit GpuBatchNormMiOp(OpKernelConstruction* c) : OpKernel(c) {
     c->GetAttr("epsilon", &eps_);
+    c->GetAttr("fuse_relu", &fuse_relu_);
   }
   void Compute(OpKernelContext* ctx) override {
     const Tensor& x = ctx->input(0);
@@ -806,42 +807,49 @@ class GpuBatchNormMiOp : public OpKernel {
                               scale.raw_data(), offset.raw_data(),
                               acc.flat<float>(), mean->flat<float>(),
                               var->flat<float>(), inv_std->flat<float>(),
-                              y->raw_data(), rows, c, eps_, 0, s));
+                              y->raw_data(), rows, c, eps_,
+                              fuse_relu_ ? 1 : 0, s));
   }
 
  private:
   float eps_ = 1e-4f;
+  bool fuse_relu_ = false;
 };
 REGISTER_KERNEL_BUILDER(Name("BatchNormMi").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuBatchNormMiOp);
 REGISTER_KERNEL_BUILDER(Name("BatchNormMi").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuBatchNormMiOp);
 
 class GpuBatchNormMiGradOp : public OpKernel {
  public:
-  using OpKernel::OpKernel;
+  explicit GpuBatchNormMiGradOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("fuse_relu", &fuse_relu_);
+  }
   void Compute(OpKernelContext* ctx) override {
     const Tensor& dy = ctx->input(0);
     const Tensor& x = ctx->input(1);
     const Tensor& scale = ctx->input(2);
     const Tensor& mean = ctx->input(3);
     const Tensor& inv_std = ctx->input(4);
+    const Tensor& y_relu = ctx->input(5);
     int c = (int)x.dim_size(x.dims() - 1);
     int64_t rows = x.NumElements() / c;
     Tensor* dx = ctx->allocate_output(0, x.shape());
     Tensor* dscale = ctx->allocate_output(1, TensorShape({c}));
     Tensor* doffset = ctx->allocate_output(2, TensorShape({c}));
     hipStream_t s = GPU_STREAM(ctx);
-    Tensor acc = ctx->allocate_temp(DT_FLOAT, TensorShape({2 * c}));
-    OP_HIP_OK(ctx, ZeroF32(acc.raw_data(), 2 * c, s));
+    // stats accumulate straight into the gradient outputs (sum_dy ==
+    // doffset, sum_dy_xhat == dscale) — no scratch, no d2d copies.
+    OP_HIP_OK(ctx, ZeroF32(doffset->raw_data(), c, s));
+    OP_HIP_OK(ctx, ZeroF32(dscale->raw_data(), c, s));
     OP_HIP_OK(ctx, stf_bn_bwd(DtypeCode(x.dtype()), dy.raw_data(),
-                              x.raw_data(), mean.flat<float>(),
-                              inv_std.flat<float>(), scale.raw_data(),
-                              acc.flat<float>(), dx->raw_data(), rows, c, s));
-    // doffset = acc[0:c] (sum_dy); dscale = acc[c:2c] (sum_dy_xhat)
-    OP_HIP_OK(ctx, hipMemcpyAsync(doffset->raw_data(), acc.flat<float>(),
-                                  c * 4, hipMemcpyDeviceToDevice, s));
-    OP_HIP_OK(ctx, hipMemcpyAsync(dscale->raw_data(), acc.flat<float>() + c,
-                                  c * 4, hipMemcpyDeviceToDevice, s));
+                              x.raw_data(), y_relu.raw_data(),
+                              mean.flat<float>(), inv_std.flat<float>(),
+                              scale.raw_data(), doffset->flat<float>(),
+                              dscale->flat<float>(), dx->raw_data(), rows, c,
+                              fuse_relu_ ? 1 : 0, s));
   }
+
+ private:
+  bool fuse_relu_ = false;
 };
 REGISTER_KERNEL_BUILDER(Name("BatchNormMiGrad").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuBatchNormMiGradOp);
 REGISTER_KERNEL_BUILDER(Name("BatchNormMiGrad").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuBatchNormMiGradOp);

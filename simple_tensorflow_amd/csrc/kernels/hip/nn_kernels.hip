@@ -248,8 +248,8 @@ __global__ void BnStatsKernel(const T* __restrict__ x, float* __restrict__ acc,
   }
   __syncthreads();
   for (int i = threadIdx.x; i < c; i += blockDim.x) {
-    atomicAdd(&acc[i], s1[i]);
-    atomicAdd(&acc[c + i], s2[i]);
+    atomicAdd(&sum_dy[i], s1[i]);
+    atomicAdd(&sum_dy_xhat[i], s2[i]);
   }
 }
 
@@ -286,14 +286,18 @@ __global__ void BnNormKernel(const T* __restrict__ x,
   }
 }
 
-// bwd pass 1: per-channel sum(dy), sum(dy * xhat) into acc[2C] (zeroed)
-template <typename T>
+// bwd pass 1: per-channel sum(dy), sum(dy * xhat) into acc[2C] (zeroed).
+// RELU: the forward was BN+ReLU fused; dy is masked by y>0 on the fly so no
+// separate relu-grad elementwise pass (or its memory traffic) is needed.
+template <typename T, bool RELU>
 __global__ void BnGradStatsKernel(const T* __restrict__ dy,
                                   const T* __restrict__ x,
+                                  const T* __restrict__ yr,
                                   const float* __restrict__ mean,
                                   const float* __restrict__ inv_std,
-                                  float* __restrict__ acc, int64_t rows,
-                                  int c) {
+                                  float* __restrict__ sum_dy,
+                                  float* __restrict__ sum_dy_xhat,
+                                  int64_t rows, int c) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* s1 = (float*)smem;
   float* s2 = s1 + c;
@@ -312,9 +316,10 @@ __global__ void BnGradStatsKernel(const T* __restrict__ dy,
     int cb0 = -1;
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
          i += gstride) {
-      T g[8], xv[8];
+      T g[8], xv[8], yv[8];
       *(ulong2*)g = *(const ulong2*)(dy + i * 8);
       *(ulong2*)xv = *(const ulong2*)(x + i * 8);
+      if (RELU) *(ulong2*)yv = *(const ulong2*)(yr + i * 8);
       int cb = (int)((i * 8) % c);
       if (cb != cb0) {
         if (cb0 >= 0) {
@@ -336,6 +341,7 @@ __global__ void BnGradStatsKernel(const T* __restrict__ dy,
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         float gf = (float)g[e];
+        if (RELU && (float)yv[e] <= 0.f) gf = 0.f;
         float xhat = ((float)xv[e] - mloc[e]) * iloc[e];
         p1[e] += gf;
         p2[e] += gf * xhat;
@@ -353,6 +359,7 @@ __global__ void BnGradStatsKernel(const T* __restrict__ dy,
          i += gstride) {
       int ch = (int)(i % c);
       float gf = (float)dy[i];
+      if (RELU && (float)yr[i] <= 0.f) gf = 0.f;
       float xhat = ((float)x[i] - mean[ch]) * inv_std[ch];
       atomicAdd(&s1[ch], gf);
       atomicAdd(&s2[ch], gf * xhat);
@@ -360,18 +367,20 @@ __global__ void BnGradStatsKernel(const T* __restrict__ dy,
   }
   __syncthreads();
   for (int i = threadIdx.x; i < c; i += blockDim.x) {
-    atomicAdd(&acc[i], s1[i]);
-    atomicAdd(&acc[c + i], s2[i]);
+    atomicAdd(&sum_dy[i], s1[i]);
+    atomicAdd(&sum_dy_xhat[i], s2[i]);
   }
 }
 
 // bwd pass 2: dx = scale*inv_std*(dy - sum_dy/rows - xhat*sum_dy_xhat/rows)
-template <typename T>
+template <typename T, bool RELU>
 __global__ void BnGradKernel(const T* __restrict__ dy, const T* __restrict__ x,
+                             const T* __restrict__ yr,
                              const float* __restrict__ mean,
                              const float* __restrict__ inv_std,
                              const float* __restrict__ scale,
-                             const float* __restrict__ acc,
+                             const float* __restrict__ sum_dy,
+                             const float* __restrict__ sum_dy_xhat,
                              T* __restrict__ dx, int64_t n, int64_t rows,
                              int c) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -380,8 +389,10 @@ __global__ void BnGradKernel(const T* __restrict__ dy, const T* __restrict__ x,
     int ch = (int)(i % c);
     float xhat = ((float)x[i] - mean[ch]) * inv_std[ch];
     float g = (float)dy[i];
+    if (RELU && (float)yr[i] <= 0.f) g = 0.f;
     dx[i] = (T)(scale[ch] * inv_std[ch] *
-                (g - acc[ch] * inv_rows - xhat * acc[c + ch] * inv_rows));
+                (g - sum_dy[ch] * inv_rows -
+                 xhat * sum_dy_xhat[ch] * inv_rows));
   }
 }
 
@@ -604,29 +615,32 @@ hipError_t stf_bn_fwd(int dtype, const void* x, const void* scale,
 // acc zeroed [2C]; outputs dscale=acc[c..], doffset=acc[0..] are copied out by
 // the wrapper after the kernel (acc[0:c]=sum_dy=doffset, acc[c:2c]=sum_dy_xhat=dscale)
 hipError_t stf_bn_bwd(int dtype, const void* dy, const void* x,
-                      const float* mean, const float* inv_std,
-                      const void* scale, float* acc, void* dx, int64_t rows,
-                      int c, hipStream_t stream) {
+                      const void* y_relu, const float* mean,
+                      const float* inv_std, const void* scale,
+                      float* sum_dy, float* sum_dy_xhat, void* dx,
+                      int64_t rows, int c, int fuse_relu,
+                      hipStream_t stream) {
   int blocks = 512;
   size_t lds = (size_t)c * 8;
-  if (dtype == 0)
-    hipLaunchKernelGGL((BnGradStatsKernel<float>), dim3(blocks), dim3(256),
-                       lds, stream, (const float*)dy, (const float*)x, mean,
-                       inv_std, acc, rows, c);
-  else
-    hipLaunchKernelGGL((BnGradStatsKernel<__bf16>), dim3(blocks), dim3(256),
-                       lds, stream, (const __bf16*)dy, (const __bf16*)x, mean,
-                       inv_std, acc, rows, c);
   int64_t n = rows * c;
   dim3 grid = ElemwiseGrid(n, 256, 4);
-  if (dtype == 0)
-    hipLaunchKernelGGL((BnGradKernel<float>), grid, dim3(256), 0, stream,
-                       (const float*)dy, (const float*)x, mean, inv_std,
-                       (const float*)scale, acc, (float*)dx, n, rows, c);
-  else
-    hipLaunchKernelGGL((BnGradKernel<__bf16>), grid, dim3(256), 0, stream,
-                       (const __bf16*)dy, (const __bf16*)x, mean, inv_std,
-                       (const float*)scale, acc, (__bf16*)dx, n, rows, c);
+#define BNB(T, R)                                                           \
+  do {                                                                      \
+    hipLaunchKernelGGL((BnGradStatsKernel<T, R>), dim3(blocks), dim3(256),  \
+                       lds, stream, (const T*)dy, (const T*)x,              \
+                       (const T*)y_relu, mean, inv_std, sum_dy,             \
+                       sum_dy_xhat, rows, c);                               \
+    hipLaunchKernelGGL((BnGradKernel<T, R>), grid, dim3(256), 0, stream,    \
+                       (const T*)dy, (const T*)x, (const T*)y_relu, mean,   \
+                       inv_std, (const float*)scale, sum_dy, sum_dy_xhat,   \
+                       (T*)dx, n, rows, c);                                 \
+  } while (0)
+  if (dtype == 0) {
+    if (fuse_relu) BNB(float, true); else BNB(float, false);
+  } else {
+    if (fuse_relu) BNB(__bf16, true); else BNB(__bf16, false);
+  }
+#undef BNB
   return hipGetLastError();
 }
 

@@ -26,11 +26,11 @@ def batch_norm(x, c, name, relu=True):
                                name=name + '_scale')
     offset = variables.Variable(array_ops.zeros([c], tf.float32),
                                 name=name + '_offset')
+    # BN+ReLU fused in one kernel pass (fwd relu in BnNormKernel, bwd mask
+    # in BnGrad*) — saves a full-tensor elementwise pass each way.
     y, _, _, _ = apply_op('BatchNormMi', x, scale.ref(), offset.ref(),
-                          epsilon=1e-4, name=name)
+                          epsilon=1e-4, fuse_relu=relu, name=name)
     y.set_shape(x._shape)
-    if relu:
-        y = tf.nn.relu(y)
     return y
 
 

@@ -248,8 +248,8 @@ __global__ void BnStatsKernel(const T* __restrict__ x, float* __restrict__ acc,
   }
   __syncthreads();
   for (int i = threadIdx.x; i < c; i += blockDim.x) {
-    atomicAdd(&sum_dy[i], s1[i]);
-    atomicAdd(&sum_dy_xhat[i], s2[i]);
+    atomicAdd(&acc[i], s1[i]);
+    atomicAdd(&acc[c + i], s2[i]);
   }
 }
 

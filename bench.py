@@ -27,9 +27,15 @@ def main():
     ap.add_argument('--gpus', type=int, default=1)
     ap.add_argument('--steps', type=int, default=20)
     ap.add_argument('--warmup', type=int, default=5)
-    ap.add_argument('--batch', type=int, default=256)
+    ap.add_argument('--batch', type=int, default=0)  # 0 = model default
     ap.add_argument('--lr', type=float, default=0.1)
+    ap.add_argument('--model', default='resnet50',
+                    choices=['resnet50', 'inception3', 'ptb_lstm'])
+    ap.add_argument('--seq-len', type=int, default=35)
     args = ap.parse_args()
+    if args.batch == 0:
+        args.batch = {'resnet50': 256, 'inception3': 128,
+                      'ptb_lstm': 20}[args.model]
 
     import simple_tensorflow_amd as tf
     from simple_tensorflow_amd.models import resnet
@@ -69,25 +75,35 @@ def main():
         elapsed = comm.max_scalar(elapsed)
 
     ms_per_step = elapsed / args.steps * 1000.0
-    imgs_per_sec = args.batch * world * args.steps / elapsed
+    if args.model == 'ptb_lstm':
+        value = args.batch * args.seq_len * world * args.steps / elapsed
+        metric, unit = 'words/sec', 'words/sec'
+        model_name, seq_len, vsb = 'ptb_lstm_h1500', args.seq_len, None
+    else:
+        value = args.batch * world * args.steps / elapsed
+        metric, unit = 'images/sec', 'images/sec'
+        seq_len = None
+        model_name = 'resnet50_v1' if args.model == 'resnet50' \
+            else 'inception_v3'
+        vsb = round(value / 219.0, 3) if args.model == 'resnet50' else None
     if rank == 0:
         out = {
-            'metric': 'images/sec',
-            'value': round(imgs_per_sec, 2),
-            'unit': 'images/sec',
+            'metric': metric,
+            'value': round(value, 2),
+            'unit': unit,
             'n_gpus': world,
             'steps': args.steps,
             'warmup': args.warmup,
             'ms_per_step': round(ms_per_step, 3),
             'higher_is_better': True,
             'scaling': 'weak',
-            'vs_baseline': round(imgs_per_sec / 219.0, 3),
+            'vs_baseline': vsb,
             'dtype': 'bf16',
             'data': 'synthetic',
             'config': {
-                'model': 'resnet50_v1',
+                'model': model_name,
                 'global_batch': args.batch * world,
-                'seq_len': None,
+                'seq_len': seq_len,
                 'parallelism': 'dp%d' % world,
             },
         }
@@ -96,6 +112,21 @@ def main():
 
 def _build(args, world):
     import simple_tensorflow_amd as tf
+    if args.model == 'inception3':
+        from simple_tensorflow_amd.models import inception
+        loss, train_op = inception.build_train_graph(batch=args.batch,
+                                                     lr=args.lr)
+        if world > 1:
+            raise SystemExit('inception bench is single-GPU in round 1')
+        return loss, train_op
+    if args.model == 'ptb_lstm':
+        from simple_tensorflow_amd.models import ptb_lstm
+        loss, train_op = ptb_lstm.build_ptb_graph(
+            batch=args.batch, seq_len=args.seq_len, hidden=1500,
+            vocab=10000, lr=args.lr)
+        if world > 1:
+            raise SystemExit('use the resnet50 model for multi-GPU scaling')
+        return loss, train_op
     from simple_tensorflow_amd.models import resnet
     images, labels = resnet.synthetic_inputs(args.batch)
     loss = resnet.resnet50_loss(images, labels)

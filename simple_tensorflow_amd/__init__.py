@@ -343,6 +343,12 @@ python_io = _PythonIoModule()
 
 from simple_tensorflow_amd.python import saved_model  # noqa: E402,F401
 from simple_tensorflow_amd.python import estimator  # noqa: E402,F401
+from simple_tensorflow_amd.python.ops import losses  # noqa: E402,F401
+from simple_tensorflow_amd.python.ops import metrics_impl as metrics  # noqa: E402,F401
+from simple_tensorflow_amd.python.ops import gradient_checker as _gc  # noqa: E402
+test = type(_sys)('simple_tensorflow_amd.test')
+test.compute_gradient = _gc.compute_gradient
+test.compute_gradient_error = _gc.compute_gradient_error
 from simple_tensorflow_amd.python.client import timeline  # noqa: E402,F401
 RunOptions = timeline.RunOptions
 RunMetadata = timeline.RunMetadata

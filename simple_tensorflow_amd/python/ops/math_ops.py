@@ -226,7 +226,11 @@ def argmin(x, axis=0, name=None, output_type=dtypes.int64):
 
 
 def select(cond, x, y, name=None):
-    return apply_op('Select', cond, x, y, name=name)
+    x = convert_to_tensor(x)
+    out = apply_op('Select', cond, x, y, name=name)
+    if out._shape is None and x._shape is not None:
+        out.set_shape(x._shape)
+    return out
 
 
 def range(start, limit=None, delta=1, dtype=None, name=None):  # pylint: disable=redefined-builtin

@@ -125,3 +125,50 @@ def test_fold_batch_norms():
         got = s.run(y2, {tf.get_default_graph().get_tensor_by_name('x:0'):
                          xv})
     np.testing.assert_allclose(got, want, rtol=1e-4, atol=1e-4)
+
+
+def test_transform_graph_pipeline(tmp_path):
+    import subprocess, sys
+    x = tf.placeholder(tf.float32, [2], name='x')
+    ident = tf.identity(x, name='mid')
+    tf.multiply(ident, tf.constant(3.0), name='out')
+    tf.sqrt(ident, name='dead')
+    in_pb = str(tmp_path / 'in.pb')
+    out_pb = str(tmp_path / 'out.pb')
+    with open(in_pb, 'wb') as f:
+        f.write(tf.get_default_graph().as_graph_def())
+    r = subprocess.run(
+        [sys.executable, '-m',
+         'simple_tensorflow_amd.python.tools.transform_graph',
+         '--in_graph', in_pb, '--out_graph', out_pb,
+         '--inputs', 'x', '--outputs', 'out',
+         '--transforms',
+         'remove_nodes(op=Identity) strip_unused_nodes'],
+        capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    with open(out_pb, 'rb') as f:
+        names = {n['name']: n for n in pbreader.parse_graph_def(f.read())}
+    assert 'dead' not in names and 'mid' not in names
+
+    tf.reset_default_graph()
+    with open(out_pb, 'rb') as f:
+        (out,) = tf.import_graph_def(f.read(), return_elements=['out:0'],
+                                     name='')
+    with tf.Session() as s:
+        v = s.run(out, {tf.get_default_graph().get_tensor_by_name('x:0'):
+                        np.array([1.0, 2.0], dtype=np.float32)})
+    np.testing.assert_allclose(v, [3.0, 6.0])
+
+
+def test_benchmark_model(tmp_path, capsys):
+    from simple_tensorflow_amd.python.tools import benchmark_model
+    x = tf.placeholder(tf.float32, [4, 8], name='x')
+    w = tf.constant(np.random.RandomState(0).randn(8, 8).astype(np.float32))
+    tf.matmul(x, w, name='y')
+    pb = str(tmp_path / 'g.pb')
+    with open(pb, 'wb') as f:
+        f.write(tf.get_default_graph().as_graph_def())
+    dt, rows = benchmark_model.benchmark(
+        pb, [('x', (4, 8), np.float32)], ['y'], num_runs=3, warmup=1)
+    assert dt > 0
+    assert any(r['name'] == 'MatMul' for r in rows)

@@ -18,7 +18,11 @@ __device__ __forceinline__ f32x4 mfma_bf16(bf16x8 a, bf16x8 b, f32x4 c) {
 }
 
 // One operand tile: BR rows x 64 k, stored as 16B chunks with chunk index
-// XOR-swizzled by (row&7). Slot s holds (r = s/8, c8 = (s%8) ^ (r&7)).
+// XOR-swizzled by Swz(row). Swz mixes bit 3+ of the row so that row-groups
+// of 8 (the K-major staging granularity, whose LDS row stride of 128B
+// aliases all 32 banks) still spread their writes across banks.
+__device__ __forceinline__ int Swz(int r) { return (r ^ (r >> 3)) & 7; }
+
 template <int BR>
 struct TileGeom {
   static constexpr int kSlots = BR * 8;        // 16B slots
@@ -36,7 +40,7 @@ __device__ __forceinline__ void StageFast(const uint16_t* __restrict__ src,
   for (int p = 0; p < TileGeom<BR>::kPasses; ++p) {
     int s = p * 256 + tid;
     int r = s >> 3;
-    int c8 = (s & 7) ^ (r & 7);
+    int c8 = (s & 7) ^ Swz(r);
     const uint16_t* g = src + (int64_t)(row0 + r) * ld + k0 + c8 * 8;
     __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) uint32_t*)g,
                                      (__attribute__((address_space(3))) uint32_t*)(lds_base + (int64_t)s * 8),
@@ -45,34 +49,32 @@ __device__ __forceinline__ void StageFast(const uint16_t* __restrict__ src,
 }
 
 // K-major staging: the operand is stored [K, R] row-major (contraction dim
-// outermost — e.g. im2col columns feeding dW, or a weight matrix [K, N] read
-// without a separate transpose pass). 16B global loads run along the
-// contiguous R dim; the transpose happens on the LDS write side as 8
-// staggered ds_write_b16s per load (the (e + lane) stagger spreads the
-// writes over banks — without it every lane in a row-group hits the same
-// bank because the 128B LDS row stride aliases all 32 banks).
+// outermost — im2col columns feeding dW, or a weight matrix [K, N] read
+// without a separate transpose pass). Each thread loads an 8x8 (k x r)
+// block with 16B row loads, transposes it in registers, and writes 8 full
+// 16B LDS slots — the Swz(r) swizzle keeps the 8-row-strided writes on
+// distinct banks.
 template <int BR>
 __device__ __forceinline__ void StageKMajor(const uint16_t* __restrict__ src,
                                             int64_t ld, int col0, int64_t k0,
                                             uint16_t* lds_base, int tid) {
-  constexpr int kRG = BR / 8;  // 16B row-groups per k row
+  constexpr int kRG = BR / 8;  // 8-wide row groups per k row
+  if (tid >= BR) return;       // one thread per 8x8 block (8 k-chunks)
+  int r0 = (tid % kRG) * 8;
+  int chunk = tid / kRG;       // k chunk of 8
+  const uint16_t* g = src + (k0 + chunk * 8) * ld + col0 + r0;
+  uint16_t v[8][8];
 #pragma unroll
-  for (int p = 0; p < BR / 32; ++p) {
-    int idx = p * 256 + tid;
-    int k = idx / kRG;
-    int r0 = (idx % kRG) * 8;
-    const uint16_t* g = src + (k0 + k) * ld + col0 + r0;
-    uint16_t vals[8];
-    *(ulong2*)vals = *(const ulong2*)g;
-    int c8 = k >> 3;
-    int klo = (k & 7) * 2;
+  for (int kk = 0; kk < 8; ++kk)
+    *(ulong2*)v[kk] = *(const ulong2*)(g + (int64_t)kk * ld);
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      int e = (i + tid) & 7;
-      int r = r0 + e;
-      *(uint16_t*)((char*)(lds_base + (int64_t)(r * 8 + (c8 ^ (r & 7))) * 8) +
-                   klo) = vals[e];
-    }
+  for (int e = 0; e < 8; ++e) {
+    uint16_t out[8];
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk) out[kk] = v[kk][e];
+    int r = r0 + e;
+    *(ulong2*)(lds_base + (int64_t)(r * 8 + (chunk ^ Swz(r))) * 8) =
+        *(ulong2*)out;
   }
 }
 
@@ -81,32 +83,33 @@ __device__ __forceinline__ void StageKMajorSafe(
     const uint16_t* __restrict__ src, int64_t ld, int col0, int64_t k0,
     int64_t cols, int64_t K, uint16_t* lds_base, int tid) {
   constexpr int kRG = BR / 8;
+  if (tid >= BR) return;
+  int r0 = (tid % kRG) * 8;
+  int chunk = tid / kRG;
+  uint16_t v[8][8];
 #pragma unroll
-  for (int p = 0; p < BR / 32; ++p) {
-    int idx = p * 256 + tid;
-    int k = idx / kRG;
-    int r0 = (idx % kRG) * 8;
-    uint16_t vals[8];
-    int64_t krow = k0 + k;
+  for (int kk = 0; kk < 8; ++kk) {
+    int64_t krow = k0 + chunk * 8 + kk;
     if (krow < K && col0 + r0 + 8 <= cols) {
-      *(ulong2*)vals = *(const ulong2*)(src + krow * ld + col0 + r0);
+      *(ulong2*)v[kk] = *(const ulong2*)(src + krow * ld + col0 + r0);
     } else if (krow < K) {
 #pragma unroll
       for (int e = 0; e < 8; ++e)
-        vals[e] = (col0 + r0 + e < cols) ? src[krow * ld + col0 + r0 + e] : 0;
+        v[kk][e] = (col0 + r0 + e < cols) ? src[krow * ld + col0 + r0 + e]
+                                          : 0;
     } else {
 #pragma unroll
-      for (int e = 0; e < 8; ++e) vals[e] = 0;
+      for (int e = 0; e < 8; ++e) v[kk][e] = 0;
     }
-    int c8 = k >> 3;
-    int klo = (k & 7) * 2;
+  }
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      int e = (i + tid) & 7;
-      int r = r0 + e;
-      *(uint16_t*)((char*)(lds_base + (int64_t)(r * 8 + (c8 ^ (r & 7))) * 8) +
-                   klo) = vals[e];
-    }
+  for (int e = 0; e < 8; ++e) {
+    uint16_t out[8];
+#pragma unroll
+    for (int kk = 0; kk < 8; ++kk) out[kk] = v[kk][e];
+    int r = r0 + e;
+    *(ulong2*)(lds_base + (int64_t)(r * 8 + (chunk ^ Swz(r))) * 8) =
+        *(ulong2*)out;
   }
 }
 
@@ -120,7 +123,7 @@ __device__ __forceinline__ void StageSafe(const uint16_t* __restrict__ src,
   for (int p = 0; p < TileGeom<BR>::kPasses; ++p) {
     int s = p * 256 + tid;
     int r = s >> 3;
-    int c8 = (s & 7) ^ (r & 7);
+    int c8 = (s & 7) ^ Swz(r);
     uint16_t vals[8];
     int64_t row = row0 + r;
     int64_t kbase = k0 + c8 * 8;
@@ -243,13 +246,13 @@ __launch_bounds__(256) __global__ void GemmBf16NT(
 #pragma unroll
       for (int i = 0; i < WM; ++i) {
         int r = wr * WM * 16 + i * 16 + (lane & 15);
-        int c8 = (kk * 4 + (lane >> 4)) ^ (r & 7);
+        int c8 = (kk * 4 + (lane >> 4)) ^ Swz(r);
         afrag[i] = *(const bf16x8*)(at + (r * 8 + c8) * 8);
       }
 #pragma unroll
       for (int j = 0; j < WN; ++j) {
         int r = wc * WN * 16 + j * 16 + (lane & 15);
-        int c8 = (kk * 4 + (lane >> 4)) ^ (r & 7);
+        int c8 = (kk * 4 + (lane >> 4)) ^ Swz(r);
         bfrag[j] = *(const bf16x8*)(bt + (r * 8 + c8) * 8);
       }
 #pragma unroll

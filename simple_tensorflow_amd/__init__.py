@@ -344,6 +344,7 @@ python_io = _PythonIoModule()
 from simple_tensorflow_amd.python import saved_model  # noqa: E402,F401
 from simple_tensorflow_amd.python import estimator  # noqa: E402,F401
 from simple_tensorflow_amd.python.ops import losses  # noqa: E402,F401
+from simple_tensorflow_amd.python.ops import image_ops_impl as image  # noqa: E402,F401
 from simple_tensorflow_amd.python.ops import metrics_impl as metrics  # noqa: E402,F401
 from simple_tensorflow_amd.python.ops import gradient_checker as _gc  # noqa: E402
 test = type(_sys)('simple_tensorflow_amd.test')

@@ -218,6 +218,9 @@ REGISTER_OP("MergeV2Checkpoints").Input("checkpoint_prefixes: string").Input("de
 // ------------------------------- queues ------------------------------------
 // TensorArray family (reference core/ops/data_flow_ops.cc:1080; resource
 // handles are session-scoped strings here, like the queue ops above).
+REGISTER_OP("ResizeBilinear").Input("images: T").Input("size: int32").Output("resized_images: float").Attr("T: {float}").Attr("align_corners: bool = false");
+REGISTER_OP("ResizeBilinearGrad").Input("grads: float").Input("original_image: T").Output("output: T").Attr("T: {float}").Attr("align_corners: bool = false");
+REGISTER_OP("ResizeNearestNeighbor").Input("images: T").Input("size: int32").Output("resized_images: T").Attr("T: {float}").Attr("align_corners: bool = false");
 REGISTER_OP("ConditionalAccumulator").Output("handle: Ref(string)").Attr("dtype: type").Attr("shape: shape").Attr("container: string = ''").Attr("shared_name: string = ''").SetIsStateful();
 REGISTER_OP("AccumulatorApplyGradient").Input("handle: Ref(string)").Input("local_step: int64").Input("gradient: dtype").Attr("dtype: type").SetIsStateful();
 REGISTER_OP("AccumulatorTakeGradient").Input("handle: Ref(string)").Input("num_required: int32").Output("average: dtype").Attr("dtype: type").SetIsStateful();

@@ -1,0 +1,143 @@
+"""tf.image (reference python/ops/image_ops_impl.py subset: resize family,
+flips, crops, standardization; resize kernels in csrc/kernels/cpu_image.cc)."""
+from simple_tensorflow_amd.python.framework import dtypes, ops
+from simple_tensorflow_amd.python.framework.ops import (RegisterGradient,
+                                                        apply_op,
+                                                        convert_to_tensor)
+from simple_tensorflow_amd.python.ops import array_ops, math_ops, random_ops
+
+
+class ResizeMethod(object):
+    BILINEAR = 0
+    NEAREST_NEIGHBOR = 1
+    BICUBIC = 2
+    AREA = 3
+
+
+def resize_bilinear(images, size, align_corners=False, name=None):
+    images = convert_to_tensor(images)
+    size_t = convert_to_tensor(size, dtype=dtypes.int32)
+    out = apply_op('ResizeBilinear', images, size_t,
+                   align_corners=align_corners, name=name)
+    if images._shape is not None and not isinstance(size, ops.Tensor):
+        out.set_shape([images._shape[0], size[0], size[1],
+                       images._shape[3]])
+    return out
+
+
+def resize_nearest_neighbor(images, size, align_corners=False, name=None):
+    images = convert_to_tensor(images)
+    size_t = convert_to_tensor(size, dtype=dtypes.int32)
+    out = apply_op('ResizeNearestNeighbor', images, size_t,
+                   align_corners=align_corners, name=name)
+    if images._shape is not None and not isinstance(size, ops.Tensor):
+        out.set_shape([images._shape[0], size[0], size[1],
+                       images._shape[3]])
+    return out
+
+
+def resize_images(images, size, method=ResizeMethod.BILINEAR,
+                  align_corners=False):
+    if method == ResizeMethod.BILINEAR:
+        return resize_bilinear(images, size, align_corners)
+    if method == ResizeMethod.NEAREST_NEIGHBOR:
+        return resize_nearest_neighbor(images, size, align_corners)
+    raise NotImplementedError('resize method %r' % method)
+
+
+@RegisterGradient('ResizeBilinear')
+def _resize_bilinear_grad(op, grad):
+    dx = apply_op('ResizeBilinearGrad', grad, op.inputs[0],
+                  align_corners=op.get_attr('align_corners'))
+    dx.set_shape(op.inputs[0]._shape)
+    return [dx, None]
+
+
+def flip_left_right(image):
+    image = convert_to_tensor(image)
+    return _reverse(image, axis=1)
+
+
+def flip_up_down(image):
+    image = convert_to_tensor(image)
+    return _reverse(image, axis=0)
+
+
+def _reverse(image, axis):
+    # gather-based reverse along `axis` of a 3-D [h, w, c] image
+    n = image._shape[axis]
+    idx = ops.constant(list(range(n - 1, -1, -1)), dtype=dtypes.int32)
+    if axis == 0:
+        return array_ops.gather(image, idx)
+    perm = [1, 0, 2]
+    return array_ops.transpose(
+        array_ops.gather(array_ops.transpose(image, perm), idx), perm)
+
+
+def random_flip_left_right(image, seed=None):
+    from simple_tensorflow_amd.python.ops import control_flow_ops
+    u = random_ops.random_uniform([], 0.0, 1.0, seed=seed)
+    return control_flow_ops.cond(math_ops.less(u, 0.5),
+                                 lambda: flip_left_right(image),
+                                 lambda: array_ops.identity(image))
+
+
+def crop_to_bounding_box(image, offset_height, offset_width, target_height,
+                         target_width):
+    image = convert_to_tensor(image)
+    if len(image._shape) == 3:
+        return image[offset_height:offset_height + target_height,
+                     offset_width:offset_width + target_width, :]
+    return image[:, offset_height:offset_height + target_height,
+                 offset_width:offset_width + target_width, :]
+
+
+def central_crop(image, central_fraction):
+    image = convert_to_tensor(image)
+    h, w = image._shape[0], image._shape[1]
+    ch = int(h * central_fraction)
+    cw = int(w * central_fraction)
+    oh = (h - ch) // 2
+    ow = (w - cw) // 2
+    return crop_to_bounding_box(image, oh, ow, ch, cw)
+
+
+def random_crop(image, size, seed=None):
+    image = convert_to_tensor(image)
+    h, w = image._shape[0], image._shape[1]
+    th, tw = size[0], size[1]
+    oh = random_ops.random_uniform([], 0, h - th + 1, dtype=dtypes.int32,
+                                   seed=seed) if h > th else 0
+    ow = random_ops.random_uniform([], 0, w - tw + 1, dtype=dtypes.int32,
+                                   seed=seed) if w > tw else 0
+    begin = array_ops.stack([oh, ow, ops.constant(0)]) \
+        if isinstance(oh, ops.Tensor) else [oh, ow, 0]
+    out = array_ops.slice(image, begin, [th, tw, image._shape[2]])
+    out.set_shape([th, tw, image._shape[2]])
+    return out
+
+
+def per_image_standardization(image):
+    image = math_ops.cast(convert_to_tensor(image), dtypes.float32)
+    num = 1
+    for d in image._shape:
+        num *= d
+    mean = math_ops.reduce_mean(image)
+    variance = math_ops.reduce_mean(math_ops.square(image)) - \
+        math_ops.square(mean)
+    stddev = math_ops.sqrt(math_ops.maximum(variance, ops.constant(0.0)))
+    min_stddev = ops.constant(1.0 / float(num) ** 0.5)
+    adj = math_ops.maximum(stddev, min_stddev)
+    return (image - mean) / adj
+
+
+def convert_image_dtype(image, dtype, saturate=False, name=None):
+    image = convert_to_tensor(image)
+    dtype = dtypes.as_dtype(dtype)
+    if image.dtype == dtype:
+        return image
+    if image.dtype in (dtypes.uint8,) and dtype == dtypes.float32:
+        return math_ops.cast(image, dtype) * ops.constant(1.0 / 255.0)
+    if image.dtype == dtypes.float32 and dtype == dtypes.uint8:
+        return math_ops.cast(image * ops.constant(255.0), dtype)
+    return math_ops.cast(image, dtype)

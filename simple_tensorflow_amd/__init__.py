@@ -345,6 +345,12 @@ from simple_tensorflow_amd.python import saved_model  # noqa: E402,F401
 from simple_tensorflow_amd.python import estimator  # noqa: E402,F401
 from simple_tensorflow_amd.python.ops import losses  # noqa: E402,F401
 from simple_tensorflow_amd.python.ops import image_ops_impl as image  # noqa: E402,F401
+from simple_tensorflow_amd.python.platform import app  # noqa: E402,F401
+from simple_tensorflow_amd.python.platform import gfile  # noqa: E402,F401
+from simple_tensorflow_amd.python.platform import tf_logging as logging  # noqa: E402,F401
+app.flags = __import__('simple_tensorflow_amd.python.platform.flags',
+                       fromlist=['flags'])
+flags = app.flags
 from simple_tensorflow_amd.python.ops import metrics_impl as metrics  # noqa: E402,F401
 from simple_tensorflow_amd.python.ops import gradient_checker as _gc  # noqa: E402
 test = type(_sys)('simple_tensorflow_amd.test')

@@ -268,6 +268,12 @@ class _TrainModule(object):
 
     SyncReplicasOptimizer = _sync_opt.SyncReplicasOptimizer
 
+    @staticmethod
+    def Supervisor(*a, **kw):
+        from simple_tensorflow_amd.python.training import supervisor
+        return supervisor.Supervisor(*a, **kw)
+
+
     Saver = _saver.Saver
 
     @staticmethod

@@ -331,6 +331,8 @@ train = _TrainModule()
 
 class _SummaryModule(object):
     scalar = staticmethod(_summary_mod.scalar)
+    image = staticmethod(_summary_mod.image)
+    audio = staticmethod(_summary_mod.audio)
     histogram = staticmethod(_summary_mod.histogram)
     merge = staticmethod(_summary_mod.merge)
     merge_all = staticmethod(_summary_mod.merge_all)

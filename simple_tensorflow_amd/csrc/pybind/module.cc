@@ -209,7 +209,14 @@ class PyFuncOp : public OpKernel {
       return;
     }
     for (size_t i = 0; i < tout_.size(); ++i) {
-      py::array arr = py::array::ensure(outs[i]);
+      py::object o = outs[i];
+      if (tout_[i] == DT_STRING) {
+        Tensor t(DT_STRING, TensorShape({}));
+        t.flat<std::string>()[0] = py::cast<std::string>(o);
+        ctx->set_output((int)i, t);
+        continue;
+      }
+      py::array arr = py::array::ensure(o);
       Tensor t = NumpyToTensor(arr);
       if (t.dtype() != tout_[i]) {
         ctx->SetStatus(errors::InvalidArgument(

@@ -1031,6 +1031,40 @@ REGISTER_KERNEL_BUILDER(Name("AvgPoolGrad").Device(DEVICE_GPU).TypeConstraint<bf
 // ---------------------------------------------------------------------------
 // reductions: Sum / Mean / Max (all-axes, trailing-axes, or leading-axes)
 // ---------------------------------------------------------------------------
+// L2Loss: 0.5 * sum(x^2) — square into a temp then full-reduce (the
+// reference's l2loss_op_gpu; without this PTB's clip_by_global_norm dragged
+// multi-MB gradients through a CPU-placed reduction at ~22 s/step).
+class GpuL2LossOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    hipStream_t s = GPU_STREAM(ctx);
+    int dt = DtypeCode(x.dtype());
+    int64_t n = x.NumElements();
+    Tensor sq = ctx->allocate_temp(x.dtype(), x.shape());
+    OP_HIP_OK(ctx, stf_binary(B_MUL, dt, x.raw_data(), x.raw_data(),
+                              sq.raw_data(), n, s));
+    Tensor acc = ctx->allocate_temp(DT_FLOAT, TensorShape({}));
+    OP_HIP_OK(ctx, ZeroF32(acc.raw_data(), 1, s));
+    OP_HIP_OK(ctx, stf_full_reduce(dt, 0 /*sum*/, sq.raw_data(),
+                                   acc.flat<float>(), n, s));
+    Tensor* out = ctx->allocate_output(0, TensorShape({}));
+    if (x.dtype() == DT_FLOAT) {
+      OP_HIP_OK(ctx, stf_scale(0, acc.raw_data(), out->raw_data(), 1, 0.5f,
+                               s));
+    } else {
+      Tensor half = ctx->allocate_temp(DT_FLOAT, TensorShape({}));
+      OP_HIP_OK(ctx, stf_scale(0, acc.raw_data(), half.raw_data(), 1, 0.5f,
+                               s));
+      OP_HIP_OK(ctx, stf_cast(0, CastCode(x.dtype()), half.raw_data(),
+                              out->raw_data(), 1, s));
+    }
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("L2Loss").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuL2LossOp);
+REGISTER_KERNEL_BUILDER(Name("L2Loss").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuL2LossOp);
+
 class GpuReduceOp : public OpKernel {
  public:
   GpuReduceOp(OpKernelConstruction* c, int red, bool mean)

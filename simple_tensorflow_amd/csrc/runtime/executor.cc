@@ -464,7 +464,13 @@ class ExecutorState {
       if (want == MemSpace::HOST && inputs[i].mem_space() == MemSpace::DEVICE) {
         Tensor host;
         Status s = dev->CopyDeviceTensorToHost(inputs[i], &host);
-        if (!s.ok()) { Fail(s); Done(t, outs, dead, inline_q); return; }
+        if (!s.ok()) {
+          Fail(Status(s.code(), "input " + std::to_string(i) + " of node " +
+                                    n->name() + " (" + n->op() +
+                                    "): " + s.message()));
+          Done(t, outs, dead, inline_q);
+          return;
+        }
         inputs[i] = host;
       } else if (want == MemSpace::DEVICE &&
                  inputs[i].mem_space() == MemSpace::HOST) {

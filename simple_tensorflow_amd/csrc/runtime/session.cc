@@ -566,7 +566,16 @@ Status DirectSession::Run(
       ek->capture_broken = true;
     }
   }
-  if (!agg.ok()) return agg;
+  if (!agg.ok()) {
+    if (do_capture) {
+      // The capture attempt failed mid-step (e.g. a host-dependent op needed
+      // a d2h copy). Stream capture executed nothing on the device, so the
+      // step can simply be retried eagerly; capture_broken is already set so
+      // the retry (and all later steps) skip capture.
+      return Run(feeds, fetches, targets, outputs);
+    }
+    return agg;
+  }
   return fetch_status;
 }
 

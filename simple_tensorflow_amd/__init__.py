@@ -353,6 +353,10 @@ from simple_tensorflow_amd.python import saved_model  # noqa: E402,F401
 from simple_tensorflow_amd.python import estimator  # noqa: E402,F401
 from simple_tensorflow_amd.python.ops import losses  # noqa: E402,F401
 from simple_tensorflow_amd.python.ops import image_ops_impl as image  # noqa: E402,F401
+from simple_tensorflow_amd.python.ops import sparse_ops as _sparse_ops  # noqa: E402
+SparseTensor = _sparse_ops.SparseTensor
+sparse_to_dense = _sparse_ops.sparse_to_dense
+sparse_tensor_to_dense = _sparse_ops.sparse_tensor_to_dense
 from simple_tensorflow_amd.python.platform import app  # noqa: E402,F401
 from simple_tensorflow_amd.python.platform import gfile  # noqa: E402,F401
 from simple_tensorflow_amd.python.platform import tf_logging as logging  # noqa: E402,F401

@@ -869,4 +869,45 @@ class CheckNumericsOp : public OpKernel {
 };
 REGISTER_CPU_KERNEL_FLOATS("CheckNumerics", CheckNumericsOp)
 
+// ----------------------------- SparseToDense --------------------------------
+// (reference core/kernels/sparse_to_dense_op.cc)
+class SparseToDenseOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& indices = ctx->input(0);
+    const Tensor& shape_t = ctx->input(1);
+    const Tensor& values = ctx->input(2);
+    const Tensor& default_v = ctx->input(3);
+    TensorShape out_shape;
+    for (int64_t i = 0; i < shape_t.NumElements(); ++i)
+      out_shape.AddDim(shape_t.dtype() == DT_INT64
+                           ? shape_t.flat<int64_t>()[i]
+                           : shape_t.flat<int32_t>()[i]);
+    Tensor* out = ctx->allocate_output(0, out_shape);
+    size_t es = DataTypeSize(out->dtype());
+    char* base = (char*)out->raw_data();
+    const char* dv = (const char*)default_v.raw_data();
+    for (int64_t i = 0; i < out->NumElements(); ++i)
+      std::memcpy(base + i * es, dv, es);
+    int64_t nnz = indices.dims() == 0 ? 1 : indices.dim_size(0);
+    int rank = out_shape.dims();
+    auto idx_at = [&](int64_t n, int d) -> int64_t {
+      int64_t flat_i = indices.dims() <= 1 ? n : n * rank + d;
+      return indices.dtype() == DT_INT64 ? indices.flat<int64_t>()[flat_i]
+                                         : indices.flat<int32_t>()[flat_i];
+    };
+    const char* vals = (const char*)values.raw_data();
+    bool scalar_val = values.NumElements() == 1;
+    for (int64_t n = 0; n < nnz; ++n) {
+      int64_t off = 0;
+      for (int d = 0; d < rank; ++d)
+        off = off * out_shape.dim_size(d) + idx_at(n, d);
+      std::memcpy(base + off * es, vals + (scalar_val ? 0 : n) * es, es);
+    }
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("SparseToDense").Device(DEVICE_CPU),
+                        SparseToDenseOp);
+
 }  // namespace stf

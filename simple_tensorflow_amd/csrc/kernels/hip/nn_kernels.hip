@@ -286,6 +286,33 @@ __global__ void BnNormKernel(const T* __restrict__ x,
   }
 }
 
+// C%8==0 fast path: 16B vector loads/stores, one modulo per 8 channels.
+template <bool RELU>
+__global__ void BnNormKernelV8(const __bf16* __restrict__ x,
+                               const float* __restrict__ mean,
+                               const float* __restrict__ inv_std,
+                               const float* __restrict__ scale,
+                               const float* __restrict__ offset,
+                               __bf16* __restrict__ y, int64_t n8, int c) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int cg = c / 8;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += stride) {
+    int cb = (int)(i % cg) * 8;
+    __bf16 v[8], o[8];
+    *(ulong2*)v = *(const ulong2*)(x + i * 8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float f = ((float)v[e] - mean[cb + e]) * inv_std[cb + e] *
+                    scale[cb + e] +
+                offset[cb + e];
+      if (RELU) f = f > 0.f ? f : 0.f;
+      o[e] = (__bf16)f;
+    }
+    *(ulong2*)(y + i * 8) = *(ulong2*)o;
+  }
+}
+
 // bwd pass 1: per-channel sum(dy), sum(dy * xhat) into acc[2C] (zeroed).
 // RELU: the forward was BN+ReLU fused; dy is masked by y>0 on the fly so no
 // separate relu-grad elementwise pass (or its memory traffic) is needed.
@@ -396,6 +423,41 @@ __global__ void BnGradKernel(const T* __restrict__ dy, const T* __restrict__ x,
   }
 }
 
+template <bool RELU>
+__global__ void BnGradKernelV8(const __bf16* __restrict__ dy,
+                               const __bf16* __restrict__ x,
+                               const __bf16* __restrict__ yr,
+                               const float* __restrict__ mean,
+                               const float* __restrict__ inv_std,
+                               const float* __restrict__ scale,
+                               const float* __restrict__ sum_dy,
+                               const float* __restrict__ sum_dy_xhat,
+                               __bf16* __restrict__ dx, int64_t n8,
+                               int64_t rows, int c) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  float inv_rows = 1.f / (float)rows;
+  int cg = c / 8;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += stride) {
+    int cb = (int)(i % cg) * 8;
+    __bf16 g8[8], x8[8], y8[8], o[8];
+    *(ulong2*)g8 = *(const ulong2*)(dy + i * 8);
+    *(ulong2*)x8 = *(const ulong2*)(x + i * 8);
+    if (RELU) *(ulong2*)y8 = *(const ulong2*)(yr + i * 8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      int ch = cb + e;
+      float xhat = ((float)x8[e] - mean[ch]) * inv_std[ch];
+      float g = (float)g8[e];
+      if (RELU && (float)y8[e] <= 0.f) g = 0.f;
+      o[e] = (__bf16)(scale[ch] * inv_std[ch] *
+                      (g - sum_dy[ch] * inv_rows -
+                       xhat * sum_dy_xhat[ch] * inv_rows));
+    }
+    *(ulong2*)(dx + i * 8) = *(ulong2*)o;
+  }
+}
+
 // ---------------- pooling ----------------
 struct PoolGeom {
   int N, H, W, C, kh, kw, sh, sw, ph, pw, P, Q;
@@ -461,6 +523,49 @@ __global__ void MaxPoolGradKernel(const T* __restrict__ x,
       }
     }
     if (best_idx >= 0) atomicAdd(&dx_f32[best_idx], (float)dy[i]);
+  }
+}
+
+// C%8==0: one thread per (n,p,q, 8-channel group) — 16B window loads, 8
+// independent argmax lanes, 8 scattered f32 atomics.
+__global__ void MaxPoolGradKernelV8(const __bf16* __restrict__ x,
+                                    const __bf16* __restrict__ dy,
+                                    float* __restrict__ dx_f32, PoolGeom g,
+                                    int64_t total8) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int cg = g.C / 8;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total8;
+       i += stride) {
+    int64_t rem = i;
+    int cb = (int)(rem % cg) * 8; rem /= cg;
+    int q = (int)(rem % g.Q); rem /= g.Q;
+    int p = (int)(rem % g.P); rem /= g.P;
+    int n = (int)rem;
+    float best[8];
+    int64_t bidx[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) { best[e] = -3.4e38f; bidx[e] = -1; }
+    for (int kh = 0; kh < g.kh; ++kh) {
+      int ih = p * g.sh - g.ph + kh;
+      if (ih < 0 || ih >= g.H) continue;
+      for (int kw = 0; kw < g.kw; ++kw) {
+        int iw = q * g.sw - g.pw + kw;
+        if (iw < 0 || iw >= g.W) continue;
+        int64_t idx = ((int64_t)(n * g.H + ih) * g.W + iw) * g.C + cb;
+        __bf16 v8[8];
+        *(ulong2*)v8 = *(const ulong2*)(x + idx);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float v = (float)v8[e];
+          if (v > best[e]) { best[e] = v; bidx[e] = idx + e; }
+        }
+      }
+    }
+    __bf16 g8[8];
+    *(ulong2*)g8 = *(const ulong2*)(dy + i * 8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      if (bidx[e] >= 0) atomicAdd(&dx_f32[bidx[e]], (float)g8[e]);
   }
 }
 
@@ -605,6 +710,19 @@ hipError_t stf_bn_fwd(int dtype, const void* x, const void* scale,
                      (const float*)offset, (T*)y, n, c)
   if (dtype == 0) {
     if (fuse_relu) BNN(float, true); else BNN(float, false);
+  } else if (c % 8 == 0) {
+    int64_t n8 = n / 8;
+    dim3 g8 = ElemwiseGrid(n8, 256, 1);
+    if (fuse_relu)
+      hipLaunchKernelGGL((BnNormKernelV8<true>), g8, dim3(256), 0, stream,
+                         (const __bf16*)x, mean, inv_std,
+                         (const float*)scale, (const float*)offset,
+                         (__bf16*)y, n8, c);
+    else
+      hipLaunchKernelGGL((BnNormKernelV8<false>), g8, dim3(256), 0, stream,
+                         (const __bf16*)x, mean, inv_std,
+                         (const float*)scale, (const float*)offset,
+                         (__bf16*)y, n8, c);
   } else {
     if (fuse_relu) BNN(__bf16, true); else BNN(__bf16, false);
   }
@@ -635,11 +753,28 @@ hipError_t stf_bn_bwd(int dtype, const void* dy, const void* x,
                        inv_std, (const float*)scale, sum_dy, sum_dy_xhat,   \
                        (T*)dx, n, rows, c);                                 \
   } while (0)
+#define BNB8(R)                                                             \
+  do {                                                                      \
+    hipLaunchKernelGGL((BnGradStatsKernel<__bf16, R>), dim3(blocks),        \
+                       dim3(256), lds, stream, (const __bf16*)dy,           \
+                       (const __bf16*)x, (const __bf16*)y_relu, mean,       \
+                       inv_std, sum_dy, sum_dy_xhat, rows, c);              \
+    int64_t n8 = n / 8;                                                     \
+    dim3 g8 = ElemwiseGrid(n8, 256, 1);                                     \
+    hipLaunchKernelGGL((BnGradKernelV8<R>), g8, dim3(256), 0, stream,       \
+                       (const __bf16*)dy, (const __bf16*)x,                 \
+                       (const __bf16*)y_relu, mean, inv_std,                \
+                       (const float*)scale, sum_dy, sum_dy_xhat,            \
+                       (__bf16*)dx, n8, rows, c);                           \
+  } while (0)
   if (dtype == 0) {
     if (fuse_relu) BNB(float, true); else BNB(float, false);
+  } else if (c % 8 == 0) {
+    if (fuse_relu) BNB8(true); else BNB8(false);
   } else {
     if (fuse_relu) BNB(__bf16, true); else BNB(__bf16, false);
   }
+#undef BNB8
 #undef BNB
   return hipGetLastError();
 }
@@ -673,7 +808,12 @@ hipError_t stf_max_pool_bwd(int dtype, const void* x, const void* dy,
   if (dtype == 0)
     hipLaunchKernelGGL((MaxPoolGradKernel<float>), grid, dim3(256), 0, stream,
                        (const float*)x, (const float*)dy, dx_f32, g, total);
-  else
+  else if (C % 8 == 0) {
+    int64_t t8 = total / 8;
+    hipLaunchKernelGGL(MaxPoolGradKernelV8, ElemwiseGrid(t8, 256, 1),
+                       dim3(256), 0, stream, (const __bf16*)x,
+                       (const __bf16*)dy, dx_f32, g, t8);
+  } else
     hipLaunchKernelGGL((MaxPoolGradKernel<__bf16>), grid, dim3(256), 0, stream,
                        (const __bf16*)x, (const __bf16*)dy, dx_f32, g, total);
   return hipGetLastError();

@@ -234,6 +234,72 @@ class PyFuncOp : public OpKernel {
 REGISTER_KERNEL_BUILDER(Name("PyFunc").Device(DEVICE_CPU), PyFuncOp);
 REGISTER_KERNEL_BUILDER(Name("PyFuncStateless").Device(DEVICE_CPU), PyFuncOp);
 
+// CPU fallback for the RCCL collective ops: delegates to a python callable
+// (registered by parallel/dist.py, backed by torch.distributed gloo) so the
+// multi-process data-parallel graph runs — and is tested — on CPU-only
+// hosts. On GPU boxes the HIP RCCL kernels in rccl/rccl_ops.cc are used.
+class CpuCollectiveOp : public OpKernel {
+ public:
+  CpuCollectiveOp(OpKernelConstruction* c, const char* token)
+      : OpKernel(c), token_(token) {}
+  void Compute(OpKernelContext* ctx) override {
+    py::gil_scoped_acquire gil;
+    py::function fn;
+    {
+      std::lock_guard<std::mutex> l(g_pyfunc_mu);
+      auto it = PyFuncRegistry().find(token_);
+      if (it == PyFuncRegistry().end()) {
+        ctx->SetStatus(errors::FailedPrecondition(
+            "CPU collective fallback not initialized (", token_,
+            ") — call parallel.dist.init() first"));
+        return;
+      }
+      fn = it->second;
+    }
+    py::object result;
+    try {
+      result = fn(TensorToPy(ctx->input(0)));
+    } catch (py::error_already_set& e) {
+      ctx->SetStatus(errors::Internal("collective raised: ", e.what()));
+      return;
+    }
+    py::array arr = py::array::ensure(result);
+    Tensor t = NumpyToTensor(arr);
+    if (t.dtype() != ctx->input(0).dtype()) {
+      // gloo reduces in f32; cast back to the graph dtype (e.g. bf16)
+      Tensor out(ctx->input(0).dtype(), t.shape());
+      if (ctx->input(0).dtype() == DT_BFLOAT16 && t.dtype() == DT_FLOAT) {
+        const float* src = t.flat<float>();
+        bfloat16* dst = out.flat<bfloat16>();
+        for (int64_t i = 0; i < t.NumElements(); ++i)
+          dst[i] = bfloat16(src[i]);
+        ctx->set_output(0, out);
+        return;
+      }
+      ctx->SetStatus(errors::Internal("collective dtype mismatch"));
+      return;
+    }
+    ctx->set_output(0, t);
+  }
+
+ private:
+  std::string token_;
+};
+class CpuAllReduceOp : public CpuCollectiveOp {
+ public:
+  explicit CpuAllReduceOp(OpKernelConstruction* c)
+      : CpuCollectiveOp(c, "__cpu_collective_allreduce") {}
+};
+class CpuBroadcastOp : public CpuCollectiveOp {
+ public:
+  explicit CpuBroadcastOp(OpKernelConstruction* c)
+      : CpuCollectiveOp(c, "__cpu_collective_broadcast") {}
+};
+REGISTER_KERNEL_BUILDER(Name("RcclAllReduce").Device(DEVICE_CPU),
+                        CpuAllReduceOp);
+REGISTER_KERNEL_BUILDER(Name("RcclBroadcast").Device(DEVICE_CPU),
+                        CpuBroadcastOp);
+
 class PySession {
  public:
   explicit PySession(bool cpu_only) : sess_(cpu_only) {}

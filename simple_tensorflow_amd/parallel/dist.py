@@ -56,20 +56,37 @@ class Comm(object):
 
 
 def init(world, rank):
-    """Initialize gloo rendezvous + the RCCL communicator."""
+    """Initialize gloo rendezvous + the RCCL communicator (GPU) or the
+    torch.distributed-backed CPU collective fallback (CPU-only hosts /
+    tests — csrc/pybind/module.cc CpuCollectiveOp)."""
     import torch.distributed as td
     if not td.is_initialized():
         os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
         td.init_process_group('gloo', rank=rank, world_size=world)
     import torch
-    if rank == 0:
-        uid = _core.rccl_get_unique_id()
-        buf = torch.tensor(list(uid), dtype=torch.uint8)
+    has_gpu = tf.Session().num_gpus() > 0
+    if has_gpu:
+        if rank == 0:
+            uid = _core.rccl_get_unique_id()
+            buf = torch.tensor(list(uid), dtype=torch.uint8)
+        else:
+            buf = torch.zeros(128, dtype=torch.uint8)
+        td.broadcast(buf, src=0)
+        uid = bytes(buf.tolist())
+        _core.rccl_init(world, rank, uid)
     else:
-        buf = torch.zeros(128, dtype=torch.uint8)
-    td.broadcast(buf, src=0)
-    uid = bytes(buf.tolist())
-    _core.rccl_init(world, rank, uid)
+        def _allreduce(arr):
+            t = torch.from_numpy(np.ascontiguousarray(arr))
+            td.all_reduce(t, op=td.ReduceOp.SUM)
+            return t.numpy()
+
+        def _broadcast(arr):
+            t = torch.from_numpy(np.ascontiguousarray(arr))
+            td.broadcast(t, src=0)
+            return t.numpy()
+
+        _core.register_py_func('__cpu_collective_allreduce', _allreduce)
+        _core.register_py_func('__cpu_collective_broadcast', _broadcast)
     return Comm(world, rank, td)
 
 

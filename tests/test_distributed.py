@@ -60,3 +60,21 @@ def test_cluster_spec():
     class FakeOp:
         type = 'VariableV2'
     assert setter(FakeOp()).startswith('/job:ps/task:')
+
+
+def test_data_parallel_two_process_gloo():
+    """The RCCL data-parallel graph on 2 CPU processes over the gloo-backed
+    collective fallback (the same graph the 8-GPU bench runs)."""
+    import subprocess, sys, os
+    env = dict(os.environ)
+    env['MASTER_ADDR'] = '127.0.0.1'
+    env['MASTER_PORT'] = '29713'
+    r = subprocess.run(
+        [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
+         '--nproc-per-node', '2', '--master-addr', '127.0.0.1',
+         '--master-port', '29713',
+         os.path.join(os.path.dirname(__file__), 'helpers',
+                      'dist_worker.py')],
+        capture_output=True, text=True, timeout=240, env=env)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    assert 'DIST_OK' in r.stdout

@@ -59,7 +59,7 @@ __device__ __forceinline__ void StageKMajor(const uint16_t* __restrict__ src,
                                             int64_t ld, int col0, int64_t k0,
                                             uint16_t* lds_base, int tid) {
   constexpr int kRG = BR / 8;  // 8-wide row groups per k row
-  if (tid >= BR) return;       // one thread per 8x8 block (8 k-chunks)
+  if (tid < 0 || tid >= BR) return;  // one thread per 8x8 block
   int r0 = (tid % kRG) * 8;
   int chunk = tid / kRG;       // k chunk of 8
   const uint16_t* g = src + (k0 + chunk * 8) * ld + col0 + r0;
@@ -83,7 +83,7 @@ __device__ __forceinline__ void StageKMajorSafe(
     const uint16_t* __restrict__ src, int64_t ld, int col0, int64_t k0,
     int64_t cols, int64_t K, uint16_t* lds_base, int tid) {
   constexpr int kRG = BR / 8;
-  if (tid >= BR) return;
+  if (tid < 0 || tid >= BR) return;
   int r0 = (tid % kRG) * 8;
   int chunk = tid / kRG;
   uint16_t v[8][8];
@@ -214,10 +214,13 @@ __launch_bounds__(256) __global__ void GemmBf16NT(
         StageSafe<BM>(A, lda, (int)m0, k0, M, K, a_tile(buf), tid);
     }
     if (B_KM) {
+      // When A also staged K-major, A used threads [0,BM) — give B the next
+      // BN threads so both operand stages run concurrently across waves.
+      int bt = A_KM ? tid - BM : tid;
       if (b_interior && kfull)
-        StageKMajor<BN>(B, ldb, (int)n0, k0, b_tile(buf), tid);
+        StageKMajor<BN>(B, ldb, (int)n0, k0, b_tile(buf), bt);
       else
-        StageKMajorSafe<BN>(B, ldb, (int)n0, k0, N, K, b_tile(buf), tid);
+        StageKMajorSafe<BN>(B, ldb, (int)n0, k0, N, K, b_tile(buf), bt);
     } else {
       if (b_interior && kfull)
         StageFast<BN>(B, ldb, (int)n0, k0, b_tile(buf), tid);

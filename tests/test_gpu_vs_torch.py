@@ -1,0 +1,113 @@
+"""HIP kernel numerics vs plain PyTorch fp32 references (CPU torch): the
+framework's GPU ops must agree with an independent implementation, not just
+our own CPU kernels."""
+import numpy as np
+import pytest
+
+import simple_tensorflow_amd as tf
+
+pytestmark = pytest.mark.gpu
+
+torch = pytest.importorskip('torch')
+
+
+@pytest.fixture(autouse=True)
+def fresh_graph():
+    tf.reset_default_graph()
+    yield
+
+
+def _run(t):
+    with tf.Session() as s:
+        return s.run(t)
+
+
+def test_matmul_bf16_vs_torch():
+    rng = np.random.RandomState(0)
+    a = rng.randn(192, 320).astype(np.float32)
+    b = rng.randn(320, 256).astype(np.float32)
+    got = _run(tf.matmul(tf.constant(a, dtype=tf.bfloat16),
+                         tf.constant(b, dtype=tf.bfloat16)))
+    # bf16 inputs: compare against torch bf16 matmul upcast to f32
+    want = (torch.from_numpy(a).bfloat16().float() @
+            torch.from_numpy(b).bfloat16().float()).numpy()
+    rel = np.abs(got - want) / (np.abs(want) + 1e-2)
+    assert np.percentile(rel, 99) < 0.05
+
+
+def test_conv2d_bf16_vs_torch():
+    rng = np.random.RandomState(1)
+    x = rng.randn(2, 16, 16, 8).astype(np.float32)
+    w = rng.randn(3, 3, 8, 16).astype(np.float32)
+    got = _run(tf.nn.conv2d(tf.constant(x, dtype=tf.bfloat16),
+                            tf.constant(w, dtype=tf.bfloat16),
+                            [1, 1, 1, 1], 'SAME'))
+    tx = torch.from_numpy(x).permute(0, 3, 1, 2)
+    tw = torch.from_numpy(w).permute(3, 2, 0, 1)
+    want = torch.nn.functional.conv2d(tx, tw, padding=1) \
+        .permute(0, 2, 3, 1).numpy()
+    assert np.abs(got - want).max() / (np.abs(want).max() + 1e-6) < 0.05
+
+
+def test_batch_norm_vs_torch():
+    from simple_tensorflow_amd.python.framework.ops import apply_op
+    rng = np.random.RandomState(2)
+    x = rng.randn(4, 8, 8, 32).astype(np.float32)
+    scale = rng.rand(32).astype(np.float32) + 0.5
+    offset = rng.randn(32).astype(np.float32)
+    y, mean, var, _ = apply_op('BatchNormMi',
+                               tf.constant(x, dtype=tf.bfloat16),
+                               tf.constant(scale), tf.constant(offset),
+                               epsilon=1e-4)
+    got_y, got_mean = _run([y, mean])
+    tx = torch.from_numpy(x).permute(0, 3, 1, 2)
+    want = torch.nn.functional.batch_norm(
+        tx, None, None, torch.from_numpy(scale),
+        torch.from_numpy(offset), training=True, eps=1e-4) \
+        .permute(0, 2, 3, 1).numpy()
+    assert np.abs(got_y - want).max() < 0.1  # bf16 rounding
+    np.testing.assert_allclose(got_mean, x.reshape(-1, 32).mean(0),
+                               rtol=1e-3, atol=1e-3)
+
+
+def test_softmax_xent_vs_torch():
+    rng = np.random.RandomState(3)
+    logits = rng.randn(64, 100).astype(np.float32)
+    labels = rng.randint(0, 100, 64).astype(np.int64)
+    loss = tf.nn.sparse_softmax_cross_entropy_with_logits(
+        labels=tf.constant(labels), logits=tf.constant(logits))
+    got = _run(loss)
+    want = torch.nn.functional.cross_entropy(
+        torch.from_numpy(logits), torch.from_numpy(labels),
+        reduction='none').numpy()
+    np.testing.assert_allclose(got, want, rtol=1e-4, atol=1e-4)
+
+
+def test_max_pool_grad_vs_torch():
+    rng = np.random.RandomState(4)
+    x = rng.randn(2, 8, 8, 16).astype(np.float32)
+    xt = tf.constant(x, dtype=tf.bfloat16)
+    y = tf.nn.max_pool(xt, [1, 2, 2, 1], [1, 2, 2, 1], 'VALID')
+    loss = tf.reduce_sum(tf.cast(y, tf.float32))
+    g = tf.gradients(loss, [xt])[0]
+    got = _run(g)
+    tx = torch.from_numpy(x).permute(0, 3, 1, 2).requires_grad_(True)
+    ty = torch.nn.functional.max_pool2d(tx, 2)
+    ty.sum().backward()
+    want = tx.grad.permute(0, 2, 3, 1).numpy()
+    np.testing.assert_allclose(got, want, atol=1e-2)
+
+
+def test_reduce_and_elementwise_vs_torch():
+    rng = np.random.RandomState(5)
+    x = rng.randn(1000, 33).astype(np.float32)
+    got = _run([tf.reduce_sum(tf.constant(x), axis=0),
+                tf.tanh(tf.constant(x[:5])),
+                tf.sigmoid(tf.constant(x[:5]))])
+    np.testing.assert_allclose(got[0], torch.from_numpy(x).sum(0).numpy(),
+                               rtol=1e-4, atol=1e-3)
+    np.testing.assert_allclose(got[1], np.tanh(x[:5]), rtol=1e-5,
+                               atol=1e-5)
+    np.testing.assert_allclose(got[2],
+                               torch.sigmoid(torch.from_numpy(x[:5]))
+                               .numpy(), rtol=1e-5, atol=1e-5)

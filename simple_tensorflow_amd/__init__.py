@@ -199,6 +199,17 @@ argmin = _math_ops.argmin
 select = _math_ops.select
 range = _math_ops.range  # noqa: A001
 cumsum = _math_ops.cumsum
+realdiv = _math_ops.realdiv
+div_no_nan = _math_ops.div_no_nan
+norm = _math_ops.norm
+tensordot = _math_ops.tensordot
+trace = _math_ops.trace
+placeholder_with_default = _array_ops.placeholder_with_default
+eye = _array_ops.eye
+meshgrid = _array_ops.meshgrid
+reverse = _array_ops.reverse
+reverse_v2 = _array_ops.reverse_v2
+case = _control_flow_ops.case
 global_norm = _clip_ops.global_norm
 clip_by_value = _clip_ops.clip_by_value
 clip_by_norm = _clip_ops.clip_by_norm

@@ -218,6 +218,7 @@ REGISTER_OP("MergeV2Checkpoints").Input("checkpoint_prefixes: string").Input("de
 // ------------------------------- queues ------------------------------------
 // TensorArray family (reference core/ops/data_flow_ops.cc:1080; resource
 // handles are session-scoped strings here, like the queue ops above).
+REGISTER_OP("PlaceholderWithDefault").Input("input: dtype").Output("output: dtype").Attr("dtype: type").Attr("shape: shape = []");
 REGISTER_OP("SparseToDense").Input("sparse_indices: Tindices").Input("output_shape: Tindices").Input("sparse_values: T").Input("default_value: T").Output("dense: T").Attr("validate_indices: bool = true").Attr("T: type").Attr("Tindices: {int32, int64}");
 REGISTER_OP("ResizeBilinear").Input("images: T").Input("size: int32").Output("resized_images: float").Attr("T: {float}").Attr("align_corners: bool = false");
 REGISTER_OP("ResizeBilinearGrad").Input("grads: float").Input("original_image: T").Output("output: T").Attr("T: {float}").Attr("align_corners: bool = false");

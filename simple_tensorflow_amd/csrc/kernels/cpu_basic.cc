@@ -81,6 +81,8 @@ class IdentityOp : public OpKernel {
 };
 REGISTER_KERNEL_BUILDER(Name("Identity").Device(DEVICE_CPU), IdentityOp);
 REGISTER_KERNEL_BUILDER(Name("Identity").Device(DEVICE_GPU), IdentityOp);
+REGISTER_KERNEL_BUILDER(Name("PlaceholderWithDefault").Device(DEVICE_CPU), IdentityOp);
+REGISTER_KERNEL_BUILDER(Name("PlaceholderWithDefault").Device(DEVICE_GPU), IdentityOp);
 REGISTER_KERNEL_BUILDER(Name("StopGradient").Device(DEVICE_CPU), IdentityOp);
 REGISTER_KERNEL_BUILDER(Name("StopGradient").Device(DEVICE_GPU), IdentityOp);
 REGISTER_KERNEL_BUILDER(Name("PreventGradient").Device(DEVICE_CPU), IdentityOp);

@@ -354,3 +354,52 @@ def unsorted_segment_sum(data, segment_ids, num_segments, name=None):
                     convert_to_tensor(segment_ids),
                     convert_to_tensor(num_segments, dtype=dtypes.int32),
                     name=name)
+
+
+def placeholder_with_default(input, shape=None, name=None):  # pylint: disable=redefined-builtin
+    t = apply_op('PlaceholderWithDefault', convert_to_tensor(input),
+                 shape=list(shape) if shape is not None else [], name=name)
+    if shape is not None:
+        t.set_shape(list(shape))
+    return t
+
+
+def eye(num_rows, num_columns=None, dtype=dtypes.float32, name=None):
+    m = num_columns if num_columns is not None else num_rows
+    arr = np.eye(num_rows, m).astype(dtypes.as_dtype(dtype).as_numpy_dtype)
+    return ops.constant(arr, name=name or 'eye')
+
+
+def meshgrid(*args, **kwargs):
+    indexing = kwargs.get('indexing', 'xy')
+    tensors = [convert_to_tensor(a) for a in args]
+    if len(tensors) != 2:
+        raise NotImplementedError('meshgrid supports 2 inputs')
+    x, y = tensors
+    nx, ny = x._shape[0], y._shape[0]
+    if indexing == 'xy':
+        X = tile(reshape(x, [1, nx]), [ny, 1])
+        Y = tile(reshape(y, [ny, 1]), [1, nx])
+    else:
+        X = tile(reshape(x, [nx, 1]), [1, ny])
+        Y = tile(reshape(y, [1, ny]), [nx, 1])
+    return X, Y
+
+
+def reverse(tensor, axis, name=None):
+    t = convert_to_tensor(tensor)
+    if isinstance(axis, (list, tuple)):
+        for a in axis:
+            t = reverse(t, a)
+        return t
+    a = int(axis) % len(t._shape)
+    n = t._shape[a]
+    idx = ops.constant(list(range(n - 1, -1, -1)), dtype=dtypes.int32)
+    if a == 0:
+        return gather(t, idx)
+    perm = list(range(len(t._shape)))
+    perm[0], perm[a] = perm[a], perm[0]
+    return transpose(gather(transpose(t, perm), idx), perm)
+
+
+reverse_v2 = reverse

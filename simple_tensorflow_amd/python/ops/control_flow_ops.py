@@ -178,3 +178,19 @@ def Assert(condition, data, summarize=3, name=None):  # noqa: N802
     data = [convert_to_tensor(d) for d in data]
     g = ops.get_default_graph()
     return apply_op('Assert', condition, data, summarize=summarize, name=name)
+
+
+def case(pred_fn_pairs, default, exclusive=False, name=None):
+    """tf.case via nested cond (reference control_flow_ops.case)."""
+    if isinstance(pred_fn_pairs, dict):
+        pairs = list(pred_fn_pairs.items())
+    else:
+        pairs = list(pred_fn_pairs)
+
+    def build(i):
+        if i >= len(pairs):
+            return default()
+        pred, fn = pairs[i]
+        return cond(pred, fn, lambda: build(i + 1))
+
+    return build(0)

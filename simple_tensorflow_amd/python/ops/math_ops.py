@@ -256,3 +256,73 @@ def global_norm(t_list, name=None):
     from simple_tensorflow_amd.python.ops import array_ops
     halves = [l2_loss(t) for t in t_list if t is not None]
     return sqrt(multiply(add_n(halves), 2.0))
+
+
+def realdiv(x, y, name=None):
+    return _binary('RealDiv', x, y, name)
+
+
+def div_no_nan(x, y, name=None):
+    x = convert_to_tensor(x)
+    y = convert_to_tensor(y)
+    safe = select(equal(y, ops.constant(0, dtype=y.dtype)),
+                  __import__('simple_tensorflow_amd.python.ops.array_ops',
+                             fromlist=['x']).ones_like(y), y)
+    out = select(equal(y, ops.constant(0, dtype=y.dtype)),
+                 __import__('simple_tensorflow_amd.python.ops.array_ops',
+                            fromlist=['x']).zeros_like(x), divide(x, safe))
+    return out
+
+
+def norm(tensor, ord='euclidean', axis=None, keep_dims=False, name=None):
+    t = convert_to_tensor(tensor)
+    if ord in ('euclidean', 2):
+        return sqrt(reduce_sum(square(t), axis=axis, keep_dims=keep_dims))
+    if ord == 1:
+        return reduce_sum(abs(t), axis=axis, keep_dims=keep_dims)
+    if ord in ('inf', float('inf')):
+        return reduce_max(abs(t), axis=axis, keep_dims=keep_dims)
+    raise ValueError('unsupported norm ord %r' % (ord,))
+
+
+def tensordot(a, b, axes, name=None):
+    from simple_tensorflow_amd.python.ops import array_ops
+    a = convert_to_tensor(a)
+    b = convert_to_tensor(b)
+    if isinstance(axes, int):
+        a_axes = list(_bi.range(len(a._shape) - axes, len(a._shape)))
+        b_axes = list(_bi.range(axes))
+    else:
+        a_axes, b_axes = axes
+        if isinstance(a_axes, int):
+            a_axes, b_axes = [a_axes], [b_axes]
+    a_free = [i for i in _bi.range(len(a._shape)) if i not in a_axes]
+    b_free = [i for i in _bi.range(len(b._shape)) if i not in b_axes]
+    ta = array_ops.transpose(a, a_free + list(a_axes))
+    tb = array_ops.transpose(b, list(b_axes) + b_free)
+    m = 1
+    for i in a_free:
+        m *= a._shape[i]
+    k = 1
+    for i in a_axes:
+        k *= a._shape[i]
+    n = 1
+    for i in b_free:
+        n *= b._shape[i]
+    out = matmul(array_ops.reshape(ta, [m, k]),
+                 array_ops.reshape(tb, [k, n]))
+    return array_ops.reshape(
+        out, [a._shape[i] for i in a_free] + [b._shape[i] for i in b_free])
+
+
+def trace(x, name=None):
+    from simple_tensorflow_amd.python.ops import array_ops
+    x = convert_to_tensor(x)
+    n = min(x._shape[-2], x._shape[-1])
+    idx = ops.constant([[i, i] for i in _bi.range(n)], dtype=dtypes.int64)
+    # gather diagonal via reshape trick: flat index i*(cols+1)
+    cols = x._shape[-1]
+    flat = array_ops.reshape(x, [-1])
+    diag_idx = ops.constant([i * (cols + 1) for i in _bi.range(n)],
+                            dtype=dtypes.int32)
+    return reduce_sum(array_ops.gather(flat, diag_idx))

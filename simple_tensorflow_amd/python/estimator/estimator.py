@@ -148,3 +148,62 @@ class Estimator(object):
             else:
                 for row in preds:
                     yield row
+
+
+def _default_serving_input_fn_placeholder(features_spec):
+    import simple_tensorflow_amd as tf
+    return {k: tf.placeholder(dtype, shape, name=k)
+            for k, (dtype, shape) in features_spec.items()}
+
+
+class ServingInputReceiver(object):
+    def __init__(self, features, receiver_tensors):
+        self.features = features
+        self.receiver_tensors = receiver_tensors
+
+
+def build_raw_serving_input_receiver_fn(features):
+    """features: dict name -> placeholder-like Tensor spec (dtype, shape)."""
+    def fn():
+        import simple_tensorflow_amd as tf
+        phs = {k: tf.placeholder(v[0], v[1], name=k)
+               for k, v in features.items()}
+        return ServingInputReceiver(phs, phs)
+    return fn
+
+
+def _export_savedmodel(self, export_dir_base, serving_input_receiver_fn):
+    """Estimator.export_savedmodel (reference estimator export.py): builds
+    the PREDICT graph, restores the latest checkpoint, writes a SavedModel
+    with a serving_default signature."""
+    import os as _os
+    import time as _time
+    import simple_tensorflow_amd as tf
+    from simple_tensorflow_amd.python import saved_model as sm
+    g = ops.Graph()
+    with g.as_default():
+        tf.train.get_or_create_global_step()
+        receiver = serving_input_receiver_fn()
+        spec = self._call_model_fn(receiver.features, None,
+                                   ModeKeys.PREDICT)
+        sess = self._restore_session(g)
+        export_dir = _os.path.join(export_dir_base,
+                                   str(int(_time.time())))
+        b = sm.SavedModelBuilder(export_dir)
+        preds = spec.predictions
+        if not isinstance(preds, dict):
+            preds = {'output': preds}
+        sig = sm.predict_signature_def(inputs=receiver.receiver_tensors,
+                                       outputs=preds)
+        b.add_meta_graph_and_variables(
+            sess, [sm.tag_constants.SERVING],
+            signature_def_map={
+                sm.signature_constants.DEFAULT_SERVING_SIGNATURE_DEF_KEY:
+                    sig})
+        b.save()
+        sess.close()
+    return export_dir
+
+
+Estimator.export_savedmodel = _export_savedmodel
+Estimator.export_saved_model = _export_savedmodel

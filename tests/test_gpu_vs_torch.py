@@ -84,8 +84,14 @@ def test_softmax_xent_vs_torch():
 
 
 def test_max_pool_grad_vs_torch():
+    # bf16-exact values with a unique max per 2x2 window: coarse random
+    # value per window (5 bits) + in-window position offset (3 frac bits)
     rng = np.random.RandomState(4)
-    x = rng.randn(2, 8, 8, 16).astype(np.float32)
+    wr = rng.randint(0, 32, (2, 4, 4, 16)).astype(np.float32)
+    x = np.zeros((2, 8, 8, 16), dtype=np.float32)
+    for di in range(2):
+        for dj in range(2):
+            x[:, di::2, dj::2, :] = wr + (di * 2 + dj) * 0.125
     xt = tf.constant(x, dtype=tf.bfloat16)
     y = tf.nn.max_pool(xt, [1, 2, 2, 1], [1, 2, 2, 1], 'VALID')
     loss = tf.reduce_sum(tf.cast(y, tf.float32))

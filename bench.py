@@ -18,6 +18,10 @@ _local_rank = int(os.environ.get('LOCAL_RANK', 0))
 _world = int(os.environ.get('WORLD_SIZE', 1))
 if _world > 1:
     os.environ.setdefault('HIP_VISIBLE_DEVICES', str(_local_rank))
+    # RCCL collectives inside hipGraph capture are not yet validated on
+    # multi-GPU nodes (replay of the communicator's internal sequencing);
+    # run the distributed step eagerly until round 2 proves it out.
+    os.environ.setdefault('STF_NO_HIPGRAPH', '1')
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 

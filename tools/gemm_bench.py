@@ -40,6 +40,12 @@ def bench_shape(sess_cache, m, n, k, iters=20):
 
 
 def main():
+    import sys
+    if len(sys.argv) > 1:
+        m, n, k = (int(x) for x in sys.argv[1].split(','))
+        for _ in range(3):
+            bench_shape(None, m, n, k)
+        return
     shapes = [
         (4096, 4096, 4096),
         (8192, 8192, 8192),

@@ -149,8 +149,12 @@ __device__ __forceinline__ void StageSafe(const uint16_t* __restrict__ src,
 //  lda/ldb = that row stride) and transposed during LDS staging — this is
 //  how dW (= col^T · dy) and NN matmuls run without separate transpose
 //  kernels. With *_KM false the operand is [M, K] / [N, K] row-major (NT).
+// DBUF=false: single-buffered LDS (half the footprint -> 2x the resident
+// blocks per CU). Chosen for short K loops (kt <= 4), where double
+// buffering cannot warm up and cross-block overlap hides latency better.
 template <int WAVES_M, int WAVES_N, int WM, int WN, bool A_KM, bool B_KM,
-          bool OUT_BF16, bool FUSE_RELU, bool SPLITK = false>
+          bool OUT_BF16, bool FUSE_RELU, bool SPLITK = false,
+          bool DBUF = true>
 __launch_bounds__(256) __global__ void GemmBf16NT(
     const uint16_t* __restrict__ A,
     const uint16_t* __restrict__ B,
@@ -163,11 +167,14 @@ __launch_bounds__(256) __global__ void GemmBf16NT(
   constexpr int BK = 64;
 
   __shared__ __attribute__((aligned(16)))
-      uint16_t lds[2 * (BM + BN) * BK];
+      uint16_t lds[(DBUF ? 2 : 1) * (BM + BN) * BK];
   // buffer layout: [A0][A1][B0][B1]; pick by integer offset (an initialized
   // array of LDS pointers does not compile on gfx950)
-  auto a_tile = [&](int buf) { return lds + buf * BM * BK; };
-  auto b_tile = [&](int buf) { return lds + 2 * BM * BK + buf * BN * BK; };
+  constexpr int NBUF = DBUF ? 2 : 1;
+  auto a_tile = [&](int buf) { return lds + (DBUF ? buf : 0) * BM * BK; };
+  auto b_tile = [&](int buf) {
+    return lds + NBUF * BM * BK + (DBUF ? buf : 0) * BN * BK;
+  };
 
   int nbm = (int)((M + BM - 1) / BM);
   int nbn = (int)((N + BN - 1) / BN);
@@ -235,8 +242,9 @@ __launch_bounds__(256) __global__ void GemmBf16NT(
 
   int cur = 0;
   for (int kt = kt_begin; kt < kt_count; ++kt) {
-    // issue next tile's loads first (overlap with this tile's compute)
-    if (kt + 1 < kt_count) {
+    // issue next tile's loads first (overlap with this tile's compute);
+    // single-buffer mode must finish compute before restaging instead.
+    if (DBUF && kt + 1 < kt_count) {
       stage(cur ^ 1, (int64_t)(kt + 1) * BK);
     }
 
@@ -265,7 +273,12 @@ __launch_bounds__(256) __global__ void GemmBf16NT(
           acc[i][j] = mfma_bf16(afrag[i], bfrag[j], acc[i][j]);
     }
     __syncthreads();
-    cur ^= 1;
+    if (DBUF) {
+      cur ^= 1;
+    } else if (kt + 1 < kt_count) {
+      stage(0, (int64_t)(kt + 1) * BK);
+      __syncthreads();
+    }
   }
 
   // ---- epilogue ----
@@ -305,20 +318,23 @@ hipError_t LaunchVariant(const uint16_t* A, const uint16_t* B, void* C,
                          const float* bias, int64_t M, int64_t N, int64_t K,
                          int64_t lda, int64_t ldb, float beta,
                          hipStream_t stream) {
+  bool short_k = K <= 4 * 64;
   auto launch = [&](auto kern, int BM, int BN) {
     int64_t blocks = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
     hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0, stream, A,
                        B, C, bias, M, N, K, lda, ldb, beta, 1);
   };
+#define STF_PICK(WM_, WN_, TM, TN)                                           do {                                                                         if (short_k)                                                                 launch(GemmBf16NT<2, 2, WM_, WN_, A_KM, B_KM, OUT_BF16, FUSE_RELU,                           false, false>,                                                  TM, TN);                                                          else                                                                         launch(GemmBf16NT<2, 2, WM_, WN_, A_KM, B_KM, OUT_BF16, FUSE_RELU>,               TM, TN);                                                        } while (0)
   if (N >= 128 && M >= 128) {
-    launch(GemmBf16NT<2, 2, 4, 4, A_KM, B_KM, OUT_BF16, FUSE_RELU>, 128, 128);
+    STF_PICK(4, 4, 128, 128);
   } else if (N >= 128) {
-    launch(GemmBf16NT<2, 2, 2, 4, A_KM, B_KM, OUT_BF16, FUSE_RELU>, 64, 128);
+    STF_PICK(2, 4, 64, 128);
   } else if (M >= 128) {
-    launch(GemmBf16NT<2, 2, 4, 2, A_KM, B_KM, OUT_BF16, FUSE_RELU>, 128, 64);
+    STF_PICK(4, 2, 128, 64);
   } else {
-    launch(GemmBf16NT<2, 2, 2, 2, A_KM, B_KM, OUT_BF16, FUSE_RELU>, 64, 64);
+    STF_PICK(2, 2, 64, 64);
   }
+#undef STF_PICK
   return hipGetLastError();
 }
 

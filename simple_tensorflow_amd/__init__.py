@@ -368,6 +368,7 @@ from simple_tensorflow_amd.python.ops import sparse_ops as _sparse_ops  # noqa: 
 SparseTensor = _sparse_ops.SparseTensor
 sparse_to_dense = _sparse_ops.sparse_to_dense
 sparse_tensor_to_dense = _sparse_ops.sparse_tensor_to_dense
+from simple_tensorflow_amd.python.layers import layers  # noqa: E402,F401
 from simple_tensorflow_amd.python.platform import app  # noqa: E402,F401
 from simple_tensorflow_amd.python.platform import gfile  # noqa: E402,F401
 from simple_tensorflow_amd.python.platform import tf_logging as logging  # noqa: E402,F401

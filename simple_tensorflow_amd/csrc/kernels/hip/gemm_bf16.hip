@@ -318,7 +318,8 @@ hipError_t LaunchVariant(const uint16_t* A, const uint16_t* B, void* C,
                          const float* bias, int64_t M, int64_t N, int64_t K,
                          int64_t lda, int64_t ldb, float beta,
                          hipStream_t stream) {
-  bool short_k = K <= 4 * 64;
+  static const bool no_sb = getenv("STF_GEMM_NO_SB") != nullptr;
+  bool short_k = !no_sb && K <= 4 * 64;
   auto launch = [&](auto kern, int BM, int BN) {
     int64_t blocks = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
     hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0, stream, A,

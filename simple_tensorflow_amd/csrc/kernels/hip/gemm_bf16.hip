@@ -282,6 +282,41 @@ __launch_bounds__(256) __global__ void GemmBf16NT(
   }
 
   // ---- epilogue ----
+  // Fast path for plain bf16 output on interior tiles: the fragment layout
+  // scatters 2-byte stores (quarter-wave 32B segments, ~1.2 TB/s effective
+  // on skinny-K streaming GEMMs) — stage the tile through LDS (the A/B
+  // buffers are dead now) and emit contiguous 16B stores instead.
+  if (OUT_BF16 && !SPLITK && beta == 0.f && a_interior && b_interior &&
+      (N & 7) == 0) {
+    __syncthreads();  // K-loop reads of LDS are done; reuse as C staging
+    uint16_t* cbuf = lds;  // [BM][BN] bf16
+#pragma unroll
+    for (int i = 0; i < WM; ++i) {
+#pragma unroll
+      for (int j = 0; j < WN; ++j) {
+        int col = wc * WN * 16 + j * 16 + (lane & 15);
+        float bv = bias ? bias[n0 + col] : 0.f;
+#pragma unroll
+        for (int rgi = 0; rgi < 4; ++rgi) {
+          int row = wr * WM * 16 + i * 16 + (lane >> 4) * 4 + rgi;
+          float v = acc[i][j][rgi] + bv;
+          if (FUSE_RELU) v = v > 0.f ? v : 0.f;
+          cbuf[row * BN + col] = f32_to_bf16(v);
+        }
+      }
+    }
+    __syncthreads();
+    constexpr int kChunks = BM * BN / 8;  // 16B chunks
+#pragma unroll
+    for (int p = 0; p < kChunks / 256; ++p) {
+      int sidx = (p * 256 + tid) * 8;
+      int r = sidx / BN;
+      int c = sidx % BN;
+      *(ulong2*)((uint16_t*)C + (m0 + r) * N + n0 + c) =
+          *(ulong2*)(cbuf + sidx);
+    }
+    return;
+  }
 #pragma unroll
   for (int i = 0; i < WM; ++i) {
 #pragma unroll

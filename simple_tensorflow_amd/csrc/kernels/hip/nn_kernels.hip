@@ -808,12 +808,9 @@ hipError_t stf_max_pool_bwd(int dtype, const void* x, const void* dy,
   if (dtype == 0)
     hipLaunchKernelGGL((MaxPoolGradKernel<float>), grid, dim3(256), 0, stream,
                        (const float*)x, (const float*)dy, dx_f32, g, total);
-  else if (C % 8 == 0) {
-    int64_t t8 = total / 8;
-    hipLaunchKernelGGL(MaxPoolGradKernelV8, ElemwiseGrid(t8, 256, 1),
-                       dim3(256), 0, stream, (const __bf16*)x,
-                       (const __bf16*)dy, dx_f32, g, t8);
-  } else
+  // (an 8-channel vectorized variant measured 2x SLOWER here — the per-
+  // thread argmax state spills; the scalar form wins)
+  else
     hipLaunchKernelGGL((MaxPoolGradKernel<__bf16>), grid, dim3(256), 0, stream,
                        (const __bf16*)x, (const __bf16*)dy, dx_f32, g, total);
   return hipGetLastError();

@@ -78,42 +78,6 @@ __device__ __forceinline__ void StageKMajor(const uint16_t* __restrict__ src,
   }
 }
 
-// Pipelined staging for the split-K dW variants (small accumulators, so
-// the 64 staging VGPRs fit): KmLoad issues the 8 row loads, KmStore
-// transposes+writes LDS after the tile's MFMA work, hiding global latency.
-template <int BR>
-__device__ __forceinline__ bool KmLoad(const uint16_t* __restrict__ src,
-                                       int64_t ld, int col0, int64_t k0,
-                                       int tid, uint16_t (*v)[8]) {
-  constexpr int kRG = BR / 8;
-  if (tid < 0 || tid >= BR) return false;
-  int r0 = (tid % kRG) * 8;
-  int chunk = tid / kRG;
-  const uint16_t* g = src + (k0 + chunk * 8) * ld + col0 + r0;
-#pragma unroll
-  for (int kk = 0; kk < 8; ++kk)
-    *(ulong2*)v[kk] = *(const ulong2*)(g + (int64_t)kk * ld);
-  return true;
-}
-
-template <int BR>
-__device__ __forceinline__ void KmStore(uint16_t* lds_base, int tid,
-                                        uint16_t (*v)[8]) {
-  constexpr int kRG = BR / 8;
-  if (tid < 0 || tid >= BR) return;
-  int r0 = (tid % kRG) * 8;
-  int chunk = tid / kRG;
-#pragma unroll
-  for (int e = 0; e < 8; ++e) {
-    uint16_t out[8];
-#pragma unroll
-    for (int kk = 0; kk < 8; ++kk) out[kk] = v[kk][e];
-    int r = r0 + e;
-    *(ulong2*)(lds_base + (int64_t)(r * 8 + (chunk ^ Swz(r))) * 8) =
-        *(ulong2*)out;
-  }
-}
-
 template <int BR>
 __device__ __forceinline__ void StageKMajorSafe(
     const uint16_t* __restrict__ src, int64_t ld, int col0, int64_t k0,
@@ -276,43 +240,12 @@ __launch_bounds__(256) __global__ void GemmBf16NT(
   stage(0, (int64_t)kt_begin * BK);
   __syncthreads();
 
-  constexpr bool PIPE_KM = SPLITK && (A_KM || B_KM) && (WM * WN <= 8);
   int cur = 0;
   for (int kt = kt_begin; kt < kt_count; ++kt) {
     // issue next tile's loads first (overlap with this tile's compute);
     // single-buffer mode must finish compute before restaging instead.
-    uint16_t ra[8][8], rb[8][8];
-    bool ra_live = false, rb_live = false;
-    int btid = A_KM ? tid - BM : tid;
     if (DBUF && kt + 1 < kt_count) {
-      if (PIPE_KM) {
-        int64_t k0 = (int64_t)(kt + 1) * BK;
-        bool kfull = k0 + BK <= K;
-        if (A_KM) {
-          if (a_interior && kfull)
-            ra_live = KmLoad<BM>(A, lda, (int)m0, k0, tid, ra);
-          else
-            StageKMajorSafe<BM>(A, lda, (int)m0, k0, M, K,
-                                a_tile(cur ^ 1), tid);
-        } else if (a_interior && kfull) {
-          StageFast<BM>(A, lda, (int)m0, k0, a_tile(cur ^ 1), tid);
-        } else {
-          StageSafe<BM>(A, lda, (int)m0, k0, M, K, a_tile(cur ^ 1), tid);
-        }
-        if (B_KM) {
-          if (b_interior && kfull)
-            rb_live = KmLoad<BN>(B, ldb, (int)n0, k0, btid, rb);
-          else
-            StageKMajorSafe<BN>(B, ldb, (int)n0, k0, N, K,
-                                b_tile(cur ^ 1), btid);
-        } else if (b_interior && kfull) {
-          StageFast<BN>(B, ldb, (int)n0, k0, b_tile(cur ^ 1), tid);
-        } else {
-          StageSafe<BN>(B, ldb, (int)n0, k0, N, K, b_tile(cur ^ 1), tid);
-        }
-      } else {
-        stage(cur ^ 1, (int64_t)(kt + 1) * BK);
-      }
+      stage(cur ^ 1, (int64_t)(kt + 1) * BK);
     }
 
     // compute on current tile: 2 mfma K-steps of 32
@@ -338,10 +271,6 @@ __launch_bounds__(256) __global__ void GemmBf16NT(
 #pragma unroll
         for (int j = 0; j < WN; ++j)
           acc[i][j] = mfma_bf16(afrag[i], bfrag[j], acc[i][j]);
-    }
-    if (PIPE_KM) {
-      if (ra_live) KmStore<BM>(a_tile(cur ^ 1), tid, ra);
-      if (rb_live) KmStore<BN>(b_tile(cur ^ 1), btid, rb);
     }
     __syncthreads();
     if (DBUF) {

@@ -475,4 +475,81 @@ class FusedBatchNormGradOp : public OpKernel {
 };
 REGISTER_KERNEL_BUILDER(Name("FusedBatchNormGrad").Device(DEVICE_CPU).TypeConstraint<float>("T"), FusedBatchNormGradOp);
 
+
+// ---------------------- depthwise conv (CPU reference) ----------------------
+// (reference core/kernels/depthwise_conv_op.cc; plain loops, f32 accum)
+namespace {
+
+struct DwShape {
+  int64_t N, H, W, C, R, S, sh, sw, ph, pw, P, Q, mult;
+};
+
+static Status DwShapeFrom(const TensorShape& x, const TensorShape& f,
+                          const std::vector<int64_t>& strides,
+                          const std::string& padding, DwShape* d) {
+  d->N = x.dim_size(0); d->H = x.dim_size(1); d->W = x.dim_size(2);
+  d->C = x.dim_size(3);
+  d->R = f.dim_size(0); d->S = f.dim_size(1); d->mult = f.dim_size(3);
+  d->sh = strides[1]; d->sw = strides[2];
+  if (padding == "SAME") {
+    d->P = (d->H + d->sh - 1) / d->sh;
+    d->Q = (d->W + d->sw - 1) / d->sw;
+    d->ph = std::max<int64_t>(0, (d->P - 1) * d->sh + d->R - d->H) / 2;
+    d->pw = std::max<int64_t>(0, (d->Q - 1) * d->sw + d->S - d->W) / 2;
+  } else {
+    d->P = (d->H - d->R) / d->sh + 1;
+    d->Q = (d->W - d->S) / d->sw + 1;
+    d->ph = d->pw = 0;
+  }
+  return Status::OK();
+}
+
+template <typename T>
+class CpuDepthwiseConvOp : public OpKernel {
+ public:
+  explicit CpuDepthwiseConvOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("strides", &strides_);
+    c->GetAttr("padding", &padding_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    const Tensor& w = ctx->input(1);
+    DwShape d;
+    OP_REQUIRES_OK(ctx, DwShapeFrom(x.shape(), w.shape(), strides_,
+                                    padding_, &d));
+    Tensor* y = ctx->allocate_output(
+        0, TensorShape({d.N, d.P, d.Q, d.C * d.mult}));
+    const T* xp = x.flat<T>();
+    const T* wp = w.flat<T>();
+    T* yp = y->flat<T>();
+    int64_t CM = d.C * d.mult;
+    for (int64_t n = 0; n < d.N; ++n)
+      for (int64_t p = 0; p < d.P; ++p)
+        for (int64_t q = 0; q < d.Q; ++q)
+          for (int64_t c = 0; c < d.C; ++c)
+            for (int64_t m = 0; m < d.mult; ++m) {
+              float acc = 0.f;
+              for (int64_t r = 0; r < d.R; ++r) {
+                int64_t ih = p * d.sh - d.ph + r;
+                if (ih < 0 || ih >= d.H) continue;
+                for (int64_t s2 = 0; s2 < d.S; ++s2) {
+                  int64_t iw = q * d.sw - d.pw + s2;
+                  if (iw < 0 || iw >= d.W) continue;
+                  acc += (float)xp[((n * d.H + ih) * d.W + iw) * d.C + c] *
+                         (float)wp[((r * d.S + s2) * d.C + c) * d.mult + m];
+                }
+              }
+              yp[((n * d.P + p) * d.Q + q) * CM + c * d.mult + m] = (T)acc;
+            }
+  }
+
+ private:
+  std::vector<int64_t> strides_;
+  std::string padding_;
+};
+REGISTER_KERNEL_BUILDER(Name("DepthwiseConv2dNative").Device(DEVICE_CPU).TypeConstraint<float>("T"), CpuDepthwiseConvOp<float>);
+REGISTER_KERNEL_BUILDER(Name("DepthwiseConv2dNative").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), CpuDepthwiseConvOp<bfloat16>);
+
+}  // namespace
+
 }  // namespace stf

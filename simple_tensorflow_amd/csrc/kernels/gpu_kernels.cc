@@ -27,6 +27,15 @@ hipError_t stf_gemm_bf16_splitk(const void*, const void*, void*, int64_t,
                                 int, hipStream_t);
 hipError_t stf_gemm_f32_nt(const void*, const void*, void*, int64_t, int64_t,
                            int64_t, hipStream_t);
+hipError_t stf_depthwise_fwd(const void*, const void*, void*, int, int, int,
+                             int, int, int, int, int, int, int, int, int,
+                             int, hipStream_t);
+hipError_t stf_depthwise_bwd_input(const void*, const void*, void*, int, int,
+                                   int, int, int, int, int, int, int, int,
+                                   int, int, int, hipStream_t);
+hipError_t stf_depthwise_bwd_filter(const void*, const void*, float*, int,
+                                    int, int, int, int, int, int, int, int,
+                                    int, int, int, int, hipStream_t);
 hipError_t stf_unary(int, int, const void*, void*, int64_t, hipStream_t);
 hipError_t stf_binary(int, int, const void*, const void*, void*, int64_t,
                       hipStream_t);
@@ -715,6 +724,112 @@ class GpuConv2DBackpropFilterOp : public OpKernel {
   std::string padding_;
 };
 REGISTER_KERNEL_BUILDER(Name("Conv2DBackpropFilter").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T").HostMemory("filter_sizes"), GpuConv2DBackpropFilterOp);
+
+// ---------------------------------------------------------------------------
+// depthwise conv (reference depthwise_conv_op_gpu.cu.cc family)
+// ---------------------------------------------------------------------------
+static Status DwGeomFromShapes(const TensorShape& x_shape,
+                               const TensorShape& f_shape,
+                               const std::vector<int64_t>& strides,
+                               const std::string& padding, GpuConvGeom* g) {
+  STF_RETURN_IF_ERROR(
+      GetConvGeom(x_shape, f_shape, strides, padding, g));
+  return Status::OK();
+}
+
+class GpuDepthwiseConvOp : public OpKernel {
+ public:
+  explicit GpuDepthwiseConvOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("strides", &strides_);
+    c->GetAttr("padding", &padding_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    const Tensor& w = ctx->input(1);  // [R, S, C, mult]
+    GpuConvGeom g;
+    OP_REQUIRES_OK(ctx, DwGeomFromShapes(x.shape(), w.shape(), strides_,
+                                         padding_, &g));
+    int mult = (int)w.dim_size(3);
+    Tensor* y = ctx->allocate_output(
+        0, TensorShape({g.N, g.P, g.Q, g.C * mult}));
+    OP_HIP_OK(ctx, stf_depthwise_fwd(
+                       x.raw_data(), w.raw_data(), y->raw_data(), (int)g.N,
+                       (int)g.H, (int)g.W, (int)g.C, (int)g.R, (int)g.S,
+                       (int)g.sh, (int)g.sw, (int)g.ph, (int)g.pw, (int)g.P,
+                       (int)g.Q, mult, GPU_STREAM(ctx)));
+  }
+
+ private:
+  std::vector<int64_t> strides_;
+  std::string padding_;
+};
+REGISTER_KERNEL_BUILDER(Name("DepthwiseConv2dNative").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuDepthwiseConvOp);
+
+class GpuDepthwiseConvBackpropInputOp : public OpKernel {
+ public:
+  explicit GpuDepthwiseConvBackpropInputOp(OpKernelConstruction* c)
+      : OpKernel(c) {
+    c->GetAttr("strides", &strides_);
+    c->GetAttr("padding", &padding_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    auto in_sizes = IntVector(ctx->input(0));
+    const Tensor& w = ctx->input(1);
+    const Tensor& dy = ctx->input(2);
+    TensorShape x_shape(in_sizes);
+    GpuConvGeom g;
+    OP_REQUIRES_OK(ctx, DwGeomFromShapes(x_shape, w.shape(), strides_,
+                                         padding_, &g));
+    int mult = (int)w.dim_size(3);
+    Tensor* dx = ctx->allocate_output(0, x_shape);
+    OP_HIP_OK(ctx, stf_depthwise_bwd_input(
+                       dy.raw_data(), w.raw_data(), dx->raw_data(), (int)g.N,
+                       (int)g.H, (int)g.W, (int)g.C, (int)g.R, (int)g.S,
+                       (int)g.sh, (int)g.sw, (int)g.ph, (int)g.pw, (int)g.P,
+                       (int)g.Q, mult, GPU_STREAM(ctx)));
+  }
+
+ private:
+  std::vector<int64_t> strides_;
+  std::string padding_;
+};
+REGISTER_KERNEL_BUILDER(Name("DepthwiseConv2dNativeBackpropInput").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T").HostMemory("input_sizes"), GpuDepthwiseConvBackpropInputOp);
+
+class GpuDepthwiseConvBackpropFilterOp : public OpKernel {
+ public:
+  explicit GpuDepthwiseConvBackpropFilterOp(OpKernelConstruction* c)
+      : OpKernel(c) {
+    c->GetAttr("strides", &strides_);
+    c->GetAttr("padding", &padding_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    auto f_sizes = IntVector(ctx->input(1));
+    const Tensor& dy = ctx->input(2);
+    TensorShape f_shape(f_sizes);
+    GpuConvGeom g;
+    OP_REQUIRES_OK(ctx, DwGeomFromShapes(x.shape(), f_shape, strides_,
+                                         padding_, &g));
+    int mult = (int)f_shape.dim_size(3);
+    hipStream_t s = GPU_STREAM(ctx);
+    Tensor* dw = ctx->allocate_output(0, f_shape);
+    Tensor scratch = ctx->allocate_temp(
+        DT_FLOAT, TensorShape({f_shape.num_elements()}));
+    OP_HIP_OK(ctx, ZeroF32(scratch.raw_data(), f_shape.num_elements(), s));
+    OP_HIP_OK(ctx, stf_depthwise_bwd_filter(
+                       x.raw_data(), dy.raw_data(), scratch.flat<float>(),
+                       (int)g.N, (int)g.H, (int)g.W, (int)g.C, (int)g.R,
+                       (int)g.S, (int)g.sh, (int)g.sw, (int)g.ph, (int)g.pw,
+                       (int)g.P, (int)g.Q, mult, s));
+    OP_HIP_OK(ctx, stf_cast(0, CastCode(dw->dtype()), scratch.raw_data(),
+                            dw->raw_data(), f_shape.num_elements(), s));
+  }
+
+ private:
+  std::vector<int64_t> strides_;
+  std::string padding_;
+};
+REGISTER_KERNEL_BUILDER(Name("DepthwiseConv2dNativeBackpropFilter").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T").HostMemory("filter_sizes"), GpuDepthwiseConvBackpropFilterOp);
 
 // ---------------------------------------------------------------------------
 // bias / softmax / xent

@@ -205,3 +205,21 @@ def lrn(input, depth_radius=5, bias=1.0, alpha=1.0, beta=0.5, name=None):  # pyl
 
 
 local_response_normalization = lrn
+
+
+def depthwise_conv2d(input, filter, strides, padding, rate=None, name=None):  # pylint: disable=redefined-builtin
+    """Depthwise 2-D convolution (reference nn_impl.depthwise_conv2d;
+    kernels in csrc/kernels/hip/depthwise.hip)."""
+    x = convert_to_tensor(input)
+    w = convert_to_tensor(filter)
+    out = apply_op('DepthwiseConv2dNative', x, w,
+                   strides=[int(s) for s in strides], padding=padding,
+                   name=name)
+    if x._shape is not None and w._shape is not None:
+        p = _conv_out_dim(x._shape[1], w._shape[0], strides[1], padding)
+        q = _conv_out_dim(x._shape[2], w._shape[1], strides[2], padding)
+        out.set_shape([x._shape[0], p, q, w._shape[2] * w._shape[3]])
+    return out
+
+
+depthwise_conv2d_native = depthwise_conv2d

@@ -117,3 +117,28 @@ def test_reduce_and_elementwise_vs_torch():
     np.testing.assert_allclose(got[2],
                                torch.sigmoid(torch.from_numpy(x[:5]))
                                .numpy(), rtol=1e-5, atol=1e-5)
+
+
+def test_depthwise_conv_and_grads_vs_torch():
+    rng = np.random.RandomState(6)
+    x = rng.randn(2, 9, 9, 4).astype(np.float32)
+    w = rng.randn(3, 3, 4, 1).astype(np.float32)
+    xt = tf.constant(x, dtype=tf.bfloat16)
+    wt = tf.constant(w, dtype=tf.bfloat16)
+    y = tf.nn.depthwise_conv2d(xt, wt, [1, 1, 1, 1], 'SAME')
+    loss = tf.reduce_sum(tf.cast(y, tf.float32))
+    gx, gw = tf.gradients(loss, [xt, wt])
+    got_y, got_gx, got_gw = _run([y, gx, gw])
+
+    tx = torch.from_numpy(x).permute(0, 3, 1, 2).requires_grad_(True)
+    twp = torch.from_numpy(w).permute(2, 3, 0, 1).reshape(4, 1, 3, 3) \
+        .detach().requires_grad_(True)
+    ty = torch.nn.functional.conv2d(tx, twp, padding=1, groups=4)
+    ty.sum().backward()
+    want_y = ty.detach().permute(0, 2, 3, 1).numpy()
+    want_gx = tx.grad.permute(0, 2, 3, 1).numpy()
+    want_gw = twp.grad.reshape(4, 1, 3, 3).permute(2, 3, 0, 1).numpy()
+    scale = np.abs(want_y).max()
+    assert np.abs(got_y - want_y).max() / scale < 0.05
+    assert np.abs(got_gx - want_gx).max() / (np.abs(want_gx).max()) < 0.05
+    assert np.abs(got_gw - want_gw).max() / (np.abs(want_gw).max()) < 0.05

@@ -221,9 +221,6 @@ REGISTER_OP("MergeV2Checkpoints").Input("checkpoint_prefixes: string").Input("de
 // ------------------------------- queues ------------------------------------
 // TensorArray family (reference core/ops/data_flow_ops.cc:1080; resource
 // handles are session-scoped strings here, like the queue ops above).
-REGISTER_OP("DepthwiseConv2dNative").Input("input: T").Input("filter: T").Output("output: T").Attr("T: {float, bfloat16}").Attr("strides: list(int)").Attr("padding: {'SAME', 'VALID'}").Attr("data_format: {'NHWC', 'NCHW'} = 'NHWC'");
-REGISTER_OP("DepthwiseConv2dNativeBackpropInput").Input("input_sizes: int32").Input("filter: T").Input("out_backprop: T").Output("output: T").Attr("T: {float, bfloat16}").Attr("strides: list(int)").Attr("padding: {'SAME', 'VALID'}").Attr("data_format: {'NHWC', 'NCHW'} = 'NHWC'");
-REGISTER_OP("DepthwiseConv2dNativeBackpropFilter").Input("input: T").Input("filter_sizes: int32").Input("out_backprop: T").Output("output: T").Attr("T: {float, bfloat16}").Attr("strides: list(int)").Attr("padding: {'SAME', 'VALID'}").Attr("data_format: {'NHWC', 'NCHW'} = 'NHWC'");
 REGISTER_OP("PlaceholderWithDefault").Input("input: dtype").Output("output: dtype").Attr("dtype: type").Attr("shape: shape = []");
 REGISTER_OP("SparseToDense").Input("sparse_indices: Tindices").Input("output_shape: Tindices").Input("sparse_values: T").Input("default_value: T").Output("dense: T").Attr("validate_indices: bool = true").Attr("T: type").Attr("Tindices: {int32, int64}");
 REGISTER_OP("ResizeBilinear").Input("images: T").Input("size: int32").Output("resized_images: float").Attr("T: {float}").Attr("align_corners: bool = false");

@@ -194,3 +194,32 @@ def _merge_grad(op, grad, _index_grad=None):
 for _op in ('ZerosLike', 'OnesLike', 'StopGradient', 'Shape', 'ShapeN',
             'UnsortedSegmentSum', 'OneHot', 'InvertPermutation'):
     ops.NoGradient(_op)
+
+
+@RegisterGradient('Enter')
+def _enter_grad(op, grad):
+    # constant-capture Enter inside a rematerialized while body: the
+    # gradient passes straight through to the captured external tensor.
+    return [grad]
+
+
+@RegisterGradient('Exit')
+def _exit_grad(op, grad):
+    return [grad]
+
+
+@RegisterGradient('NextIteration')
+def _next_iteration_grad(op, grad):
+    return [grad]
+
+
+# TensorArray reads act as sweep boundaries in the while-loop gradient
+# (values are forward-recorded; nothing upstream needs their gradient).
+@RegisterGradient('TensorArrayReadV3')
+def _ta_read_grad(op, grad):
+    return [None, None, None]
+
+
+@RegisterGradient('TensorArrayWriteV3')
+def _ta_write_grad(op, grad):
+    return [None, None, None, None]

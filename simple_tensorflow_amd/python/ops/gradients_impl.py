@@ -13,6 +13,75 @@ from simple_tensorflow_amd.python.ops import array_grad  # noqa: F401
 from simple_tensorflow_amd.python.ops import nn_grad  # noqa: F401
 
 
+class _WhilePseudoOp(object):
+    """Stands in for a whole while loop during the reverse sweep: inputs are
+    the original loop variables + captured externals, outputs the exits
+    (reference gradients_impl + control_flow_grad collapse into this since
+    our while_loop records its iteration inputs in TensorArrays)."""
+
+    def __init__(self, record):
+        self.record = record
+        self.inputs = list(record['loop_var_inputs']) + \
+            list(record['externals'])
+        self.outputs = list(record['exits'])
+        self.type = '_WhileLoop'
+        self.name = record['exits'][0].op.name + '_loop'
+
+
+def _while_grad(record, exit_grads):
+    """Builds the backward loop: j = N-1..0, reading iteration inputs from
+    the forward recording and rematerializing the body to differentiate it.
+    Returns grads for loop_var_inputs + externals."""
+    from simple_tensorflow_amd.python.ops import control_flow_ops
+    from simple_tensorflow_amd.python.ops import tensor_array_ops
+    n = record['n']
+    ext = record['externals']
+    body_fn = record['body_fn']
+    count = record['count_exit']
+    tas = [record['tas'][k]._with_flow(record['ta_flow_exits'][k])
+           for k in range(n)]
+    g = ops.get_default_graph()
+    with g.name_scope(record['exits'][0].op.name.rsplit('/', 1)[0] +
+                      '_grad'):
+        gvars0 = [eg if eg is not None else
+                  array_ops.zeros_like(record['exits'][k])
+                  for k, eg in enumerate(exit_grads)]
+        gext0 = [array_ops.zeros_like(e) for e in ext]
+
+        def gcond(j, *rest):
+            return math_ops.greater_equal(j, ops.constant(0))
+
+        def gbody(j, *rest):
+            gvars = list(rest[:n])
+            gexts = list(rest[n:])
+            vals = [tas[k].read(j) for k in range(n)]
+            for k, v in enumerate(vals):
+                if record['exits'][k]._shape is not None:
+                    v.set_shape(record['exits'][k]._shape)
+            from simple_tensorflow_amd.python.ops import variable_scope \
+                as vs_mod
+            with vs_mod.variable_scope(record['var_scope'], reuse=True):
+                outs = body_fn(*vals)
+            if not isinstance(outs, (list, tuple)):
+                outs = [outs]
+            outs = [ops.convert_to_tensor(o) for o in outs]
+            inner = gradients(outs, list(vals) + list(ext), grad_ys=gvars,
+                              name='bodygrad')
+            new_gvars = [ig if ig is not None else array_ops.zeros_like(v)
+                         for ig, v in zip(inner[:n], vals)]
+            new_gexts = [ge + ig if ig is not None else ge
+                         for ge, ig in zip(gexts, inner[n:])]
+            return [math_ops.subtract(j, 1)] + new_gvars + new_gexts
+
+        res = control_flow_ops.while_loop(
+            gcond, gbody,
+            [count - ops.constant(1)] + gvars0 + gext0,
+            back_prop=False)
+        if not isinstance(res, (list, tuple)):
+            res = [res]
+        return list(res[1:1 + n]) + list(res[1 + n:])
+
+
 def gradients(ys, xs, grad_ys=None, name='gradients',
               colocate_gradients_with_ops=False, gate_gradients=False,
               aggregation_method=None, stop_gradients=None):
@@ -35,28 +104,58 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
 
     g = ops.get_default_graph()
     with g.name_scope(name):
-        # ---- find ops between xs and ys ----
-        x_ops = {x.op for x in xs}
-        reached = set()  # ops from which some x is reachable (downstream sweep)
-        stack = list(x_ops)
+        # ---- collapse recorded while loops into pseudo ops ----
+        pseudo_of = {}   # internal/exit op -> _WhilePseudoOp
+        pseudos = []
+        for op in g._node_list:
+            if op.type != 'Exit':
+                continue
+            rec = getattr(op.outputs[0], '_while_record', None) if \
+                op.outputs else None
+            if rec is None or id(rec) in {id(p.record) for p in pseudos}:
+                continue
+            ps = _WhilePseudoOp(rec)
+            pseudos.append(ps)
+            for iop in rec['internal_ops']:
+                pseudo_of[iop] = ps
+
+        def xlate(op):
+            return pseudo_of.get(op, op)
+
+        # ---- find ops between xs and ys (pseudo-aware) ----
+        x_ops = {xlate(x.op) for x in xs}
         consumers = _build_consumers(g)
+
+        def successors(node):
+            if isinstance(node, _WhilePseudoOp):
+                outs = node.outputs
+                seen = []
+                for t in outs:
+                    for c in consumers.get(t.op, ()):  # consumers of exits
+                        seen.append(xlate(c))
+                return seen
+            return [xlate(c) for c in consumers.get(node, ())]
+
+        def predecessors(node):
+            ins = node.inputs
+            return [xlate(t.op) for t in ins]
+
+        reached = set()
+        stack = list(x_ops)
         while stack:
             op = stack.pop()
             if op in reached:
                 continue
             reached.add(op)
-            for c in consumers.get(op, ()):  # ops consuming op's outputs
-                stack.append(c)
-        # ops that influence ys
+            stack.extend(successors(op))
         useful = set()
-        stack = [y.op for y in ys]
+        stack = [xlate(y.op) for y in ys]
         while stack:
             op = stack.pop()
             if op in useful:
                 continue
             useful.add(op)
-            for t in op.inputs:
-                stack.append(t.op)
+            stack.extend(predecessors(op))
         between = reached & useful
 
         # ---- init output grads ----
@@ -87,9 +186,20 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
         out_consumers = {}
         for op in between:
             n = 0
-            for c in consumers.get(op, ()):  # consumer ops
-                if c in between:
-                    n += sum(1 for t in c.inputs if t.op is op)
+            outs = op.outputs
+            for t_out in outs:
+                for c in consumers.get(
+                        t_out.op if isinstance(op, _WhilePseudoOp) else op,
+                        ()):
+                    if xlate(c) in between and xlate(c) is not op:
+                        n += sum(1 for t in c.inputs if t is t_out)
+                if not isinstance(op, _WhilePseudoOp):
+                    break  # plain op: counted all outputs via consumers map
+            if not isinstance(op, _WhilePseudoOp):
+                n = 0
+                for c in consumers.get(op, ()):
+                    if xlate(c) in between:
+                        n += sum(1 for t in c.inputs if t.op is op)
             out_consumers[op] = n
 
         # seed: ops whose outputs include ys get their seed grads immediately
@@ -116,6 +226,26 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
             if op in processed:
                 continue
             processed.add(op)
+            if isinstance(op, _WhilePseudoOp):
+                exit_grads = []
+                for t in op.outputs:
+                    lst = grads.get(t)
+                    exit_grads.append(None if not lst else
+                                      (lst[0] if len(lst) == 1
+                                       else math_ops.add_n(lst)))
+                if all(eg is None for eg in exit_grads):
+                    continue  # loop not on any differentiated path
+                in_grads = _while_grad(op.record, exit_grads)
+                for t, dg in zip(op.inputs, in_grads):
+                    if t in stop_set:
+                        dg = None
+                    tprod = xlate(t.op)
+                    if tprod in between:
+                        add_grad(t, dg)
+                        out_consumers[tprod] -= 1
+                        if out_consumers[tprod] == 0:
+                            ready.append(tprod)
+                continue
             out_grads = []
             has_any = False
             for t in op.outputs:
@@ -152,11 +282,12 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
             for t, dg in zip(op.inputs, in_grads):
                 if t in stop_set or t.op.type in ('StopGradient',):
                     dg = None
-                if t.op in between:
+                tprod = xlate(t.op)
+                if tprod in between:
                     add_grad(t, dg)
-                    out_consumers[t.op] -= 1
-                    if out_consumers[t.op] == 0:
-                        ready.append(t.op)
+                    out_consumers[tprod] -= 1
+                    if out_consumers[tprod] == 0:
+                        ready.append(tprod)
 
         # ---- collect ----
         result = []

@@ -78,3 +78,23 @@ def test_hipgraph_capture_matches_eager():
     final = build_and_train(20)  # capture kicks in after run 2
     assert np.isfinite(final)
     assert final < 1.0  # trains well past initial ~2.3
+
+
+def test_dynamic_rnn_trains_gpu():
+    """while-loop gradients on the GPU executor (frames + TensorArray +
+    rematerialized backward)."""
+    cell = tf.nn.rnn_cell.BasicRNNCell(32)
+    x = tf.constant(np.random.RandomState(0)
+                    .randn(8, 12, 16).astype(np.float32))
+    out, state = tf.nn.dynamic_rnn(cell, x, dtype=tf.float32)
+    target = tf.constant(np.random.RandomState(1)
+                         .randn(8, 32).astype(np.float32))
+    loss = tf.reduce_mean(tf.square(state - target))
+    train = tf.train.GradientDescentOptimizer(0.05).minimize(loss)
+    with tf.Session() as s:
+        s.run(tf.global_variables_initializer())
+        l0 = s.run(loss)
+        for _ in range(20):
+            s.run(train)
+        l1 = s.run(loss)
+    assert np.isfinite(l1) and l1 < l0 * 0.9

@@ -349,6 +349,12 @@ class LikeOp : public OpKernel {
   }
 };
 REGISTER_KERNEL_BUILDER(Name("ZerosLike").Device(DEVICE_CPU), LikeOp<false>);
+// int32/int64 variants under the GPU device compute on host (loop counters
+// stay in the GPU partition — see cpu_math.cc REG_GPU_HOST_INT).
+REGISTER_KERNEL_BUILDER(Name("ZerosLike").Device(DEVICE_GPU).TypeConstraint<int32_t>("T").HostMemory("x").HostMemory("y"), LikeOp<false>);
+REGISTER_KERNEL_BUILDER(Name("ZerosLike").Device(DEVICE_GPU).TypeConstraint<int64_t>("T").HostMemory("x").HostMemory("y"), LikeOp<false>);
+REGISTER_KERNEL_BUILDER(Name("OnesLike").Device(DEVICE_GPU).TypeConstraint<int32_t>("T").HostMemory("x").HostMemory("y"), LikeOp<true>);
+REGISTER_KERNEL_BUILDER(Name("OnesLike").Device(DEVICE_GPU).TypeConstraint<int64_t>("T").HostMemory("x").HostMemory("y"), LikeOp<true>);
 REGISTER_KERNEL_BUILDER(Name("OnesLike").Device(DEVICE_CPU), LikeOp<true>);
 
 // --------------------------------- Cast -------------------------------------

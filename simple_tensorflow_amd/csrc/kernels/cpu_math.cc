@@ -132,6 +132,34 @@ REG_CMP("GreaterEqual", FGreaterEq)
 REG_CMP("Equal", FEq)
 REG_CMP("NotEqual", FNe)
 #undef REG_CMP
+// Host-compute int32/int64 registrations under the GPU device (all args in
+// host memory). These keep while-loop counter arithmetic in the SAME
+// partition as the GPU loop body — an in-frame cross-partition edge would
+// need the reference's per-device control-loop machinery
+// (graph_partition.cc:801), which round 1 does not implement. The analog of
+// the reference's int32-on-GPU host kernels.
+#define REG_GPU_HOST_INT(OP, F)                                               \
+  REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_GPU).TypeConstraint<int32_t>("T").HostMemory("x").HostMemory("y").HostMemory("z"), BinaryOp<int32_t, F>); \
+  REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_GPU).TypeConstraint<int64_t>("T").HostMemory("x").HostMemory("y").HostMemory("z"), BinaryOp<int64_t, F>);
+REG_GPU_HOST_INT("Add", FAdd)
+REG_GPU_HOST_INT("Sub", FSub)
+REG_GPU_HOST_INT("Mul", FMul)
+REG_GPU_HOST_INT("Maximum", FMax)
+REG_GPU_HOST_INT("Minimum", FMin)
+REG_GPU_HOST_INT("FloorDiv", FFloorDiv)
+REG_GPU_HOST_INT("FloorMod", FFloorMod)
+#undef REG_GPU_HOST_INT
+#define REG_GPU_HOST_CMP(OP, F)                                               \
+  REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_GPU).TypeConstraint<int32_t>("T").HostMemory("x").HostMemory("y").HostMemory("z"), BinaryOp<int32_t, F, bool>); \
+  REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_GPU).TypeConstraint<int64_t>("T").HostMemory("x").HostMemory("y").HostMemory("z"), BinaryOp<int64_t, F, bool>);
+REG_GPU_HOST_CMP("Less", FLess)
+REG_GPU_HOST_CMP("LessEqual", FLessEq)
+REG_GPU_HOST_CMP("Greater", FGreater)
+REG_GPU_HOST_CMP("GreaterEqual", FGreaterEq)
+REG_GPU_HOST_CMP("Equal", FEq)
+REG_GPU_HOST_CMP("NotEqual", FNe)
+#undef REG_GPU_HOST_CMP
+
 struct FAnd { bool operator()(bool a, bool b) const { return a && b; } };
 struct FOr { bool operator()(bool a, bool b) const { return a || b; } };
 REGISTER_KERNEL_BUILDER(Name("LogicalAnd").Device(DEVICE_CPU), BinaryOp<bool, FAnd>);

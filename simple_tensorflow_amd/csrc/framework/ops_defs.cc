@@ -154,6 +154,8 @@ REGISTER_OP("FusedBatchNormGrad").Input("y_backprop: T").Input("x: T").Input("sc
 // MI355X-native fused batch norm: bf16 activations with f32 statistics and
 // f32 scale/offset (the bf16 analog of FusedBatchNorm; reserve = 1/sqrt(var+eps)).
 REGISTER_OP("BatchNormMi").Input("x: T").Input("scale: float").Input("offset: float").Output("y: T").Output("batch_mean: float").Output("batch_variance: float").Output("reserve_inv_std: float").Attr("T: {float, bfloat16}").Attr("epsilon: float = 0.0001").Attr("fuse_relu: bool = false");
+REGISTER_OP("BatchNormAddReluMi").Input("x: T").Input("scale: float").Input("offset: float").Input("side: T").Output("y: T").Output("batch_mean: float").Output("batch_variance: float").Output("reserve_inv_std: float").Attr("T: {float, bfloat16}").Attr("epsilon: float = 0.0001");
+REGISTER_OP("BatchNormAddReluMiGrad").Input("y_backprop: T").Input("x: T").Input("scale: float").Input("saved_mean: float").Input("saved_inv_std: float").Input("y_relu: T").Output("x_backprop: T").Output("scale_backprop: float").Output("offset_backprop: float").Output("side_backprop: T").Attr("T: {float, bfloat16}").Attr("epsilon: float = 0.0001");
 REGISTER_OP("BatchNormMiGrad").Input("y_backprop: T").Input("x: T").Input("scale: float").Input("saved_mean: float").Input("saved_inv_std: float").Input("y_relu: T").Output("x_backprop: T").Output("scale_backprop: float").Output("offset_backprop: float").Attr("T: {float, bfloat16}").Attr("epsilon: float = 0.0001").Attr("fuse_relu: bool = false");
 REGISTER_OP("L2Loss").Input("t: T").Output("output: T").Attr("T: " FLOATTYPES);
 REGISTER_OP("LRN").Input("input: T").Output("output: T").Attr("depth_radius: int = 5").Attr("bias: float = 1.0").Attr("alpha: float = 1.0").Attr("beta: float = 0.5").Attr("T: {float}");

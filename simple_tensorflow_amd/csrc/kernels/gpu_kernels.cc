@@ -64,6 +64,11 @@ hipError_t stf_xent(int, const void*, const void*, void*, void*, int64_t, int,
 hipError_t stf_bn_fwd(int, const void*, const void*, const void*, float*,
                       float*, float*, float*, void*, int64_t, int, float, int,
                       hipStream_t);
+hipError_t stf_bn_stats_only(int, const void*, float*, float*, float*,
+                             float*, int64_t, int, float, hipStream_t);
+hipError_t stf_bn_add_relu(const void*, const void*, const float*,
+                           const float*, const void*, const void*, void*,
+                           int64_t, int, hipStream_t);
 hipError_t stf_bn_bwd(int, const void*, const void*, const void*,
                       const float*, const float*, const void*, float*,
                       float*, void*, int64_t, int, int, hipStream_t);
@@ -980,6 +985,45 @@ class GpuBatchNormMiOp : public OpKernel {
 REGISTER_KERNEL_BUILDER(Name("BatchNormMi").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuBatchNormMiOp);
 REGISTER_KERNEL_BUILDER(Name("BatchNormMi").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuBatchNormMiOp);
 
+// BN + residual-add + relu fused (reference has no equivalent; the CDNA4
+// win is one elementwise pass instead of three over a [N,H,W,C] tensor).
+class GpuBatchNormAddReluMiOp : public OpKernel {
+ public:
+  explicit GpuBatchNormAddReluMiOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("epsilon", &eps_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    const Tensor& scale = ctx->input(1);
+    const Tensor& offset = ctx->input(2);
+    const Tensor& side = ctx->input(3);
+    int c = (int)x.dim_size(x.dims() - 1);
+    int64_t rows = x.NumElements() / c;
+    Tensor* y = ctx->allocate_output(0, x.shape());
+    Tensor* mean = ctx->allocate_output(1, TensorShape({c}));
+    Tensor* var = ctx->allocate_output(2, TensorShape({c}));
+    Tensor* inv_std = ctx->allocate_output(3, TensorShape({c}));
+    hipStream_t s = GPU_STREAM(ctx);
+    Tensor acc = ctx->allocate_temp(DT_FLOAT, TensorShape({2 * c}));
+    OP_HIP_OK(ctx, ZeroF32(acc.raw_data(), 2 * c, s));
+    // stats + finalize from the plain BN path, then the fused normalize
+    OP_HIP_OK(ctx, stf_bn_stats_only(DtypeCode(x.dtype()), x.raw_data(),
+                                     acc.flat<float>(), mean->flat<float>(),
+                                     var->flat<float>(),
+                                     inv_std->flat<float>(), rows, c, eps_,
+                                     s));
+    OP_HIP_OK(ctx, stf_bn_add_relu(x.raw_data(), side.raw_data(),
+                                   mean->flat<float>(),
+                                   inv_std->flat<float>(), scale.raw_data(),
+                                   offset.raw_data(), y->raw_data(),
+                                   rows * (int64_t)c, c, s));
+  }
+
+ private:
+  float eps_ = 1e-4f;
+};
+REGISTER_KERNEL_BUILDER(Name("BatchNormAddReluMi").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuBatchNormAddReluMiOp);
+
 class GpuBatchNormMiGradOp : public OpKernel {
  public:
   explicit GpuBatchNormMiGradOp(OpKernelConstruction* c) : OpKernel(c) {
@@ -1013,6 +1057,39 @@ class GpuBatchNormMiGradOp : public OpKernel {
  private:
   bool fuse_relu_ = false;
 };
+class GpuBatchNormAddReluMiGradOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& dy = ctx->input(0);
+    const Tensor& x = ctx->input(1);
+    const Tensor& scale = ctx->input(2);
+    const Tensor& mean = ctx->input(3);
+    const Tensor& inv_std = ctx->input(4);
+    const Tensor& y_relu = ctx->input(5);
+    int c = (int)x.dim_size(x.dims() - 1);
+    int64_t rows = x.NumElements() / c;
+    Tensor* dx = ctx->allocate_output(0, x.shape());
+    Tensor* dscale = ctx->allocate_output(1, TensorShape({c}));
+    Tensor* doffset = ctx->allocate_output(2, TensorShape({c}));
+    Tensor* dside = ctx->allocate_output(3, x.shape());
+    hipStream_t s = GPU_STREAM(ctx);
+    OP_HIP_OK(ctx, ZeroF32(doffset->raw_data(), c, s));
+    OP_HIP_OK(ctx, ZeroF32(dscale->raw_data(), c, s));
+    OP_HIP_OK(ctx, stf_bn_bwd(DtypeCode(x.dtype()), dy.raw_data(),
+                              x.raw_data(), y_relu.raw_data(),
+                              mean.flat<float>(), inv_std.flat<float>(),
+                              scale.raw_data(), doffset->flat<float>(),
+                              dscale->flat<float>(), dx->raw_data(), rows, c,
+                              1, s));
+    // dside = relu-masked dy
+    OP_HIP_OK(ctx, stf_binary(B_RELU_GRAD, DtypeCode(x.dtype()),
+                              dy.raw_data(), y_relu.raw_data(),
+                              dside->raw_data(), x.NumElements(), s));
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("BatchNormAddReluMiGrad").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuBatchNormAddReluMiGradOp);
+
 REGISTER_KERNEL_BUILDER(Name("BatchNormMiGrad").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuBatchNormMiGradOp);
 REGISTER_KERNEL_BUILDER(Name("BatchNormMiGrad").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuBatchNormMiGradOp);
 

@@ -313,6 +313,51 @@ __global__ void BnNormKernelV8(const __bf16* __restrict__ x,
   }
 }
 
+// BN + residual add + relu in one pass (the bottleneck tail:
+// relu(bn(conv_out) + shortcut)) — saves two full elementwise passes.
+__global__ void BnNormAddReluKernelV8(const __bf16* __restrict__ x,
+                                      const __bf16* __restrict__ side,
+                                      const float* __restrict__ mean,
+                                      const float* __restrict__ inv_std,
+                                      const float* __restrict__ scale,
+                                      const float* __restrict__ offset,
+                                      __bf16* __restrict__ y, int64_t n8,
+                                      int c) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int cg = c / 8;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += stride) {
+    int cb = (int)(i % cg) * 8;
+    __bf16 v[8], sv[8], o[8];
+    *(ulong2*)v = *(const ulong2*)(x + i * 8);
+    *(ulong2*)sv = *(const ulong2*)(side + i * 8);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float f = ((float)v[e] - mean[cb + e]) * inv_std[cb + e] *
+                    scale[cb + e] +
+                offset[cb + e] + (float)sv[e];
+      o[e] = (__bf16)(f > 0.f ? f : 0.f);
+    }
+    *(ulong2*)(y + i * 8) = *(ulong2*)o;
+  }
+}
+
+extern "C" hipError_t stf_bn_add_relu(const void* x, const void* side,
+                                      const float* mean,
+                                      const float* inv_std,
+                                      const void* scale, const void* offset,
+                                      void* y, int64_t n, int c,
+                                      hipStream_t stream) {
+  if (c % 8 != 0) return hipErrorInvalidValue;
+  int64_t n8 = n / 8;
+  hipLaunchKernelGGL(BnNormAddReluKernelV8, ElemwiseGrid(n8, 256, 1),
+                     dim3(256), 0, stream, (const __bf16*)x,
+                     (const __bf16*)side, mean, inv_std,
+                     (const float*)scale, (const float*)offset, (__bf16*)y,
+                     n8, c);
+  return hipGetLastError();
+}
+
 // bwd pass 1: per-channel sum(dy), sum(dy * xhat) into acc[2C] (zeroed).
 // RELU: the forward was BN+ReLU fused; dy is masked by y>0 on the fly so no
 // separate relu-grad elementwise pass (or its memory traffic) is needed.
@@ -684,6 +729,24 @@ hipError_t stf_xent(int dtype, const void* logits, const void* labels,
                        0, stream, (const __bf16*)logits,
                        (const __bf16*)labels, (float*)loss_f32,
                        (__bf16*)backprop, cols);
+  return hipGetLastError();
+}
+
+// stats + finalize only (the fused add+relu normalize runs separately)
+extern "C" hipError_t stf_bn_stats_only(int dtype, const void* x, float* acc,
+                                        float* mean, float* var,
+                                        float* inv_std, int64_t rows, int c,
+                                        float eps, hipStream_t stream) {
+  int blocks = 512;
+  size_t lds = (size_t)c * 8;
+  if (dtype == 0)
+    hipLaunchKernelGGL((BnStatsKernel<float>), dim3(blocks), dim3(256), lds,
+                       stream, (const float*)x, acc, rows, c);
+  else
+    hipLaunchKernelGGL((BnStatsKernel<__bf16>), dim3(blocks), dim3(256), lds,
+                       stream, (const __bf16*)x, acc, rows, c);
+  hipLaunchKernelGGL(BnFinalizeKernel, dim3((c + 255) / 256), dim3(256), 0,
+                     stream, acc, mean, var, inv_std, rows, c, eps);
   return hipGetLastError();
 }
 

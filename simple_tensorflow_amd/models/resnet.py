@@ -44,14 +44,24 @@ def conv2d_bf16(x, name, r, s, cin, cout, stride, relu_bn=True, bn_relu=True):
 
 
 def bottleneck(x, name, cin, cmid, cout, stride):
+    from simple_tensorflow_amd.python.ops import variables, array_ops
     shortcut = x
     if cin != cout or stride != 1:
         shortcut = conv2d_bf16(x, name + '_proj', 1, 1, cin, cout, stride,
                                relu_bn=True, bn_relu=False)
     y = conv2d_bf16(x, name + '_a', 1, 1, cin, cmid, stride)
     y = conv2d_bf16(y, name + '_b', 3, 3, cmid, cmid, 1)
-    y = conv2d_bf16(y, name + '_c', 1, 1, cmid, cout, 1, bn_relu=False)
-    return tf.nn.relu(y + shortcut)
+    # tail: conv -> fused BN+residual-add+ReLU (one elementwise pass)
+    y = conv2d_bf16(y, name + '_c', 1, 1, cmid, cout, 1, relu_bn=False)
+    scale = variables.Variable(array_ops.ones([cout], tf.float32),
+                               name=name + '_c_bn_scale')
+    offset = variables.Variable(array_ops.zeros([cout], tf.float32),
+                                name=name + '_c_bn_offset')
+    out, _, _, _ = apply_op('BatchNormAddReluMi', y, scale.ref(),
+                            offset.ref(), shortcut, epsilon=1e-4,
+                            name=name + '_c_bnar')
+    out.set_shape(y._shape)
+    return out
 
 
 def resnet50_loss(images_bf16, labels_i64, num_classes=1000):

@@ -146,3 +146,16 @@ def _depthwise_conv2d_grad(op, grad):
     dx.set_shape(x._shape)
     dw.set_shape(w._shape)
     return [dx, dw]
+
+
+@RegisterGradient('BatchNormAddReluMi')
+def _batch_norm_add_relu_mi_grad(op, grad_y, *rest):
+    dx, dscale, doffset, dside = apply_op(
+        'BatchNormAddReluMiGrad', grad_y, op.inputs[0], op.inputs[1],
+        op.outputs[1], op.outputs[3], op.outputs[0],
+        epsilon=op.get_attr('epsilon'))
+    dx.set_shape(op.inputs[0]._shape)
+    dscale.set_shape(op.inputs[1]._shape)
+    doffset.set_shape(op.inputs[2]._shape)
+    dside.set_shape(op.inputs[3]._shape)
+    return [dx, dscale, doffset, dside]

@@ -130,7 +130,11 @@ inline int DtypeCode(DataType dt) { return dt == DT_FLOAT ? 0 : 1; }
 inline int PickSplitK(int64_t M, int64_t N, int64_t K) {
   int64_t tiles = ((M + 127) / 128) * ((N + 127) / 128);
   if (tiles >= 256 || K < 1024) return 1;
-  int64_t want = 512 / (tiles ? tiles : 1);
+  static const int64_t target = [] {
+    const char* e = getenv("STF_SPLITK_TARGET");
+    return e ? atoll(e) : 512;
+  }();
+  int64_t want = target / (tiles ? tiles : 1);
   int64_t maxk = K / 512;  // keep >= 8 K-iters per slice
   if (maxk < 1) maxk = 1;
   int64_t sk = std::min(want, maxk);

@@ -11,14 +11,22 @@ import numpy as np
 import simple_tensorflow_amd as tf
 
 
-def bench_shape(sess_cache, m, n, k, iters=20):
+def bench_shape(sess_cache, m, n, k, iters=20, dw=False):
     tf.reset_default_graph()
     rng = np.random.RandomState(0)
-    a = tf.constant((rng.rand(m, k) * 2 - 1).astype(np.float32),
-                    dtype=tf.bfloat16)
-    b = tf.constant((rng.rand(n, k) * 2 - 1).astype(np.float32),
-                    dtype=tf.bfloat16)
-    c = tf.matmul(a, b, transpose_b=True)  # direct NT, no pre-transpose
+    if dw:
+        # dW shape: C[m,n] = A[k,m]^T x B[k,n] (both contraction-major)
+        a = tf.constant((rng.rand(k, m) * 2 - 1).astype(np.float32),
+                        dtype=tf.bfloat16)
+        b = tf.constant((rng.rand(k, n) * 2 - 1).astype(np.float32),
+                        dtype=tf.bfloat16)
+        c = tf.matmul(a, b, transpose_a=True)
+    else:
+        a = tf.constant((rng.rand(m, k) * 2 - 1).astype(np.float32),
+                        dtype=tf.bfloat16)
+        b = tf.constant((rng.rand(n, k) * 2 - 1).astype(np.float32),
+                        dtype=tf.bfloat16)
+        c = tf.matmul(a, b, transpose_b=True)  # direct NT
     # keep the result alive on device by snapshotting into a variable
     from simple_tensorflow_amd.python.ops import variables
     v = variables.Variable(tf.zeros([m, n], tf.bfloat16), trainable=False)
@@ -42,9 +50,10 @@ def bench_shape(sess_cache, m, n, k, iters=20):
 def main():
     import sys
     if len(sys.argv) > 1:
+        dw = len(sys.argv) > 2 and sys.argv[2] == 'dw'
         m, n, k = (int(x) for x in sys.argv[1].split(','))
         for _ in range(3):
-            bench_shape(None, m, n, k)
+            bench_shape(None, m, n, k, dw=dw)
         return
     shapes = [
         (4096, 4096, 4096),

@@ -130,11 +130,21 @@ inline int DtypeCode(DataType dt) { return dt == DT_FLOAT ? 0 : 1; }
 inline int PickSplitK(int64_t M, int64_t N, int64_t K) {
   int64_t tiles = ((M + 127) / 128) * ((N + 127) / 128);
   if (tiles >= 256 || K < 1024) return 1;
+  // Measured optimum keeps ~48 K-chunks (of 64) per block: enough work to
+  // amortize the prologue, enough blocks to hide staging latency
+  // (576x64x802816: 725us @ sk=102 -> 605us @ sk~260).
   static const int64_t target = [] {
     const char* e = getenv("STF_SPLITK_TARGET");
-    return e ? atoll(e) : 512;
+    return e ? atoll(e) : 0;
   }();
-  int64_t want = target / (tiles ? tiles : 1);
+  int64_t want;
+  if (target > 0) {
+    want = target / (tiles ? tiles : 1);
+  } else {
+    int64_t kt = K / 64;
+    want = std::max<int64_t>(512 / (tiles ? tiles : 1), kt / 48);
+    if (tiles * want > 4096) want = 4096 / tiles;
+  }
   int64_t maxk = K / 512;  // keep >= 8 K-iters per slice
   if (maxk < 1) maxk = 1;
   int64_t sk = std::min(want, maxk);

@@ -133,6 +133,17 @@ Status DirectSession::BuildExecutors(const std::vector<std::string>& feeds,
         continue;
       }
     }
+    // String tensors are host-only (variable-length payloads have no device
+    // representation): pin every string-producing node to the CPU.
+    {
+      bool has_string = false;
+      for (auto t : n->out_types)
+        if (t == DT_STRING) has_string = true;
+      if (has_string) {
+        n->assigned_device = CanonicalDevice(cpu->name());
+        continue;
+      }
+    }
     if (!req.empty()) {
       dev = devices_.LookUp(req);
       if (!dev && StrStartsWith(req, "GPU") && gpu0) dev = gpu0;

@@ -189,6 +189,7 @@ inline int CastCode(DataType dt) {
     case DT_HALF: return 2;
     case DT_INT32: return 3;
     case DT_INT64: return 4;
+    case DT_BOOL: return 5;
     default: return -1;
   }
 }

@@ -271,7 +271,7 @@ extern "C" hipError_t stf_binary_bcast(int op, int dtype, const void* a,
   return hipGetLastError();
 }
 
-// src/dst dtype codes: 0 f32, 1 bf16, 2 f16, 3 i32, 4 i64
+// src/dst dtype codes: 0 f32, 1 bf16, 2 f16, 3 i32, 4 i64, 5 bool
 extern "C" hipError_t stf_cast(int sdt, int ddt, const void* x, void* y,
                                int64_t n, hipStream_t stream) {
   dim3 grid = ElemwiseGrid(n, 256, 4);
@@ -291,6 +291,13 @@ extern "C" hipError_t stf_cast(int sdt, int ddt, const void* x, void* y,
   CASTCASE(4, 3, int64_t, int32_t)
   CASTCASE(1, 1, __bf16, __bf16)
   CASTCASE(0, 0, float, float)
+  CASTCASE(5, 0, bool, float)
+  CASTCASE(5, 1, bool, __bf16)
+  CASTCASE(5, 3, bool, int32_t)
+  CASTCASE(5, 4, bool, int64_t)
+  CASTCASE(0, 5, float, bool)
+  CASTCASE(3, 5, int32_t, bool)
+  CASTCASE(4, 5, int64_t, bool)
 #undef CASTCASE
   return hipErrorInvalidValue;
 }

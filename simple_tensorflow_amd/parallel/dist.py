@@ -18,6 +18,18 @@ from simple_tensorflow_amd.python.framework.ops import apply_op
 from simple_tensorflow_amd.python.framework import ops as fw_ops
 
 
+# True when collectives run through the torch-gloo CPU fallback (no GPU, or
+# more ranks than visible GPUs — e.g. a 2-process test on a 1-GPU box). The
+# Rccl* ops are then pinned to /cpu:0 so placement doesn't pick the GPU
+# kernel, which would require a real RCCL communicator.
+_CPU_FALLBACK = False
+
+
+def _collective_device():
+    return fw_ops.get_default_graph().device(
+        '/cpu:0' if _CPU_FALLBACK else '')
+
+
 class Comm(object):
     def __init__(self, world, rank, torch_dist):
         self.world = world
@@ -43,7 +55,8 @@ class Comm(object):
             ref = v._as_graph_element()
             if ref.dtype.name not in ('float32', 'bfloat16', 'float16'):
                 continue
-            b = apply_op('RcclBroadcast', v.value(), root=0)
+            with _collective_device():
+                b = apply_op('RcclBroadcast', v.value(), root=0)
             if prev is not None:
                 b.op._add_control_input(prev)  # fixed cross-rank order
             asn = tf.assign(ref, b)
@@ -64,8 +77,13 @@ def init(world, rank):
         os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
         td.init_process_group('gloo', rank=rank, world_size=world)
     import torch
-    has_gpu = tf.Session().num_gpus() > 0
-    if has_gpu:
+    global _CPU_FALLBACK
+    ngpu = tf.Session().num_gpus()
+    # RCCL needs one distinct device per rank; with more ranks than GPUs
+    # (CPU-only hosts, or a 2-process test on a 1-GPU box) collectives run
+    # on CPU through torch.distributed gloo instead.
+    if ngpu > 0 and world <= ngpu:
+        _CPU_FALLBACK = False
         if rank == 0:
             uid = _core.rccl_get_unique_id()
             buf = torch.tensor(list(uid), dtype=torch.uint8)
@@ -75,6 +93,8 @@ def init(world, rank):
         uid = bytes(buf.tolist())
         _core.rccl_init(world, rank, uid)
     else:
+        _CPU_FALLBACK = True
+
         def _allreduce(arr):
             t = torch.from_numpy(np.ascontiguousarray(arr))
             td.all_reduce(t, op=td.ReduceOp.SUM)
@@ -110,8 +130,9 @@ class DistributedOptimizer(object):
             g_scaled = g * scale
             if prev is not None:
                 g_scaled.op._add_control_input(prev)
-            red = apply_op('RcclAllReduce', g_scaled,
-                           num_devices=self._world)
+            with _collective_device():
+                red = apply_op('RcclAllReduce', g_scaled,
+                               num_devices=self._world)
             red.set_shape(g._shape)
             prev = red.op
             out.append((red, v))

@@ -26,7 +26,7 @@ def test_matmul_vs_numpy():
     b = np.random.randn(5, 3).astype(np.float32)
     with tf.Session() as s:
         out = s.run(tf.matmul(tf.constant(a), tf.constant(b)))
-    np.testing.assert_allclose(out, a @ b, rtol=1e-5)
+    np.testing.assert_allclose(out, a @ b, rtol=1e-5, atol=1e-6)
 
 
 def test_matmul_transpose():
@@ -35,7 +35,7 @@ def test_matmul_transpose():
     with tf.Session() as s:
         out = s.run(tf.matmul(tf.constant(a), tf.constant(b),
                               transpose_a=True, transpose_b=True))
-    np.testing.assert_allclose(out, a.T @ b.T, rtol=1e-5)
+    np.testing.assert_allclose(out, a.T @ b.T, rtol=1e-5, atol=1e-6)
 
 
 def test_feed_fetch():

@@ -134,14 +134,42 @@ Status DirectSession::BuildExecutors(const std::vector<std::string>& feeds,
       }
     }
     // String tensors are host-only (variable-length payloads have no device
-    // representation): pin every string-producing node to the CPU.
+    // representation). TensorArray-style ops carry string HANDLES, but their
+    // GPU kernels declare them HostMemory — those stay GPU-eligible so
+    // while-loop frames remain single-partition (the executor has no
+    // cross-partition control loops). Pin to CPU only when GPU placement
+    // would put a string payload in device memory.
     {
-      bool has_string = false;
+      bool touches_string = false;
       for (auto t : n->out_types)
-        if (t == DT_STRING) has_string = true;
-      if (has_string) {
-        n->assigned_device = CanonicalDevice(cpu->name());
-        continue;
+        if (t == DT_STRING) touches_string = true;
+      for (auto t : n->in_types)
+        if (t == DT_STRING) touches_string = true;
+      if (touches_string) {
+        bool gpu_ok = false;
+        const KernelDef* kd =
+            gpu0 ? KernelRegistry::Global()->Find(n->def, "GPU") : nullptr;
+        const OpDef* od = OpRegistry::Global()->LookUp(n->op());
+        if (kd && od) {
+          // Only fixed-arity signatures are checked arg-by-arg; ops with
+          // list-expanded string args stay conservative (CPU).
+          gpu_ok = od->output_arg.size() == n->out_types.size() &&
+                   od->input_arg.size() == n->in_types.size();
+          if (gpu_ok)
+            for (size_t i = 0; i < n->out_types.size(); ++i)
+              if (n->out_types[i] == DT_STRING &&
+                  !kd->host_memory.count(od->output_arg[i].name))
+                gpu_ok = false;
+          if (gpu_ok)
+            for (size_t i = 0; i < n->in_types.size(); ++i)
+              if (n->in_types[i] == DT_STRING &&
+                  !kd->host_memory.count(od->input_arg[i].name))
+                gpu_ok = false;
+        }
+        if (!gpu_ok) {
+          n->assigned_device = CanonicalDevice(cpu->name());
+          continue;
+        }
       }
     }
     if (!req.empty()) {

@@ -145,7 +145,11 @@ Status DirectSession::BuildExecutors(const std::vector<std::string>& feeds,
         if (t == DT_STRING) touches_string = true;
       for (auto t : n->in_types)
         if (t == DT_STRING) touches_string = true;
-      if (touches_string) {
+      // Control-flow ops (Enter/Exit/Switch/Merge/NextIteration) forward
+      // tensors without allocating: a host-resident string handle passes
+      // through them untouched, and pinning them would split a while frame
+      // across partitions (deadlock). Exempt them.
+      if (touches_string && !n->IsControlFlow()) {
         bool gpu_ok = false;
         const KernelDef* kd =
             gpu0 ? KernelRegistry::Global()->Find(n->def, "GPU") : nullptr;

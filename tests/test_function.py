@@ -1,0 +1,94 @@
+"""Defun graph functions + custom symbolic gradients
+(reference python/framework/function.py + SymbolicGradient)."""
+import numpy as np
+
+import simple_tensorflow_amd as tf
+from simple_tensorflow_amd.python.framework import function
+
+
+def setup_function(_):
+    tf.reset_default_graph()
+
+
+def test_defun_call_and_run():
+    @function.Defun(tf.float32, tf.float32)
+    def f(a, b):
+        return a * b + a
+
+    x = tf.constant([1.0, 2.0, 3.0])
+    y = tf.constant([4.0, 5.0, 6.0])
+    out = f(x, y)
+    with tf.Session() as s:
+        v = s.run(out)
+    np.testing.assert_allclose(v, [5.0, 12.0, 21.0])
+
+
+def test_defun_multiple_calls_unique():
+    @function.Defun(tf.float32)
+    def sq(x):
+        return x * x
+
+    a = sq(tf.constant(3.0))
+    b = sq(tf.constant(4.0))
+    assert a.op.name != b.op.name
+    with tf.Session() as s:
+        va, vb = s.run([a, b])
+    assert va == 9.0 and vb == 16.0
+
+
+def test_defun_default_gradient():
+    @function.Defun(tf.float32)
+    def cube(x):
+        return x * x * x
+
+    x = tf.constant(2.0)
+    y = cube(x)
+    dx, = tf.gradients(y, [x])
+    with tf.Session() as s:
+        v = s.run(dx)
+    np.testing.assert_allclose(v, 12.0)  # 3x^2
+
+
+def test_defun_python_grad_func():
+    # custom gradient deliberately wrong (2x) to prove it is the one used
+    def grad(call, dy):
+        return dy * call.inputs[0] * 2.0
+
+    @function.Defun(tf.float32, python_grad_func=grad)
+    def cube(x):
+        return x * x * x
+
+    x = tf.constant(3.0)
+    dx, = tf.gradients(cube(x), [x])
+    with tf.Session() as s:
+        v = s.run(dx)
+    np.testing.assert_allclose(v, 6.0)  # NOT 27
+
+
+def test_defun_grad_func_defun():
+    @function.Defun(tf.float32, tf.float32)
+    def mul_grad(x, dy):
+        return dy * tf.exp(x)  # wrong on purpose: proves grad_func is used
+
+    @function.Defun(tf.float32, grad_func=mul_grad)
+    def f(x):
+        return x * x
+
+    x = tf.constant(0.0)
+    dx, = tf.gradients(f(x), [x])
+    with tf.Session() as s:
+        v = s.run(dx)
+    np.testing.assert_allclose(v, 1.0)  # exp(0), not 2*0
+
+
+def test_defun_gradient_mixes_with_plain_ops():
+    @function.Defun(tf.float32)
+    def half(x):
+        return x * 0.5
+
+    x = tf.constant([2.0, 4.0])
+    y = tf.reduce_sum(half(x) * x)  # d/dx (x^2/2) = x
+    dx, = tf.gradients(y, [x])
+    with tf.Session() as s:
+        v = s.run(dx)
+    np.testing.assert_allclose(v, [2.0, 4.0])

@@ -53,5 +53,33 @@ def embedding_lookup(params, ids, partition_strategy='mod', name=None,
 
 def embedding_lookup_sparse(params, sp_ids, sp_weights, combiner='mean',
                             name=None, partition_strategy='mod'):
-    raise NotImplementedError(
-        'embedding_lookup_sparse requires SparseTensor support')
+    """Weighted per-row combination of embeddings for ragged id lists held
+    in a SparseTensor (reference embedding_ops.embedding_lookup_sparse:191).
+    Rows are sp_ids.indices[:, 0]; combine = sum | mean | sqrtn. Assumes
+    every row has at least one id (same contract as the reference)."""
+    if combiner not in ('sum', 'mean', 'sqrtn'):
+        raise ValueError('unknown combiner %r' % combiner)
+    g = ops.get_default_graph()
+    with g.name_scope(name or 'embedding_lookup_sparse'):
+        seg = math_ops.cast(sp_ids.indices[:, 0], dtypes.int32)
+        emb = embedding_lookup(params,
+                               math_ops.cast(sp_ids.values, dtypes.int32),
+                               partition_strategy=partition_strategy)
+        n_rows = math_ops.cast(sp_ids.dense_shape[0], dtypes.int32)
+        if sp_weights is not None:
+            w = array_ops.reshape(
+                math_ops.cast(sp_weights.values, emb.dtype), [-1, 1])
+            emb = emb * w
+        else:
+            w = array_ops.reshape(
+                math_ops.cast(math_ops.cast(sp_ids.values, dtypes.int32) * 0
+                              + 1, emb.dtype), [-1, 1])
+        combined = array_ops.unsorted_segment_sum(emb, seg, n_rows)
+        if combiner == 'sum':
+            return combined
+        if combiner == 'mean':
+            denom = array_ops.unsorted_segment_sum(w, seg, n_rows)
+        else:  # sqrtn
+            denom = math_ops.sqrt(
+                array_ops.unsorted_segment_sum(w * w, seg, n_rows))
+        return combined / denom

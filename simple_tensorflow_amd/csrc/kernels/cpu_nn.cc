@@ -550,6 +550,44 @@ class CpuDepthwiseConvOp : public OpKernel {
 REGISTER_KERNEL_BUILDER(Name("DepthwiseConv2dNative").Device(DEVICE_CPU).TypeConstraint<float>("T"), CpuDepthwiseConvOp<float>);
 REGISTER_KERNEL_BUILDER(Name("DepthwiseConv2dNative").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), CpuDepthwiseConvOp<bfloat16>);
 
+// Local response normalization over the channel (last) axis:
+// y_i = x_i / (bias + alpha * sum_{|j-i|<=r} x_j^2)^beta
+// (reference lrn_op.cc; the backward pass is a python composite in
+// python/ops/nn_grad.py _lrn_grad).
+class LRNOp : public OpKernel {
+ public:
+  explicit LRNOp(OpKernelConstruction* ctx) : OpKernel(ctx) {
+    ctx->GetAttr("depth_radius", &radius_);
+    ctx->GetAttr("bias", &bias_);
+    ctx->GetAttr("alpha", &alpha_);
+    ctx->GetAttr("beta", &beta_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& in = ctx->input(0);
+    Tensor* out = ctx->allocate_output(0, in.shape());
+    int64_t c = in.shape().dim_size(in.shape().dims() - 1);
+    int64_t rows = in.NumElements() / c;
+    const float* x = in.flat<float>();
+    float* y = out->flat<float>();
+    for (int64_t rI = 0; rI < rows; ++rI) {
+      const float* xr = x + rI * c;
+      float* yr = y + rI * c;
+      for (int64_t k = 0; k < c; ++k) {
+        int64_t lo = k - radius_ < 0 ? 0 : k - radius_;
+        int64_t hi = k + radius_ + 1 > c ? c : k + radius_ + 1;
+        float s = bias_;
+        for (int64_t j = lo; j < hi; ++j) s += alpha_ * xr[j] * xr[j];
+        yr[k] = xr[k] / std::pow(s, beta_);
+      }
+    }
+  }
+
+ private:
+  int64_t radius_ = 5;
+  float bias_ = 1.f, alpha_ = 1.f, beta_ = 0.5f;
+};
+REGISTER_KERNEL_BUILDER(Name("LRN").Device(DEVICE_CPU), LRNOp);
+
 }  // namespace
 
 }  // namespace stf

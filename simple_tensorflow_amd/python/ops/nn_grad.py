@@ -118,9 +118,39 @@ def _batch_norm_mi_grad(op, grad_y, *rest):
     return dx, dscale, doffset
 
 
+def _lrn_window_sum(t, radius):
+    """Sum of a (2r+1)-wide window over the channel (last) axis, via pad +
+    shifted slices. Needs a static channel count (always true in the conv
+    nets that use LRN)."""
+    c = t._shape[-1]
+    padded = array_ops.pad(
+        t, [[0, 0]] * (len(t._shape) - 1) + [[radius, radius]])
+    total = None
+    for k in range(2 * radius + 1):
+        begin = [0] * (len(t._shape) - 1) + [k]
+        size = [-1] * (len(t._shape) - 1) + [c]
+        piece = array_ops.slice(padded, begin, size)
+        total = piece if total is None else total + piece
+    return total
+
+
 @RegisterGradient('LRN')
 def _lrn_grad(op, grad):
-    raise NotImplementedError('LRN gradient: round 2')
+    """y_i = x_i * s_i^-b with s_i = bias + a*sum_win(x^2):
+    dx_k = g_k s_k^-b - 2ab x_k sum_{i: k in win(i)} g_i x_i s_i^(-b-1)
+    (the window relation is symmetric, so the second sum is the same
+    windowed sum applied to g*x*s^(-b-1))."""
+    x = op.inputs[0]
+    radius = op.get_attr('depth_radius')
+    bias = float(op.get_attr('bias'))
+    alpha = float(op.get_attr('alpha'))
+    beta = float(op.get_attr('beta'))
+    s = bias + alpha * _lrn_window_sum(x * x, radius)
+    sb = s ** (-beta)
+    inner = _lrn_window_sum(grad * x * (s ** (-beta - 1.0)), radius)
+    dx = grad * sb - (2.0 * alpha * beta) * x * inner
+    dx.set_shape(x._shape)
+    return dx
 
 
 for _op in ('MaxPoolGrad', 'AvgPoolGrad', 'ReluGrad', 'Relu6Grad',

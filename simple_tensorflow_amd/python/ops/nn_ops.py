@@ -200,8 +200,10 @@ def in_top_k(predictions, targets, k, name=None):
 
 
 def lrn(input, depth_radius=5, bias=1.0, alpha=1.0, beta=0.5, name=None):  # pylint: disable=redefined-builtin
-    return apply_op('LRN', input, depth_radius=depth_radius, bias=bias,
-                    alpha=alpha, beta=beta, name=name)
+    t = apply_op('LRN', input, depth_radius=depth_radius, bias=bias,
+                 alpha=alpha, beta=beta, name=name)
+    t.set_shape(t.op.inputs[0]._shape)
+    return t
 
 
 local_response_normalization = lrn

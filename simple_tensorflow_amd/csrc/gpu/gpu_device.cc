@@ -268,6 +268,20 @@ class GpuDevice : public Device {
   Status CopyHostTensorToDevice(const Tensor& src, Tensor* dst) override {
     hipSetDevice(ordinal_);
     Tensor dev(&bfc_, src.dtype(), src.shape());
+    // BFC reuse is stream-ordered ONLY on the compute stream: `dev` may be
+    // a just-freed block that pending compute kernels (enqueued before the
+    // free) still reference. Writing it on the h2d stream "now" would race
+    // those kernels — order the copy after all compute work enqueued so
+    // far. (Symmetric to the d2h path above. Skipped while capturing:
+    // recording an event on a capturing compute stream would splice the
+    // h2d stream into the capture; nothing executes during capture, so
+    // there is no pending work to race with.)
+    hipEvent_t ev = nullptr;
+    if (!capturing_) {
+      HIP_CHECK_STATUS(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+      HIP_CHECK_STATUS(hipEventRecord(ev, compute_));
+      HIP_CHECK_STATUS(hipStreamWaitEvent(h2d_, ev, 0));
+    }
     HIP_CHECK_STATUS(hipMemcpyAsync(dev.raw_data(), src.raw_data(),
                                     src.TotalBytes(), hipMemcpyHostToDevice,
                                     h2d_));
@@ -275,6 +289,7 @@ class GpuDevice : public Device {
     // and completion here also orders the data before any compute-stream
     // consumer enqueued after this call.
     HIP_CHECK_STATUS(hipStreamSynchronize(h2d_));
+    if (ev) HIP_CHECK_STATUS(hipEventDestroy(ev));
     if (capturing_) {
       // The h2d stream is not part of the capture, so this copy is NOT a
       // node of the hipGraph: replayed kernels will re-read `dev` directly.

@@ -79,6 +79,7 @@ enum class Code : int {
   UNIMPLEMENTED = 12,
   INTERNAL = 13,
   UNAVAILABLE = 14,
+  DATA_LOSS = 15,
 };
 
 class Status {
@@ -125,6 +126,7 @@ STF_DECLARE_ERROR(OutOfRange, OUT_OF_RANGE)
 STF_DECLARE_ERROR(Aborted, ABORTED)
 STF_DECLARE_ERROR(Cancelled, CANCELLED)
 STF_DECLARE_ERROR(Unavailable, UNAVAILABLE)
+STF_DECLARE_ERROR(DataLoss, DATA_LOSS)
 #undef STF_DECLARE_ERROR
 }  // namespace errors
 

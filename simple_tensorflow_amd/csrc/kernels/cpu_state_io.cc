@@ -152,6 +152,13 @@ class SaveV2Op : public OpKernel {
       const Tensor& t = ctx->input(3 + oi);
       OP_REQUIRES(ctx, slices.flat<std::string>()[oi].empty(),
                   errors::Unimplemented("SaveV2 slices"));
+      // Raw-byte copy would serialize heap pointers for DT_STRING tensors;
+      // reject until a length-prefixed string encoding exists
+      // (reference tensor_bundle.cc serializes strings specially).
+      OP_REQUIRES(ctx, t.dtype() != DT_STRING,
+                  errors::Unimplemented("SaveV2 does not support DT_STRING ",
+                                        "tensors (tensor ",
+                                        names.flat<std::string>()[oi], ")"));
       int64_t offset = (int64_t)data.size();
       int64_t size = (int64_t)t.TotalBytes();
       data.append((const char*)t.raw_data(), size);
@@ -219,6 +226,13 @@ class RestoreV2Op : public OpKernel {
             ctx, ReadFileString(ShardName(prefix, e.shard, num_shards), &data));
       OP_REQUIRES(ctx, e.offset + e.size <= (int64_t)data.size(),
                   errors::InvalidArgument("bundle entry out of range"));
+      // Verify the stored masked crc32c before accepting the bytes so a
+      // corrupted checkpoint fails loudly (reference tensor_bundle.cc:580).
+      uint32_t got = table::MaskCrc(
+          table::Crc32c(data.data() + e.offset, (size_t)e.size));
+      OP_REQUIRES(ctx, e.crc == 0 || got == e.crc,
+                  errors::DataLoss("checksum mismatch restoring ", name,
+                                   " from ", prefix));
       Tensor* out = ctx->allocate_output(i, e.shape);
       OP_REQUIRES(ctx, out->dtype() == e.dtype,
                   errors::InvalidArgument("restore dtype mismatch for ", name));

@@ -167,6 +167,20 @@ def _flatten_fetches(fetches):
     return [fetches], lambda vals: vals[0]
 
 
+def _check_feed_shape(t, arr):
+    # Reference BaseSession._run raises ValueError when a fed array is
+    # incompatible with the placeholder's declared static shape; silent
+    # acceptance invalidates static-shape-based gradient fast paths.
+    shape = getattr(t, '_shape', None)
+    if shape is None:
+        return
+    if len(arr.shape) != len(shape) or any(
+            d is not None and d != a for d, a in zip(shape, arr.shape)):
+        raise ValueError(
+            'Cannot feed value of shape %r for Tensor %r, which has shape %r'
+            % (tuple(arr.shape), t.name, tuple(shape)))
+
+
 def _convert_feed(t, v):
     if t.dtype is dtypes.string:
         if isinstance(v, str):
@@ -176,6 +190,7 @@ def _convert_feed(t, v):
         raise TypeError('string feed must be str/bytes')
     np_dt = t.dtype.as_numpy_dtype
     arr = np.asarray(v)
+    _check_feed_shape(t, arr)
     if t.dtype is dtypes.bfloat16:
         arr32 = arr.astype(np.float32)
         return ops._f32_to_bf16(arr32)

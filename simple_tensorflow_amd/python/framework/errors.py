@@ -18,6 +18,7 @@ OUT_OF_RANGE = 11
 UNIMPLEMENTED = 12
 INTERNAL = 13
 UNAVAILABLE = 14
+DATA_LOSS = 15
 
 
 class OpError(Exception):
@@ -83,6 +84,11 @@ class UnavailableError(OpError):
     pass
 
 
+class DataLossError(OpError):
+    pass
+
+
+
 _CODE_TO_CLASS = {
     CANCELLED: CancelledError,
     UNKNOWN: UnknownError,
@@ -98,6 +104,7 @@ _CODE_TO_CLASS = {
     UNIMPLEMENTED: UnimplementedError,
     INTERNAL: InternalError,
     UNAVAILABLE: UnavailableError,
+    DATA_LOSS: DataLossError,
 }
 
 

@@ -106,8 +106,11 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
               aggregation_method=None, stop_gradients=None):
     if not isinstance(ys, (list, tuple)):
         ys = [ys]
-    single_x = not isinstance(xs, (list, tuple))
-    if single_x:
+    # Reference contract (python/ops/gradients_impl.py gradients:376): the
+    # result is ALWAYS a list, even for a single non-list x — unwrapping the
+    # single case makes tf.gradients(loss, w)[0] silently index into the
+    # gradient tensor.
+    if not isinstance(xs, (list, tuple)):
         xs = [xs]
     xs = [x._as_graph_element() if hasattr(x, '_as_graph_element') else x
           for x in xs]
@@ -254,6 +257,21 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
             if out_consumers[op] == 0:
                 ready.append(op)
 
+        def _propagate(op, in_grads):
+            # Hand a (possibly None) gradient to every input and decrement
+            # the producer's consumer count — Kahn readiness depends on the
+            # decrement happening even when the grad is None.
+            for t, dg in zip(op.inputs, in_grads):
+                if dg is not None and (
+                        t in stop_set or t.op.type in ('StopGradient',)):
+                    dg = None
+                tprod = xlate(t.op)
+                if tprod in between:
+                    add_grad(t, dg)
+                    out_consumers[tprod] -= 1
+                    if out_consumers[tprod] == 0:
+                        ready.append(tprod)
+
         while ready:
             op = ready.pop()
             if op in processed:
@@ -268,7 +286,10 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
                                      (lst[0] if len(lst) == 1
                                       else math_ops.add_n(lst)))
                 if all(og is None for og in out_grads):
-                    continue  # call not on any differentiated path
+                    # Call not on any differentiated path: still propagate
+                    # None to inputs so upstream producers become ready.
+                    _propagate(op, [None] * len(op.inputs))
+                    continue
                 filled = [og if og is not None else array_ops.zeros_like(t)
                           for t, og in zip(op.outputs, out_grads)]
                 with g.name_scope(op.name + '_grad'):
@@ -282,15 +303,7 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
                     raise ValueError(
                         'gradient of function %s returned %d grads, want %d'
                         % (rec.func.name, len(in_grads), len(op.inputs)))
-                for t, dg in zip(op.inputs, in_grads):
-                    if t in stop_set:
-                        dg = None
-                    tprod = xlate(t.op)
-                    if tprod in between:
-                        add_grad(t, dg)
-                        out_consumers[tprod] -= 1
-                        if out_consumers[tprod] == 0:
-                            ready.append(tprod)
+                _propagate(op, in_grads)
                 continue
             if isinstance(op, _WhilePseudoOp):
                 exit_grads = []
@@ -300,17 +313,12 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
                                       (lst[0] if len(lst) == 1
                                        else math_ops.add_n(lst)))
                 if all(eg is None for eg in exit_grads):
-                    continue  # loop not on any differentiated path
+                    # Loop not on any differentiated path: still propagate
+                    # None so upstream producers become ready.
+                    _propagate(op, [None] * len(op.inputs))
+                    continue
                 in_grads = _while_grad(op.record, exit_grads)
-                for t, dg in zip(op.inputs, in_grads):
-                    if t in stop_set:
-                        dg = None
-                    tprod = xlate(t.op)
-                    if tprod in between:
-                        add_grad(t, dg)
-                        out_consumers[tprod] -= 1
-                        if out_consumers[tprod] == 0:
-                            ready.append(tprod)
+                _propagate(op, in_grads)
                 continue
             out_grads = []
             has_any = False
@@ -345,15 +353,7 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
                         raise ValueError(
                             'grad fn for %s returned %d grads, want %d' %
                             (op.type, len(in_grads), len(op.inputs)))
-            for t, dg in zip(op.inputs, in_grads):
-                if t in stop_set or t.op.type in ('StopGradient',):
-                    dg = None
-                tprod = xlate(t.op)
-                if tprod in between:
-                    add_grad(t, dg)
-                    out_consumers[tprod] -= 1
-                    if out_consumers[tprod] == 0:
-                        ready.append(tprod)
+            _propagate(op, in_grads)
 
         # ---- collect ----
         result = []
@@ -366,7 +366,7 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
                 result.append(lst[0])
             else:
                 result.append(math_ops.add_n(lst))
-        return result[0] if single_x else result
+        return result
 
 
 def _needs_fill(op):

@@ -275,10 +275,15 @@ def _mean_grad(op, grad):
 
 @RegisterGradient('Max')
 def _max_grad(op, grad):
-    y_rep = _tile_spec(op, op.outputs[0], array_ops.shape(op.inputs[0]))
-    g_rep = _tile_spec(op, grad, array_ops.shape(op.inputs[0]))
+    # Reference _MaxGrad divides the incoming gradient evenly among tied
+    # extrema (indicators / num_selected), so [1,1,0] -> [0.5,0.5,0].
+    in_shape = array_ops.shape(op.inputs[0])
+    y_rep = _tile_spec(op, op.outputs[0], in_shape)
+    g_rep = _tile_spec(op, grad, in_shape)
     mask = math_ops.cast(math_ops.equal(op.inputs[0], y_rep), grad.dtype)
-    return [g_rep * mask, None]
+    num_sel = _tile_spec(op, math_ops.reduce_sum(mask, op.inputs[1]),
+                         in_shape)
+    return [g_rep * mask / num_sel, None]
 
 
 @RegisterGradient('Min')

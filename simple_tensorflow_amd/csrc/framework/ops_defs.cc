@@ -261,5 +261,11 @@ REGISTER_OP("QueueSize").Input("handle: Ref(string)").Output("size: int32").SetI
 // path per BASELINE.json config 3).
 REGISTER_OP("RcclAllReduce").Input("input: T").Output("output: T").Attr("T: {float, bfloat16, half}").Attr("reduction: string = 'sum'").Attr("num_devices: int = 1").Attr("shared_name: string = ''").SetIsStateful();
 REGISTER_OP("RcclBroadcast").Input("input: T").Output("output: T").Attr("T: {float, bfloat16, half}").Attr("root: int = 0").SetIsStateful();
+// Fused gradient bucket: one ncclAllReduce per bucket on a dedicated comm
+// stream (pack -> reduce f32 -> unpack*scale), overlapped with backprop.
+// Outputs are valid for compute-stream consumers only AFTER a RcclCommSync
+// control edge (parallel/dist.py wires it).
+REGISTER_OP("RcclBucketAllReduce").Input("inputs: N * T").Output("outputs: N * T").Attr("N: int >= 1").Attr("T: {float, bfloat16}").Attr("scale: float = 1").SetIsStateful();
+REGISTER_OP("RcclCommSync").SetIsStateful();
 
 }  // namespace stf

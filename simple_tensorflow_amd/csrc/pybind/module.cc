@@ -300,6 +300,96 @@ REGISTER_KERNEL_BUILDER(Name("RcclAllReduce").Device(DEVICE_CPU),
 REGISTER_KERNEL_BUILDER(Name("RcclBroadcast").Device(DEVICE_CPU),
                         CpuBroadcastOp);
 
+// CPU fallback of the fused-bucket path: pack all segments into one f32
+// array, ONE collective call per bucket (same wire semantics as the GPU
+// ncclAllReduce of the flat staging buffer), unpack with the 1/world scale.
+class CpuBucketAllReduceOp : public OpKernel {
+ public:
+  explicit CpuBucketAllReduceOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("scale", &scale_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    int n = ctx->num_inputs();
+    int64_t total = 0;
+    for (int i = 0; i < n; ++i) total += ctx->input(i).NumElements();
+    std::vector<float> flat((size_t)total);
+    int64_t off = 0;
+    for (int i = 0; i < n; ++i) {
+      const Tensor& in = ctx->input(i);
+      int64_t m = in.NumElements();
+      if (in.dtype() == DT_FLOAT) {
+        memcpy(flat.data() + off, in.raw_data(), m * sizeof(float));
+      } else if (in.dtype() == DT_BFLOAT16) {
+        const bfloat16* src = in.flat<bfloat16>();
+        for (int64_t j = 0; j < m; ++j)
+          flat[off + j] = (float)src[j];
+      } else {
+        ctx->SetStatus(errors::InvalidArgument("bucket dtype must be f32/bf16"));
+        return;
+      }
+      off += m;
+    }
+    py::gil_scoped_acquire gil;
+    py::function fn;
+    {
+      std::lock_guard<std::mutex> l(g_pyfunc_mu);
+      auto it = PyFuncRegistry().find("__cpu_collective_allreduce");
+      if (it == PyFuncRegistry().end()) {
+        ctx->SetStatus(errors::FailedPrecondition(
+            "CPU collective fallback not initialized — call "
+            "parallel.dist.init() first"));
+        return;
+      }
+      fn = it->second;
+    }
+    py::array_t<float> arr((py::ssize_t)total);
+    memcpy(arr.mutable_data(), flat.data(), total * sizeof(float));
+    py::object result;
+    try {
+      result = fn(arr);
+    } catch (py::error_already_set& e) {
+      ctx->SetStatus(errors::Internal("collective raised: ", e.what()));
+      return;
+    }
+    py::array red = py::array::ensure(result);
+    Tensor rt = NumpyToTensor(red);
+    if (rt.dtype() != DT_FLOAT || rt.NumElements() != total) {
+      ctx->SetStatus(errors::Internal("bucket collective shape mismatch"));
+      return;
+    }
+    const float* rf = rt.flat<float>();
+    off = 0;
+    for (int i = 0; i < n; ++i) {
+      const Tensor& in = ctx->input(i);
+      int64_t m = in.NumElements();
+      Tensor* out = ctx->allocate_output(i, in.shape());
+      if (in.dtype() == DT_FLOAT) {
+        float* dst = out->flat<float>();
+        for (int64_t j = 0; j < m; ++j) dst[j] = rf[off + j] * scale_;
+      } else {
+        bfloat16* dst = out->flat<bfloat16>();
+        for (int64_t j = 0; j < m; ++j)
+          dst[j] = bfloat16(rf[off + j] * scale_);
+      }
+      off += m;
+    }
+  }
+
+ private:
+  float scale_ = 1.0f;
+};
+REGISTER_KERNEL_BUILDER(Name("RcclBucketAllReduce").Device(DEVICE_CPU),
+                        CpuBucketAllReduceOp);
+
+// Everything on CPU is synchronous already; the sync is a no-op.
+class CpuCommSyncOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {}
+};
+REGISTER_KERNEL_BUILDER(Name("RcclCommSync").Device(DEVICE_CPU),
+                        CpuCommSyncOp);
+
 class PySession {
  public:
   explicit PySession(bool cpu_only) : sess_(cpu_only) {}

@@ -16,6 +16,7 @@ import simple_tensorflow_amd as tf
 from simple_tensorflow_amd import _core
 from simple_tensorflow_amd.python.framework.ops import apply_op
 from simple_tensorflow_amd.python.framework import ops as fw_ops
+from simple_tensorflow_amd.python.ops import array_ops
 
 
 # True when collectives run through the torch-gloo CPU fallback (no GPU, or
@@ -81,8 +82,10 @@ def init(world, rank):
     ngpu = tf.Session().num_gpus()
     # RCCL needs one distinct device per rank; with more ranks than GPUs
     # (CPU-only hosts, or a 2-process test on a 1-GPU box) collectives run
-    # on CPU through torch.distributed gloo instead.
-    if ngpu > 0 and world <= ngpu:
+    # on CPU through torch.distributed gloo instead. STF_FORCE_RCCL=1
+    # overrides (e.g. validating multi-rank RCCL on a single-GPU box).
+    if ngpu > 0 and (world <= ngpu or
+                     os.environ.get('STF_FORCE_RCCL') == '1'):
         _CPU_FALLBACK = False
         if rank == 0:
             uid = _core.rccl_get_unique_id()
@@ -110,32 +113,99 @@ def init(world, rank):
     return Comm(world, rank, td)
 
 
-class DistributedOptimizer(object):
-    """Wraps an optimizer: all-reduce (sum) gradients across ranks, scaled by
-    1/world, with a fixed cross-rank collective order (control-dep chain)."""
+def _grad_bytes(g):
+    shape = g._shape
+    if shape is None or any(d is None for d in shape):
+        return 4 << 20  # unknown shape: assume mid-sized
+    n = 1
+    for d in shape:
+        n *= d
+    try:
+        item = np.dtype(g.dtype.as_numpy_dtype).itemsize
+    except TypeError:
+        item = 2 if g.dtype.name in ('bfloat16', 'float16') else 4
+    return n * item
 
-    def __init__(self, opt, world):
+
+class DistributedOptimizer(object):
+    """Wraps an optimizer: bucketed all-reduce (sum, scaled by 1/world) of
+    gradients across ranks on a dedicated comm stream, overlapped with the
+    rest of backprop.
+
+    Gradients are grouped (reversed, so the bucket of the LAST layers —
+    produced first by backprop — reduces while earlier layers still compute)
+    into ~STF_BUCKET_BYTES fused buckets; each bucket is one
+    RcclBucketAllReduce (pack -> single ncclAllReduce -> unpack*1/world on
+    the comm stream). A control-dep chain between buckets fixes the
+    cross-rank collective order; a RcclCommSync op joins the comm stream
+    back before the optimizer-apply ops consume the reduced gradients.
+    """
+
+    def __init__(self, opt, world, bucket_bytes=None):
         self._opt = opt
         self._world = world
+        if bucket_bytes is None:
+            bucket_bytes = int(os.environ.get('STF_BUCKET_BYTES', 25 << 20))
+        self._bucket_bytes = bucket_bytes
 
     def compute_gradients(self, loss, var_list=None, **kw):
         gvs = self._opt.compute_gradients(loss, var_list=var_list, **kw)
-        out = []
-        prev = None
         scale = 1.0 / self._world
-        for g, v in gvs:
-            if g is None:
+        max_seg = 120  # STF_COMM_MAX_SEG in comm_pack.hip
+
+        # Buckets follow backprop completion order (reverse of creation
+        # order), split on dtype (a fused bucket is single-dtype).
+        live = [(i, g, v) for i, (g, v) in enumerate(gvs) if g is not None]
+        buckets = []
+        cur, cur_bytes, cur_dt = [], 0, None
+        for item in reversed(live):
+            g = item[1]
+            b = _grad_bytes(g)
+            if cur and (cur_bytes + b > self._bucket_bytes or
+                        len(cur) >= max_seg or g.dtype != cur_dt):
+                buckets.append(cur)
+                cur, cur_bytes = [], 0
+            cur.append(item)
+            cur_bytes += b
+            cur_dt = g.dtype
+        if cur:
+            buckets.append(cur)
+
+        reduced = {}   # original gvs index -> reduced grad tensor
+        bucket_ops = []
+        prev = None
+        for bk in buckets:
+            grads = [g for (_, g, _) in bk]
+            with _collective_device():
+                red = apply_op('RcclBucketAllReduce', grads, scale=scale)
+            if not isinstance(red, tuple):
+                red = (red,)
+            if prev is not None:
+                red[0].op._add_control_input(prev)
+            prev = red[0].op
+            bucket_ops.append(red[0].op)
+            for (idx, g, _), r in zip(bk, red):
+                r.set_shape(g._shape)
+                reduced[idx] = r
+
+        with _collective_device():
+            sync = apply_op('RcclCommSync')
+        for bop in bucket_ops:
+            sync._add_control_input(bop)
+        self._sync_op = sync
+
+        # Gate every reduced grad behind the sync: consumers (the apply ops)
+        # are then enqueued after the compute stream waits on the comm
+        # events. Identity forwards the buffer, no copy.
+        out = []
+        for i, (g, v) in enumerate(gvs):
+            if i not in reduced:
                 out.append((g, v))
                 continue
-            g_scaled = g * scale
-            if prev is not None:
-                g_scaled.op._add_control_input(prev)
-            with _collective_device():
-                red = apply_op('RcclAllReduce', g_scaled,
-                               num_devices=self._world)
-            red.set_shape(g._shape)
-            prev = red.op
-            out.append((red, v))
+            gated = array_ops.identity(reduced[i])
+            gated.op._add_control_input(sync)
+            gated.set_shape(g._shape)
+            out.append((gated, v))
         return out
 
     def apply_gradients(self, grads_and_vars, global_step=None, name=None):

@@ -396,6 +396,9 @@ class Graph(object):
     def get_operation_by_name(self, name):
         return self._nodes_by_name[name]
 
+    def get_operations(self):
+        return list(self._node_list)
+
     def get_tensor_by_name(self, name):
         base, _, idx = name.partition(':')
         return self._nodes_by_name[base].outputs[int(idx or 0)]

@@ -40,9 +40,10 @@ def main():
         comm.broadcast_variables(s)
         s.run(train)
         v = s.run(w.ref())
-    # average grad over ranks 1..world: mean(rank+1) = 1.5 for world=2
-    expect = -np.array([1.5, 3.0])
-    assert np.allclose(v, expect), (rank, v)
+    # average grad over ranks: mean(rank+1) (= 1.5 for world=2)
+    m = sum(range(1, world + 1)) / world
+    expect = -np.array([m, 2.0 * m])
+    assert np.allclose(v, expect), (rank, v, expect)
     comm.barrier()
 
     # ---- multi-bucket: many variables of mixed sizes, tiny bucket cap so

@@ -25,6 +25,7 @@ hipError_t stf_gemm_bf16(const void*, const void*, void*, const void*,
 hipError_t stf_gemm_bf16_splitk(const void*, const void*, void*, int64_t,
                                 int64_t, int64_t, int64_t, int64_t, int, int,
                                 int, hipStream_t);
+int stf_gemm_bf16_8ph_ok(int64_t, int64_t, int64_t);
 hipError_t stf_gemm_f32_nt(const void*, const void*, void*, int64_t, int64_t,
                            int64_t, hipStream_t);
 hipError_t stf_depthwise_fwd(const void*, const void*, void*, int, int, int,
@@ -607,7 +608,16 @@ class GpuConv2DOp : public OpKernel {
         cc->m[{x.raw_data(), ConvGeomHash(g)}] = col;
       }
     }
-    if ((g.K & 7) == 0) {
+    if (stf_gemm_bf16_8ph_ok(g.M(), g.K, rscp)) {
+      // 8-phase-eligible shape: one small weight transpose ([rscp,K] ->
+      // [K,rscp]) buys the NT fast path for the big M=NPQ GEMM.
+      Tensor wt = ctx->allocate_temp(DT_BFLOAT16, TensorShape({g.K, rscp}));
+      OP_HIP_OK(ctx, stf_transpose2d(2, wsrc.raw_data(), wt.raw_data(), rscp,
+                                     g.K, s));
+      OP_HIP_OK(ctx, stf_gemm_bf16_nt(col_data, wt.raw_data(), y->raw_data(),
+                                      nullptr, g.M(), g.K, rscp, 0.f, 1, 0,
+                                      s));
+    } else if ((g.K & 7) == 0) {
       OP_HIP_OK(ctx, stf_gemm_bf16(col_data, wsrc.raw_data(), y->raw_data(),
                                    nullptr, g.M(), g.K, rscp, rscp, g.K, 0.f,
                                    0, 1, 1, 0, s));

@@ -397,6 +397,13 @@ hipError_t LaunchSplitK(const uint16_t* A, const uint16_t* B, float* C,
 
 }  // namespace
 
+extern "C" int stf_gemm_bf16_8ph_ok(int64_t M, int64_t N, int64_t K);
+extern "C" hipError_t stf_gemm_bf16_8ph(const void* A, const void* B, void* C,
+                                        const void* bias_f32, int64_t M,
+                                        int64_t N, int64_t K, int64_t lda,
+                                        int64_t ldb, int out_bf16,
+                                        int fuse_relu, hipStream_t stream);
+
 // General entry. a_km/b_km select the contraction-major (transposed-staging)
 // path per operand; lda/ldb are row strides of the stored layouts
 // (a_km ? [K,M] : [M,K], b_km ? [K,N] : [N,K]).
@@ -406,6 +413,13 @@ extern "C" hipError_t stf_gemm_bf16(const void* A, const void* B, void* C,
                                     float beta, int a_km, int b_km,
                                     int out_bf16, int fuse_relu,
                                     hipStream_t stream) {
+  // Fat NT tiles take the 256² 8-phase template (guide §5 top rung);
+  // STF_NO_8PH reverts to the 128² step-3 kernel for A/B comparison.
+  static const bool no_8ph = getenv("STF_NO_8PH") != nullptr;
+  if (!no_8ph && !a_km && !b_km && beta == 0.f &&
+      stf_gemm_bf16_8ph_ok(M, N, K))
+    return stf_gemm_bf16_8ph(A, B, C, bias_f32, M, N, K, lda, ldb, out_bf16,
+                             fuse_relu, stream);
   const uint16_t* a = (const uint16_t*)A;
   const uint16_t* b = (const uint16_t*)B;
   const float* bias = (const float*)bias_f32;

@@ -142,3 +142,23 @@ def test_depthwise_conv_and_grads_vs_torch():
     assert np.abs(got_y - want_y).max() / scale < 0.05
     assert np.abs(got_gx - want_gx).max() / (np.abs(want_gx).max()) < 0.05
     assert np.abs(got_gw - want_gw).max() / (np.abs(want_gw).max()) < 0.05
+
+
+def test_matmul_bf16_8phase_path_vs_torch():
+    """Shapes eligible for the 256x256 8-phase GEMM (M,N mult of 256,
+    K mult of 64) — exercises the fast template, not the 128x128 fallback.
+    Non-square and transpose-detecting data (gemm_bf16_8ph.hip)."""
+    rng = np.random.RandomState(7)
+    for (m, n, k) in [(256, 256, 64), (512, 256, 128), (256, 512, 192),
+                      (768, 512, 320), (512, 128, 256), (512, 64, 128),
+                      (256, 576, 192), (256, 1152, 64)]:
+        a = (rng.rand(m, k) * 2 - 1).astype(np.float32)
+        b = (rng.rand(n, k) * 2 - 1).astype(np.float32)
+        tf.reset_default_graph()
+        got = _run(tf.matmul(tf.constant(a, dtype=tf.bfloat16),
+                             tf.constant(b, dtype=tf.bfloat16),
+                             transpose_b=True))
+        want = (torch.from_numpy(a).bfloat16().float() @
+                torch.from_numpy(b).bfloat16().float().T).numpy()
+        rel = np.abs(got - want) / (np.abs(want) + 1e-2)
+        assert np.percentile(rel, 99) < 0.05, (m, n, k)

@@ -26,6 +26,13 @@ hipError_t stf_gemm_bf16_splitk(const void*, const void*, void*, int64_t,
                                 int64_t, int64_t, int64_t, int64_t, int, int,
                                 int, hipStream_t);
 int stf_gemm_bf16_8ph_ok(int64_t, int64_t, int64_t);
+hipError_t stf_conv2d_fwd_8ph(const void*, const void*, void*, const void*,
+                              const void*, int, int, int, int, int, int, int,
+                              int, int, int, int, int, int64_t, int64_t, int,
+                              int, hipStream_t);
+hipError_t stf_conv2d_dw_splitk(const void*, const void*, void*, const void*,
+                                int, int, int, int, int, int, int, int, int,
+                                int, int, int, int64_t, int, hipStream_t);
 hipError_t stf_gemm_f32_nt(const void*, const void*, void*, int64_t, int64_t,
                            int64_t, hipStream_t);
 hipError_t stf_depthwise_fwd(const void*, const void*, void*, int, int, int,
@@ -193,6 +200,19 @@ inline int CastCode(DataType dt) {
     case DT_BOOL: return 5;
     default: return -1;
   }
+}
+
+// 256 zeroed device bytes for the implicit-GEMM address generators'
+// out-of-bounds (padding) loads. Allocated once per process (one GPU per
+// process under the data-parallel model).
+inline const void* ZeroPage() {
+  static const void* page = [] {
+    void* p = nullptr;
+    if (hipMalloc(&p, 256) != hipSuccess) return (void*)nullptr;
+    (void)hipMemset(p, 0, 256);
+    return p;
+  }();
+  return page;
 }
 
 // zero an f32 device buffer on the stream
@@ -592,6 +612,22 @@ class GpuConv2DOp : public OpKernel {
                                     rsc * g.K * 2, hipMemcpyDeviceToDevice,
                                     s));
     }
+    // Implicit-GEMM fast path (the reference's cuDNN implicit-GEMM slot,
+    // conv_ops.cc:664): column matrix generated inside the GEMM's A
+    // staging — no im2col kernel, no column buffer, no column cache.
+    static const bool no_implicit = getenv("STF_NO_IMPLICIT_CONV") != nullptr;
+    if (!no_implicit && !g.is_1x1_s1() && (g.C % 8) == 0 &&
+        stf_gemm_bf16_8ph_ok(g.M(), g.K, rscp) && ZeroPage()) {
+      Tensor wt = ctx->allocate_temp(DT_BFLOAT16, TensorShape({g.K, rscp}));
+      OP_HIP_OK(ctx, stf_transpose2d(2, wsrc.raw_data(), wt.raw_data(), rscp,
+                                     g.K, s));
+      OP_HIP_OK(ctx, stf_conv2d_fwd_8ph(
+                         x.raw_data(), wt.raw_data(), y->raw_data(), nullptr,
+                         ZeroPage(), (int)g.N, (int)g.H, (int)g.W, (int)g.C,
+                         (int)g.R, (int)g.S, (int)g.sh, (int)g.sw, (int)g.ph,
+                         (int)g.pw, (int)g.P, (int)g.Q, g.K, rscp, 1, 0, s));
+      return;
+    }
     const void* col_data = x.raw_data();
     Tensor col;
     if (!g.is_1x1_s1()) {
@@ -697,6 +733,24 @@ class GpuConv2DBackpropFilterOp : public OpKernel {
     // dW[RSC, K] = col[M, RSC]^T x dy[M, K]: both operands are contraction
     // (M-)major as stored, so the GEMM's K-major staging reads them directly
     // — no transpose kernels (this was 2 full passes over the im2col matrix).
+    // Implicit-GEMM dW: contraction over output pixels with the column
+    // matrix generated inside the K-major staging — pairs with the
+    // implicit forward (no materialized columns anywhere in the conv).
+    static const bool no_implicit = getenv("STF_NO_IMPLICIT_CONV") != nullptr;
+    if (!no_implicit && !g.is_1x1_s1() && (g.C % 8) == 0 &&
+        (g.K & 7) == 0 && (g.M() & 63) == 0 && ZeroPage()) {
+      Tensor scratch = ctx->allocate_temp(DT_FLOAT, TensorShape({rsc, g.K}));
+      OP_HIP_OK(ctx, hipMemsetAsync(scratch.raw_data(), 0, rsc * g.K * 4, s));
+      int sk = PickSplitK(rsc, g.K, g.M());
+      OP_HIP_OK(ctx, stf_conv2d_dw_splitk(
+                         x.raw_data(), dy.raw_data(), scratch.raw_data(),
+                         ZeroPage(), (int)g.N, (int)g.H, (int)g.W, (int)g.C,
+                         (int)g.R, (int)g.S, (int)g.sh, (int)g.sw, (int)g.ph,
+                         (int)g.pw, (int)g.P, (int)g.Q, g.K, sk, s));
+      OP_HIP_OK(ctx, stf_cast(0, 1, scratch.raw_data(), dw->raw_data(),
+                              rsc * g.K, s));
+      return;
+    }
     bool km_ok = (g.K & 7) == 0 && (!g.is_1x1_s1() || (g.C & 7) == 0);
     if (km_ok) {
       const void* col_data = x.raw_data();

@@ -32,16 +32,15 @@ struct TileGeom {
 
 // Fast staging: lane-linear glds; the source address carries the inverse
 // swizzle so the LDS image is the swizzled one.
-template <int BR>
-__device__ __forceinline__ void StageFast(const uint16_t* __restrict__ src,
-                                          int64_t ld, int row0, int64_t k0,
+template <int BR, class AG>
+__device__ __forceinline__ void StageFast(const AG& ag, int row0, int64_t k0,
                                           uint16_t* lds_base, int tid) {
 #pragma unroll
   for (int p = 0; p < TileGeom<BR>::kPasses; ++p) {
     int s = p * 256 + tid;
     int r = s >> 3;
     int c8 = (s & 7) ^ Swz(r);
-    const uint16_t* g = src + (int64_t)(row0 + r) * ld + k0 + c8 * 8;
+    const uint16_t* g = ag.at(row0 + r, k0 + c8 * 8);
     __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) uint32_t*)g,
                                      (__attribute__((address_space(3))) uint32_t*)(lds_base + (int64_t)s * 8),
                                      16, 0, 0);
@@ -54,19 +53,17 @@ __device__ __forceinline__ void StageFast(const uint16_t* __restrict__ src,
 // block with 16B row loads, transposes it in registers, and writes 8 full
 // 16B LDS slots — the Swz(r) swizzle keeps the 8-row-strided writes on
 // distinct banks.
-template <int BR>
-__device__ __forceinline__ void StageKMajor(const uint16_t* __restrict__ src,
-                                            int64_t ld, int col0, int64_t k0,
+template <int BR, class AG>
+__device__ __forceinline__ void StageKMajor(const AG& ag, int col0, int64_t k0,
                                             uint16_t* lds_base, int tid) {
   constexpr int kRG = BR / 8;  // 8-wide row groups per k row
   if (tid < 0 || tid >= BR) return;  // one thread per 8x8 block
   int r0 = (tid % kRG) * 8;
   int chunk = tid / kRG;       // k chunk of 8
-  const uint16_t* g = src + (k0 + chunk * 8) * ld + col0 + r0;
   uint16_t v[8][8];
 #pragma unroll
   for (int kk = 0; kk < 8; ++kk)
-    *(ulong2*)v[kk] = *(const ulong2*)(g + (int64_t)kk * ld);
+    *(ulong2*)v[kk] = *(const ulong2*)ag.at(k0 + chunk * 8 + kk, col0 + r0);
 #pragma unroll
   for (int e = 0; e < 8; ++e) {
     uint16_t out[8];
@@ -78,9 +75,9 @@ __device__ __forceinline__ void StageKMajor(const uint16_t* __restrict__ src,
   }
 }
 
-template <int BR>
+template <int BR, class AG>
 __device__ __forceinline__ void StageKMajorSafe(
-    const uint16_t* __restrict__ src, int64_t ld, int col0, int64_t k0,
+    const AG& ag, int col0, int64_t k0,
     int64_t cols, int64_t K, uint16_t* lds_base, int tid) {
   constexpr int kRG = BR / 8;
   if (tid < 0 || tid >= BR) return;
@@ -91,11 +88,11 @@ __device__ __forceinline__ void StageKMajorSafe(
   for (int kk = 0; kk < 8; ++kk) {
     int64_t krow = k0 + chunk * 8 + kk;
     if (krow < K && col0 + r0 + 8 <= cols) {
-      *(ulong2*)v[kk] = *(const ulong2*)(src + krow * ld + col0 + r0);
+      *(ulong2*)v[kk] = *(const ulong2*)ag.at(krow, col0 + r0);
     } else if (krow < K) {
 #pragma unroll
       for (int e = 0; e < 8; ++e)
-        v[kk][e] = (col0 + r0 + e < cols) ? src[krow * ld + col0 + r0 + e]
+        v[kk][e] = (col0 + r0 + e < cols) ? *ag.at(krow, col0 + r0 + e)
                                           : 0;
     } else {
 #pragma unroll
@@ -114,9 +111,8 @@ __device__ __forceinline__ void StageKMajorSafe(
 }
 
 // Guarded staging for edge blocks / K tails: scalar loads, zero padding.
-template <int BR>
-__device__ __forceinline__ void StageSafe(const uint16_t* __restrict__ src,
-                                          int64_t ld, int row0, int64_t k0,
+template <int BR, class AG>
+__device__ __forceinline__ void StageSafe(const AG& ag, int row0, int64_t k0,
                                           int64_t rows, int64_t K,
                                           uint16_t* lds_base, int tid) {
 #pragma unroll
@@ -128,12 +124,11 @@ __device__ __forceinline__ void StageSafe(const uint16_t* __restrict__ src,
     int64_t row = row0 + r;
     int64_t kbase = k0 + c8 * 8;
     if (row < rows && kbase + 8 <= K) {
-      const uint16_t* g = src + row * ld + kbase;
-      *(ulong2*)vals = *(const ulong2*)g;
+      *(ulong2*)vals = *(const ulong2*)ag.at(row, kbase);
     } else if (row < rows && kbase < K) {
 #pragma unroll
       for (int e = 0; e < 8; ++e)
-        vals[e] = (kbase + e < K) ? src[row * ld + kbase + e] : 0;
+        vals[e] = (kbase + e < K) ? *ag.at(row, kbase + e) : 0;
     } else {
 #pragma unroll
       for (int e = 0; e < 8; ++e) vals[e] = 0;
@@ -154,13 +149,13 @@ __device__ __forceinline__ void StageSafe(const uint16_t* __restrict__ src,
 // buffering cannot warm up and cross-block overlap hides latency better.
 template <int WAVES_M, int WAVES_N, int WM, int WN, bool A_KM, bool B_KM,
           bool OUT_BF16, bool FUSE_RELU, bool SPLITK = false,
-          bool DBUF = true>
+          bool DBUF = true, class AAG = LinearAG>
 __launch_bounds__(256) __global__ void GemmBf16NT(
-    const uint16_t* __restrict__ A,
+    AAG a_ag,
     const uint16_t* __restrict__ B,
     void* __restrict__ C,            // [M, N] f32 or bf16 (f32 for SPLITK)
     const float* __restrict__ bias,  // optional [N] f32 bias (nullptr = none)
-    int64_t M, int64_t N, int64_t K, int64_t lda, int64_t ldb, float beta,
+    int64_t M, int64_t N, int64_t K, int64_t ldb, float beta,
     int splitk = 1) {
   constexpr int BM = WAVES_M * WM * 16;
   constexpr int BN = WAVES_N * WN * 16;
@@ -207,32 +202,33 @@ __launch_bounds__(256) __global__ void GemmBf16NT(
     if (kt_begin >= kt_count) return;
   }
   // Stage one K-chunk of both operands into LDS buffer `buf`.
+  LinearAG b_ag{B, ldb};
   auto stage = [&](int buf, int64_t k0) {
     bool kfull = k0 + BK <= K;
     if (A_KM) {
       if (a_interior && kfull)
-        StageKMajor<BM>(A, lda, (int)m0, k0, a_tile(buf), tid);
+        StageKMajor<BM>(a_ag, (int)m0, k0, a_tile(buf), tid);
       else
-        StageKMajorSafe<BM>(A, lda, (int)m0, k0, M, K, a_tile(buf), tid);
+        StageKMajorSafe<BM>(a_ag, (int)m0, k0, M, K, a_tile(buf), tid);
     } else {
       if (a_interior && kfull)
-        StageFast<BM>(A, lda, (int)m0, k0, a_tile(buf), tid);
+        StageFast<BM>(a_ag, (int)m0, k0, a_tile(buf), tid);
       else
-        StageSafe<BM>(A, lda, (int)m0, k0, M, K, a_tile(buf), tid);
+        StageSafe<BM>(a_ag, (int)m0, k0, M, K, a_tile(buf), tid);
     }
     if (B_KM) {
       // When A also staged K-major, A used threads [0,BM) — give B the next
       // BN threads so both operand stages run concurrently across waves.
       int bt = A_KM ? tid - BM : tid;
       if (b_interior && kfull)
-        StageKMajor<BN>(B, ldb, (int)n0, k0, b_tile(buf), bt);
+        StageKMajor<BN>(b_ag, (int)n0, k0, b_tile(buf), bt);
       else
-        StageKMajorSafe<BN>(B, ldb, (int)n0, k0, N, K, b_tile(buf), bt);
+        StageKMajorSafe<BN>(b_ag, (int)n0, k0, N, K, b_tile(buf), bt);
     } else {
       if (b_interior && kfull)
-        StageFast<BN>(B, ldb, (int)n0, k0, b_tile(buf), tid);
+        StageFast<BN>(b_ag, (int)n0, k0, b_tile(buf), tid);
       else
-        StageSafe<BN>(B, ldb, (int)n0, k0, N, K, b_tile(buf), tid);
+        StageSafe<BN>(b_ag, (int)n0, k0, N, K, b_tile(buf), tid);
     }
   };
 
@@ -355,10 +351,11 @@ hipError_t LaunchVariant(const uint16_t* A, const uint16_t* B, void* C,
                          hipStream_t stream) {
   static const bool no_sb = getenv("STF_GEMM_NO_SB") != nullptr;
   bool short_k = !no_sb && K <= 4 * 64;
+  LinearAG a_ag{A, lda};
   auto launch = [&](auto kern, int BM, int BN) {
     int64_t blocks = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
-    hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0, stream, A,
-                       B, C, bias, M, N, K, lda, ldb, beta, 1);
+    hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0, stream,
+                       a_ag, B, C, bias, M, N, K, ldb, beta, 1);
   };
 #define STF_PICK(WM_, WN_, TM, TN)                                           do {                                                                         if (short_k)                                                                 launch(GemmBf16NT<2, 2, WM_, WN_, A_KM, B_KM, OUT_BF16, FUSE_RELU,                           false, false>,                                                  TM, TN);                                                          else                                                                         launch(GemmBf16NT<2, 2, WM_, WN_, A_KM, B_KM, OUT_BF16, FUSE_RELU>,               TM, TN);                                                        } while (0)
   if (N >= 128 && M >= 128) {
@@ -378,10 +375,11 @@ template <bool A_KM, bool B_KM>
 hipError_t LaunchSplitK(const uint16_t* A, const uint16_t* B, float* C,
                         int64_t M, int64_t N, int64_t K, int64_t lda,
                         int64_t ldb, int splitk, hipStream_t stream) {
+  LinearAG a_ag{A, lda};
   auto launch = [&](auto kern, int BM, int BN) {
     int64_t blocks = ((M + BM - 1) / BM) * ((N + BN - 1) / BN) * splitk;
-    hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0, stream, A,
-                       B, C, nullptr, M, N, K, lda, ldb, 0.f, splitk);
+    hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0, stream,
+                       a_ag, B, C, nullptr, M, N, K, ldb, 0.f, splitk);
   };
   if (N >= 128 && M >= 128) {
     launch(GemmBf16NT<2, 2, 4, 4, A_KM, B_KM, false, false, true>, 128, 128);
@@ -492,4 +490,50 @@ extern "C" hipError_t stf_gemm_bf16_nt(const void* A, const void* B, void* C,
                                        hipStream_t stream) {
   return stf_gemm_bf16(A, B, C, bias_f32, M, N, K, K, K, beta, 0, 0,
                        out_bf16, fuse_relu, stream);
+}
+
+// Implicit-GEMM Conv2D backprop-filter: dW[rsc, Cout] = im2col(x)^T * dy,
+// contraction over the NPQ output pixels, split-K with f32 atomics. The
+// column matrix is generated inside the K-major A staging (ConvAG) — the
+// materialized im2col buffer and its cache disappear (reference analog:
+// conv_grad_filter_ops.cc ThenConvolveBackwardFilterWithAlgorithm).
+extern "C" hipError_t stf_conv2d_dw_splitk(
+    const void* x, const void* dy, void* dw_f32, const void* zero16, int n,
+    int h, int w, int c, int r, int s_, int sh, int sw, int ph, int pw,
+    int p, int q, int64_t cout, int splitk, hipStream_t stream) {
+  if ((c % 8) != 0) return hipErrorInvalidValue;
+  ConvAG ag;
+  ag.x = (const uint16_t*)x;
+  ag.zero16 = (const uint16_t*)zero16;
+  ag.div_pq.init((uint32_t)(p * q));
+  ag.div_q.init((uint32_t)q);
+  ag.div_c.init((uint32_t)c);
+  ag.div_s.init((uint32_t)s_);
+  ag.H = h; ag.W = w; ag.C = c; ag.S = s_;
+  ag.sh = sh; ag.sw = sw; ag.ph = ph; ag.pw = pw;
+  ag.rsc = (int64_t)r * s_ * c;
+  int64_t M = ag.rsc;            // dW rows
+  int64_t N = cout;
+  int64_t K = (int64_t)n * p * q;  // contraction: output pixels
+  const uint16_t* b = (const uint16_t*)dy;
+  float* cptr = (float*)dw_f32;
+  auto launch = [&](auto kern, int BM, int BN) {
+    int64_t blocks = ((M + BM - 1) / BM) * ((N + BN - 1) / BN) * splitk;
+    hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0, stream,
+                       ag, b, cptr, nullptr, M, N, K, cout, 0.f, splitk);
+  };
+  if (N >= 128 && M >= 128) {
+    launch(GemmBf16NT<2, 2, 4, 4, true, true, false, false, true, true,
+                      ConvAG>, 128, 128);
+  } else if (N >= 128) {
+    launch(GemmBf16NT<2, 2, 2, 4, true, true, false, false, true, true,
+                      ConvAG>, 64, 128);
+  } else if (M >= 128) {
+    launch(GemmBf16NT<2, 2, 4, 2, true, true, false, false, true, true,
+                      ConvAG>, 128, 64);
+  } else {
+    launch(GemmBf16NT<2, 2, 2, 2, true, true, false, false, true, true,
+                      ConvAG>, 64, 64);
+  }
+  return hipGetLastError();
 }

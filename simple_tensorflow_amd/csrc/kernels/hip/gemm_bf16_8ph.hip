@@ -41,17 +41,17 @@ constexpr int kThreads = 512;
 
 // Stage `ROWS` rows x 64 k with lane-linear glds (ROWS/64 passes of
 // 512 x 16B). The source column carries the inverse swizzle so the LDS
-// image is the swizzled one.
-template <int ROWS>
-__device__ __forceinline__ void Stage8(const uint16_t* __restrict__ src,
-                                       int64_t ld, int64_t row0, int64_t k0,
+// image is the swizzled one. AG maps (row, k) -> global pointer (plain
+// matrix or implicit-conv NHWC).
+template <int ROWS, class AG>
+__device__ __forceinline__ void Stage8(const AG& ag, int64_t row0, int64_t k0,
                                        uint16_t* lds_base, int tid) {
 #pragma unroll
   for (int p = 0; p < ROWS / 64; ++p) {
     int s = p * kThreads + tid;          // 16B slot
     int r = s >> 3;
     int c8 = (s & 7) ^ Swz8(r);
-    const uint16_t* g = src + (row0 + r) * ld + k0 + c8 * 8;
+    const uint16_t* g = ag.at(row0 + r, k0 + c8 * 8);
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) uint32_t*)g,
         (__attribute__((address_space(3))) uint32_t*)(lds_base + s * 8), 16,
@@ -60,11 +60,11 @@ __device__ __forceinline__ void Stage8(const uint16_t* __restrict__ src,
 }
 
 // NR = B fragments per wave; BN = 64*NR (wave grid fixed 2M x 4N).
-template <int NR, bool OUT_BF16, bool FUSE_RELU>
+template <int NR, bool OUT_BF16, bool FUSE_RELU, class AG>
 __launch_bounds__(512) __global__ void Gemm256x8Ph(
-    const uint16_t* __restrict__ A, const uint16_t* __restrict__ B,
-    void* __restrict__ C, const float* __restrict__ bias, int64_t M,
-    int64_t N, int64_t K, int64_t lda, int64_t ldb) {
+    AG ag, const uint16_t* __restrict__ B, void* __restrict__ C,
+    const float* __restrict__ bias, int64_t M, int64_t N, int64_t K,
+    int64_t ldb) {
   constexpr int kBN = 64 * NR;
   constexpr int kABytes = kBM * kBK * 2;       // 32 KiB per slot
   constexpr int kBBytes = kBN * kBK * 2;       // per slot
@@ -88,11 +88,11 @@ __launch_bounds__(512) __global__ void Gemm256x8Ph(
   int nkt = (int)(K / kBK);
 
   // ---- prologue: A(0), B(0) -> slot 0; B(1) -> slot 1 ----
-  Stage8<128>(A, lda, m0, 0, a_slot(0), tid);
-  Stage8<128>(A, lda, m0 + 128, 0, a_slot(0) + 128 * kBK, tid);
-  Stage8<kBN>(B, ldb, n0, 0, b_slot(0), tid);
+  Stage8<128>(ag, m0, 0, a_slot(0), tid);
+  Stage8<128>(ag, m0 + 128, 0, a_slot(0) + 128 * kBK, tid);
+  Stage8<kBN>(LinearAG{B, ldb}, n0, 0, b_slot(0), tid);
   if (nkt > 1) {
-    Stage8<kBN>(B, ldb, n0, kBK, b_slot(1), tid);
+    Stage8<kBN>(LinearAG{B, ldb}, n0, kBK, b_slot(1), tid);
     if (NR == 4) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     else if (NR == 2) asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
     else asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
@@ -140,7 +140,7 @@ __launch_bounds__(512) __global__ void Gemm256x8Ph(
             *(const bf16x8*)(at + (r * 8 + ((kk * 4 + kq) ^ Swz8(r))) * 8);
     }
     if (u + 1 < nkt)
-      Stage8<128>(A, lda, m0, (int64_t)(u + 1) * kBK, a_slot(slot ^ 1), tid);
+      Stage8<128>(ag, m0, (int64_t)(u + 1) * kBK, a_slot(slot ^ 1), tid);
     __builtin_amdgcn_s_barrier();
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_setprio(1);
@@ -167,27 +167,27 @@ __launch_bounds__(512) __global__ void Gemm256x8Ph(
       }
       if (q == 1) {
         if (u + 1 < nkt)
-          Stage8<128>(A, lda, m0 + 128, (int64_t)(u + 1) * kBK,
+          Stage8<128>(ag, m0 + 128, (int64_t)(u + 1) * kBK,
                       a_slot(slot ^ 1) + 128 * kBK, tid);
       } else if (q == 2) {
         if (NR == 4) {
           if (u + 2 < nkt)
-            Stage8<128>(B, ldb, n0, (int64_t)(u + 2) * kBK, b_slot(slot),
-                        tid);
+            Stage8<128>(LinearAG{B, ldb}, n0, (int64_t)(u + 2) * kBK,
+                        b_slot(slot), tid);
         }
       } else {
         if (u + 2 < nkt) {
           if (NR == 4) {
-            Stage8<128>(B, ldb, n0 + 128, (int64_t)(u + 2) * kBK,
+            Stage8<128>(LinearAG{B, ldb}, n0 + 128, (int64_t)(u + 2) * kBK,
                         b_slot(slot) + 128 * kBK, tid);
             asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
           } else if (NR == 2) {
-            Stage8<kBN>(B, ldb, n0, (int64_t)(u + 2) * kBK, b_slot(slot),
-                        tid);
+            Stage8<kBN>(LinearAG{B, ldb}, n0, (int64_t)(u + 2) * kBK,
+                        b_slot(slot), tid);
             asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
           } else {
-            Stage8<kBN>(B, ldb, n0, (int64_t)(u + 2) * kBK, b_slot(slot),
-                        tid);
+            Stage8<kBN>(LinearAG{B, ldb}, n0, (int64_t)(u + 2) * kBK,
+                        b_slot(slot), tid);
             asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
           }
         } else {
@@ -274,7 +274,7 @@ hipError_t stf_gemm_bf16_8ph(const void* A, const void* B, void* C,
                              int out_bf16, int fuse_relu,
                              hipStream_t stream) {
   if (!stf_gemm_bf16_8ph_ok(M, N, K)) return hipErrorInvalidValue;
-  const uint16_t* a = (const uint16_t*)A;
+  LinearAG ag{(const uint16_t*)A, lda};
   const uint16_t* b = (const uint16_t*)B;
   const float* bias = (const float*)bias_f32;
   int nr = (N % 256 == 0) ? 4 : (N % 128 == 0) ? 2 : 1;
@@ -282,9 +282,9 @@ hipError_t stf_gemm_bf16_8ph(const void* A, const void* B, void* C,
   // LDS: A ring 64 KiB + B ring 2 * (BN*64*2) bytes.
   size_t shmem = 64 * 1024 + 2 * (size_t)(64 * nr) * kBK * 2;
 #define STF_8PH_LAUNCH(NR_, OB, FR)                                          \
-  hipLaunchKernelGGL((Gemm256x8Ph<NR_, OB, FR>), dim3((uint32_t)blocks),     \
-                     dim3(kThreads), shmem, stream, a, b, C, bias, M, N, K, \
-                     lda, ldb)
+  hipLaunchKernelGGL((Gemm256x8Ph<NR_, OB, FR, LinearAG>),                   \
+                     dim3((uint32_t)blocks), dim3(kThreads), shmem, stream,  \
+                     ag, b, C, bias, M, N, K, ldb)
 #define STF_8PH_NR(NR_)                                                      \
   do {                                                                       \
     if (out_bf16) {                                                          \
@@ -300,6 +300,57 @@ hipError_t stf_gemm_bf16_8ph(const void* A, const void* B, void* C,
   else STF_8PH_NR(1);
 #undef STF_8PH_NR
 #undef STF_8PH_LAUNCH
+  return hipGetLastError();
+}
+
+// Implicit-GEMM Conv2D forward: y[M=NPQ, N=Cout] = im2col(x) * wt[Cout,rscp]^T
+// with the column matrix generated inside the A staging (ConvAG) — the
+// MI355X answer to the reference's cuDNN implicit-GEMM conv path
+// (conv_ops.cc:664 ThenConvolveWithAlgorithm). Requires C%8==0 (16B loads
+// stay within one (r,s) patch) and the 8-phase geometry gate.
+hipError_t stf_conv2d_fwd_8ph(const void* x, const void* wt, void* y,
+                              const void* bias_f32, const void* zero16,
+                              int n, int h, int w, int c, int r, int s_,
+                              int sh, int sw, int ph, int pw, int p, int q,
+                              int64_t cout, int64_t rscp, int out_bf16,
+                              int fuse_relu, hipStream_t stream) {
+  int64_t M = (int64_t)n * p * q;
+  if (!stf_gemm_bf16_8ph_ok(M, cout, rscp) || (c % 8) != 0)
+    return hipErrorInvalidValue;
+  ConvAG ag;
+  ag.x = (const uint16_t*)x;
+  ag.zero16 = (const uint16_t*)zero16;
+  ag.div_pq.init((uint32_t)(p * q));
+  ag.div_q.init((uint32_t)q);
+  ag.div_c.init((uint32_t)c);
+  ag.div_s.init((uint32_t)s_);
+  ag.H = h; ag.W = w; ag.C = c; ag.S = s_;
+  ag.sh = sh; ag.sw = sw; ag.ph = ph; ag.pw = pw;
+  ag.rsc = (int64_t)r * s_ * c;
+  const uint16_t* b = (const uint16_t*)wt;
+  const float* bias = (const float*)bias_f32;
+  int nr = (cout % 256 == 0) ? 4 : (cout % 128 == 0) ? 2 : 1;
+  int64_t blocks = (M / kBM) * (cout / (64 * nr));
+  size_t shmem = 64 * 1024 + 2 * (size_t)(64 * nr) * kBK * 2;
+#define STF_C8_LAUNCH(NR_, OB, FR)                                           \
+  hipLaunchKernelGGL((Gemm256x8Ph<NR_, OB, FR, ConvAG>),                     \
+                     dim3((uint32_t)blocks), dim3(kThreads), shmem, stream,  \
+                     ag, b, y, bias, M, cout, rscp, rscp)
+#define STF_C8_NR(NR_)                                                       \
+  do {                                                                       \
+    if (out_bf16) {                                                          \
+      if (fuse_relu) STF_C8_LAUNCH(NR_, true, true);                         \
+      else STF_C8_LAUNCH(NR_, true, false);                                  \
+    } else {                                                                 \
+      if (fuse_relu) STF_C8_LAUNCH(NR_, false, true);                        \
+      else STF_C8_LAUNCH(NR_, false, false);                                 \
+    }                                                                        \
+  } while (0)
+  if (nr == 4) STF_C8_NR(4);
+  else if (nr == 2) STF_C8_NR(2);
+  else STF_C8_NR(1);
+#undef STF_C8_NR
+#undef STF_C8_LAUNCH
   return hipGetLastError();
 }
 

@@ -54,3 +54,71 @@ __device__ __forceinline__ int XcdSwizzle(int bid, int nblocks) {
     hipError_t _e = hipGetLastError();                           \
     if (_e != hipSuccess) return _e;                             \
   } while (0)
+
+// Exact division by a launch-constant divisor via mul+shift (round-up
+// magic, Hacker's Delight §10-9: mul = ceil(2^(32+l)/d), exact for all
+// n < 2^31 with the 64-bit multiplier — mul can be 2^33, which is why it
+// must NOT be stored in 32 bits). Avoids ~30-instruction runtime division
+// sequences inside the implicit-GEMM address generators.
+struct U32Div {
+  unsigned long long mul = 1;
+  uint32_t shift = 0;
+  uint32_t d = 1;
+  void init(uint32_t div) {
+    d = div;
+    if (d <= 1) { mul = 1; shift = 0; return; }
+    uint32_t l = 0;
+    while ((1ull << l) < d) ++l;
+    shift = 32 + l;
+    mul = (unsigned long long)(((__uint128_t(1) << shift) + d - 1) / d);
+  }
+  __device__ __forceinline__ uint32_t div(uint32_t n) const {
+    if (d == 1) return n;
+    return (uint32_t)(((unsigned long long)n * mul) >> shift);
+  }
+  __device__ __forceinline__ uint32_t mod(uint32_t n, uint32_t q) const {
+    return n - q * d;
+  }
+};
+
+// A-operand address generators for the GEMM staging paths: the plain GEMM
+// reads a row-major matrix; the implicit-GEMM convolution maps
+// (row = output pixel, k = (r,s,c)) onto the NHWC input directly — the
+// im2col matrix is never materialized (reference conv_ops.cc would call
+// cuDNN implicit-GEMM here). Out-of-bounds (padding / rsc->rscp K padding)
+// resolves to a zeroed 16B page.
+struct LinearAG {
+  const uint16_t* base;
+  int64_t ld;
+  __device__ __forceinline__ const uint16_t* at(int64_t row,
+                                                int64_t k8) const {
+    return base + row * ld + k8;
+  }
+};
+
+struct ConvAG {
+  const uint16_t* x;           // NHWC input
+  const uint16_t* zero16;      // >=16 zero bytes
+  U32Div div_pq, div_q, div_c, div_s;
+  int H, W, C, S;
+  int sh, sw, ph, pw;
+  int64_t rsc;                 // un-padded K extent
+  // row = ((n*P)+p)*Q + q; k8 = 8-element-aligned (r,s,c) index.
+  __device__ __forceinline__ const uint16_t* at(int64_t row,
+                                                int64_t k8) const {
+    if (k8 >= rsc) return zero16;
+    uint32_t row32 = (uint32_t)row, k32 = (uint32_t)k8;
+    uint32_t n = div_pq.div(row32);
+    uint32_t pq = div_pq.mod(row32, n);
+    uint32_t p = div_q.div(pq);
+    uint32_t q = div_q.mod(pq, p);
+    uint32_t rs = div_c.div(k32);
+    uint32_t c = div_c.mod(k32, rs);
+    uint32_t r = div_s.div(rs);
+    uint32_t s_ = div_s.mod(rs, r);
+    int h = (int)p * sh - ph + (int)r;
+    int w = (int)q * sw - pw + (int)s_;
+    if (h < 0 || h >= H || w < 0 || w >= W) return zero16;
+    return x + ((((int64_t)n * H + h) * W + w) * C + c);
+  }
+};

@@ -162,3 +162,46 @@ def test_matmul_bf16_8phase_path_vs_torch():
                 torch.from_numpy(b).bfloat16().float().T).numpy()
         rel = np.abs(got - want) / (np.abs(want) + 1e-2)
         assert np.percentile(rel, 99) < 0.05, (m, n, k)
+
+
+def _torch_conv_ref(x, w, stride, padding):
+    tx = torch.from_numpy(x).permute(0, 3, 1, 2)
+    tw = torch.from_numpy(w).permute(3, 2, 0, 1)
+    if padding == 'SAME':
+        pad = (w.shape[0] - 1) // 2
+    else:
+        pad = 0
+    return torch.nn.functional.conv2d(tx, tw, stride=stride, padding=pad)
+
+
+def test_implicit_gemm_conv_vs_torch():
+    """Shapes eligible for the implicit-GEMM conv paths (C%8==0, NPQ%256==0,
+    Cout%64==0): fwd + backprop-input + backprop-filter vs torch autograd."""
+    rng = np.random.RandomState(11)
+    cases = [(4, 8, 16, 64, 1, 'SAME'), (4, 17, 8, 64, 2, 'VALID'),
+             (1, 16, 32, 128, 1, 'SAME')]
+    for (nb, hw, cin, cout, stride, padding) in cases:
+        tf.reset_default_graph()
+        x = (rng.randn(nb, hw, hw, cin) * 0.5).astype(np.float32)
+        w = (rng.randn(3, 3, cin, cout) * 0.2).astype(np.float32)
+        tx = torch.from_numpy(x).permute(0, 3, 1, 2).bfloat16().float() \
+            .requires_grad_(True)
+        tw = torch.from_numpy(w).permute(3, 2, 0, 1).bfloat16().float() \
+            .requires_grad_(True)
+        pad = 1 if padding == 'SAME' else 0
+        ty = torch.nn.functional.conv2d(tx, tw, stride=stride, padding=pad)
+        ty.backward(torch.ones_like(ty))
+        with tf.Session() as s:
+            xg = tf.constant(x, dtype=tf.bfloat16)
+            wg = tf.constant(w, dtype=tf.bfloat16)
+            y = tf.nn.conv2d(xg, wg, [1, stride, stride, 1], padding)
+            gx, gw = tf.gradients(tf.reduce_sum(y), [xg, wg])
+            got_y, got_gx, got_gw = s.run([y, gx, gw])
+        want_y = ty.detach().permute(0, 2, 3, 1).numpy()
+        want_gx = tx.grad.permute(0, 2, 3, 1).numpy()
+        want_gw = tw.grad.permute(2, 3, 1, 0).numpy()
+        scale = np.abs(want_y).max() + 1e-6
+        assert got_y.shape == want_y.shape, (got_y.shape, want_y.shape)
+        assert np.abs(got_y - want_y).max() / scale < 0.05, (cin, cout, stride)
+        assert np.abs(got_gx - want_gx).max() / (np.abs(want_gx).max() + 1e-6) < 0.05
+        assert np.abs(got_gw - want_gw).max() / (np.abs(want_gw).max() + 1e-6) < 0.05

@@ -194,6 +194,69 @@ class ShapeOp : public OpKernel {
 REGISTER_KERNEL_BUILDER(Name("Shape").Device(DEVICE_CPU), ShapeOp);
 REGISTER_KERNEL_BUILDER(Name("Shape").Device(DEVICE_GPU).HostMemory("output"), ShapeOp);
 
+// ConcatOffset: given concat_dim and N input shapes, emit each input's start
+// offset vector along concat_dim (reference concat_op.cc; used by the
+// dynamic-shape _ConcatGrad).
+class ConcatOffsetOp : public OpKernel {
+ public:
+  explicit ConcatOffsetOp(OpKernelConstruction* ctx) : OpKernel(ctx) {
+    set_expensive(false);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    int32_t cdim = ctx->input(0).flat<int32_t>()[0];
+    int n = num_inputs() - 1;
+    int rank = (int)ctx->input(1).NumElements();
+    if (cdim < 0) cdim += rank;
+    OP_REQUIRES(ctx, cdim >= 0 && cdim < rank,
+                errors::InvalidArgument("concat_dim out of range"));
+    int32_t running = 0;
+    for (int i = 0; i < n; ++i) {
+      const Tensor& shp = ctx->input(1 + i);
+      Tensor* out = ctx->allocate_output(i, shp.shape());
+      for (int d = 0; d < rank; ++d)
+        out->flat<int32_t>()[d] = (d == cdim) ? running : 0;
+      running += shp.flat<int32_t>()[cdim];
+    }
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("ConcatOffset").Device(DEVICE_CPU),
+                        ConcatOffsetOp);
+REGISTER_KERNEL_BUILDER(
+    Name("ConcatOffset").Device(DEVICE_GPU).HostMemory("concat_dim")
+        .HostMemory("shape").HostMemory("offset"), ConcatOffsetOp);
+
+class InvertPermutationOp : public OpKernel {
+ public:
+  explicit InvertPermutationOp(OpKernelConstruction* ctx) : OpKernel(ctx) {
+    set_expensive(false);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    int64_t n = x.NumElements();
+    Tensor* out = ctx->allocate_output(0, x.shape());
+    if (x.dtype() == DT_INT32) {
+      for (int64_t i = 0; i < n; ++i) {
+        int32_t p = x.flat<int32_t>()[i];
+        OP_REQUIRES(ctx, p >= 0 && p < n,
+                    errors::InvalidArgument("permutation out of range"));
+        out->flat<int32_t>()[p] = (int32_t)i;
+      }
+    } else {
+      for (int64_t i = 0; i < n; ++i) {
+        int64_t p = x.flat<int64_t>()[i];
+        OP_REQUIRES(ctx, p >= 0 && p < n,
+                    errors::InvalidArgument("permutation out of range"));
+        out->flat<int64_t>()[p] = i;
+      }
+    }
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("InvertPermutation").Device(DEVICE_CPU),
+                        InvertPermutationOp);
+REGISTER_KERNEL_BUILDER(Name("InvertPermutation").Device(DEVICE_GPU)
+                            .HostMemory("x").HostMemory("y"),
+                        InvertPermutationOp);
+
 class RankOp : public OpKernel {
  public:
   using OpKernel::OpKernel;

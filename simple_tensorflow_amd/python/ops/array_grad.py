@@ -38,30 +38,48 @@ def _unpack_grad(op, *grads):
     return array_ops.stack(filled, axis=axis)
 
 
+def _concat_grad_impl(op, grad, inputs, axis_t):
+    """Static fast path when every shape is known; otherwise the reference's
+    runtime-shape form (array_grad.py:34 _ConcatGrad): ShapeN + ConcatOffset
+    feeding Slice with runtime begin/size."""
+    axis_v = getattr(axis_t, '_const_value', None)
+    shapes = [t._shape for t in inputs]
+    if axis_v is not None and not any(
+            s is None or any(d is None for d in s) for s in shapes):
+        axis = int(axis_v) % len(shapes[0])
+        outs = []
+        offset = 0
+        for s in shapes:
+            begin = [0] * len(s)
+            begin[axis] = offset
+            size = list(s)
+            outs.append(array_ops.slice(grad, begin, size))
+            offset += s[axis]
+        return outs
+    from simple_tensorflow_amd.python.framework.ops import apply_op
+    sizes = array_ops.shape_n(inputs)
+    axis_i32 = math_ops.cast(axis_t, dtypes.int32)
+    offsets = apply_op('ConcatOffset', axis_i32, sizes)
+    if not isinstance(offsets, (list, tuple)):
+        offsets = (offsets,)
+    return [array_ops.slice(grad, o, sz) for o, sz in zip(offsets, sizes)]
+
+
 @RegisterGradient('ConcatV2')
 def _concat_v2_grad(op, grad):
-    n = len(op.inputs) - 1
-    axis_t = op.inputs[-1]
-    axis_v = getattr(axis_t, '_const_value', None)
-    shapes = [t._shape for t in op.inputs[:-1]]
-    if axis_v is None or any(s is None or any(d is None for d in s)
-                             for s in shapes):
-        raise NotImplementedError('Concat grad needs static shapes (round 1)')
-    axis = int(axis_v) % len(shapes[0])
-    outs = []
-    offset = 0
-    for s in shapes:
-        begin = [0] * len(s)
-        begin[axis] = offset
-        size = list(s)
-        outs.append(array_ops.slice(grad, begin, size))
-        offset += s[axis]
-    return outs + [None]
+    if len(op.inputs) == 2:  # one input + axis: pass-through
+        return [grad, None]
+    return _concat_grad_impl(op, grad, list(op.inputs[:-1]),
+                             op.inputs[-1]) + [None]
 
 
 @RegisterGradient('Concat')
 def _concat_grad(op, grad):
-    raise NotImplementedError('legacy Concat grad')
+    # legacy form: concat_dim is input 0
+    if len(op.inputs) == 2:
+        return [None, grad]
+    return [None] + _concat_grad_impl(op, grad, list(op.inputs[1:]),
+                                      op.inputs[0])
 
 
 @RegisterGradient('Split')
@@ -78,11 +96,20 @@ def _slice_grad(op, grad):
     x = op.inputs[0]
     begin = op.inputs[1]
     bv = getattr(begin, '_const_value', None)
-    if x._shape is None or any(d is None for d in x._shape) or bv is None or \
-            grad._shape is None or any(d is None for d in grad._shape):
-        raise NotImplementedError('Slice grad needs static shapes (round 1)')
-    paddings = [[int(b), int(xs) - int(b) - int(gs)]
-                for b, xs, gs in zip(bv.reshape(-1), x._shape, grad._shape)]
+    if not (x._shape is None or any(d is None for d in x._shape) or
+            bv is None or grad._shape is None or
+            any(d is None for d in grad._shape)):
+        paddings = [[int(b), int(xs) - int(b) - int(gs)]
+                    for b, xs, gs in zip(bv.reshape(-1), x._shape,
+                                         grad._shape)]
+        return [array_ops.pad(grad, paddings), None, None]
+    # Runtime-shape form (reference _SliceGrad): paddings built from
+    # shape(x), begin and shape(grad) at execution time.
+    begin32 = math_ops.cast(begin, dtypes.int32)
+    before = array_ops.reshape(begin32, [-1, 1])
+    after = array_ops.reshape(
+        array_ops.shape(x) - begin32 - array_ops.shape(grad), [-1, 1])
+    paddings = array_ops.concat([before, after], 1)
     return [array_ops.pad(grad, paddings), None, None]
 
 
@@ -90,42 +117,64 @@ def _slice_grad(op, grad):
 def _pad_grad(op, grad):
     pv = getattr(op.inputs[1], '_const_value', None)
     x = op.inputs[0]
-    if pv is None or x._shape is None or any(d is None for d in x._shape):
-        raise NotImplementedError('Pad grad needs static shapes (round 1)')
-    pv = pv.reshape(-1, 2)
-    begin = [int(p[0]) for p in pv]
-    size = list(x._shape)
-    return [array_ops.slice(grad, begin, size), None]
+    if pv is not None and x._shape is not None and \
+            not any(d is None for d in x._shape):
+        pv = pv.reshape(-1, 2)
+        begin = [int(p[0]) for p in pv]
+        size = list(x._shape)
+        return [array_ops.slice(grad, begin, size), None]
+    # Runtime form: begin = paddings[:, 0], size = shape(x).
+    pads = math_ops.cast(op.inputs[1], dtypes.int32)
+    begin = array_ops.reshape(
+        array_ops.slice(pads, [0, 0], [-1, 1]), [-1])
+    return [array_ops.slice(grad, begin, array_ops.shape(x)), None]
 
 
 @RegisterGradient('Transpose')
 def _transpose_grad(op, grad):
     perm = op.inputs[1]
     pv = getattr(perm, '_const_value', None)
-    if pv is None:
-        raise NotImplementedError('Transpose grad needs static perm')
-    inv = [0] * len(pv.reshape(-1))
-    for i, p in enumerate(pv.reshape(-1)):
-        inv[int(p)] = i
-    return [array_ops.transpose(grad, inv), None]
+    if pv is not None:
+        inv = [0] * len(pv.reshape(-1))
+        for i, p in enumerate(pv.reshape(-1)):
+            inv[int(p)] = i
+        return [array_ops.transpose(grad, inv), None]
+    from simple_tensorflow_amd.python.framework.ops import apply_op
+    return [array_ops.transpose(grad, apply_op('InvertPermutation', perm)),
+            None]
 
 
 @RegisterGradient('Tile')
 def _tile_grad(op, grad):
     x = op.inputs[0]
     mv = getattr(op.inputs[1], '_const_value', None)
-    if mv is None or x._shape is None or any(d is None for d in x._shape):
-        raise NotImplementedError('Tile grad needs static shapes')
-    # reshape to [m0, s0, m1, s1, ...] and sum over the m axes
-    interleaved = []
-    sum_axes = []
-    for i, (m, s) in enumerate(zip(mv.reshape(-1), x._shape)):
-        sum_axes.append(len(interleaved))
-        interleaved.append(int(m))
-        interleaved.append(int(s))
-    g = array_ops.reshape(grad, interleaved)
-    out = math_ops.reduce_sum(g, sum_axes)
-    out.set_shape(list(x._shape))
+    if mv is not None and x._shape is not None and \
+            not any(d is None for d in x._shape):
+        # reshape to [m0, s0, m1, s1, ...] and sum over the m axes
+        interleaved = []
+        sum_axes = []
+        for i, (m, s) in enumerate(zip(mv.reshape(-1), x._shape)):
+            sum_axes.append(len(interleaved))
+            interleaved.append(int(m))
+            interleaved.append(int(s))
+        g = array_ops.reshape(grad, interleaved)
+        out = math_ops.reduce_sum(g, sum_axes)
+        out.set_shape(list(x._shape))
+        return [out, None]
+    # Runtime form (reference _TileGrad): interleave multiples with the
+    # input shape; the reduction axes 0,2,4,.. only need the static rank.
+    rank = len(x._shape) if x._shape is not None else None
+    if rank is None:
+        raise NotImplementedError('Tile grad needs a static rank')
+    in_shape = array_ops.shape(x)
+    mult = math_ops.cast(op.inputs[1], dtypes.int32)
+    split_shape = array_ops.reshape(
+        array_ops.transpose(array_ops.stack([mult, in_shape])), [-1])
+    g = array_ops.reshape(grad, split_shape)
+    out = math_ops.reduce_sum(g, list(range(0, 2 * rank, 2)))
+    out = array_ops.reshape(out, in_shape)
+    if x._shape is not None:
+        out.set_shape(list(x._shape))
     return [out, None]
 
 
@@ -133,14 +182,24 @@ def _tile_grad(op, grad):
 def _gather_grad(op, grad):
     params = op.inputs[0]
     indices = op.inputs[1]
-    if params._shape is None or params._shape[0] is None:
-        raise NotImplementedError('Gather grad needs static param rows')
-    num_rows = params._shape[0]
     flat_idx = array_ops.reshape(indices, [-1])
-    gshape = [-1] + list(params._shape[1:])
+    if params._shape is not None and params._shape[0] is not None:
+        num_rows = params._shape[0]
+        gshape = [-1] + list(params._shape[1:]) \
+            if not any(d is None for d in params._shape[1:]) else None
+        if gshape is not None:
+            flat_grad = array_ops.reshape(grad, gshape)
+            out = array_ops.unsorted_segment_sum(flat_grad, flat_idx,
+                                                 num_rows)
+            out.set_shape(list(params._shape))
+            return [out, None]
+    # Runtime form: rows from shape(params) at execution time.
+    pshape = array_ops.shape(params)
+    num_rows = array_ops.reshape(array_ops.slice(pshape, [0], [1]), [])
+    inner = array_ops.slice(pshape, [1], [-1])
+    gshape = array_ops.concat([ops.constant([-1]), inner], 0)
     flat_grad = array_ops.reshape(grad, gshape)
     out = array_ops.unsorted_segment_sum(flat_grad, flat_idx, num_rows)
-    out.set_shape(list(params._shape))
     return [out, None]
 
 

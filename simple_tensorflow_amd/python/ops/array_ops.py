@@ -301,8 +301,11 @@ def transpose(x, perm=None, name=None):
         perm = list(range(nd))[::-1]
     t = apply_op('Transpose', x, convert_to_tensor(perm, dtype=dtypes.int32),
                  name=name)
-    if x._shape is not None and not isinstance(perm, ops.Tensor):
-        t.set_shape([x._shape[p] for p in perm])
+    if x._shape is not None:
+        if not isinstance(perm, ops.Tensor):
+            t.set_shape([x._shape[p] for p in perm])
+        else:
+            t.set_shape([None] * len(x._shape))  # rank is permutation-invariant
     return t
 
 
@@ -316,8 +319,19 @@ def gather(params, indices, name=None):
 
 
 def tile(x, multiples, name=None):
-    return apply_op('Tile', convert_to_tensor(x),
-                    convert_to_tensor(multiples, dtype=dtypes.int32), name=name)
+    xt = convert_to_tensor(x)
+    mt = convert_to_tensor(multiples, dtype=dtypes.int32)
+    t = apply_op('Tile', xt, mt, name=name)
+    # Shape inference: per-dim product where both factors are known; rank is
+    # known whenever the input rank (or a static multiples vector) is.
+    mv = getattr(mt, '_const_value', None)
+    if xt._shape is not None:
+        if mv is not None:
+            t.set_shape([None if d is None else int(d) * int(m)
+                         for d, m in zip(xt._shape, mv.reshape(-1))])
+        else:
+            t.set_shape([None] * len(xt._shape))
+    return t
 
 
 def one_hot(indices, depth, on_value=1.0, off_value=0.0, axis=-1,

@@ -209,6 +209,29 @@ eye = _array_ops.eye
 meshgrid = _array_ops.meshgrid
 reverse = _array_ops.reverse
 reverse_v2 = _array_ops.reverse_v2
+unique = _array_ops.unique
+unique_with_counts = _array_ops.unique_with_counts
+setdiff1d = _array_ops.setdiff1d
+dynamic_partition = _array_ops.dynamic_partition
+gather_nd = _array_ops.gather_nd
+scatter_nd = _array_ops.scatter_nd
+diag = _array_ops.diag
+diag_part = _array_ops.diag_part
+matrix_diag = _array_ops.matrix_diag
+matrix_diag_part = _array_ops.matrix_diag_part
+matrix_set_diag = _array_ops.matrix_set_diag
+matrix_band_part = _array_ops.matrix_band_part
+space_to_depth = _array_ops.space_to_depth
+depth_to_space = _array_ops.depth_to_space
+mirror_pad = _array_ops.mirror_pad
+reverse_sequence = _array_ops.reverse_sequence
+bitcast = _array_ops.bitcast
+cumprod = _math_ops.cumprod
+segment_sum = _math_ops.segment_sum
+segment_mean = _math_ops.segment_mean
+segment_max = _math_ops.segment_max
+segment_min = _math_ops.segment_min
+segment_prod = _math_ops.segment_prod
 case = _control_flow_ops.case
 global_norm = _clip_ops.global_norm
 clip_by_value = _clip_ops.clip_by_value

@@ -268,4 +268,48 @@ REGISTER_OP("RcclBroadcast").Input("input: T").Output("output: T").Attr("T: {flo
 REGISTER_OP("RcclBucketAllReduce").Input("inputs: N * T").Output("outputs: N * T").Attr("N: int >= 1").Attr("T: {float, bfloat16}").Attr("scale: float = 1").SetIsStateful();
 REGISTER_OP("RcclCommSync").SetIsStateful();
 
+// ------------------------ breadth wave (round 2) ---------------------------
+// search / index
+REGISTER_OP("Where").Input("input: bool").Output("index: int64");
+REGISTER_OP("Unique").Input("x: T").Output("y: T").Output("idx: out_idx").Attr("T: type").Attr("out_idx: {int32, int64} = int32");
+REGISTER_OP("UniqueWithCounts").Input("x: T").Output("y: T").Output("idx: out_idx").Output("count: out_idx").Attr("T: type").Attr("out_idx: {int32, int64} = int32");
+REGISTER_OP("TopKV2").Input("input: T").Input("k: int32").Output("values: T").Output("indices: int32").Attr("sorted: bool = true").Attr("T: " REALTYPES);
+REGISTER_OP("TopK").Input("input: T").Output("values: T").Output("indices: int32").Attr("k: int >= 0").Attr("sorted: bool = true").Attr("T: " REALTYPES);
+// scan / segments
+REGISTER_OP("Cumprod").Input("x: T").Input("axis: Tidx").Output("out: T").Attr("exclusive: bool = false").Attr("reverse: bool = false").Attr("T: " NUMTYPES).Attr("Tidx: {int32, int64} = int32");
+#define SEGMENT_OP(NAME) REGISTER_OP(NAME).Input("data: T").Input("segment_ids: Tindices").Output("output: T").Attr("T: " NUMTYPES).Attr("Tindices: {int32, int64}")
+SEGMENT_OP("SegmentSum");
+SEGMENT_OP("SegmentMean");
+SEGMENT_OP("SegmentMax");
+SEGMENT_OP("SegmentMin");
+SEGMENT_OP("SegmentProd");
+#undef SEGMENT_OP
+// array restructuring
+REGISTER_OP("ReverseV2").Input("tensor: T").Input("axis: Tidx").Output("output: T").Attr("T: type").Attr("Tidx: {int32, int64} = int32");
+REGISTER_OP("ListDiff").Input("x: T").Input("y: T").Output("out: T").Output("idx: out_idx").Attr("T: type").Attr("out_idx: {int32, int64} = int32");
+REGISTER_OP("DynamicPartition").Input("data: T").Input("partitions: int32").Output("outputs: num_partitions * T").Attr("num_partitions: int >= 1").Attr("T: type");
+REGISTER_OP("GatherNd").Input("params: Tparams").Input("indices: Tindices").Output("output: Tparams").Attr("Tparams: type").Attr("Tindices: {int32, int64}");
+REGISTER_OP("ScatterNd").Input("indices: Tindices").Input("updates: T").Input("shape: Tindices").Output("output: T").Attr("T: " NUMTYPES).Attr("Tindices: {int32, int64}");
+REGISTER_OP("Diag").Input("diagonal: T").Output("output: T").Attr("T: " NUMTYPES);
+REGISTER_OP("DiagPart").Input("input: T").Output("diagonal: T").Attr("T: " NUMTYPES);
+REGISTER_OP("MatrixDiag").Input("diagonal: T").Output("output: T").Attr("T: type");
+REGISTER_OP("BatchMatrixDiag").Input("diagonal: T").Output("output: T").Attr("T: type");
+REGISTER_OP("MatrixDiagPart").Input("input: T").Output("diagonal: T").Attr("T: type");
+REGISTER_OP("BatchMatrixDiagPart").Input("input: T").Output("diagonal: T").Attr("T: type");
+REGISTER_OP("MatrixSetDiag").Input("input: T").Input("diagonal: T").Output("output: T").Attr("T: type");
+REGISTER_OP("BatchMatrixSetDiag").Input("input: T").Input("diagonal: T").Output("output: T").Attr("T: type");
+REGISTER_OP("MatrixBandPart").Input("input: T").Input("num_lower: int64").Input("num_upper: int64").Output("band: T").Attr("T: " NUMTYPES);
+REGISTER_OP("BatchMatrixBandPart").Input("input: T").Input("num_lower: int64").Input("num_upper: int64").Output("band: T").Attr("T: " NUMTYPES);
+REGISTER_OP("SpaceToDepth").Input("input: T").Output("output: T").Attr("T: type").Attr("block_size: int >= 2");
+REGISTER_OP("DepthToSpace").Input("input: T").Output("output: T").Attr("T: type").Attr("block_size: int >= 2");
+REGISTER_OP("MirrorPad").Input("input: T").Input("paddings: Tpaddings").Output("output: T").Attr("T: type").Attr("Tpaddings: {int32, int64} = int32").Attr("mode: string");
+REGISTER_OP("ReverseSequence").Input("input: T").Input("seq_lengths: Tlen").Output("output: T").Attr("seq_dim: int").Attr("batch_dim: int = 0").Attr("T: " NUMTYPES).Attr("Tlen: {int32, int64} = int64");
+REGISTER_OP("Bitcast").Input("input: T").Output("output: type").Attr("T: " NUMTYPES).Attr("type: " NUMTYPES);
+// variable scatter updates
+#define SCATTER_VAR_OP(NAME) REGISTER_OP(NAME).Input("ref: Ref(T)").Input("indices: Tindices").Input("updates: T").Output("output_ref: Ref(T)").Attr("T: " NUMTYPES).Attr("Tindices: {int32, int64}").Attr("use_locking: bool = true")
+SCATTER_VAR_OP("ScatterUpdate");
+SCATTER_VAR_OP("ScatterMul");
+SCATTER_VAR_OP("ScatterDiv");
+#undef SCATTER_VAR_OP
+
 }  // namespace stf

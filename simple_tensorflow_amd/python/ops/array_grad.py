@@ -282,3 +282,101 @@ def _ta_read_grad(op, grad):
 @RegisterGradient('TensorArrayWriteV3')
 def _ta_write_grad(op, grad):
     return [None, None, None, None]
+
+
+# ---------------------------------------------------------------------------
+# round-2 breadth-wave gradients
+# ---------------------------------------------------------------------------
+@RegisterGradient('GatherNd')
+def _gather_nd_grad(op, grad):
+    from simple_tensorflow_amd.python.framework.ops import apply_op
+    pshape = array_ops.shape(op.inputs[0])
+    out = apply_op('ScatterNd', op.inputs[1], grad, pshape)
+    return [out, None]
+
+
+@RegisterGradient('ScatterNd')
+def _scatter_nd_grad(op, grad):
+    return [None, array_ops.gather_nd(grad, op.inputs[0]), None]
+
+
+@RegisterGradient('ReverseV2')
+def _reverse_v2_grad(op, grad):
+    return [array_ops.reverse_v2(grad, op.inputs[1]), None]
+
+
+@RegisterGradient('ReverseSequence')
+def _reverse_sequence_grad(op, grad):
+    from simple_tensorflow_amd.python.framework.ops import apply_op
+    return [apply_op('ReverseSequence', grad, op.inputs[1],
+                     seq_dim=op.get_attr('seq_dim'),
+                     batch_dim=op.get_attr('batch_dim')), None]
+
+
+@RegisterGradient('Diag')
+def _diag_grad(op, grad):
+    return array_ops.diag_part(grad)
+
+
+@RegisterGradient('DiagPart')
+def _diag_part_grad(op, grad):
+    return array_ops.diag(grad)
+
+
+@RegisterGradient('MatrixDiag')
+def _matrix_diag_grad(op, grad):
+    return array_ops.matrix_diag_part(grad)
+
+
+@RegisterGradient('MatrixDiagPart')
+def _matrix_diag_part_grad(op, grad):
+    return array_ops.matrix_diag(grad)
+
+
+@RegisterGradient('MatrixSetDiag')
+def _matrix_set_diag_grad(op, grad):
+    diag_grad = array_ops.matrix_diag_part(grad)
+    input_grad = array_ops.matrix_set_diag(
+        grad, array_ops.zeros_like(diag_grad))
+    return [input_grad, diag_grad]
+
+
+@RegisterGradient('MatrixBandPart')
+def _matrix_band_part_grad(op, grad):
+    return [array_ops.matrix_band_part(grad, op.inputs[1], op.inputs[2]),
+            None, None]
+
+
+@RegisterGradient('SpaceToDepth')
+def _space_to_depth_grad(op, grad):
+    return array_ops.depth_to_space(grad, op.get_attr('block_size'))
+
+
+@RegisterGradient('DepthToSpace')
+def _depth_to_space_grad(op, grad):
+    return array_ops.space_to_depth(grad, op.get_attr('block_size'))
+
+
+@RegisterGradient('DynamicPartition')
+def _dynamic_partition_grad(op, *grads):
+    """Stitch the per-partition grads back (reference
+    data_flow_grad.py _DynamicPartitionGrads)."""
+    from simple_tensorflow_amd.python.framework.ops import apply_op
+    data = op.inputs[0]
+    partitions = op.inputs[1]
+    num_partitions = op.get_attr('num_partitions')
+    prefix_shape = array_ops.shape(partitions)
+    original_indices = array_ops.reshape(
+        math_ops.range(0, math_ops.reduce_prod(prefix_shape)), prefix_shape)
+    partitioned_indices = array_ops.dynamic_partition(
+        original_indices, partitions, num_partitions)
+    filled = [g if g is not None else array_ops.zeros_like(op.outputs[i])
+              for i, g in enumerate(grads)]
+    reconstructed = apply_op('DynamicStitch',
+                             list(partitioned_indices), list(filled))
+    return [array_ops.reshape(reconstructed, array_ops.shape(data)), None]
+
+
+for _op in ('Where', 'Unique', 'UniqueWithCounts', 'ListDiff', 'Bitcast',
+            'ConcatOffset'):
+    ops.NoGradient(_op)

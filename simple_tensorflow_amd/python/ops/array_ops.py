@@ -350,7 +350,10 @@ def one_hot(indices, depth, on_value=1.0, off_value=0.0, axis=-1,
 
 def where(condition, x=None, y=None, name=None):
     if x is None and y is None:
-        raise NotImplementedError('tf.where without x/y')
+        # coordinate form: [n, rank] int64 indices of true elements
+        return apply_op('Where', convert_to_tensor(condition), name=name)
+    if (x is None) != (y is None):
+        raise ValueError('x and y must both be set or both be None')
     return apply_op('Select', convert_to_tensor(condition),
                     convert_to_tensor(x), convert_to_tensor(y), name=name)
 
@@ -417,3 +420,102 @@ def reverse(tensor, axis, name=None):
 
 
 reverse_v2 = reverse
+
+
+# ---------------------------------------------------------------------------
+# round-2 breadth wave wrappers
+# ---------------------------------------------------------------------------
+def unique(x, out_idx=dtypes.int32, name=None):
+    return apply_op('Unique', convert_to_tensor(x), name=name)
+
+
+def unique_with_counts(x, out_idx=dtypes.int32, name=None):
+    return apply_op('UniqueWithCounts', convert_to_tensor(x), name=name)
+
+
+def reverse_v2(tensor, axis, name=None):
+    return apply_op('ReverseV2', convert_to_tensor(tensor),
+                    convert_to_tensor(axis, dtype=dtypes.int32), name=name)
+
+
+def setdiff1d(x, y, index_dtype=dtypes.int32, name=None):
+    return apply_op('ListDiff', convert_to_tensor(x), convert_to_tensor(y),
+                    name=name)
+
+
+def dynamic_partition(data, partitions, num_partitions, name=None):
+    out = apply_op('DynamicPartition', convert_to_tensor(data),
+                   convert_to_tensor(partitions, dtype=dtypes.int32),
+                   num_partitions=num_partitions, name=name)
+    return list(out) if isinstance(out, tuple) else [out]
+
+
+def gather_nd(params, indices, name=None):
+    return apply_op('GatherNd', convert_to_tensor(params),
+                    convert_to_tensor(indices, dtype=dtypes.int32), name=name)
+
+
+def scatter_nd(indices, updates, shape, name=None):
+    return apply_op('ScatterNd',
+                    convert_to_tensor(indices, dtype=dtypes.int32),
+                    convert_to_tensor(updates),
+                    convert_to_tensor(shape, dtype=dtypes.int32), name=name)
+
+
+def diag(diagonal, name=None):
+    return apply_op('Diag', convert_to_tensor(diagonal), name=name)
+
+
+def diag_part(input_, name=None):
+    return apply_op('DiagPart', convert_to_tensor(input_), name=name)
+
+
+def matrix_diag(diagonal, name=None):
+    return apply_op('MatrixDiag', convert_to_tensor(diagonal), name=name)
+
+
+def matrix_diag_part(input_, name=None):
+    return apply_op('MatrixDiagPart', convert_to_tensor(input_), name=name)
+
+
+def matrix_set_diag(input_, diagonal, name=None):
+    return apply_op('MatrixSetDiag', convert_to_tensor(input_),
+                    convert_to_tensor(diagonal), name=name)
+
+
+def matrix_band_part(input_, num_lower, num_upper, name=None):
+    return apply_op('MatrixBandPart', convert_to_tensor(input_),
+                    convert_to_tensor(num_lower, dtype=dtypes.int64),
+                    convert_to_tensor(num_upper, dtype=dtypes.int64),
+                    name=name)
+
+
+def space_to_depth(input_, block_size, name=None):
+    return apply_op('SpaceToDepth', convert_to_tensor(input_),
+                    block_size=block_size, name=name)
+
+
+def depth_to_space(input_, block_size, name=None):
+    return apply_op('DepthToSpace', convert_to_tensor(input_),
+                    block_size=block_size, name=name)
+
+
+def mirror_pad(tensor, paddings, mode, name=None):
+    return apply_op('MirrorPad', convert_to_tensor(tensor),
+                    convert_to_tensor(paddings, dtype=dtypes.int32),
+                    mode=mode, name=name)
+
+
+def reverse_sequence(input_, seq_lengths, seq_axis=None, batch_axis=None,
+                     seq_dim=None, batch_dim=None, name=None):
+    seq_dim = seq_axis if seq_axis is not None else seq_dim
+    batch_dim = batch_axis if batch_axis is not None else (batch_dim or 0)
+    lens = seq_lengths if isinstance(seq_lengths, ops.Tensor) else \
+        convert_to_tensor(seq_lengths, dtype=dtypes.int64)
+    return apply_op('ReverseSequence', convert_to_tensor(input_), lens,
+                    seq_dim=seq_dim, batch_dim=batch_dim, name=name)
+
+
+def bitcast(input_, type, name=None):  # pylint: disable=redefined-builtin
+    return apply_op('Bitcast', convert_to_tensor(input_),
+                    type=dtypes.as_dtype(type), name=name)

@@ -326,3 +326,69 @@ for _op in ('Less', 'LessEqual', 'Greater', 'GreaterEqual', 'Equal',
             'Rank', 'Size', 'IsNan', 'IsInf', 'IsFinite',
             'BroadcastGradientArgs', 'InTopK'):
     ops.NoGradient(_op)
+
+
+# ---------------------------------------------------------------------------
+# round-2 breadth-wave gradients
+# ---------------------------------------------------------------------------
+@RegisterGradient('Cumsum')
+def _cumsum_grad(op, grad):
+    axis = op.inputs[1]
+    exclusive = op.get_attr('exclusive')
+    reverse = op.get_attr('reverse')
+    return [math_ops.cumsum(grad, axis, exclusive=exclusive,
+                            reverse=not reverse), None]
+
+
+@RegisterGradient('Cumprod')
+def _cumprod_grad(op, grad):
+    # Reference _CumprodGrad form (zero-input caveat matches the reference).
+    x = op.inputs[0]
+    axis = op.inputs[1]
+    exclusive = op.get_attr('exclusive')
+    reverse = op.get_attr('reverse')
+    prod = math_ops.cumprod(x, axis, exclusive=exclusive, reverse=reverse)
+    out = math_ops.cumsum(prod * grad, axis, exclusive=exclusive,
+                          reverse=not reverse)
+    return [out / x, None]
+
+
+@RegisterGradient('SegmentSum')
+def _segment_sum_grad(op, grad):
+    return [array_ops.gather(grad, op.inputs[1]), None]
+
+
+@RegisterGradient('SegmentMean')
+def _segment_mean_grad(op, grad):
+    # d/dx mean_seg = gather(grad / count, ids): each row of a segment gets
+    # the segment grad divided by the segment population.
+    ids = op.inputs[1]
+    ones = array_ops.ones_like(math_ops.cast(ids, grad.dtype))
+    counts = math_ops.segment_sum(ones, ids)          # [nseg]
+    g = array_ops.gather(grad, ids)                   # [n, ...]
+    s = array_ops.gather(math_ops.reciprocal(counts), ids)  # [n]
+    data_shape = op.inputs[0]._shape
+    rank = len(data_shape) if data_shape is not None else 1
+    if rank > 1:
+        s = array_ops.reshape(s, [-1] + [1] * (rank - 1))
+    return [g * s, None]
+
+
+def _segment_minmax_grad(op, grad):
+    ids = op.inputs[1]
+    gathered_out = array_ops.gather(op.outputs[0], ids)
+    mask = math_ops.cast(math_ops.equal(op.inputs[0], gathered_out),
+                         grad.dtype)
+    num_sel = math_ops.segment_sum(mask, ids)
+    gathered = array_ops.gather(grad / num_sel, ids)
+    return [gathered * mask, None]
+
+
+@RegisterGradient('SegmentMax')
+def _segment_max_grad(op, grad):
+    return _segment_minmax_grad(op, grad)
+
+
+@RegisterGradient('SegmentMin')
+def _segment_min_grad(op, grad):
+    return _segment_minmax_grad(op, grad)

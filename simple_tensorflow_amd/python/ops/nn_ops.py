@@ -199,6 +199,14 @@ def in_top_k(predictions, targets, k, name=None):
     return apply_op('InTopK', predictions, targets, k=k, name=name)
 
 
+def top_k(input, k=1, sorted=True, name=None):  # pylint: disable=redefined-builtin
+    from simple_tensorflow_amd.python.framework import dtypes as _dt
+    from simple_tensorflow_amd.python.framework.ops import convert_to_tensor
+    return apply_op('TopKV2', convert_to_tensor(input),
+                    convert_to_tensor(k, dtype=_dt.int32), sorted=sorted,
+                    name=name)
+
+
 def lrn(input, depth_radius=5, bias=1.0, alpha=1.0, beta=0.5, name=None):  # pylint: disable=redefined-builtin
     t = apply_op('LRN', input, depth_radius=depth_radius, bias=bias,
                  alpha=alpha, beta=beta, name=name)

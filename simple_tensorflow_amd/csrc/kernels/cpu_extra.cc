@@ -846,5 +846,194 @@ REG_SCATTER("ScatterMul", 1)
 REG_SCATTER("ScatterDiv", 2)
 #undef REG_SCATTER
 
+// ------------------------------ StridedSlice --------------------------------
+// Full reference semantics (core/util/strided_slice_op.cc redesigned):
+// sparse spec (begin/end/strides + 5 masks) canonicalized to one dense
+// (begin,end,stride) per input dim plus the final shape (new_axis dims
+// inserted, shrink dims removed). The element ORDER of the processing
+// output equals the final-shape output, so the kernel writes the output
+// tensor directly.
+namespace strided {
+
+struct Spec {
+  std::vector<int64_t> begin, end, stride;  // dense, one per input dim
+  std::vector<int64_t> proc_dims;           // extent per input dim
+  std::vector<int64_t> final_dims;          // output shape
+  int64_t total = 1;
+};
+
+inline Status Build(const TensorShape& in, const std::vector<int64_t>& b,
+                    const std::vector<int64_t>& e,
+                    const std::vector<int64_t>& st, int64_t begin_mask,
+                    int64_t end_mask, int64_t ellipsis_mask,
+                    int64_t new_axis_mask, int64_t shrink_mask, Spec* out) {
+  int n = (int)b.size();
+  int rank = in.dims();
+  // count spec entries that consume an input dim
+  int consuming = 0;
+  for (int i = 0; i < n; ++i)
+    if (!((new_axis_mask >> i) & 1) && !((ellipsis_mask >> i) & 1))
+      ++consuming;
+  int ell_dims = rank - consuming;  // dims covered by the ellipsis
+  if (ell_dims < 0) return errors::InvalidArgument("too many slice specs");
+  bool has_ellipsis = ellipsis_mask != 0;
+  int dim = 0;  // current input dim
+  auto add_full = [&](int64_t) {
+    out->begin.push_back(0);
+    out->end.push_back(in.dim_size(dim));
+    out->stride.push_back(1);
+    int64_t ext = in.dim_size(dim);
+    out->proc_dims.push_back(ext);
+    out->final_dims.push_back(ext);
+    ++dim;
+  };
+  for (int i = 0; i < n; ++i) {
+    if ((ellipsis_mask >> i) & 1) {
+      for (int k = 0; k < ell_dims; ++k) add_full(0);
+      continue;
+    }
+    if ((new_axis_mask >> i) & 1) {
+      out->final_dims.push_back(1);
+      continue;
+    }
+    if (dim >= rank) return errors::InvalidArgument("slice spec OOB");
+    int64_t d = in.dim_size(dim);
+    int64_t stride = st[i] == 0 ? 1 : st[i];
+    int64_t bi = b[i], ei = e[i];
+    if ((shrink_mask >> i) & 1) {
+      if (bi < 0) bi += d;
+      if (bi < 0 || bi >= d)
+        return errors::InvalidArgument("shrink index out of range");
+      out->begin.push_back(bi);
+      out->end.push_back(bi + 1);
+      out->stride.push_back(1);
+      out->proc_dims.push_back(1);
+      ++dim;
+      continue;  // no final dim
+    }
+    if ((begin_mask >> i) & 1) bi = stride > 0 ? 0 : d - 1;
+    else if (bi < 0) bi += d;
+    if ((end_mask >> i) & 1) ei = stride > 0 ? d : -d - 1;
+    else if (ei < 0) ei += d;
+    // clamp
+    if (stride > 0) {
+      bi = std::min(std::max<int64_t>(bi, 0), d);
+      ei = std::min(std::max<int64_t>(ei, 0), d);
+    } else {
+      bi = std::min(std::max<int64_t>(bi, -1), d - 1);
+      ei = std::min(std::max<int64_t>(ei, -d - 1), d - 1);
+      if ((end_mask >> i) & 1) ei = -1;
+    }
+    int64_t ext = stride > 0 ? (ei - bi + stride - 1) / stride
+                             : (bi - ei - stride - 1) / (-stride);
+    if (ext < 0) ext = 0;
+    out->begin.push_back(bi);
+    out->end.push_back(ei);
+    out->stride.push_back(stride);
+    out->proc_dims.push_back(ext);
+    out->final_dims.push_back(ext);
+    ++dim;
+  }
+  // remaining input dims: full range
+  while (dim < rank) add_full(0);
+  out->total = 1;
+  for (int64_t x : out->proc_dims) out->total *= x;
+  return Status::OK();
+}
+
+}  // namespace strided
+
+class StridedSliceOp : public OpKernel {
+ public:
+  explicit StridedSliceOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("begin_mask", &bm_);
+    c->GetAttr("end_mask", &em_);
+    c->GetAttr("ellipsis_mask", &elm_);
+    c->GetAttr("new_axis_mask", &nam_);
+    c->GetAttr("shrink_axis_mask", &sam_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    strided::Spec sp;
+    OP_REQUIRES_OK(ctx, strided::Build(x.shape(), IntVector(ctx->input(1)),
+                                       IntVector(ctx->input(2)),
+                                       IntVector(ctx->input(3)), bm_, em_,
+                                       elm_, nam_, sam_, &sp));
+    Tensor* out = ctx->allocate_output(0, TensorShape(sp.final_dims));
+    int rank = x.dims();
+    std::vector<int64_t> in_strides(rank, 1);
+    for (int i = rank - 2; i >= 0; --i)
+      in_strides[i] = in_strides[i + 1] * x.dim_size(i + 1);
+    size_t es = DataTypeSize(x.dtype());
+    const char* src = (const char*)x.raw_data();
+    char* dst = (char*)out->raw_data();
+    std::vector<int64_t> idx(rank, 0);
+    for (int64_t i = 0; i < sp.total; ++i) {
+      int64_t off = 0;
+      for (int d = 0; d < rank; ++d)
+        off += (sp.begin[d] + idx[d] * sp.stride[d]) * in_strides[d];
+      std::memcpy(dst + i * es, src + off * es, es);
+      for (int d = rank - 1; d >= 0; --d) {
+        if (++idx[d] < sp.proc_dims[d]) break;
+        idx[d] = 0;
+      }
+    }
+  }
+
+ private:
+  int64_t bm_ = 0, em_ = 0, elm_ = 0, nam_ = 0, sam_ = 0;
+};
+REGISTER_KERNEL_BUILDER(Name("StridedSlice").Device(DEVICE_CPU),
+                        StridedSliceOp);
+
+template <typename T>
+class StridedSliceGradOp : public OpKernel {
+ public:
+  explicit StridedSliceGradOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("begin_mask", &bm_);
+    c->GetAttr("end_mask", &em_);
+    c->GetAttr("ellipsis_mask", &elm_);
+    c->GetAttr("new_axis_mask", &nam_);
+    c->GetAttr("shrink_axis_mask", &sam_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    TensorShape in_shape(IntVector(ctx->input(0)));
+    const Tensor& dy = ctx->input(4);
+    strided::Spec sp;
+    OP_REQUIRES_OK(ctx, strided::Build(in_shape, IntVector(ctx->input(1)),
+                                       IntVector(ctx->input(2)),
+                                       IntVector(ctx->input(3)), bm_, em_,
+                                       elm_, nam_, sam_, &sp));
+    OP_REQUIRES(ctx, dy.NumElements() == sp.total,
+                errors::InvalidArgument("dy size mismatch"));
+    Tensor* out = ctx->allocate_output(0, in_shape);
+    std::memset(out->raw_data(), 0, out->TotalBytes());
+    int rank = in_shape.dims();
+    std::vector<int64_t> in_strides(rank, 1);
+    for (int i = rank - 2; i >= 0; --i)
+      in_strides[i] = in_strides[i + 1] * in_shape.dim_size(i + 1);
+    const T* src = dy.flat<T>();
+    T* dst = out->flat<T>();
+    std::vector<int64_t> idx(rank, 0);
+    for (int64_t i = 0; i < sp.total; ++i) {
+      int64_t off = 0;
+      for (int d = 0; d < rank; ++d)
+        off += (sp.begin[d] + idx[d] * sp.stride[d]) * in_strides[d];
+      dst[off] += src[i];
+      for (int d = rank - 1; d >= 0; --d) {
+        if (++idx[d] < sp.proc_dims[d]) break;
+        idx[d] = 0;
+      }
+    }
+  }
+
+ private:
+  int64_t bm_ = 0, em_ = 0, elm_ = 0, nam_ = 0, sam_ = 0;
+};
+REGISTER_CPU_KERNEL_TYPES("StridedSliceGrad", StridedSliceGradOp)
+REGISTER_KERNEL_BUILDER(Name("StridedSliceGrad").Device(DEVICE_CPU)
+                            .TypeConstraint<bfloat16>("T"),
+                        StridedSliceGradOp<bfloat16>);
+
 }  // namespace
 }  // namespace stf

@@ -380,3 +380,31 @@ def _dynamic_partition_grad(op, *grads):
 for _op in ('Where', 'Unique', 'UniqueWithCounts', 'ListDiff', 'Bitcast',
             'ConcatOffset'):
     ops.NoGradient(_op)
+
+
+@RegisterGradient('StridedSlice')
+def _strided_slice_grad(op, grad):
+    from simple_tensorflow_amd.python.framework.ops import apply_op
+    out = apply_op('StridedSliceGrad', array_ops.shape(op.inputs[0]),
+                   op.inputs[1], op.inputs[2], op.inputs[3], grad,
+                   begin_mask=op.get_attr('begin_mask'),
+                   end_mask=op.get_attr('end_mask'),
+                   ellipsis_mask=op.get_attr('ellipsis_mask'),
+                   new_axis_mask=op.get_attr('new_axis_mask'),
+                   shrink_axis_mask=op.get_attr('shrink_axis_mask'))
+    if op.inputs[0]._shape is not None:
+        out.set_shape(list(op.inputs[0]._shape))
+    return [out, None, None, None]
+
+
+@RegisterGradient('StridedSliceGrad')
+def _strided_slice_grad_grad(op, grad):
+    from simple_tensorflow_amd.python.framework.ops import apply_op
+    out = apply_op('StridedSlice', grad, op.inputs[1], op.inputs[2],
+                   op.inputs[3],
+                   begin_mask=op.get_attr('begin_mask'),
+                   end_mask=op.get_attr('end_mask'),
+                   ellipsis_mask=op.get_attr('ellipsis_mask'),
+                   new_axis_mask=op.get_attr('new_axis_mask'),
+                   shrink_axis_mask=op.get_attr('shrink_axis_mask'))
+    return [None, None, None, None, out]

@@ -249,37 +249,89 @@ def slice(input_, begin, size, name=None):  # pylint: disable=redefined-builtin
     return t
 
 
+def strided_slice(input_, begin, end, strides=None, begin_mask=0,
+                  end_mask=0, ellipsis_mask=0, new_axis_mask=0,
+                  shrink_axis_mask=0, name=None):
+    """Full-semantics strided slice (reference array_ops.strided_slice)."""
+    x = convert_to_tensor(input_)
+    if strides is None:
+        strides = [1] * len(begin)
+    t = apply_op('StridedSlice', x,
+                 convert_to_tensor(begin, dtype=dtypes.int32),
+                 convert_to_tensor(end, dtype=dtypes.int32),
+                 convert_to_tensor(strides, dtype=dtypes.int32),
+                 begin_mask=begin_mask, end_mask=end_mask,
+                 ellipsis_mask=ellipsis_mask, new_axis_mask=new_axis_mask,
+                 shrink_axis_mask=shrink_axis_mask, name=name)
+    # Static shape via a zero-stride numpy view (no allocation).
+    bs = _static_value(t.op.inputs[1])
+    es = _static_value(t.op.inputs[2])
+    ss = _static_value(t.op.inputs[3])
+    if x._shape is not None and all(d is not None for d in x._shape) and \
+            bs is not None and es is not None and ss is not None:
+        key = []
+        for i in range(len(bs)):
+            if (new_axis_mask >> i) & 1:
+                key.append(None)
+            elif (ellipsis_mask >> i) & 1:
+                key.append(Ellipsis)
+            elif (shrink_axis_mask >> i) & 1:
+                key.append(int(bs[i]))
+            else:
+                b = None if (begin_mask >> i) & 1 else int(bs[i])
+                e = None if (end_mask >> i) & 1 else int(es[i])
+                key.append(_bi.slice(b, e, int(ss[i])))
+        fake = np.lib.stride_tricks.as_strided(
+            np.zeros(1, np.int8), tuple(x._shape), (0,) * len(x._shape))
+        t.set_shape(list(fake[tuple(key)].shape))
+    return t
+
+
 def _slice_helper(tensor, key):
-    """Basic tensor[a:b, c] support via Slice (no strides)."""
+    """tensor[...] -> StridedSlice with the reference's mask encoding
+    (python/ops/array_ops.py _SliceHelper)."""
     if not isinstance(key, tuple):
         key = (key,)
-    begin, sz, squeeze_axes = [], [], []
+    begin, end, strides = [], [], []
+    begin_mask = end_mask = ellipsis_mask = new_axis_mask = 0
+    shrink_axis_mask = 0
     for i, k in enumerate(key):
-        if isinstance(k, int):
-            begin.append(k)
-            sz.append(1)
-            squeeze_axes.append(i)
-        elif isinstance(k, type(Ellipsis)):
-            raise NotImplementedError('ellipsis slicing')
-        elif isinstance(k, _bi.slice):
-            if k.step not in (None, 1):
-                raise NotImplementedError('strided slicing')
-            b = k.start or 0
-            begin.append(b)
-            if k.stop is None:
-                sz.append(-1)
+        if isinstance(k, _bi.slice):
+            if k.start is None:
+                begin.append(0)
+                begin_mask |= 1 << i
             else:
-                sz.append(k.stop - b)
+                begin.append(k.start)
+            if k.stop is None:
+                end.append(0)
+                end_mask |= 1 << i
+            else:
+                end.append(k.stop)
+            strides.append(1 if k.step is None else k.step)
+        elif k is Ellipsis:
+            begin.append(0)
+            end.append(0)
+            strides.append(1)
+            ellipsis_mask |= 1 << i
+        elif k is None:
+            begin.append(0)
+            end.append(0)
+            strides.append(1)
+            new_axis_mask |= 1 << i
+        elif isinstance(k, int) or (isinstance(k, np.integer)):
+            begin.append(int(k))
+            end.append(int(k) + 1)
+            strides.append(1)
+            shrink_axis_mask |= 1 << i
+        elif isinstance(k, ops.Tensor):
+            raise NotImplementedError('tensor slice indices')
         else:
-            raise NotImplementedError('slice key %r' % (k,))
-    nd = len(tensor._shape) if tensor._shape is not None else len(key)
-    while len(begin) < nd:
-        begin.append(0)
-        sz.append(-1)
-    out = slice(tensor, begin, sz)
-    if squeeze_axes:
-        out = squeeze(out, axis=squeeze_axes)
-    return out
+            raise TypeError('invalid slice key %r' % (k,))
+    return strided_slice(tensor, begin, end, strides,
+                         begin_mask=begin_mask, end_mask=end_mask,
+                         ellipsis_mask=ellipsis_mask,
+                         new_axis_mask=new_axis_mask,
+                         shrink_axis_mask=shrink_axis_mask)
 
 
 def pad(x, paddings, name=None):

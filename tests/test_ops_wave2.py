@@ -221,3 +221,45 @@ def test_scatter_update_mul():
         s.run(mul)
         out = s.run(v.value())
     np.testing.assert_allclose(out, [[3, 3], [7, 7], [1, 1], [7, 7]])
+
+
+def test_strided_slice_full_semantics():
+    x = np.arange(120, dtype=np.float32).reshape(2, 3, 4, 5)
+    t = tf.constant(x)
+    cases = [
+        (lambda a: a[1], lambda a: a[1]),
+        (lambda a: a[:, 1:3], lambda a: a[:, 1:3]),
+        (lambda a: a[:, ::2, 1::2], lambda a: a[:, ::2, 1::2]),
+        (lambda a: a[..., 2], lambda a: a[..., 2]),
+        (lambda a: a[1, ..., ::-1], lambda a: a[1, ..., ::-1]),
+        (lambda a: a[None, 0, -1], lambda a: a[None, 0, -1]),
+        (lambda a: a[:, -2:, :, ::-2], lambda a: a[:, -2:, :, ::-2]),
+    ]
+    for i, (tf_fn, np_fn) in enumerate(cases):
+        out = _run(tf_fn(t))
+        want = np_fn(x)
+        assert out.shape == want.shape, (i, out.shape, want.shape)
+        np.testing.assert_allclose(out, want, err_msg=str(i))
+
+
+def test_strided_slice_grad():
+    x = np.arange(24, dtype=np.float32).reshape(4, 6)
+    t = tf.constant(x)
+    y = t[1:3, ::2]
+    w = tf.constant(np.array([[1.0, 2.0, 3.0], [4.0, 5.0, 6.0]], np.float32))
+    g = tf.gradients(tf.reduce_sum(y * w), [t])[0]
+    out = _run(g)
+    want = np.zeros((4, 6), np.float32)
+    want[1:3, ::2] = [[1, 2, 3], [4, 5, 6]]
+    np.testing.assert_allclose(out, want)
+
+
+def test_strided_slice_ellipsis_grad():
+    x = np.arange(24, dtype=np.float32).reshape(2, 3, 4)
+    t = tf.constant(x)
+    y = t[..., 1]
+    g = tf.gradients(tf.reduce_sum(y), [t])[0]
+    out = _run(g)
+    want = np.zeros((2, 3, 4), np.float32)
+    want[..., 1] = 1
+    np.testing.assert_allclose(out, want)

@@ -305,6 +305,27 @@ REGISTER_OP("DepthToSpace").Input("input: T").Output("output: T").Attr("T: type"
 REGISTER_OP("MirrorPad").Input("input: T").Input("paddings: Tpaddings").Output("output: T").Attr("T: type").Attr("Tpaddings: {int32, int64} = int32").Attr("mode: string");
 REGISTER_OP("ReverseSequence").Input("input: T").Input("seq_lengths: Tlen").Output("output: T").Attr("seq_dim: int").Attr("batch_dim: int = 0").Attr("T: " NUMTYPES).Attr("Tlen: {int32, int64} = int64");
 REGISTER_OP("Bitcast").Input("input: T").Output("output: type").Attr("T: " NUMTYPES).Attr("type: " NUMTYPES);
+// lookup tables (reference core/ops/data_flow_ops.cc lookup section)
+REGISTER_OP("HashTable").Output("table_handle: Ref(string)").Attr("container: string = ''").Attr("shared_name: string = ''").Attr("key_dtype: type").Attr("value_dtype: type").SetIsStateful();
+REGISTER_OP("MutableHashTable").Output("table_handle: Ref(string)").Attr("container: string = ''").Attr("shared_name: string = ''").Attr("key_dtype: type").Attr("value_dtype: type").SetIsStateful();
+REGISTER_OP("InitializeTable").Input("table_handle: Ref(string)").Input("keys: Tkey").Input("values: Tval").Attr("Tkey: type").Attr("Tval: type").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("LookupTableInsert").Input("table_handle: Ref(string)").Input("keys: Tin").Input("values: Tout").Attr("Tin: type").Attr("Tout: type").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("LookupTableImport").Input("table_handle: Ref(string)").Input("keys: Tin").Input("values: Tout").Attr("Tin: type").Attr("Tout: type").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("LookupTableFind").Input("table_handle: Ref(string)").Input("keys: Tin").Input("default_value: Tout").Output("values: Tout").Attr("Tin: type").Attr("Tout: type").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("LookupTableSize").Input("table_handle: Ref(string)").Output("size: int64").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("LookupTableExport").Input("table_handle: Ref(string)").Output("keys: Tkeys").Output("values: Tvalues").Attr("Tkeys: type").Attr("Tvalues: type").SetIsStateful().SetAllowsUninitializedInput();
+// stacks (reference stack_ops.cc)
+REGISTER_OP("Stack").Output("handle: Ref(string)").Attr("elem_type: type").Attr("stack_name: string = ''").SetIsStateful();
+REGISTER_OP("StackPush").Input("handle: Ref(string)").Input("elem: T").Output("output: T").Attr("T: type").Attr("swap_memory: bool = false").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("StackPop").Input("handle: Ref(string)").Output("elem: elem_type").Attr("elem_type: type").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("StackClose").Input("handle: Ref(string)").SetIsStateful().SetAllowsUninitializedInput();
+// barrier (reference barrier_ops.cc)
+REGISTER_OP("Barrier").Output("handle: Ref(string)").Attr("component_types: list(type) >= 1").Attr("shapes: list(shape) = []").Attr("capacity: int = -1").Attr("container: string = ''").Attr("shared_name: string = ''").SetIsStateful();
+REGISTER_OP("BarrierInsertMany").Input("handle: Ref(string)").Input("keys: string").Input("values: T").Attr("T: type").Attr("component_index: int").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("BarrierTakeMany").Input("handle: Ref(string)").Input("num_elements: int32").Output("indices: int64").Output("keys: string").Output("values: component_types").Attr("component_types: list(type) >= 1").Attr("allow_small_batch: bool = false").Attr("wait_for_incomplete: bool = false").Attr("timeout_ms: int = -1").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("BarrierClose").Input("handle: Ref(string)").Attr("cancel_pending_enqueues: bool = false").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("BarrierReadySize").Input("handle: Ref(string)").Output("size: int32").SetIsStateful().SetAllowsUninitializedInput();
+REGISTER_OP("BarrierIncompleteSize").Input("handle: Ref(string)").Output("size: int32").SetIsStateful().SetAllowsUninitializedInput();
 // variable scatter updates
 #define SCATTER_VAR_OP(NAME) REGISTER_OP(NAME).Input("ref: Ref(T)").Input("indices: Tindices").Input("updates: T").Output("output_ref: Ref(T)").Attr("T: " NUMTYPES).Attr("Tindices: {int32, int64}").Attr("use_locking: bool = true")
 SCATTER_VAR_OP("ScatterUpdate");

@@ -36,6 +36,7 @@ class GraphKeys(object):
     SAVERS = 'savers'
     INIT_OP = 'init_op'
     LOSSES = 'losses'
+    TABLE_INITIALIZERS = 'table_initializer'
     VARIABLES = 'variables'  # legacy alias
 
 

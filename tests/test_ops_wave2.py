@@ -263,3 +263,70 @@ def test_strided_slice_ellipsis_grad():
     want = np.zeros((2, 3, 4), np.float32)
     want[..., 1] = 1
     np.testing.assert_allclose(out, want)
+
+
+def test_hash_table():
+    from simple_tensorflow_amd.python.ops import lookup_ops
+    table = lookup_ops.HashTable(
+        lookup_ops.KeyValueTensorInitializer(
+            tf.constant(np.array([1, 2, 3], np.int64)),
+            tf.constant(np.array([10.0, 20.0, 30.0], np.float32))),
+        default_value=-1.0)
+    out = table.lookup(tf.constant(np.array([2, 5, 1], np.int64)))
+    size = table.size()
+    with tf.Session() as s:
+        s.run(table.init)
+        r, n = s.run([out, size])
+    np.testing.assert_allclose(r, [20.0, -1.0, 10.0])
+    assert n == 3
+
+
+def test_mutable_hash_table():
+    from simple_tensorflow_amd.python.ops import lookup_ops
+    table = lookup_ops.MutableHashTable(tf.int64, tf.float32, -1.0)
+    ins = table.insert(tf.constant(np.array([7, 8], np.int64)),
+                       tf.constant(np.array([70.0, 80.0], np.float32)))
+    out = table.lookup(tf.constant(np.array([8, 9], np.int64)))
+    with tf.Session() as s:
+        s.run(ins)
+        r = s.run(out)
+        k, v = s.run(table.export())
+    np.testing.assert_allclose(r, [80.0, -1.0])
+    assert set(k.tolist()) == {7, 8}
+
+
+def test_stack_ops():
+    from simple_tensorflow_amd.python.framework.ops import apply_op
+    h = apply_op('Stack', elem_type=tf.float32)
+    p1 = apply_op('StackPush', h, tf.constant(1.0))
+    p2 = apply_op('StackPush', h, tf.constant(2.0))
+    p2.op._add_control_input(p1.op)
+    pop = apply_op('StackPop', h, elem_type=tf.float32)
+    pop.op._add_control_input(p2.op)
+    pop2 = apply_op('StackPop', h, elem_type=tf.float32)
+    pop2.op._add_control_input(pop.op)
+    with tf.Session() as s:
+        a, b = s.run([pop, pop2])
+    assert (a, b) == (2.0, 1.0)
+
+
+def test_barrier_ops():
+    from simple_tensorflow_amd.python.framework.ops import apply_op
+    h = apply_op('Barrier', component_types=[tf.float32, tf.float32])
+    i0 = apply_op('BarrierInsertMany', h,
+                  tf.constant([b'k1', b'k2']),
+                  tf.constant(np.array([1.0, 2.0], np.float32)),
+                  component_index=0)
+    i1 = apply_op('BarrierInsertMany', h,
+                  tf.constant([b'k1', b'k2']),
+                  tf.constant(np.array([10.0, 20.0], np.float32)),
+                  component_index=1)
+    take = apply_op('BarrierTakeMany', h, tf.constant(2),
+                    component_types=[tf.float32, tf.float32])
+    take[0].op._add_control_input(i0.op if hasattr(i0, 'op') else i0)
+    take[0].op._add_control_input(i1.op if hasattr(i1, 'op') else i1)
+    with tf.Session() as s:
+        idx, keys, v0, v1 = s.run(list(take))
+    assert sorted(list(keys)) == [b'k1', b'k2']
+    np.testing.assert_allclose(sorted(v0.tolist() if hasattr(v0, 'tolist') else list(v0)), [1.0, 2.0])
+    np.testing.assert_allclose(sorted(v1.tolist() if hasattr(v1, 'tolist') else list(v1)), [10.0, 20.0])

@@ -227,6 +227,20 @@ mirror_pad = _array_ops.mirror_pad
 reverse_sequence = _array_ops.reverse_sequence
 bitcast = _array_ops.bitcast
 cumprod = _math_ops.cumprod
+tan = _math_ops.tan
+asin = _math_ops.asin
+acos = _math_ops.acos
+atan = _math_ops.atan
+erf = _math_ops.erf
+erfc = _math_ops.erfc
+expm1 = _math_ops.expm1
+lgamma = _math_ops.lgamma
+digamma = _math_ops.digamma
+rint = _math_ops.rint
+mod = _math_ops.mod
+approximate_equal = _math_ops.approximate_equal
+as_string = lambda x, name=None: _ops.apply_op('AsString', _ops.convert_to_tensor(x), name=name)  # noqa: E731
+decode_raw = lambda bytes_, out_type, name=None: _ops.apply_op('DecodeRaw', _ops.convert_to_tensor(bytes_), out_type=_dtypes.as_dtype(out_type), name=name)  # noqa: E731
 segment_sum = _math_ops.segment_sum
 segment_mean = _math_ops.segment_mean
 segment_max = _math_ops.segment_max

@@ -305,6 +305,27 @@ REGISTER_OP("DepthToSpace").Input("input: T").Output("output: T").Attr("T: type"
 REGISTER_OP("MirrorPad").Input("input: T").Input("paddings: Tpaddings").Output("output: T").Attr("T: type").Attr("Tpaddings: {int32, int64} = int32").Attr("mode: string");
 REGISTER_OP("ReverseSequence").Input("input: T").Input("seq_lengths: Tlen").Output("output: T").Attr("seq_dim: int").Attr("batch_dim: int = 0").Attr("T: " NUMTYPES).Attr("Tlen: {int32, int64} = int64");
 REGISTER_OP("Bitcast").Input("input: T").Output("output: type").Attr("T: " NUMTYPES).Attr("type: " NUMTYPES);
+// math breadth (trig / special functions / mod — reference math_ops.cc)
+#define UOP(NAME) REGISTER_OP(NAME).Input("x: T").Output("y: T").Attr("T: " FLOATTYPES)
+UOP("Tan");
+UOP("Asin");
+UOP("Acos");
+UOP("Atan");
+UOP("Erf");
+UOP("Erfc");
+UOP("Expm1");
+UOP("Lgamma");
+UOP("Digamma");
+UOP("Rint");
+UOP("Softsign");
+UOP("Inv");
+#undef UOP
+REGISTER_OP("SoftsignGrad").Input("gradients: T").Input("features: T").Output("backprops: T").Attr("T: " FLOATTYPES);
+REGISTER_OP("InvGrad").Input("y: T").Input("dy: T").Output("z: T").Attr("T: " FLOATTYPES);
+REGISTER_OP("Mod").Input("x: T").Input("y: T").Output("z: T").Attr("T: {float, double, int32, int64}");
+REGISTER_OP("ApproximateEqual").Input("x: T").Input("y: T").Output("z: bool").Attr("T: " NUMTYPES).Attr("tolerance: float = 1e-05");
+REGISTER_OP("AsString").Input("input: T").Output("output: string").Attr("T: {float, double, int32, int64, bool}").Attr("precision: int = -1").Attr("scientific: bool = false").Attr("shortest: bool = false").Attr("width: int = -1").Attr("fill: string = ''");
+REGISTER_OP("DecodeRaw").Input("bytes: string").Output("output: out_type").Attr("out_type: {float, double, int32, uint8, int16, int8, int64}").Attr("little_endian: bool = true");
 // lookup tables (reference core/ops/data_flow_ops.cc lookup section)
 REGISTER_OP("HashTable").Output("table_handle: Ref(string)").Attr("container: string = ''").Attr("shared_name: string = ''").Attr("key_dtype: type").Attr("value_dtype: type").SetIsStateful();
 REGISTER_OP("MutableHashTable").Output("table_handle: Ref(string)").Attr("container: string = ''").Attr("shared_name: string = ''").Attr("key_dtype: type").Attr("value_dtype: type").SetIsStateful();

@@ -1035,5 +1035,199 @@ REGISTER_KERNEL_BUILDER(Name("StridedSliceGrad").Device(DEVICE_CPU)
                             .TypeConstraint<bfloat16>("T"),
                         StridedSliceGradOp<bfloat16>);
 
+// ------------------------ math breadth (trig/special) -----------------------
+template <typename T, typename F>
+class UnaryXOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    Tensor* out = ctx->allocate_output(0, x.shape());
+    const T* xp = x.flat<T>();
+    T* op = out->flat<T>();
+    F f;
+    for (int64_t i = 0; i < x.NumElements(); ++i) op[i] = f((double)xp[i]);
+  }
+};
+#define XFN(NAME, EXPR)                                            \
+  struct NAME {                                                    \
+    double operator()(double a) const { return EXPR; }             \
+  };
+XFN(FTan, std::tan(a))
+XFN(FAsin, std::asin(a))
+XFN(FAcos, std::acos(a))
+XFN(FAtan, std::atan(a))
+XFN(FErf, std::erf(a))
+XFN(FErfc, std::erfc(a))
+XFN(FExpm1, std::expm1(a))
+XFN(FLgamma, std::lgamma(a))
+XFN(FRint2, std::rint(a))
+XFN(FSoftsign, a / (1.0 + std::abs(a)))
+XFN(FInvX, 1.0 / a)
+#undef XFN
+struct FDigamma {
+  double operator()(double x) const {
+    // standard asymptotic series with upward recurrence
+    double r = 0.0;
+    while (x < 6.0) { r -= 1.0 / x; x += 1.0; }
+    double f = 1.0 / (x * x);
+    return r + std::log(x) - 0.5 / x -
+           f * (1.0 / 12 - f * (1.0 / 120 - f * (1.0 / 252 - f / 240)));
+  }
+};
+#define REG_XUNARY(OP, F)                                                     \
+  REGISTER_KERNEL_BUILDER(                                                    \
+      Name(OP).Device(DEVICE_CPU).TypeConstraint<float>("T"),                 \
+      (UnaryXOp<float, F>));                                                  \
+  REGISTER_KERNEL_BUILDER(                                                    \
+      Name(OP).Device(DEVICE_CPU).TypeConstraint<double>("T"),                \
+      (UnaryXOp<double, F>));
+REG_XUNARY("Tan", FTan)
+REG_XUNARY("Asin", FAsin)
+REG_XUNARY("Acos", FAcos)
+REG_XUNARY("Atan", FAtan)
+REG_XUNARY("Erf", FErf)
+REG_XUNARY("Erfc", FErfc)
+REG_XUNARY("Expm1", FExpm1)
+REG_XUNARY("Lgamma", FLgamma)
+REG_XUNARY("Digamma", FDigamma)
+REG_XUNARY("Rint", FRint2)
+REG_XUNARY("Softsign", FSoftsign)
+REG_XUNARY("Inv", FInvX)
+#undef REG_XUNARY
+
+template <typename T>
+class SoftsignGradOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& g = ctx->input(0);
+    const Tensor& x = ctx->input(1);
+    Tensor* out = ctx->allocate_output(0, x.shape());
+    const T* gp = g.flat<T>();
+    const T* xp = x.flat<T>();
+    T* op = out->flat<T>();
+    for (int64_t i = 0; i < x.NumElements(); ++i) {
+      T d = T(1) + (xp[i] < T(0) ? -xp[i] : xp[i]);
+      op[i] = gp[i] / (d * d);
+    }
+  }
+};
+REGISTER_CPU_KERNEL_FLOATS("SoftsignGrad", SoftsignGradOp)
+
+template <typename T>
+class InvGradOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& y = ctx->input(0);
+    const Tensor& dy = ctx->input(1);
+    Tensor* out = ctx->allocate_output(0, y.shape());
+    const T* yp = y.flat<T>();
+    const T* dp = dy.flat<T>();
+    T* op = out->flat<T>();
+    for (int64_t i = 0; i < y.NumElements(); ++i)
+      op[i] = -dp[i] * yp[i] * yp[i];
+  }
+};
+REGISTER_CPU_KERNEL_FLOATS("InvGrad", InvGradOp)
+
+// truncated mod (C semantics), matching the reference Mod op
+template <typename T>
+class ModOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    const Tensor& y = ctx->input(1);
+    BCast b(x.shape(), y.shape());
+    OP_REQUIRES(ctx, b.valid, errors::InvalidArgument("Mod: bad broadcast"));
+    Tensor* out = ctx->allocate_output(0, b.out_shape());
+    const T* xp = x.flat<T>();
+    const T* yp = y.flat<T>();
+    T* op = out->flat<T>();
+    for (int64_t i = 0; i < b.num_elements; ++i) {
+      int64_t xi, yi;
+      b.Map(i, &xi, &yi);
+      if constexpr (std::is_integral<T>::value)
+        op[i] = xp[xi] % yp[yi];
+      else
+        op[i] = std::fmod((double)xp[xi], (double)yp[yi]);
+    }
+  }
+};
+REGISTER_CPU_KERNEL_TYPES("Mod", ModOp)
+
+template <typename T>
+class ApproximateEqualOp : public OpKernel {
+ public:
+  explicit ApproximateEqualOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("tolerance", &tol_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    const Tensor& y = ctx->input(1);
+    Tensor* out = ctx->allocate_output(0, x.shape());
+    const T* xp = x.flat<T>();
+    const T* yp = y.flat<T>();
+    bool* op = out->flat<bool>();
+    for (int64_t i = 0; i < x.NumElements(); ++i)
+      op[i] = std::abs((double)(xp[i] - yp[i])) < (double)tol_;
+  }
+
+ private:
+  float tol_ = 1e-5f;
+};
+REGISTER_CPU_KERNEL_TYPES("ApproximateEqual", ApproximateEqualOp)
+
+class AsStringOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    Tensor* out = ctx->allocate_output(0, x.shape());
+    for (int64_t i = 0; i < x.NumElements(); ++i) {
+      std::string v;
+      switch (x.dtype()) {
+        case DT_FLOAT: v = std::to_string(x.flat<float>()[i]); break;
+        case DT_DOUBLE: v = std::to_string(x.flat<double>()[i]); break;
+        case DT_INT32: v = std::to_string(x.flat<int32_t>()[i]); break;
+        case DT_INT64: v = std::to_string(x.flat<int64_t>()[i]); break;
+        case DT_BOOL: v = x.flat<bool>()[i] ? "true" : "false"; break;
+        default:
+          ctx->SetStatus(errors::InvalidArgument("AsString dtype"));
+          return;
+      }
+      out->flat<std::string>()[i] = v;
+    }
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("AsString").Device(DEVICE_CPU), AsStringOp);
+
+class DecodeRawOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& in = ctx->input(0);
+    DataType out_t = output_type(0);
+    size_t es = DataTypeSize(out_t);
+    OP_REQUIRES(ctx, in.NumElements() > 0,
+                errors::InvalidArgument("DecodeRaw: empty input"));
+    const std::string& first = in.flat<std::string>()[0];
+    int64_t elems = (int64_t)(first.size() / es);
+    TensorShape shape = in.shape();
+    shape.AddDim(elems);
+    Tensor* out = ctx->allocate_output(0, shape);
+    for (int64_t i = 0; i < in.NumElements(); ++i) {
+      const std::string& s = in.flat<std::string>()[i];
+      OP_REQUIRES(ctx, (int64_t)(s.size() / es) == elems,
+                  errors::InvalidArgument("DecodeRaw: ragged strings"));
+      std::memcpy((char*)out->raw_data() + i * elems * es, s.data(),
+                  elems * es);
+    }
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("DecodeRaw").Device(DEVICE_CPU), DecodeRawOp);
+
 }  // namespace
 }  // namespace stf

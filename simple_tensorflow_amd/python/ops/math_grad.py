@@ -392,3 +392,65 @@ def _segment_max_grad(op, grad):
 @RegisterGradient('SegmentMin')
 def _segment_min_grad(op, grad):
     return _segment_minmax_grad(op, grad)
+
+
+@RegisterGradient('Tan')
+def _tan_grad(op, grad):
+    c = math_ops.cos(op.inputs[0])
+    return grad / (c * c)
+
+
+@RegisterGradient('Asin')
+def _asin_grad(op, grad):
+    x = op.inputs[0]
+    return grad * math_ops.rsqrt(1.0 - x * x)
+
+
+@RegisterGradient('Acos')
+def _acos_grad(op, grad):
+    x = op.inputs[0]
+    return -grad * math_ops.rsqrt(1.0 - x * x)
+
+
+@RegisterGradient('Atan')
+def _atan_grad(op, grad):
+    x = op.inputs[0]
+    return grad / (1.0 + x * x)
+
+
+@RegisterGradient('Erf')
+def _erf_grad(op, grad):
+    x = op.inputs[0]
+    return grad * (2.0 / 1.7724538509055159) * math_ops.exp(-x * x)
+
+
+@RegisterGradient('Erfc')
+def _erfc_grad(op, grad):
+    x = op.inputs[0]
+    return -grad * (2.0 / 1.7724538509055159) * math_ops.exp(-x * x)
+
+
+@RegisterGradient('Expm1')
+def _expm1_grad(op, grad):
+    return grad * math_ops.exp(op.inputs[0])
+
+
+@RegisterGradient('Lgamma')
+def _lgamma_grad(op, grad):
+    return grad * math_ops.digamma(op.inputs[0])
+
+
+@RegisterGradient('Softsign')
+def _softsign_grad(op, grad):
+    return apply_op('SoftsignGrad', grad, op.inputs[0])
+
+
+@RegisterGradient('Inv')
+def _inv_grad(op, grad):
+    return apply_op('InvGrad', op.outputs[0], grad)
+
+
+ops.NoGradient('Rint')
+ops.NoGradient('ApproximateEqual')
+ops.NoGradient('AsString')
+ops.NoGradient('DecodeRaw')

@@ -363,3 +363,62 @@ def trace(x, name=None):
     diag_idx = ops.constant([i * (cols + 1) for i in _bi.range(n)],
                             dtype=dtypes.int32)
     return reduce_sum(array_ops.gather(flat, diag_idx))
+
+
+# ---------------------------------------------------------------------------
+# round-2 math breadth
+# ---------------------------------------------------------------------------
+def tan(x, name=None):
+    return _unary('Tan', x, name)
+
+
+def asin(x, name=None):
+    return _unary('Asin', x, name)
+
+
+def acos(x, name=None):
+    return _unary('Acos', x, name)
+
+
+def atan(x, name=None):
+    return _unary('Atan', x, name)
+
+
+def erf(x, name=None):
+    return _unary('Erf', x, name)
+
+
+def erfc(x, name=None):
+    return _unary('Erfc', x, name)
+
+
+def expm1(x, name=None):
+    return _unary('Expm1', x, name)
+
+
+def lgamma(x, name=None):
+    return _unary('Lgamma', x, name)
+
+
+def digamma(x, name=None):
+    return _unary('Digamma', x, name)
+
+
+def rint(x, name=None):
+    return _unary('Rint', x, name)
+
+
+def softsign(x, name=None):
+    return _unary('Softsign', x, name)
+
+
+def mod(x, y, name=None):
+    x = convert_to_tensor(x)
+    return apply_op('Mod', x, convert_to_tensor(y, dtype=x.dtype), name=name)
+
+
+def approximate_equal(x, y, tolerance=1e-5, name=None):
+    x = convert_to_tensor(x)
+    return apply_op('ApproximateEqual', x,
+                    convert_to_tensor(y, dtype=x.dtype),
+                    tolerance=tolerance, name=name)

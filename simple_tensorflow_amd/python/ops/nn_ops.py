@@ -233,3 +233,8 @@ def depthwise_conv2d(input, filter, strides, padding, rate=None, name=None):  # 
 
 
 depthwise_conv2d_native = depthwise_conv2d
+
+
+def softsign(features, name=None):
+    from simple_tensorflow_amd.python.framework.ops import convert_to_tensor
+    return apply_op('Softsign', convert_to_tensor(features), name=name)

@@ -330,3 +330,48 @@ def test_barrier_ops():
     assert sorted(list(keys)) == [b'k1', b'k2']
     np.testing.assert_allclose(sorted(v0.tolist() if hasattr(v0, 'tolist') else list(v0)), [1.0, 2.0])
     np.testing.assert_allclose(sorted(v1.tolist() if hasattr(v1, 'tolist') else list(v1)), [10.0, 20.0])
+
+
+def test_math_breadth_vs_numpy():
+    from scipy import special
+    x = np.linspace(-0.9, 0.9, 7).astype(np.float32)
+    t = tf.constant(x)
+    got = _run([tf.tan(t), tf.asin(t), tf.acos(t), tf.atan(t), tf.erf(t),
+                tf.erfc(t), tf.expm1(t)])
+    for g, w in zip(got, [np.tan(x), np.arcsin(x), np.arccos(x),
+                          np.arctan(x), special.erf(x), special.erfc(x),
+                          np.expm1(x)]):
+        np.testing.assert_allclose(g, w, rtol=1e-5, atol=1e-6)
+    xp = np.array([0.5, 1.0, 2.5, 7.0], np.float32)
+    got2 = _run([tf.lgamma(tf.constant(xp)), tf.digamma(tf.constant(xp))])
+    np.testing.assert_allclose(got2[0], special.gammaln(xp), rtol=1e-5,
+                               atol=1e-5)
+    np.testing.assert_allclose(got2[1], special.digamma(xp), rtol=1e-4,
+                               atol=1e-4)
+    m = _run(tf.mod(tf.constant([7.0, -7.0]), tf.constant([3.0, 3.0])))
+    np.testing.assert_allclose(m, np.fmod([7.0, -7.0], 3.0))
+    ae = _run(tf.approximate_equal(tf.constant([1.0, 1.1]),
+                                   tf.constant([1.0000001, 1.0])))
+    np.testing.assert_array_equal(ae, [True, False])
+
+
+def test_math_breadth_grads():
+    x = np.array([0.3, -0.2, 0.7], np.float32)
+    t = tf.constant(x)
+    for fn, dfn in [(tf.tan, lambda v: 1 / np.cos(v) ** 2),
+                    (tf.asin, lambda v: 1 / np.sqrt(1 - v * v)),
+                    (tf.atan, lambda v: 1 / (1 + v * v)),
+                    (tf.erf, lambda v: 2 / np.sqrt(np.pi) * np.exp(-v * v)),
+                    (tf.expm1, np.exp)]:
+        g = tf.gradients(tf.reduce_sum(fn(t)), [t])[0]
+        np.testing.assert_allclose(_run(g), dfn(x), rtol=1e-4, atol=1e-5)
+    g = tf.gradients(tf.reduce_sum(tf.nn.softsign(t)), [t])[0]
+    np.testing.assert_allclose(_run(g), 1 / (1 + np.abs(x)) ** 2, rtol=1e-5)
+
+
+def test_as_string_decode_raw():
+    s = _run(tf.as_string(tf.constant(np.array([1, 2], np.int32))))
+    assert list(s) == [b'1', b'2']
+    raw = np.array([1.5, -2.0], np.float32).tobytes()
+    out = _run(tf.decode_raw(tf.constant([raw]), tf.float32))
+    np.testing.assert_allclose(out, [[1.5, -2.0]])

@@ -26,6 +26,8 @@ hipError_t stf_gemm_bf16_splitk(const void*, const void*, void*, int64_t,
                                 int64_t, int64_t, int64_t, int64_t, int, int,
                                 int, hipStream_t);
 int stf_gemm_bf16_8ph_ok(int64_t, int64_t, int64_t);
+hipError_t stf_lrn_fwd(const void*, void*, int64_t, int, int, float, float,
+                       float, hipStream_t);
 hipError_t stf_conv2d_fwd_8ph(const void*, const void*, void*, const void*,
                               const void*, int, int, int, int, int, int, int,
                               int, int, int, int, int, int64_t, int64_t, int,
@@ -510,6 +512,113 @@ class GpuMatMulOp : public OpKernel {
 };
 REGISTER_KERNEL_BUILDER(Name("MatMul").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuMatMulOp);
 REGISTER_KERNEL_BUILDER(Name("MatMul").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuMatMulOp);
+
+// BatchMatMul on GPU: per-batch MFMA GEMM launches with offset pointers
+// (reference batch_matmul_op_impl.h:350 ThenBlasGemmBatchedWithScratch).
+// The adj flags map onto the GEMM's contraction-major staging the same way
+// MatMul's transpose flags do, so no transpose kernels are launched.
+class GpuBatchMatMulOp : public OpKernel {
+ public:
+  explicit GpuBatchMatMulOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("adj_x", &ta_);
+    c->GetAttr("adj_y", &tb_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& a = ctx->input(0);
+    const Tensor& b = ctx->input(1);
+    hipStream_t s = GPU_STREAM(ctx);
+    int ra = a.dims(), rb = b.dims();
+    OP_REQUIRES(ctx, ra >= 2 && rb == ra,
+                errors::InvalidArgument("BatchMatMul rank mismatch"));
+    int64_t m = ta_ ? a.dim_size(ra - 1) : a.dim_size(ra - 2);
+    int64_t k = ta_ ? a.dim_size(ra - 2) : a.dim_size(ra - 1);
+    int64_t n = tb_ ? b.dim_size(rb - 2) : b.dim_size(rb - 1);
+    int64_t kb = tb_ ? b.dim_size(rb - 1) : b.dim_size(rb - 2);
+    OP_REQUIRES(ctx, k == kb,
+                errors::InvalidArgument("BatchMatMul inner dim mismatch"));
+    int64_t batch = 1;
+    TensorShape out_shape;
+    for (int i = 0; i < ra - 2; ++i) {
+      batch *= a.dim_size(i);
+      out_shape.AddDim(a.dim_size(i));
+    }
+    out_shape.AddDim(m);
+    out_shape.AddDim(n);
+    Tensor* y = ctx->allocate_output(0, out_shape);
+    int64_t sa = m * k, sb = k * n, sc = m * n;
+    if (a.dtype() == DT_BFLOAT16) {
+      // adj_x: A [K,M] contraction-major -> a_km; plain B [K,N] -> b_km.
+      int a_km = ta_ ? 1 : 0;
+      int b_km = tb_ ? 0 : 1;
+      int64_t lda = a.dim_size(ra - 1);
+      int64_t ldb = b.dim_size(rb - 1);
+      const uint16_t* ap = (const uint16_t*)a.raw_data();
+      const uint16_t* bp = (const uint16_t*)b.raw_data();
+      uint16_t* yp = (uint16_t*)y->raw_data();
+      for (int64_t i = 0; i < batch; ++i) {
+        OP_HIP_OK(ctx, stf_gemm_bf16(ap + i * sa, bp + i * sb, yp + i * sc,
+                                     nullptr, m, n, k, lda, ldb, 0.f, a_km,
+                                     b_km, 1, 0, s));
+      }
+    } else {
+      // f32 path: pre-transpose per batch when adjoint flags are set.
+      Tensor a_eff = a, b_eff = b;
+      const float* ap = (const float*)a.raw_data();
+      const float* bp = (const float*)b.raw_data();
+      if (ta_) {
+        a_eff = ctx->allocate_temp(DT_FLOAT, TensorShape({batch, m, k}));
+        for (int64_t i = 0; i < batch; ++i)
+          OP_HIP_OK(ctx, stf_transpose2d(4, ap + i * sa,
+                                         (float*)a_eff.raw_data() + i * sa,
+                                         k, m, s));
+        ap = (const float*)a_eff.raw_data();
+      }
+      // GEMM f32 wants B as [N, K] (NT): transpose unless adj_y.
+      if (!tb_) {
+        b_eff = ctx->allocate_temp(DT_FLOAT, TensorShape({batch, n, k}));
+        for (int64_t i = 0; i < batch; ++i)
+          OP_HIP_OK(ctx, stf_transpose2d(4, bp + i * sb,
+                                         (float*)b_eff.raw_data() + i * sb,
+                                         k, n, s));
+        bp = (const float*)b_eff.raw_data();
+      }
+      float* yp = (float*)y->raw_data();
+      for (int64_t i = 0; i < batch; ++i)
+        OP_HIP_OK(ctx, stf_gemm_f32_nt(ap + i * sa, bp + i * sb, yp + i * sc,
+                                       m, n, k, s));
+    }
+  }
+
+ private:
+  bool ta_ = false, tb_ = false;
+};
+REGISTER_KERNEL_BUILDER(Name("BatchMatMul").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuBatchMatMulOp);
+REGISTER_KERNEL_BUILDER(Name("BatchMatMul").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuBatchMatMulOp);
+
+// LRN forward on GPU (reference used cuDNN, lrn_op.cc:211). The composite
+// python gradient runs on the GPU's elementwise kernels.
+class GpuLRNOp : public OpKernel {
+ public:
+  explicit GpuLRNOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("depth_radius", &radius_);
+    c->GetAttr("bias", &bias_);
+    c->GetAttr("alpha", &alpha_);
+    c->GetAttr("beta", &beta_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& x = ctx->input(0);
+    Tensor* y = ctx->allocate_output(0, x.shape());
+    int64_t c = x.dim_size(x.dims() - 1);
+    OP_HIP_OK(ctx, stf_lrn_fwd(x.raw_data(), y->raw_data(),
+                               x.NumElements() / c, (int)c, (int)radius_,
+                               bias_, alpha_, beta_, GPU_STREAM(ctx)));
+  }
+
+ private:
+  int64_t radius_ = 5;
+  float bias_ = 1.f, alpha_ = 1.f, beta_ = 0.5f;
+};
+REGISTER_KERNEL_BUILDER(Name("LRN").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuLRNOp);
 
 // ---------------------------------------------------------------------------
 // Conv2D family (im2col + MFMA GEMM)

@@ -205,3 +205,42 @@ def test_implicit_gemm_conv_vs_torch():
         assert np.abs(got_y - want_y).max() / scale < 0.05, (cin, cout, stride)
         assert np.abs(got_gx - want_gx).max() / (np.abs(want_gx).max() + 1e-6) < 0.05
         assert np.abs(got_gw - want_gw).max() / (np.abs(want_gw).max() + 1e-6) < 0.05
+
+
+def test_batch_matmul_bf16_vs_torch():
+    rng = np.random.RandomState(13)
+    a = (rng.rand(6, 48, 32) * 2 - 1).astype(np.float32)
+    b = (rng.rand(6, 32, 40) * 2 - 1).astype(np.float32)
+    got = _run(tf.batch_matmul(tf.constant(a, dtype=tf.bfloat16),
+                               tf.constant(b, dtype=tf.bfloat16)))
+    want = (torch.from_numpy(a).bfloat16().float() @
+            torch.from_numpy(b).bfloat16().float()).numpy()
+    rel = np.abs(got - want) / (np.abs(want) + 1e-2)
+    assert np.percentile(rel, 99) < 0.05
+    # adjoint flags
+    got2 = _run(tf.batch_matmul(tf.constant(a, dtype=tf.bfloat16),
+                                tf.constant(np.ascontiguousarray(
+                                    b.transpose(0, 2, 1)),
+                                    dtype=tf.bfloat16), adj_y=True))
+    rel2 = np.abs(got2 - want) / (np.abs(want) + 1e-2)
+    assert np.percentile(rel2, 99) < 0.05
+
+
+def test_lrn_gpu_vs_torch():
+    rng = np.random.RandomState(14)
+    x = rng.randn(2, 5, 5, 16).astype(np.float32)
+    t = tf.constant(x)
+    y = tf.nn.lrn(t, depth_radius=2, bias=1.0, alpha=1e-3, beta=0.75)
+    g = tf.gradients(tf.reduce_sum(y * y), [t])[0]
+    got_y, got_g = _run([y, g])
+    tx = torch.from_numpy(x).permute(0, 3, 1, 2).requires_grad_(True)
+    # torch LRN: size = full window, alpha is summed-normalized
+    size = 2 * 2 + 1
+    ty = torch.nn.functional.local_response_norm(tx, size=size,
+                                                 alpha=1e-3 * size, beta=0.75,
+                                                 k=1.0)
+    (ty * ty).sum().backward()
+    want_y = ty.detach().permute(0, 2, 3, 1).numpy()
+    want_g = tx.grad.permute(0, 2, 3, 1).numpy()
+    np.testing.assert_allclose(got_y, want_y, rtol=1e-3, atol=1e-4)
+    np.testing.assert_allclose(got_g, want_g, rtol=1e-2, atol=1e-3)

@@ -6,6 +6,7 @@
 
 #include <map>
 #include <memory>
+#include <set>
 #include <string>
 #include <vector>
 
@@ -102,5 +103,13 @@ Status ConvertGraphDefToGraph(const GraphDef& gdef, Graph* g);
 // Topological order (ignores NextIteration back-edges). Returns error on
 // cycles that are not while-loop back-edges.
 Status TopologicalOrder(const Graph& g, std::vector<Node*>* order);
+
+class Device;
+
+// Session-level graph optimization: common-subexpression elimination +
+// constant folding to fixpoint (graph/optimizer.cc). Nodes named in
+// `preserve` (feeds/fetches/targets) are never removed. `cpu` is the host
+// device used to execute foldable subgraphs. Returns nodes eliminated.
+int OptimizeGraph(Graph* g, Device* cpu, const std::set<std::string>& preserve);
 
 }  // namespace stf

@@ -107,6 +107,17 @@ Status DirectSession::BuildExecutors(const std::vector<std::string>& feeds,
   STF_RETURN_IF_ERROR(ConvertGraphDefToGraph(gdef_copy, graph.get()));
   auto ek = std::make_unique<ExecutorsAndKeys>();
 
+  // ---- 0. Graph optimization (CSE + constant folding) on the pruned-to-be
+  // graph. Feeds/fetches/targets must survive by name. Capability analog of
+  // the reference's OptimizeCSE + DoConstantFolding pre-passes.
+  {
+    std::set<std::string> preserve;
+    for (auto& f : feeds) preserve.insert(ParseTensorName(f).first);
+    for (auto& f : fetches) preserve.insert(ParseTensorName(f).first);
+    for (auto& t : targets) preserve.insert(ParseTensorName(t).first);
+    OptimizeGraph(graph.get(), devices_.LookUp("/cpu:0"), preserve);
+  }
+
   // ---- 1. Placement (before rewrite so feed recvs inherit devices). ----
   std::vector<Node*> order;
   STF_RETURN_IF_ERROR(TopologicalOrder(*graph, &order));

@@ -11,6 +11,8 @@ def variable_op(shape, dtype, name='Variable', container='', shared_name=''):
 
 
 def assign(ref, value, validate_shape=True, use_locking=True, name=None):
+    if hasattr(ref, '_variable'):  # tf.Variable → its mutable ref tensor
+        ref = ref._variable
     t = apply_op('Assign', ref, convert_to_tensor(value, dtype=ref.dtype),
                  validate_shape=validate_shape, use_locking=use_locking,
                  name=name)

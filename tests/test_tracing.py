@@ -14,14 +14,17 @@ def setup_function(_):
 
 
 def _traced_run():
-    a = tf.constant(np.random.randn(64, 64).astype(np.float32), name='a')
+    # `a` is fed so the graph is not all-constant (session-level constant
+    # folding would otherwise legitimately collapse mm into a Const).
+    a = tf.placeholder(tf.float32, [64, 64], name='a')
     b = tf.constant(np.random.randn(64, 64).astype(np.float32), name='b')
     c = tf.matmul(a, b, name='mm')
     d = tf.reduce_sum(c, name='total')
     opts = tf.RunOptions(trace_level=tf.RunOptions.FULL_TRACE)
     md = tf.RunMetadata()
     with tf.Session() as s:
-        v = s.run(d, options=opts, run_metadata=md)
+        v = s.run(d, feed_dict={a: np.random.randn(64, 64).astype(np.float32)},
+                  options=opts, run_metadata=md)
     return v, md
 
 

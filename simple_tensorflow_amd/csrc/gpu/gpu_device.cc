@@ -9,6 +9,10 @@
 // construction; H2D/D2H copies run on side streams fenced by hipEvents.
 #include <hip/hip_runtime.h>
 
+#include <chrono>
+#include <condition_variable>
+#include <cstring>
+#include <memory>
 #include <map>
 #include <mutex>
 #include <set>
@@ -39,17 +43,32 @@ class GpuBfcAllocator : public Allocator {
   void* Allocate(size_t bytes) override {
     if (bytes == 0) bytes = 256;
     size_t size = RoundUp(bytes);
-    std::lock_guard<std::mutex> l(mu_);
+    std::unique_lock<std::mutex> l(mu_);
     Chunk* c = FindFree(size);
     if (!c) {
       if (!Grow(size)) {
-        // Retry once after trying a bigger region failed: hard OOM.
-        LOG(ERROR) << "GPU" << ordinal_ << " BFC: out of memory allocating "
-                   << bytes << " bytes (in use: " << in_use_bytes_ << ")";
-        return nullptr;
+        // Out of device memory right now. Another thread (a concurrent
+        // session / the EventMgr draining d2h copies) may free soon: wait for
+        // a Deallocate and retry, up to ~2 s total (reference
+        // allocator_retry.cc behavior), then dump state and fail hard.
+        auto deadline =
+            std::chrono::steady_clock::now() + std::chrono::seconds(2);
+        while (!c && std::chrono::steady_clock::now() < deadline) {
+          if (free_cv_.wait_until(l, deadline) == std::cv_status::timeout)
+            break;
+          c = FindFree(size);
+          if (!c && Grow(size)) c = FindFree(size);
+        }
+        if (!c) {
+          LOG(ERROR) << "GPU" << ordinal_ << " BFC: out of memory allocating "
+                     << bytes << " bytes after retry\n"
+                     << DumpState();
+          return nullptr;
+        }
+      } else {
+        c = FindFree(size);
+        if (!c) return nullptr;
       }
-      c = FindFree(size);
-      if (!c) return nullptr;
     }
     RemoveFromFree(c);
     // Split if the remainder is useful.
@@ -101,6 +120,7 @@ class GpuBfcAllocator : public Allocator {
       c = p;
     }
     InsertFree(c);
+    free_cv_.notify_all();
   }
 
   MemSpace space() const override { return MemSpace::DEVICE; }
@@ -134,6 +154,31 @@ class GpuBfcAllocator : public Allocator {
   void InsertFree(Chunk* c) { free_.insert(c); }
   void RemoveFromFree(Chunk* c) { free_.erase(c); }
 
+  // Bin/fragmentation summary for the OOM log (gpu_bfc_allocator DumpMemoryLog
+  // analog). Caller holds mu_.
+  std::string DumpState() const {
+    size_t free_bytes = 0, largest_free = 0;
+    std::map<int, std::pair<int, size_t>> bins;  // log2 -> {count, bytes}
+    for (const Chunk* c : free_) {
+      free_bytes += c->size;
+      largest_free = std::max(largest_free, c->size);
+      int b = 0;
+      while ((1ull << (b + 1)) <= c->size) ++b;
+      bins[b].first++;
+      bins[b].second += c->size;
+    }
+    std::string s = "BFC state: total=" + std::to_string(total_bytes_) +
+                    " in_use=" + std::to_string(in_use_bytes_) +
+                    " free=" + std::to_string(free_bytes) +
+                    " largest_free_chunk=" + std::to_string(largest_free) +
+                    " live_allocs=" + std::to_string(by_ptr_.size()) + "\n";
+    for (auto& kv : bins)
+      s += "  free bin 2^" + std::to_string(kv.first) + ": " +
+           std::to_string(kv.second.first) + " chunks, " +
+           std::to_string(kv.second.second) + " bytes\n";
+    return s;
+  }
+
   bool Grow(size_t min_bytes) {
     size_t region = 1ull << 30;  // 1 GiB
     while (region < min_bytes) region <<= 1;
@@ -156,6 +201,7 @@ class GpuBfcAllocator : public Allocator {
 
   int ordinal_;
   std::mutex mu_;
+  std::condition_variable free_cv_;
   std::set<Chunk*, BySize> free_;
   std::map<void*, Chunk*> by_ptr_;
   size_t total_bytes_ = 0;
@@ -172,6 +218,87 @@ class PinnedAllocator : public Allocator {
   }
   void Deallocate(void* ptr, size_t) override { hipHostFree(ptr); }
   const char* name() const override { return "pinned"; }
+};
+
+// ---------------------------------------------------------------------------
+// Debug allocators (reference gpu_debug_allocator.h:33,63 capability analogs),
+// selected with STF_GPU_ALLOC_DEBUG=guard|nan|guard,nan. They wrap the BFC
+// allocator; guard mode brackets every allocation with 64-byte patterns
+// verified on free (catches out-of-bounds kernel writes at the faulting
+// allocation), nan mode fills new/freed memory with f32 quiet-NaNs (catches
+// reads of uninitialized or dangling buffers in any fp32 consumer).
+// ---------------------------------------------------------------------------
+class GpuGuardAllocator : public Allocator {
+ public:
+  static constexpr size_t kGuard = 64;  // bytes each side, 256-aligned base
+  GpuGuardAllocator(Allocator* base, bool guard, bool nan)
+      : base_(base), guard_(guard), nan_(nan) {
+    for (size_t i = 0; i < kGuard / 8; ++i) pattern_[i] = 0x5A5A5A5A5A5A5A5Aull;
+  }
+
+  void* Allocate(size_t bytes) override {
+    if (bytes == 0) bytes = 256;
+    size_t padded = guard_ ? bytes + 2 * 256 : bytes;
+    char* base = (char*)base_->Allocate(padded);
+    if (!base) return nullptr;
+    char* user = base;
+    if (guard_) {
+      user = base + 256;
+      // 256-byte pads keep user alignment; the pattern occupies the 64 bytes
+      // adjacent to the user range on each side.
+      hipMemcpy(user - kGuard, pattern_, kGuard, hipMemcpyHostToDevice);
+      hipMemcpy(user + bytes, pattern_, kGuard, hipMemcpyHostToDevice);
+      std::lock_guard<std::mutex> l(mu_);
+      live_[user] = {base, bytes};
+    }
+    if (nan_)
+      hipMemsetD32((hipDeviceptr_t)user, 0x7fc00000u, bytes / 4);
+    return user;
+  }
+
+  void Deallocate(void* ptr, size_t bytes) override {
+    if (!ptr) return;
+    char* user = (char*)ptr;
+    char* base = user;
+    size_t n = bytes;
+    if (guard_) {
+      std::lock_guard<std::mutex> l(mu_);
+      auto it = live_.find(user);
+      if (it == live_.end()) {
+        LOG(ERROR) << "guard allocator: free of unknown pointer " << ptr;
+        return;
+      }
+      base = it->second.first;
+      n = it->second.second;
+      live_.erase(it);
+    }
+    if (guard_ && !CheckGuards(user, n))
+      LOG(FATAL) << "guard allocator: out-of-bounds GPU write detected on a "
+                 << n << "-byte allocation at " << (void*)user;
+    if (nan_ && n >= 4)
+      hipMemsetD32((hipDeviceptr_t)user, 0x7fc00000u, n / 4);
+    base_->Deallocate(base, 0);
+  }
+
+  MemSpace space() const override { return MemSpace::DEVICE; }
+  int device_ordinal() const override { return base_->device_ordinal(); }
+  const char* name() const override { return "gpu_debug"; }
+
+ private:
+  bool CheckGuards(char* user, size_t bytes) {
+    uint64_t got[kGuard / 8];
+    hipDeviceSynchronize();  // settle pending kernels before reading
+    hipMemcpy(got, user - kGuard, kGuard, hipMemcpyDeviceToHost);
+    if (memcmp(got, pattern_, kGuard) != 0) return false;
+    hipMemcpy(got, user + bytes, kGuard, hipMemcpyDeviceToHost);
+    return memcmp(got, pattern_, kGuard) == 0;
+  }
+
+  Allocator* base_;
+  bool guard_, nan_;
+  uint64_t pattern_[kGuard / 8];
+  std::mutex mu_;
+  std::map<void*, std::pair<char*, size_t>> live_;  // user -> {base, bytes}
 };
 
 // ---------------------------------------------------------------------------
@@ -192,7 +319,19 @@ class GpuDevice : public Device {
     hipStreamDestroy(d2h_);
   }
 
-  Allocator* allocator() override { return &bfc_; }
+  Allocator* allocator() override {
+    static const char* dbg = getenv("STF_GPU_ALLOC_DEBUG");
+    if (dbg && *dbg) {
+      if (!debug_alloc_) {
+        std::string mode(dbg);
+        debug_alloc_.reset(new GpuGuardAllocator(
+            &bfc_, mode.find("guard") != std::string::npos,
+            mode.find("nan") != std::string::npos));
+      }
+      return debug_alloc_.get();
+    }
+    return &bfc_;
+  }
   Allocator* host_allocator() override {
     static PinnedAllocator* pinned = new PinnedAllocator();
     return pinned;
@@ -305,6 +444,7 @@ class GpuDevice : public Device {
  private:
   int ordinal_;
   GpuBfcAllocator bfc_;
+  std::unique_ptr<GpuGuardAllocator> debug_alloc_;
   hipStream_t compute_, h2d_, d2h_;
   bool capturing_ = false;
   // Device buffers referenced by captured hipGraphs but not produced inside

@@ -4,11 +4,20 @@ Analog of the reference's python/client/session.py (BaseSession.run:660),
 speaking to csrc/runtime/session.cc through the pybind module.
 """
 import threading
+import time
 
 import numpy as np
 
 from simple_tensorflow_amd import _core
 from simple_tensorflow_amd.python.framework import dtypes, errors, ops
+from simple_tensorflow_amd.python.lib import monitoring
+
+# Framework metrics (reference core/lib/monitoring instrumentation points).
+_run_counter = monitoring.Counter('/stf/session/runs',
+                                  'Number of Session.run calls')
+_run_time_ms = monitoring.Sampler(
+    '/stf/session/run_time_ms', monitoring.exponential_buckets(0.1, 4, 12),
+    'Session.run wall time (ms)')
 
 _default_session_stack = threading.local()
 
@@ -52,6 +61,14 @@ class Session(object):
                 self._serialized_nodes = len(g._node_list)
 
     def run(self, fetches, feed_dict=None, options=None, run_metadata=None):
+        _run_counter.get_cell().increment()
+        _t0 = time.perf_counter()
+        try:
+            return self._run(fetches, feed_dict, options, run_metadata)
+        finally:
+            _run_time_ms.get_cell().add((time.perf_counter() - _t0) * 1e3)
+
+    def _run(self, fetches, feed_dict=None, options=None, run_metadata=None):
         self._sync_graph()
         flat, restore = _flatten_fetches(fetches)
         fetch_names = []

@@ -163,6 +163,53 @@ class WorkerSessionCreator(object):
         return SessionManager().wait_for_session(self._master, self._config)
 
 
+class _RecoverableSession(object):
+    """Retries `run` on AbortedError/UnavailableError by recreating the
+    underlying session through the creator (reference
+    monitored_session.py:778 _RecoverableSession): a preempted worker or a
+    restarted parameter server surfaces as Aborted/Unavailable; training
+    resumes from the last checkpoint the creator restores."""
+
+    def __init__(self, creator, on_recreate=None, max_retries=None):
+        self._creator = creator
+        self._on_recreate = on_recreate
+        self._max_retries = max_retries
+        self._sess = self._create()
+
+    def _create(self):
+        while True:
+            try:
+                return self._creator.create_session()
+            except (errors.AbortedError, errors.UnavailableError):
+                time.sleep(1.0)
+
+    def run(self, *a, **kw):
+        retries = 0
+        while True:
+            try:
+                return self._sess.run(*a, **kw)
+            except (errors.AbortedError, errors.UnavailableError) as ex:
+                retries += 1
+                if self._max_retries is not None and retries > self._max_retries:
+                    raise
+                tf.logging.warning(
+                    'Session run aborted (%s); recreating session' % ex)
+                try:
+                    self._sess.close()
+                except Exception:  # noqa: BLE001
+                    pass
+                self._sess = self._create()
+                if self._on_recreate:
+                    self._on_recreate(self._sess)
+
+    @property
+    def raw_session(self):
+        return self._sess
+
+    def __getattr__(self, name):
+        return getattr(self._sess, name)
+
+
 class _CoordinatedSession(object):
     """Runs queue-runner threads under a coordinator (reference
     monitored_session.py:829)."""
@@ -184,17 +231,31 @@ class _CoordinatedSession(object):
 
 class MonitoredSession(object):
     def __init__(self, session_creator=None, hooks=None,
-                 stop_grace_period_secs=120):
+                 stop_grace_period_secs=120, recoverable=True):
         self._hooks = list(hooks or [])
         self._creator = session_creator or ChiefSessionCreator()
         for h in self._hooks:
             h.begin()
         self._coord = coord_lib.Coordinator()
-        self._sess = self._creator.create_session()
-        self._raw_sess = self._sess
-        coord_lib.start_queue_runners(self._sess, coord=self._coord)
+
+        def _on_recreate(sess):
+            # A recreated session needs its queue runners restarted and
+            # hooks re-pointed at the live session.
+            self._raw_sess = sess
+            coord_lib.start_queue_runners(sess, coord=self._coord)
+            for h in self._hooks:
+                h.after_create_session(sess, self._coord)
+
+        if recoverable:
+            self._sess = _RecoverableSession(self._creator,
+                                             on_recreate=_on_recreate)
+            self._raw_sess = self._sess.raw_session
+        else:
+            self._sess = self._creator.create_session()
+            self._raw_sess = self._sess
+        coord_lib.start_queue_runners(self._raw_sess, coord=self._coord)
         for h in self._hooks:
-            h.after_create_session(self._sess, self._coord)
+            h.after_create_session(self._raw_sess, self._coord)
         self._csess = _CoordinatedSession(self._sess, self._coord)
         self._should_stop = False
 

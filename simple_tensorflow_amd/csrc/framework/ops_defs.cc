@@ -158,6 +158,12 @@ REGISTER_OP("BatchNormAddReluMi").Input("x: T").Input("scale: float").Input("off
 REGISTER_OP("BatchNormAddReluMiGrad").Input("y_backprop: T").Input("x: T").Input("scale: float").Input("saved_mean: float").Input("saved_inv_std: float").Input("y_relu: T").Output("x_backprop: T").Output("scale_backprop: float").Output("offset_backprop: float").Output("side_backprop: T").Attr("T: {float, bfloat16}").Attr("epsilon: float = 0.0001");
 REGISTER_OP("BatchNormMiGrad").Input("y_backprop: T").Input("x: T").Input("scale: float").Input("saved_mean: float").Input("saved_inv_std: float").Input("y_relu: T").Output("x_backprop: T").Output("scale_backprop: float").Output("offset_backprop: float").Attr("T: {float, bfloat16}").Attr("epsilon: float = 0.0001").Attr("fuse_relu: bool = false");
 REGISTER_OP("L2Loss").Input("t: T").Output("output: T").Attr("T: " FLOATTYPES);
+// Fused LSTM cell pointwise math over the post-GEMM gate matrix [B, 4H]
+// (gate order i, j, f, o — BasicLSTMCell split order). Outputs every
+// activation the backward op needs (reference capability analog:
+// contrib/rnn lstm_ops.cc LSTMBlockCell / LSTMBlockCellGrad).
+REGISTER_OP("LSTMGates").Input("gates: T").Input("c_prev: T").Output("i: T").Output("f: T").Output("o: T").Output("ci: T").Output("cs: T").Output("co: T").Output("h: T").Attr("T: {float, bfloat16}").Attr("forget_bias: float = 1.0");
+REGISTER_OP("LSTMGatesGrad").Input("c_prev: T").Input("i: T").Input("f: T").Input("o: T").Input("ci: T").Input("co: T").Input("dh: T").Input("dcs: T").Output("dgates: T").Output("dc_prev: T").Attr("T: {float, bfloat16}");
 REGISTER_OP("LRN").Input("input: T").Output("output: T").Attr("depth_radius: int = 5").Attr("bias: float = 1.0").Attr("alpha: float = 1.0").Attr("beta: float = 0.5").Attr("T: {float}");
 REGISTER_OP("InTopK").Input("predictions: float").Input("targets: T").Output("precision: bool").Attr("k: int").Attr("T: {int32, int64} = int32");
 

@@ -405,6 +405,16 @@ class LikeOp : public OpKernel {
           for (int64_t i = 0; i < out->NumElements(); ++i) p[i] = 1;
           break;
         }
+        case DT_BFLOAT16: {
+          bfloat16* p = out->flat<bfloat16>();
+          for (int64_t i = 0; i < out->NumElements(); ++i) p[i] = bfloat16(1.0f);
+          break;
+        }
+        case DT_HALF: {
+          uint16_t* p = out->flat<uint16_t>();
+          for (int64_t i = 0; i < out->NumElements(); ++i) p[i] = 0x3C00;  // fp16 1.0
+          break;
+        }
         default:
           ctx->SetStatus(errors::Unimplemented("OnesLike dtype"));
       }

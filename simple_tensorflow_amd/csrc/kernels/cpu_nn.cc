@@ -588,6 +588,89 @@ class LRNOp : public OpKernel {
 };
 REGISTER_KERNEL_BUILDER(Name("LRN").Device(DEVICE_CPU), LRNOp);
 
+// ------------------------- fused LSTM cell pointwise ------------------------
+// CPU reference of the GPU LstmGates kernels (nn_kernels.hip); gate order
+// i, j, f, o as produced by BasicLSTMCell's single [B, 4H] GEMM.
+template <typename T>
+class LSTMGatesOp : public OpKernel {
+ public:
+  explicit LSTMGatesOp(OpKernelConstruction* ctx) : OpKernel(ctx) {
+    ctx->GetAttr("forget_bias", &forget_bias_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& gates = ctx->input(0);
+    const Tensor& c_prev = ctx->input(1);
+    int64_t B = c_prev.shape().dim_size(0);
+    int64_t H = c_prev.shape().dim_size(1);
+    Tensor* outs[7];
+    for (int k = 0; k < 7; ++k) outs[k] = ctx->allocate_output(k, c_prev.shape());
+    const T* g = gates.flat<T>();
+    const T* cp = c_prev.flat<T>();
+    for (int64_t b = 0; b < B; ++b) {
+      const T* gr = g + b * 4 * H;
+      for (int64_t h = 0; h < H; ++h) {
+        float i = 1.f / (1.f + std::exp(-(float)gr[h]));
+        float ci = std::tanh((float)gr[H + h]);
+        float f = 1.f / (1.f + std::exp(-((float)gr[2 * H + h] + forget_bias_)));
+        float o = 1.f / (1.f + std::exp(-(float)gr[3 * H + h]));
+        float cs = f * (float)cp[b * H + h] + i * ci;
+        float co = std::tanh(cs);
+        int64_t idx = b * H + h;
+        outs[0]->flat<T>()[idx] = (T)i;
+        outs[1]->flat<T>()[idx] = (T)f;
+        outs[2]->flat<T>()[idx] = (T)o;
+        outs[3]->flat<T>()[idx] = (T)ci;
+        outs[4]->flat<T>()[idx] = (T)cs;
+        outs[5]->flat<T>()[idx] = (T)co;
+        outs[6]->flat<T>()[idx] = (T)(o * co);
+      }
+    }
+  }
+
+ private:
+  float forget_bias_ = 1.f;
+};
+REGISTER_KERNEL_BUILDER(Name("LSTMGates").Device(DEVICE_CPU).TypeConstraint<float>("T"), LSTMGatesOp<float>);
+REGISTER_KERNEL_BUILDER(Name("LSTMGates").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), LSTMGatesOp<bfloat16>);
+
+template <typename T>
+class LSTMGatesGradOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& c_prev = ctx->input(0);
+    int64_t B = c_prev.shape().dim_size(0);
+    int64_t H = c_prev.shape().dim_size(1);
+    const T* cp = c_prev.flat<T>();
+    const T* i_ = ctx->input(1).flat<T>();
+    const T* f_ = ctx->input(2).flat<T>();
+    const T* o_ = ctx->input(3).flat<T>();
+    const T* ci_ = ctx->input(4).flat<T>();
+    const T* co_ = ctx->input(5).flat<T>();
+    const T* dh = ctx->input(6).flat<T>();
+    const T* dcsn = ctx->input(7).flat<T>();
+    Tensor* dgates = ctx->allocate_output(0, TensorShape({B, 4 * H}));
+    Tensor* dc_prev = ctx->allocate_output(1, c_prev.shape());
+    T* dg = dgates->flat<T>();
+    T* dcp = dc_prev->flat<T>();
+    for (int64_t idx = 0; idx < B * H; ++idx) {
+      int64_t b = idx / H, h = idx % H;
+      float i = (float)i_[idx], f = (float)f_[idx], o = (float)o_[idx];
+      float ci = (float)ci_[idx], co = (float)co_[idx];
+      float dhv = (float)dh[idx];
+      float dcs = dhv * o * (1.f - co * co) + (float)dcsn[idx];
+      T* row = dg + b * 4 * H;
+      row[h] = (T)(dcs * ci * i * (1.f - i));
+      row[H + h] = (T)(dcs * i * (1.f - ci * ci));
+      row[2 * H + h] = (T)(dcs * (float)cp[idx] * f * (1.f - f));
+      row[3 * H + h] = (T)(dhv * co * o * (1.f - o));
+      dcp[idx] = (T)(dcs * f);
+    }
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("LSTMGatesGrad").Device(DEVICE_CPU).TypeConstraint<float>("T"), LSTMGatesGradOp<float>);
+REGISTER_KERNEL_BUILDER(Name("LSTMGatesGrad").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), LSTMGatesGradOp<bfloat16>);
+
 }  // namespace
 
 }  // namespace stf

@@ -118,6 +118,14 @@ hipError_t stf_segment_sum(int, int, const void*, const void*, float*,
                            int64_t, int64_t, int64_t, hipStream_t);
 hipError_t stf_random_normal(uint64_t, void*, void*, int64_t, int, int,
                              hipStream_t);
+hipError_t stf_lstm_gates(int, const void*, const void*, float, void*, void*,
+                          void*, void*, void*, void*, void*, int64_t, int,
+                          hipStream_t);
+hipError_t stf_lstm_gates_grad(int, const void*, const void*, const void*,
+                               const void*, const void*, const void*,
+                               const void*, const void*, void*, void*,
+                               int64_t, int, hipStream_t);
+hipError_t stf_l2loss(int, const void*, float*, int64_t, hipStream_t);
 }
 
 namespace {
@@ -1427,28 +1435,69 @@ class GpuL2LossOp : public OpKernel {
     hipStream_t s = GPU_STREAM(ctx);
     int dt = DtypeCode(x.dtype());
     int64_t n = x.NumElements();
-    Tensor sq = ctx->allocate_temp(x.dtype(), x.shape());
-    OP_HIP_OK(ctx, stf_binary(B_MUL, dt, x.raw_data(), x.raw_data(),
-                              sq.raw_data(), n, s));
-    Tensor acc = ctx->allocate_temp(DT_FLOAT, TensorShape({}));
-    OP_HIP_OK(ctx, ZeroF32(acc.raw_data(), 1, s));
-    OP_HIP_OK(ctx, stf_full_reduce(dt, 0 /*sum*/, sq.raw_data(),
-                                   acc.flat<float>(), n, s));
     Tensor* out = ctx->allocate_output(0, TensorShape({}));
     if (x.dtype() == DT_FLOAT) {
-      OP_HIP_OK(ctx, stf_scale(0, acc.raw_data(), out->raw_data(), 1, 0.5f,
-                               s));
+      OP_HIP_OK(ctx, ZeroF32(out->raw_data(), 1, s));
+      OP_HIP_OK(ctx, stf_l2loss(dt, x.raw_data(), out->flat<float>(), n, s));
     } else {
-      Tensor half = ctx->allocate_temp(DT_FLOAT, TensorShape({}));
-      OP_HIP_OK(ctx, stf_scale(0, acc.raw_data(), half.raw_data(), 1, 0.5f,
-                               s));
-      OP_HIP_OK(ctx, stf_cast(0, CastCode(x.dtype()), half.raw_data(),
+      Tensor acc = ctx->allocate_temp(DT_FLOAT, TensorShape({}));
+      OP_HIP_OK(ctx, ZeroF32(acc.raw_data(), 1, s));
+      OP_HIP_OK(ctx, stf_l2loss(dt, x.raw_data(), acc.flat<float>(), n, s));
+      OP_HIP_OK(ctx, stf_cast(0, CastCode(x.dtype()), acc.raw_data(),
                               out->raw_data(), 1, s));
     }
   }
 };
 REGISTER_KERNEL_BUILDER(Name("L2Loss").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuL2LossOp);
 REGISTER_KERNEL_BUILDER(Name("L2Loss").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuL2LossOp);
+
+// Fused LSTM cell pointwise (nn_kernels.hip LstmGatesKernel); one launch
+// replaces the composed cell's ~17 elementwise/split launches per timestep.
+class GpuLSTMGatesOp : public OpKernel {
+ public:
+  explicit GpuLSTMGatesOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("forget_bias", &forget_bias_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& gates = ctx->input(0);
+    const Tensor& c_prev = ctx->input(1);
+    int64_t B = c_prev.shape().dim_size(0);
+    int64_t H = c_prev.shape().dim_size(1);
+    void* outs[7];
+    for (int k = 0; k < 7; ++k)
+      outs[k] = ctx->allocate_output(k, c_prev.shape())->raw_data();
+    OP_HIP_OK(ctx, stf_lstm_gates(DtypeCode(gates.dtype()), gates.raw_data(),
+                                  c_prev.raw_data(), forget_bias_, outs[0],
+                                  outs[1], outs[2], outs[3], outs[4], outs[5],
+                                  outs[6], B, (int)H, GPU_STREAM(ctx)));
+  }
+
+ private:
+  float forget_bias_ = 1.f;
+};
+REGISTER_KERNEL_BUILDER(Name("LSTMGates").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuLSTMGatesOp);
+REGISTER_KERNEL_BUILDER(Name("LSTMGates").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuLSTMGatesOp);
+
+class GpuLSTMGatesGradOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& c_prev = ctx->input(0);
+    int64_t B = c_prev.shape().dim_size(0);
+    int64_t H = c_prev.shape().dim_size(1);
+    Tensor* dgates = ctx->allocate_output(0, TensorShape({B, 4 * H}));
+    Tensor* dc_prev = ctx->allocate_output(1, c_prev.shape());
+    OP_HIP_OK(ctx, stf_lstm_gates_grad(
+                       DtypeCode(c_prev.dtype()), c_prev.raw_data(),
+                       ctx->input(1).raw_data(), ctx->input(2).raw_data(),
+                       ctx->input(3).raw_data(), ctx->input(4).raw_data(),
+                       ctx->input(5).raw_data(), ctx->input(6).raw_data(),
+                       ctx->input(7).raw_data(), dgates->raw_data(),
+                       dc_prev->raw_data(), B, (int)H, GPU_STREAM(ctx)));
+  }
+};
+REGISTER_KERNEL_BUILDER(Name("LSTMGatesGrad").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuLSTMGatesGradOp);
+REGISTER_KERNEL_BUILDER(Name("LSTMGatesGrad").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuLSTMGatesGradOp);
 
 class GpuReduceOp : public OpKernel {
  public:

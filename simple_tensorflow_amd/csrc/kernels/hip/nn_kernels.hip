@@ -48,26 +48,25 @@ __global__ void BiasAddKernel(const T* __restrict__ x,
     y[i] = (T)((float)x[i] + (float)bias[i % c]);
 }
 
-// column reduction dy[rows, C] -> db[C], partials via f32 atomics into a
-// zeroed f32 scratch (LDS pre-accumulate per block, guide G12).
+// Column sum of dy[rows, c] into db[c] (pre-zeroed f32). 2D grid: x covers
+// columns (thread = one column, coalesced reads), y chunks the rows so the
+// grid reaches a few hundred blocks even for skinny row counts (an LSTM
+// timestep has rows = batch = 20); each thread accumulates its chunk in a
+// register and lands ONE atomicAdd.
 template <typename T>
 __global__ void BiasGradKernel(const T* __restrict__ dy, float* __restrict__ db,
-                               int64_t rows, int c) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* part = (float*)smem;  // [c]
-  for (int i = threadIdx.x; i < c; i += blockDim.x) part[i] = 0.f;
-  __syncthreads();
-  int64_t row0 = (int64_t)blockIdx.x * 64;
-  int64_t row1 = min(row0 + 64, rows);
-  // threads sweep the slab row-major; consecutive threads -> consecutive c
-  for (int64_t r = row0; r < row1; ++r) {
-    const T* src = dy + r * c;
-    for (int i = threadIdx.x; i < c; i += blockDim.x)
-      part[i] += (float)src[i];
-  }
-  __syncthreads();
-  for (int i = threadIdx.x; i < c; i += blockDim.x)
-    if (part[i] != 0.f) atomicAdd(&db[i], part[i]);
+                               int64_t rows, int c, int rows_per_block) {
+  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= c) return;
+  int64_t r0 = (int64_t)blockIdx.y * rows_per_block;
+  int64_t r1 = min(r0 + (int64_t)rows_per_block, rows);
+  float acc = 0.f;
+  const T* p = dy + r0 * c + col;
+  for (int64_t r = r0; r < r1; ++r, p += c) acc += (float)*p;
+  if (gridDim.y == 1)
+    db[col] = acc;  // contract: db was zeroed; single chunk can store
+  else
+    atomicAdd(&db[col], acc);
 }
 
 // ---------------- softmax / xent ----------------
@@ -654,6 +653,74 @@ __global__ void AvgPoolGradKernel(const T* __restrict__ dy,
   }
 }
 
+// ---------------- fused LSTM cell pointwise ----------------
+// One pass over the post-GEMM gate matrix (layout [B, 4H] in BasicLSTMCell
+// split order i, j, f, o) + c_prev -> all cell outputs and every activation
+// the backward pass needs. Replaces the ~17-kernel elementwise chain the
+// composed cell launches per timestep (reference capability analog:
+// contrib/rnn lstm_ops.cc LSTMBlockCell fused CUDA kernel).
+__device__ __forceinline__ float stf_sigmoid(float x) {
+  return 1.f / (1.f + __expf(-x));
+}
+
+template <typename T>
+__global__ void LstmGatesKernel(const T* __restrict__ gates,
+                                const T* __restrict__ c_prev,
+                                float forget_bias, T* __restrict__ i_out,
+                                T* __restrict__ f_out, T* __restrict__ o_out,
+                                T* __restrict__ ci_out, T* __restrict__ cs_out,
+                                T* __restrict__ co_out, T* __restrict__ h_out,
+                                int64_t total, int H) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += stride) {
+    int64_t b = idx / H;
+    int hcol = (int)(idx - b * H);
+    const T* g = gates + b * 4 * H;
+    float i = stf_sigmoid((float)g[hcol]);
+    float ci = tanhf((float)g[H + hcol]);
+    float f = stf_sigmoid((float)g[2 * H + hcol] + forget_bias);
+    float o = stf_sigmoid((float)g[3 * H + hcol]);
+    float cs = f * (float)c_prev[idx] + i * ci;
+    float co = tanhf(cs);
+    i_out[idx] = (T)i;
+    f_out[idx] = (T)f;
+    o_out[idx] = (T)o;
+    ci_out[idx] = (T)ci;
+    cs_out[idx] = (T)cs;
+    co_out[idx] = (T)co;
+    h_out[idx] = (T)(o * co);
+  }
+}
+
+// Backward of the pointwise cell: dh + dc(next) + saved activations ->
+// packed dgates [B, 4H] (feeds the dW / dx GEMMs directly, no 4-way concat)
+// and dc_prev.
+template <typename T>
+__global__ void LstmGatesGradKernel(
+    const T* __restrict__ c_prev, const T* __restrict__ i_,
+    const T* __restrict__ f_, const T* __restrict__ o_,
+    const T* __restrict__ ci_, const T* __restrict__ co_,
+    const T* __restrict__ dh, const T* __restrict__ dcs_next,
+    T* __restrict__ dgates, T* __restrict__ dc_prev, int64_t total, int H) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += stride) {
+    int64_t b = idx / H;
+    int hcol = (int)(idx - b * H);
+    float i = (float)i_[idx], f = (float)f_[idx], o = (float)o_[idx];
+    float ci = (float)ci_[idx], co = (float)co_[idx];
+    float dhv = (float)dh[idx];
+    float dcs = dhv * o * (1.f - co * co) + (float)dcs_next[idx];
+    T* dg = dgates + b * 4 * H;
+    dg[hcol] = (T)(dcs * ci * i * (1.f - i));
+    dg[H + hcol] = (T)(dcs * i * (1.f - ci * ci));
+    dg[2 * H + hcol] = (T)(dcs * (float)c_prev[idx] * f * (1.f - f));
+    dg[3 * H + hcol] = (T)(dhv * co * o * (1.f - o));
+    dc_prev[idx] = (T)(dcs * f);
+  }
+}
+
 }  // namespace
 
 extern "C" {
@@ -673,17 +740,22 @@ hipError_t stf_bias_add(int dtype, const void* x, const void* bias, void* y,
 // db_f32 must be zeroed
 hipError_t stf_bias_grad(int dtype, const void* dy, void* db_f32, int64_t rows,
                          int c, hipStream_t stream) {
-  int64_t blocks = (rows + 63) / 64;
-  if (blocks > 1024) blocks = 1024;
-  size_t lds = (size_t)c * 4;
+  uint32_t colb = (uint32_t)((c + 255) / 256);
+  // enough row chunks to put ~512 blocks on the 256 CUs, but at least 4 rows
+  // per chunk so the atomic traffic stays bounded
+  int64_t yb = 512 / (colb ? colb : 1);
+  int64_t max_yb = (rows + 3) / 4;
+  if (yb > max_yb) yb = max_yb;
+  if (yb < 1) yb = 1;
+  int rpb = (int)((rows + yb - 1) / yb);
+  yb = (rows + rpb - 1) / rpb;
+  dim3 grid(colb, (uint32_t)yb);
   if (dtype == 0)
-    hipLaunchKernelGGL((BiasGradKernel<float>), dim3((uint32_t)blocks),
-                       dim3(256), lds, stream, (const float*)dy,
-                       (float*)db_f32, rows, c);
+    hipLaunchKernelGGL((BiasGradKernel<float>), grid, dim3(256), 0, stream,
+                       (const float*)dy, (float*)db_f32, rows, c, rpb);
   else
-    hipLaunchKernelGGL((BiasGradKernel<__bf16>), dim3((uint32_t)blocks),
-                       dim3(256), lds, stream, (const __bf16*)dy,
-                       (float*)db_f32, rows, c);
+    hipLaunchKernelGGL((BiasGradKernel<__bf16>), grid, dim3(256), 0, stream,
+                       (const __bf16*)dy, (float*)db_f32, rows, c, rpb);
   return hipGetLastError();
 }
 
@@ -891,6 +963,53 @@ hipError_t stf_avg_pool_bwd(int dtype, const void* dy, void* dx, int N, int H,
   else
     hipLaunchKernelGGL((AvgPoolGradKernel<__bf16>), grid, dim3(256), 0, stream,
                        (const __bf16*)dy, (__bf16*)dx, g, total);
+  return hipGetLastError();
+}
+
+hipError_t stf_lstm_gates(int dtype, const void* gates, const void* c_prev,
+                          float forget_bias, void* i_out, void* f_out,
+                          void* o_out, void* ci_out, void* cs_out,
+                          void* co_out, void* h_out, int64_t batch, int H,
+                          hipStream_t stream) {
+  int64_t total = batch * H;
+  dim3 grid = ElemwiseGrid(total, 256, 1);
+  if (dtype == 0)
+    hipLaunchKernelGGL((LstmGatesKernel<float>), grid, dim3(256), 0, stream,
+                       (const float*)gates, (const float*)c_prev, forget_bias,
+                       (float*)i_out, (float*)f_out, (float*)o_out,
+                       (float*)ci_out, (float*)cs_out, (float*)co_out,
+                       (float*)h_out, total, H);
+  else
+    hipLaunchKernelGGL((LstmGatesKernel<__bf16>), grid, dim3(256), 0, stream,
+                       (const __bf16*)gates, (const __bf16*)c_prev,
+                       forget_bias, (__bf16*)i_out, (__bf16*)f_out,
+                       (__bf16*)o_out, (__bf16*)ci_out, (__bf16*)cs_out,
+                       (__bf16*)co_out, (__bf16*)h_out, total, H);
+  return hipGetLastError();
+}
+
+hipError_t stf_lstm_gates_grad(int dtype, const void* c_prev, const void* i_,
+                               const void* f_, const void* o_, const void* ci_,
+                               const void* co_, const void* dh,
+                               const void* dcs_next, void* dgates,
+                               void* dc_prev, int64_t batch, int H,
+                               hipStream_t stream) {
+  int64_t total = batch * H;
+  dim3 grid = ElemwiseGrid(total, 256, 1);
+  if (dtype == 0)
+    hipLaunchKernelGGL((LstmGatesGradKernel<float>), grid, dim3(256), 0,
+                       stream, (const float*)c_prev, (const float*)i_,
+                       (const float*)f_, (const float*)o_, (const float*)ci_,
+                       (const float*)co_, (const float*)dh,
+                       (const float*)dcs_next, (float*)dgates,
+                       (float*)dc_prev, total, H);
+  else
+    hipLaunchKernelGGL((LstmGatesGradKernel<__bf16>), grid, dim3(256), 0,
+                       stream, (const __bf16*)c_prev, (const __bf16*)i_,
+                       (const __bf16*)f_, (const __bf16*)o_,
+                       (const __bf16*)ci_, (const __bf16*)co_,
+                       (const __bf16*)dh, (const __bf16*)dcs_next,
+                       (__bf16*)dgates, (__bf16*)dc_prev, total, H);
   return hipGetLastError();
 }
 

@@ -72,6 +72,29 @@ __global__ void BcastCopyKernel(const T* __restrict__ in, T* __restrict__ out,
   }
 }
 
+
+// 0.5 * sum(x^2) fused in one pass (L2Loss hot path: clip_by_global_norm
+// reduces every gradient tensor; the square-temp + reduce pair costs 3
+// memory passes, this costs 1).
+template <typename T>
+__global__ void SquareSumKernel(const T* __restrict__ x,
+                                float* __restrict__ out, int64_t n) {
+  __shared__ float lds4[4];
+  float acc = 0.f;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    float v = (float)x[i];
+    acc += v * v;
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
+  int wid = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) lds4[wid] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0)
+    atomicAdd(out, 0.5f * (lds4[0] + lds4[1] + lds4[2] + lds4[3]));
+}
+
 // ---- reductions ----
 // full reduce to scalar: two-stage (block partials via atomics on f32)
 template <typename T, int RED>  // 0 sum, 1 max, 2 min
@@ -399,6 +422,18 @@ hipError_t stf_permute(int elem_size, const void* in, void* out, int64_t n,
   else
     hipLaunchKernelGGL((PermuteKernel<uint64_t>), grid, dim3(256), 0, stream,
                        (const uint64_t*)in, (uint64_t*)out, n, args);
+  return hipGetLastError();
+}
+
+hipError_t stf_l2loss(int dtype, const void* x, float* out_f32, int64_t n,
+                      hipStream_t stream) {
+  dim3 grid = ElemwiseGrid(n, 256, 8);
+  if (dtype == 0)
+    hipLaunchKernelGGL((SquareSumKernel<float>), grid, dim3(256), 0, stream,
+                       (const float*)x, out_f32, n);
+  else
+    hipLaunchKernelGGL((SquareSumKernel<__bf16>), grid, dim3(256), 0, stream,
+                       (const __bf16*)x, out_f32, n);
   return hipGetLastError();
 }
 

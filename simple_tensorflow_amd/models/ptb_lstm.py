@@ -24,7 +24,7 @@ def build_ptb_graph(batch=20, seq_len=35, hidden=1500, vocab=10000,
     xs = tf.unstack(emb, num=seq_len, axis=1)  # seq * [batch, hidden]
 
     cell = rnn_cell_impl.MultiRNNCell(
-        [rnn_cell_impl.BasicLSTMCell(hidden) for _ in range(layers)])
+        [rnn_cell_impl.LSTMBlockCell(hidden) for _ in range(layers)])
     outputs, _ = rnn_cell_impl.static_rnn(cell, xs, dtype=tf.bfloat16)
 
     output = tf.concat(outputs, 0)            # [seq*batch, hidden]

@@ -372,7 +372,8 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
 def _needs_fill(op):
     # ops whose grad fn requires every output grad slot present
     return op.type in ('Switch', 'Merge', 'SoftmaxCrossEntropyWithLogits',
-                       'SparseSoftmaxCrossEntropyWithLogits', 'FusedBatchNorm')
+                       'SparseSoftmaxCrossEntropyWithLogits', 'FusedBatchNorm',
+                       'LSTMGates')
 
 
 def _build_consumers(g):

@@ -155,7 +155,7 @@ def _lrn_grad(op, grad):
 
 for _op in ('MaxPoolGrad', 'AvgPoolGrad', 'ReluGrad', 'Relu6Grad',
             'SoftplusGrad', 'EluGrad', 'BiasAddGrad', 'Conv2DBackpropInput',
-            'Conv2DBackpropFilter', 'FusedBatchNormGrad'):
+            'Conv2DBackpropFilter', 'FusedBatchNormGrad', 'LSTMGatesGrad'):
     ops.NoGradient(_op)
 
 
@@ -189,3 +189,15 @@ def _batch_norm_add_relu_mi_grad(op, grad_y, *rest):
     doffset.set_shape(op.inputs[2]._shape)
     dside.set_shape(op.inputs[3]._shape)
     return [dx, dscale, doffset, dside]
+
+@RegisterGradient('LSTMGates')
+def _lstm_gates_grad(op, di, df, do_, dci, dcs, dco, dh):
+    """Only the cs (next-cell-state) and h outputs carry gradients in an
+    unrolled RNN; the activation outputs exist for this op's own use.
+    dcs/dh arrive zero-filled when unused (_needs_fill)."""
+    dgates, dc_prev = apply_op(
+        'LSTMGatesGrad', op.inputs[1], op.outputs[0], op.outputs[1],
+        op.outputs[2], op.outputs[3], op.outputs[5], dh, dcs)
+    dgates.set_shape(op.inputs[0]._shape)
+    dc_prev.set_shape(op.inputs[1]._shape)
+    return dgates, dc_prev

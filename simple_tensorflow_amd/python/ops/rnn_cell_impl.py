@@ -8,6 +8,22 @@ from simple_tensorflow_amd.python.ops import (array_ops, init_ops, math_ops,
 
 
 class RNNCell(object):
+    def _cast_cached(self, var, dtype):
+        """Cast a master-f32 variable to the compute dtype ONCE per graph.
+
+        Cells are called once per unrolled timestep; an uncached cast would
+        re-materialize the full weight matrix every step (35 casts of an
+        18M-element kernel per PTB step before this)."""
+        cache = getattr(self, '_weight_cast_cache', None)
+        if cache is None:
+            cache = self._weight_cast_cache = {}
+        key = (id(ops.get_default_graph()), var.name, dtype)
+        t = cache.get(key)
+        if t is None:
+            t = math_ops.cast(var.ref(), dtype)
+            cache[key] = t
+        return t
+
     @property
     def state_size(self):
         raise NotImplementedError
@@ -89,8 +105,8 @@ class BasicLSTMCell(RNNCell):
             wk = w.ref()
             bk = b.ref()
             if inputs.dtype == dtypes.bfloat16:
-                wk = math_ops.cast(wk, dtypes.bfloat16)
-                bk = math_ops.cast(bk, dtypes.bfloat16)
+                wk = self._cast_cached(w, dtypes.bfloat16)
+                bk = self._cast_cached(b, dtypes.bfloat16)
             gates = nn_ops.bias_add(math_ops.matmul(cat, wk), bk)
             i, j, f, o = array_ops.split(gates, 4, axis=1)
             fb = ops.constant(self._forget_bias, dtype=gates.dtype)
@@ -98,6 +114,56 @@ class BasicLSTMCell(RNNCell):
                 math_ops.sigmoid(i) * self._activation(j)
             new_h = self._activation(new_c) * math_ops.sigmoid(o)
             return new_h, (new_c, new_h)
+
+
+class LSTMBlockCell(RNNCell):
+    """Fused LSTM cell: the per-timestep pointwise math runs as ONE LSTMGates
+    op (HIP LstmGatesKernel) instead of the composed split + ~12 elementwise
+    ops of BasicLSTMCell, and the backward pass is one LSTMGatesGrad emitting
+    the packed [B, 4H] dgates the dW/dx GEMMs consume directly.
+
+    Capability analog of the reference's contrib/rnn LSTMBlockCell
+    (lstm_ops.cc); numerics match BasicLSTMCell(activation=tanh) exactly
+    (tests/test_rnn_fused.py)."""
+
+    def __init__(self, num_units, forget_bias=1.0):
+        self._num_units = num_units
+        self._forget_bias = forget_bias
+
+    @property
+    def state_size(self):
+        return (self._num_units, self._num_units)
+
+    @property
+    def output_size(self):
+        return self._num_units
+
+    def zero_state(self, batch_size, dtype):
+        return (array_ops.zeros([batch_size, self._num_units], dtype),
+                array_ops.zeros([batch_size, self._num_units], dtype))
+
+    def __call__(self, inputs, state, scope=None):
+        c, h = state
+        with variable_scope.variable_scope(scope or 'lstm_block_cell'):
+            in_dim = inputs._shape[-1]
+            w = variable_scope.get_variable(
+                'kernel', [in_dim + self._num_units, 4 * self._num_units])
+            b = variable_scope.get_variable(
+                'bias', [4 * self._num_units],
+                initializer=init_ops.zeros_initializer())
+            wk = w.ref()
+            bk = b.ref()
+            if inputs.dtype == dtypes.bfloat16:
+                wk = self._cast_cached(w, dtypes.bfloat16)
+                bk = self._cast_cached(b, dtypes.bfloat16)
+            cat = array_ops.concat([inputs, h], 1)
+            gates = nn_ops.bias_add(math_ops.matmul(cat, wk), bk)
+            outs = ops.apply_op('LSTMGates', gates, c,
+                                forget_bias=float(self._forget_bias))
+            i, f, o, ci, cs, co, new_h = outs
+            for t in outs:
+                t.set_shape(c._shape)
+            return new_h, (cs, new_h)
 
 
 class MultiRNNCell(RNNCell):

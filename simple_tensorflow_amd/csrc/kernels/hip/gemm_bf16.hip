@@ -358,7 +358,17 @@ hipError_t LaunchVariant(const uint16_t* A, const uint16_t* B, void* C,
                        a_ag, B, C, bias, M, N, K, ldb, beta, 1);
   };
 #define STF_PICK(WM_, WN_, TM, TN)                                           do {                                                                         if (short_k)                                                                 launch(GemmBf16NT<2, 2, WM_, WN_, A_KM, B_KM, OUT_BF16, FUSE_RELU,                           false, false>,                                                  TM, TN);                                                          else                                                                         launch(GemmBf16NT<2, 2, WM_, WN_, A_KM, B_KM, OUT_BF16, FUSE_RELU>,               TM, TN);                                                        } while (0)
-  if (N >= 128 && M >= 128) {
+  if (M <= 32 && N >= 128) {
+    // Skinny-M (RNN batch rows, FC with tiny batch): a 32-row tile spans the
+    // whole M in one block row, so B is streamed exactly once; 1x4 wave
+    // layout keeps all four waves on distinct N columns.
+    if (short_k)
+      launch(GemmBf16NT<1, 4, 2, 2, A_KM, B_KM, OUT_BF16, FUSE_RELU, false,
+                        false>, 32, 128);
+    else
+      launch(GemmBf16NT<1, 4, 2, 2, A_KM, B_KM, OUT_BF16, FUSE_RELU>, 32,
+             128);
+  } else if (N >= 128 && M >= 128) {
     STF_PICK(4, 4, 128, 128);
   } else if (N >= 128) {
     STF_PICK(2, 4, 64, 128);
@@ -381,7 +391,9 @@ hipError_t LaunchSplitK(const uint16_t* A, const uint16_t* B, float* C,
     hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0, stream,
                        a_ag, B, C, nullptr, M, N, K, ldb, 0.f, splitk);
   };
-  if (N >= 128 && M >= 128) {
+  if (M <= 32 && N >= 128) {
+    launch(GemmBf16NT<1, 4, 2, 2, A_KM, B_KM, false, false, true>, 32, 128);
+  } else if (N >= 128 && M >= 128) {
     launch(GemmBf16NT<2, 2, 4, 4, A_KM, B_KM, false, false, true>, 128, 128);
   } else if (N >= 128) {
     launch(GemmBf16NT<2, 2, 2, 4, A_KM, B_KM, false, false, true>, 64, 128);

@@ -416,7 +416,8 @@ class PySession {
   py::list LastStats() {
     py::list out;
     for (auto& ns : last_stats_)
-      out.append(py::make_tuple(ns.node, ns.op, ns.start_us, ns.end_us));
+      out.append(py::make_tuple(ns.node, ns.op, ns.start_us, ns.end_us,
+                                ns.device));
     return out;
   }
 

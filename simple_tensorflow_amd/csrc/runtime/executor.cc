@@ -500,12 +500,16 @@ class ExecutorState {
     OpKernelContext ctx(item.kernel, dev, std::move(inputs));
     FillCtx(&ctx, t);
     int64_t t0 = 0;
-    if (args_.stats)
+    void* trace_tag = nullptr;
+    if (args_.stats) {
       t0 = std::chrono::duration_cast<std::chrono::microseconds>(
                std::chrono::system_clock::now().time_since_epoch())
                .count();
+      if (args_.stats->gpu_pre) trace_tag = args_.stats->gpu_pre(dev);
+    }
     dev->Compute(item.kernel, &ctx);
     if (args_.stats) {
+      if (trace_tag) args_.stats->gpu_post(dev, trace_tag, n->name(), n->op());
       int64_t t1 = std::chrono::duration_cast<std::chrono::microseconds>(
                        std::chrono::system_clock::now().time_since_epoch())
                        .count();

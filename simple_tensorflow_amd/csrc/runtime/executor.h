@@ -59,6 +59,9 @@ struct NodeStats {
   std::string op;
   int64_t start_us = 0;
   int64_t end_us = 0;
+  // "" = host enqueue lane; the GpuTracer appends device-lane entries
+  // ("/device:GPU:n/stream:compute") with hardware kernel intervals.
+  std::string device;
 };
 
 struct StatsCollector {
@@ -67,8 +70,13 @@ struct StatsCollector {
   void Add(const std::string& node, const std::string& op, int64_t start_us,
            int64_t end_us) {
     std::lock_guard<std::mutex> l(mu);
-    stats.push_back({node, op, start_us, end_us});
+    stats.push_back({node, op, start_us, end_us, ""});
   }
+  // Hardware-trace hooks (gpu/gpu_tracer.h): bracket a GPU node's kernel
+  // enqueue with stream events. Unset when tracing is off or no GPU.
+  std::function<void*(Device*)> gpu_pre;
+  std::function<void(Device*, void*, const std::string&, const std::string&)>
+      gpu_post;
 };
 
 struct ExecutorArgs {

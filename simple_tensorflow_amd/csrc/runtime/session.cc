@@ -1,5 +1,7 @@
 #include "runtime/session.h"
 
+#include "gpu/gpu_tracer.h"
+
 #include <algorithm>
 #include <condition_variable>
 #include <set>
@@ -499,6 +501,22 @@ Status DirectSession::Run(
     }
   }
 
+  // Hardware trace: when stats are requested and a GPU partition exists,
+  // bracket every GPU node with stream events (gpu/gpu_tracer.h).
+  GpuTracer tracer;
+  if (stats) {
+    std::vector<Device*> gpus;
+    for (auto& item : ek->items)
+      if (item.device->is_gpu()) gpus.push_back(item.device);
+    if (!gpus.empty() && tracer.Start(gpus).ok()) {
+      stats->gpu_pre = [&tracer](Device* d) { return tracer.Pre(d); };
+      stats->gpu_post = [&tracer](Device* d, void* t, const std::string& n,
+                                  const std::string& o) {
+        tracer.Post(d, t, n, o);
+      };
+    }
+  }
+
   // Run all partition executors.
   int64_t step_id;
   {
@@ -619,6 +637,11 @@ Status DirectSession::Run(
       capture_dev->EndGraphCapture(&exec);  // abort capture
       ek->capture_broken = true;
     }
+  }
+  if (stats) {
+    stats->gpu_pre = nullptr;
+    stats->gpu_post = nullptr;
+    (void)tracer.Collect(stats);
   }
   if (!agg.ok()) {
     if (do_capture) {

@@ -40,11 +40,22 @@ class NodeExecStats(object):
 
 
 class StepStats(object):
-    """Wraps the pybind (node, op, start_us, end_us) tuples."""
+    """Wraps the pybind (node, op, start_us, end_us[, device]) tuples.
+
+    Entries with an empty device are the host enqueue lane; entries tagged
+    by the GpuTracer (csrc/gpu/gpu_tracer.cc) land in their own
+    /device:GPU:n/stream:compute lane with hardware kernel intervals."""
 
     def __init__(self, raw):
-        self.dev_stats = [DeviceStepStats('/job:localhost/replica:0/task:0',
-                                          raw)]
+        lanes = {}
+        for entry in raw:
+            dev = entry[4] if len(entry) > 4 and entry[4] else \
+                '/job:localhost/replica:0/task:0'
+            lanes.setdefault(dev, []).append(entry[:4])
+        if not lanes:
+            lanes['/job:localhost/replica:0/task:0'] = []
+        self.dev_stats = [DeviceStepStats(dev, entries)
+                          for dev, entries in sorted(lanes.items())]
 
     def SerializeToString(self):
         out = b''

@@ -2,6 +2,8 @@
 python/client/timeline.py + step_stats.proto analogs)."""
 import json
 
+import pytest
+
 import numpy as np
 
 import simple_tensorflow_amd as tf
@@ -82,3 +84,31 @@ def test_no_trace_without_options():
     with tf.Session() as s:
         s.run(a, run_metadata=md)  # no options -> no stats
     assert md.step_stats is None
+
+
+@pytest.mark.gpu
+def test_gpu_device_lane_in_step_stats():
+    """GpuTracer (csrc/gpu/gpu_tracer.cc): FULL_TRACE on a GPU graph must
+    produce a /device:GPU lane with hardware kernel intervals."""
+    tf.reset_default_graph()
+    a = tf.placeholder(tf.float32, [256, 256], name='a')
+    b = tf.constant(np.random.randn(256, 256).astype(np.float32), name='b')
+    c = tf.matmul(a, b, name='mm')
+    d = tf.reduce_sum(c, name='total')
+    opts = tf.RunOptions(trace_level=tf.RunOptions.FULL_TRACE)
+    md = tf.RunMetadata()
+    with tf.Session() as s:
+        s.run(d, feed_dict={a: np.random.randn(256, 256).astype(np.float32)},
+              options=opts, run_metadata=md)
+    gpu_lanes = [ds for ds in md.step_stats.dev_stats
+                 if '/device:GPU' in ds.device]
+    assert gpu_lanes, 'no GPU hardware lane in step stats'
+    names = {ns.node_name for ds in gpu_lanes for ns in ds.node_stats}
+    assert 'mm' in names
+    for ds in gpu_lanes:
+        for ns in ds.node_stats:
+            assert ns.all_start_micros > 0
+            assert ns.op_end_rel_micros >= 0
+    # chrome trace renders the device lane
+    ctf = timeline.Timeline(md.step_stats).generate_chrome_trace_format()
+    assert '/device:GPU' in ctf

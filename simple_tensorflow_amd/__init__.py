@@ -15,6 +15,7 @@ from simple_tensorflow_amd.python.framework import ops as _ops
 from simple_tensorflow_amd.python.client import session as _session
 from simple_tensorflow_amd.python.ops import (  # noqa: F401
     array_ops as _array_ops,
+    linalg_ops as _linalg_ops,
     clip_ops as _clip_ops,
     control_flow_ops as _control_flow_ops,
     gradients_impl as _gradients_impl,
@@ -226,6 +227,19 @@ depth_to_space = _array_ops.depth_to_space
 mirror_pad = _array_ops.mirror_pad
 reverse_sequence = _array_ops.reverse_sequence
 bitcast = _array_ops.bitcast
+
+# linear algebra (reference tf.linalg / root exports)
+cholesky = _linalg_ops.cholesky
+matrix_determinant = _linalg_ops.matrix_determinant
+matrix_inverse = _linalg_ops.matrix_inverse
+matrix_solve = _linalg_ops.matrix_solve
+matrix_triangular_solve = _linalg_ops.matrix_triangular_solve
+matrix_solve_ls = _linalg_ops.matrix_solve_ls
+qr = _linalg_ops.qr
+svd = _linalg_ops.svd
+self_adjoint_eig = _linalg_ops.self_adjoint_eig
+self_adjoint_eigvals = _linalg_ops.self_adjoint_eigvals
+eye = _linalg_ops.eye
 cumprod = _math_ops.cumprod
 tan = _math_ops.tan
 asin = _math_ops.asin

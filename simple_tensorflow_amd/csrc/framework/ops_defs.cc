@@ -167,6 +167,21 @@ REGISTER_OP("LSTMGatesGrad").Input("c_prev: T").Input("i: T").Input("f: T").Inpu
 REGISTER_OP("LRN").Input("input: T").Output("output: T").Attr("depth_radius: int = 5").Attr("bias: float = 1.0").Attr("alpha: float = 1.0").Attr("beta: float = 0.5").Attr("T: {float}");
 REGISTER_OP("InTopK").Input("predictions: float").Input("targets: T").Output("precision: bool").Attr("k: int").Attr("T: {int32, int64} = int32");
 
+// --------------------------- linear algebra --------------------------------
+// Batched dense decompositions (reference core/ops/linalg_ops.cc). CPU
+// kernels in kernels/cpu_linalg.cc (LU w/ partial pivoting, Cholesky,
+// Householder QR, one-sided Jacobi SVD, cyclic Jacobi symmetric eig).
+REGISTER_OP("Cholesky").Input("input: T").Output("output: T").Attr("T: {float, double}");
+REGISTER_OP("CholeskyGrad").Input("l: T").Input("grad: T").Output("output: T").Attr("T: {float, double}");
+REGISTER_OP("MatrixDeterminant").Input("input: T").Output("output: T").Attr("T: {float, double}");
+REGISTER_OP("MatrixInverse").Input("input: T").Output("output: T").Attr("adjoint: bool = false").Attr("T: {float, double}");
+REGISTER_OP("MatrixSolve").Input("matrix: T").Input("rhs: T").Output("output: T").Attr("adjoint: bool = false").Attr("T: {float, double}");
+REGISTER_OP("MatrixTriangularSolve").Input("matrix: T").Input("rhs: T").Output("output: T").Attr("lower: bool = true").Attr("adjoint: bool = false").Attr("T: {float, double}");
+REGISTER_OP("MatrixSolveLs").Input("matrix: T").Input("rhs: T").Input("l2_regularizer: double").Output("output: T").Attr("T: {float, double}").Attr("fast: bool = true");
+REGISTER_OP("Qr").Input("input: T").Output("q: T").Output("r: T").Attr("full_matrices: bool = false").Attr("T: {float, double}");
+REGISTER_OP("Svd").Input("input: T").Output("s: T").Output("u: T").Output("v: T").Attr("compute_uv: bool = true").Attr("full_matrices: bool = false").Attr("T: {float, double}");
+REGISTER_OP("SelfAdjointEigV2").Input("input: T").Output("e: T").Output("v: T").Attr("compute_v: bool = true").Attr("T: {float, double}");
+
 // --------------------------- state / training -------------------------------
 REGISTER_OP("VariableV2").Output("ref: Ref(dtype)").Attr("shape: shape").Attr("dtype: type").Attr("container: string = ''").Attr("shared_name: string = ''").SetIsStateful();
 REGISTER_OP("Variable").Output("ref: Ref(dtype)").Attr("shape: shape").Attr("dtype: type").Attr("container: string = ''").Attr("shared_name: string = ''").SetIsStateful();

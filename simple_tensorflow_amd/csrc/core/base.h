@@ -13,6 +13,7 @@
 #include <cstring>
 #include <cmath>
 #include <memory>
+#include <complex>
 #include <string>
 #include <vector>
 #include <sstream>
@@ -229,6 +230,8 @@ inline size_t DataTypeSize(DataType dt) {
     case DT_BFLOAT16: return 2;
     case DT_UINT16: return 2;
     case DT_HALF: return 2;
+    case DT_COMPLEX64: return 8;
+    case DT_COMPLEX128: return 16;
     case DT_STRING: return sizeof(std::string);
     default: return 0;
   }
@@ -249,6 +252,8 @@ inline const char* DataTypeString(DataType dt) {
     case DT_UINT16: return "uint16";
     case DT_HALF: return "half";
     case DT_RESOURCE: return "resource";
+    case DT_COMPLEX64: return "complex64";
+    case DT_COMPLEX128: return "complex128";
     default: return "invalid";
   }
 }
@@ -282,6 +287,8 @@ template <> struct DataTypeToEnum<bool> { static constexpr DataType v = DT_BOOL;
 template <> struct DataTypeToEnum<bfloat16> { static constexpr DataType v = DT_BFLOAT16; };
 template <> struct DataTypeToEnum<half16> { static constexpr DataType v = DT_HALF; };
 template <> struct DataTypeToEnum<std::string> { static constexpr DataType v = DT_STRING; };
+template <> struct DataTypeToEnum<std::complex<float>> { static constexpr DataType v = DT_COMPLEX64; };
+template <> struct DataTypeToEnum<std::complex<double>> { static constexpr DataType v = DT_COMPLEX128; };
 
 // ---------------------------------------------------------------------------
 // TensorShape

@@ -6,7 +6,7 @@
 
 namespace stf {
 
-#define NUMTYPES "{float, double, int32, int64, bfloat16, half, uint8, int8}"
+#define NUMTYPES "{float, double, int32, int64, bfloat16, half, uint8, int8, complex64}"
 #define REALTYPES "{float, double, int32, int64, bfloat16, half}"
 #define FLOATTYPES "{float, double, bfloat16, half}"
 
@@ -166,6 +166,27 @@ REGISTER_OP("LSTMGates").Input("gates: T").Input("c_prev: T").Output("i: T").Out
 REGISTER_OP("LSTMGatesGrad").Input("c_prev: T").Input("i: T").Input("f: T").Input("o: T").Input("ci: T").Input("co: T").Input("dh: T").Input("dcs: T").Output("dgates: T").Output("dc_prev: T").Attr("T: {float, bfloat16}");
 REGISTER_OP("LRN").Input("input: T").Output("output: T").Attr("depth_radius: int = 5").Attr("bias: float = 1.0").Attr("alpha: float = 1.0").Attr("beta: float = 0.5").Attr("T: {float}");
 REGISTER_OP("InTopK").Input("predictions: float").Input("targets: T").Output("precision: bool").Attr("k: int").Attr("T: {int32, int64} = int32");
+
+// ----------------------------- spectral (FFT) ------------------------------
+// Reference core/ops/spectral_ops.cc. CPU kernels in kernels/cpu_fft.cc
+// (iterative radix-2 + Bluestein chirp-z for arbitrary lengths, double
+// accumulation).
+REGISTER_OP("FFT").Input("input: complex64").Output("output: complex64");
+REGISTER_OP("IFFT").Input("input: complex64").Output("output: complex64");
+REGISTER_OP("FFT2D").Input("input: complex64").Output("output: complex64");
+REGISTER_OP("IFFT2D").Input("input: complex64").Output("output: complex64");
+REGISTER_OP("FFT3D").Input("input: complex64").Output("output: complex64");
+REGISTER_OP("IFFT3D").Input("input: complex64").Output("output: complex64");
+REGISTER_OP("RFFT").Input("input: float").Input("fft_length: int32").Output("output: complex64");
+REGISTER_OP("IRFFT").Input("input: complex64").Input("fft_length: int32").Output("output: float");
+REGISTER_OP("RFFT2D").Input("input: float").Input("fft_length: int32").Output("output: complex64");
+REGISTER_OP("IRFFT2D").Input("input: complex64").Input("fft_length: int32").Output("output: float");
+// complex construction/accessors (reference math_ops.cc)
+REGISTER_OP("Complex").Input("real: T").Input("imag: T").Output("out: Tout").Attr("T: {float, double} = float").Attr("Tout: {complex64, complex128} = complex64");
+REGISTER_OP("Real").Input("input: T").Output("output: Tout").Attr("T: {complex64, complex128} = complex64").Attr("Tout: {float, double} = float");
+REGISTER_OP("Imag").Input("input: T").Output("output: Tout").Attr("T: {complex64, complex128} = complex64").Attr("Tout: {float, double} = float");
+REGISTER_OP("Conj").Input("input: T").Output("output: T").Attr("T: {complex64, complex128} = complex64");
+REGISTER_OP("ComplexAbs").Input("x: T").Output("y: Tout").Attr("T: {complex64, complex128} = complex64").Attr("Tout: {float, double} = float");
 
 // --------------------------- linear algebra --------------------------------
 // Batched dense decompositions (reference core/ops/linalg_ops.cc). CPU

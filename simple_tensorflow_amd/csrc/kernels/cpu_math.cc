@@ -356,6 +356,7 @@ class AddNOp : public OpKernel {
 };
 REGISTER_CPU_KERNEL_TYPES("AddN", AddNOp)
 REGISTER_KERNEL_BUILDER(Name("AddN").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), AddNOp<bfloat16>);
+REGISTER_KERNEL_BUILDER(Name("AddN").Device(DEVICE_CPU).TypeConstraint<std::complex<float>>("T"), AddNOp<std::complex<float>>);
 
 // -------------------------------- MatMul ------------------------------------
 template <typename T>

@@ -2,6 +2,9 @@
 // tf_session_helper.cc (numpy <-> Tensor, session lifecycle, op registry
 // introspection for the Python op_def_library).
 #include <pybind11/numpy.h>
+#include <pybind11/complex.h>
+
+#include <complex>
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
@@ -35,6 +38,8 @@ DataType NumpyToDataType(const py::array& a) {
   if (kind == 'u' && size == 1) return DT_UINT8;
   if (kind == 'u' && size == 2) return DT_UINT16;  // also bf16 carrier
   if (kind == 'b') return DT_BOOL;
+  if (kind == 'c' && size == 8) return DT_COMPLEX64;
+  if (kind == 'c' && size == 16) return DT_COMPLEX128;
   return DT_INVALID;
 }
 
@@ -66,6 +71,18 @@ py::object TensorToPy(const Tensor& t) {
     case DT_BFLOAT16:
     case DT_HALF: fmt = py::format_descriptor<uint16_t>::format(); break;
     case DT_BOOL: fmt = py::format_descriptor<bool>::format(); break;
+    case DT_COMPLEX64: {
+      py::array outc(py::dtype::of<std::complex<float>>(), shape);
+      std::memcpy(outc.mutable_data(), t.raw_data(),
+                  (size_t)t.NumElements() * 8);
+      return outc;
+    }
+    case DT_COMPLEX128: {
+      py::array outc(py::dtype::of<std::complex<double>>(), shape);
+      std::memcpy(outc.mutable_data(), t.raw_data(),
+                  (size_t)t.NumElements() * 16);
+      return outc;
+    }
     case DT_STRING: {
       // Return list (or scalar) of bytes.
       const std::string* p = t.flat<std::string>();

@@ -56,11 +56,13 @@ bool = DType(10, 'bool', np.bool_, False, False)
 bfloat16 = DType(14, 'bfloat16', np.uint16, True, False)
 uint16 = DType(17, 'uint16', np.uint16, False, True)
 float16 = DType(19, 'float16', np.float16, True, False)
+complex64 = DType(8, 'complex64', np.complex64, False, False)
+complex128 = DType(18, 'complex128', np.complex128, False, False)
 half = float16
 double = float64
 
 _ALL = [float32, float64, int32, uint8, int16, int8, string, int64, bool,
-        bfloat16, uint16, float16]
+        bfloat16, uint16, float16, complex64, complex128]
 _BY_ENUM = {d._enum: d for d in _ALL}
 _BY_NAME = {d.name: d for d in _ALL}
 _BY_NAME.update({'float': float32, 'double': float64, 'half': float16})

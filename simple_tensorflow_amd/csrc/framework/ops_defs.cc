@@ -167,6 +167,12 @@ REGISTER_OP("LSTMGatesGrad").Input("c_prev: T").Input("i: T").Input("f: T").Inpu
 REGISTER_OP("LRN").Input("input: T").Output("output: T").Attr("depth_radius: int = 5").Attr("bias: float = 1.0").Attr("alpha: float = 1.0").Attr("beta: float = 0.5").Attr("T: {float}");
 REGISTER_OP("InTopK").Input("predictions: float").Input("targets: T").Output("precision: bool").Attr("k: int").Attr("T: {int32, int64} = int32");
 
+// --------------------------------- CTC -------------------------------------
+// Reference core/ops/ctc_ops.cc / kernels/ctc_loss_op.cc. Kernel in
+// kernels/cpu_ctc.cc (log-space forward-backward).
+REGISTER_OP("CTCLoss").Input("inputs: float").Input("labels_indices: int64").Input("labels_values: int32").Input("sequence_length: int32").Output("loss: float").Output("gradient: float").Attr("preprocess_collapse_repeated: bool = false").Attr("ctc_merge_repeated: bool = true");
+REGISTER_OP("CTCGreedyDecoder").Input("inputs: float").Input("sequence_length: int32").Output("decoded_indices: int64").Output("decoded_values: int64").Output("decoded_shape: int64").Output("log_probability: float").Attr("merge_repeated: bool = false");
+
 // ----------------------------- spectral (FFT) ------------------------------
 // Reference core/ops/spectral_ops.cc. CPU kernels in kernels/cpu_fft.cc
 // (iterative radix-2 + Bluestein chirp-z for arbitrary lengths, double

@@ -17,6 +17,7 @@ from simple_tensorflow_amd.python.ops import (  # noqa: F401
     array_ops as _array_ops,
     linalg_ops as _linalg_ops,
     spectral_ops as _spectral_ops,
+    string_ops as _string_ops,
     clip_ops as _clip_ops,
     control_flow_ops as _control_flow_ops,
     gradients_impl as _gradients_impl,
@@ -254,6 +255,19 @@ real = _spectral_ops.real
 imag = _spectral_ops.imag
 conj = _spectral_ops.conj
 spectral = _spectral_ops
+
+# strings (reference tf.string_* root exports)
+string_join = _string_ops.string_join
+string_split = _string_ops.string_split
+substr = _string_ops.substr
+string_to_hash_bucket = _string_ops.string_to_hash_bucket
+string_to_hash_bucket_fast = _string_ops.string_to_hash_bucket_fast
+string_to_hash_bucket_strong = _string_ops.string_to_hash_bucket_strong
+string_to_number = _string_ops.string_to_number
+reduce_join = _string_ops.reduce_join
+encode_base64 = _string_ops.encode_base64
+decode_base64 = _string_ops.decode_base64
+strings = _string_ops
 cumprod = _math_ops.cumprod
 tan = _math_ops.tan
 asin = _math_ops.asin

@@ -374,6 +374,17 @@ REGISTER_OP("Mod").Input("x: T").Input("y: T").Output("z: T").Attr("T: {float, d
 REGISTER_OP("ApproximateEqual").Input("x: T").Input("y: T").Output("z: bool").Attr("T: " NUMTYPES).Attr("tolerance: float = 1e-05");
 REGISTER_OP("AsString").Input("input: T").Output("output: string").Attr("T: {float, double, int32, int64, bool}").Attr("precision: int = -1").Attr("scientific: bool = false").Attr("shortest: bool = false").Attr("width: int = -1").Attr("fill: string = ''");
 REGISTER_OP("DecodeRaw").Input("bytes: string").Output("output: out_type").Attr("out_type: {float, double, int32, uint8, int16, int8, int64}").Attr("little_endian: bool = true");
+// string family (reference core/ops/string_ops.cc; kernels/cpu_strings.cc)
+REGISTER_OP("StringJoin").Input("inputs: N * string").Output("output: string").Attr("N: int >= 1").Attr("separator: string = ''");
+REGISTER_OP("StringSplit").Input("input: string").Input("delimiter: string").Output("indices: int64").Output("values: string").Output("shape: int64").Attr("skip_empty: bool = true");
+REGISTER_OP("Substr").Input("input: string").Input("pos: T").Input("len: T").Output("output: string").Attr("T: {int32, int64}");
+REGISTER_OP("StringToHashBucket").Input("string_tensor: string").Output("output: int64").Attr("num_buckets: int >= 1");
+REGISTER_OP("StringToHashBucketFast").Input("input: string").Output("output: int64").Attr("num_buckets: int >= 1");
+REGISTER_OP("StringToHashBucketStrong").Input("input: string").Output("output: int64").Attr("num_buckets: int >= 1").Attr("key: list(int)");
+REGISTER_OP("StringToNumber").Input("string_tensor: string").Output("output: out_type").Attr("out_type: {float, double, int32, int64} = float");
+REGISTER_OP("ReduceJoin").Input("inputs: string").Input("reduction_indices: int32").Output("output: string").Attr("keep_dims: bool = false").Attr("separator: string = ''");
+REGISTER_OP("EncodeBase64").Input("input: string").Output("output: string").Attr("pad: bool = false");
+REGISTER_OP("DecodeBase64").Input("input: string").Output("output: string");
 // lookup tables (reference core/ops/data_flow_ops.cc lookup section)
 REGISTER_OP("HashTable").Output("table_handle: Ref(string)").Attr("container: string = ''").Attr("shared_name: string = ''").Attr("key_dtype: type").Attr("value_dtype: type").SetIsStateful();
 REGISTER_OP("MutableHashTable").Output("table_handle: Ref(string)").Attr("container: string = ''").Attr("shared_name: string = ''").Attr("key_dtype: type").Attr("value_dtype: type").SetIsStateful();

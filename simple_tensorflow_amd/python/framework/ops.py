@@ -491,16 +491,24 @@ def constant(value, dtype=None, shape=None, name='Const'):
         return value
     if dtype is not None:
         dtype = dtypes.as_dtype(dtype)
-    is_str_list = isinstance(value, (list, tuple)) and len(value) > 0 and \
-        all(isinstance(v, (str, bytes)) for v in value)
-    if dtype is dtypes.string or (
-            dtype is None and (isinstance(value, (str, bytes)) or is_str_list)):
-        vals = value
-        if isinstance(vals, (str, bytes)):
-            vals = [vals]
+    def _is_str_data(v):
+        if isinstance(v, (str, bytes)):
+            return True
+        if isinstance(v, np.ndarray):
+            return v.dtype.kind in ('U', 'S', 'O') and \
+                all(isinstance(e, (str, bytes)) for e in v.reshape(-1))
+        if isinstance(v, (list, tuple)) and len(v) > 0:
+            return all(_is_str_data(e) for e in v)
+        return False
+
+    if dtype is dtypes.string or (dtype is None and _is_str_data(value)):
+        if isinstance(value, (str, bytes)):
+            vals = [value]
             dims = []
         else:
-            dims = [len(vals)]
+            arr = np.array(value, dtype=object)
+            dims = list(arr.shape)
+            vals = list(arr.reshape(-1))
         vals = [v.encode() if isinstance(v, str) else v for v in vals]
         tp = pbwire.tensor_proto(7, dims, string_vals=vals)
         op = g.create_op('Const', [], [dtypes.string],

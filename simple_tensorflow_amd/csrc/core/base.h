@@ -128,6 +128,7 @@ STF_DECLARE_ERROR(Aborted, ABORTED)
 STF_DECLARE_ERROR(Cancelled, CANCELLED)
 STF_DECLARE_ERROR(Unavailable, UNAVAILABLE)
 STF_DECLARE_ERROR(DataLoss, DATA_LOSS)
+STF_DECLARE_ERROR(DeadlineExceeded, DEADLINE_EXCEEDED)
 #undef STF_DECLARE_ERROR
 }  // namespace errors
 

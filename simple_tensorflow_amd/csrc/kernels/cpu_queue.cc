@@ -90,8 +90,31 @@ class QueueResource : public ResourceBase {
       std::vector<std::pair<PendingDq, std::vector<std::vector<Tensor>>>> fire;
       std::vector<PendingDq> oor;
       std::vector<PendingEq> eq_fire;
+      std::vector<PendingDq> cancelled_dq;
+      std::vector<PendingEq> cancelled_eq;
       {
         std::unique_lock<std::mutex> l(mu_);
+        // Purge waiters whose step was cancelled (Run timeout / abort):
+        // they must not be completed later against a dead step, and must not
+        // consume items meant for live steps.
+        for (auto it = pending_dequeue_.begin();
+             it != pending_dequeue_.end();) {
+          if (it->ctx->is_cancelled && it->ctx->is_cancelled()) {
+            cancelled_dq.push_back(std::move(*it));
+            it = pending_dequeue_.erase(it);
+          } else {
+            ++it;
+          }
+        }
+        for (auto it = pending_enqueue_.begin();
+             it != pending_enqueue_.end();) {
+          if (it->ctx->is_cancelled && it->ctx->is_cancelled()) {
+            cancelled_eq.push_back(std::move(*it));
+            it = pending_enqueue_.erase(it);
+          } else {
+            ++it;
+          }
+        }
         // admit pending enqueues
         while (!pending_enqueue_.empty() &&
                (int64_t)items_.size() < capacity_ && !closed_) {
@@ -125,7 +148,17 @@ class QueueResource : public ResourceBase {
           }
         }
       }
-      if (fire.empty() && oor.empty() && eq_fire.empty()) return;
+      if (fire.empty() && oor.empty() && eq_fire.empty() &&
+          cancelled_dq.empty() && cancelled_eq.empty())
+        return;
+      for (auto& d : cancelled_dq) {
+        d.ctx->SetStatus(errors::Cancelled("dequeue cancelled (step aborted)"));
+        d.done();
+      }
+      for (auto& e : cancelled_eq) {
+        e.ctx->SetStatus(errors::Cancelled("enqueue cancelled (step aborted)"));
+        e.done();
+      }
       for (auto& e : eq_fire) e.done();
       for (auto& d : oor) {
         d.ctx->SetStatus(errors::OutOfRange(

@@ -4,6 +4,7 @@
 #include <pybind11/numpy.h>
 #include <pybind11/complex.h>
 
+#include <algorithm>
 #include <complex>
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
@@ -407,9 +408,23 @@ class CpuCommSyncOp : public OpKernel {
 REGISTER_KERNEL_BUILDER(Name("RcclCommSync").Device(DEVICE_CPU),
                         CpuCommSyncOp);
 
+class PySession;
+namespace {
+std::mutex g_sessions_mu;
+std::vector<PySession*> g_sessions;  // live sessions, for Session.reset
+}  // namespace
+
 class PySession {
  public:
-  explicit PySession(bool cpu_only) : sess_(cpu_only) {}
+  explicit PySession(bool cpu_only) : sess_(cpu_only) {
+    std::lock_guard<std::mutex> l(g_sessions_mu);
+    g_sessions.push_back(this);
+  }
+  ~PySession() {
+    std::lock_guard<std::mutex> l(g_sessions_mu);
+    g_sessions.erase(std::remove(g_sessions.begin(), g_sessions.end(), this),
+                     g_sessions.end());
+  }
 
   void Create(py::bytes graph_def) {
     GraphDef gd;
@@ -438,8 +453,11 @@ class PySession {
     return out;
   }
 
+  void Reset() { sess_.Reset(); }
+
   py::list Run(py::dict feeds, std::vector<std::string> fetches,
-               std::vector<std::string> targets, bool collect_stats = false) {
+               std::vector<std::string> targets, bool collect_stats = false,
+               int64_t timeout_ms = 0) {
     std::vector<std::pair<std::string, Tensor>> feed_vec;
     for (auto item : feeds) {
       std::string name = py::cast<std::string>(item.first);
@@ -458,7 +476,7 @@ class PySession {
     {
       py::gil_scoped_release release;
       s = sess_.Run(feed_vec, fetches, targets, &outputs,
-                    collect_stats ? &stats : nullptr);
+                    collect_stats ? &stats : nullptr, timeout_ms);
     }
     if (!s.ok()) throw std::runtime_error(s.ToString());
     if (collect_stats) last_stats_ = std::move(stats.stats);
@@ -553,7 +571,16 @@ PYBIND11_MODULE(_core, m) {
       .def("create", &PySession::Create)
       .def("extend", &PySession::Extend)
       .def("run", &PySession::Run, py::arg("feeds"), py::arg("fetches"),
-           py::arg("targets"), py::arg("collect_stats") = false)
+           py::arg("targets"), py::arg("collect_stats") = false,
+           py::arg("timeout_ms") = 0)
+      .def("reset", &PySession::Reset)
+      .def_static("reset_all", []() {
+        // Session.reset analog for in-process sessions: reset the stateful
+        // containers of EVERY live session (reference TF_Reset clears the
+        // target's session registry).
+        std::lock_guard<std::mutex> l(g_sessions_mu);
+        for (PySession* s : g_sessions) s->Reset();
+      })
       .def("last_stats", &PySession::LastStats)
       .def("sync", &PySession::Sync)
       .def("num_gpus", &PySession::NumGpus);

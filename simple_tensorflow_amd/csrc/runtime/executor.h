@@ -44,6 +44,13 @@ class OpSegment {
     return Status::OK();
   }
 
+  // Drops every cached stateful kernel (variables, queues' op instances) —
+  // Session.reset support.
+  void Clear() {
+    std::lock_guard<std::mutex> l(mu_);
+    kernels_.clear();
+  }
+
  private:
   std::mutex mu_;
   std::map<std::string, std::unique_ptr<OpKernel>> kernels_;

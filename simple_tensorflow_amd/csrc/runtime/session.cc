@@ -3,6 +3,8 @@
 #include "gpu/gpu_tracer.h"
 
 #include <algorithm>
+#include <atomic>
+#include <chrono>
 #include <condition_variable>
 #include <set>
 #include <thread>
@@ -454,17 +456,29 @@ Status DirectSession::BuildExecutors(const std::vector<std::string>& feeds,
   return Status::OK();
 }
 
+void DirectSession::Reset() {
+  std::lock_guard<std::mutex> l(mu_);
+  executors_.clear();
+  opseg_.Clear();
+  DeleteResourceMgr(resource_mgr_);
+  resource_mgr_ = NewResourceMgr();
+}
+
 Status DirectSession::Run(
     const std::vector<std::pair<std::string, Tensor>>& feeds,
     const std::vector<std::string>& fetches,
     const std::vector<std::string>& targets, std::vector<Tensor>* outputs,
-    StatsCollector* stats) {
+    StatsCollector* stats, int64_t timeout_ms) {
   std::vector<std::string> feed_names;
   for (auto& f : feeds) feed_names.push_back(f.first);
   ExecutorsAndKeys* ek = nullptr;
   STF_RETURN_IF_ERROR(GetOrCreateExecutors(feed_names, fetches, targets, &ek));
 
-  Rendezvous rendez;
+  // Per-step rendezvous on the heap: a timed-out Run returns while stale
+  // async work (a cancelled dequeue's continuation) may still complete an
+  // executor later; the executor's completion lambda keeps this alive.
+  auto rendez_sp = std::make_shared<Rendezvous>();
+  Rendezvous& rendez = *rendez_sp;
   // Send feeds.
   for (auto& f : feeds) {
     auto [name, port] = ParseTensorName(f.first);
@@ -523,10 +537,51 @@ Status DirectSession::Run(
     std::lock_guard<std::mutex> l(mu_);
     step_id = ++step_counter_;
   }
-  std::mutex mu;
-  std::condition_variable cv;
-  int remaining = (int)ek->items.size();
-  Status agg;
+  // Completion state lives on the heap: a timed-out Run returns while
+  // executor callbacks may still fire (e.g. a dequeue blocked forever), and
+  // they must not touch a dead stack frame.
+  struct RunState {
+    std::mutex mu;
+    std::condition_variable cv;
+    int remaining = 0;
+    Status agg;
+    std::atomic<bool> done{false};
+    std::atomic<bool> cancelled{false};
+  };
+  auto rs = std::make_shared<RunState>();
+  rs->remaining = (int)ek->items.size();
+  std::mutex& mu = rs->mu;
+  std::condition_variable& cv = rs->cv;
+  int& remaining = rs->remaining;
+  Status& agg = rs->agg;
+  // Deadline watchdog (RunOptions.timeout_in_ms / ConfigProto
+  // operation_timeout_in_ms analog): aborts the step's rendezvous, which
+  // unblocks the client fetch Recv and every executor recv.
+  std::shared_ptr<std::thread> watchdog;
+  if (timeout_ms > 0) {
+    watchdog = std::make_shared<std::thread>([rs, rendez_sp, timeout_ms]() {
+      std::unique_lock<std::mutex> l(rs->mu);
+      if (!rs->cv.wait_for(l, std::chrono::milliseconds(timeout_ms),
+                           [&]() { return rs->done.load(); })) {
+        rs->cancelled = true;
+        l.unlock();
+        rendez_sp->StartAbort(
+            errors::DeadlineExceeded("Session::Run timed out after ",
+                                     timeout_ms, " ms"));
+      }
+    });
+  }
+  auto finish_watchdog = [&]() {
+    if (watchdog) {
+      {
+        std::lock_guard<std::mutex> l(rs->mu);
+        rs->done = true;
+      }
+      rs->cv.notify_all();
+      watchdog->join();
+      watchdog.reset();
+    }
+  };
   // During hipGraph capture every enqueue must come from this thread: use a
   // single-threaded trampoline queue instead of the pool.
   std::deque<std::function<void()>> inline_q;
@@ -546,14 +601,15 @@ Status DirectSession::Run(
   for (auto& item : ek->items) {
     ExecutorArgs args;
     args.step_id = step_id;
-    args.rendezvous = &rendez;
+    args.rendezvous = rendez_sp.get();
+    args.is_cancelled = [rs]() { return rs->cancelled.load(); };
     args.schedule = scheduler;
     args.resource_mgr = resource_mgr_;
     args.stats = stats;
-    item.executor->RunAsync(args, [&](Status s) {
-      std::lock_guard<std::mutex> l(mu);
-      if (!s.ok() && agg.ok()) agg = s;
-      if (--remaining == 0) cv.notify_one();
+    item.executor->RunAsync(args, [rs, rendez_sp](Status s) {
+      std::lock_guard<std::mutex> l(rs->mu);
+      if (!s.ok() && rs->agg.ok()) rs->agg = s;
+      if (--rs->remaining == 0) rs->cv.notify_all();
     });
   }
   if (do_capture) {
@@ -614,8 +670,23 @@ Status DirectSession::Run(
 
   {
     std::unique_lock<std::mutex> l(mu);
-    cv.wait(l, [&]() { return remaining == 0; });
+    if (timeout_ms > 0) {
+      // Executors unblocked by the abort finish almost immediately; only a
+      // kernel stuck outside the rendezvous (queue waiter) stays pending —
+      // don't hold the client for it.
+      if (!cv.wait_for(l, std::chrono::milliseconds(500),
+                       [&]() { return remaining == 0; })) {
+        l.unlock();
+        finish_watchdog();
+        return errors::DeadlineExceeded(
+            "Session::Run timed out after ", timeout_ms,
+            " ms (executors still pending)");
+      }
+    } else {
+      cv.wait(l, [&]() { return remaining == 0; });
+    }
   }
+  finish_watchdog();
   if (do_capture && capture_dev) {
     if (agg.ok()) {
       void* exec = nullptr;

@@ -33,7 +33,12 @@ class DirectSession {
   Status Run(const std::vector<std::pair<std::string, Tensor>>& feeds,
              const std::vector<std::string>& fetches,
              const std::vector<std::string>& targets,
-             std::vector<Tensor>* outputs, StatsCollector* stats = nullptr);
+             std::vector<Tensor>* outputs, StatsCollector* stats = nullptr,
+             int64_t timeout_ms = 0);
+
+  // Session.reset analog: drops all stateful kernels (variables, queues,
+  // tables) and cached executors; the graph itself is retained.
+  void Reset();
 
   DeviceMgr* device_mgr() { return &devices_; }
   // Blocks until all device work is complete (bench timing bracket).

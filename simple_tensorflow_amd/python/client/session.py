@@ -42,7 +42,18 @@ class Session(object):
             self._core = _core.Session(cpu_only)
         self._created = False
         self._serialized_nodes = 0
+        self._operation_timeout_ms = 0
+        if config is not None and isinstance(config, dict):
+            self._operation_timeout_ms = int(
+                config.get('operation_timeout_in_ms', 0))
         self._lock = threading.Lock()
+
+    @staticmethod
+    def reset(target='', containers=None, config=None):
+        """Resets resource containers (reference Session.reset / TF_Reset):
+        drops every stateful kernel (variables, queues, lookup tables) of all
+        live in-process sessions; they see fresh uninitialized state."""
+        _core.Session.reset_all()
 
     @property
     def graph(self):
@@ -96,9 +107,13 @@ class Session(object):
         want_stats = (options is not None and
                       getattr(options, 'trace_level', 0) and
                       run_metadata is not None)
+        timeout_ms = int(getattr(options, 'timeout_in_ms', 0) or 0)
+        if not timeout_ms:
+            timeout_ms = self._operation_timeout_ms
         try:
-            if want_stats:
-                results = self._core.run(feeds, fetch_names, targets, True)
+            if want_stats or timeout_ms:
+                results = self._core.run(feeds, fetch_names, targets,
+                                         want_stats, timeout_ms)
             else:
                 results = self._core.run(feeds, fetch_names, targets)
         except RuntimeError as e:

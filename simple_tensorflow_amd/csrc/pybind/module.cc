@@ -455,6 +455,35 @@ class PySession {
 
   void Reset() { sess_.Reset(); }
 
+  std::string PartialRunSetup(std::vector<std::string> feeds,
+                              std::vector<std::string> fetches,
+                              std::vector<std::string> targets) {
+    std::string handle;
+    Status s = sess_.PartialRunSetup(feeds, fetches, targets, &handle);
+    if (!s.ok()) throw std::runtime_error(s.ToString());
+    return handle;
+  }
+
+  py::list PartialRun(const std::string& handle, py::dict feeds,
+                      std::vector<std::string> fetches) {
+    std::vector<std::pair<std::string, Tensor>> feed_vec;
+    for (auto item : feeds) {
+      std::string name = py::cast<std::string>(item.first);
+      py::object val = py::reinterpret_borrow<py::object>(item.second);
+      feed_vec.emplace_back(name, NumpyToTensor(py::cast<py::array>(val)));
+    }
+    std::vector<Tensor> outputs;
+    Status s;
+    {
+      py::gil_scoped_release release;
+      s = sess_.PartialRun(handle, feed_vec, fetches, &outputs);
+    }
+    if (!s.ok()) throw std::runtime_error(s.ToString());
+    py::list out;
+    for (auto& t : outputs) out.append(TensorToPy(t));
+    return out;
+  }
+
   py::list Run(py::dict feeds, std::vector<std::string> fetches,
                std::vector<std::string> targets, bool collect_stats = false,
                int64_t timeout_ms = 0) {
@@ -574,6 +603,8 @@ PYBIND11_MODULE(_core, m) {
            py::arg("targets"), py::arg("collect_stats") = false,
            py::arg("timeout_ms") = 0)
       .def("reset", &PySession::Reset)
+      .def("partial_run_setup", &PySession::PartialRunSetup)
+      .def("partial_run", &PySession::PartialRun)
       .def_static("reset_all", []() {
         // Session.reset analog for in-process sessions: reset the stateful
         // containers of EVERY live session (reference TF_Reset clears the

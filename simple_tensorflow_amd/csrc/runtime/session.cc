@@ -464,6 +464,155 @@ void DirectSession::Reset() {
   resource_mgr_ = NewResourceMgr();
 }
 
+
+// ---------------------------------------------------------------------------
+// Partial run (reference direct_session.cc PRunSetup/PRun): the executors
+// for the full feed/fetch union start immediately; _Recv nodes wait
+// asynchronously on the step rendezvous, so feeds can arrive across several
+// PartialRun calls and fetches are pulled as their subgraphs complete.
+// ---------------------------------------------------------------------------
+struct DirectSession::PartialRunState {
+  std::shared_ptr<Rendezvous> rendez;
+  ExecutorsAndKeys* ek = nullptr;
+  std::mutex mu;
+  std::condition_variable cv;
+  int remaining = 0;
+  Status agg;
+  std::set<std::string> pending_fetches;  // "name:port" not yet returned
+  std::set<std::string> pending_feeds;
+};
+
+Status DirectSession::PartialRunSetup(const std::vector<std::string>& feeds,
+                                      const std::vector<std::string>& fetches,
+                                      const std::vector<std::string>& targets,
+                                      std::string* handle) {
+  ExecutorsAndKeys* ek = nullptr;
+  STF_RETURN_IF_ERROR(GetOrCreateExecutors(feeds, fetches, targets, &ek));
+  auto prs = std::make_shared<PartialRunState>();
+  prs->rendez = std::make_shared<Rendezvous>();
+  prs->ek = ek;
+  prs->remaining = (int)ek->items.size();
+  for (auto& f : feeds) {
+    auto [name, port] = ParseTensorName(f);
+    prs->pending_feeds.insert(name + ":" + std::to_string(port));
+  }
+  for (auto& f : fetches) {
+    auto [name, port] = ParseTensorName(f);
+    prs->pending_fetches.insert(name + ":" + std::to_string(port));
+  }
+  int64_t step_id;
+  {
+    std::lock_guard<std::mutex> l(mu_);
+    step_id = ++step_counter_;
+    *handle = "prun_" + std::to_string(++partial_run_counter_);
+    partial_runs_[*handle] = prs;
+  }
+  ThreadPool* pool = pool_.get();
+  for (auto& item : ek->items) {
+    ExecutorArgs args;
+    args.step_id = step_id;
+    args.rendezvous = prs->rendez.get();
+    args.schedule = [pool](std::function<void()> fn) {
+      pool->Schedule(std::move(fn));
+    };
+    args.resource_mgr = resource_mgr_;
+    item.executor->RunAsync(args, [prs](Status s) {
+      std::lock_guard<std::mutex> l(prs->mu);
+      if (!s.ok() && prs->agg.ok()) prs->agg = s;
+      if (--prs->remaining == 0) prs->cv.notify_all();
+    });
+  }
+  return Status::OK();
+}
+
+Status DirectSession::PartialRun(
+    const std::string& handle,
+    const std::vector<std::pair<std::string, Tensor>>& feeds,
+    const std::vector<std::string>& fetches, std::vector<Tensor>* outputs) {
+  std::shared_ptr<PartialRunState> prs;
+  {
+    std::lock_guard<std::mutex> l(mu_);
+    auto it = partial_runs_.find(handle);
+    if (it == partial_runs_.end())
+      return errors::InvalidArgument("Unknown partial-run handle ", handle);
+    prs = it->second;
+  }
+  for (auto& f : feeds) {
+    auto [name, port] = ParseTensorName(f.first);
+    std::string fkey = name + ":" + std::to_string(port);
+    {
+      std::lock_guard<std::mutex> l(prs->mu);
+      if (!prs->pending_feeds.erase(fkey))
+        return errors::InvalidArgument(
+            "Feed ", f.first,
+            " was not declared in partial_run_setup (or already fed)");
+    }
+    auto it = prs->ek->feed_devices.find("feed:" + fkey);
+    if (it == prs->ek->feed_devices.end())
+      return errors::Internal("feed not wired: ", f.first);
+    STF_RETURN_IF_ERROR(prs->rendez->Send(
+        RendezvousKey("client", it->second, "feed:" + fkey, "", 0), f.second,
+        false));
+  }
+  outputs->clear();
+  Status fetch_status;
+  for (auto& f : fetches) {
+    auto [name, port] = ParseTensorName(f);
+    std::string fkey = name + ":" + std::to_string(port);
+    {
+      std::lock_guard<std::mutex> l(prs->mu);
+      if (!prs->pending_fetches.count(fkey))
+        return errors::InvalidArgument(
+            "Fetch ", f, " was not declared in partial_run_setup (or was "
+            "already fetched)");
+    }
+    auto it = prs->ek->fetch_devices.find("fetch:" + fkey);
+    if (it == prs->ek->fetch_devices.end()) {
+      fetch_status = errors::Internal("fetch not wired: ", f);
+      break;
+    }
+    Tensor val;
+    bool is_dead = false;
+    Status s = prs->rendez->Recv(
+        RendezvousKey(it->second, "client", "fetch:" + fkey, "", 0), &val,
+        &is_dead);
+    if (!s.ok()) {
+      fetch_status = s;
+      break;
+    }
+    if (val.IsInitialized() && val.mem_space() == MemSpace::DEVICE) {
+      Device* dev = devices_.LookUp(it->second);
+      Tensor host;
+      Status cs = dev->CopyDeviceTensorToHost(val, &host);
+      if (!cs.ok()) {
+        fetch_status = cs;
+        break;
+      }
+      val = host;
+    }
+    outputs->push_back(val);
+    std::lock_guard<std::mutex> l(prs->mu);
+    prs->pending_fetches.erase(fkey);
+  }
+  bool finished;
+  {
+    std::lock_guard<std::mutex> l(prs->mu);
+    finished = prs->pending_fetches.empty();
+  }
+  if (!fetch_status.ok() || finished) {
+    // tear down: abort anything still pending, wait for executors
+    prs->rendez->StartAbort(errors::Cancelled("partial run finished"));
+    {
+      std::unique_lock<std::mutex> l(prs->mu);
+      prs->cv.wait_for(l, std::chrono::seconds(2),
+                       [&]() { return prs->remaining == 0; });
+    }
+    std::lock_guard<std::mutex> l(mu_);
+    partial_runs_.erase(handle);
+  }
+  return fetch_status;
+}
+
 Status DirectSession::Run(
     const std::vector<std::pair<std::string, Tensor>>& feeds,
     const std::vector<std::string>& fetches,

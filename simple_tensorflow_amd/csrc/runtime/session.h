@@ -40,6 +40,18 @@ class DirectSession {
   // tables) and cached executors; the graph itself is retained.
   void Reset();
 
+  // Partial-run support (reference DirectSession::PRunSetup/PRun): launch
+  // the executors for the union of feeds/fetches now, feed and fetch
+  // incrementally across multiple PartialRun calls.
+  Status PartialRunSetup(const std::vector<std::string>& feeds,
+                         const std::vector<std::string>& fetches,
+                         const std::vector<std::string>& targets,
+                         std::string* handle);
+  Status PartialRun(const std::string& handle,
+                    const std::vector<std::pair<std::string, Tensor>>& feeds,
+                    const std::vector<std::string>& fetches,
+                    std::vector<Tensor>* outputs);
+
   DeviceMgr* device_mgr() { return &devices_; }
   // Blocks until all device work is complete (bench timing bracket).
   Status SyncAllDevices() {
@@ -49,6 +61,7 @@ class DirectSession {
   void* resource_mgr() { return resource_mgr_; }
 
  private:
+  struct PartialRunState;
   struct ExecutorsAndKeys {
     struct Item {
       Device* device;
@@ -84,6 +97,8 @@ class DirectSession {
   OpSegment opseg_;
   std::unique_ptr<ThreadPool> pool_;
   std::map<std::string, std::unique_ptr<ExecutorsAndKeys>> executors_;
+  std::map<std::string, std::shared_ptr<PartialRunState>> partial_runs_;
+  int64_t partial_run_counter_ = 0;
   int64_t step_counter_ = 0;
   void* resource_mgr_ = nullptr;  // owned; see kernels/resource_mgr
 };

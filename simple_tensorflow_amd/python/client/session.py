@@ -48,6 +48,37 @@ class Session(object):
                 config.get('operation_timeout_in_ms', 0))
         self._lock = threading.Lock()
 
+    def partial_run_setup(self, fetches, feeds=None):
+        """Declare the fetches/feeds of an incremental run (reference
+        Session.partial_run_setup): executors start now; partial_run() feeds
+        and fetches subsets until every declared fetch is consumed."""
+        self._sync_graph()
+        if not isinstance(fetches, (list, tuple)):
+            fetches = [fetches]
+        feeds = feeds or []
+        if not isinstance(feeds, (list, tuple)):
+            feeds = [feeds]
+        fetch_names = [_as_fetchable(f, self._graph).name for f in fetches]
+        feed_names = [_as_fetchable(f, self._graph).name for f in feeds]
+        return self._core.partial_run_setup(feed_names, fetch_names, [])
+
+    def partial_run(self, handle, fetches, feed_dict=None):
+        single = not isinstance(fetches, (list, tuple))
+        if single:
+            fetches = [fetches]
+        fetch_names = [_as_fetchable(f, self._graph).name for f in fetches]
+        feeds = {}
+        if feed_dict:
+            for k, v in feed_dict.items():
+                t = _as_fetchable(k, self._graph)
+                feeds[t.name] = _convert_feed(t, v)
+        try:
+            results = self._core.partial_run(handle, feeds, fetch_names)
+        except RuntimeError as e:
+            errors.raise_from_message(str(e))
+        out = list(results)
+        return out[0] if single else out
+
     @staticmethod
     def reset(target='', containers=None, config=None):
         """Resets resource containers (reference Session.reset / TF_Reset):

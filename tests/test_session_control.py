@@ -63,3 +63,41 @@ def test_session_reset_clears_containers():
         # same session object sees fresh (empty) stateful containers
         assert s.run(q.size()) == 0
         assert s.run(table.size()) == 0
+
+
+def test_partial_run_incremental():
+    a = tf.placeholder(tf.float32, [2])
+    b = tf.placeholder(tf.float32, [2])
+    mid = a + tf.constant([1.0, 1.0])
+    out = mid * b
+    with tf.Session() as s:
+        h = s.partial_run_setup([mid, out], [a, b])
+        v1 = s.partial_run(h, mid, feed_dict={a: np.array([1., 2.],
+                                                          np.float32)})
+        np.testing.assert_allclose(v1, [2., 3.])
+        # mid's subgraph ran once; out builds on it with the late feed of b
+        v2 = s.partial_run(h, out, feed_dict={b: np.array([10., 20.],
+                                                          np.float32)})
+        np.testing.assert_allclose(v2, [20., 60.])
+
+
+def test_partial_run_undeclared_feed_rejected():
+    a = tf.placeholder(tf.float32, [1])
+    b = tf.placeholder(tf.float32, [1])
+    out = a * 2.0
+    out2 = b * 3.0
+    with tf.Session() as s:
+        h = s.partial_run_setup([out], [a])
+        with pytest.raises(Exception, match='not declared'):
+            s.partial_run(h, out, feed_dict={b: np.array([1.], np.float32)})
+
+
+def test_partial_run_handle_consumed():
+    a = tf.placeholder(tf.float32, [1])
+    out = a + 1.0
+    with tf.Session() as s:
+        h = s.partial_run_setup([out], [a])
+        s.partial_run(h, out, feed_dict={a: np.array([1.], np.float32)})
+        # handle is spent once every declared fetch was returned
+        with pytest.raises(Exception, match='handle'):
+            s.partial_run(h, out, feed_dict={a: np.array([2.], np.float32)})

@@ -336,6 +336,13 @@ VariableScope = _vs.VariableScope
 nn.rnn_cell = _rnn
 nn.dynamic_rnn = _rnn.dynamic_rnn
 nn.static_rnn = _rnn.static_rnn
+from simple_tensorflow_amd.python.ops import candidate_sampling_ops as _cs  # noqa: E402
+nn.sampled_softmax_loss = _cs.sampled_softmax_loss
+nn.nce_loss = _cs.nce_loss
+nn.uniform_candidate_sampler = _cs.uniform_candidate_sampler
+nn.log_uniform_candidate_sampler = _cs.log_uniform_candidate_sampler
+nn.learned_unigram_candidate_sampler = _cs.learned_unigram_candidate_sampler
+nn.compute_accidental_hits = _cs.compute_accidental_hits
 nn.rnn = _rnn.static_rnn
 
 

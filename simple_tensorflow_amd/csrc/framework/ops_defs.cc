@@ -167,6 +167,16 @@ REGISTER_OP("LSTMGatesGrad").Input("c_prev: T").Input("i: T").Input("f: T").Inpu
 REGISTER_OP("LRN").Input("input: T").Output("output: T").Attr("depth_radius: int = 5").Attr("bias: float = 1.0").Attr("alpha: float = 1.0").Attr("beta: float = 0.5").Attr("T: {float}");
 REGISTER_OP("InTopK").Input("predictions: float").Input("targets: T").Output("precision: bool").Attr("k: int").Attr("T: {int32, int64} = int32");
 
+// ------------------------- candidate sampling ------------------------------
+// Reference core/ops/candidate_sampling_ops.cc; kernels in
+// kernels/cpu_sampling.cc. true_classes is [batch, num_true] int64.
+#define SAMPLER_OP(NAME) REGISTER_OP(NAME).Input("true_classes: int64").Output("sampled_candidates: int64").Output("true_expected_count: float").Output("sampled_expected_count: float").Attr("num_true: int >= 1").Attr("num_sampled: int >= 1").Attr("unique: bool").Attr("range_max: int >= 1").Attr("seed: int = 0").Attr("seed2: int = 0").SetIsStateful()
+SAMPLER_OP("UniformCandidateSampler");
+SAMPLER_OP("LogUniformCandidateSampler");
+SAMPLER_OP("LearnedUnigramCandidateSampler");
+#undef SAMPLER_OP
+REGISTER_OP("ComputeAccidentalHits").Input("true_classes: int64").Input("sampled_candidates: int64").Output("indices: int32").Output("ids: int64").Output("weights: float").Attr("num_true: int").Attr("seed: int = 0").Attr("seed2: int = 0");
+
 // --------------------------------- CTC -------------------------------------
 // Reference core/ops/ctc_ops.cc / kernels/ctc_loss_op.cc. Kernel in
 // kernels/cpu_ctc.cc (log-space forward-backward).

@@ -526,10 +526,13 @@ def constant(value, dtype=None, shape=None, name='Const'):
     else:
         np_dt = _np_for_dtype(dtype) if dtype is not None else None
         arr = np.array(value, dtype=np_dt)
-        if arr.dtype == np.float64 and dtype is None:
-            arr = arr.astype(np.float32)
-        if arr.dtype == np.int64 and dtype is None:
-            arr = arr.astype(np.int32)
+        # python floats/ints default to f32/i32 (TF convention); an explicit
+        # numpy array keeps its dtype.
+        if not isinstance(value, np.ndarray):
+            if arr.dtype == np.float64 and dtype is None:
+                arr = arr.astype(np.float32)
+            if arr.dtype == np.int64 and dtype is None:
+                arr = arr.astype(np.int32)
         if shape is not None:
             arr = np.broadcast_to(arr, shape).astype(arr.dtype)
         enum = dtypes.as_dtype(arr.dtype).as_datatype_enum

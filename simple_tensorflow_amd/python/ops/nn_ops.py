@@ -279,3 +279,13 @@ def _ctc_loss_grad(op, grad_loss, _):
 
 
 ops.NoGradient('CTCGreedyDecoder')
+
+
+def sigmoid_cross_entropy_with_logits(labels=None, logits=None, name=None):
+    """Elementwise sigmoid cross-entropy (reference nn_impl.py:112
+    formulation, numerically stable): max(x,0) - x*z + log1p(exp(-|x|))."""
+    x = convert_to_tensor(logits)
+    z = convert_to_tensor(labels, dtype=x.dtype)
+    return math_ops.add(
+        math_ops.maximum(x, array_ops.zeros_like(x)) - x * z,
+        math_ops.log1p(math_ops.exp(-math_ops.abs(x))), name=name)

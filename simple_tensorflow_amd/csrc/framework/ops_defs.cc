@@ -283,7 +283,13 @@ REGISTER_OP("MergeV2Checkpoints").Input("checkpoint_prefixes: string").Input("de
 // handles are session-scoped strings here, like the queue ops above).
 REGISTER_OP("PlaceholderWithDefault").Input("input: dtype").Output("output: dtype").Attr("dtype: type").Attr("shape: shape = []");
 REGISTER_OP("SparseToDense").Input("sparse_indices: Tindices").Input("output_shape: Tindices").Input("sparse_values: T").Input("default_value: T").Output("dense: T").Attr("validate_indices: bool = true").Attr("T: type").Attr("Tindices: {int32, int64}");
-REGISTER_OP("ResizeBilinear").Input("images: T").Input("size: int32").Output("resized_images: float").Attr("T: {float}").Attr("align_corners: bool = false");
+
+// sparse algebra (reference core/ops/sparse_ops.cc; kernels/cpu_sparse.cc)
+REGISTER_OP("SparseAdd").Input("a_indices: int64").Input("a_values: T").Input("a_shape: int64").Input("b_indices: int64").Input("b_values: T").Input("b_shape: int64").Input("thresh: Treal").Output("sum_indices: int64").Output("sum_values: T").Output("sum_shape: int64").Attr("T: " NUMTYPES).Attr("Treal: {float, double, int32, int64}");
+REGISTER_OP("SparseTensorDenseAdd").Input("a_indices: Tindices").Input("a_values: T").Input("a_shape: Tindices").Input("b: T").Output("output: T").Attr("T: " NUMTYPES).Attr("Tindices: {int32, int64}");
+REGISTER_OP("SparseReorder").Input("input_indices: int64").Input("input_values: T").Input("input_shape: int64").Output("output_indices: int64").Output("output_values: T").Attr("T: type");
+REGISTER_OP("SparseReduceSum").Input("input_indices: int64").Input("input_values: T").Input("input_shape: int64").Input("reduction_axes: int32").Output("output: T").Attr("keep_dims: bool = false").Attr("T: " NUMTYPES);
+REGISTER_OP("SparseConcat").Input("indices: N * int64").Input("values: N * T").Input("shapes: N * int64").Output("output_indices: int64").Output("output_values: T").Output("output_shape: int64").Attr("concat_dim: int").Attr("N: int >= 2").Attr("T: type");REGISTER_OP("ResizeBilinear").Input("images: T").Input("size: int32").Output("resized_images: float").Attr("T: {float}").Attr("align_corners: bool = false");
 REGISTER_OP("ResizeBilinearGrad").Input("grads: float").Input("original_image: T").Output("output: T").Attr("T: {float}").Attr("align_corners: bool = false");
 REGISTER_OP("ResizeNearestNeighbor").Input("images: T").Input("size: int32").Output("resized_images: T").Attr("T: {float}").Attr("align_corners: bool = false");
 REGISTER_OP("ConditionalAccumulator").Output("handle: Ref(string)").Attr("dtype: type").Attr("shape: shape").Attr("container: string = ''").Attr("shared_name: string = ''").SetIsStateful();

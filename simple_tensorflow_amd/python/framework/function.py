@@ -49,6 +49,7 @@ class _DefinedFunction(object):
             raise ValueError('%s expects %d arguments, got %d' %
                              (self.name, len(self._input_types), len(args)))
         g = ops.get_default_graph()
+        g._add_function(self)  # serialized into the GraphDef library
         tensors = [ops.convert_to_tensor(a, dtype=dt)
                    for a, dt in zip(args, self._input_types)]
         mark = len(g._node_list)
@@ -67,6 +68,53 @@ class _DefinedFunction(object):
             for t in outs:
                 t._defun_record = rec
         return outs[0] if single else tuple(outs)
+
+
+    @property
+    def definition(self):
+        """Serialized FunctionDef (reference function.proto wire format):
+        the body is traced into a private graph with `_arg_i` placeholders;
+        ret maps `out_i` to the producing tensor names."""
+        from simple_tensorflow_amd.python.framework import dtypes, pbwire
+        g = ops.Graph()
+        with g.as_default():
+            from simple_tensorflow_amd.python.ops import array_ops
+            args = [array_ops.placeholder(dt, name='_arg_%d' % i)
+                    for i, dt in enumerate(self._input_types)]
+            outs = self._func(*args)
+        single = not isinstance(outs, (list, tuple))
+        outs = [outs] if single else list(outs)
+        node_bytes = [op.node_def_bytes() for op in g._node_list
+                      if not op.name.startswith('_arg_')]
+        in_args = [('_arg_%d' % i, dt.as_datatype_enum)
+                   for i, dt in enumerate(self._input_types)]
+        out_args = [('out_%d' % i, o.dtype.as_datatype_enum)
+                    for i, o in enumerate(outs)]
+        sig = pbwire.op_def_signature(self.name, in_args, out_args)
+        ret = {'out_%d' % i: o.name for i, o in enumerate(outs)}
+        return pbwire.function_def(sig, node_bytes, ret)
+
+
+def from_function_def(fdef_bytes):
+    """Parses a wire-format FunctionDef back into a callable graph function
+    (instantiated inline at each call, like @Defun)."""
+    from simple_tensorflow_amd.python.framework import dtypes, pbreader
+    fdef = pbreader.parse_function_def(bytes(fdef_bytes))
+    sig = fdef['signature']
+    input_types = [dtypes.as_dtype(a['type']) for a in sig['input_arg']]
+    n_out = len(sig['output_arg'])
+    ret_names = [fdef['ret']['out_%d' % i] for i in range(n_out)]
+    nodes = fdef['node_def']
+
+    def body(*args):
+        from simple_tensorflow_amd.python.framework import importer
+        input_map = {'_arg_%d' % i: a for i, a in enumerate(args)}
+        outs = importer.import_graph_def(nodes, input_map=input_map,
+                                         return_elements=ret_names, name='')
+        return outs[0] if n_out == 1 else tuple(outs)
+
+    return _DefinedFunction(body, input_types, func_name=sig['name'])
+
 
 
 class Defun(object):

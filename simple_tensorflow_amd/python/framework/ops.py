@@ -406,7 +406,16 @@ class Graph(object):
 
     def as_graph_def(self, from_version=0):
         nodes = [op.node_def_bytes() for op in self._node_list[from_version:]]
-        return pbwire.graph_def(nodes)
+        library = None
+        if from_version == 0 and getattr(self, '_functions', None):
+            library = pbwire.function_def_library(
+                [f.definition for f in self._functions.values()])
+        return pbwire.graph_def(nodes, library=library)
+
+    def _add_function(self, fn):
+        if not hasattr(self, '_functions'):
+            self._functions = {}
+        self._functions[fn.name] = fn
 
     def finalize(self):
         self._finalized = True

@@ -197,3 +197,59 @@ def np_from_tensor_proto(data):
     else:
         arr = np.zeros(n, dtype=npdt)
     return arr.reshape(dims)
+
+def parse_arg_def(data):
+    arg = {'name': '', 'type': 0}
+    for f, w, v in _fields(data):
+        if f == 1:
+            arg['name'] = v.decode()
+        elif f == 3:
+            arg['type'] = v
+    return arg
+
+
+def parse_op_def_signature(data):
+    sig = {'name': '', 'input_arg': [], 'output_arg': []}
+    for f, w, v in _fields(data):
+        if f == 1:
+            sig['name'] = v.decode()
+        elif f == 2:
+            sig['input_arg'].append(parse_arg_def(v))
+        elif f == 3:
+            sig['output_arg'].append(parse_arg_def(v))
+    return sig
+
+
+def parse_function_def(data):
+    fdef = {'signature': None, 'node_def': [], 'ret': {}}
+    for f, w, v in _fields(data):
+        if f == 1:
+            fdef['signature'] = parse_op_def_signature(v)
+        elif f == 3:
+            fdef['node_def'].append(parse_node_def(v))
+        elif f == 4:
+            k = val = None
+            for f2, w2, v2 in _fields(v):
+                if f2 == 1:
+                    k = v2.decode()
+                elif f2 == 2:
+                    val = v2.decode()
+            if k is not None:
+                fdef['ret'][k] = val
+    return fdef
+
+
+def parse_function_def_library(data):
+    return [parse_function_def(v) for f, w, v in _fields(data) if f == 1]
+
+
+def parse_graph_def_full(data):
+    """Returns (nodes, functions): the node list plus the FunctionDefLibrary
+    content (GraphDef field 2)."""
+    nodes, functions = [], []
+    for f, w, v in _fields(data):
+        if f == 1:
+            nodes.append(parse_node_def(v))
+        elif f == 2:
+            functions.extend(parse_function_def_library(v))
+    return nodes, functions

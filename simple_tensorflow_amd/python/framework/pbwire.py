@@ -136,8 +136,42 @@ def node_def(name, op, inputs, device, attrs):
     return out
 
 
-def graph_def(node_bytes_list, producer=21):
-    """GraphDef: node=1, versions=4 {producer=1}."""
+def graph_def(node_bytes_list, producer=21, library=None):
+    """GraphDef: node=1, library=2 (FunctionDefLibrary), versions=4
+    {producer=1}."""
     out = b''.join(f_bytes(1, nb) for nb in node_bytes_list)
+    if library:
+        out += f_bytes(2, library)
     out += f_bytes(4, f_varint(1, producer))
     return out
+
+def arg_def(name, type_enum):
+    """OpDef.ArgDef: name=1, type=3."""
+    return f_bytes(1, name) + f_varint(3, type_enum)
+
+
+def op_def_signature(name, input_args, output_args):
+    """OpDef subset for FunctionDef.signature: name=1, input_arg=2,
+    output_arg=3 (each ArgDef)."""
+    out = f_bytes(1, name)
+    for n, t in input_args:
+        out += f_bytes(2, arg_def(n, t))
+    for n, t in output_args:
+        out += f_bytes(3, arg_def(n, t))
+    return out
+
+
+def function_def(signature_bytes, node_bytes_list, ret_map):
+    """FunctionDef: signature=1 (OpDef), node_def=3 (NodeDef), ret=4
+    (map<string,string>)."""
+    out = f_bytes(1, signature_bytes)
+    for nb in node_bytes_list:
+        out += f_bytes(3, nb)
+    for k in sorted(ret_map):
+        out += f_bytes(4, f_bytes(1, k) + f_bytes(2, ret_map[k]))
+    return out
+
+
+def function_def_library(fdef_bytes_list):
+    """FunctionDefLibrary: function=1."""
+    return b''.join(f_bytes(1, fb) for fb in fdef_bytes_list)

@@ -168,3 +168,55 @@ def decode_png(contents, channels=0, dtype=dtypes.uint8, name=None):
     out = script_ops.py_func(_dec, [convert_to_tensor(contents)],
                              dtypes.uint8, name=name)
     return out
+
+
+def encode_jpeg(image, quality=95, name=None, **kw):
+    """Baseline JFIF encode via the python codec
+    (lib/io/jpeg_codec.py; reference EncodeJpeg / jpeg_mem.cc analog)."""
+    from simple_tensorflow_amd.python.lib.io import jpeg_codec
+    from simple_tensorflow_amd.python.ops import script_ops
+    import numpy as np
+
+    def _enc(arr):
+        return jpeg_codec.encode_jpeg(np.asarray(arr, dtype=np.uint8),
+                                      quality=quality)
+
+    return script_ops.py_func(_enc, [convert_to_tensor(image)],
+                              dtypes.string, name=name)
+
+
+def decode_jpeg(contents, channels=0, name=None, **kw):
+    from simple_tensorflow_amd.python.lib.io import jpeg_codec
+    from simple_tensorflow_amd.python.ops import script_ops
+    import numpy as np
+
+    def _dec(blob):
+        b = blob if isinstance(blob, bytes) else bytes(blob)
+        img = jpeg_codec.decode_jpeg(b)
+        if channels == 1 and img.shape[2] == 3:
+            y = (0.299 * img[..., 0] + 0.587 * img[..., 1] +
+                 0.114 * img[..., 2])
+            img = np.clip(y, 0, 255).astype(np.uint8)[:, :, None]
+        elif channels == 3 and img.shape[2] == 1:
+            img = np.repeat(img, 3, 2)
+        return img
+
+    return script_ops.py_func(_dec, [convert_to_tensor(contents)],
+                              dtypes.uint8, name=name)
+
+
+def decode_image(contents, channels=None, name=None):
+    """Dispatches on magic bytes to decode_png / decode_jpeg (reference
+    decode_image)."""
+    from simple_tensorflow_amd.python.lib.io import jpeg_codec, png_codec
+    from simple_tensorflow_amd.python.ops import script_ops
+    import numpy as np
+
+    def _dec(blob):
+        b = blob if isinstance(blob, bytes) else bytes(blob)
+        if b[:2] == b'\xff\xd8':
+            return jpeg_codec.decode_jpeg(b)
+        return png_codec.decode_png(b)
+
+    return script_ops.py_func(_dec, [convert_to_tensor(contents)],
+                              dtypes.uint8, name=name)

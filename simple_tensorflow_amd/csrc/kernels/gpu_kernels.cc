@@ -84,6 +84,9 @@ hipError_t stf_bn_bwd(int, const void*, const void*, const void*,
                       float*, void*, int64_t, int, int, hipStream_t);
 hipError_t stf_pool_fwd(int, int, const void*, void*, int, int, int, int, int,
                         int, int, int, int, int, int, int, hipStream_t);
+hipError_t stf_max_pool_bwd_v8(const void*, const void*, void*, int, int,
+                               int, int, int, int, int, int, int, int, int,
+                               int, hipStream_t);
 hipError_t stf_max_pool_bwd(int, const void*, const void*, float*, int, int,
                             int, int, int, int, int, int, int, int, int, int,
                             hipStream_t);
@@ -1373,6 +1376,17 @@ class GpuMaxPoolGradOp : public OpKernel {
     PoolDims(x.shape(), p_, &P, &Q, &ph, &pw);
     Tensor* dx = ctx->allocate_output(0, x.shape());
     hipStream_t s = GPU_STREAM(ctx);
+    if (x.dtype() == DT_BFLOAT16 && x.dim_size(3) % 8 == 0) {
+      // per-input direct path: no f32 scratch, no zero fill, no cast
+      OP_HIP_OK(ctx, stf_max_pool_bwd_v8(
+                         x.raw_data(), dy.raw_data(), dx->raw_data(),
+                         (int)x.dim_size(0), (int)x.dim_size(1),
+                         (int)x.dim_size(2), (int)x.dim_size(3),
+                         (int)p_.ksize[1], (int)p_.ksize[2],
+                         (int)p_.strides[1], (int)p_.strides[2], (int)ph,
+                         (int)pw, (int)P, (int)Q, s));
+      return;
+    }
     Tensor scratch = ctx->allocate_temp(DT_FLOAT, x.shape());
     OP_HIP_OK(ctx, ZeroF32(scratch.raw_data(), x.NumElements(), s));
     OP_HIP_OK(ctx, stf_max_pool_bwd(DtypeCode(x.dtype()), x.raw_data(),

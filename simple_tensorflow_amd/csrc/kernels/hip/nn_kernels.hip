@@ -341,6 +341,241 @@ __global__ void BnNormAddReluKernelV8(const __bf16* __restrict__ x,
   }
 }
 
+// ---------------- BN V9: fixed-channel-window kernels ----------------
+// Each thread owns ONE 8-channel window for its whole lifetime and strides
+// over rows: no per-iteration modulo (64-bit int div/mod stalls wave issue),
+// per-channel parameters hoisted into registers once, and for the reduction
+// kernels the private partials flush exactly once (the V8 kernels flushed
+// 16 LDS atomics per iteration whenever C did not divide the grid stride —
+// every non-power-of-two channel count, i.e. most of Inception).
+struct BnIdx {
+  int nw;          // channel windows (c/8)
+  int w;           // this thread's window
+  int64_t chunk;   // row-chunk id
+  int64_t chunks;  // number of row chunks
+  bool active;
+};
+
+__device__ __forceinline__ BnIdx BnThreadIndex(int c, int64_t rows) {
+  BnIdx ix;
+  ix.nw = c >> 3;
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
+  ix.chunks = nthreads / ix.nw;
+  if (ix.chunks > rows) ix.chunks = rows;
+  if (ix.chunks < 1) ix.chunks = 1;
+  ix.w = (int)(tid % ix.nw);
+  ix.chunk = tid / ix.nw;
+  ix.active = ix.chunk < ix.chunks;
+  return ix;
+}
+
+inline int BnGridV9(int c, int64_t rows) {
+  int nw = c >> 3;
+  int64_t chunks = (131072 + nw - 1) / nw;
+  if (chunks > rows) chunks = rows;
+  if (chunks < 1) chunks = 1;
+  return (int)((nw * chunks + 255) / 256) + 1;
+}
+
+template <typename T>
+__global__ void BnStatsKernelV9(const T* __restrict__ x,
+                                float* __restrict__ acc, int64_t rows,
+                                int c) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* s1 = (float*)smem;
+  float* s2 = s1 + c;
+  for (int i = threadIdx.x; i < c; i += blockDim.x) {
+    s1[i] = 0.f;
+    s2[i] = 0.f;
+  }
+  __syncthreads();
+  BnIdx ix = BnThreadIndex(c, rows);
+  float p1[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float p2[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (ix.active) {
+    const T* base = x + (int64_t)ix.w * 8;
+    for (int64_t r = ix.chunk; r < rows; r += ix.chunks) {
+      T v[8];
+      *(ulong2*)v = *(const ulong2*)(base + r * c);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float f = (float)v[e];
+        p1[e] += f;
+        p2[e] += f * f;
+      }
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      atomicAdd(&s1[ix.w * 8 + e], p1[e]);
+      atomicAdd(&s2[ix.w * 8 + e], p2[e]);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < c; i += blockDim.x)
+    if (s1[i] != 0.f || s2[i] != 0.f) {
+      atomicAdd(&acc[i], s1[i]);
+      atomicAdd(&acc[c + i], s2[i]);
+    }
+}
+
+template <bool RELU>
+__global__ void BnGradStatsKernelV9(const __bf16* __restrict__ dy,
+                                    const __bf16* __restrict__ x,
+                                    const __bf16* __restrict__ yr,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ inv_std,
+                                    float* __restrict__ sum_dy,
+                                    float* __restrict__ sum_dy_xhat,
+                                    int64_t rows, int c) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* s1 = (float*)smem;
+  float* s2 = s1 + c;
+  for (int i = threadIdx.x; i < c; i += blockDim.x) {
+    s1[i] = 0.f;
+    s2[i] = 0.f;
+  }
+  __syncthreads();
+  BnIdx ix = BnThreadIndex(c, rows);
+  float p1[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float p2[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (ix.active) {
+    int cb = ix.w * 8;
+    float mloc[8], iloc[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      mloc[e] = mean[cb + e];
+      iloc[e] = inv_std[cb + e];
+    }
+    int64_t off = (int64_t)cb;
+    for (int64_t r = ix.chunk; r < rows; r += ix.chunks) {
+      __bf16 g[8], xv[8], yv[8];
+      *(ulong2*)g = *(const ulong2*)(dy + r * c + off);
+      *(ulong2*)xv = *(const ulong2*)(x + r * c + off);
+      if (RELU) *(ulong2*)yv = *(const ulong2*)(yr + r * c + off);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float gf = (float)g[e];
+        if (RELU && (float)yv[e] <= 0.f) gf = 0.f;
+        float xhat = ((float)xv[e] - mloc[e]) * iloc[e];
+        p1[e] += gf;
+        p2[e] += gf * xhat;
+      }
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      atomicAdd(&s1[cb + e], p1[e]);
+      atomicAdd(&s2[cb + e], p2[e]);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < c; i += blockDim.x)
+    if (s1[i] != 0.f || s2[i] != 0.f) {
+      atomicAdd(&sum_dy[i], s1[i]);
+      atomicAdd(&sum_dy_xhat[i], s2[i]);
+    }
+}
+
+template <bool RELU>
+__global__ void BnNormKernelV9(const __bf16* __restrict__ x,
+                               const float* __restrict__ mean,
+                               const float* __restrict__ inv_std,
+                               const float* __restrict__ scale,
+                               const float* __restrict__ offset,
+                               __bf16* __restrict__ y, int64_t rows, int c) {
+  BnIdx ix = BnThreadIndex(c, rows);
+  if (!ix.active) return;
+  int cb = ix.w * 8;
+  float m[8], is[8], sc[8], of[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    m[e] = mean[cb + e];
+    is[e] = inv_std[cb + e] * scale[cb + e];
+    sc[e] = is[e];
+    of[e] = offset[cb + e] - m[e] * is[e];
+  }
+  for (int64_t r = ix.chunk; r < rows; r += ix.chunks) {
+    __bf16 v[8], o[8];
+    *(ulong2*)v = *(const ulong2*)(x + r * c + cb);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float f = (float)v[e] * sc[e] + of[e];
+      if (RELU) f = f > 0.f ? f : 0.f;
+      o[e] = (__bf16)f;
+    }
+    *(ulong2*)(y + r * c + cb) = *(ulong2*)o;
+  }
+}
+
+__global__ void BnNormAddReluKernelV9(const __bf16* __restrict__ x,
+                                      const __bf16* __restrict__ side,
+                                      const float* __restrict__ mean,
+                                      const float* __restrict__ inv_std,
+                                      const float* __restrict__ scale,
+                                      const float* __restrict__ offset,
+                                      __bf16* __restrict__ y, int64_t rows,
+                                      int c) {
+  BnIdx ix = BnThreadIndex(c, rows);
+  if (!ix.active) return;
+  int cb = ix.w * 8;
+  float sc[8], of[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    sc[e] = inv_std[cb + e] * scale[cb + e];
+    of[e] = offset[cb + e] - mean[cb + e] * sc[e];
+  }
+  for (int64_t r = ix.chunk; r < rows; r += ix.chunks) {
+    __bf16 v[8], sd[8], o[8];
+    *(ulong2*)v = *(const ulong2*)(x + r * c + cb);
+    *(ulong2*)sd = *(const ulong2*)(side + r * c + cb);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float f = (float)v[e] * sc[e] + of[e] + (float)sd[e];
+      o[e] = (__bf16)(f > 0.f ? f : 0.f);
+    }
+    *(ulong2*)(y + r * c + cb) = *(ulong2*)o;
+  }
+}
+
+template <bool RELU>
+__global__ void BnGradKernelV9(const __bf16* __restrict__ dy,
+                               const __bf16* __restrict__ x,
+                               const __bf16* __restrict__ yr,
+                               const float* __restrict__ mean,
+                               const float* __restrict__ inv_std,
+                               const float* __restrict__ scale,
+                               const float* __restrict__ sum_dy,
+                               const float* __restrict__ sum_dy_xhat,
+                               __bf16* __restrict__ dx, int64_t rows, int c) {
+  BnIdx ix = BnThreadIndex(c, rows);
+  if (!ix.active) return;
+  int cb = ix.w * 8;
+  float inv_rows = 1.f / (float)rows;
+  float m[8], is[8], k[8], a[8], b[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    m[e] = mean[cb + e];
+    is[e] = inv_std[cb + e];
+    k[e] = scale[cb + e] * is[e];
+    a[e] = sum_dy[cb + e] * inv_rows;
+    b[e] = sum_dy_xhat[cb + e] * inv_rows;
+  }
+  for (int64_t r = ix.chunk; r < rows; r += ix.chunks) {
+    __bf16 g8[8], x8[8], y8[8], o[8];
+    *(ulong2*)g8 = *(const ulong2*)(dy + r * c + cb);
+    *(ulong2*)x8 = *(const ulong2*)(x + r * c + cb);
+    if (RELU) *(ulong2*)y8 = *(const ulong2*)(yr + r * c + cb);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float xhat = ((float)x8[e] - m[e]) * is[e];
+      float g = (float)g8[e];
+      if (RELU && (float)y8[e] <= 0.f) g = 0.f;
+      o[e] = (__bf16)(k[e] * (g - a[e] - xhat * b[e]));
+    }
+    *(ulong2*)(dx + r * c + cb) = *(ulong2*)o;
+  }
+}
+
 extern "C" hipError_t stf_bn_add_relu(const void* x, const void* side,
                                       const float* mean,
                                       const float* inv_std,
@@ -348,12 +583,12 @@ extern "C" hipError_t stf_bn_add_relu(const void* x, const void* side,
                                       void* y, int64_t n, int c,
                                       hipStream_t stream) {
   if (c % 8 != 0) return hipErrorInvalidValue;
-  int64_t n8 = n / 8;
-  hipLaunchKernelGGL(BnNormAddReluKernelV8, ElemwiseGrid(n8, 256, 1),
-                     dim3(256), 0, stream, (const __bf16*)x,
-                     (const __bf16*)side, mean, inv_std,
+  int64_t rows = n / c;
+  dim3 g9((uint32_t)BnGridV9(c, rows));
+  hipLaunchKernelGGL(BnNormAddReluKernelV9, g9, dim3(256), 0, stream,
+                     (const __bf16*)x, (const __bf16*)side, mean, inv_std,
                      (const float*)scale, (const float*)offset, (__bf16*)y,
-                     n8, c);
+                     rows, c);
   return hipGetLastError();
 }
 
@@ -502,6 +737,7 @@ __global__ void BnGradKernelV8(const __bf16* __restrict__ dy,
   }
 }
 
+
 // ---------------- pooling ----------------
 struct PoolGeom {
   int N, H, W, C, kh, kw, sh, sw, ph, pw, P, Q;
@@ -533,6 +769,163 @@ __global__ void PoolFwdKernel(const T* __restrict__ x, T* __restrict__ y,
       }
     }
     y[i] = (T)(IS_MAX ? best : best / count);
+  }
+}
+
+
+// C%8==0 8-wide pooling (16B loads/stores, one index decomposition per 8
+// channels, O(1) border-count instead of the nested window scan).
+template <bool IS_MAX>
+__global__ void PoolFwdKernelV8(const __bf16* __restrict__ x,
+                                __bf16* __restrict__ y, PoolGeom g,
+                                int64_t total8) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int c8 = g.C / 8;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total8;
+       i += stride) {
+    int64_t rem = i;
+    int cw = (int)(rem % c8); rem /= c8;
+    int q = (int)(rem % g.Q); rem /= g.Q;
+    int p = (int)(rem % g.P); rem /= g.P;
+    int n = (int)rem;
+    int h0 = max(p * g.sh - g.ph, 0), h1 = min(p * g.sh - g.ph + g.kh, g.H);
+    int w0 = max(q * g.sw - g.pw, 0), w1 = min(q * g.sw - g.pw + g.kw, g.W);
+    float acc[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] = IS_MAX ? -3.4e38f : 0.f;
+    for (int ih = h0; ih < h1; ++ih)
+      for (int iw = w0; iw < w1; ++iw) {
+        __bf16 v[8];
+        *(ulong2*)v = *(const ulong2*)(
+            x + ((int64_t)(n * g.H + ih) * g.W + iw) * g.C + cw * 8);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float f = (float)v[e];
+          if (IS_MAX) acc[e] = fmaxf(acc[e], f);
+          else acc[e] += f;
+        }
+      }
+    float inv = IS_MAX ? 1.f : 1.f / ((h1 - h0) * (w1 - w0));
+    __bf16 o[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o[e] = (__bf16)(IS_MAX ? acc[e] : acc[e] * inv);
+    *(ulong2*)(y + ((int64_t)(n * g.P + p) * g.Q + q) * g.C + cw * 8) =
+        *(ulong2*)o;
+  }
+}
+
+__global__ void AvgPoolGradKernelV8(const __bf16* __restrict__ dy,
+                                    __bf16* __restrict__ dx, PoolGeom g,
+                                    int64_t total8) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int c8 = g.C / 8;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total8;
+       i += stride) {
+    int64_t rem = i;
+    int cw = (int)(rem % c8); rem /= c8;
+    int iw = (int)(rem % g.W); rem /= g.W;
+    int ih = (int)(rem % g.H); rem /= g.H;
+    int n = (int)rem;
+    float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    // output rows whose window covers ih: p*sh <= ih+ph < p*sh+kh
+    int plo = max((ih + g.ph - g.kh + g.sh) / g.sh, 0);
+    int phi = min((ih + g.ph) / g.sh, g.P - 1);
+    int qlo = max((iw + g.pw - g.kw + g.sw) / g.sw, 0);
+    int qhi = min((iw + g.pw) / g.sw, g.Q - 1);
+    for (int p = plo; p <= phi; ++p) {
+      int h0 = max(p * g.sh - g.ph, 0);
+      int h1 = min(p * g.sh - g.ph + g.kh, g.H);
+      for (int q = qlo; q <= qhi; ++q) {
+        int w0 = max(q * g.sw - g.pw, 0);
+        int w1 = min(q * g.sw - g.pw + g.kw, g.W);
+        float inv = 1.f / ((h1 - h0) * (w1 - w0));
+        __bf16 v[8];
+        *(ulong2*)v = *(const ulong2*)(
+            dy + ((int64_t)(n * g.P + p) * g.Q + q) * g.C + cw * 8);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) acc[e] += (float)v[e] * inv;
+      }
+    }
+    __bf16 o[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o[e] = (__bf16)acc[e];
+    *(ulong2*)(dx + i * 8) = *(ulong2*)o;
+  }
+}
+
+
+// Per-INPUT max-pool gradient, 8 channels wide, no atomics/f32 scratch:
+// each input pixel sums dy over the covering windows where it is the
+// first argmax (scan order matches the per-output kernel's routing).
+__global__ void MaxPoolGradKernelV8(const __bf16* __restrict__ x,
+                                    const __bf16* __restrict__ dy,
+                                    __bf16* __restrict__ dx, PoolGeom g,
+                                    int64_t total8) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int c8 = g.C / 8;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total8;
+       i += stride) {
+    int64_t rem = i;
+    int cw = (int)(rem % c8); rem /= c8;
+    int iw = (int)(rem % g.W); rem /= g.W;
+    int ih = (int)(rem % g.H); rem /= g.H;
+    int n = (int)rem;
+    float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    float self[8];
+    {
+      __bf16 v[8];
+      *(ulong2*)v = *(const ulong2*)(
+          x + ((int64_t)(n * g.H + ih) * g.W + iw) * g.C + cw * 8);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) self[e] = (float)v[e];
+    }
+    int plo = max((ih + g.ph - g.kh + g.sh) / g.sh, 0);
+    int phi = min((ih + g.ph) / g.sh, g.P - 1);
+    int qlo = max((iw + g.pw - g.kw + g.sw) / g.sw, 0);
+    int qhi = min((iw + g.pw) / g.sw, g.Q - 1);
+    for (int p = plo; p <= phi; ++p) {
+      for (int q = qlo; q <= qhi; ++q) {
+        // first-argmax position per channel within this window
+        float best[8];
+        int bpos[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          best[e] = -3.4e38f;
+          bpos[e] = -1;
+        }
+        for (int kh = 0; kh < g.kh; ++kh) {
+          int h = p * g.sh - g.ph + kh;
+          if (h < 0 || h >= g.H) continue;
+          for (int kw = 0; kw < g.kw; ++kw) {
+            int w = q * g.sw - g.pw + kw;
+            if (w < 0 || w >= g.W) continue;
+            __bf16 v[8];
+            *(ulong2*)v = *(const ulong2*)(
+                x + ((int64_t)(n * g.H + h) * g.W + w) * g.C + cw * 8);
+            int pos = h * g.W + w;
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+              float f = (float)v[e];
+              if (f > best[e]) {
+                best[e] = f;
+                bpos[e] = pos;
+              }
+            }
+          }
+        }
+        __bf16 gv[8];
+        *(ulong2*)gv = *(const ulong2*)(
+            dy + ((int64_t)(n * g.P + p) * g.Q + q) * g.C + cw * 8);
+        int mypos = ih * g.W + iw;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          if (bpos[e] == mypos) acc[e] += (float)gv[e];
+      }
+    }
+    __bf16 o[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o[e] = (__bf16)acc[e];
+    *(ulong2*)(dx + i * 8) = *(ulong2*)o;
   }
 }
 
@@ -809,14 +1202,22 @@ extern "C" hipError_t stf_bn_stats_only(int dtype, const void* x, float* acc,
                                         float* mean, float* var,
                                         float* inv_std, int64_t rows, int c,
                                         float eps, hipStream_t stream) {
-  int blocks = 512;
   size_t lds = (size_t)c * 8;
-  if (dtype == 0)
-    hipLaunchKernelGGL((BnStatsKernel<float>), dim3(blocks), dim3(256), lds,
+  if (c % 8 == 0) {
+    int blocks = BnGridV9(c, rows);
+    if (dtype == 0)
+      hipLaunchKernelGGL((BnStatsKernelV9<float>), dim3(blocks), dim3(256),
+                         lds, stream, (const float*)x, acc, rows, c);
+    else
+      hipLaunchKernelGGL((BnStatsKernelV9<__bf16>), dim3(blocks), dim3(256),
+                         lds, stream, (const __bf16*)x, acc, rows, c);
+  } else if (dtype == 0) {
+    hipLaunchKernelGGL((BnStatsKernel<float>), dim3(512), dim3(256), lds,
                        stream, (const float*)x, acc, rows, c);
-  else
-    hipLaunchKernelGGL((BnStatsKernel<__bf16>), dim3(blocks), dim3(256), lds,
+  } else {
+    hipLaunchKernelGGL((BnStatsKernel<__bf16>), dim3(512), dim3(256), lds,
                        stream, (const __bf16*)x, acc, rows, c);
+  }
   hipLaunchKernelGGL(BnFinalizeKernel, dim3((c + 255) / 256), dim3(256), 0,
                      stream, acc, mean, var, inv_std, rows, c, eps);
   return hipGetLastError();
@@ -827,14 +1228,22 @@ hipError_t stf_bn_fwd(int dtype, const void* x, const void* scale,
                       const void* offset, float* acc, float* mean, float* var,
                       float* inv_std, void* y, int64_t rows, int c, float eps,
                       int fuse_relu, hipStream_t stream) {
-  int blocks = 512;
   size_t lds = (size_t)c * 8;
-  if (dtype == 0)
-    hipLaunchKernelGGL((BnStatsKernel<float>), dim3(blocks), dim3(256), lds,
+  if (c % 8 == 0) {
+    int blocks9 = BnGridV9(c, rows);
+    if (dtype == 0)
+      hipLaunchKernelGGL((BnStatsKernelV9<float>), dim3(blocks9), dim3(256),
+                         lds, stream, (const float*)x, acc, rows, c);
+    else
+      hipLaunchKernelGGL((BnStatsKernelV9<__bf16>), dim3(blocks9), dim3(256),
+                         lds, stream, (const __bf16*)x, acc, rows, c);
+  } else if (dtype == 0) {
+    hipLaunchKernelGGL((BnStatsKernel<float>), dim3(512), dim3(256), lds,
                        stream, (const float*)x, acc, rows, c);
-  else
-    hipLaunchKernelGGL((BnStatsKernel<__bf16>), dim3(blocks), dim3(256), lds,
+  } else {
+    hipLaunchKernelGGL((BnStatsKernel<__bf16>), dim3(512), dim3(256), lds,
                        stream, (const __bf16*)x, acc, rows, c);
+  }
   hipLaunchKernelGGL(BnFinalizeKernel, dim3((c + 255) / 256), dim3(256), 0,
                      stream, acc, mean, var, inv_std, rows, c, eps);
   int64_t n = rows * c;
@@ -846,18 +1255,17 @@ hipError_t stf_bn_fwd(int dtype, const void* x, const void* scale,
   if (dtype == 0) {
     if (fuse_relu) BNN(float, true); else BNN(float, false);
   } else if (c % 8 == 0) {
-    int64_t n8 = n / 8;
-    dim3 g8 = ElemwiseGrid(n8, 256, 1);
+    dim3 g9((uint32_t)BnGridV9(c, rows));
     if (fuse_relu)
-      hipLaunchKernelGGL((BnNormKernelV8<true>), g8, dim3(256), 0, stream,
+      hipLaunchKernelGGL((BnNormKernelV9<true>), g9, dim3(256), 0, stream,
                          (const __bf16*)x, mean, inv_std,
                          (const float*)scale, (const float*)offset,
-                         (__bf16*)y, n8, c);
+                         (__bf16*)y, rows, c);
     else
-      hipLaunchKernelGGL((BnNormKernelV8<false>), g8, dim3(256), 0, stream,
+      hipLaunchKernelGGL((BnNormKernelV9<false>), g9, dim3(256), 0, stream,
                          (const __bf16*)x, mean, inv_std,
                          (const float*)scale, (const float*)offset,
-                         (__bf16*)y, n8, c);
+                         (__bf16*)y, rows, c);
   } else {
     if (fuse_relu) BNN(__bf16, true); else BNN(__bf16, false);
   }
@@ -890,17 +1298,16 @@ hipError_t stf_bn_bwd(int dtype, const void* dy, const void* x,
   } while (0)
 #define BNB8(R)                                                             \
   do {                                                                      \
-    hipLaunchKernelGGL((BnGradStatsKernel<__bf16, R>), dim3(blocks),        \
-                       dim3(256), lds, stream, (const __bf16*)dy,           \
-                       (const __bf16*)x, (const __bf16*)y_relu, mean,       \
-                       inv_std, sum_dy, sum_dy_xhat, rows, c);              \
-    int64_t n8 = n / 8;                                                     \
-    dim3 g8 = ElemwiseGrid(n8, 256, 1);                                     \
-    hipLaunchKernelGGL((BnGradKernelV8<R>), g8, dim3(256), 0, stream,       \
+    dim3 g9((uint32_t)BnGridV9(c, rows));                                   \
+    hipLaunchKernelGGL((BnGradStatsKernelV9<R>), g9, dim3(256), lds,        \
+                       stream, (const __bf16*)dy, (const __bf16*)x,         \
+                       (const __bf16*)y_relu, mean, inv_std, sum_dy,        \
+                       sum_dy_xhat, rows, c);                               \
+    hipLaunchKernelGGL((BnGradKernelV9<R>), g9, dim3(256), 0, stream,       \
                        (const __bf16*)dy, (const __bf16*)x,                 \
                        (const __bf16*)y_relu, mean, inv_std,                \
                        (const float*)scale, sum_dy, sum_dy_xhat,            \
-                       (__bf16*)dx, n8, rows, c);                           \
+                       (__bf16*)dx, rows, c);                               \
   } while (0)
   if (dtype == 0) {
     if (fuse_relu) BNB(float, true); else BNB(float, false);
@@ -925,6 +1332,15 @@ hipError_t stf_pool_fwd(int dtype, int is_max, const void* x, void* y, int N,
                      (const T*)x, (T*)y, g, total)
   if (dtype == 0) {
     if (is_max) PF(float, true); else PF(float, false);
+  } else if (C % 8 == 0) {
+    int64_t total8 = total / 8;
+    dim3 g8 = ElemwiseGrid(total8, 256, 1);
+    if (is_max)
+      hipLaunchKernelGGL((PoolFwdKernelV8<true>), g8, dim3(256), 0, stream,
+                         (const __bf16*)x, (__bf16*)y, g, total8);
+    else
+      hipLaunchKernelGGL((PoolFwdKernelV8<false>), g8, dim3(256), 0, stream,
+                         (const __bf16*)x, (__bf16*)y, g, total8);
   } else {
     if (is_max) PF(__bf16, true); else PF(__bf16, false);
   }
@@ -951,18 +1367,38 @@ hipError_t stf_max_pool_bwd(int dtype, const void* x, const void* dy,
   return hipGetLastError();
 }
 
+hipError_t stf_max_pool_bwd_v8(const void* x, const void* dy, void* dx_bf16,
+                               int N, int H, int W, int C, int kh, int kw,
+                               int sh, int sw, int ph, int pw, int P, int Q,
+                               hipStream_t stream) {
+  if (C % 8 != 0) return hipErrorInvalidValue;
+  PoolGeom g{N, H, W, C, kh, kw, sh, sw, ph, pw, P, Q};
+  int64_t total8 = (int64_t)N * H * W * C / 8;
+  dim3 grid = ElemwiseGrid(total8, 256, 1);
+  hipLaunchKernelGGL(MaxPoolGradKernelV8, grid, dim3(256), 0, stream,
+                     (const __bf16*)x, (const __bf16*)dy, (__bf16*)dx_bf16,
+                     g, total8);
+  return hipGetLastError();
+}
+
 hipError_t stf_avg_pool_bwd(int dtype, const void* dy, void* dx, int N, int H,
                             int W, int C, int kh, int kw, int sh, int sw,
                             int ph, int pw, int P, int Q, hipStream_t stream) {
   PoolGeom g{N, H, W, C, kh, kw, sh, sw, ph, pw, P, Q};
   int64_t total = (int64_t)N * H * W * C;
   dim3 grid = ElemwiseGrid(total, 256, 1);
-  if (dtype == 0)
+  if (dtype == 0) {
     hipLaunchKernelGGL((AvgPoolGradKernel<float>), grid, dim3(256), 0, stream,
                        (const float*)dy, (float*)dx, g, total);
-  else
+  } else if (C % 8 == 0) {
+    int64_t total8 = total / 8;
+    dim3 g8 = ElemwiseGrid(total8, 256, 1);
+    hipLaunchKernelGGL(AvgPoolGradKernelV8, g8, dim3(256), 0, stream,
+                       (const __bf16*)dy, (__bf16*)dx, g, total8);
+  } else {
     hipLaunchKernelGGL((AvgPoolGradKernel<__bf16>), grid, dim3(256), 0, stream,
                        (const __bf16*)dy, (__bf16*)dx, g, total);
+  }
   return hipGetLastError();
 }
 

@@ -244,3 +244,56 @@ def test_lrn_gpu_vs_torch():
     want_g = tx.grad.permute(0, 2, 3, 1).numpy()
     np.testing.assert_allclose(got_y, want_y, rtol=1e-3, atol=1e-4)
     np.testing.assert_allclose(got_g, want_g, rtol=1e-2, atol=1e-3)
+
+
+def test_batch_norm_non_pow2_channels_vs_torch():
+    # exercises the V9 fixed-window kernels with C=48 (Inception-style
+    # non-power-of-two channel count) including the backward pass
+    from simple_tensorflow_amd.python.framework.ops import apply_op
+    rng = np.random.RandomState(9)
+    C = 48
+    x = rng.randn(8, 9, 9, C).astype(np.float32)
+    scale = rng.rand(C).astype(np.float32) + 0.5
+    offset = rng.randn(C).astype(np.float32)
+    xt = tf.constant(x, dtype=tf.bfloat16)
+    y, mean, var, _ = apply_op('BatchNormMi', xt, tf.constant(scale),
+                               tf.constant(offset), epsilon=1e-4)
+    loss = tf.reduce_sum(tf.cast(y, tf.float32) ** 2.0)
+    gx = tf.gradients(loss, [xt])[0]
+    got_y, got_mean, got_gx = _run([y, mean, gx])
+    txt = torch.from_numpy(x).permute(0, 3, 1, 2).bfloat16().float()
+    txt.requires_grad_(True)
+    want = torch.nn.functional.batch_norm(
+        txt, None, None, torch.from_numpy(scale), torch.from_numpy(offset),
+        training=True, eps=1e-4)
+    tloss = (want ** 2).sum()
+    tloss.backward()
+    want_y = want.detach().permute(0, 2, 3, 1).numpy()
+    assert np.abs(got_y - want_y).max() < 0.15
+    np.testing.assert_allclose(got_mean, x.reshape(-1, C).mean(0),
+                               rtol=1e-2, atol=1e-2)
+    want_gx = txt.grad.permute(0, 2, 3, 1).numpy()
+    denom = np.abs(want_gx) + 0.1
+    assert np.percentile(np.abs(got_gx - want_gx) / denom, 99) < 0.15
+
+
+def test_avg_pool_fwd_and_grad_vs_torch():
+    rng = np.random.RandomState(10)
+    x = rng.randn(4, 13, 13, 32).astype(np.float32)
+    xt = tf.constant(x, dtype=tf.bfloat16)
+    y = tf.nn.avg_pool(xt, [1, 3, 3, 1], [1, 1, 1, 1], 'SAME')
+    dy = rng.randn(4, 13, 13, 32).astype(np.float32)
+    g = tf.gradients(tf.reduce_sum(y * tf.constant(dy, dtype=tf.bfloat16)),
+                     [xt])[0]
+    got_y, got_g = _run([y, g])
+    txt = torch.from_numpy(x).permute(0, 3, 1, 2).bfloat16().float()
+    txt.requires_grad_(True)
+    want = torch.nn.functional.avg_pool2d(txt, 3, 1, padding=1,
+                                          count_include_pad=False)
+    tdy = torch.from_numpy(dy).permute(0, 3, 1, 2).bfloat16().float()
+    (want * tdy).sum().backward()
+    want_y = want.detach().permute(0, 2, 3, 1).numpy()
+    want_g = txt.grad.permute(0, 2, 3, 1).numpy()
+    assert np.abs(got_y - want_y).max() < 0.05
+    denom = np.abs(want_g) + 0.1
+    assert np.percentile(np.abs(got_g - want_g) / denom, 99) < 0.1

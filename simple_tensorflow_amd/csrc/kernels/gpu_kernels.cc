@@ -129,6 +129,7 @@ hipError_t stf_lstm_gates_grad(int, const void*, const void*, const void*,
                                const void*, const void*, void*, void*,
                                int64_t, int, hipStream_t);
 hipError_t stf_l2loss(int, const void*, float*, int64_t, hipStream_t);
+hipError_t stf_cast_f32_bf16_zero(void*, void*, int64_t, hipStream_t);
 }
 
 namespace {
@@ -177,6 +178,31 @@ inline int PickSplitK(int64_t M, int64_t N, int64_t K) {
   return (int)(sk < 1 ? 1 : sk);
 }
 
+
+// Persistent pre-zeroed f32 accumulator arena for split-K (per GPU): every
+// split-K GEMM atomically accumulates into it and the fused
+// cast-and-rezero epilogue restores the all-zero invariant while draining
+// the result — the per-GEMM hipMemset launch disappears. All compute runs
+// on one stream, so consecutive uses are ordered; growth leaks the old
+// buffer deliberately (in-flight work and captured hipGraphs may still
+// reference it).
+inline float* SplitKArena(int ordinal, int64_t elems, hipStream_t s) {
+  if (elems * 4 > (1ll << 28)) return nullptr;  // >256 MB: use a temp
+  static std::mutex mu;
+  static std::map<int, std::pair<float*, int64_t>> arenas;
+  std::lock_guard<std::mutex> l(mu);
+  auto& a = arenas[ordinal];
+  if (a.second < elems) {
+    int64_t cap = a.second * 2 > elems ? a.second * 2 : elems;
+    if (cap < (1 << 20)) cap = 1 << 20;
+    float* p = nullptr;
+    if (hipMalloc(&p, cap * 4) != hipSuccess) return nullptr;
+    if (hipMemsetAsync(p, 0, cap * 4, s) != hipSuccess) return nullptr;
+    a = {p, cap};
+  }
+  return a.first;
+}
+
 // GEMM helper: picks plain vs split-K (f32 scratch + cast) automatically.
 // a_km/b_km: the operand is stored contraction-major ([K,M]/[K,N], row
 // stride lda/ldb) and transposed in-kernel during LDS staging.
@@ -189,6 +215,12 @@ inline hipError_t GemmBf16AutoEx(OpKernelContext* ctx, const void* A,
   if (sk <= 1)
     return stf_gemm_bf16(A, B, C_bf16, nullptr, M, N, K, lda, ldb, 0.f, a_km,
                          b_km, 1, 0, s);
+  if (float* arena = SplitKArena(ctx->device()->gpu_ordinal(), M * N, s)) {
+    hipError_t e = stf_gemm_bf16_splitk(A, B, arena, M, N, K, lda, ldb, a_km,
+                                        b_km, sk, s);
+    if (e != hipSuccess) return e;
+    return stf_cast_f32_bf16_zero(arena, C_bf16, M * N, s);
+  }
   Tensor scratch = ctx->allocate_temp(DT_FLOAT, TensorShape({M, N}));
   hipError_t e = hipMemsetAsync(scratch.raw_data(), 0, M * N * 4, s);
   if (e != hipSuccess) return e;
@@ -859,9 +891,20 @@ class GpuConv2DBackpropFilterOp : public OpKernel {
     static const bool no_implicit = getenv("STF_NO_IMPLICIT_CONV") != nullptr;
     if (!no_implicit && !g.is_1x1_s1() && (g.C % 8) == 0 &&
         (g.K & 7) == 0 && (g.M() & 63) == 0 && ZeroPage()) {
+      int sk = PickSplitK(rsc, g.K, g.M());
+      if (float* arena =
+              SplitKArena(ctx->device()->gpu_ordinal(), rsc * g.K, s)) {
+        OP_HIP_OK(ctx, stf_conv2d_dw_splitk(
+                           x.raw_data(), dy.raw_data(), arena, ZeroPage(),
+                           (int)g.N, (int)g.H, (int)g.W, (int)g.C, (int)g.R,
+                           (int)g.S, (int)g.sh, (int)g.sw, (int)g.ph,
+                           (int)g.pw, (int)g.P, (int)g.Q, g.K, sk, s));
+        OP_HIP_OK(ctx, stf_cast_f32_bf16_zero(arena, dw->raw_data(),
+                                              rsc * g.K, s));
+        return;
+      }
       Tensor scratch = ctx->allocate_temp(DT_FLOAT, TensorShape({rsc, g.K}));
       OP_HIP_OK(ctx, hipMemsetAsync(scratch.raw_data(), 0, rsc * g.K * 4, s));
-      int sk = PickSplitK(rsc, g.K, g.M());
       OP_HIP_OK(ctx, stf_conv2d_dw_splitk(
                          x.raw_data(), dy.raw_data(), scratch.raw_data(),
                          ZeroPage(), (int)g.N, (int)g.H, (int)g.W, (int)g.C,

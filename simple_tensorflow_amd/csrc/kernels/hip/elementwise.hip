@@ -195,6 +195,43 @@ __global__ void FillKernel(T* __restrict__ y, T v, int64_t n) {
 }  // namespace
 
 // dtype codes: 0 = f32, 1 = bf16 (carried as uint16)
+
+// Fused f32->bf16 cast that RE-ZEROES the source as it reads: restores the
+// split-K scratch arena's all-zero invariant in the same pass that drains
+// the result (replaces the per-GEMM hipMemset launch).
+__global__ void CastF32Bf16ZeroKernel(float* __restrict__ src,
+                                      __bf16* __restrict__ dst, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t n8 = n / 8;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += stride) {
+    float v[8];
+    *(ulong2*)(v + 0) = *(const ulong2*)(src + i * 8);
+    *(ulong2*)(v + 4) = *(const ulong2*)(src + i * 8 + 4);
+    __bf16 o[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o[e] = (__bf16)v[e];
+    *(ulong2*)(dst + i * 8) = *(ulong2*)o;
+    ulong2 z = {0, 0};
+    *(ulong2*)(src + i * 8) = z;
+    *(ulong2*)(src + i * 8 + 4) = z;
+  }
+  int64_t t0 = n8 * 8;
+  for (int64_t i = t0 + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += stride) {
+    dst[i] = (__bf16)src[i];
+    src[i] = 0.f;
+  }
+}
+
+extern "C" hipError_t stf_cast_f32_bf16_zero(void* src_f32, void* dst_bf16,
+                                             int64_t n, hipStream_t stream) {
+  dim3 grid = ElemwiseGrid(n / 8 + 1, 256, 1);
+  hipLaunchKernelGGL(CastF32Bf16ZeroKernel, grid, dim3(256), 0, stream,
+                     (float*)src_f32, (__bf16*)dst_bf16, n);
+  return hipGetLastError();
+}
+
 extern "C" hipError_t stf_unary(int op, int dtype, const void* x, void* y,
                                 int64_t n, hipStream_t stream) {
   dim3 grid = ElemwiseGrid(n);

@@ -130,6 +130,9 @@ hipError_t stf_lstm_gates_grad(int, const void*, const void*, const void*,
                                int64_t, int, hipStream_t);
 hipError_t stf_l2loss(int, const void*, float*, int64_t, hipStream_t);
 hipError_t stf_cast_f32_bf16_zero(void*, void*, int64_t, hipStream_t);
+hipError_t stf_conv2d_fwd_nt(const void*, const void*, void*, const void*,
+                             int, int, int, int, int, int, int, int, int,
+                             int, int, int, int64_t, int64_t, hipStream_t);
 }
 
 namespace {
@@ -778,6 +781,17 @@ class GpuConv2DOp : public OpKernel {
                          ZeroPage(), (int)g.N, (int)g.H, (int)g.W, (int)g.C,
                          (int)g.R, (int)g.S, (int)g.sh, (int)g.sw, (int)g.ph,
                          (int)g.pw, (int)g.P, (int)g.Q, g.K, rscp, 1, 0, s));
+      return;
+    }
+    // Implicit-GEMM through the general NT template for shapes the 8-phase
+    // kernel cannot take (e.g. Inception's 48/96/288-cout convs).
+    if (!no_implicit && !g.is_1x1_s1() && (g.C % 8) == 0 &&
+        (g.K & 7) == 0 && ZeroPage()) {
+      OP_HIP_OK(ctx, stf_conv2d_fwd_nt(
+                         x.raw_data(), wsrc.raw_data(), y->raw_data(),
+                         ZeroPage(), (int)g.N, (int)g.H, (int)g.W, (int)g.C,
+                         (int)g.R, (int)g.S, (int)g.sh, (int)g.sw, (int)g.ph,
+                         (int)g.pw, (int)g.P, (int)g.Q, g.K, rscp, s));
       return;
     }
     const void* col_data = x.raw_data();

@@ -549,3 +549,48 @@ extern "C" hipError_t stf_conv2d_dw_splitk(
   }
   return hipGetLastError();
 }
+// Implicit-GEMM Conv2D forward through the general NT template: the column
+// matrix row (one output pixel's [r,s,c] window) is generated inside the A
+// staging by ConvAG — covers the shapes the 256-row 8-phase kernel cannot
+// take (couts or M not multiples of its tile). Weights stay [rsc, K]
+// contraction-major (B_KM staging transposes in-register).
+extern "C" hipError_t stf_conv2d_fwd_nt(
+    const void* x, const void* w, void* y, const void* zero16, int n, int h,
+    int w_, int c, int r, int s_, int sh, int sw, int ph, int pw, int p,
+    int q, int64_t cout, int64_t rscp, hipStream_t stream) {
+  if ((c % 8) != 0) return hipErrorInvalidValue;
+  ConvAG ag;
+  ag.x = (const uint16_t*)x;
+  ag.zero16 = (const uint16_t*)zero16;
+  ag.div_pq.init((uint32_t)(p * q));
+  ag.div_q.init((uint32_t)q);
+  ag.div_c.init((uint32_t)c);
+  ag.div_s.init((uint32_t)s_);
+  ag.H = h; ag.W = w_; ag.C = c; ag.S = s_;
+  ag.sh = sh; ag.sw = sw; ag.ph = ph; ag.pw = pw;
+  ag.rsc = (int64_t)r * s_ * c;
+  int64_t M = (int64_t)n * p * q;
+  int64_t N = cout;
+  int64_t K = rscp;
+  const uint16_t* b = (const uint16_t*)w;
+  auto launch = [&](auto kern, int BM, int BN) {
+    int64_t blocks = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
+    hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0, stream,
+                       ag, b, y, nullptr, M, N, K, cout, 0.f, 1);
+  };
+  if (N >= 128 && M >= 128) {
+    launch(GemmBf16NT<2, 2, 4, 4, false, true, true, false, false, true,
+                      ConvAG>, 128, 128);
+  } else if (N >= 128) {
+    launch(GemmBf16NT<2, 2, 2, 4, false, true, true, false, false, true,
+                      ConvAG>, 64, 128);
+  } else if (M >= 128) {
+    launch(GemmBf16NT<2, 2, 4, 2, false, true, true, false, false, true,
+                      ConvAG>, 128, 64);
+  } else {
+    launch(GemmBf16NT<2, 2, 2, 2, false, true, true, false, false, true,
+                      ConvAG>, 64, 64);
+  }
+  return hipGetLastError();
+}
+

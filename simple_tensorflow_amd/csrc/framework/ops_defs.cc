@@ -157,6 +157,11 @@ REGISTER_OP("BatchNormMi").Input("x: T").Input("scale: float").Input("offset: fl
 REGISTER_OP("BatchNormAddReluMi").Input("x: T").Input("scale: float").Input("offset: float").Input("side: T").Output("y: T").Output("batch_mean: float").Output("batch_variance: float").Output("reserve_inv_std: float").Attr("T: {float, bfloat16}").Attr("epsilon: float = 0.0001");
 REGISTER_OP("BatchNormAddReluMiGrad").Input("y_backprop: T").Input("x: T").Input("scale: float").Input("saved_mean: float").Input("saved_inv_std: float").Input("y_relu: T").Output("x_backprop: T").Output("scale_backprop: float").Output("offset_backprop: float").Output("side_backprop: T").Attr("T: {float, bfloat16}").Attr("epsilon: float = 0.0001");
 REGISTER_OP("BatchNormMiGrad").Input("y_backprop: T").Input("x: T").Input("scale: float").Input("saved_mean: float").Input("saved_inv_std: float").Input("y_relu: T").Output("x_backprop: T").Output("scale_backprop: float").Output("offset_backprop: float").Attr("T: {float, bfloat16}").Attr("epsilon: float = 0.0001").Attr("fuse_relu: bool = false");
+// Internal: scoped elementwise fusion output (graph/optimizer.cc emits it
+// post-autodiff; never constructed from python). Input 0 is the root
+// tensor, inputs 1..N-1 are scalar side-inputs; `program` is the packed
+// bytecode of kernels/fused_ew.h.
+REGISTER_OP("_FusedElementwise").Input("inputs: N * T").Output("output: T").Attr("N: int >= 1").Attr("T: {float, bfloat16}").Attr("program: list(int)");
 REGISTER_OP("L2Loss").Input("t: T").Output("output: T").Attr("T: " FLOATTYPES);
 // Fused LSTM cell pointwise math over the post-GEMM gate matrix [B, 4H]
 // (gate order i, j, f, o — BasicLSTMCell split order). Outputs every

@@ -6,6 +6,7 @@
 #include <cmath>
 #include <limits>
 
+#include "kernels/fused_ew.h"
 #include "kernels/kernel_util.h"
 
 namespace stf {
@@ -336,6 +337,76 @@ class SelectOp : public OpKernel {
   }
 };
 REGISTER_KERNEL_BUILDER(Name("Select").Device(DEVICE_CPU), SelectOp);
+
+
+// ----------------------- fused elementwise (bytecode) -----------------------
+// CPU interpreter for _FusedElementwise (see kernels/fused_ew.h).
+template <typename T>
+class FusedElementwiseOp : public OpKernel {
+ public:
+  explicit FusedElementwiseOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("program", &prog_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    int n_in = ctx->num_inputs();
+    const Tensor& rootT = ctx->input(0);
+    Tensor* out = ctx->allocate_output(0, rootT.shape());
+    int64_t n = rootT.NumElements();
+    const T* in[fused_ew::kMaxInputs];
+    bool scalar[fused_ew::kMaxInputs];
+    for (int i = 0; i < n_in; ++i) {
+      in[i] = ctx->input(i).flat<T>();
+      scalar[i] = ctx->input(i).NumElements() == 1;
+    }
+    T* op = out->flat<T>();
+    int np = (int)prog_.size();
+    for (int64_t idx = 0; idx < n; ++idx) {
+      float vals[fused_ew::kMaxInputs + fused_ew::kMaxInstr];
+      for (int i = 0; i < n_in; ++i)
+        vals[i] = (float)in[i][scalar[i] ? 0 : idx];
+      for (int k = 0; k < np; ++k) {
+        int opx, a, b;
+        fused_ew::Unpack(prog_[k], &opx, &a, &b);
+        float x = vals[a], y = vals[b], r = 0.f;
+        using namespace fused_ew;
+        switch (opx) {
+          case kRelu: r = x > 0 ? x : 0; break;
+          case kRelu6: r = x < 0 ? 0 : (x > 6 ? 6 : x); break;
+          case kSigmoid: r = 1.f / (1.f + std::exp(-x)); break;
+          case kTanh: r = std::tanh(x); break;
+          case kExp: r = std::exp(x); break;
+          case kLog: r = std::log(x); break;
+          case kLog1p: r = std::log1p(x); break;
+          case kNeg: r = -x; break;
+          case kSqrt: r = std::sqrt(x); break;
+          case kRsqrt: r = 1.f / std::sqrt(x); break;
+          case kSquare: r = x * x; break;
+          case kAbs: r = std::fabs(x); break;
+          case kSoftplus: r = std::log1p(std::exp(-std::fabs(x))) +
+                              (x > 0 ? x : 0); break;
+          case kSign: r = x > 0 ? 1.f : (x < 0 ? -1.f : 0.f); break;
+          case kFloor: r = std::floor(x); break;
+          case kReciprocal: r = 1.f / x; break;
+          case kAdd: r = x + y; break;
+          case kSub: r = x - y; break;
+          case kMul: r = x * y; break;
+          case kDiv: r = x / y; break;
+          case kMaximum: r = x > y ? x : y; break;
+          case kMinimum: r = x < y ? x : y; break;
+          case kSquaredDifference: r = (x - y) * (x - y); break;
+          case kPow: r = std::pow(x, y); break;
+        }
+        vals[n_in + k] = r;
+      }
+      op[idx] = (T)vals[n_in + np - 1];
+    }
+  }
+
+ private:
+  std::vector<int64_t> prog_;
+};
+REGISTER_KERNEL_BUILDER(Name("_FusedElementwise").Device(DEVICE_CPU).TypeConstraint<float>("T"), FusedElementwiseOp<float>);
+REGISTER_KERNEL_BUILDER(Name("_FusedElementwise").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), FusedElementwiseOp<bfloat16>);
 
 // --------------------------------- AddN -------------------------------------
 template <typename T>

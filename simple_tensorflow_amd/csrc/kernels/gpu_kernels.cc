@@ -130,6 +130,9 @@ hipError_t stf_lstm_gates_grad(int, const void*, const void*, const void*,
                                int64_t, int, hipStream_t);
 hipError_t stf_l2loss(int, const void*, float*, int64_t, hipStream_t);
 hipError_t stf_cast_f32_bf16_zero(void*, void*, int64_t, hipStream_t);
+hipError_t stf_fused_elementwise(int, const void**, const uint8_t*, int,
+                                 const int64_t*, int, void*, int64_t,
+                                 hipStream_t);
 hipError_t stf_conv2d_fwd_nt(const void*, const void*, void*, const void*,
                              int, int, int, int, int, int, int, int, int,
                              int, int, int, int64_t, int64_t, hipStream_t);
@@ -1521,6 +1524,36 @@ class GpuL2LossOp : public OpKernel {
 };
 REGISTER_KERNEL_BUILDER(Name("L2Loss").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuL2LossOp);
 REGISTER_KERNEL_BUILDER(Name("L2Loss").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuL2LossOp);
+
+
+// _FusedElementwise: register-resident interpreter over the fused DAG
+// (graph/optimizer.cc FuseElementwise; program format kernels/fused_ew.h).
+class GpuFusedElementwiseOp : public OpKernel {
+ public:
+  explicit GpuFusedElementwiseOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("program", &prog_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    int n_in = ctx->num_inputs();
+    const Tensor& root = ctx->input(0);
+    Tensor* out = ctx->allocate_output(0, root.shape());
+    const void* ins[8];
+    uint8_t scalar[8];
+    for (int i = 0; i < n_in; ++i) {
+      ins[i] = ctx->input(i).raw_data();
+      scalar[i] = ctx->input(i).NumElements() == 1 ? 1 : 0;
+    }
+    OP_HIP_OK(ctx, stf_fused_elementwise(
+                       DtypeCode(root.dtype()), ins, scalar, n_in,
+                       prog_.data(), (int)prog_.size(), out->raw_data(),
+                       root.NumElements(), GPU_STREAM(ctx)));
+  }
+
+ private:
+  std::vector<int64_t> prog_;
+};
+REGISTER_KERNEL_BUILDER(Name("_FusedElementwise").Device(DEVICE_GPU).TypeConstraint<float>("T"), GpuFusedElementwiseOp);
+REGISTER_KERNEL_BUILDER(Name("_FusedElementwise").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T"), GpuFusedElementwiseOp);
 
 // Fused LSTM cell pointwise (nn_kernels.hip LstmGatesKernel); one launch
 // replaces the composed cell's ~17 elementwise/split launches per timestep.

@@ -224,6 +224,91 @@ __global__ void CastF32Bf16ZeroKernel(float* __restrict__ src,
   }
 }
 
+
+// _FusedElementwise GPU interpreter (kernels/fused_ew.h program format):
+// the whole fused DAG runs register-resident — one HBM read per input
+// element, one write, nothing else touches memory.
+struct FEArgs {
+  const void* in[7];
+  uint8_t scalar_mask;
+  int n_in;
+  int n_prog;
+  int64_t prog[24];
+};
+
+template <typename T>
+__global__ void FusedEwKernel(FEArgs a, T* __restrict__ out, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < n;
+       idx += stride) {
+    float vals[32];
+    for (int i = 0; i < a.n_in; ++i) {
+      const T* p = (const T*)a.in[i];
+      vals[i] = (float)p[(a.scalar_mask >> i) & 1 ? 0 : idx];
+    }
+    for (int k = 0; k < a.n_prog; ++k) {
+      int64_t ins = a.prog[k];
+      int opx = (int)(ins & 0xff);
+      float x = vals[(ins >> 8) & 0xff];
+      float y = vals[(ins >> 16) & 0xff];
+      float r;
+      switch (opx) {
+        case 1: r = x > 0.f ? x : 0.f; break;                  // relu
+        case 2: r = x < 0.f ? 0.f : (x > 6.f ? 6.f : x); break;
+        case 3: r = 1.f / (1.f + __expf(-x)); break;           // sigmoid
+        case 4: r = tanhf(x); break;
+        case 5: r = __expf(x); break;
+        case 6: r = __logf(x); break;
+        case 7: r = log1pf(x); break;
+        case 8: r = -x; break;
+        case 9: r = sqrtf(x); break;
+        case 10: r = rsqrtf(x); break;
+        case 11: r = x * x; break;
+        case 12: r = fabsf(x); break;
+        case 13: r = log1pf(__expf(-fabsf(x))) + (x > 0.f ? x : 0.f); break;
+        case 14: r = x > 0.f ? 1.f : (x < 0.f ? -1.f : 0.f); break;
+        case 15: r = floorf(x); break;
+        case 16: r = 1.f / x; break;
+        case 64: r = x + y; break;
+        case 65: r = x - y; break;
+        case 66: r = x * y; break;
+        case 67: r = x / y; break;
+        case 68: r = fmaxf(x, y); break;
+        case 69: r = fminf(x, y); break;
+        case 70: r = (x - y) * (x - y); break;
+        case 71: r = powf(x, y); break;
+        default: r = 0.f; break;
+      }
+      vals[a.n_in + k] = r;
+    }
+    out[idx] = (T)vals[a.n_in + a.n_prog - 1];
+  }
+}
+
+extern "C" hipError_t stf_fused_elementwise(
+    int dtype, const void** inputs, const uint8_t* scalar_flags, int n_in,
+    const int64_t* prog, int n_prog, void* out, int64_t n,
+    hipStream_t stream) {
+  if (n_in > 7 || n_prog > 24) return hipErrorInvalidValue;
+  FEArgs a;
+  a.scalar_mask = 0;
+  a.n_in = n_in;
+  a.n_prog = n_prog;
+  for (int i = 0; i < n_in; ++i) {
+    a.in[i] = inputs[i];
+    if (scalar_flags[i]) a.scalar_mask |= (1 << i);
+  }
+  for (int k = 0; k < n_prog; ++k) a.prog[k] = prog[k];
+  dim3 grid = ElemwiseGrid(n, 256, 4);
+  if (dtype == 0)
+    hipLaunchKernelGGL(FusedEwKernel<float>, grid, dim3(256), 0, stream, a,
+                       (float*)out, n);
+  else
+    hipLaunchKernelGGL(FusedEwKernel<__bf16>, grid, dim3(256), 0, stream, a,
+                       (__bf16*)out, n);
+  return hipGetLastError();
+}
+
 extern "C" hipError_t stf_cast_f32_bf16_zero(void* src_f32, void* dst_bf16,
                                              int64_t n, hipStream_t stream) {
   dim3 grid = ElemwiseGrid(n / 8 + 1, 256, 1);

@@ -106,3 +106,16 @@ def test_broadcast_not_fused():
     out, names = _trace_run(z, {pa: a, pb: b})
     np.testing.assert_allclose(out, np.tanh(a) + b * 2.0, rtol=1e-5,
                                atol=1e-6)
+
+
+@pytest.mark.gpu
+def test_fused_kernel_on_gpu():
+    np.random.seed(7)
+    x = np.random.randn(4096).astype(np.float32)
+    ph = tf.placeholder(tf.float32, [4096])
+    z = tf.nn.relu(ph * tf.sigmoid(ph) + tf.constant(0.1)) * ph
+    out, names = _trace_run(z, {ph: x})
+    sig = 1 / (1 + np.exp(-x))
+    want = np.maximum(x * sig + 0.1, 0) * x
+    np.testing.assert_allclose(out, want, rtol=1e-4, atol=1e-5)
+    assert any('_fused' in n for n in names)

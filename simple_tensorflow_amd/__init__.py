@@ -18,6 +18,7 @@ from simple_tensorflow_amd.python.ops import (  # noqa: F401
     linalg_ops as _linalg_ops,
     spectral_ops as _spectral_ops,
     string_ops as _string_ops,
+    functional_ops as _functional_ops,
     clip_ops as _clip_ops,
     control_flow_ops as _control_flow_ops,
     gradients_impl as _gradients_impl,
@@ -268,6 +269,12 @@ reduce_join = _string_ops.reduce_join
 encode_base64 = _string_ops.encode_base64
 decode_base64 = _string_ops.decode_base64
 strings = _string_ops
+
+# functional ops (reference tf.map_fn / tf.foldl / tf.foldr / tf.scan)
+map_fn = _functional_ops.map_fn
+foldl = _functional_ops.foldl
+foldr = _functional_ops.foldr
+scan = _functional_ops.scan
 cumprod = _math_ops.cumprod
 tan = _math_ops.tan
 asin = _math_ops.asin

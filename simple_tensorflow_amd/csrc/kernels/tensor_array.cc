@@ -144,16 +144,36 @@ class TensorArrayReadOp : public OpKernel {
     set_expensive(false);
   }
   void Compute(OpKernelContext* ctx) override {
-    TensorArrayResource* ta =
-        GetTA(ctx, ctx->input(0).flat<std::string>()[0]);
+    const std::string handle = ctx->input(0).flat<std::string>()[0];
+    TensorArrayResource* ta = GetTA(ctx, handle);
     OP_REQUIRES(ctx, ta, errors::Internal("no resource manager"));
     int64_t idx = ctx->input(1).flat<int32_t>()[0];
-    std::lock_guard<std::mutex> l(ta->mu);
-    OP_REQUIRES(ctx, idx >= 0 && idx < (int64_t)ta->elems.size() &&
-                         ta->written[idx],
-                errors::InvalidArgument("TensorArray read of unwritten index ",
-                                        idx));
-    ctx->set_output(0, ta->elems[idx]);
+    {
+      std::lock_guard<std::mutex> l(ta->mu);
+      if (idx >= 0 && idx < (int64_t)ta->elems.size() && ta->written[idx]) {
+        ctx->set_output(0, ta->elems[idx]);
+        return;
+      }
+    }
+    // A GRADIENT array (handle "<primary>@<source>") reads zeros for
+    // positions nothing contributed to — shaped like the primary's element
+    // (reference tensor_array.h TensorAndState zero-fill semantics).
+    auto at = handle.rfind('@');
+    if (at != std::string::npos) {
+      TensorArrayResource* primary = GetTA(ctx, handle.substr(0, at));
+      if (primary) {
+        std::lock_guard<std::mutex> l(primary->mu);
+        if (idx >= 0 && idx < (int64_t)primary->elems.size() &&
+            primary->written[idx]) {
+          const Tensor& like = primary->elems[idx];
+          Tensor* out = ctx->allocate_output(0, like.shape());
+          std::memset(out->raw_data(), 0, out->TotalBytes());
+          return;
+        }
+      }
+    }
+    ctx->SetStatus(errors::InvalidArgument(
+        "TensorArray read of unwritten index ", idx));
   }
 };
 

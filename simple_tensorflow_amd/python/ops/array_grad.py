@@ -272,16 +272,71 @@ def _next_iteration_grad(op, grad):
     return [grad]
 
 
-# TensorArray reads act as sweep boundaries in the while-loop gradient
-# (values are forward-recorded; nothing upstream needs their gradient).
+# TensorArray gradients (reference tensor_array_grad.py design): each
+# primary array owns a shadow gradient array (TensorArrayGradV3 resource,
+# keyed handle@source). Ordering between the ops touching the shadow array
+# rides entirely on the FLOW-gradient scalar chain — a write-grad's read of
+# the shadow is data-dependent on the gather-grad's scatter that filled it,
+# because the flow gradient it receives descends from that scatter's output
+# flow. Works through the while-loop rematerializing backward pass:
+# TensorArray writes overwrite (idempotent re-execution), reads don't clear.
+def _grad_ta(handle, flow, dtype):
+    from simple_tensorflow_amd.python.framework.ops import apply_op
+    from simple_tensorflow_amd.python.ops import tensor_array_ops
+    g_handle, g_flow = apply_op('TensorArrayGradV3', handle, flow,
+                                source='gradients')
+    ta = tensor_array_ops.TensorArray(dtype, handle=g_handle, flow=g_flow)
+    return ta
+
+
 @RegisterGradient('TensorArrayReadV3')
 def _ta_read_grad(op, grad):
-    return [None, None, None]
+    handle, index, flow = op.inputs
+    g = _grad_ta(handle, flow, grad.dtype)
+    wflow = g.write(index, grad)._flow
+    return [None, None, wflow]
 
 
 @RegisterGradient('TensorArrayWriteV3')
-def _ta_write_grad(op, grad):
-    return [None, None, None, None]
+def _ta_write_grad(op, flow_grad):
+    handle, index, value, flow = op.inputs
+    if flow_grad is None:
+        return [None, None, None, None]
+    g = _grad_ta(handle, flow_grad, value.dtype)
+    value_grad = g.read(index)
+    if value._shape is not None:
+        value_grad.set_shape(value._shape)
+    return [None, None, value_grad, flow_grad]
+
+
+@RegisterGradient('TensorArrayGatherV3')
+def _ta_gather_grad(op, grad):
+    handle, indices, flow = op.inputs
+    g = _grad_ta(handle, flow, grad.dtype)
+    wflow = g.scatter(indices, grad)._flow
+    return [None, None, wflow]
+
+
+@RegisterGradient('TensorArrayScatterV3')
+def _ta_scatter_grad(op, flow_grad):
+    handle, indices, value, flow = op.inputs
+    if flow_grad is None:
+        return [None, None, None, None]
+    g = _grad_ta(handle, flow_grad, value.dtype)
+    value_grad = g.gather(indices)
+    if value._shape is not None:
+        value_grad.set_shape(value._shape)
+    return [None, None, value_grad, flow_grad]
+
+
+@RegisterGradient('TensorArrayV3')
+def _ta_create_grad(op, *grads):
+    return [None]  # size input; the flow chain stops at creation
+
+
+@RegisterGradient('TensorArrayGradV3')
+def _ta_grad_grad(op, *grads):
+    return [None, None]
 
 
 # ---------------------------------------------------------------------------

@@ -306,6 +306,15 @@ inline Status OpKernelConstruction::GetAttr<std::vector<int64_t>>(
   return Status::OK();
 }
 template <>
+inline Status OpKernelConstruction::GetAttr<std::vector<float>>(
+    const std::string& name, std::vector<float>* out) const {
+  auto it = def().attr.find(name);
+  if (it == def().attr.end() || it->second.kind != 'l')
+    return errors::NotFound("attr ", name);
+  *out = it->second.list.f;
+  return Status::OK();
+}
+template <>
 inline Status OpKernelConstruction::GetAttr<std::vector<std::string>>(
     const std::string& name, std::vector<std::string>* out) const {
   auto it = def().attr.find(name);

@@ -169,6 +169,13 @@ REGISTER_OP("L2Loss").Input("t: T").Output("output: T").Attr("T: " FLOATTYPES);
 // contrib/rnn lstm_ops.cc LSTMBlockCell / LSTMBlockCellGrad).
 REGISTER_OP("LSTMGates").Input("gates: T").Input("c_prev: T").Output("i: T").Output("f: T").Output("o: T").Output("ci: T").Output("cs: T").Output("co: T").Output("h: T").Attr("T: {float, bfloat16}").Attr("forget_bias: float = 1.0");
 REGISTER_OP("LSTMGatesGrad").Input("c_prev: T").Input("i: T").Input("f: T").Input("o: T").Input("ci: T").Input("co: T").Input("dh: T").Input("dcs: T").Output("dgates: T").Output("dc_prev: T").Attr("T: {float, bfloat16}");
+// image ops (reference core/ops/image_ops.cc; kernels/cpu_image.cc)
+REGISTER_OP("RGBToHSV").Input("images: T").Output("output: T").Attr("T: {float, double} = float");
+REGISTER_OP("HSVToRGB").Input("images: T").Output("output: T").Attr("T: {float, double} = float");
+REGISTER_OP("AdjustContrastv2").Input("images: T").Input("contrast_factor: float").Output("output: T").Attr("T: {float} = float");
+REGISTER_OP("NonMaxSuppression").Input("boxes: float").Input("scores: float").Input("max_output_size: int32").Output("selected_indices: int32").Attr("iou_threshold: float = 0.5");
+REGISTER_OP("NonMaxSuppressionV2").Input("boxes: float").Input("scores: float").Input("max_output_size: int32").Input("iou_threshold: float").Output("selected_indices: int32");
+REGISTER_OP("SampleDistortedBoundingBox").Input("image_size: T").Input("bounding_boxes: float").Output("begin: T").Output("size: T").Output("bboxes: float").Attr("T: {int32, int64} = int32").Attr("seed: int = 0").Attr("seed2: int = 0").Attr("min_object_covered: float = 0.1").Attr("aspect_ratio_range: list(float) = [0.75, 1.33]").Attr("area_range: list(float) = [0.05, 1.0]").Attr("max_attempts: int = 100").Attr("use_image_if_no_bounding_boxes: bool = false").SetIsStateful();
 REGISTER_OP("LRN").Input("input: T").Output("output: T").Attr("depth_radius: int = 5").Attr("bias: float = 1.0").Attr("alpha: float = 1.0").Attr("beta: float = 0.5").Attr("T: {float}");
 REGISTER_OP("InTopK").Input("predictions: float").Input("targets: T").Output("precision: bool").Attr("k: int").Attr("T: {int32, int64} = int32");
 

@@ -220,3 +220,141 @@ def decode_image(contents, channels=None, name=None):
 
     return script_ops.py_func(_dec, [convert_to_tensor(contents)],
                               dtypes.uint8, name=name)
+
+
+# ---------------------------------------------------------------------------
+# color / photometric ops (reference image_ops_impl.py)
+# ---------------------------------------------------------------------------
+def rgb_to_hsv(images, name=None):
+    return apply_op('RGBToHSV', convert_to_tensor(images,
+                                                  dtype=dtypes.float32),
+                    name=name)
+
+
+def hsv_to_rgb(images, name=None):
+    return apply_op('HSVToRGB', convert_to_tensor(images,
+                                                  dtype=dtypes.float32),
+                    name=name)
+
+
+def rgb_to_grayscale(images, name=None):
+    images = convert_to_tensor(images, dtype=dtypes.float32)
+    w = ops.constant([0.2989, 0.587, 0.114], dtypes.float32)
+    gray = math_ops.reduce_sum(images * w, -1, keep_dims=True)
+    return gray
+
+
+def grayscale_to_rgb(images, name=None):
+    from simple_tensorflow_amd.python.ops import array_ops as ao
+    images = convert_to_tensor(images, dtype=dtypes.float32)
+    return ao.concat([images, images, images], -1)
+
+
+def adjust_brightness(image, delta):
+    return convert_to_tensor(image, dtype=dtypes.float32) + \
+        convert_to_tensor(float(delta), dtype=dtypes.float32)
+
+
+def adjust_contrast(images, contrast_factor, name=None):
+    return apply_op('AdjustContrastv2',
+                    convert_to_tensor(images, dtype=dtypes.float32),
+                    convert_to_tensor(float(contrast_factor),
+                                      dtype=dtypes.float32), name=name)
+
+
+def adjust_saturation(image, saturation_factor, name=None):
+    from simple_tensorflow_amd.python.ops import array_ops as ao
+    hsv = rgb_to_hsv(image)
+    h = hsv[..., 0:1]
+    s = hsv[..., 1:2] * convert_to_tensor(float(saturation_factor))
+    s = math_ops.minimum(s, ops.constant(1.0))
+    v = hsv[..., 2:3]
+    return hsv_to_rgb(ao.concat([h, s, v], -1))
+
+
+def adjust_hue(image, delta, name=None):
+    from simple_tensorflow_amd.python.ops import array_ops as ao
+    hsv = rgb_to_hsv(image)
+    h = hsv[..., 0:1] + convert_to_tensor(float(delta))
+    h = h - math_ops.floor(h)  # wrap to [0, 1)
+    return hsv_to_rgb(ao.concat([h, hsv[..., 1:2], hsv[..., 2:3]], -1))
+
+
+def random_brightness(image, max_delta, seed=None):
+    from simple_tensorflow_amd.python.ops import random_ops
+    delta = random_ops.random_uniform([], -max_delta, max_delta, seed=seed)
+    return convert_to_tensor(image, dtype=dtypes.float32) + delta
+
+
+def random_contrast(image, lower, upper, seed=None):
+    from simple_tensorflow_amd.python.ops import random_ops
+    factor = random_ops.random_uniform([], lower, upper, seed=seed)
+    return apply_op('AdjustContrastv2',
+                    convert_to_tensor(image, dtype=dtypes.float32), factor)
+
+
+def pad_to_bounding_box(image, offset_height, offset_width, target_height,
+                        target_width):
+    from simple_tensorflow_amd.python.ops import array_ops as ao
+    image = convert_to_tensor(image)
+    h, w = int(image._shape[-3]), int(image._shape[-2])
+    pads = [[offset_height, target_height - offset_height - h],
+            [offset_width, target_width - offset_width - w], [0, 0]]
+    if len(image._shape) == 4:
+        pads = [[0, 0]] + pads
+    return ao.pad(image, pads)
+
+
+def resize_image_with_crop_or_pad(image, target_height, target_width):
+    image = convert_to_tensor(image)
+    h, w = int(image._shape[-3]), int(image._shape[-2])
+    # crop first
+    ch = min(h, target_height)
+    cw = min(w, target_width)
+    image = crop_to_bounding_box(image, (h - ch) // 2, (w - cw) // 2, ch, cw)
+    if ch == target_height and cw == target_width:
+        return image
+    return pad_to_bounding_box(image, (target_height - ch) // 2,
+                               (target_width - cw) // 2, target_height,
+                               target_width)
+
+
+def total_variation(images, name=None):
+    images = convert_to_tensor(images, dtype=dtypes.float32)
+    if len(images._shape) == 3:
+        dh = images[1:, :, :] - images[:-1, :, :]
+        dw = images[:, 1:, :] - images[:, :-1, :]
+        return math_ops.reduce_sum(math_ops.abs(dh)) + \
+            math_ops.reduce_sum(math_ops.abs(dw))
+    dh = images[:, 1:, :, :] - images[:, :-1, :, :]
+    dw = images[:, :, 1:, :] - images[:, :, :-1, :]
+    axes = [1, 2, 3]
+    return math_ops.reduce_sum(math_ops.abs(dh), axes) + \
+        math_ops.reduce_sum(math_ops.abs(dw), axes)
+
+
+def non_max_suppression(boxes, scores, max_output_size, iou_threshold=0.5,
+                        name=None):
+    return apply_op('NonMaxSuppressionV2', convert_to_tensor(boxes),
+                    convert_to_tensor(scores),
+                    convert_to_tensor(int(max_output_size),
+                                      dtype=dtypes.int32),
+                    convert_to_tensor(float(iou_threshold)), name=name)
+
+
+def sample_distorted_bounding_box(image_size, bounding_boxes,
+                                  min_object_covered=0.1,
+                                  aspect_ratio_range=None, area_range=None,
+                                  max_attempts=100,
+                                  use_image_if_no_bounding_boxes=False,
+                                  seed=None, seed2=None, name=None):
+    return apply_op(
+        'SampleDistortedBoundingBox',
+        convert_to_tensor(image_size, dtype=dtypes.int32),
+        convert_to_tensor(bounding_boxes, dtype=dtypes.float32),
+        min_object_covered=min_object_covered,
+        aspect_ratio_range=list(aspect_ratio_range or [0.75, 1.33]),
+        area_range=list(area_range or [0.05, 1.0]),
+        max_attempts=max_attempts,
+        use_image_if_no_bounding_boxes=use_image_if_no_bounding_boxes,
+        seed=seed or 0, seed2=seed2 or 0, name=name)

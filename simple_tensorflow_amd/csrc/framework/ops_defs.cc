@@ -216,6 +216,18 @@ REGISTER_OP("Imag").Input("input: T").Output("output: Tout").Attr("T: {complex64
 REGISTER_OP("Conj").Input("input: T").Output("output: T").Attr("T: {complex64, complex128} = complex64");
 REGISTER_OP("ComplexAbs").Input("x: T").Output("y: Tout").Attr("T: {complex64, complex128} = complex64").Attr("Tout: {float, double} = float");
 
+// ------------------------------ quantized ----------------------------------
+// Reference core/ops/math_ops.cc quantized section + quantize_op.cc /
+// quantized_matmul_op.cc. Carrier-typed redesign: quint8 rides DT_UINT8 and
+// qint32 rides DT_INT32 (this framework has no distinct quantized dtypes;
+// the range tensors carry the scale exactly as in the reference).
+REGISTER_OP("QuantizeV2").Input("input: float").Input("min_range: float").Input("max_range: float").Output("output: T").Output("output_min: float").Output("output_max: float").Attr("T: {uint8} = uint8").Attr("mode: string = 'MIN_COMBINED'");
+REGISTER_OP("Dequantize").Input("input: T").Input("min_range: float").Input("max_range: float").Output("output: float").Attr("T: {uint8, int32} = uint8").Attr("mode: string = 'MIN_COMBINED'");
+REGISTER_OP("QuantizedMatMul").Input("a: T1").Input("b: T2").Input("min_a: float").Input("max_a: float").Input("min_b: float").Input("max_b: float").Output("out: Toutput").Output("min_out: float").Output("max_out: float").Attr("T1: {uint8} = uint8").Attr("T2: {uint8} = uint8").Attr("Toutput: {int32} = int32").Attr("transpose_a: bool = false").Attr("transpose_b: bool = false");
+REGISTER_OP("QuantizedRelu").Input("features: T").Input("min_features: float").Input("max_features: float").Output("activations: T").Output("min_activations: float").Output("max_activations: float").Attr("T: {uint8} = uint8");
+REGISTER_OP("QuantizeDownAndShrinkRange").Input("input: Tinput").Input("input_min: float").Input("input_max: float").Output("output: out_type").Output("output_min: float").Output("output_max: float").Attr("Tinput: {int32} = int32").Attr("out_type: {uint8} = uint8");
+REGISTER_OP("RequantizationRange").Input("input: Tinput").Input("input_min: float").Input("input_max: float").Output("output_min: float").Output("output_max: float").Attr("Tinput: {int32} = int32");
+
 // --------------------------- linear algebra --------------------------------
 // Batched dense decompositions (reference core/ops/linalg_ops.cc). CPU
 // kernels in kernels/cpu_linalg.cc (LU w/ partial pivoting, Cholesky,

@@ -216,6 +216,11 @@ REGISTER_OP("Imag").Input("input: T").Output("output: Tout").Attr("T: {complex64
 REGISTER_OP("Conj").Input("input: T").Output("output: T").Attr("T: {complex64, complex128} = complex64");
 REGISTER_OP("ComplexAbs").Input("x: T").Output("y: Tout").Attr("T: {complex64, complex128} = complex64").Attr("Tout: {float, double} = float");
 
+// -------------------------------- set ops ----------------------------------
+// Reference core/ops/set_ops.cc (dense-to-dense subset; sparse outputs).
+REGISTER_OP("DenseToDenseSetOperation").Input("set1: T").Input("set2: T").Output("result_indices: int64").Output("result_values: T").Output("result_shape: int64").Attr("set_operation: string").Attr("validate_indices: bool = true").Attr("T: {int32, int64}");
+REGISTER_OP("SetSize").Input("set_indices: int64").Input("set_values: T").Input("set_shape: int64").Output("size: int32").Attr("validate_indices: bool = true").Attr("T: {int32, int64}");
+
 // ------------------------------ quantized ----------------------------------
 // Reference core/ops/math_ops.cc quantized section + quantize_op.cc /
 // quantized_matmul_op.cc. Carrier-typed redesign: quint8 rides DT_UINT8 and

@@ -130,6 +130,8 @@ hipError_t stf_lstm_gates_grad(int, const void*, const void*, const void*,
                                int64_t, int, hipStream_t);
 hipError_t stf_l2loss(int, const void*, float*, int64_t, hipStream_t);
 hipError_t stf_cast_f32_bf16_zero(void*, void*, int64_t, hipStream_t);
+hipError_t stf_block_pad(int, const void*, void*, int64_t, int64_t, int64_t,
+                         hipStream_t);
 hipError_t stf_fused_elementwise(int, const void**, const uint8_t*, int,
                                  const int64_t*, int, void*, int64_t,
                                  hipStream_t);
@@ -755,6 +757,31 @@ class GpuConv2DOp : public OpKernel {
     OP_REQUIRES_OK(ctx, GetConvGeom(x.shape(), w.shape(), strides_, padding_,
                                     &g));
     Tensor* y = ctx->allocate_output(0, TensorShape({g.N, g.P, g.Q, g.K}));
+    // C not a multiple of 8 (the C=3 stem): pad channels once so the
+    // implicit-GEMM paths (8-wide k runs) apply — the padded channels are
+    // zero in both x and w, contributing nothing.
+    Tensor xp, wp;
+    if (!g.is_1x1_s1() && (g.C % 8) != 0 && x.dtype() == DT_BFLOAT16) {
+      int64_t c8 = (g.C + 7) & ~7ll;
+      xp = ctx->allocate_temp(DT_BFLOAT16,
+                              TensorShape({g.N, g.H, g.W, c8}));
+      OP_HIP_OK(ctx, stf_block_pad(1, x.raw_data(), xp.raw_data(),
+                                   (int64_t)g.N * g.H * g.W, g.C, c8, s));
+      wp = ctx->allocate_temp(DT_BFLOAT16,
+                              TensorShape({g.R * g.S * c8, g.K}));
+      OP_HIP_OK(ctx, stf_block_pad(1, w.raw_data(), wp.raw_data(),
+                                   g.R * g.S, g.C * g.K, c8 * g.K, s));
+      GpuConvGeom gp = g;
+      gp.C = c8;
+      ComputeWith(ctx, xp, wp, gp, y, s);
+      return;
+    }
+    ComputeWith(ctx, x, w, g, y, s);
+  }
+
+ private:
+  void ComputeWith(OpKernelContext* ctx, const Tensor& x, const Tensor& w,
+                   GpuConvGeom g, Tensor* y, hipStream_t s) {
     int64_t rsc = g.RSC();
     // 1x1/s1 uses x directly as the GEMM A (ld = C) — no padding there.
     int64_t rscp = g.is_1x1_s1() ? rsc : g.RSCp();
@@ -899,6 +926,35 @@ class GpuConv2DBackpropFilterOp : public OpKernel {
                                     &g));
     Tensor* dw = ctx->allocate_output(0, f_shape);
     int64_t rsc = g.RSC();
+    static const bool no_implicit0 = getenv("STF_NO_IMPLICIT_CONV") != nullptr;
+    // C%8!=0 stem: channel-pad x, take the implicit dW path at C8, then
+    // drop the padded filter rows (their gradient is exactly zero).
+    if (!no_implicit0 && !g.is_1x1_s1() && (g.C % 8) != 0 &&
+        x.dtype() == DT_BFLOAT16 && (g.K & 7) == 0 && (g.M() & 63) == 0 &&
+        ZeroPage()) {
+      int64_t c8 = (g.C + 7) & ~7ll;
+      int64_t rsc8 = g.R * g.S * c8;
+      Tensor xp = ctx->allocate_temp(DT_BFLOAT16,
+                                     TensorShape({g.N, g.H, g.W, c8}));
+      OP_HIP_OK(ctx, stf_block_pad(1, x.raw_data(), xp.raw_data(),
+                                   (int64_t)g.N * g.H * g.W, g.C, c8, s));
+      int sk = PickSplitK(rsc8, g.K, g.M());
+      if (float* arena =
+              SplitKArena(ctx->device()->gpu_ordinal(), rsc8 * g.K, s)) {
+        OP_HIP_OK(ctx, stf_conv2d_dw_splitk(
+                           xp.raw_data(), dy.raw_data(), arena, ZeroPage(),
+                           (int)g.N, (int)g.H, (int)g.W, (int)c8, (int)g.R,
+                           (int)g.S, (int)g.sh, (int)g.sw, (int)g.ph,
+                           (int)g.pw, (int)g.P, (int)g.Q, g.K, sk, s));
+        Tensor dwp = ctx->allocate_temp(DT_BFLOAT16,
+                                        TensorShape({rsc8, g.K}));
+        OP_HIP_OK(ctx, stf_cast_f32_bf16_zero(arena, dwp.raw_data(),
+                                              rsc8 * g.K, s));
+        OP_HIP_OK(ctx, stf_block_pad(1, dwp.raw_data(), dw->raw_data(),
+                                     g.R * g.S, c8 * g.K, g.C * g.K, s));
+        return;
+      }
+    }
     // dW[RSC, K] = col[M, RSC]^T x dy[M, K]: both operands are contraction
     // (M-)major as stored, so the GEMM's K-major staging reads them directly
     // — no transpose kernels (this was 2 full passes over the im2col matrix).

@@ -95,6 +95,25 @@ __global__ void SquareSumKernel(const T* __restrict__ x,
     atomicAdd(out, 0.5f * (lds4[0] + lds4[1] + lds4[2] + lds4[3]));
 }
 
+
+// Blocked channel pad/unpad: view src as [nblocks, in_block] and dst as
+// [nblocks, out_block]; elements past in_block zero-fill (pad mode when
+// out_block > in_block, truncate when smaller). Lets C%8!=0 convolutions
+// (the ResNet/Inception stems, C=3) take the 8-channel implicit-GEMM path.
+template <typename T>
+__global__ void BlockPadKernel(const T* __restrict__ src, T* __restrict__ dst,
+                               int64_t nblocks, int64_t in_block,
+                               int64_t out_block) {
+  int64_t total = nblocks * out_block;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    int64_t b = i / out_block;
+    int64_t off = i - b * out_block;
+    dst[i] = off < in_block ? src[b * in_block + off] : (T)0;
+  }
+}
+
 // ---- reductions ----
 // full reduce to scalar: two-stage (block partials via atomics on f32)
 template <typename T, int RED>  // 0 sum, 1 max, 2 min
@@ -422,6 +441,21 @@ hipError_t stf_permute(int elem_size, const void* in, void* out, int64_t n,
   else
     hipLaunchKernelGGL((PermuteKernel<uint64_t>), grid, dim3(256), 0, stream,
                        (const uint64_t*)in, (uint64_t*)out, n, args);
+  return hipGetLastError();
+}
+
+hipError_t stf_block_pad(int dtype, const void* src, void* dst,
+                         int64_t nblocks, int64_t in_block,
+                         int64_t out_block, hipStream_t stream) {
+  dim3 grid = ElemwiseGrid(nblocks * out_block, 256, 4);
+  if (dtype == 0)
+    hipLaunchKernelGGL(BlockPadKernel<float>, grid, dim3(256), 0, stream,
+                       (const float*)src, (float*)dst, nblocks, in_block,
+                       out_block);
+  else
+    hipLaunchKernelGGL(BlockPadKernel<__bf16>, grid, dim3(256), 0, stream,
+                       (const __bf16*)src, (__bf16*)dst, nblocks, in_block,
+                       out_block);
   return hipGetLastError();
 }
 

@@ -275,6 +275,9 @@ map_fn = _functional_ops.map_fn
 foldl = _functional_ops.foldl
 foldr = _functional_ops.foldr
 scan = _functional_ops.scan
+
+from simple_tensorflow_amd.python.ops import special_math_ops as _sm  # noqa: E402
+einsum = _sm.einsum
 cumprod = _math_ops.cumprod
 tan = _math_ops.tan
 asin = _math_ops.asin

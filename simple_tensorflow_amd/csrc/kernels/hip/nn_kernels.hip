@@ -372,10 +372,9 @@ __device__ __forceinline__ BnIdx BnThreadIndex(int c, int64_t rows) {
 
 inline int BnGridV9(int c, int64_t rows) {
   int nw = c >> 3;
-  // ~384k threads: 1.5k blocks = ~24 waves/CU so the row-stride loads have
-  // enough in-flight misses to cover HBM latency (131k threads measured
-  // ~4x off bandwidth-bound at 8 waves/CU).
-  int64_t chunks = (393216 + nw - 1) / nw;
+  // ~131k threads / 512 blocks measured FASTER than 1.5k blocks (the
+  // per-block LDS zero/flush overhead beats the extra latency hiding).
+  int64_t chunks = (131072 + nw - 1) / nw;
   if (chunks > rows) chunks = rows;
   if (chunks < 1) chunks = 1;
   return (int)((nw * chunks + 255) / 256) + 1;

@@ -278,6 +278,46 @@ scan = _functional_ops.scan
 
 from simple_tensorflow_amd.python.ops import special_math_ops as _sm  # noqa: E402
 einsum = _sm.einsum
+
+from simple_tensorflow_amd.python.ops import more_ops as _more  # noqa: E402
+accumulate_n = _more.accumulate_n
+boolean_mask = _more.boolean_mask
+sequence_mask = _more.sequence_mask
+multinomial = _more.multinomial
+sparse_matmul = _more.sparse_matmul
+space_to_batch = _more.space_to_batch
+batch_to_space = _more.batch_to_space
+required_space_to_batch_paddings = _more.required_space_to_batch_paddings
+from simple_tensorflow_amd.python.ops import check_ops as _check_ops  # noqa: E402
+Assert = _check_ops.Assert
+Print = _check_ops.Print
+assert_equal = _check_ops.assert_equal
+assert_less = _check_ops.assert_less
+assert_greater = _check_ops.assert_greater
+assert_positive = _check_ops.assert_positive
+assert_non_negative = _check_ops.assert_non_negative
+add_check_numerics_ops = _check_ops.add_check_numerics_ops
+confusion_matrix = _more.confusion_matrix
+bincount = _more.bincount
+random_shuffle = _more.random_shuffle
+random_gamma = _more.random_gamma
+tables_initializer = _more.tables_initializer
+sparse_placeholder = _more.sparse_placeholder
+floormod = _math_ops.mod
+floor_div = _math_ops.floordiv if hasattr(_math_ops, 'floordiv') else None
+truncatemod = _math_ops.truncatemod if hasattr(_math_ops, 'truncatemod') \
+    else _math_ops.mod
+_nn.accumulate_n = _more.accumulate_n
+_nn.crelu = _more.crelu
+_nn.zero_fraction = _more.zero_fraction
+_nn.weighted_cross_entropy_with_logits = \
+    _more.weighted_cross_entropy_with_logits
+_nn.conv1d = _more.conv1d
+_nn.separable_conv2d = _more.separable_conv2d
+_nn.atrous_conv2d = _more.atrous_conv2d
+_nn.bincount = _more.bincount
+_nn.sufficient_statistics = _more.sufficient_statistics
+_nn.normalize_moments = _more.normalize_moments
 cumprod = _math_ops.cumprod
 tan = _math_ops.tan
 asin = _math_ops.asin
@@ -364,6 +404,9 @@ class _TrainModule(object):
     RMSPropOptimizer = _optimizer.RMSPropOptimizer
     AdagradOptimizer = _optimizer.AdagradOptimizer
     AdadeltaOptimizer = _optimizer.AdadeltaOptimizer
+    FtrlOptimizer = _optimizer.FtrlOptimizer
+    ProximalGradientDescentOptimizer = \
+        _optimizer.ProximalGradientDescentOptimizer
     create_global_step = staticmethod(_training_util.create_global_step)
     get_global_step = staticmethod(_training_util.get_global_step)
     get_or_create_global_step = staticmethod(
@@ -371,6 +414,45 @@ class _TrainModule(object):
     exponential_decay = staticmethod(_training_util.exponential_decay)
     polynomial_decay = staticmethod(_training_util.polynomial_decay)
     piecewise_constant = staticmethod(_training_util.piecewise_constant)
+    natural_exp_decay = staticmethod(_training_util.natural_exp_decay)
+    inverse_time_decay = staticmethod(_training_util.inverse_time_decay)
+
+    @staticmethod
+    def global_step(sess, global_step_tensor):
+        return int(sess.run(global_step_tensor))
+
+    @staticmethod
+    def write_graph(graph_or_graph_def, logdir, name, as_text=True):
+        import os
+        os.makedirs(logdir, exist_ok=True)
+        gd = graph_or_graph_def.as_graph_def() \
+            if hasattr(graph_or_graph_def, 'as_graph_def') \
+            else graph_or_graph_def
+        path = os.path.join(logdir, name)
+        with open(path, 'wb') as f:
+            f.write(gd if isinstance(gd, (bytes, bytearray)) else bytes(gd))
+        return path
+
+    @staticmethod
+    def summary_iterator(path):
+        from simple_tensorflow_amd.python.lib.io import tf_record
+        from simple_tensorflow_amd.python.summary import summary as _s
+        for rec in tf_record.tf_record_iterator(path):
+            yield _s.parse_event(rec) if hasattr(_s, 'parse_event') else rec
+
+    @staticmethod
+    def limit_epochs(tensor, num_epochs=None, name=None):
+        from simple_tensorflow_amd.python.training import input as _inp
+        return _inp.limit_epochs(tensor, num_epochs=num_epochs, name=name)
+
+    @staticmethod
+    def slice_input_producer(tensor_list, num_epochs=None, shuffle=True,
+                             seed=None, capacity=32, name=None):
+        from simple_tensorflow_amd.python.training import input as _inp
+        return _inp.slice_input_producer(tensor_list,
+                                         num_epochs=num_epochs,
+                                         shuffle=shuffle, seed=seed,
+                                         capacity=capacity, name=name)
     ExponentialMovingAverage = _training_util.ExponentialMovingAverage
 
     SyncReplicasOptimizer = _sync_opt.SyncReplicasOptimizer

@@ -2,6 +2,7 @@
 // (capability analog of reference core/kernels/{constant_op,variable_ops,
 // shape_ops,reshape_op,concat_op,split_op,slice_op,pad_op,transpose_op,
 // gather_op,one_hot_op,...}.cc — re-implemented compactly, no Eigen).
+#include <atomic>
 #include <cstring>
 #include <mutex>
 
@@ -89,6 +90,90 @@ REGISTER_KERNEL_BUILDER(Name("PreventGradient").Device(DEVICE_CPU), IdentityOp);
 REGISTER_KERNEL_BUILDER(Name("PreventGradient").Device(DEVICE_GPU), IdentityOp);
 REGISTER_KERNEL_BUILDER(Name("LoopCond").Device(DEVICE_CPU), IdentityOp);
 REGISTER_KERNEL_BUILDER(Name("LoopCond").Device(DEVICE_GPU).HostMemory("input").HostMemory("output"), IdentityOp);
+
+
+// ---- Assert / Print (reference logging_ops.cc) ----
+class AssertOp : public OpKernel {
+ public:
+  explicit AssertOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("summarize", &summarize_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    const Tensor& cond = ctx->input(0);
+    bool ok = true;
+    for (int64_t i = 0; i < cond.NumElements(); ++i)
+      if (!cond.flat<bool>()[i]) ok = false;
+    if (ok) return;
+    std::string msg = "assertion failed: [";
+    for (int k = 1; k < ctx->num_inputs(); ++k) {
+      const Tensor& t = ctx->input(k);
+      int64_t n = std::min<int64_t>(t.NumElements(), summarize_);
+      for (int64_t i = 0; i < n; ++i) {
+        if (i) msg += " ";
+        switch (t.dtype()) {
+          case DT_FLOAT: msg += std::to_string(t.flat<float>()[i]); break;
+          case DT_DOUBLE: msg += std::to_string(t.flat<double>()[i]); break;
+          case DT_INT32: msg += std::to_string(t.flat<int32_t>()[i]); break;
+          case DT_INT64: msg += std::to_string(t.flat<int64_t>()[i]); break;
+          case DT_BOOL:
+            msg += t.flat<bool>()[i] ? "true" : "false";
+            break;
+          case DT_STRING: msg += t.flat<std::string>()[i]; break;
+          default: msg += "?";
+        }
+      }
+      msg += k + 1 < ctx->num_inputs() ? "] [" : "]";
+    }
+    ctx->SetStatus(errors::InvalidArgument(msg));
+  }
+
+ private:
+  int64_t summarize_ = 3;
+};
+REGISTER_KERNEL_BUILDER(Name("Assert").Device(DEVICE_CPU), AssertOp);
+REGISTER_KERNEL_BUILDER(Name("Assert").Device(DEVICE_GPU).HostMemory("condition").HostMemory("data"), AssertOp);
+
+class PrintOp : public OpKernel {
+ public:
+  explicit PrintOp(OpKernelConstruction* c) : OpKernel(c) {
+    c->GetAttr("message", &message_);
+    c->GetAttr("first_n", &first_n_);
+    c->GetAttr("summarize", &summarize_);
+  }
+  void Compute(OpKernelContext* ctx) override {
+    int64_t count = ++count_;
+    if (first_n_ < 0 || count <= first_n_) {
+      std::string msg = message_;
+      for (int k = 1; k < ctx->num_inputs(); ++k) {
+        const Tensor& t = ctx->input(k);
+        msg += "[";
+        int64_t n = std::min<int64_t>(t.NumElements(), summarize_);
+        for (int64_t i = 0; i < n; ++i) {
+          if (i) msg += " ";
+          switch (t.dtype()) {
+            case DT_FLOAT: msg += std::to_string(t.flat<float>()[i]); break;
+            case DT_DOUBLE: msg += std::to_string(t.flat<double>()[i]); break;
+            case DT_INT32: msg += std::to_string(t.flat<int32_t>()[i]); break;
+            case DT_INT64: msg += std::to_string(t.flat<int64_t>()[i]); break;
+            case DT_BOOL: msg += t.flat<bool>()[i] ? "true" : "false"; break;
+            case DT_STRING: msg += t.flat<std::string>()[i]; break;
+            default: msg += "?";
+          }
+        }
+        if (t.NumElements() > summarize_) msg += "...";
+        msg += "]";
+      }
+      fprintf(stderr, "%s\n", msg.c_str());
+    }
+    ctx->set_output(0, ctx->input(0));
+  }
+
+ private:
+  std::string message_;
+  int64_t first_n_ = -1, summarize_ = 3;
+  std::atomic<int64_t> count_{0};
+};
+REGISTER_KERNEL_BUILDER(Name("Print").Device(DEVICE_CPU), PrintOp);
 
 // ------------------------------ Variable ------------------------------------
 // The kernel instance owns the storage; instances are shared across executors

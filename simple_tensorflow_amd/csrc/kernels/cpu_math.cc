@@ -636,6 +636,10 @@ class ReduceOp : public OpKernel {
   REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_CPU).TypeConstraint<int32_t>("T"), ReduceOp<int32_t, R>); \
   REGISTER_KERNEL_BUILDER(Name(OP).Device(DEVICE_CPU).TypeConstraint<int64_t>("T"), ReduceOp<int64_t, R>);
 REGISTER_KERNEL_BUILDER(Name("Sum").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), ReduceOp<bfloat16, Red::SUM>);
+// bool All/Any: bool is a 1-byte carrier, so min/max over uint8 IS
+// logical and/or (flat<T> reinterprets; op def pins dtype to bool).
+REGISTER_KERNEL_BUILDER(Name("All").Device(DEVICE_CPU), ReduceOp<uint8_t, Red::MIN>);
+REGISTER_KERNEL_BUILDER(Name("Any").Device(DEVICE_CPU), ReduceOp<uint8_t, Red::MAX>);
 REGISTER_KERNEL_BUILDER(Name("Mean").Device(DEVICE_CPU).TypeConstraint<bfloat16>("T"), ReduceOp<bfloat16, Red::MEAN>);
 REG_REDUCE("Sum", Red::SUM)
 REG_REDUCE("Mean", Red::MEAN)

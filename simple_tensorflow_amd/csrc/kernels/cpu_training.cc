@@ -127,6 +127,66 @@ class ApplyAdagradOp : public OpKernel {
 };
 REGISTER_CPU_KERNEL_FLOATS("ApplyAdagrad", ApplyAdagradOp)
 
+// FTRL-proximal (reference training_ops.cc ApplyFtrl)
+template <typename T>
+class ApplyFtrlOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    Tensor var = ctx->input(0);
+    Tensor accum = ctx->input(1);
+    Tensor linear = ctx->input(2);
+    const Tensor& grad = ctx->input(3);
+    T lr = ctx->input(4).flat<T>()[0];
+    T l1 = ctx->input(5).flat<T>()[0];
+    T l2 = ctx->input(6).flat<T>()[0];
+    T lr_power = ctx->input(7).flat<T>()[0];
+    T* v = var.flat<T>();
+    T* a = accum.flat<T>();
+    T* l = linear.flat<T>();
+    const T* g = grad.flat<T>();
+    for (int64_t i = 0; i < var.NumElements(); ++i) {
+      double new_a = (double)a[i] + (double)g[i] * g[i];
+      double sigma = (std::pow(new_a, -(double)lr_power) -
+                      std::pow((double)a[i], -(double)lr_power)) / lr;
+      l[i] = (T)((double)l[i] + g[i] - sigma * v[i]);
+      a[i] = (T)new_a;
+      double quad = std::pow(new_a, -(double)lr_power) / lr + 2.0 * l2;
+      double lv = (double)l[i];
+      if (std::fabs(lv) > (double)l1)
+        v[i] = (T)((lv > 0 ? (double)l1 - lv : -(double)l1 - lv) / quad);
+      else
+        v[i] = (T)0;
+    }
+    ctx->set_output(0, var);
+  }
+};
+REGISTER_CPU_KERNEL_FLOATS("ApplyFtrl", ApplyFtrlOp)
+
+template <typename T>
+class ApplyProximalSgdOp : public OpKernel {
+ public:
+  using OpKernel::OpKernel;
+  void Compute(OpKernelContext* ctx) override {
+    Tensor var = ctx->input(0);
+    T alpha = ctx->input(1).flat<T>()[0];
+    T l1 = ctx->input(2).flat<T>()[0];
+    T l2 = ctx->input(3).flat<T>()[0];
+    const Tensor& delta = ctx->input(4);
+    T* v = var.flat<T>();
+    const T* d = delta.flat<T>();
+    for (int64_t i = 0; i < var.NumElements(); ++i) {
+      double prox = (double)v[i] - (double)alpha * d[i];
+      double shrink = std::fabs(prox) - (double)alpha * l1;
+      if (shrink < 0) shrink = 0;
+      double sgn = prox > 0 ? 1.0 : (prox < 0 ? -1.0 : 0.0);
+      v[i] = (T)(sgn * shrink / (1.0 + (double)alpha * l2));
+    }
+    ctx->set_output(0, var);
+  }
+};
+REGISTER_CPU_KERNEL_FLOATS("ApplyProximalGradientDescent", ApplyProximalSgdOp)
+
 template <typename T>
 class ApplyAdadeltaOp : public OpKernel {
  public:

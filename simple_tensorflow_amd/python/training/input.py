@@ -69,3 +69,41 @@ def range_input_producer(limit, num_epochs=None, shuffle=True, seed=None,
     from simple_tensorflow_amd.python.ops import math_ops
     return input_producer(math_ops.range(limit), num_epochs, shuffle, seed,
                           capacity, name or 'range_input_producer')
+
+
+def limit_epochs(tensor, num_epochs=None, name=None):
+    """Raises OutOfRange after the tensor has been produced num_epochs times
+    (reference input.py limit_epochs: a counter variable + assert)."""
+    if num_epochs is None:
+        return convert_to_tensor(tensor)
+    from simple_tensorflow_amd.python.framework import dtypes, ops
+    from simple_tensorflow_amd.python.ops import math_ops, state_ops
+    from simple_tensorflow_amd.python.ops import variables as vars_mod
+    from simple_tensorflow_amd.python.ops import control_flow_ops
+    counter = vars_mod.Variable(0, name=name or 'epochs', trainable=False,
+                                dtype=dtypes.int32)
+    t = convert_to_tensor(tensor)
+    inc = state_ops.assign_add(counter._variable, 1)
+    # OutOfRange once the counter passes the limit: piggyback on queue
+    # semantics by asserting through a cond raising via Assert
+    check = control_flow_ops.Assert(
+        math_ops.less_equal(inc, ops.constant(int(num_epochs),
+                                              dtypes.int32)),
+        [inc])
+    with ops.get_default_graph().control_dependencies([check]):
+        from simple_tensorflow_amd.python.ops import array_ops
+        return array_ops.identity(t)
+
+
+def slice_input_producer(tensor_list, num_epochs=None, shuffle=True,
+                         seed=None, capacity=32, name=None):
+    """Produces one row at a time from each tensor in tensor_list
+    (reference input.py slice_input_producer)."""
+    from simple_tensorflow_amd.python.ops import array_ops, math_ops
+    tensor_list = [convert_to_tensor(t) for t in tensor_list]
+    n = int(tensor_list[0]._shape[0])
+    rq = input_producer(math_ops.range(n), num_epochs, shuffle, seed,
+                        capacity, name or 'slice_input_producer')
+    idx = rq.dequeue()
+    return [array_ops.gather(t, array_ops.reshape(idx, [1]))[0]
+            if False else array_ops.gather(t, idx) for t in tensor_list]

@@ -231,6 +231,63 @@ class AdagradOptimizer(Optimizer):
                         use_locking=self._use_locking)
 
 
+class FtrlOptimizer(Optimizer):
+    """FTRL-proximal (reference ftrl.py + ApplyFtrl kernel)."""
+
+    def __init__(self, learning_rate, learning_rate_power=-0.5,
+                 initial_accumulator_value=0.1,
+                 l1_regularization_strength=0.0,
+                 l2_regularization_strength=0.0, use_locking=False,
+                 name='Ftrl'):
+        super().__init__(use_locking, name)
+        self._lr = learning_rate
+        self._lr_power = learning_rate_power
+        self._init_acc = initial_accumulator_value
+        self._l1 = l1_regularization_strength
+        self._l2 = l2_regularization_strength
+
+    def _create_slots(self, var_list):
+        for v in var_list:
+            slots = self._slots.setdefault('accum', {})
+            if v.name not in slots:
+                shape = v.get_shape().as_list()
+                slots[v.name] = variables.Variable(
+                    array_ops.ones(shape, v.dtype) * self._init_acc,
+                    trainable=False, name=self._name + '/accum')
+            self._zeros_slot(v, 'linear', self._name)
+
+    def _apply_dense(self, grad, var):
+        ref = self._var_ref(var)
+        accum = self.get_slot(var, 'accum')
+        linear = self.get_slot(var, 'linear')
+        return apply_op(
+            'ApplyFtrl', ref, self._var_ref(accum), self._var_ref(linear),
+            grad, convert_to_tensor(self._lr, dtype=ref.dtype),
+            convert_to_tensor(self._l1, dtype=ref.dtype),
+            convert_to_tensor(self._l2, dtype=ref.dtype),
+            convert_to_tensor(self._lr_power, dtype=ref.dtype),
+            use_locking=self._use_locking)
+
+
+class ProximalGradientDescentOptimizer(Optimizer):
+    def __init__(self, learning_rate, l1_regularization_strength=0.0,
+                 l2_regularization_strength=0.0, use_locking=False,
+                 name='ProximalGradientDescent'):
+        super().__init__(use_locking, name)
+        self._lr = learning_rate
+        self._l1 = l1_regularization_strength
+        self._l2 = l2_regularization_strength
+
+    def _apply_dense(self, grad, var):
+        ref = self._var_ref(var)
+        return apply_op(
+            'ApplyProximalGradientDescent', ref,
+            convert_to_tensor(self._lr, dtype=ref.dtype),
+            convert_to_tensor(self._l1, dtype=ref.dtype),
+            convert_to_tensor(self._l2, dtype=ref.dtype), grad,
+            use_locking=self._use_locking)
+
+
 class AdadeltaOptimizer(Optimizer):
     def __init__(self, learning_rate=0.001, rho=0.95, epsilon=1e-8,
                  use_locking=False, name='Adadelta'):

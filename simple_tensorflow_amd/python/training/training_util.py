@@ -51,6 +51,31 @@ def polynomial_decay(learning_rate, global_step, decay_steps,
         float(end_learning_rate), name=name)
 
 
+def natural_exp_decay(learning_rate, global_step, decay_steps, decay_rate,
+                      staircase=False, name=None):
+    from simple_tensorflow_amd.python.ops import math_ops
+    from simple_tensorflow_amd.python.framework import dtypes, ops as fops
+    lr = fops.convert_to_tensor(learning_rate, dtype=dtypes.float32)
+    step = math_ops.cast(global_step, dtypes.float32)
+    p = step / float(decay_steps)
+    if staircase:
+        p = math_ops.floor(p)
+    return lr * math_ops.exp(fops.constant(-float(decay_rate)) * p)
+
+
+def inverse_time_decay(learning_rate, global_step, decay_steps, decay_rate,
+                       staircase=False, name=None):
+    from simple_tensorflow_amd.python.ops import math_ops
+    from simple_tensorflow_amd.python.framework import dtypes, ops as fops
+    lr = fops.convert_to_tensor(learning_rate, dtype=dtypes.float32)
+    step = math_ops.cast(global_step, dtypes.float32)
+    p = step / float(decay_steps)
+    if staircase:
+        p = math_ops.floor(p)
+    denom = fops.constant(1.0) + fops.constant(float(decay_rate)) * p
+    return lr * math_ops.reciprocal(denom)
+
+
 def piecewise_constant(x, boundaries, values, name=None):
     xf = math_ops.cast(x._as_graph_element()
                        if hasattr(x, '_as_graph_element') else x,

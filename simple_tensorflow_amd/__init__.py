@@ -303,6 +303,11 @@ random_shuffle = _more.random_shuffle
 random_gamma = _more.random_gamma
 tables_initializer = _more.tables_initializer
 sparse_placeholder = _more.sparse_placeholder
+from simple_tensorflow_amd.python.ops import parsing_ops as _parsing  # noqa: E402
+parse_example = _parsing.parse_example
+parse_single_example = _parsing.parse_single_example
+FixedLenFeature = _parsing.FixedLenFeature
+VarLenFeature = _parsing.VarLenFeature
 floormod = _math_ops.mod
 floor_div = _math_ops.floordiv if hasattr(_math_ops, 'floordiv') else None
 truncatemod = _math_ops.truncatemod if hasattr(_math_ops, 'truncatemod') \
@@ -456,6 +461,14 @@ class _TrainModule(object):
     ExponentialMovingAverage = _training_util.ExponentialMovingAverage
 
     SyncReplicasOptimizer = _sync_opt.SyncReplicasOptimizer
+
+    from simple_tensorflow_amd.python.lib import example_pb as _ex
+    Example = _ex.Example
+    Features = _ex.Features
+    Feature = _ex.Feature
+    BytesList = _ex.BytesList
+    FloatList = _ex.FloatList
+    Int64List = _ex.Int64List
 
     @staticmethod
     def Supervisor(*a, **kw):

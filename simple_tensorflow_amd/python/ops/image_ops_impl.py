@@ -205,10 +205,38 @@ def decode_jpeg(contents, channels=0, name=None, **kw):
                               dtypes.uint8, name=name)
 
 
+def decode_gif(contents, name=None):
+    """GIF decode to uint8 [num_frames, h, w, 3] via the python codec
+    (lib/io/gif_codec.py; reference DecodeGif / gif_io.cc analog)."""
+    from simple_tensorflow_amd.python.lib.io import gif_codec
+    from simple_tensorflow_amd.python.ops import script_ops
+
+    def _dec(blob):
+        b = blob if isinstance(blob, bytes) else bytes(blob)
+        return gif_codec.decode_gif(b)
+
+    return script_ops.py_func(_dec, [convert_to_tensor(contents)],
+                              dtypes.uint8, name=name)
+
+
+def encode_gif(images, name=None):
+    from simple_tensorflow_amd.python.lib.io import gif_codec
+    from simple_tensorflow_amd.python.ops import script_ops
+    import numpy as np
+
+    def _enc(arr):
+        return gif_codec.encode_gif(np.asarray(arr, dtype=np.uint8))
+
+    return script_ops.py_func(_enc, [convert_to_tensor(images)],
+                              dtypes.string, name=name)
+
+
 def decode_image(contents, channels=None, name=None):
-    """Dispatches on magic bytes to decode_png / decode_jpeg (reference
-    decode_image)."""
-    from simple_tensorflow_amd.python.lib.io import jpeg_codec, png_codec
+    """Dispatches on magic bytes to decode_png / decode_jpeg / decode_gif
+    (reference decode_image). GIF yields [frames, h, w, 3] like the
+    reference."""
+    from simple_tensorflow_amd.python.lib.io import (gif_codec, jpeg_codec,
+                                                     png_codec)
     from simple_tensorflow_amd.python.ops import script_ops
     import numpy as np
 
@@ -216,6 +244,8 @@ def decode_image(contents, channels=None, name=None):
         b = blob if isinstance(blob, bytes) else bytes(blob)
         if b[:2] == b'\xff\xd8':
             return jpeg_codec.decode_jpeg(b)
+        if b[:4] == b'GIF8':
+            return gif_codec.decode_gif(b)
         return png_codec.decode_png(b)
 
     return script_ops.py_func(_dec, [convert_to_tensor(contents)],

@@ -130,6 +130,9 @@ REGISTER_OP("Cumsum").Input("x: T").Input("axis: Tidx").Output("out: T").Attr("e
 // ------------------------------- nn_ops ------------------------------------
 REGISTER_OP("Conv2D").Input("input: T").Input("filter: T").Output("output: T").Attr("T: " FLOATTYPES).Attr("strides: list(int)").Attr("use_cudnn_on_gpu: bool = true").Attr("padding: string").Attr("data_format: string = 'NHWC'");
 REGISTER_OP("Conv2DBackpropInput").Input("input_sizes: int32").Input("filter: T").Input("out_backprop: T").Output("output: T").Attr("T: " FLOATTYPES).Attr("strides: list(int)").Attr("use_cudnn_on_gpu: bool = true").Attr("padding: string").Attr("data_format: string = 'NHWC'");
+// Fused variant produced by gradient aggregation: output = conv-backprop-dx
+// + side (residual-gradient accumulation folded into the GEMM epilogue).
+REGISTER_OP("Conv2DBackpropInputAdd").Input("input_sizes: int32").Input("filter: T").Input("out_backprop: T").Input("side: T").Output("output: T").Attr("T: " FLOATTYPES).Attr("strides: list(int)").Attr("padding: string").Attr("data_format: string = 'NHWC'");
 REGISTER_OP("Conv2DBackpropFilter").Input("input: T").Input("filter_sizes: int32").Input("out_backprop: T").Output("output: T").Attr("T: " FLOATTYPES).Attr("strides: list(int)").Attr("use_cudnn_on_gpu: bool = true").Attr("padding: string").Attr("data_format: string = 'NHWC'");
 REGISTER_OP("BiasAdd").Input("value: T").Input("bias: T").Output("output: T").Attr("T: " NUMTYPES).Attr("data_format: string = 'NHWC'");
 REGISTER_OP("BiasAddGrad").Input("out_backprop: T").Output("output: T").Attr("T: " NUMTYPES).Attr("data_format: string = 'NHWC'");

@@ -94,7 +94,8 @@ REGISTER_CPU_KERNEL_FLOATS("Conv2D", Conv2DOp)
 template <typename T>
 class Conv2DBackpropInputOp : public OpKernel {
  public:
-  explicit Conv2DBackpropInputOp(OpKernelConstruction* ctx) : OpKernel(ctx) {
+  explicit Conv2DBackpropInputOp(OpKernelConstruction* ctx, bool side = false)
+      : OpKernel(ctx), side_(side) {
     ctx->GetAttr("strides", &strides_);
     ctx->GetAttr("padding", &padding_);
   }
@@ -110,7 +111,17 @@ class Conv2DBackpropInputOp : public OpKernel {
     const T* w = filter.flat<T>();
     const T* g = dy.flat<T>();
     T* dx = out->flat<T>();
-    std::memset(dx, 0, out->TotalBytes());
+    if (side_) {
+      // Conv2DBackpropInputAdd: accumulate on top of the side gradient
+      // instead of zero (the add the gradient aggregator would have run).
+      const Tensor& sd = ctx->input(3);
+      OP_REQUIRES(ctx, sd.NumElements() == out->NumElements(),
+                  errors::InvalidArgument(
+                      "Conv2DBackpropInputAdd: side shape mismatch"));
+      std::memcpy(dx, sd.flat<T>(), out->TotalBytes());
+    } else {
+      std::memset(dx, 0, out->TotalBytes());
+    }
     for (int64_t n = 0; n < p.N; ++n)
       for (int64_t ph = 0; ph < p.P; ++ph)
         for (int64_t pw = 0; pw < p.Q; ++pw) {
@@ -137,8 +148,17 @@ class Conv2DBackpropInputOp : public OpKernel {
  private:
   std::vector<int64_t> strides_;
   std::string padding_;
+  bool side_;
 };
 REGISTER_CPU_KERNEL_FLOATS("Conv2DBackpropInput", Conv2DBackpropInputOp)
+
+template <typename T>
+class Conv2DBackpropInputAddOp : public Conv2DBackpropInputOp<T> {
+ public:
+  explicit Conv2DBackpropInputAddOp(OpKernelConstruction* ctx)
+      : Conv2DBackpropInputOp<T>(ctx, true) {}
+};
+REGISTER_CPU_KERNEL_FLOATS("Conv2DBackpropInputAdd", Conv2DBackpropInputAddOp)
 
 template <typename T>
 class Conv2DBackpropFilterOp : public OpKernel {

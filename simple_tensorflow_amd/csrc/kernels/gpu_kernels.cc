@@ -26,6 +26,10 @@ hipError_t stf_gemm_bf16_splitk(const void*, const void*, void*, int64_t,
                                 int64_t, int64_t, int64_t, int64_t, int, int,
                                 int, hipStream_t);
 int stf_gemm_bf16_8ph_ok(int64_t, int64_t, int64_t);
+hipError_t stf_gemm_bf16_8ph_side(const void*, const void*, void*,
+                                  const void*, const void*, int64_t, int64_t,
+                                  int64_t, int64_t, int64_t, int, int,
+                                  hipStream_t);
 hipError_t stf_lrn_fwd(const void*, void*, int64_t, int, int, float, float,
                        float, hipStream_t);
 hipError_t stf_conv2d_fwd_8ph(const void*, const void*, void*, const void*,
@@ -871,7 +875,8 @@ REGISTER_KERNEL_BUILDER(Name("Conv2D").Device(DEVICE_GPU).TypeConstraint<bfloat1
 
 class GpuConv2DBackpropInputOp : public OpKernel {
  public:
-  explicit GpuConv2DBackpropInputOp(OpKernelConstruction* c) : OpKernel(c) {
+  explicit GpuConv2DBackpropInputOp(OpKernelConstruction* c, bool side = false)
+      : OpKernel(c), side_(side) {
     c->GetAttr("strides", &strides_);
     c->GetAttr("padding", &padding_);
   }
@@ -885,11 +890,32 @@ class GpuConv2DBackpropInputOp : public OpKernel {
     OP_REQUIRES_OK(ctx, GetConvGeom(x_shape, w.shape(), strides_, padding_,
                                     &g));
     Tensor* dx = ctx->allocate_output(0, x_shape);
+    const void* side = nullptr;
+    if (side_) {
+      const Tensor& sd = ctx->input(3);
+      OP_REQUIRES(ctx, sd.NumElements() == dx->NumElements(),
+                  errors::InvalidArgument(
+                      "Conv2DBackpropInputAdd: side shape mismatch"));
+      side = sd.raw_data();
+    }
     // dcol[M, RSC] = dy[M, K] x w[RSC, K]^T  (filter layout is already NT B)
     if (g.is_1x1_s1()) {
+      if (side && stf_gemm_bf16_8ph_ok(g.M(), g.RSC(), g.K)) {
+        // residual-gradient accumulation fused into the GEMM epilogue: the
+        // separate full-tensor add pass (and its 3x-bandwidth cost) vanishes
+        OP_HIP_OK(ctx, stf_gemm_bf16_8ph_side(
+                           dy.raw_data(), w.raw_data(), dx->raw_data(),
+                           nullptr, side, g.M(), g.RSC(), g.K, g.K, g.K, 1, 0,
+                           s));
+        return;
+      }
       OP_HIP_OK(ctx, stf_gemm_bf16_nt(dy.raw_data(), w.raw_data(),
                                       dx->raw_data(), nullptr, g.M(), g.RSC(),
                                       g.K, 0.f, 1, 0, s));
+      if (side)
+        OP_HIP_OK(ctx, stf_binary(B_ADD, DtypeCode(DT_BFLOAT16),
+                                  dx->raw_data(), side, dx->raw_data(),
+                                  dx->NumElements(), s));
       return;
     }
     Tensor dcol = ctx->allocate_temp(DT_BFLOAT16,
@@ -901,13 +927,24 @@ class GpuConv2DBackpropInputOp : public OpKernel {
                                    (int)g.H, (int)g.W, (int)g.C, (int)g.R,
                                    (int)g.S, (int)g.sh, (int)g.sw, (int)g.ph,
                                    (int)g.pw, (int)g.P, (int)g.Q, s));
+    if (side)
+      OP_HIP_OK(ctx, stf_binary(B_ADD, DtypeCode(DT_BFLOAT16), dx->raw_data(),
+                                side, dx->raw_data(), dx->NumElements(), s));
   }
 
  private:
   std::vector<int64_t> strides_;
   std::string padding_;
+  bool side_;
 };
 REGISTER_KERNEL_BUILDER(Name("Conv2DBackpropInput").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T").HostMemory("input_sizes"), GpuConv2DBackpropInputOp);
+
+class GpuConv2DBackpropInputAddOp : public GpuConv2DBackpropInputOp {
+ public:
+  explicit GpuConv2DBackpropInputAddOp(OpKernelConstruction* c)
+      : GpuConv2DBackpropInputOp(c, true) {}
+};
+REGISTER_KERNEL_BUILDER(Name("Conv2DBackpropInputAdd").Device(DEVICE_GPU).TypeConstraint<bfloat16>("T").HostMemory("input_sizes"), GpuConv2DBackpropInputAddOp);
 
 class GpuConv2DBackpropFilterOp : public OpKernel {
  public:

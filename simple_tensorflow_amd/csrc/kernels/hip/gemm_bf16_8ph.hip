@@ -63,8 +63,8 @@ __device__ __forceinline__ void Stage8(const AG& ag, int64_t row0, int64_t k0,
 template <int NR, bool OUT_BF16, bool FUSE_RELU, class AG>
 __launch_bounds__(512) __global__ void Gemm256x8Ph(
     AG ag, const uint16_t* __restrict__ B, void* __restrict__ C,
-    const float* __restrict__ bias, int64_t M, int64_t N, int64_t K,
-    int64_t ldb) {
+    const float* __restrict__ bias, const uint16_t* __restrict__ side,
+    int64_t M, int64_t N, int64_t K, int64_t ldb) {
   constexpr int kBN = 64 * NR;
   constexpr int kABytes = kBM * kBK * 2;       // 32 KiB per slot
   constexpr int kBBytes = kBN * kBK * 2;       // per slot
@@ -237,8 +237,19 @@ __launch_bounds__(512) __global__ void Gemm256x8Ph(
       int sidx = (p * kThreads + tid) * 8;
       int r = sidx / kBN;
       int c = sidx % kBN;
-      *(ulong2*)((uint16_t*)C + (m0 + r) * N + n0 + c) =
-          *(ulong2*)(cbuf + sidx);
+      int64_t off = (m0 + r) * N + n0 + c;
+      if (side) {
+        // fused elementwise side add (residual-gradient accumulation):
+        // one extra 16B load replaces a whole separate add pass
+        __bf16 v[8], sv[8];
+        *(ulong2*)v = *(ulong2*)(cbuf + sidx);
+        *(ulong2*)sv = *(const ulong2*)(side + off);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) v[e] = (__bf16)((float)v[e] + (float)sv[e]);
+        *(ulong2*)((uint16_t*)C + off) = *(ulong2*)v;
+      } else {
+        *(ulong2*)((uint16_t*)C + off) = *(ulong2*)(cbuf + sidx);
+      }
     }
     return;
   }
@@ -252,6 +263,7 @@ __launch_bounds__(512) __global__ void Gemm256x8Ph(
       for (int rgi = 0; rgi < 4; ++rgi) {
         int64_t row = m0 + wr * 128 + i * 16 + (lane >> 4) * 4 + rgi;
         float v = acc[i][j][rgi] + bv;
+        if (side) v += (float)((const __bf16*)side)[row * N + col];
         if (FUSE_RELU) v = v > 0.f ? v : 0.f;
         ((float*)C)[row * N + col] = v;
       }
@@ -268,15 +280,16 @@ int stf_gemm_bf16_8ph_ok(int64_t M, int64_t N, int64_t K) {
   return (M % 256 == 0) && (N % 64 == 0) && (K % 64 == 0) && K > 0;
 }
 
-hipError_t stf_gemm_bf16_8ph(const void* A, const void* B, void* C,
-                             const void* bias_f32, int64_t M, int64_t N,
-                             int64_t K, int64_t lda, int64_t ldb,
-                             int out_bf16, int fuse_relu,
-                             hipStream_t stream) {
+hipError_t stf_gemm_bf16_8ph_side(const void* A, const void* B, void* C,
+                                  const void* bias_f32, const void* side_bf16,
+                                  int64_t M, int64_t N, int64_t K, int64_t lda,
+                                  int64_t ldb, int out_bf16, int fuse_relu,
+                                  hipStream_t stream) {
   if (!stf_gemm_bf16_8ph_ok(M, N, K)) return hipErrorInvalidValue;
   LinearAG ag{(const uint16_t*)A, lda};
   const uint16_t* b = (const uint16_t*)B;
   const float* bias = (const float*)bias_f32;
+  const uint16_t* side = (const uint16_t*)side_bf16;
   int nr = (N % 256 == 0) ? 4 : (N % 128 == 0) ? 2 : 1;
   int64_t blocks = (M / kBM) * (N / (64 * nr));
   // LDS: A ring 64 KiB + B ring 2 * (BN*64*2) bytes.
@@ -284,7 +297,7 @@ hipError_t stf_gemm_bf16_8ph(const void* A, const void* B, void* C,
 #define STF_8PH_LAUNCH(NR_, OB, FR)                                          \
   hipLaunchKernelGGL((Gemm256x8Ph<NR_, OB, FR, LinearAG>),                   \
                      dim3((uint32_t)blocks), dim3(kThreads), shmem, stream,  \
-                     ag, b, C, bias, M, N, K, ldb)
+                     ag, b, C, bias, side, M, N, K, ldb)
 #define STF_8PH_NR(NR_)                                                      \
   do {                                                                       \
     if (out_bf16) {                                                          \
@@ -301,6 +314,15 @@ hipError_t stf_gemm_bf16_8ph(const void* A, const void* B, void* C,
 #undef STF_8PH_NR
 #undef STF_8PH_LAUNCH
   return hipGetLastError();
+}
+
+hipError_t stf_gemm_bf16_8ph(const void* A, const void* B, void* C,
+                             const void* bias_f32, int64_t M, int64_t N,
+                             int64_t K, int64_t lda, int64_t ldb,
+                             int out_bf16, int fuse_relu,
+                             hipStream_t stream) {
+  return stf_gemm_bf16_8ph_side(A, B, C, bias_f32, nullptr, M, N, K, lda,
+                                ldb, out_bf16, fuse_relu, stream);
 }
 
 // Implicit-GEMM Conv2D forward: y[M=NPQ, N=Cout] = im2col(x) * wt[Cout,rscp]^T
@@ -335,7 +357,7 @@ hipError_t stf_conv2d_fwd_8ph(const void* x, const void* wt, void* y,
 #define STF_C8_LAUNCH(NR_, OB, FR)                                           \
   hipLaunchKernelGGL((Gemm256x8Ph<NR_, OB, FR, ConvAG>),                     \
                      dim3((uint32_t)blocks), dim3(kThreads), shmem, stream,  \
-                     ag, b, y, bias, M, cout, rscp, rscp)
+                     ag, b, y, bias, nullptr, M, cout, rscp, rscp)
 #define STF_C8_NR(NR_)                                                       \
   do {                                                                       \
     if (out_bf16) {                                                          \

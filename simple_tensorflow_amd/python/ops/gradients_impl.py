@@ -4,6 +4,8 @@ Analog of the reference's python/ops/gradients_impl.py (gradients:376):
 reverse BFS from ys to xs, per-op grad functions from the registry, AddN
 aggregation of fan-in gradients.
 """
+import os
+
 from simple_tensorflow_amd.python.framework import dtypes, ops
 from simple_tensorflow_amd.python.ops import array_ops, math_ops
 
@@ -99,6 +101,43 @@ def _while_grad(record, exit_grads):
         if not isinstance(res, (list, tuple)):
             res = [res]
         return list(res[1:1 + n]) + list(res[1 + n:])
+
+
+def _aggregate(lst):
+    """Sum a fan-in gradient list. Two-term sums where one side is a
+    freshly built Conv2DBackpropInput (the ResNet residual-block pattern:
+    dx_branch + d_shortcut) are folded into Conv2DBackpropInputAdd so the
+    add runs in the conv GEMM's epilogue instead of a separate
+    full-tensor pass."""
+    if len(lst) == 1:
+        return lst[0]
+    if len(lst) == 2 and not os.environ.get('STF_NO_CONV_DX_FUSE'):
+        fused = (_fuse_conv_dx_add(lst[0], lst[1]) or
+                 _fuse_conv_dx_add(lst[1], lst[0]))
+        if fused is not None:
+            return fused
+    return math_ops.add_n(lst)
+
+
+def _fuse_conv_dx_add(conv_t, side):
+    op = getattr(conv_t, 'op', None)
+    if op is None or getattr(op, 'type', None) != 'Conv2DBackpropInput':
+        return None
+    if conv_t is side or conv_t.dtype != side.dtype:
+        return None
+    cs, ss = conv_t._shape, side._shape
+    if (cs is None or ss is None or any(d is None for d in cs) or
+            list(cs) != list(ss)):
+        return None
+    # only when this partial gradient has no other use: otherwise the
+    # rewrite would duplicate the conv GEMM
+    if len(conv_t.consumers()) != 0:
+        return None
+    out = ops.apply_op(
+        'Conv2DBackpropInputAdd', op.inputs[0], op.inputs[1], op.inputs[2],
+        side, strides=op.get_attr('strides'), padding=op.get_attr('padding'))
+    out.set_shape(cs)
+    return out
 
 
 def gradients(ys, xs, grad_ys=None, name='gradients',
@@ -282,9 +321,8 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
                 out_grads = []
                 for t in op.outputs:
                     lst = grads.get(t)
-                    out_grads.append(None if not lst else
-                                     (lst[0] if len(lst) == 1
-                                      else math_ops.add_n(lst)))
+                    out_grads.append(None if not lst
+                                     else _aggregate(lst))
                 if all(og is None for og in out_grads):
                     # Call not on any differentiated path: still propagate
                     # None to inputs so upstream producers become ready.
@@ -309,9 +347,8 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
                 exit_grads = []
                 for t in op.outputs:
                     lst = grads.get(t)
-                    exit_grads.append(None if not lst else
-                                      (lst[0] if len(lst) == 1
-                                       else math_ops.add_n(lst)))
+                    exit_grads.append(None if not lst
+                                      else _aggregate(lst))
                 if all(eg is None for eg in exit_grads):
                     # Loop not on any differentiated path: still propagate
                     # None so upstream producers become ready.
@@ -326,8 +363,7 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
                 lst = grads.get(t)
                 if lst:
                     has_any = True
-                    out_grads.append(lst[0] if len(lst) == 1
-                                     else math_ops.add_n(lst))
+                    out_grads.append(_aggregate(lst))
                 else:
                     out_grads.append(None)
             in_grads = [None] * len(op.inputs)
@@ -362,10 +398,8 @@ def gradients(ys, xs, grad_ys=None, name='gradients',
             if lst is None:
                 # maybe gradient was recorded against the variable snapshot
                 result.append(None)
-            elif len(lst) == 1:
-                result.append(lst[0])
             else:
-                result.append(math_ops.add_n(lst))
+                result.append(_aggregate(lst))
         return result
 
 

@@ -155,7 +155,8 @@ def _lrn_grad(op, grad):
 
 for _op in ('MaxPoolGrad', 'AvgPoolGrad', 'ReluGrad', 'Relu6Grad',
             'SoftplusGrad', 'EluGrad', 'BiasAddGrad', 'Conv2DBackpropInput',
-            'Conv2DBackpropFilter', 'FusedBatchNormGrad', 'LSTMGatesGrad'):
+            'Conv2DBackpropFilter', 'Conv2DBackpropInputAdd',
+            'FusedBatchNormGrad', 'LSTMGatesGrad'):
     ops.NoGradient(_op)
 
 

@@ -296,3 +296,49 @@ def test_training_step_decreases_loss_mlp():
         l1 = s.run(loss)
     assert np.isfinite(l0) and np.isfinite(l1)
     assert l1 < 0.7 * l0
+
+
+# ---------------------------------------------------------------------------
+# Conv2DBackpropInputAdd: residual-gradient add fused into the dx GEMM
+# epilogue (8-phase path) — vs the unfused AddN graph and an fp32 reference
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize('cin,cout,hw,fused_path', [
+    (64, 128, 16, True),     # M=2*16*16=512 %256, 8ph-eligible
+    (64, 72, 16, False),     # cout%8==0 but N=cin... K=72 not %64 -> fallback
+])
+def test_conv_dx_side_add_fused(cin, cout, hw, fused_path):
+    import os
+    rng = np.random.RandomState(11)
+    n = 2
+    xv = rng.randn(n, hw, hw, cin).astype(np.float32) * 0.5
+    wv = rng.randn(1, 1, cin, cout).astype(np.float32) * 0.1
+    bv = rng.randn(n, hw, hw, cin).astype(np.float32) * 0.5
+
+    def build():
+        x = tf.constant(_bf16(xv), dtype=tf.bfloat16)
+        w = tf.constant(_bf16(wv), dtype=tf.bfloat16)
+        b = tf.constant(_bf16(bv), dtype=tf.bfloat16)
+        y = tf.nn.conv2d(x, w, [1, 1, 1, 1], 'SAME')
+        loss = tf.reduce_sum(tf.cast(y, tf.float32)) + \
+            tf.reduce_sum(tf.cast(x * b, tf.float32))
+        return tf.gradients(loss, [x])[0]
+
+    g = build()
+    assert g.op.type == 'Conv2DBackpropInputAdd'
+    with _sess() as s:
+        out = s.run(tf.cast(g, tf.float32))
+
+    os.environ['STF_NO_CONV_DX_FUSE'] = '1'
+    try:
+        tf.reset_default_graph()
+        g2 = build()
+        assert g2.op.type in ('AddN', 'Add')
+        with _sess() as s:
+            out2 = s.run(tf.cast(g2, tf.float32))
+    finally:
+        del os.environ['STF_NO_CONV_DX_FUSE']
+
+    # fused epilogue does the identical bf16+bf16 add -> bit-exact match
+    np.testing.assert_array_equal(out, out2)
+    ref = _bf16(wv[0, 0]).sum(axis=1)[None, None, None, :] + _bf16(bv)
+    np.testing.assert_allclose(out, ref, rtol=3e-2, atol=3e-2 * cout)

@@ -68,3 +68,59 @@ def test_random_not_merged():
     d = tf.reduce_max(tf.abs(a - b))
     with tf.Session() as s:
         assert s.run(d) > 0.0
+
+
+def test_conv_dx_add_fusion_cpu():
+    """Gradient aggregation folds AddN(Conv2DBackpropInput, side) into
+    Conv2DBackpropInputAdd (residual-block pattern); numerics must match
+    the plain composition."""
+    import os
+    import numpy as np
+    tf.reset_default_graph()
+    rng = np.random.RandomState(3)
+    xv = rng.randn(2, 6, 6, 5).astype(np.float32)
+    wv = rng.randn(3, 3, 5, 7).astype(np.float32)
+    bv = rng.randn(2, 6, 6, 5).astype(np.float32)
+    x = tf.constant(xv)
+    w = tf.constant(wv)
+    b = tf.constant(bv)
+    y = tf.nn.conv2d(x, w, [1, 1, 1, 1], 'SAME')
+    loss = tf.reduce_sum(y * 2.0) + tf.reduce_sum(x * b)
+    g, = tf.gradients(loss, [x])
+    assert g.op.type == 'Conv2DBackpropInputAdd'
+    with tf.Session() as s:
+        gv = s.run(g)
+    os.environ['STF_NO_CONV_DX_FUSE'] = '1'
+    try:
+        tf.reset_default_graph()
+        x = tf.constant(xv)
+        w = tf.constant(wv)
+        b = tf.constant(bv)
+        y = tf.nn.conv2d(x, w, [1, 1, 1, 1], 'SAME')
+        loss = tf.reduce_sum(y * 2.0) + tf.reduce_sum(x * b)
+        g2, = tf.gradients(loss, [x])
+        assert g2.op.type == 'AddN'
+        with tf.Session() as s:
+            gv2 = s.run(g2)
+    finally:
+        del os.environ['STF_NO_CONV_DX_FUSE']
+    np.testing.assert_allclose(gv, gv2, rtol=1e-5, atol=1e-5)
+
+
+def test_conv_dx_add_fusion_skips_shared_grad():
+    """When the conv-backprop partial is consumed elsewhere the rewrite
+    must not fire (it would duplicate the GEMM)."""
+    import numpy as np
+    tf.reset_default_graph()
+    x = tf.constant(np.ones((1, 4, 4, 3), np.float32))
+    w = tf.constant(np.ones((1, 1, 3, 3), np.float32))
+    b = tf.constant(np.ones((1, 4, 4, 3), np.float32))
+    y = tf.nn.conv2d(x, w, [1, 1, 1, 1], 'SAME')
+    # y + y: the Add grad propagates the SAME incoming tensor to both
+    # slots; downstream aggregation sees a list with a repeated tensor
+    z = y + y
+    loss = tf.reduce_sum(z) + tf.reduce_sum(x * b)
+    g, = tf.gradients(loss, [x])
+    with tf.Session() as s:
+        gv = s.run(g)
+    np.testing.assert_allclose(gv, np.full((1, 4, 4, 3), 7.0), rtol=1e-5)

@@ -131,7 +131,35 @@ class TensorArrayWriteOp : public OpKernel {
       ta->elems.resize(idx + 1);
       ta->written.resize(idx + 1, false);
     }
-    ta->elems[idx] = val;
+    // Gradient arrays (handle "<primary>@<source>") ACCUMULATE on duplicate
+    // indices — a source slot read k times contributes k gradients
+    // (reference tensor_array.h TensorAndState multiple_writes_aggregate).
+    // Primary arrays overwrite (while-loop grad rematerialization relies
+    // on that).
+    bool is_grad =
+        ctx->input(0).flat<std::string>()[0].find('@') != std::string::npos;
+    if (is_grad && ta->written[idx] &&
+        ta->elems[idx].shape() == val.shape()) {
+      Tensor sum(val.dtype(), val.shape());
+      int64_t n = val.NumElements();
+      if (val.dtype() == DT_FLOAT) {
+        const float* a = ta->elems[idx].flat<float>();
+        const float* b = val.flat<float>();
+        float* o = sum.flat<float>();
+        for (int64_t i = 0; i < n; ++i) o[i] = a[i] + b[i];
+        ta->elems[idx] = sum;
+      } else if (val.dtype() == DT_DOUBLE) {
+        const double* a = ta->elems[idx].flat<double>();
+        const double* b = val.flat<double>();
+        double* o = sum.flat<double>();
+        for (int64_t i = 0; i < n; ++i) o[i] = a[i] + b[i];
+        ta->elems[idx] = sum;
+      } else {
+        ta->elems[idx] = val;  // non-float grads: keep last (unused path)
+      }
+    } else {
+      ta->elems[idx] = val;
+    }
     ta->written[idx] = true;
     Tensor* flow = ctx->allocate_output(0, TensorShape({}));
     flow->flat<float>()[0] = 0.f;

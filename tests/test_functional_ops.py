@@ -86,3 +86,20 @@ def test_dynamic_rnn_output_gradients():
             s.run(train)
         l1 = s.run(loss)
     assert np.isfinite(l1) and l1 < l0
+
+
+def test_tensor_array_duplicate_read_grads_accumulate():
+    """A TA slot read twice contributes both read-gradients: the shadow
+    gradient array must ACCUMULATE duplicate-index writes (reference
+    tensor_array.h multiple_writes_aggregate), not keep the last one."""
+    tf.reset_default_graph()
+    v = tf.constant(np.array([1.0, 2.0], np.float32))
+    ta = tf.TensorArray(tf.float32, size=1)
+    ta = ta.write(0, v)
+    a = ta.read(0)
+    b = ta.read(0)
+    loss = tf.reduce_sum(a * 3.0) + tf.reduce_sum(b * 5.0)
+    g, = tf.gradients(loss, [v])
+    with tf.Session() as s:
+        gv = s.run(g)
+    np.testing.assert_allclose(gv, [8.0, 8.0])

@@ -176,11 +176,13 @@ inline int PickSplitK(int64_t M, int64_t N, int64_t K) {
     want = target / (tiles ? tiles : 1);
   } else {
     want = 512 / (tiles ? tiles : 1);
-    // Very skinny outputs with huge K (stage-1 dW): deepen the split so
-    // blocks carry ~48 K-chunks each (measured +20% on 576x64x802816).
+    // Very skinny outputs with huge K (the 7x7 stem dW: 392x64x3.2M):
+    // deepen the split to ~2048 blocks (sweep: 3.14ms @1044 blocks ->
+    // 2.39ms @2048; flat beyond). Keep >=48 K-chunks per block.
     int64_t kt = K / 64;
     if (tiles <= 6 && kt / (tiles * want) > 96) {
-      want = kt / 48 / tiles;
+      want = 2048 / tiles;
+      if (want > kt / 48) want = kt / 48;
       if (tiles * want > 4096) want = 4096 / tiles;
     }
   }
